@@ -1,0 +1,7 @@
+from .engine import LLMEngine, choose_nsplit
+from .scheduler import PageAllocator, ScheduleBatch, Scheduler, SchedulerConfig
+from .sequence import SamplingParams, Sequence, SeqStatus
+
+__all__ = ["LLMEngine", "choose_nsplit", "PageAllocator", "ScheduleBatch",
+           "Scheduler", "SchedulerConfig", "SamplingParams", "Sequence",
+           "SeqStatus"]

@@ -1,0 +1,275 @@
+"""The serving engine: continuous batching over the paged-KV Llama model.
+
+One engine instance drives one GPU (DP replica) or one TP rank group.
+Decode steps are hipGraph-captured per batch-size bucket: the whole
+(embed -> L x layer -> norm -> lm_head -> sample) step replays with only
+device-buffer updates between replays (guide: capture launch-bound inner
+loops in hipGraphs).
+"""
+from __future__ import annotations
+
+import math
+import time
+
+import torch
+
+from .. import ops
+from ..models.llama import (AttnMetadata, KVCache, LlamaConfig,
+                            LlamaForCausalLM)
+from .scheduler import ScheduleBatch, Scheduler, SchedulerConfig
+from .sequence import SamplingParams, Sequence, SeqStatus
+
+DECODE_BUCKETS = (1, 2, 4, 8, 16, 32, 64, 128, 256)
+
+
+def _bucket_for(n: int, max_bs: int) -> int:
+    for b in DECODE_BUCKETS:
+        if n <= b:
+            return min(b, max_bs)
+    return max_bs
+
+
+def choose_nsplit(bs: int, hk: int) -> int:
+    """Decode attention split-K factor: fill >=512 workgroups on 256 CUs."""
+    target = 512
+    ns = max(1, min(16, target // max(1, bs * hk)))
+    return 1 << (ns.bit_length() - 1)
+
+
+class LLMEngine:
+    def __init__(self, cfg: LlamaConfig, device="cuda",
+                 dtype=torch.bfloat16, page_size: int = 16,
+                 kv_memory_frac: float = 0.80, num_pages: int | None = None,
+                 max_num_seqs: int = 64, max_prefill_tokens: int = 8192,
+                 enable_graphs: bool = True, eos_id: int = 2, seed: int = 0,
+                 model: LlamaForCausalLM | None = None,
+                 tp_group=None):
+        self.cfg = cfg
+        self.device = torch.device(device)
+        self.is_gpu = self.device.type == "cuda"
+        self.tp_group = tp_group
+        if model is None:
+            model = LlamaForCausalLM(cfg, device=device, dtype=dtype).init_random(seed)
+        self.model = model.eval()
+
+        if num_pages is None:
+            if self.is_gpu:
+                free, _total = torch.cuda.mem_get_info(self.device)
+                budget = int(free * kv_memory_frac)
+            else:
+                budget = 64 << 20
+            num_pages = max(16, budget // KVCache.bytes_per_page(cfg, page_size))
+        max_pages_per_seq = (cfg.max_position + page_size - 1) // page_size
+        self.max_pages_per_seq = max_pages_per_seq
+        self.kv = KVCache(cfg, num_pages, page_size, self.device, dtype)
+        self.sched = Scheduler(SchedulerConfig(
+            max_num_seqs=max_num_seqs, max_prefill_tokens=max_prefill_tokens,
+            page_size=page_size, num_pages=num_pages))
+        self.page_size = page_size
+        self.eos_id = eos_id
+        self.max_num_seqs = max_num_seqs
+        self._next_id = 0
+        self._finished: dict[int, Sequence] = {}
+        self.sampler = ops.SamplerState(max_num_seqs, self.device, seed=seed or 0x5EED)
+        self.enable_graphs = enable_graphs and self.is_gpu
+        self._graphs: dict[int, dict] = {}
+        self.metrics = {"prefill_tokens": 0, "decode_tokens": 0, "steps": 0,
+                        "prefill_steps": 0, "decode_steps": 0}
+        # static decode buffers (shared across graph buckets; sized to max)
+        B = max_num_seqs
+        dev = self.device
+        self._dec = {
+            "ids": torch.zeros(B, dtype=torch.int32, device=dev),
+            "pos": torch.zeros(B, dtype=torch.int32, device=dev),
+            "slots": torch.zeros(B, dtype=torch.int64, device=dev),
+            "bt": torch.zeros(B, max_pages_per_seq, dtype=torch.int32, device=dev),
+            "lens": torch.ones(B, dtype=torch.int32, device=dev),
+            "temps": torch.zeros(B, dtype=torch.float32, device=dev),
+            "tokens": torch.zeros(B, dtype=torch.int32, device=dev),
+        }
+        self._host = {k: torch.zeros_like(v, device="cpu").pin_memory()
+                      if self.is_gpu else torch.zeros_like(v)
+                      for k, v in self._dec.items()}
+
+    # ------------------------------------------------------------- requests
+    def add_request(self, prompt_ids: list[int], sampling: SamplingParams | None = None,
+                    on_token=None) -> int | None:
+        sampling = sampling or SamplingParams()
+        seq = Sequence(self._next_id, list(prompt_ids), sampling,
+                       on_token=on_token, arrival_ns=time.monotonic_ns())
+        if len(seq.prompt_ids) + sampling.max_tokens > self.cfg.max_position:
+            raise ValueError("request exceeds model max_position")
+        if not self.sched.add(seq):
+            return None  # backpressure
+        self._next_id += 1
+        return seq.seq_id
+
+    def get_finished(self, seq_id: int) -> Sequence | None:
+        return self._finished.pop(seq_id, None)
+
+    def has_work(self) -> bool:
+        return self.sched.has_work()
+
+    # ------------------------------------------------------------- stepping
+    @torch.inference_mode()
+    def step(self) -> list[tuple[int, int, bool]]:
+        """Run one engine iteration.  Returns [(seq_id, token, done)]."""
+        batch = self.sched.schedule()
+        if batch is None:
+            return []
+        self.metrics["steps"] += 1
+        if batch.is_prefill:
+            self.metrics["prefill_steps"] += 1
+            tokens = self._step_prefill(batch)
+        else:
+            self.metrics["decode_steps"] += 1
+            tokens = self._step_decode(batch)
+        # host-side bookkeeping
+        events = []
+        now = time.monotonic_ns()
+        for seq, tok in zip(batch.seqs, tokens):
+            if not seq.output_ids:
+                seq.first_token_ns = now
+            done = seq.append(tok, self.eos_id)
+            if done:
+                seq.finish_ns = now
+                self.sched.finish(seq)
+                self._finished[seq.seq_id] = seq
+            if seq.on_token is not None:
+                seq.on_token(tok, done)
+            events.append((seq.seq_id, tok, done))
+        return events
+
+    def _slot(self, seq: Sequence, tok_idx: int) -> int:
+        return seq.pages[tok_idx // self.page_size] * self.page_size + \
+            tok_idx % self.page_size
+
+    def _step_prefill(self, batch: ScheduleBatch) -> list[int]:
+        dev = self.device
+        seqs = batch.seqs
+        ids, pos, slots, lens = [], [], [], []
+        for seq in seqs:
+            n = len(seq.prompt_ids)
+            ids.extend(seq.prompt_ids)
+            pos.extend(range(n))
+            slots.extend(self._slot(seq, i) for i in range(n))
+            lens.append(n)
+        self.metrics["prefill_tokens"] += len(ids)
+        cu_list = [0]
+        for ln in lens:
+            cu_list.append(cu_list[-1] + ln)
+        cu = torch.tensor(cu_list, dtype=torch.int32, device=dev)
+        ids_t = torch.tensor(ids, dtype=torch.int32, device=dev)
+        pos_t = torch.tensor(pos, dtype=torch.int32, device=dev)
+        slots_t = torch.tensor(slots, dtype=torch.int64, device=dev)
+        md = AttnMetadata(is_prefill=True, slots=slots_t, cu_seqlens=cu,
+                          seq_lens=lens)
+        last_rows = torch.tensor(
+            [int(cu[i + 1]) - 1 for i in range(len(seqs))],
+            dtype=torch.int32, device=dev)
+        logits = self.model(ids_t, pos_t, self.kv, md, logit_rows=last_rows)
+        temps = torch.tensor([s.sampling.temperature for s in seqs],
+                             dtype=torch.float32, device=dev)
+        toks = ops.sample(logits, temps, self.sampler)
+        return toks.cpu().tolist()
+
+    # -- decode path (graph-captured on GPU) --------------------------------
+    def _fill_decode_buffers(self, seqs: list[Sequence], bs: int) -> None:
+        h = self._host
+        for i, seq in enumerate(seqs):
+            n = seq.num_tokens
+            h["ids"][i] = seq.last_token
+            h["pos"][i] = n - 1
+            h["slots"][i] = self._slot(seq, n - 1)
+            h["lens"][i] = n
+            npg = len(seq.pages)
+            h["bt"][i, :npg] = torch.tensor(seq.pages, dtype=torch.int32)
+            h["temps"][i] = seq.sampling.temperature
+        for i in range(len(seqs), bs):  # dummy lanes -> null page 0
+            h["ids"][i] = 0
+            h["pos"][i] = 0
+            h["slots"][i] = 0
+            h["lens"][i] = 1
+            h["bt"][i, 0] = 0
+            h["temps"][i] = 0.0
+        d = self._dec
+        nb = self.is_gpu
+        for k in ("ids", "pos", "slots", "lens", "temps"):
+            d[k][:bs].copy_(h[k][:bs], non_blocking=nb)
+        d["bt"][:bs].copy_(h["bt"][:bs], non_blocking=nb)
+
+    def _decode_forward(self, bs: int, nsplit: int, scratch):
+        d = self._dec
+        md = AttnMetadata(is_prefill=False, slots=d["slots"][:bs],
+                          block_table=d["bt"][:bs], seq_lens_t=d["lens"][:bs],
+                          nsplit=nsplit, decode_scratch=scratch)
+        logits = self.model(d["ids"][:bs], d["pos"][:bs], self.kv, md)
+        ops.sample(logits, d["temps"][:bs], self.sampler, out=d["tokens"][:bs])
+
+    def _make_scratch(self, bs: int, nsplit: int):
+        if nsplit <= 1:
+            return None
+        cfg = self.cfg
+        G = cfg.num_heads // cfg.num_kv_heads
+        po = torch.empty(bs, cfg.num_kv_heads, nsplit, G, cfg.head_dim,
+                         dtype=torch.float32, device=self.device)
+        pml = torch.empty(bs, cfg.num_kv_heads, nsplit, G, 2,
+                          dtype=torch.float32, device=self.device)
+        return po, pml
+
+    def _get_graph(self, bs: int):
+        entry = self._graphs.get(bs)
+        if entry is not None:
+            return entry
+        nsplit = choose_nsplit(bs, self.cfg.num_kv_heads)
+        scratch = self._make_scratch(bs, nsplit)
+        # warm up allocator/kernels on a side stream, then capture
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self._decode_forward(bs, nsplit, scratch)
+        torch.cuda.current_stream().wait_stream(s)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._decode_forward(bs, nsplit, scratch)
+        entry = {"graph": g, "nsplit": nsplit, "scratch": scratch}
+        self._graphs[bs] = entry
+        return entry
+
+    def _step_decode(self, batch: ScheduleBatch) -> list[int]:
+        seqs = batch.seqs
+        n = len(seqs)
+        self.metrics["decode_tokens"] += n
+        bs = _bucket_for(n, self.max_num_seqs)
+        self._fill_decode_buffers(seqs, bs)
+        if self.enable_graphs:
+            self._get_graph(bs)["graph"].replay()
+        else:
+            nsplit = choose_nsplit(bs, self.cfg.num_kv_heads) if self.is_gpu else 1
+            self._decode_forward(bs, nsplit, self._make_scratch(bs, nsplit))
+        toks = self._dec["tokens"][:n]
+        if self.is_gpu:
+            self._host["tokens"][:n].copy_(toks, non_blocking=True)
+            torch.cuda.current_stream().synchronize()
+            return self._host["tokens"][:n].tolist()
+        return toks.tolist()
+
+    # ------------------------------------------------------------- generate
+    def generate(self, prompts: list[list[int]],
+                 sampling: SamplingParams | None = None) -> list[list[int]]:
+        """Synchronous batch generate (used by tests and app.ai())."""
+        idmap = {}
+        for p in prompts:
+            rid = self.add_request(p, sampling)
+            if rid is None:
+                raise RuntimeError("engine queue full")
+            idmap[rid] = None
+        while self.has_work() and any(v is None for v in idmap.values()):
+            self.step()
+            for rid in list(idmap):
+                if idmap[rid] is None:
+                    fin = self.get_finished(rid)
+                    if fin is not None:
+                        idmap[rid] = fin.output_ids
+        return [idmap[rid] for rid in sorted(idmap)]

@@ -1,0 +1,157 @@
+"""Continuous-batching scheduler + KV page allocator (Python reference).
+
+The native C++ implementation (agentfield_amd/native/scheduler.cpp, exposed
+via agentfield_amd._native) is the production path; this module defines the
+semantics and is the oracle for its tests.  The engine picks _native when the
+extension is built.
+
+Admission control is the GPU-side face of the control plane's bounded async
+queue (reference: execute.go async worker pool): a bounded
+waiting queue, a KV-page budget sized for 288 GB HBM3E, and preemption of the
+youngest running sequence when decode runs out of pages.
+"""
+from __future__ import annotations
+
+from collections import deque
+from dataclasses import dataclass, field
+
+from .sequence import Sequence, SeqStatus
+
+
+class PageAllocator:
+    """Free-list page allocator.  Page 0 is reserved as the null page used by
+    padded (dummy) decode lanes in hipGraph buckets."""
+
+    def __init__(self, num_pages: int):
+        self.num_pages = num_pages
+        self.free_list = list(range(num_pages - 1, 0, -1))
+
+    @property
+    def num_free(self) -> int:
+        return len(self.free_list)
+
+    def alloc(self, n: int) -> list[int]:
+        if n > len(self.free_list):
+            raise MemoryError(f"KV allocator: need {n} pages, {len(self.free_list)} free")
+        out = [self.free_list.pop() for _ in range(n)]
+        return out
+
+    def free(self, pages: list[int]) -> None:
+        self.free_list.extend(pages)
+
+
+@dataclass
+class SchedulerConfig:
+    max_num_seqs: int = 64
+    max_prefill_tokens: int = 8192
+    page_size: int = 16
+    num_pages: int = 1024
+    max_waiting: int = 4096
+
+
+@dataclass
+class ScheduleBatch:
+    is_prefill: bool
+    seqs: list[Sequence] = field(default_factory=list)
+
+
+class Scheduler:
+    def __init__(self, cfg: SchedulerConfig):
+        self.cfg = cfg
+        self.alloc = PageAllocator(cfg.num_pages)
+        self.waiting: deque[Sequence] = deque()
+        self.running: list[Sequence] = []
+        self.n_preempted = 0
+
+    # -- queue interface -------------------------------------------------
+    def add(self, seq: Sequence) -> bool:
+        """Returns False on backpressure (queue full -> control plane 503s)."""
+        if len(self.waiting) >= self.cfg.max_waiting:
+            return False
+        self.waiting.append(seq)
+        return True
+
+    def has_work(self) -> bool:
+        return bool(self.waiting or self.running)
+
+    def num_queued(self) -> int:
+        return len(self.waiting)
+
+    def num_running(self) -> int:
+        return len(self.running)
+
+    # -- page bookkeeping ------------------------------------------------
+    def _pages_needed(self, ntokens: int) -> int:
+        return (ntokens + self.cfg.page_size - 1) // self.cfg.page_size
+
+    def _grow(self, seq: Sequence) -> bool:
+        """Ensure capacity for one more token; returns False if OOM."""
+        need = self._pages_needed(seq.num_tokens + 1)
+        if need > len(seq.pages):
+            if self.alloc.num_free < 1:
+                return False
+            seq.pages.extend(self.alloc.alloc(1))
+        return True
+
+    def release(self, seq: Sequence) -> None:
+        self.alloc.free(seq.pages)
+        seq.pages = []
+
+    def _preempt_last(self) -> None:
+        victim = self.running.pop()
+        self.release(victim)
+        victim.output_ids.clear()       # recompute from scratch on readmit
+        victim.status = SeqStatus.WAITING
+        self.waiting.appendleft(victim)
+        self.n_preempted += 1
+
+    # -- the scheduling step ---------------------------------------------
+    def schedule(self) -> ScheduleBatch | None:
+        cfg = self.cfg
+        # 1) admit prefills while budget allows
+        batch: list[Sequence] = []
+        tokens = 0
+        while (self.waiting and len(self.running) + len(batch) < cfg.max_num_seqs):
+            cand = self.waiting[0]
+            ntok = len(cand.prompt_ids)
+            if batch and tokens + ntok > cfg.max_prefill_tokens:
+                break
+            need = self._pages_needed(ntok)
+            if need > self.alloc.num_free:
+                break
+            self.waiting.popleft()
+            cand.pages = self.alloc.alloc(need)
+            cand.status = SeqStatus.RUNNING
+            batch.append(cand)
+            tokens += ntok
+        if batch:
+            self.running.extend(batch)
+            return ScheduleBatch(is_prefill=True, seqs=batch)
+
+        # 2) otherwise decode everything running (grow pages, preempt on OOM)
+        if not self.running:
+            return None
+        i = 0
+        while i < len(self.running):
+            seq = self.running[i]
+            while not self._grow(seq):
+                if self.running[-1] is seq:
+                    # can't preempt self and nothing else to free: defer
+                    self.running.pop()
+                    self.release(seq)
+                    seq.output_ids.clear()
+                    seq.status = SeqStatus.WAITING
+                    self.waiting.appendleft(seq)
+                    self.n_preempted += 1
+                    i -= 1
+                    break
+                self._preempt_last()
+            i += 1
+        if not self.running:
+            return None
+        return ScheduleBatch(is_prefill=False, seqs=list(self.running))
+
+    def finish(self, seq: Sequence) -> None:
+        seq.status = SeqStatus.FINISHED
+        self.running.remove(seq)
+        self.release(seq)

@@ -1,0 +1,56 @@
+"""Request/sequence state for the serving engine."""
+from __future__ import annotations
+
+import enum
+from dataclasses import dataclass, field
+from typing import Callable
+
+
+@dataclass
+class SamplingParams:
+    max_tokens: int = 128
+    temperature: float = 0.0
+    stop_token_ids: tuple = ()
+    ignore_eos: bool = False
+
+
+class SeqStatus(enum.Enum):
+    WAITING = 0
+    RUNNING = 1
+    FINISHED = 2
+    PREEMPTED = 3
+
+
+@dataclass
+class Sequence:
+    seq_id: int
+    prompt_ids: list[int]
+    sampling: SamplingParams
+    status: SeqStatus = SeqStatus.WAITING
+    output_ids: list[int] = field(default_factory=list)
+    pages: list[int] = field(default_factory=list)
+    finish_reason: str | None = None
+    on_token: Callable | None = None     # streaming callback (token_id, done)
+    arrival_ns: int = 0
+    first_token_ns: int = 0
+    finish_ns: int = 0
+
+    @property
+    def num_tokens(self) -> int:
+        return len(self.prompt_ids) + len(self.output_ids)
+
+    @property
+    def last_token(self) -> int:
+        return self.output_ids[-1] if self.output_ids else self.prompt_ids[-1]
+
+    def append(self, tok: int, eos_id: int) -> bool:
+        """Append a generated token; returns True when the sequence finished."""
+        self.output_ids.append(tok)
+        sp = self.sampling
+        if not sp.ignore_eos and (tok == eos_id or tok in sp.stop_token_ids):
+            self.finish_reason = "stop"
+            return True
+        if len(self.output_ids) >= sp.max_tokens:
+            self.finish_reason = "length"
+            return True
+        return False

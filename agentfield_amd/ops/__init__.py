@@ -1,0 +1,237 @@
+"""MI355X-native ops: thin Python wrappers over the HIP kernel library.
+
+Every op has two paths:
+  * CUDA (= ROCm/HIP) tensors -> the hand-written gfx950 kernel, always.
+    There is no eager fallback on GPU; a missing/failed kernel raises.
+  * CPU tensors -> a plain fp32 torch reference (used by CPU-only tests and
+    as the numerics oracle for the GPU kernels).
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+
+from . import _lib
+from ._lib import AfOpsError, native_loaded
+from . import reference as ref
+
+
+def _on_gpu(*ts: torch.Tensor | None) -> bool:
+    return any(t is not None and t.is_cuda for t in ts)
+
+
+# ---------------------------------------------------------------- rmsnorm
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5,
+            residual: torch.Tensor | None = None):
+    """Fused (residual-add) RMSNorm.
+
+    Returns (out, new_residual) when residual is given (residual updated
+    in-place to x+residual), else out.
+    """
+    if not _on_gpu(x):
+        return ref.rmsnorm(x, weight, eps, residual)
+    T, H = x.shape
+    out = torch.empty_like(x)
+    rc = _lib.lib().af_rmsnorm(
+        _lib.ptr(out), _lib.ptr(residual), _lib.ptr(x), _lib.ptr(residual),
+        _lib.ptr(weight), eps, T, H, _lib.cur_stream())
+    _lib.check(rc, "af_rmsnorm")
+    return (out, residual) if residual is not None else out
+
+
+# ---------------------------------------------------------------- rope
+def rope_table(max_pos: int, head_dim: int, theta: float = 500000.0,
+               device="cpu") -> torch.Tensor:
+    """Host-precomputed fp32 [max_pos, head_dim] table: [cos | sin] halves."""
+    half = head_dim // 2
+    inv = 1.0 / (theta ** (torch.arange(0, half, dtype=torch.float64) / half))
+    t = torch.arange(max_pos, dtype=torch.float64)
+    freqs = torch.outer(t, inv)
+    return torch.cat([freqs.cos(), freqs.sin()], dim=-1).float().to(device)
+
+
+def rope(q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor,
+         table: torch.Tensor) -> None:
+    """In-place rotate-half RoPE on q [T,Hq,D] and k [T,Hk,D]."""
+    if not _on_gpu(q):
+        ref.rope(q, k, positions, table)
+        return
+    T, Hq, D = q.shape
+    Hk = k.shape[1]
+    rc = _lib.lib().af_rope(
+        _lib.ptr(q), _lib.ptr(k), _lib.ptr(positions.int()), _lib.ptr(table),
+        T, Hq, Hk, D, _lib.cur_stream())
+    _lib.check(rc, "af_rope")
+
+
+# ---------------------------------------------------------------- activation
+def silu_and_mul(gate_up: torch.Tensor) -> torch.Tensor:
+    """gate_up [T, 2I] -> silu(gate)*up [T, I]."""
+    if not _on_gpu(gate_up):
+        return ref.silu_and_mul(gate_up)
+    T, I2 = gate_up.shape
+    out = torch.empty(T, I2 // 2, dtype=gate_up.dtype, device=gate_up.device)
+    rc = _lib.lib().af_silu_mul(_lib.ptr(out), _lib.ptr(gate_up), T, I2 // 2,
+                                _lib.cur_stream())
+    _lib.check(rc, "af_silu_mul")
+    return out
+
+
+# ---------------------------------------------------------------- cache
+def reshape_and_cache(k: torch.Tensor, v: torch.Tensor, kcache: torch.Tensor,
+                      vcache: torch.Tensor, slots: torch.Tensor) -> None:
+    """Scatter k/v [T,Hk,D] into paged caches [npages,Hk,page,D] at slots [T]."""
+    if not _on_gpu(k):
+        return ref.reshape_and_cache(k, v, kcache, vcache, slots)
+    T = k.shape[0]
+    _, Hk, page, D = kcache.shape
+    rc = _lib.lib().af_reshape_and_cache(
+        _lib.ptr(k), _lib.ptr(v), _lib.ptr(kcache), _lib.ptr(vcache),
+        _lib.ptr(slots), T, Hk, D, page, _lib.cur_stream())
+    _lib.check(rc, "af_reshape_and_cache")
+
+
+def embedding(ids: torch.Tensor, table: torch.Tensor) -> torch.Tensor:
+    if not _on_gpu(table):
+        return table[ids.long()]
+    T = ids.numel()
+    V, H = table.shape
+    out = torch.empty(T, H, dtype=table.dtype, device=table.device)
+    rc = _lib.lib().af_embedding(_lib.ptr(out), _lib.ptr(table),
+                                 _lib.ptr(ids.int()), T, H, _lib.cur_stream())
+    _lib.check(rc, "af_embedding")
+    return out
+
+
+# ---------------------------------------------------------------- attention
+def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
+                block_table: torch.Tensor, seq_lens: torch.Tensor,
+                scale: float | None = None, nsplit: int = 1,
+                scratch: tuple[torch.Tensor, torch.Tensor] | None = None,
+                out: torch.Tensor | None = None) -> torch.Tensor:
+    """Paged decode attention. q [B,Hq,D] -> out [B,Hq,D]."""
+    B, Hq, D = q.shape
+    _, Hk, page, _ = kcache.shape
+    if scale is None:
+        scale = 1.0 / math.sqrt(D)
+    if not _on_gpu(q):
+        return ref.attn_decode(q, kcache, vcache, block_table, seq_lens, scale)
+    if out is None:
+        out = torch.empty_like(q)
+    G = Hq // Hk
+    if nsplit > 1:
+        if scratch is None:
+            po = torch.empty(B, Hk, nsplit, G, D, dtype=torch.float32, device=q.device)
+            pml = torch.empty(B, Hk, nsplit, G, 2, dtype=torch.float32, device=q.device)
+        else:
+            po, pml = scratch
+    else:
+        po = pml = None
+    rc = _lib.lib().af_attn_decode(
+        _lib.ptr(out), _lib.ptr(po), _lib.ptr(pml), _lib.ptr(q),
+        _lib.ptr(kcache), _lib.ptr(vcache), _lib.ptr(block_table),
+        _lib.ptr(seq_lens), scale, B, Hq, Hk, D, page, block_table.shape[1],
+        nsplit, _lib.cur_stream())
+    _lib.check(rc, "af_attn_decode")
+    return out
+
+
+def prefill_tiles(seq_lens: list[int], rows_per_wg: int):
+    """Host-side tile map for varlen prefill: (tile_seq[], tile_q0[])."""
+    ts, tq = [], []
+    for s, ln in enumerate(seq_lens):
+        for q0 in range(0, ln, rows_per_wg):
+            ts.append(s)
+            tq.append(q0)
+    return ts, tq
+
+
+def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                 cu_seqlens: torch.Tensor, seq_lens: list[int],
+                 scale: float | None = None) -> torch.Tensor:
+    """Causal varlen prefill attention.  q [T,Hq,D], k/v [T,Hk,D]."""
+    T, Hq, D = q.shape
+    Hk = k.shape[1]
+    if scale is None:
+        scale = 1.0 / math.sqrt(D)
+    if not _on_gpu(q):
+        return ref.attn_prefill(q, k, v, cu_seqlens, scale)
+    G = Hq // Hk
+    rows_per_wg = 16 * max(1, 4 // G)
+    ts, tq = prefill_tiles(seq_lens, rows_per_wg)
+    dev = q.device
+    tile_seq = torch.tensor(ts, dtype=torch.int32, device=dev)
+    tile_q0 = torch.tensor(tq, dtype=torch.int32, device=dev)
+    out = torch.empty_like(q)
+    rc = _lib.lib().af_attn_prefill(
+        _lib.ptr(out), _lib.ptr(q), _lib.ptr(k), _lib.ptr(v),
+        _lib.ptr(cu_seqlens), _lib.ptr(tile_seq), _lib.ptr(tile_q0),
+        scale, len(ts), Hq, Hk, D, _lib.cur_stream())
+    _lib.check(rc, "af_attn_prefill")
+    return out
+
+
+# ---------------------------------------------------------------- gemm
+def gemm_bf16(a: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """C[M,N] = a[M,K] @ w[N,K]^T via the hand-written MFMA kernel."""
+    if not _on_gpu(a):
+        return (a.float() @ w.float().t()).to(a.dtype)
+    M, K = a.shape
+    N = w.shape[0]
+    c = torch.empty(M, N, dtype=a.dtype, device=a.device)
+    rc = _lib.lib().af_gemm_bf16(_lib.ptr(c), _lib.ptr(a), _lib.ptr(w),
+                                 M, N, K, _lib.cur_stream())
+    _lib.check(rc, "af_gemm_bf16")
+    return c
+
+
+# ---------------------------------------------------------------- sampling
+class SamplerState:
+    """Device-side scratch for graph-capturable sampling."""
+
+    def __init__(self, max_b: int, device, seed: int = 0x5EED):
+        self.pv = torch.empty(max_b, 16, dtype=torch.float32, device=device)
+        self.pi = torch.empty(max_b, 16, dtype=torch.int32, device=device)
+        self.step = torch.zeros(1, dtype=torch.int32, device=device)
+        self.seed = seed
+
+
+def sample(logits: torch.Tensor, temps: torch.Tensor, state: SamplerState,
+           out: torch.Tensor | None = None) -> torch.Tensor:
+    """Greedy (temp==0) or Gumbel-max temperature sampling.  logits [B,V] bf16."""
+    B, V = logits.shape
+    if not _on_gpu(logits):
+        res = ref.sample_greedy(logits)
+        if out is not None:
+            out[:B].copy_(res)
+            return out
+        return res
+    if out is None:
+        out = torch.empty(B, dtype=torch.int32, device=logits.device)
+    rc = _lib.lib().af_sample(
+        _lib.ptr(out), _lib.ptr(state.pv), _lib.ptr(state.pi), _lib.ptr(logits),
+        _lib.ptr(temps), _lib.ptr(state.step), state.seed, B, V, _lib.cur_stream())
+    _lib.check(rc, "af_sample")
+    return out
+
+
+def gather_rows(x: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
+    if not _on_gpu(x):
+        return x[idx.long()]
+    B = idx.numel()
+    H = x.shape[1]
+    out = torch.empty(B, H, dtype=x.dtype, device=x.device)
+    rc = _lib.lib().af_gather_rows(_lib.ptr(out), _lib.ptr(x), _lib.ptr(idx.int()),
+                                   B, H, _lib.cur_stream())
+    _lib.check(rc, "af_gather_rows")
+    return out
+
+
+def mfma_probe(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """Single-wave 16x16x32 MFMA with the assumed layouts (GPU layout test)."""
+    d = torch.empty(16, 16, dtype=torch.float32, device=a.device)
+    rc = _lib.lib().af_mfma_probe(_lib.ptr(d), _lib.ptr(a), _lib.ptr(b),
+                                  _lib.cur_stream())
+    _lib.check(rc, "af_mfma_probe")
+    return d

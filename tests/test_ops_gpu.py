@@ -1,0 +1,230 @@
+"""GPU numerics tests: every HIP kernel vs the plain-torch fp32 reference.
+
+All inputs are asymmetric random (guide G9: symmetric inputs can't detect
+operand/output transposes).
+"""
+import math
+
+import pytest
+import torch
+
+from agentfield_amd import ops
+from agentfield_amd.ops import reference as ref
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda"
+
+
+def rnd(*shape, dtype=torch.bfloat16, seed=None, scale=1.0):
+    if seed is not None:
+        torch.manual_seed(seed)
+    return (torch.randn(*shape, dtype=torch.float32, device=DEV) * scale).to(dtype)
+
+
+def close(a, b, atol=2e-2, rtol=2e-2):
+    a, b = a.float().cpu(), b.float().cpu()
+    ok = torch.allclose(a, b, atol=atol, rtol=rtol)
+    if not ok:
+        d = (a - b).abs()
+        print("max abs diff", d.max().item(), "mean", d.mean().item())
+    return ok
+
+
+def test_native_lib_loads():
+    assert ops.native_loaded(), "libafops.so must load on a GPU box"
+
+
+def test_axpy_plumbing():
+    from agentfield_amd.ops import _lib
+    y = torch.ones(100, dtype=torch.float32, device=DEV)
+    x = torch.full((100,), 2.0, device=DEV)
+    rc = _lib.lib().af_axpy(_lib.ptr(y), _lib.ptr(x), 3.0, 100, _lib.cur_stream())
+    _lib.check(rc, "af_axpy")
+    torch.cuda.synchronize()
+    assert torch.allclose(y.cpu(), torch.full((100,), 7.0))
+
+
+def test_mfma_probe_layouts():
+    """Assumed A/B/C fragment layouts for mfma_f32_16x16x32_bf16."""
+    a = rnd(16, 32, seed=1)
+    b = rnd(32, 16, seed=2)
+    d = ops.mfma_probe(a, b)
+    torch.cuda.synchronize()
+    want = a.float() @ b.float()
+    assert close(d, want, atol=1e-1), "MFMA fragment layout mismatch"
+
+
+def test_rmsnorm():
+    x = rnd(33, 4096, seed=3)
+    w = rnd(4096, seed=4)
+    out = ops.rmsnorm(x, w, 1e-5)
+    torch.cuda.synchronize()
+    assert close(out, ref.rmsnorm(x.cpu(), w.cpu(), 1e-5))
+
+
+def test_rmsnorm_fused_residual():
+    x = rnd(17, 512, seed=5)
+    w = rnd(512, seed=6)
+    res = rnd(17, 512, seed=7)
+    res_ref = res.cpu().clone()
+    out, new_res = ops.rmsnorm(x, w, 1e-5, residual=res)
+    torch.cuda.synchronize()
+    out_ref, res_ref = ref.rmsnorm(x.cpu(), w.cpu(), 1e-5, residual=res_ref)
+    assert close(out, out_ref)
+    assert close(new_res, res_ref)
+
+
+def test_rope():
+    T, Hq, Hk, D = 9, 4, 2, 128
+    q = rnd(T, Hq, D, seed=8)
+    k = rnd(T, Hk, D, seed=9)
+    qc, kc = q.cpu().clone(), k.cpu().clone()
+    pos = torch.randint(0, 100, (T,), dtype=torch.int32, device=DEV)
+    tab = ops.rope_table(128, D, device=DEV)
+    ops.rope(q, k, pos, tab)
+    torch.cuda.synchronize()
+    ref.rope(qc, kc, pos.cpu(), tab.cpu())
+    assert close(q, qc) and close(k, kc)
+
+
+def test_silu_mul():
+    x = rnd(31, 1024, seed=10)
+    out = ops.silu_and_mul(x)
+    torch.cuda.synchronize()
+    assert close(out, ref.silu_and_mul(x.cpu()))
+
+
+def test_reshape_and_cache():
+    T, Hk, D, page, npages = 11, 2, 128, 4, 8
+    k, v = rnd(T, Hk, D, seed=11), rnd(T, Hk, D, seed=12)
+    kc = torch.zeros(npages, Hk, page, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    slots = torch.randperm(npages * page, device=DEV)[:T].to(torch.int64)
+    ops.reshape_and_cache(k, v, kc, vc, slots)
+    torch.cuda.synchronize()
+    kc_ref = torch.zeros_like(kc, device="cpu")
+    vc_ref = torch.zeros_like(vc, device="cpu")
+    ref.reshape_and_cache(k.cpu(), v.cpu(), kc_ref, vc_ref, slots.cpu())
+    assert close(kc, kc_ref, atol=0) and close(vc, vc_ref, atol=0)
+
+
+def _decode_case(B, Hq, Hk, lens, page=16, nsplit=1, seed=13):
+    D = 128
+    torch.manual_seed(seed)
+    maxp = (max(lens) + page - 1) // page
+    npages = 1 + sum((l + page - 1) // page for l in lens)
+    q = rnd(B, Hq, D, seed=seed)
+    kc = rnd(npages, Hk, page, D, seed=seed + 1)
+    vc = rnd(npages, Hk, page, D, seed=seed + 2)
+    perm = torch.randperm(npages - 1) + 1
+    bt = torch.zeros(B, maxp, dtype=torch.int32)
+    at = 0
+    for b, l in enumerate(lens):
+        n = (l + page - 1) // page
+        bt[b, :n] = perm[at:at + n]
+        at += n
+    bt = bt.to(DEV)
+    lens_t = torch.tensor(lens, dtype=torch.int32, device=DEV)
+    out = ops.attn_decode(q, kc, vc, bt, lens_t, nsplit=nsplit)
+    torch.cuda.synchronize()
+    want = ref.attn_decode(q.cpu(), kc.cpu(), vc.cpu(), bt.cpu(),
+                           lens_t.cpu(), 1.0 / math.sqrt(D))
+    assert close(out, want), f"decode mismatch B={B} G={Hq//Hk} nsplit={nsplit}"
+
+
+def test_attn_decode_g4():
+    _decode_case(3, 8, 2, [5, 33, 17])
+
+
+def test_attn_decode_g1_g8():
+    _decode_case(2, 4, 4, [20, 64], seed=20)
+    _decode_case(2, 8, 1, [31, 7], seed=21)
+
+
+def test_attn_decode_split():
+    _decode_case(2, 8, 2, [100, 230], nsplit=4, seed=22)
+    _decode_case(1, 4, 1, [511], nsplit=8, seed=23)
+
+
+def _prefill_case(Hq, Hk, lens, seed=30):
+    D = 128
+    T = sum(lens)
+    q = rnd(T, Hq, D, seed=seed)
+    k = rnd(T, Hk, D, seed=seed + 1)
+    v = rnd(T, Hk, D, seed=seed + 2)
+    cu = [0]
+    for l in lens:
+        cu.append(cu[-1] + l)
+    cu_t = torch.tensor(cu, dtype=torch.int32, device=DEV)
+    out = ops.attn_prefill(q, k, v, cu_t, lens)
+    torch.cuda.synchronize()
+    want = ref.attn_prefill(q.cpu(), k.cpu(), v.cpu(), cu_t.cpu(),
+                            1.0 / math.sqrt(D))
+    assert close(out, want), f"prefill mismatch G={Hq//Hk} lens={lens}"
+
+
+def test_attn_prefill_g4():
+    _prefill_case(8, 2, [48, 17, 100])
+
+
+def test_attn_prefill_g1():
+    _prefill_case(2, 2, [33, 64], seed=40)
+
+
+def test_attn_prefill_g2_g8():
+    _prefill_case(4, 2, [40, 23], seed=50)
+    _prefill_case(8, 1, [129], seed=60)
+
+
+def test_attn_prefill_long():
+    _prefill_case(8, 2, [1024], seed=70)
+
+
+def test_gemm():
+    for (M, N, K) in [(128, 128, 64), (256, 512, 256), (100, 300, 128),
+                      (512, 4096, 4096)]:
+        a = rnd(M, K, seed=M + N, scale=0.5)
+        w = rnd(N, K, seed=M + N + 1, scale=0.5)
+        c = ops.gemm_bf16(a, w)
+        torch.cuda.synchronize()
+        want = (a.float() @ w.float().t())
+        # bf16 accumulation tolerance scales with K
+        tol = 0.1 + 0.02 * math.sqrt(K)
+        assert close(c, want, atol=tol, rtol=5e-2), f"gemm {M}x{N}x{K}"
+
+
+def test_sample_greedy_matches_argmax():
+    torch.manual_seed(99)
+    B, V = 5, 128256
+    logits = rnd(B, V, seed=99)
+    st = ops.SamplerState(B, DEV)
+    temps = torch.zeros(B, dtype=torch.float32, device=DEV)
+    toks = ops.sample(logits, temps, st)
+    torch.cuda.synchronize()
+    assert toks.cpu().tolist() == logits.float().argmax(-1).cpu().tolist()
+
+
+def test_sample_temperature_varies_with_step():
+    B, V = 2, 1000
+    logits = torch.zeros(B, V, dtype=torch.bfloat16, device=DEV)
+    st = ops.SamplerState(B, DEV)
+    temps = torch.full((B,), 1.0, dtype=torch.float32, device=DEV)
+    draws = set()
+    for _ in range(8):
+        t = ops.sample(logits, temps, st)
+        torch.cuda.synchronize()
+        draws.add(tuple(t.cpu().tolist()))
+    assert len(draws) > 1, "uniform logits should sample different tokens per step"
+
+
+def test_embedding_and_gather():
+    tab = rnd(64, 256, seed=101)
+    ids = torch.randint(0, 64, (10,), dtype=torch.int32, device=DEV)
+    out = ops.embedding(ids, tab)
+    torch.cuda.synchronize()
+    assert close(out, tab.cpu()[ids.cpu().long()], atol=0)
+    rows = torch.tensor([3, 7, 9], dtype=torch.int32, device=DEV)
+    g = ops.gather_rows(out, rows)
+    torch.cuda.synchronize()
+    assert close(g, out.cpu()[rows.cpu().long()], atol=0)
