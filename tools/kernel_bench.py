@@ -1,0 +1,108 @@
+#!/usr/bin/env python3
+"""Per-kernel microbenchmarks on MI355X: our HIP kernels vs roofline and
+vs hipBLASLt (torch.matmul) where applicable.  Prints one JSON line per case.
+"""
+import argparse
+import json
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+import torch
+
+from agentfield_amd import ops
+
+DEV = "cuda"
+
+
+def timeit(fn, iters=20, warmup=5):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+
+def bench_gemm():
+    for (M, N, K) in [(8192, 8192, 8192), (4096, 4096, 4096),
+                      (8192, 14336 * 2, 4096), (16, 4096, 4096)]:
+        a = torch.randn(M, K, dtype=torch.bfloat16, device=DEV)
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV)
+        tf = 2.0 * M * N * K / 1e12
+        t_ours = timeit(lambda: ops.gemm_bf16(a, w))
+        t_blas = timeit(lambda: a @ w.t())
+        print(json.dumps({"op": "gemm_bf16", "MNK": [M, N, K],
+                          "ours_tflops": round(tf / t_ours, 1),
+                          "hipblaslt_tflops": round(tf / t_blas, 1)}))
+
+
+def bench_prefill(Hq=32, Hk=8, D=128):
+    for (B, S) in [(16, 512), (4, 2048), (1, 8192)]:
+        T = B * S
+        q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device=DEV)
+        k = torch.randn(T, Hk, D, dtype=torch.bfloat16, device=DEV)
+        v = torch.randn_like(k)
+        cu = torch.arange(0, T + 1, S, dtype=torch.int32, device=DEV)
+        lens = [S] * B
+        # causal flops: 2 * (QK + PV) * 0.5 * S^2 * D * Hq per seq
+        tf = 2.0 * 2 * 0.5 * S * S * D * Hq * B / 1e12
+        t = timeit(lambda: ops.attn_prefill(q, k, v, cu, lens))
+        print(json.dumps({"op": "attn_prefill", "B": B, "S": S,
+                          "tflops": round(tf / t, 1), "ms": round(t * 1e3, 3)}))
+
+
+def bench_decode(Hq=32, Hk=8, D=128, page=16):
+    for (B, L, nsplit) in [(16, 576, 4), (16, 4096, 4), (64, 1024, 1),
+                           (256, 1024, 1), (1, 8192, 16)]:
+        npages = B * ((L + page - 1) // page) + 1
+        q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=DEV)
+        kc = torch.randn(npages, Hk, page, D, dtype=torch.bfloat16, device=DEV)
+        vc = torch.randn_like(kc)
+        maxp = (L + page - 1) // page
+        bt = torch.arange(1, npages, dtype=torch.int32, device=DEV).reshape(B, maxp)
+        lens = torch.full((B,), L, dtype=torch.int32, device=DEV)
+        gb = 2.0 * B * L * Hk * D * 2 / 1e9  # K+V bytes read
+        t = timeit(lambda: ops.attn_decode(q, kc, vc, bt, lens, nsplit=nsplit))
+        print(json.dumps({"op": "attn_decode", "B": B, "L": L, "nsplit": nsplit,
+                          "gbps": round(gb / t, 1), "us": round(t * 1e6, 1)}))
+
+
+def bench_norm_rope(H=4096):
+    for T in (16, 8192):
+        x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+        w = torch.randn(H, dtype=torch.bfloat16, device=DEV)
+        res = torch.randn_like(x)
+        gb = 4.0 * T * H * 2 / 1e9  # read x,res; write out,res
+        t = timeit(lambda: ops.rmsnorm(x, w, 1e-5, residual=res))
+        print(json.dumps({"op": "rmsnorm_fused", "T": T, "gbps": round(gb / t, 1),
+                          "us": round(t * 1e6, 1)}))
+        gu = torch.randn(T, 28672, dtype=torch.bfloat16, device=DEV)
+        gb2 = (T * 28672 + T * 14336) * 2 / 1e9
+        t2 = timeit(lambda: ops.silu_and_mul(gu))
+        print(json.dumps({"op": "silu_mul", "T": T, "gbps": round(gb2 / t2, 1),
+                          "us": round(t2 * 1e6, 1)}))
+
+
+def bench_sample(B=64, V=128256):
+    logits = torch.randn(B, V, dtype=torch.bfloat16, device=DEV)
+    st = ops.SamplerState(B, DEV)
+    temps = torch.full((B,), 0.8, device=DEV)
+    t = timeit(lambda: ops.sample(logits, temps, st))
+    print(json.dumps({"op": "sample", "B": B, "V": V, "us": round(t * 1e6, 1)}))
+
+
+if __name__ == "__main__":
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--only", default=None)
+    args = ap.parse_args()
+    torch.manual_seed(0)
+    for name, fn in [("gemm", bench_gemm), ("prefill", bench_prefill),
+                     ("decode", bench_decode), ("norm", bench_norm_rope),
+                     ("sample", bench_sample)]:
+        if args.only in (None, name):
+            fn()
