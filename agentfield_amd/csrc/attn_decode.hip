@@ -1,22 +1,25 @@
 // Paged decode attention (one query token per sequence) with GQA and
 // split-K over the KV length ("flash-decoding" style).
 //
-// Shapes: q   [B, Hq, D=128] bf16     (post-RoPE)
+// Shapes: q   [B, rows at stride qs] bf16, head h at offset h*D (post-RoPE)
 //         kc  [npages, Hk, page_size, D] bf16
 //         vc  [npages, Hk, page_size, D] bf16
 //         bt  [B, max_pages] i32      block table
 //         len [B] i32                 total tokens per seq (incl. current)
-//         out [B, Hq, D] bf16
-// GQA group G = Hq/Hk, 1 <= G <= 8.
+//         out [B, Hq*D] bf16 (contiguous)
+// GQA group G = Hq/Hk in {1,2,4,8}.  D = 128.
 //
-// Work decomposition (memory-bound op, CDNA4-shaped):
+// Work decomposition (memory-bound; CDNA4-shaped):
 //   grid (B, Hk, NSPLIT), block = 256 threads = 4 waves.
-//   Each wave walks tokens of its split range with stride 4; a wave's 64
-//   lanes cover the 128-dim head (2 dims/lane) so every K/V row is one
-//   coalesced 256 B wave read (u16x2 per lane).  Scores for the G query
-//   heads sharing this KV head are wave-reduced; online softmax keeps
-//   (m, l, o[G][2]) in registers.  Wave partials merge through LDS; split
-//   partials merge in a second kernel (af_attn_decode_combine).
+//   A wave iteration covers FOUR tokens: lane = tg*16 + dl, token subgroup
+//   tg = lane>>4, dim-group dl = lane&15 owning dims [dl*8, dl*8+8).  Each
+//   K/V row is a 16-lane x 16 B coalesced read; within-page consecutive
+//   tokens make the full wave read 1 KiB contiguous.  The score reduction is
+//   4 xor-shuffles within the 16-lane group + 2 across groups — ~2 shuffles
+//   per token vs 6 for a whole-wave-per-token layout, with 4x the load ILP.
+//   Online softmax state (m, l) is tracked wave-wide over the 4-token tile;
+//   o accumulates per-lane (8 dims) and is tg-reduced once at the end.
+//   Wave partials merge through LDS; split partials merge in a second kernel.
 #include "common.h"
 
 #define AD_D 128
@@ -28,10 +31,10 @@ __global__ void __launch_bounds__(256) attn_decode_kernel(
     u16* __restrict__ out, float* __restrict__ po, float* __restrict__ pml,
     const u16* __restrict__ q, const u16* __restrict__ kc, const u16* __restrict__ vc,
     const i32* __restrict__ bt, const i32* __restrict__ len,
-    float scale, int Hk, int page_size, int max_pages, int nsplit) {
+    float scale, int Hk, int page_size, int max_pages, int nsplit, i64 qs) {
   const int b = blockIdx.x, kvh = blockIdx.y, split = blockIdx.z;
   const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  const int d0 = lane * 2;
+  const int tg = lane >> 4, dl = lane & 15;
   const int Hq = Hk * G;
 
   const int L = len[b];
@@ -39,39 +42,69 @@ __global__ void __launch_bounds__(256) attn_decode_kernel(
   const int t0 = split * chunk;
   const int t1 = min(L, t0 + chunk);
 
-  // Q for the G heads of this group, pre-scaled.
-  float qr[G][2];
+  // Q (8 dims per lane per head), pre-scaled
+  float qr[G][8];
 #pragma unroll
   for (int g = 0; g < G; ++g) {
-    const u16* qp = q + ((size_t)b * Hq + kvh * G + g) * AD_D + d0;
-    qr[g][0] = bf2f(qp[0]) * scale;
-    qr[g][1] = bf2f(qp[1]) * scale;
+    const u16* qp = q + (size_t)b * qs + (kvh * G + g) * AD_D + dl * 8;
+    s16x8 qv = *reinterpret_cast<const s16x8*>(qp);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) qr[g][j] = bf2f((u16)qv[j]) * scale;
   }
 
-  float m[G], l[G], acc[G][2];
+  float m[G], l[G], acc[G][8];
 #pragma unroll
-  for (int g = 0; g < G; ++g) { m[g] = AF_NEG_INF; l[g] = 0.f; acc[g][0] = 0.f; acc[g][1] = 0.f; }
+  for (int g = 0; g < G; ++g) {
+    m[g] = AF_NEG_INF;
+    l[g] = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[g][j] = 0.f;
+  }
 
   const i32* btrow = bt + (size_t)b * max_pages;
-  for (int t = t0 + wid; t < t1; t += 4) {
-    const i64 page = btrow[t / page_size];
-    const size_t base = (((size_t)page * Hk + kvh) * page_size + (t % page_size)) * AD_D + d0;
-    const u16x2 kv = *reinterpret_cast<const u16x2*>(kc + base);
-    const float k0 = bf2f(kv.x), k1 = bf2f(kv.y);
-    const u16x2 vv = *reinterpret_cast<const u16x2*>(vc + base);
-    const float v0 = bf2f(vv.x), v1 = bf2f(vv.y);
+  // wave w handles tokens t0 + i*16 + w*4 + tg
+  for (int tb = t0 + wid * 4; tb < t1; tb += 16) {
+    const int t = tb + tg;
+    const bool valid = t < t1;
+    const int tc = valid ? t : (t1 - 1);
+    const i64 page = btrow[tc / page_size];
+    const size_t base =
+        (((size_t)page * Hk + kvh) * page_size + (tc % page_size)) * AD_D + dl * 8;
+    const s16x8 kv8 = *reinterpret_cast<const s16x8*>(kc + base);
+    const s16x8 vv8 = *reinterpret_cast<const s16x8*>(vc + base);
+    float kf[8], vf[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { kf[j] = bf2f((u16)kv8[j]); vf[j] = bf2f((u16)vv8[j]); }
 #pragma unroll
     for (int g = 0; g < G; ++g) {
-      const float s = wave_sum_f32(qr[g][0] * k0 + qr[g][1] * k1);
-      const float mn = fmaxf(m[g], s);
-      const float corr = __expf(m[g] - mn);
-      const float p = __expf(s - mn);
-      l[g] = l[g] * corr + p;
-      acc[g][0] = acc[g][0] * corr + p * v0;
-      acc[g][1] = acc[g][1] * corr + p * v1;
+      float d = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) d += qr[g][j] * kf[j];
+      // 16-lane reduce -> s uniform within token group
+      float s = group16_sum_f32(d);
+      if (!valid) s = AF_NEG_INF;
+      // tile max across the 4 token groups
+      float tm = fmaxf(s, __shfl_xor(s, 16, 64));
+      tm = fmaxf(tm, __shfl_xor(tm, 32, 64));
+      const float mn = fmaxf(m[g], tm);
+      const float corr = (m[g] <= AF_NEG_INF) ? 0.f : __expf(m[g] - mn);
+      const float p = (s <= AF_NEG_INF) ? 0.f : __expf(s - mn);
+      float psum = p + __shfl_xor(p, 16, 64);
+      psum += __shfl_xor(psum, 32, 64);
+      l[g] = l[g] * corr + psum;
       m[g] = mn;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[g][j] = acc[g][j] * corr + p * vf[j];
     }
   }
+  // fold the 4 token-groups' partial o (same dims, disjoint tokens)
+#pragma unroll
+  for (int g = 0; g < G; ++g)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      acc[g][j] += __shfl_xor(acc[g][j], 16, 64);
+      acc[g][j] += __shfl_xor(acc[g][j], 32, 64);
+    }
 
   // ---- merge 4 wave partials through LDS ----
   __shared__ float s_ml[4][G][2];
@@ -80,10 +113,11 @@ __global__ void __launch_bounds__(256) attn_decode_kernel(
 #pragma unroll
     for (int g = 0; g < G; ++g) { s_ml[wid][g][0] = m[g]; s_ml[wid][g][1] = l[g]; }
   }
+  if (tg == 0) {
 #pragma unroll
-  for (int g = 0; g < G; ++g) {
-    s_o[wid][g][d0] = acc[g][0];
-    s_o[wid][g][d0 + 1] = acc[g][1];
+    for (int g = 0; g < G; ++g)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) s_o[wid][g][dl * 8 + j] = acc[g][j];
   }
   __syncthreads();
   if (wid != 0) return;
@@ -93,23 +127,29 @@ __global__ void __launch_bounds__(256) attn_decode_kernel(
     float M = AF_NEG_INF;
 #pragma unroll
     for (int w = 0; w < 4; ++w) M = fmaxf(M, s_ml[w][g][0]);
-    float L2 = 0.f, o0 = 0.f, o1 = 0.f;
+    float L2 = 0.f, o8[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o8[j] = 0.f;
 #pragma unroll
     for (int w = 0; w < 4; ++w) {
       const float c = (s_ml[w][g][0] <= AF_NEG_INF) ? 0.f : __expf(s_ml[w][g][0] - M);
       L2 += s_ml[w][g][1] * c;
-      o0 += s_o[w][g][d0] * c;
-      o1 += s_o[w][g][d0 + 1] * c;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) o8[j] += s_o[w][g][dl * 8 + j] * c;
     }
+    if (tg != 0) continue;
     if (nsplit == 1) {
       const float inv = (L2 > 0.f) ? 1.f / L2 : 0.f;
-      u16* op = out + ((size_t)b * Hq + kvh * G + g) * AD_D + d0;
-      op[0] = f2bf(o0 * inv);
-      op[1] = f2bf(o1 * inv);
+      u16* op = out + (size_t)b * (Hq * AD_D) + (kvh * G + g) * AD_D + dl * 8;
+      s16x8 ov;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ov[j] = (short)f2bf(o8[j] * inv);
+      *reinterpret_cast<s16x8*>(op) = ov;
     } else {
       const size_t pbase = ((((size_t)b * Hk + kvh) * nsplit + split) * G + g);
-      float* od = po + pbase * AD_D + d0;
-      od[0] = o0; od[1] = o1;
+      float* od = po + pbase * AD_D + dl * 8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) od[j] = o8[j];
       if (lane == 0) { pml[pbase * 2] = M; pml[pbase * 2 + 1] = L2; }
     }
   }
@@ -139,7 +179,7 @@ __global__ void __launch_bounds__(64) attn_decode_combine_kernel(
     o1 += po[pbase * AD_D + d0 + 1] * c;
   }
   const float inv = (L2 > 0.f) ? 1.f / L2 : 0.f;
-  u16* op = out + ((size_t)b * Hq + h) * AD_D + d0;
+  u16* op = out + (size_t)b * (Hq * AD_D) + h * AD_D + d0;
   op[0] = f2bf(o0 * inv);
   op[1] = f2bf(o1 * inv);
 }
@@ -148,7 +188,7 @@ AF_EXPORT int af_attn_decode(void* out, void* po, void* pml, const void* q,
                              const void* kc, const void* vc, const void* bt,
                              const void* len, float scale, int B, int Hq, int Hk,
                              int D, int page_size, int max_pages, int nsplit,
-                             void* stream) {
+                             i64 qs, void* stream) {
   if (D != AD_D) return 9002;
   const int G = Hq / Hk;
   if (G < 1 || G > AD_MAXG || G * Hk != Hq) return 9003;
@@ -159,7 +199,7 @@ AF_EXPORT int af_attn_decode(void* out, void* po, void* pml, const void* q,
   attn_decode_kernel<GG><<<grid, blk, 0, st>>>(                                  \
       (u16*)out, (float*)po, (float*)pml, (const u16*)q, (const u16*)kc,         \
       (const u16*)vc, (const i32*)bt, (const i32*)len, scale, Hk, page_size,     \
-      max_pages, nsplit)
+      max_pages, nsplit, qs)
   switch (G) {
     case 1: AF_LAUNCH(1); break;
     case 2: AF_LAUNCH(2); break;

@@ -46,7 +46,7 @@ __global__ void __launch_bounds__(256) attn_prefill_kernel(
     u16* __restrict__ out, const u16* __restrict__ q, const u16* __restrict__ k,
     const u16* __restrict__ v, const i32* __restrict__ cu_seqlens,
     const i32* __restrict__ tile_seq, const i32* __restrict__ tile_q0,
-    float scale, int Hq, int Hk) {
+    float scale, int Hq, int Hk, i64 qs, i64 ks, i64 vs) {
   const int tile = blockIdx.x, kvh = blockIdx.y;
   const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
   const int seq = tile_seq[tile];
@@ -76,7 +76,7 @@ __global__ void __launch_bounds__(256) attn_prefill_kernel(
 #pragma unroll
   for (int hl = 0; hl < HLOOP; ++hl) {
     const int qh = kvh * G + g0 + hl * 4;
-    const u16* qp = q + ((size_t)(seq_start + qrow_frag) * Hq + qh) * AP_D +
+    const u16* qp = q + (size_t)(seq_start + qrow_frag) * qs + qh * AP_D +
                     (lane >> 4) * 8;
 #pragma unroll
     for (int c = 0; c < 4; ++c) q_frag[hl][c] = lds_read8(qp + c * 32);
@@ -104,14 +104,15 @@ __global__ void __launch_bounds__(256) attn_prefill_kernel(
       const int tok = i / (AP_D / 8);
       const int d8 = (i % (AP_D / 8)) * 8;
       const int tg = kv0 + tok;
-      const size_t src = ((size_t)(seq_start + min(tg, len - 1)) * Hk + kvh) * AP_D + d8;
+      const size_t kvrow = (size_t)(seq_start + min(tg, len - 1));
       {  // K: row-major swizzled, garbage beyond len is masked later
-        s16x8 kv8 = lds_read8(k + src);
+        s16x8 kv8 = lds_read8(k + kvrow * ks + kvh * AP_D + d8);
         const int byte = tok * (AP_D * 2) + ((d8 * 2) ^ ((tok & 7) << 4));
         *reinterpret_cast<s16x8*>(reinterpret_cast<char*>(k_lds) + byte) = kv8;
       }
       {  // V: transposed, zero-filled beyond len (0 * P avoids NaN)
-        s16x8 vv8 = (tg < len) ? lds_read8(v + src) : s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+        s16x8 vv8 = (tg < len) ? lds_read8(v + kvrow * vs + kvh * AP_D + d8)
+                               : s16x8{0, 0, 0, 0, 0, 0, 0, 0};
 #pragma unroll
         for (int j = 0; j < 8; ++j) v_lds[d8 + j][tok] = (u16)vv8[j];
       }
@@ -190,7 +191,8 @@ __global__ void __launch_bounds__(256) attn_prefill_kernel(
 AF_EXPORT int af_attn_prefill(void* out, const void* q, const void* k, const void* v,
                               const void* cu_seqlens, const void* tile_seq,
                               const void* tile_q0, float scale, int ntiles,
-                              int Hq, int Hk, int D, void* stream) {
+                              int Hq, int Hk, int D, i64 qs, i64 ks, i64 vs,
+                              void* stream) {
   if (D != AP_D) return 9002;
   const int G = Hq / Hk;
   if (G * Hk != Hq) return 9003;
@@ -201,7 +203,7 @@ AF_EXPORT int af_attn_prefill(void* out, const void* q, const void* k, const voi
   attn_prefill_kernel<GG><<<grid, blk, 0, st>>>(                                \
       (u16*)out, (const u16*)q, (const u16*)k, (const u16*)v,                   \
       (const i32*)cu_seqlens, (const i32*)tile_seq, (const i32*)tile_q0,        \
-      scale, Hq, Hk)
+      scale, Hq, Hk, qs, ks, vs)
   switch (G) {
     case 1: AF_LAUNCH(1); break;
     case 2: AF_LAUNCH(2); break;

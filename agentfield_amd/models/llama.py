@@ -124,22 +124,21 @@ class LlamaAttention(nn.Module):
 
     def forward(self, x, positions, rope_tab, kv: KVCache, md: AttnMetadata):
         cfg = self.cfg
-        T = x.shape[0]
         qkv = x @ self.qkv.t()
-        q, k, v = qkv.split([cfg.q_size, cfg.kv_size, cfg.kv_size], dim=-1)
-        q = q.view(T, cfg.num_heads, cfg.head_dim).contiguous()
-        k = k.view(T, cfg.num_kv_heads, cfg.head_dim).contiguous()
-        v = v.view(T, cfg.num_kv_heads, cfg.head_dim).contiguous()
-        ops.rope(q, k, positions, rope_tab)
-        ops.reshape_and_cache(k, v, kv.k[self.layer_idx], kv.v[self.layer_idx],
-                              md.slots)
+        # strided views straight into the fused projection (no copies)
+        q = qkv[:, :cfg.q_size]
+        k = qkv[:, cfg.q_size:cfg.q_size + cfg.kv_size]
+        v = qkv[:, cfg.q_size + cfg.kv_size:]
+        ops.rope_cache(q, k, v, positions, rope_tab,
+                       kv.k[self.layer_idx], kv.v[self.layer_idx], md.slots)
         if md.is_prefill:
-            o = ops.attn_prefill(q, k, v, md.cu_seqlens, md.seq_lens, self.scale)
+            o = ops.attn_prefill(q, k, v, md.cu_seqlens, md.seq_lens,
+                                 self.scale, head_dim=cfg.head_dim)
         else:
             o = ops.attn_decode(q, kv.k[self.layer_idx], kv.v[self.layer_idx],
                                 md.block_table, md.seq_lens_t, self.scale,
                                 nsplit=md.nsplit, scratch=md.decode_scratch)
-        return o.view(T, cfg.q_size) @ self.o.t()
+        return o @ self.o.t()
 
 
 class LlamaMLP(nn.Module):

@@ -51,18 +51,31 @@ def rope_table(max_pos: int, head_dim: int, theta: float = 500000.0,
     return torch.cat([freqs.cos(), freqs.sin()], dim=-1).float().to(device)
 
 
-def rope(q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor,
-         table: torch.Tensor) -> None:
-    """In-place rotate-half RoPE on q [T,Hq,D] and k [T,Hk,D]."""
+def rope_cache(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+               positions: torch.Tensor, table: torch.Tensor,
+               kcache: torch.Tensor, vcache: torch.Tensor,
+               slots: torch.Tensor) -> None:
+    """Fused in-place RoPE on q/k + paged-cache append of k/v.
+
+    q [T, Hq*D], k/v [T, Hk*D] — rows may be strided (views straight out of
+    the fused QKV projection); caches [npages, Hk, page, D].
+    """
+    _, Hk, page, D = kcache.shape
+    T = q.shape[0]
+    Hq = q.shape[1] // D
+    assert q.stride(1) == 1 and k.stride(1) == 1 and v.stride(1) == 1
     if not _on_gpu(q):
-        ref.rope(q, k, positions, table)
+        q3 = q.unflatten(-1, (Hq, D))
+        k3 = k.unflatten(-1, (Hk, D))
+        ref.rope(q3, k3, positions, table)
+        ref.reshape_and_cache(k3, v.unflatten(-1, (Hk, D)), kcache, vcache, slots)
         return
-    T, Hq, D = q.shape
-    Hk = k.shape[1]
-    rc = _lib.lib().af_rope(
-        _lib.ptr(q), _lib.ptr(k), _lib.ptr(positions.int()), _lib.ptr(table),
-        T, Hq, Hk, D, _lib.cur_stream())
-    _lib.check(rc, "af_rope")
+    rc = _lib.lib().af_rope_cache(
+        _lib.ptr(q), _lib.ptr(k), _lib.ptr(v), _lib.ptr(positions.int()),
+        _lib.ptr(table), _lib.ptr(kcache), _lib.ptr(vcache), _lib.ptr(slots),
+        T, Hq, Hk, D, q.stride(0), k.stride(0), v.stride(0), page,
+        _lib.cur_stream())
+    _lib.check(rc, "af_rope_cache")
 
 
 # ---------------------------------------------------------------- activation
@@ -110,15 +123,17 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
                 scale: float | None = None, nsplit: int = 1,
                 scratch: tuple[torch.Tensor, torch.Tensor] | None = None,
                 out: torch.Tensor | None = None) -> torch.Tensor:
-    """Paged decode attention. q [B,Hq,D] -> out [B,Hq,D]."""
-    B, Hq, D = q.shape
-    _, Hk, page, _ = kcache.shape
+    """Paged decode attention. q [B, Hq*D] (strided rows ok) -> out [B, Hq*D]."""
+    _, Hk, page, D = kcache.shape
+    B = q.shape[0]
+    Hq = q.shape[1] // D
     if scale is None:
         scale = 1.0 / math.sqrt(D)
     if not _on_gpu(q):
-        return ref.attn_decode(q, kcache, vcache, block_table, seq_lens, scale)
+        return ref.attn_decode(q.unflatten(-1, (Hq, D)), kcache, vcache,
+                               block_table, seq_lens, scale).flatten(1)
     if out is None:
-        out = torch.empty_like(q)
+        out = torch.empty(B, Hq * D, dtype=q.dtype, device=q.device)
     G = Hq // Hk
     if nsplit > 1:
         if scratch is None:
@@ -132,7 +147,7 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
         _lib.ptr(out), _lib.ptr(po), _lib.ptr(pml), _lib.ptr(q),
         _lib.ptr(kcache), _lib.ptr(vcache), _lib.ptr(block_table),
         _lib.ptr(seq_lens), scale, B, Hq, Hk, D, page, block_table.shape[1],
-        nsplit, _lib.cur_stream())
+        nsplit, q.stride(0), _lib.cur_stream())
     _lib.check(rc, "af_attn_decode")
     return out
 
@@ -149,25 +164,32 @@ def prefill_tiles(seq_lens: list[int], rows_per_wg: int):
 
 def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                  cu_seqlens: torch.Tensor, seq_lens: list[int],
-                 scale: float | None = None) -> torch.Tensor:
-    """Causal varlen prefill attention.  q [T,Hq,D], k/v [T,Hk,D]."""
-    T, Hq, D = q.shape
-    Hk = k.shape[1]
+                 scale: float | None = None, head_dim: int = 128) -> torch.Tensor:
+    """Causal varlen prefill attention.  q [T, Hq*D], k/v [T, Hk*D]
+    (strided rows ok) -> out [T, Hq*D] contiguous."""
+    D = head_dim
+    T = q.shape[0]
+    Hq = q.shape[1] // D
+    Hk = k.shape[1] // D
     if scale is None:
         scale = 1.0 / math.sqrt(D)
     if not _on_gpu(q):
-        return ref.attn_prefill(q, k, v, cu_seqlens, scale)
+        return ref.attn_prefill(q.unflatten(-1, (Hq, D)),
+                                k.unflatten(-1, (Hk, D)),
+                                v.unflatten(-1, (Hk, D)),
+                                cu_seqlens, scale).flatten(1)
     G = Hq // Hk
     rows_per_wg = 16 * max(1, 4 // G)
     ts, tq = prefill_tiles(seq_lens, rows_per_wg)
     dev = q.device
     tile_seq = torch.tensor(ts, dtype=torch.int32, device=dev)
     tile_q0 = torch.tensor(tq, dtype=torch.int32, device=dev)
-    out = torch.empty_like(q)
+    out = torch.empty(T, Hq * D, dtype=q.dtype, device=q.device)
     rc = _lib.lib().af_attn_prefill(
         _lib.ptr(out), _lib.ptr(q), _lib.ptr(k), _lib.ptr(v),
         _lib.ptr(cu_seqlens), _lib.ptr(tile_seq), _lib.ptr(tile_q0),
-        scale, len(ts), Hq, Hk, D, _lib.cur_stream())
+        scale, len(ts), Hq, Hk, D, q.stride(0), k.stride(0), v.stride(0),
+        _lib.cur_stream())
     _lib.check(rc, "af_attn_prefill")
     return out
 

@@ -43,20 +43,23 @@ def _declare(l: ctypes.CDLL) -> None:
     f = ctypes.c_float
     u32 = ctypes.c_uint32
     l.af_rmsnorm.argtypes = [p, p, p, p, p, f, i, i, p]
-    l.af_rope.argtypes = [p, p, p, p, i, i, i, i, p]
+    l.af_rope_cache.argtypes = [p, p, p, p, p, p, p, p, i, i, i, i,
+                                i64, i64, i64, i, p]
     l.af_silu_mul.argtypes = [p, p, i64, i64, p]
     l.af_add.argtypes = [p, p, p, i64, p]
     l.af_reshape_and_cache.argtypes = [p, p, p, p, p, i, i, i, i, p]
     l.af_embedding.argtypes = [p, p, p, i, i, p]
-    l.af_attn_decode.argtypes = [p, p, p, p, p, p, p, p, f, i, i, i, i, i, i, i, p]
-    l.af_attn_prefill.argtypes = [p, p, p, p, p, p, p, f, i, i, i, i, p]
+    l.af_attn_decode.argtypes = [p, p, p, p, p, p, p, p, f, i, i, i, i, i, i, i,
+                                 i64, p]
+    l.af_attn_prefill.argtypes = [p, p, p, p, p, p, p, f, i, i, i, i,
+                                  i64, i64, i64, p]
     l.af_gemm_bf16.argtypes = [p, p, p, i, i, i, p]
     l.af_sample.argtypes = [p, p, p, p, p, p, u32, i, i, p]
     l.af_gather_rows.argtypes = [p, p, p, i, i, p]
     l.af_mfma_probe.argtypes = [p, p, p, p]
     l.af_axpy.argtypes = [p, p, f, i, p]
     l.af_device_sync.argtypes = []
-    for fn in ("af_rmsnorm", "af_rope", "af_silu_mul", "af_add",
+    for fn in ("af_rmsnorm", "af_rope_cache", "af_silu_mul", "af_add",
                "af_reshape_and_cache", "af_embedding", "af_attn_decode",
                "af_attn_prefill", "af_gemm_bf16", "af_sample",
                "af_gather_rows", "af_mfma_probe", "af_axpy", "af_device_sync"):

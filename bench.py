@@ -32,7 +32,7 @@ def parse_args():
     p.add_argument("--steps", type=int, default=4)
     p.add_argument("--warmup", type=int, default=1)
     p.add_argument("--model", default="llama-3-8b")
-    p.add_argument("--calls", type=int, default=16,
+    p.add_argument("--calls", type=int, default=64,
                    help="reasoner calls per rank per step")
     p.add_argument("--prompt-len", type=int, default=512)
     p.add_argument("--gen-len", type=int, default=64)

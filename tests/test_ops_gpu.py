@@ -75,17 +75,30 @@ def test_rmsnorm_fused_residual():
     assert close(new_res, res_ref)
 
 
-def test_rope():
-    T, Hq, Hk, D = 9, 4, 2, 128
-    q = rnd(T, Hq, D, seed=8)
-    k = rnd(T, Hk, D, seed=9)
-    qc, kc = q.cpu().clone(), k.cpu().clone()
+def test_rope_cache_fused_strided():
+    """Fused RoPE+cache on strided rows (views of a fused QKV buffer)."""
+    T, Hq, Hk, D, page, npages = 9, 4, 2, 128, 4, 8
+    qkv = rnd(T, (Hq + 2 * Hk) * D, seed=8)
+    qkv_c = qkv.cpu().clone()
+    q = qkv[:, :Hq * D]
+    k = qkv[:, Hq * D:(Hq + Hk) * D]
+    v = qkv[:, (Hq + Hk) * D:]
     pos = torch.randint(0, 100, (T,), dtype=torch.int32, device=DEV)
     tab = ops.rope_table(128, D, device=DEV)
-    ops.rope(q, k, pos, tab)
+    kc = torch.zeros(npages, Hk, page, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    slots = torch.randperm(npages * page, device=DEV)[:T].to(torch.int64)
+    ops.rope_cache(q, k, v, pos, tab, kc, vc, slots)
     torch.cuda.synchronize()
-    ref.rope(qc, kc, pos.cpu(), tab.cpu())
-    assert close(q, qc) and close(k, kc)
+    # CPU reference on the same strided layout
+    qr = qkv_c[:, :Hq * D]
+    kr = qkv_c[:, Hq * D:(Hq + Hk) * D]
+    vr = qkv_c[:, (Hq + Hk) * D:]
+    kcr = torch.zeros_like(kc, device="cpu")
+    vcr = torch.zeros_like(vc, device="cpu")
+    ops.rope_cache(qr, kr, vr, pos.cpu(), tab.cpu(), kcr, vcr, slots.cpu())
+    assert close(q, qr) and close(k, kr)
+    assert close(kc, kcr, atol=0) and close(vc, vcr, atol=0)
 
 
 def test_silu_mul():
@@ -114,7 +127,7 @@ def _decode_case(B, Hq, Hk, lens, page=16, nsplit=1, seed=13):
     torch.manual_seed(seed)
     maxp = (max(lens) + page - 1) // page
     npages = 1 + sum((l + page - 1) // page for l in lens)
-    q = rnd(B, Hq, D, seed=seed)
+    q = rnd(B, Hq * D, seed=seed)
     kc = rnd(npages, Hk, page, D, seed=seed + 1)
     vc = rnd(npages, Hk, page, D, seed=seed + 2)
     perm = torch.randperm(npages - 1) + 1
@@ -128,8 +141,8 @@ def _decode_case(B, Hq, Hk, lens, page=16, nsplit=1, seed=13):
     lens_t = torch.tensor(lens, dtype=torch.int32, device=DEV)
     out = ops.attn_decode(q, kc, vc, bt, lens_t, nsplit=nsplit)
     torch.cuda.synchronize()
-    want = ref.attn_decode(q.cpu(), kc.cpu(), vc.cpu(), bt.cpu(),
-                           lens_t.cpu(), 1.0 / math.sqrt(D))
+    want = ref.attn_decode(q.cpu().unflatten(-1, (Hq, D)), kc.cpu(), vc.cpu(),
+                           bt.cpu(), lens_t.cpu(), 1.0 / math.sqrt(D)).flatten(1)
     assert close(out, want), f"decode mismatch B={B} G={Hq//Hk} nsplit={nsplit}"
 
 
@@ -150,17 +163,19 @@ def test_attn_decode_split():
 def _prefill_case(Hq, Hk, lens, seed=30):
     D = 128
     T = sum(lens)
-    q = rnd(T, Hq, D, seed=seed)
-    k = rnd(T, Hk, D, seed=seed + 1)
-    v = rnd(T, Hk, D, seed=seed + 2)
+    q = rnd(T, Hq * D, seed=seed)
+    k = rnd(T, Hk * D, seed=seed + 1)
+    v = rnd(T, Hk * D, seed=seed + 2)
     cu = [0]
     for l in lens:
         cu.append(cu[-1] + l)
     cu_t = torch.tensor(cu, dtype=torch.int32, device=DEV)
     out = ops.attn_prefill(q, k, v, cu_t, lens)
     torch.cuda.synchronize()
-    want = ref.attn_prefill(q.cpu(), k.cpu(), v.cpu(), cu_t.cpu(),
-                            1.0 / math.sqrt(D))
+    want = ref.attn_prefill(q.cpu().unflatten(-1, (Hq, D)),
+                            k.cpu().unflatten(-1, (Hk, D)),
+                            v.cpu().unflatten(-1, (Hk, D)), cu_t.cpu(),
+                            1.0 / math.sqrt(D)).flatten(1)
     assert close(out, want), f"prefill mismatch G={Hq//Hk} lens={lens}"
 
 

@@ -44,8 +44,8 @@ def bench_gemm():
 def bench_prefill(Hq=32, Hk=8, D=128):
     for (B, S) in [(16, 512), (4, 2048), (1, 8192)]:
         T = B * S
-        q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device=DEV)
-        k = torch.randn(T, Hk, D, dtype=torch.bfloat16, device=DEV)
+        q = torch.randn(T, Hq * D, dtype=torch.bfloat16, device=DEV)
+        k = torch.randn(T, Hk * D, dtype=torch.bfloat16, device=DEV)
         v = torch.randn_like(k)
         cu = torch.arange(0, T + 1, S, dtype=torch.int32, device=DEV)
         lens = [S] * B
@@ -60,7 +60,7 @@ def bench_decode(Hq=32, Hk=8, D=128, page=16):
     for (B, L, nsplit) in [(16, 576, 4), (16, 4096, 4), (64, 1024, 1),
                            (256, 1024, 1), (1, 8192, 16)]:
         npages = B * ((L + page - 1) // page) + 1
-        q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=DEV)
+        q = torch.randn(B, Hq * D, dtype=torch.bfloat16, device=DEV)
         kc = torch.randn(npages, Hk, page, D, dtype=torch.bfloat16, device=DEV)
         vc = torch.randn_like(kc)
         maxp = (L + page - 1) // page
