@@ -1,0 +1,711 @@
+"""The control-plane HTTP server: the full agent API + UI API surface
+(reference parity: internal/server/server.go setupRoutes).
+
+Execution hot path: parse target -> execution record ->
+HTTP call to the agent node -> 200-sync or 202+status-callback -> event-bus
+wake of sync waiters -> webhook + VC on terminal status.
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+from contextlib import asynccontextmanager
+
+import httpx
+from fastapi import FastAPI, Request, Response, WebSocket, WebSocketDisconnect
+from fastapi.responses import JSONResponse, StreamingResponse
+
+from . import status as st
+from .did import DIDService, Keystore, VCService
+from .events import Buses
+from .ids import new_execution_id, new_run_id
+from .services import (ExecutionCleanup, HealthMonitor, Metrics,
+                       PayloadStore, PresenceManager, WebhookDispatcher)
+from .storage import Storage
+from .workflow import aggregate_status, build_dag
+
+
+class Config:
+    def __init__(self, **kw):
+        self.db_path = kw.get("db_path", ":memory:")
+        self.payload_dir = kw.get("payload_dir")
+        self.keystore_path = kw.get("keystore_path")
+        self.sync_timeout = kw.get("sync_timeout", 90.0)
+        self.agent_timeout = kw.get("agent_timeout", 90.0)
+        self.async_workers = kw.get("async_workers", 8)
+        self.async_queue_capacity = kw.get("async_queue_capacity", 1024)
+        self.did_enabled = kw.get("did_enabled", True)
+        self.auto_vc = kw.get("auto_vc", False)
+        self.presence_ttl = kw.get("presence_ttl", 300.0)
+        self.health_interval = kw.get("health_interval", 60.0)
+        self.background_services = kw.get("background_services", True)
+
+
+class ControlPlane:
+    def __init__(self, cfg: Config | None = None):
+        self.cfg = cfg or Config()
+        self.storage = Storage(self.cfg.db_path)
+        self.buses = Buses()
+        self.metrics = Metrics()
+        self.payloads = PayloadStore(self.cfg.payload_dir)
+        self.webhooks = WebhookDispatcher(self.storage, self.metrics)
+        self.presence = PresenceManager(self.storage, self.buses,
+                                        ttl=self.cfg.presence_ttl)
+        self.health = HealthMonitor(self.storage, self.buses,
+                                    interval=self.cfg.health_interval)
+        self.cleanup = ExecutionCleanup(self.storage)
+        self.dids = DIDService(self.storage, Keystore(self.cfg.keystore_path)) \
+            if self.cfg.did_enabled else None
+        self.vcs = VCService(self.storage, self.dids) if self.dids else None
+        self.client: httpx.AsyncClient | None = None
+        self._async_q: asyncio.Queue | None = None
+        self._workers: list[asyncio.Task] = []
+        self.started_at = time.time()
+
+    async def start(self):
+        self.client = httpx.AsyncClient(timeout=self.cfg.agent_timeout)
+        self._async_q = asyncio.Queue(self.cfg.async_queue_capacity)
+        for _ in range(self.cfg.async_workers):
+            self._workers.append(asyncio.create_task(self._async_worker()))
+        await self.webhooks.start()
+        if self.cfg.background_services:
+            await self.presence.start()
+            await self.health.start()
+            await self.cleanup.start()
+
+    async def stop(self):
+        for t in self._workers:
+            t.cancel()
+        await self.webhooks.stop()
+        await self.presence.stop()
+        await self.health.stop()
+        await self.cleanup.stop()
+        if self.client:
+            await self.client.aclose()
+
+    # ----------------------------------------------------- execution core
+    @staticmethod
+    def parse_target(target: str) -> tuple[str, str]:
+        if "." not in target:
+            raise ValueError("target must be node_id.reasoner_id")
+        node_id, reasoner_id = target.split(".", 1)
+        return node_id, reasoner_id
+
+    @staticmethod
+    def _validate_webhook(wh: dict) -> str | None:
+        if not wh.get("url"):
+            return "webhook.url required"
+        if len(wh.get("secret") or "") > 4096:
+            return "webhook secret too large"
+        headers = wh.get("headers") or {}
+        if len(headers) > 20:
+            return "too many webhook headers"
+        for k, v in headers.items():
+            if len(str(v)) > 512 or len(str(k)) > 512:
+                return "webhook header too large"
+        return None
+
+    def _target_type(self, node: dict, reasoner_id: str) -> str:
+        skills = {s.get("id") for s in node.get("skills", [])}
+        return "skill" if reasoner_id in skills else "reasoner"
+
+    def prepare_execution(self, target: str, body: dict,
+                          headers) -> tuple[dict, dict | None]:
+        node_id, reasoner_id = self.parse_target(target)
+        node = self.storage.get_node(node_id)
+        if node is None:
+            return None, {"status_code": 404,
+                          "error": f"agent node '{node_id}' not found"}
+        webhook = body.get("webhook")
+        if webhook:
+            err = self._validate_webhook(webhook)
+            if err:
+                return None, {"status_code": 400, "error": err}
+        run_id = headers.get("x-run-id") or new_run_id()
+        rec = {
+            "id": new_execution_id(),
+            "run_id": run_id,
+            "parent_execution_id": headers.get("x-parent-execution-id"),
+            "node_id": node_id,
+            "reasoner_id": reasoner_id,
+            "target_type": self._target_type(node, reasoner_id),
+            "status": st.RUNNING,
+            "input": body.get("input", {}),
+            "session_id": headers.get("x-session-id"),
+            "actor_id": headers.get("x-actor-id"),
+            "webhook_registered": bool(webhook),
+            "_node": node,
+            "_context": body.get("context"),
+        }
+        self.storage.create_execution(rec)
+        self.storage.upsert_run(run_id, st.RUNNING, rec["id"])
+        if webhook:
+            self.storage.register_webhook(rec["id"], webhook["url"],
+                                          webhook.get("secret", ""),
+                                          webhook.get("headers"))
+        return rec, None
+
+    async def call_agent(self, rec: dict) -> tuple[int, dict | None, str | None]:
+        node = rec["_node"]
+        path = "skills" if rec["target_type"] == "skill" else "reasoners"
+        url = f"{node['base_url'].rstrip('/')}/{path}/{rec['reasoner_id']}"
+        headers = {
+            "X-Execution-ID": rec["id"],
+            "X-Run-ID": rec["run_id"],
+            "Content-Type": "application/json",
+        }
+        if rec.get("parent_execution_id"):
+            headers["X-Parent-Execution-ID"] = rec["parent_execution_id"]
+        if rec.get("session_id"):
+            headers["X-Session-ID"] = rec["session_id"]
+        if rec.get("actor_id"):
+            headers["X-Actor-ID"] = rec["actor_id"]
+        payload = dict(rec.get("input") or {})
+        try:
+            resp = await self.client.post(url, json=payload, headers=headers)
+        except Exception as e:
+            return 0, None, f"agent unreachable: {e}"
+        if resp.status_code == 202:
+            return 202, None, None
+        if resp.status_code == 200:
+            try:
+                return 200, resp.json(), None
+            except ValueError:
+                return 200, {"raw": resp.text}, None
+        return resp.status_code, None, f"agent HTTP {resp.status_code}: {resp.text[:300]}"
+
+    def complete_execution(self, execution_id: str, status: str, result=None,
+                           error: str | None = None,
+                           duration_ms: float | None = None) -> dict | None:
+        status = st.normalize(status)
+        self.storage.update_execution_result(execution_id, status, result,
+                                             error, duration_ms)
+        rec = self.storage.get_execution(execution_id)
+        if rec is None:
+            return None
+        if rec.get("run_id"):
+            sibs = self.storage.executions_by_run(rec["run_id"])
+            self.storage.upsert_run(
+                rec["run_id"], aggregate_status([e["status"] for e in sibs]))
+        if rec.get("duration_ms"):
+            self.metrics.step_duration.observe(rec["duration_ms"] / 1000.0)
+        self.buses.execution.publish({
+            "execution_id": execution_id, "status": status,
+            "terminal": st.is_terminal(status), "run_id": rec.get("run_id"),
+        })
+        if st.is_terminal(status):
+            self.webhooks.notify(rec)
+            if self.cfg.auto_vc and self.vcs:
+                try:
+                    target_did = None
+                    if rec.get("node_id"):
+                        d = self.storage.did_for_subject("agent", rec["node_id"])
+                        target_did = d["did"] if d else None
+                    self.vcs.issue_execution_vc(rec, target_did=target_did)
+                except Exception:
+                    pass
+        return rec
+
+    async def _async_worker(self):
+        while True:
+            rec = await self._async_q.get()
+            self.metrics.queue_depth.set(self._async_q.qsize())
+            self.metrics.worker_inflight.inc()
+            try:
+                code, result, err = await self.call_agent(rec)
+                if code == 200:
+                    self.complete_execution(rec["id"], st.COMPLETED,
+                                            result=result)
+                elif code != 202:
+                    self.complete_execution(rec["id"], st.FAILED, error=err)
+                # 202: agent will deliver the terminal status callback
+            finally:
+                self.metrics.worker_inflight.dec()
+
+    @staticmethod
+    def envelope(rec: dict) -> dict:
+        return {
+            "execution_id": rec["id"],
+            "run_id": rec.get("run_id"),
+            "status": rec.get("status"),
+            "result": rec.get("result"),
+            "error_message": rec.get("error_message"),
+            "duration_ms": rec.get("duration_ms"),
+            "finished_at": rec.get("finished_at"),
+            "webhook_registered": bool(rec.get("webhook_registered")),
+        }
+
+
+def _sse(event: dict) -> str:
+    return f"data: {json.dumps(event)}\n\n"
+
+
+def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
+    cp = cp or ControlPlane(Config(**cfg_kw))
+
+    @asynccontextmanager
+    async def lifespan(app: FastAPI):
+        await cp.start()
+        yield
+        await cp.stop()
+
+    app = FastAPI(title="agentfield-amd control plane", lifespan=lifespan)
+    app.state.cp = cp
+
+    # ------------------------------------------------------------- basics
+    @app.get("/api/v1/health")
+    async def health():
+        return {"status": "healthy", "uptime_s": time.time() - cp.started_at,
+                "version": "0.1.0"}
+
+    @app.get("/metrics")
+    async def metrics():
+        return Response(cp.metrics.render(), media_type="text/plain")
+
+    # -------------------------------------------------------------- nodes
+    @app.post("/api/v1/nodes/register")
+    async def register_node(req: Request):
+        body = await req.json()
+        node_id = body.get("id") or body.get("node_id")
+        if not node_id:
+            return JSONResponse({"error": "id required"}, status_code=400)
+        body["id"] = node_id
+        cp.storage.upsert_node(body)
+        cp.storage.set_node_status(node_id, "active")
+        cp.buses.node.publish({"type": "node.registered", "node_id": node_id})
+        out = {"status": "registered", "node_id": node_id,
+               "base_url": body.get("base_url")}
+        if cp.dids:
+            reasoners = [r.get("id") for r in body.get("reasoners", [])]
+            skills = [s.get("id") for s in body.get("skills", [])]
+            out["identity"] = cp.dids.register_node(node_id, reasoners, skills)
+        return out
+
+    @app.get("/api/v1/nodes")
+    async def list_nodes():
+        return {"nodes": cp.storage.list_nodes()}
+
+    @app.get("/api/v1/nodes/{node_id}")
+    async def get_node(node_id: str):
+        node = cp.storage.get_node(node_id)
+        if node is None:
+            return JSONResponse({"error": "not found"}, status_code=404)
+        return node
+
+    @app.post("/api/v1/nodes/{node_id}/heartbeat")
+    async def heartbeat(node_id: str, req: Request):
+        try:
+            body = await req.json()
+        except Exception:
+            body = {}
+        node = cp.storage.get_node(node_id)
+        if node is None:
+            return JSONResponse({"error": "not registered",
+                                 "action": "re-register"}, status_code=404)
+        status = st.normalize(body.get("status") or "active")
+        cp.storage.touch_heartbeat(node_id, "active" if status in
+                                   ("active", "healthy", "completed") else None)
+        return {"status": "ok"}
+
+    @app.post("/api/v1/nodes/{node_id}/status")
+    async def node_status(node_id: str, req: Request):
+        body = await req.json()
+        from .status import valid_node_transition
+        node = cp.storage.get_node(node_id)
+        if node is None:
+            return JSONResponse({"error": "not found"}, status_code=404)
+        nxt = body.get("status")
+        if nxt not in ("registered", "starting", "active", "inactive",
+                       "unhealthy", "stopping", "stopped"):
+            return JSONResponse({"error": f"invalid status {nxt}"},
+                                status_code=400)
+        if not valid_node_transition(node["status"], nxt):
+            return JSONResponse(
+                {"error": f"invalid transition {node['status']} -> {nxt}"},
+                status_code=409)
+        cp.storage.set_node_status(node_id, nxt)
+        cp.buses.node.publish({"type": "node.status", "node_id": node_id,
+                               "status": nxt})
+        return {"status": "ok"}
+
+    @app.delete("/api/v1/nodes/{node_id}")
+    async def delete_node(node_id: str):
+        cp.storage.delete_node(node_id)
+        cp.buses.node.publish({"type": "node.removed", "node_id": node_id})
+        return {"status": "ok"}
+
+    # ------------------------------------------------------------ execute
+    @app.post("/api/v1/execute/{target}")
+    async def execute_sync(target: str, req: Request):
+        try:
+            body = await req.json()
+        except Exception:
+            body = {}
+        rec, err = cp.prepare_execution(target, body, req.headers)
+        if err:
+            return JSONResponse({"error": err["error"]},
+                                status_code=err["status_code"])
+        t0 = time.time()
+        code, result, errmsg = await cp.call_agent(rec)
+        if code == 200:
+            final = cp.complete_execution(rec["id"], st.COMPLETED, result=result,
+                                          duration_ms=(time.time() - t0) * 1e3)
+        elif code == 202:
+            cp.metrics.waiters_inflight.inc()
+            try:
+                ev = await cp.buses.wait_for_execution(rec["id"],
+                                                       cp.cfg.sync_timeout)
+            finally:
+                cp.metrics.waiters_inflight.dec()
+            if ev is None:
+                final = cp.complete_execution(rec["id"], st.TIMEOUT,
+                                              error="sync wait timed out")
+            else:
+                final = cp.storage.get_execution(rec["id"])
+        else:
+            final = cp.complete_execution(rec["id"], st.FAILED, error=errmsg)
+        resp = JSONResponse(cp.envelope(final))
+        resp.headers["X-Execution-ID"] = rec["id"]
+        resp.headers["X-Run-ID"] = rec["run_id"]
+        return resp
+
+    @app.post("/api/v1/execute/async/{target}")
+    async def execute_async(target: str, req: Request):
+        try:
+            body = await req.json()
+        except Exception:
+            body = {}
+        rec, err = cp.prepare_execution(target, body, req.headers)
+        if err:
+            return JSONResponse({"error": err["error"]},
+                                status_code=err["status_code"])
+        try:
+            cp._async_q.put_nowait(rec)
+        except asyncio.QueueFull:
+            cp.metrics.backpressure.inc()
+            cp.complete_execution(rec["id"], st.FAILED,
+                                  error="async queue is full")
+            return JSONResponse({"error": "queue is full"}, status_code=503)
+        cp.metrics.queue_depth.set(cp._async_q.qsize())
+        return JSONResponse({
+            "execution_id": rec["id"], "run_id": rec["run_id"],
+            "workflow_id": rec["run_id"], "status": "queued",
+            "target": target, "type": rec["target_type"],
+            "created_at": time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime()),
+            "enqueued_at": time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime()),
+        }, status_code=202,
+            headers={"X-Execution-ID": rec["id"], "X-Run-ID": rec["run_id"]})
+
+    @app.get("/api/v1/executions/{execution_id}")
+    async def get_execution(execution_id: str):
+        rec = cp.storage.get_execution(execution_id)
+        if rec is None:
+            return JSONResponse({"error": "not found"}, status_code=404)
+        return cp.envelope(rec)
+
+    @app.post("/api/v1/executions/batch-status")
+    async def batch_status(req: Request):
+        body = await req.json()
+        ids = body.get("execution_ids") or body.get("ids") or []
+        found = cp.storage.batch_status(ids)
+        return {"executions": {i: cp.envelope(found[i]) for i in found},
+                "missing": [i for i in ids if i not in found]}
+
+    @app.post("/api/v1/executions/{execution_id}/status")
+    async def execution_status_callback(execution_id: str, req: Request):
+        """Agent-side terminal/progress status ingestion (A.2)."""
+        body = await req.json()
+        status = st.normalize(body.get("status", ""))
+        if status not in (st.RUNNING, st.PENDING) and not st.is_terminal(status):
+            return JSONResponse({"error": f"unknown status '{body.get('status')}'"},
+                                status_code=400)
+        if st.is_terminal(status):
+            cp.complete_execution(execution_id, status, body.get("result"),
+                                  body.get("error"), body.get("duration_ms"))
+        return {"status": "ok"}
+
+    # ------------------------------------------------- workflow events/DAG
+    @app.post("/api/v1/workflow/executions/events")
+    async def workflow_events(req: Request):
+        ev = await req.json()
+        if not ev.get("execution_id"):
+            return JSONResponse({"error": "execution_id required"},
+                                status_code=400)
+        cp.storage.upsert_workflow_event(ev)
+        run_id = ev.get("run_id") or ev.get("workflow_id")
+        if run_id:
+            sibs = cp.storage.executions_by_run(run_id)
+            cp.storage.upsert_run(run_id,
+                                  aggregate_status([e["status"] for e in sibs]))
+        return {"status": "ok"}
+
+    # -------------------------------------------------------------- memory
+    def _scope(body: dict, headers) -> tuple[str, str]:
+        """Scope resolution (A.5): explicit wins, else header priority."""
+        if body.get("scope"):
+            return body["scope"], body.get("scope_id") or "global"
+        for hdr, scope in (("x-workflow-id", "workflow"),
+                           ("x-run-id", "workflow"),
+                           ("x-session-id", "session"),
+                           ("x-actor-id", "actor")):
+            v = headers.get(hdr)
+            if v:
+                return scope, v
+        return "global", "global"
+
+    @app.post("/api/v1/memory/set")
+    async def memory_set(req: Request):
+        body = await req.json()
+        scope, sid = _scope(body, req.headers)
+        cp.storage.memory_set(scope, sid, body["key"], body.get("value"))
+        cp.buses.memory.publish({"op": "set", "scope": scope, "scope_id": sid,
+                                 "key": body["key"], "value": body.get("value"),
+                                 "at": time.time()})
+        return {"status": "ok", "scope": scope, "scope_id": sid}
+
+    @app.post("/api/v1/memory/get")
+    async def memory_get(req: Request):
+        body = await req.json()
+        scope, sid = _scope(body, req.headers)
+        value = cp.storage.memory_get(scope, sid, body["key"])
+        return {"key": body["key"], "value": value, "found": value is not None,
+                "scope": scope, "scope_id": sid}
+
+    @app.post("/api/v1/memory/delete")
+    async def memory_delete(req: Request):
+        body = await req.json()
+        scope, sid = _scope(body, req.headers)
+        deleted = cp.storage.memory_delete(scope, sid, body["key"])
+        if deleted:
+            cp.buses.memory.publish({"op": "delete", "scope": scope,
+                                     "scope_id": sid, "key": body["key"],
+                                     "at": time.time()})
+        return {"deleted": deleted}
+
+    @app.get("/api/v1/memory/list")
+    async def memory_list(req: Request):
+        scope, sid = _scope(dict(req.query_params), req.headers)
+        keys = cp.storage.memory_list(scope, sid,
+                                      req.query_params.get("prefix", ""))
+        return {"keys": keys, "scope": scope, "scope_id": sid}
+
+    @app.post("/api/v1/memory/vector/set")
+    async def vector_set(req: Request):
+        body = await req.json()
+        scope, sid = _scope(body, req.headers)
+        cp.storage.vector_set(scope, sid, body["key"], body["embedding"],
+                              body.get("metadata"))
+        return {"status": "ok"}
+
+    @app.post("/api/v1/memory/vector/search")
+    async def vector_search(req: Request):
+        body = await req.json()
+        scope, sid = _scope(body, req.headers)
+        res = cp.storage.vector_search(scope, sid, body["embedding"],
+                                       body.get("top_k", 5),
+                                       body.get("metric", "cosine"),
+                                       body.get("filters"))
+        return {"results": res}
+
+    @app.post("/api/v1/memory/vector/delete")
+    async def vector_delete(req: Request):
+        body = await req.json()
+        scope, sid = _scope(body, req.headers)
+        return {"deleted": cp.storage.vector_delete(scope, sid, body["key"])}
+
+    @app.get("/api/v1/memory/events/history")
+    async def memory_history(req: Request):
+        since = float(req.query_params.get("since", 0))
+        return {"events": cp.storage.memory_events_since(
+            since, req.query_params.get("scope"))}
+
+    @app.websocket("/api/v1/memory/events/ws")
+    async def memory_ws(ws: WebSocket):
+        await ws.accept()
+        sid, q = cp.buses.memory.subscribe()
+        try:
+            while True:
+                ev = await q.get()
+                await ws.send_json(ev)
+        except (WebSocketDisconnect, Exception):
+            pass
+        finally:
+            cp.buses.memory.unsubscribe(sid)
+
+    @app.get("/api/v1/memory/events/sse")
+    async def memory_sse():
+        sid, q = cp.buses.memory.subscribe()
+
+        async def gen():
+            try:
+                while True:
+                    try:
+                        ev = await asyncio.wait_for(q.get(), 15.0)
+                        yield _sse(ev)
+                    except asyncio.TimeoutError:
+                        yield ": keepalive\n\n"
+            finally:
+                cp.buses.memory.unsubscribe(sid)
+        return StreamingResponse(gen(), media_type="text/event-stream")
+
+    # ------------------------------------------------------------- DID/VC
+    @app.post("/api/v1/did/register")
+    async def did_register(req: Request):
+        if not cp.dids:
+            return JSONResponse({"error": "DID disabled"}, status_code=400)
+        body = await req.json()
+        node_id = body.get("node_id") or body.get("agent_id")
+        identity = cp.dids.register_node(
+            node_id, body.get("reasoners", []), body.get("skills", []))
+        return {"status": "ok", "identity": identity}
+
+    @app.get("/api/v1/did/resolve/{did}")
+    async def did_resolve(did: str):
+        doc = cp.dids.resolve(did) if cp.dids else None
+        if doc is None:
+            return JSONResponse({"error": "not found"}, status_code=404)
+        return {"did": did, "document": doc}
+
+    @app.get("/api/v1/did/document/{did}")
+    async def did_document_ep(did: str):
+        doc = cp.dids.resolve(did) if cp.dids else None
+        if doc is None:
+            return JSONResponse({"error": "not found"}, status_code=404)
+        return doc
+
+    @app.get("/api/v1/did/status")
+    async def did_status():
+        if not cp.dids:
+            return {"enabled": False}
+        return {"enabled": True, "root_did": cp.dids.root_did,
+                "agents": len(cp.storage.list_dids("agent")),
+                "components": len(cp.storage.list_dids("component"))}
+
+    @app.post("/api/v1/did/verify")
+    async def did_verify(req: Request):
+        body = await req.json()
+        return VCService.verify_document(body.get("credential") or body)
+
+    @app.post("/api/v1/execution/vc")
+    async def create_execution_vc(req: Request):
+        if not cp.vcs:
+            return JSONResponse({"error": "DID disabled"}, status_code=400)
+        body = await req.json()
+        rec = cp.storage.get_execution(body.get("execution_id", ""))
+        if rec is None:
+            return JSONResponse({"error": "execution not found"},
+                                status_code=404)
+        doc = cp.vcs.issue_execution_vc(rec, body.get("caller_did"),
+                                        body.get("target_did"))
+        return {"status": "ok", "vc": doc}
+
+    @app.get("/api/v1/executions/{execution_id}/vc")
+    async def get_execution_vc(execution_id: str):
+        rec = cp.storage.vc_for_execution(execution_id)
+        if rec is None:
+            return JSONResponse({"error": "not found"}, status_code=404)
+        return {"vc": rec["document"],
+                "verification": cp.vcs.verify_execution(execution_id)}
+
+    @app.get("/api/v1/did/workflow/{run_id}/vc-chain")
+    async def vc_chain(run_id: str):
+        if not cp.vcs:
+            return JSONResponse({"error": "DID disabled"}, status_code=400)
+        return cp.vcs.workflow_chain(run_id)
+
+    @app.get("/api/v1/did/export/vcs")
+    async def export_vcs(req: Request):
+        run_id = req.query_params.get("workflow_id")
+        vcs = cp.storage.vcs_for_run(run_id) if run_id else []
+        return {"credentials": [v["document"] for v in vcs]}
+
+    # ------------------------------------------------------------- UI API
+    @app.get("/api/ui/v1/dashboard/summary")
+    async def dashboard():
+        nodes = cp.storage.list_nodes()
+        recents = cp.storage.list_executions(limit=200)
+        by_status: dict[str, int] = {}
+        for e in recents:
+            by_status[e["status"]] = by_status.get(e["status"], 0) + 1
+        return {
+            "nodes": {"total": len(nodes),
+                      "active": sum(1 for n in nodes if n["status"] == "active")},
+            "executions": {"recent": len(recents), "by_status": by_status},
+            "uptime_s": time.time() - cp.started_at,
+        }
+
+    @app.get("/api/ui/v1/nodes")
+    async def ui_nodes():
+        return {"nodes": cp.storage.list_nodes()}
+
+    @app.get("/api/ui/v1/executions")
+    async def ui_executions(req: Request):
+        return {"executions": cp.storage.list_executions(
+            limit=int(req.query_params.get("limit", 100)),
+            node_id=req.query_params.get("node_id"),
+            status=req.query_params.get("status"))}
+
+    @app.get("/api/ui/v1/reasoners")
+    async def ui_reasoners():
+        out = []
+        for n in cp.storage.list_nodes():
+            for r in n.get("reasoners", []):
+                out.append({"node_id": n["id"], **(r if isinstance(r, dict)
+                                                  else {"id": r})})
+        return {"reasoners": out}
+
+    @app.get("/api/ui/v1/executions/events")
+    async def execution_sse():
+        sid, q = cp.buses.execution.subscribe()
+
+        async def gen():
+            try:
+                while True:
+                    try:
+                        ev = await asyncio.wait_for(q.get(), 15.0)
+                        yield _sse(ev)
+                    except asyncio.TimeoutError:
+                        yield ": keepalive\n\n"
+            finally:
+                cp.buses.execution.unsubscribe(sid)
+        return StreamingResponse(gen(), media_type="text/event-stream")
+
+    @app.get("/api/ui/v1/nodes/events")
+    async def node_sse():
+        sid, q = cp.buses.node.subscribe()
+
+        async def gen():
+            try:
+                while True:
+                    try:
+                        ev = await asyncio.wait_for(q.get(), 15.0)
+                        yield _sse(ev)
+                    except asyncio.TimeoutError:
+                        yield ": keepalive\n\n"
+            finally:
+                cp.buses.node.unsubscribe(sid)
+        return StreamingResponse(gen(), media_type="text/event-stream")
+
+    @app.get("/api/ui/v1/workflows/{run_id}/dag")
+    async def workflow_dag(run_id: str, req: Request):
+        execs = cp.storage.executions_by_run(run_id)
+        if not execs:
+            return JSONResponse({"error": "not found"}, status_code=404)
+        light = req.query_params.get("lightweight") in ("1", "true")
+        return build_dag(execs, lightweight=light)
+
+    @app.get("/api/ui/v2/workflow-runs")
+    async def workflow_runs(req: Request):
+        return {"runs": cp.storage.list_runs(
+            int(req.query_params.get("limit", 50)))}
+
+    @app.get("/api/ui/v2/workflow-runs/{run_id}")
+    async def workflow_run(run_id: str):
+        run = cp.storage.get_run(run_id)
+        if run is None:
+            return JSONResponse({"error": "not found"}, status_code=404)
+        execs = cp.storage.executions_by_run(run_id)
+        return {"run": run, "dag": build_dag(execs, lightweight=True)}
+
+    return app

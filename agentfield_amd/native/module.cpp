@@ -1,0 +1,252 @@
+// agentfield_amd._native — CPU-side native runtime components:
+//   * Ed25519 keygen/sign/verify via OpenSSL libcrypto (DID/VC audit path;
+//     reference does this in Go crypto/ed25519 — SURVEY.md C21/C22)
+//   * the continuous-batching scheduler + KV page allocator (the production
+//     counterpart of engine/scheduler.py, which stays as the tested oracle)
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <openssl/evp.h>
+
+#include <deque>
+#include <stdexcept>
+#include <string>
+#include <unordered_map>
+#include <vector>
+
+namespace py = pybind11;
+
+// ------------------------------- ed25519 ----------------------------------
+static py::bytes ed25519_pubkey(py::bytes seed) {
+  std::string s = seed;
+  if (s.size() != 32) throw std::invalid_argument("seed must be 32 bytes");
+  EVP_PKEY* k = EVP_PKEY_new_raw_private_key(EVP_PKEY_ED25519, nullptr,
+                                             (const unsigned char*)s.data(), 32);
+  if (!k) throw std::runtime_error("ed25519 key import failed");
+  unsigned char pub[32];
+  size_t plen = 32;
+  EVP_PKEY_get_raw_public_key(k, pub, &plen);
+  EVP_PKEY_free(k);
+  return py::bytes((const char*)pub, plen);
+}
+
+static py::bytes ed25519_sign(py::bytes seed, py::bytes msg) {
+  std::string s = seed, m = msg;
+  if (s.size() != 32) throw std::invalid_argument("seed must be 32 bytes");
+  EVP_PKEY* k = EVP_PKEY_new_raw_private_key(EVP_PKEY_ED25519, nullptr,
+                                             (const unsigned char*)s.data(), 32);
+  if (!k) throw std::runtime_error("ed25519 key import failed");
+  EVP_MD_CTX* ctx = EVP_MD_CTX_new();
+  unsigned char sig[64];
+  size_t slen = 64;
+  int ok = EVP_DigestSignInit(ctx, nullptr, nullptr, nullptr, k) == 1 &&
+           EVP_DigestSign(ctx, sig, &slen, (const unsigned char*)m.data(),
+                          m.size()) == 1;
+  EVP_MD_CTX_free(ctx);
+  EVP_PKEY_free(k);
+  if (!ok) throw std::runtime_error("ed25519 sign failed");
+  return py::bytes((const char*)sig, slen);
+}
+
+static bool ed25519_verify(py::bytes pub, py::bytes msg, py::bytes sig) {
+  std::string p = pub, m = msg, g = sig;
+  if (p.size() != 32 || g.size() != 64) return false;
+  EVP_PKEY* k = EVP_PKEY_new_raw_public_key(EVP_PKEY_ED25519, nullptr,
+                                            (const unsigned char*)p.data(), 32);
+  if (!k) return false;
+  EVP_MD_CTX* ctx = EVP_MD_CTX_new();
+  bool ok = EVP_DigestVerifyInit(ctx, nullptr, nullptr, nullptr, k) == 1 &&
+            EVP_DigestVerify(ctx, (const unsigned char*)g.data(), g.size(),
+                             (const unsigned char*)m.data(), m.size()) == 1;
+  EVP_MD_CTX_free(ctx);
+  EVP_PKEY_free(k);
+  return ok;
+}
+
+// ------------------------- scheduler / allocator ---------------------------
+// Semantics mirror engine/scheduler.py exactly (it is the tested oracle).
+struct SeqState {
+  int prompt_len = 0;
+  int num_tokens = 0;   // prompt + generated
+  std::vector<int> pages;
+};
+
+struct ScheduleResult {
+  bool has_work = false;
+  bool is_prefill = false;
+  std::vector<long> seq_ids;
+  std::vector<long> preempted;  // seqs whose generated tokens must be dropped
+};
+
+class NativeScheduler {
+ public:
+  NativeScheduler(int max_num_seqs, int max_prefill_tokens, int page_size,
+                  int num_pages, int max_waiting)
+      : max_num_seqs_(max_num_seqs), max_prefill_tokens_(max_prefill_tokens),
+        page_size_(page_size), num_pages_(num_pages), max_waiting_(max_waiting) {
+    for (int p = num_pages - 1; p >= 1; --p) free_list_.push_back(p);
+  }
+
+  bool add(long seq_id, int prompt_len) {
+    if ((int)waiting_.size() >= max_waiting_) return false;
+    SeqState st;
+    st.prompt_len = prompt_len;
+    st.num_tokens = prompt_len;
+    seqs_[seq_id] = st;
+    waiting_.push_back(seq_id);
+    return true;
+  }
+
+  int num_free() const { return (int)free_list_.size(); }
+  int num_queued() const { return (int)waiting_.size(); }
+  int num_running() const { return (int)running_.size(); }
+  long n_preempted() const { return n_preempted_; }
+  bool has_work() const { return !waiting_.empty() || !running_.empty(); }
+
+  std::vector<int> pages(long seq_id) const {
+    auto it = seqs_.find(seq_id);
+    return it == seqs_.end() ? std::vector<int>{} : it->second.pages;
+  }
+
+  int num_tokens(long seq_id) const {
+    auto it = seqs_.find(seq_id);
+    return it == seqs_.end() ? 0 : it->second.num_tokens;
+  }
+
+  void note_token(long seq_id) {  // engine appended one generated token
+    auto it = seqs_.find(seq_id);
+    if (it != seqs_.end()) it->second.num_tokens += 1;
+  }
+
+  void finish(long seq_id) {
+    auto it = seqs_.find(seq_id);
+    if (it == seqs_.end()) return;
+    release(it->second);
+    for (size_t i = 0; i < running_.size(); ++i)
+      if (running_[i] == seq_id) { running_.erase(running_.begin() + i); break; }
+    seqs_.erase(it);
+  }
+
+  ScheduleResult schedule() {
+    ScheduleResult r;
+    // 1) admit prefills
+    int tokens = 0;
+    std::vector<long> batch;
+    while (!waiting_.empty() &&
+           (int)(running_.size() + batch.size()) < max_num_seqs_) {
+      long cand = waiting_.front();
+      SeqState& st = seqs_[cand];
+      int ntok = st.prompt_len;
+      if (!batch.empty() && tokens + ntok > max_prefill_tokens_) break;
+      int need = pages_needed(ntok);
+      if (need > (int)free_list_.size()) break;
+      waiting_.pop_front();
+      alloc_into(st, need);
+      batch.push_back(cand);
+      tokens += ntok;
+    }
+    if (!batch.empty()) {
+      for (long s : batch) running_.push_back(s);
+      r.has_work = true;
+      r.is_prefill = true;
+      r.seq_ids = batch;
+      return r;
+    }
+    // 2) decode with growth + preemption
+    if (running_.empty()) return r;
+    size_t i = 0;
+    while (i < running_.size()) {
+      long sid = running_[i];
+      SeqState& st = seqs_[sid];
+      while (!grow(st)) {
+        long victim = running_.back();
+        if (victim == sid) {
+          running_.pop_back();
+          preempt(sid, r);
+          // i stays (element shifted out)
+          goto next_outer;
+        }
+        running_.pop_back();
+        preempt(victim, r);
+      }
+      ++i;
+    next_outer:;
+    }
+    if (running_.empty()) return r;
+    r.has_work = true;
+    r.is_prefill = false;
+    r.seq_ids = running_;
+    return r;
+  }
+
+ private:
+  int pages_needed(int ntok) const {
+    return (ntok + page_size_ - 1) / page_size_;
+  }
+
+  void alloc_into(SeqState& st, int n) {
+    for (int j = 0; j < n; ++j) {
+      st.pages.push_back(free_list_.back());
+      free_list_.pop_back();
+    }
+  }
+
+  bool grow(SeqState& st) {
+    int need = pages_needed(st.num_tokens + 1);
+    if (need > (int)st.pages.size()) {
+      if (free_list_.empty()) return false;
+      alloc_into(st, 1);
+    }
+    return true;
+  }
+
+  void release(SeqState& st) {
+    for (int p : st.pages) free_list_.push_back(p);
+    st.pages.clear();
+  }
+
+  void preempt(long sid, ScheduleResult& r) {
+    SeqState& st = seqs_[sid];
+    release(st);
+    st.num_tokens = st.prompt_len;  // recompute from scratch
+    waiting_.push_front(sid);
+    n_preempted_ += 1;
+    r.preempted.push_back(sid);
+  }
+
+  int max_num_seqs_, max_prefill_tokens_, page_size_, num_pages_, max_waiting_;
+  std::vector<int> free_list_;
+  std::deque<long> waiting_;
+  std::vector<long> running_;
+  std::unordered_map<long, SeqState> seqs_;
+  long n_preempted_ = 0;
+};
+
+PYBIND11_MODULE(_native, m) {
+  m.doc() = "agentfield_amd native runtime (ed25519 via libcrypto, scheduler)";
+  m.def("ed25519_pubkey", &ed25519_pubkey);
+  m.def("ed25519_sign", &ed25519_sign);
+  m.def("ed25519_verify", &ed25519_verify);
+
+  py::class_<ScheduleResult>(m, "ScheduleResult")
+      .def_readonly("has_work", &ScheduleResult::has_work)
+      .def_readonly("is_prefill", &ScheduleResult::is_prefill)
+      .def_readonly("seq_ids", &ScheduleResult::seq_ids)
+      .def_readonly("preempted", &ScheduleResult::preempted);
+
+  py::class_<NativeScheduler>(m, "NativeScheduler")
+      .def(py::init<int, int, int, int, int>(), py::arg("max_num_seqs"),
+           py::arg("max_prefill_tokens"), py::arg("page_size"),
+           py::arg("num_pages"), py::arg("max_waiting") = 4096)
+      .def("add", &NativeScheduler::add)
+      .def("schedule", &NativeScheduler::schedule)
+      .def("finish", &NativeScheduler::finish)
+      .def("note_token", &NativeScheduler::note_token)
+      .def("pages", &NativeScheduler::pages)
+      .def("num_tokens", &NativeScheduler::num_tokens)
+      .def("num_free", &NativeScheduler::num_free)
+      .def("num_queued", &NativeScheduler::num_queued)
+      .def("num_running", &NativeScheduler::num_running)
+      .def("n_preempted", &NativeScheduler::n_preempted)
+      .def("has_work", &NativeScheduler::has_work);
+}

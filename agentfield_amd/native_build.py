@@ -1,0 +1,32 @@
+"""Build agentfield_amd._native (pybind11 + libcrypto) in-tree."""
+from __future__ import annotations
+
+import subprocess
+import sys
+import sysconfig
+from pathlib import Path
+
+PKG = Path(__file__).resolve().parent
+SRC = PKG / "native" / "module.cpp"
+EXT = sysconfig.get_config_var("EXT_SUFFIX") or ".so"
+OUT = PKG / f"_native{EXT}"
+
+
+def build(force: bool = False, verbose: bool = True) -> Path:
+    if not force and OUT.exists() and OUT.stat().st_mtime > SRC.stat().st_mtime:
+        return OUT
+    import pybind11
+    cmd = [
+        "g++", "-O2", "-shared", "-fPIC", "-std=c++17",
+        f"-I{pybind11.get_include()}",
+        f"-I{sysconfig.get_paths()['include']}",
+        str(SRC), "-lcrypto", "-o", str(OUT),
+    ]
+    if verbose:
+        print("[native]", " ".join(cmd), file=sys.stderr)
+    subprocess.run(cmd, check=True)
+    return OUT
+
+
+if __name__ == "__main__":
+    build(force="--force" in sys.argv)

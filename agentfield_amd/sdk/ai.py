@@ -1,0 +1,243 @@
+"""app.ai() — the model seam, served by the in-process MI355X engine.
+
+Where the reference forwards app.ai() to external providers through LiteLLM
+(everything below get_litellm_params is replaced), here the
+call goes: tokenize -> continuous-batching engine (hipGraph decode loop on
+the local GPU) -> detokenize.  The public signature and the hierarchical
+config merge (agent defaults < method < runtime) match the reference.
+"""
+from __future__ import annotations
+
+import json
+import os
+import queue
+import threading
+from dataclasses import dataclass, field, replace
+
+from ..engine import LLMEngine, SamplingParams
+from ..models import CONFIGS
+
+
+@dataclass
+class AIConfig:
+    model: str = "llama-3-8b"
+    temperature: float = 0.0
+    max_tokens: int = 256
+    system_prompt: str | None = None
+    stop: tuple = ()
+    timeout: float = 600.0
+    device: str | None = None           # None -> cuda if available
+    fallback_models: tuple = ()
+    extra: dict = field(default_factory=dict)
+
+    def merged(self, **overrides) -> "AIConfig":
+        clean = {k: v for k, v in overrides.items() if v is not None}
+        extra = {**self.extra, **clean.pop("extra", {})}
+        known = {k: v for k, v in clean.items() if hasattr(self, k)}
+        return replace(self, extra=extra, **known)
+
+
+class ByteTokenizer:
+    """Offline-safe default tokenizer: UTF-8 bytes + specials.  Real
+    deployments point AGENTFIELD_TOKENIZER at a HF tokenizer directory."""
+    BOS, EOS, PAD = 1, 2, 3
+    OFFSET = 4
+
+    def __init__(self, vocab_size: int = 128256):
+        self.vocab_size = vocab_size
+
+    @property
+    def eos_id(self) -> int:
+        return self.EOS
+
+    def encode(self, text: str, add_bos: bool = True) -> list[int]:
+        ids = [b + self.OFFSET for b in text.encode("utf-8")]
+        return ([self.BOS] if add_bos else []) + ids
+
+    def decode(self, ids: list[int]) -> str:
+        data = bytes(i - self.OFFSET for i in ids
+                     if self.OFFSET <= i < self.OFFSET + 256)
+        return data.decode("utf-8", errors="replace")
+
+
+def load_tokenizer(path: str | None = None):
+    path = path or os.environ.get("AGENTFIELD_TOKENIZER")
+    if path:
+        from transformers import AutoTokenizer
+
+        class _HF:
+            def __init__(self, tok):
+                self.tok = tok
+                self.eos_id = tok.eos_token_id
+
+            def encode(self, text, add_bos=True):
+                return self.tok.encode(text)
+
+            def decode(self, ids):
+                return self.tok.decode(ids, skip_special_tokens=True)
+        return _HF(AutoTokenizer.from_pretrained(path))
+    return ByteTokenizer()
+
+
+class EngineRunner:
+    """Single driver thread stepping one engine; ai() calls submit prompts
+    and block on a per-request event.  This is the in-process counterpart of
+    the control plane's async queue feeding the scheduler."""
+
+    def __init__(self, engine: LLMEngine, tokenizer=None):
+        self.engine = engine
+        self.tokenizer = tokenizer or ByteTokenizer(engine.cfg.vocab_size)
+        self._submit: queue.Queue = queue.Queue()
+        self._waiters: dict[int, dict] = {}
+        self._lock = threading.Lock()
+        self._wake = threading.Event()
+        self._stop = False
+        self._thread = threading.Thread(target=self._loop, daemon=True,
+                                        name="af-engine")
+        self._thread.start()
+
+    def submit(self, prompt_ids: list[int], sampling: SamplingParams,
+               stream_q: queue.Queue | None = None) -> dict:
+        waiter = {"done": threading.Event(), "output": None, "error": None,
+                  "stream": stream_q}
+        self._submit.put((prompt_ids, sampling, waiter))
+        self._wake.set()
+        return waiter
+
+    def generate_text(self, prompt: str, cfg: AIConfig) -> str:
+        ids = self.tokenizer.encode(prompt)
+        limit = self.engine.cfg.max_position - 8
+        if len(ids) + cfg.max_tokens > limit:
+            ids = ids[-(limit - cfg.max_tokens):]  # token-aware trim
+        sp = SamplingParams(max_tokens=cfg.max_tokens,
+                            temperature=cfg.temperature)
+        w = self.submit(ids, sp)
+        if not w["done"].wait(cfg.timeout):
+            raise TimeoutError("engine generate timed out")
+        if w["error"]:
+            raise RuntimeError(w["error"])
+        return self.tokenizer.decode(w["output"])
+
+    def stream_text(self, prompt: str, cfg: AIConfig):
+        ids = self.tokenizer.encode(prompt)
+        sp = SamplingParams(max_tokens=cfg.max_tokens,
+                            temperature=cfg.temperature)
+        sq: queue.Queue = queue.Queue()
+        self.submit(ids, sp, stream_q=sq)
+        while True:
+            tok, done = sq.get(timeout=cfg.timeout)
+            if tok is not None:
+                yield self.tokenizer.decode([tok])
+            if done:
+                return
+
+    def _loop(self):
+        eng = self.engine
+        pending: dict[int, dict] = {}
+        while not self._stop:
+            moved = False
+            while True:
+                try:
+                    ids, sp, waiter = self._submit.get_nowait()
+                except queue.Empty:
+                    break
+                rid = eng.add_request(ids, sp)
+                if rid is None:
+                    waiter["error"] = "engine queue full"
+                    waiter["done"].set()
+                else:
+                    pending[rid] = waiter
+                moved = True
+            if eng.has_work():
+                for (rid, tok, done) in eng.step():
+                    w = pending.get(rid)
+                    if w and w["stream"] is not None:
+                        w["stream"].put((tok, done))
+                moved = True
+                for rid in list(pending):
+                    fin = eng.get_finished(rid)
+                    if fin is not None:
+                        w = pending.pop(rid)
+                        w["output"] = fin.output_ids
+                        w["done"].set()
+            if not moved:
+                self._wake.wait(0.005)
+                self._wake.clear()
+
+    def shutdown(self):
+        self._stop = True
+        self._wake.set()
+
+
+_runners: dict[str, EngineRunner] = {}
+_runners_lock = threading.Lock()
+
+
+def get_runner(cfg: AIConfig) -> EngineRunner:
+    """Process-wide engine registry: one engine per model name."""
+    import torch
+    key = cfg.model
+    with _runners_lock:
+        if key in _runners:
+            return _runners[key]
+        device = cfg.device or ("cuda" if torch.cuda.is_available() else "cpu")
+        model_cfg = CONFIGS[cfg.model]
+        kw = {}
+        if device == "cpu":
+            kw = {"num_pages": 512, "max_num_seqs": 8, "enable_graphs": False,
+                  "dtype": torch.float32}
+        eng = LLMEngine(model_cfg, device=device, **kw)
+        runner = EngineRunner(eng, load_tokenizer())
+        _runners[key] = runner
+        return runner
+
+
+def set_runner(model: str, runner: EngineRunner) -> None:
+    with _runners_lock:
+        _runners[model] = runner
+
+
+class AgentAI:
+    """The ai() callable bound to an Agent (hierarchical config merge)."""
+
+    def __init__(self, default_config: AIConfig | None = None):
+        self.config = default_config or AIConfig(
+            model=os.environ.get("AGENTFIELD_AI_MODEL", "llama-3-8b"))
+
+    def __call__(self, *prompt_parts, system: str | None = None,
+                 user: str | None = None, schema=None, stream: bool = False,
+                 **overrides):
+        cfg = self.config.merged(**overrides)
+        parts = []
+        sys_p = system or cfg.system_prompt
+        if sys_p:
+            parts.append(f"<|system|>\n{sys_p}")
+        body = user if user is not None else "\n".join(str(p) for p in prompt_parts)
+        if schema is not None:
+            schema_json = (schema.model_json_schema()
+                           if hasattr(schema, "model_json_schema") else schema)
+            parts.append(f"<|system|>\nRespond ONLY with JSON matching this "
+                         f"schema:\n{json.dumps(schema_json)}")
+        parts.append(f"<|user|>\n{body}\n<|assistant|>\n")
+        prompt = "\n".join(parts)
+        runner = get_runner(cfg)
+        if stream:
+            return runner.stream_text(prompt, cfg)
+        text = runner.generate_text(prompt, cfg)
+        if schema is not None:
+            try:
+                data = json.loads(text)
+                if hasattr(schema, "model_validate"):
+                    return schema.model_validate(data)
+                return data
+            except (ValueError, TypeError):
+                return text  # schema validation failed; return raw text
+        return text
+
+    def with_audio(self, *a, **kw):
+        raise NotImplementedError(
+            "multimodal inputs need a multimodal model; the Llama family "
+            "configs shipped with this framework are text-only")
+
+    with_vision = with_audio
+    with_multimodal = with_audio

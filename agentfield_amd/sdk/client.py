@@ -1,0 +1,150 @@
+"""Control-plane REST client used by the SDK (reference parity: client.py
+AgentFieldClient + async_execution_manager.py in compact form: submit,
+sync execute, batch-status polling with adaptive backoff, SSE nudge)."""
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+
+import httpx
+
+
+class AgentFieldClient:
+    def __init__(self, base_url: str, timeout: float = 95.0):
+        self.base_url = base_url.rstrip("/")
+        self.timeout = timeout
+        self._client = httpx.Client(timeout=timeout)
+        self._aclient: httpx.AsyncClient | None = None
+
+    # -------------------------------------------------------- registration
+    def register_agent(self, node: dict) -> dict:
+        r = self._client.post(f"{self.base_url}/api/v1/nodes/register",
+                              json=node)
+        r.raise_for_status()
+        return r.json()
+
+    def heartbeat(self, node_id: str, payload: dict | None = None) -> bool:
+        try:
+            r = self._client.post(
+                f"{self.base_url}/api/v1/nodes/{node_id}/heartbeat",
+                json=payload or {"status": "active"})
+            return r.status_code == 200
+        except httpx.HTTPError:
+            return False
+
+    def update_node_status(self, node_id: str, status: str) -> bool:
+        try:
+            r = self._client.post(
+                f"{self.base_url}/api/v1/nodes/{node_id}/status",
+                json={"status": status})
+            return r.status_code == 200
+        except httpx.HTTPError:
+            return False
+
+    # ----------------------------------------------------------- execution
+    def execute_sync(self, target: str, input: dict,
+                     headers: dict | None = None,
+                     webhook: dict | None = None) -> dict:
+        body = {"input": input}
+        if webhook:
+            body["webhook"] = webhook
+        r = self._client.post(f"{self.base_url}/api/v1/execute/{target}",
+                              json=body, headers=headers or {})
+        r.raise_for_status()
+        return r.json()
+
+    def execute_async(self, target: str, input: dict,
+                      headers: dict | None = None,
+                      webhook: dict | None = None) -> dict:
+        body = {"input": input}
+        if webhook:
+            body["webhook"] = webhook
+        r = self._client.post(f"{self.base_url}/api/v1/execute/async/{target}",
+                              json=body, headers=headers or {})
+        if r.status_code == 503:
+            raise RuntimeError("control plane queue is full (backpressure)")
+        r.raise_for_status()
+        return r.json()
+
+    def get_execution(self, execution_id: str) -> dict | None:
+        r = self._client.get(f"{self.base_url}/api/v1/executions/{execution_id}")
+        return r.json() if r.status_code == 200 else None
+
+    def batch_status(self, ids: list[str]) -> dict:
+        r = self._client.post(f"{self.base_url}/api/v1/executions/batch-status",
+                              json={"execution_ids": ids})
+        r.raise_for_status()
+        return r.json()
+
+    def wait_for_result(self, execution_id: str, timeout: float = 300.0,
+                        poll_initial: float = 0.05, poll_max: float = 2.0) -> dict:
+        """Adaptive polling (age-based backoff, reference async_config.py)."""
+        deadline = time.time() + timeout
+        poll = poll_initial
+        while time.time() < deadline:
+            rec = self.get_execution(execution_id)
+            if rec and rec.get("status") in ("completed", "failed", "timeout",
+                                             "cancelled"):
+                return rec
+            time.sleep(poll)
+            poll = min(poll * 1.5, poll_max)
+        raise TimeoutError(f"execution {execution_id} did not finish")
+
+    def report_status(self, execution_id: str, status: str, result=None,
+                      error: str | None = None,
+                      duration_ms: float | None = None) -> bool:
+        try:
+            r = self._client.post(
+                f"{self.base_url}/api/v1/executions/{execution_id}/status",
+                json={"execution_id": execution_id, "status": status,
+                      "result": result, "error": error,
+                      "duration_ms": duration_ms})
+            return r.status_code == 200
+        except httpx.HTTPError:
+            return False
+
+    def workflow_event(self, event: dict) -> bool:
+        try:
+            r = self._client.post(
+                f"{self.base_url}/api/v1/workflow/executions/events", json=event)
+            return r.status_code == 200
+        except httpx.HTTPError:
+            return False
+
+    # -------------------------------------------------------------- memory
+    def memory_op(self, op: str, body: dict, headers: dict | None = None):
+        r = self._client.post(f"{self.base_url}/api/v1/memory/{op}", json=body,
+                              headers=headers or {})
+        r.raise_for_status()
+        return r.json()
+
+    def memory_list(self, params: dict, headers: dict | None = None):
+        r = self._client.get(f"{self.base_url}/api/v1/memory/list",
+                             params=params, headers=headers or {})
+        r.raise_for_status()
+        return r.json()
+
+    # --------------------------------------------------------------- DID/VC
+    def did_register(self, node_id: str, reasoners: list[str],
+                     skills: list[str]) -> dict:
+        r = self._client.post(f"{self.base_url}/api/v1/did/register",
+                              json={"node_id": node_id, "reasoners": reasoners,
+                                    "skills": skills})
+        r.raise_for_status()
+        return r.json()
+
+    def create_execution_vc(self, execution_id: str, **kw) -> dict:
+        r = self._client.post(f"{self.base_url}/api/v1/execution/vc",
+                              json={"execution_id": execution_id, **kw})
+        r.raise_for_status()
+        return r.json()
+
+    def vc_chain(self, run_id: str) -> dict:
+        r = self._client.get(
+            f"{self.base_url}/api/v1/did/workflow/{run_id}/vc-chain")
+        r.raise_for_status()
+        return r.json()
+
+    def close(self):
+        self._client.close()

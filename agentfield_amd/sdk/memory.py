@@ -1,0 +1,162 @@
+"""Hierarchical shared memory client (reference parity: memory.py — KV +
+vector + change events, scoped global/workflow/session/actor with default
+scope resolved from the current execution context)."""
+from __future__ import annotations
+
+import fnmatch
+import json
+import threading
+import time
+
+from .execution_context import current_context
+
+
+class ScopedMemory:
+    def __init__(self, iface: "MemoryInterface", scope: str,
+                 scope_id: str | None = None):
+        self._iface = iface
+        self._scope = scope
+        self._scope_id = scope_id
+
+    def _body(self, **kw) -> dict:
+        body = dict(kw)
+        body["scope"] = self._scope
+        sid = self._scope_id or self._iface._scope_id_for(self._scope)
+        if sid:
+            body["scope_id"] = sid
+        return body
+
+    def set(self, key: str, value):
+        return self._iface.client.memory_op("set", self._body(key=key, value=value))
+
+    def get(self, key: str, default=None):
+        r = self._iface.client.memory_op("get", self._body(key=key))
+        return r["value"] if r.get("found") else default
+
+    def delete(self, key: str) -> bool:
+        return self._iface.client.memory_op("delete", self._body(key=key))["deleted"]
+
+    def keys(self, prefix: str = "") -> list[str]:
+        body = self._body()
+        return self._iface.client.memory_list(
+            {"prefix": prefix, **body})["keys"]
+
+    # vector ops share the scope
+    def vector_set(self, key: str, embedding, metadata=None):
+        return self._iface.client.memory_op(
+            "vector/set", self._body(key=key, embedding=list(embedding),
+                                     metadata=metadata))
+
+    def vector_search(self, embedding, top_k: int = 5, metric: str = "cosine",
+                      filters=None):
+        return self._iface.client.memory_op(
+            "vector/search", self._body(embedding=list(embedding), top_k=top_k,
+                                        metric=metric, filters=filters))["results"]
+
+    def vector_delete(self, key: str) -> bool:
+        return self._iface.client.memory_op("vector/delete",
+                                            self._body(key=key))["deleted"]
+
+
+class MemoryInterface:
+    """app.memory — default scope follows the current execution context
+    (workflow > session > actor > global)."""
+
+    def __init__(self, client, node_id: str):
+        self.client = client
+        self.node_id = node_id
+        self._watchers: list[tuple[str, callable]] = []
+        self._watch_thread: threading.Thread | None = None
+        self._stop = threading.Event()
+
+    def _scope_id_for(self, scope: str) -> str | None:
+        ctx = current_context()
+        if ctx is None:
+            return None
+        return {"workflow": ctx.run_id, "session": ctx.session_id,
+                "actor": ctx.actor_id}.get(scope)
+
+    def _default_scope(self) -> ScopedMemory:
+        ctx = current_context()
+        if ctx and ctx.run_id:
+            return self.workflow
+        if ctx and ctx.session_id:
+            return self.session
+        if ctx and ctx.actor_id:
+            return self.actor
+        return self.globals
+
+    @property
+    def workflow(self) -> ScopedMemory:
+        return ScopedMemory(self, "workflow")
+
+    @property
+    def session(self) -> ScopedMemory:
+        return ScopedMemory(self, "session")
+
+    @property
+    def actor(self) -> ScopedMemory:
+        return ScopedMemory(self, "actor")
+
+    @property
+    def globals(self) -> ScopedMemory:
+        return ScopedMemory(self, "global", "global")
+
+    # default-scope conveniences
+    def set(self, key: str, value):
+        return self._default_scope().set(key, value)
+
+    def get(self, key: str, default=None):
+        return self._default_scope().get(key, default)
+
+    def delete(self, key: str) -> bool:
+        return self._default_scope().delete(key)
+
+    def keys(self, prefix: str = "") -> list[str]:
+        return self._default_scope().keys(prefix)
+
+    # ------------------------------------------------------ change events
+    def on_change(self, pattern: str = "*"):
+        """Decorator: watch memory keys (glob pattern) via history polling
+        over /api/v1/memory/events/history."""
+        def deco(fn):
+            self._watchers.append((pattern, fn))
+            self._ensure_watch_thread()
+            return fn
+        return deco
+
+    def _ensure_watch_thread(self):
+        if self._watch_thread is not None:
+            return
+        def loop():
+            since = time.time()
+            base = self.client.base_url
+            import httpx
+            while not self._stop.wait(1.0):
+                try:
+                    r = httpx.get(f"{base}/api/v1/memory/events/history",
+                                  params={"since": since}, timeout=5.0)
+                    events = r.json().get("events", [])
+                except Exception:
+                    continue
+                for ev in events:
+                    since = max(since, ev.get("at", since))
+                    for pattern, fn in self._watchers:
+                        if fnmatch.fnmatch(ev.get("key", ""), pattern):
+                            try:
+                                val = ev.get("value")
+                                if isinstance(val, str):
+                                    try:
+                                        val = json.loads(val)
+                                    except ValueError:
+                                        pass
+                                fn({"key": ev.get("key"), "op": ev.get("op"),
+                                    "value": val, "scope": ev.get("scope")})
+                            except Exception:
+                                pass
+        self._watch_thread = threading.Thread(target=loop, daemon=True,
+                                              name="af-memory-watch")
+        self._watch_thread.start()
+
+    def stop(self):
+        self._stop.set()

@@ -1,0 +1,341 @@
+"""End-to-end control-plane tests: real localhost servers, real agent apps
+(the reference's integration-tier pattern)."""
+import hashlib
+import hmac
+import json
+import threading
+import time
+
+import httpx
+import pytest
+from fastapi import FastAPI, Request
+
+from agentfield_amd.controlplane import ControlPlane, create_app
+from agentfield_amd.controlplane.server import Config
+from agentfield_amd.sdk import Agent
+
+from helpers import AppServer, wait_until
+
+
+@pytest.fixture(scope="module")
+def cp_server():
+    cp = ControlPlane(Config(background_services=False, sync_timeout=15.0))
+    srv = AppServer(create_app(cp)).start().wait_healthy()
+    yield srv, cp
+    srv.stop()
+
+
+@pytest.fixture(scope="module")
+def greeting_agent(cp_server):
+    srv, _cp = cp_server
+    app = Agent("greeter", agentfield_url=srv.base_url, auto_register=False)
+
+    @app.reasoner()
+    def greet(name: str):
+        return {"greeting": f"hello {name}"}
+
+    @app.reasoner()
+    def fail_always(x: int):
+        raise ValueError("intentional boom")
+
+    @app.skill()
+    def add(a: int, b: int):
+        return a + b
+
+    @app.reasoner()
+    def relay(name: str):
+        # nested cross-agent call -> DAG edge
+        inner = app.call("greeter.greet", name=name.upper())
+        return {"relayed": inner}
+
+    agent_srv = AppServer(app).start()
+    app.base_url = agent_srv.base_url
+    assert app.register()
+    yield agent_srv, app
+    agent_srv.stop()
+
+
+def test_health_and_metrics(cp_server):
+    srv, _ = cp_server
+    r = httpx.get(srv.base_url + "/api/v1/health")
+    assert r.status_code == 200 and r.json()["status"] == "healthy"
+    m = httpx.get(srv.base_url + "/metrics").text
+    for name in ("agentfield_gateway_queue_depth", "agentfield_worker_inflight",
+                 "agentfield_gateway_backpressure_total",
+                 "agentfield_step_duration_seconds"):
+        assert name in m
+
+
+def test_register_lists_node_and_identity(cp_server, greeting_agent):
+    srv, _ = cp_server
+    _, app = greeting_agent
+    r = httpx.get(srv.base_url + "/api/v1/nodes").json()
+    ids = [n["id"] for n in r["nodes"]]
+    assert "greeter" in ids
+    assert app.identity.get("agent_did", "").startswith("did:key:z")
+    assert "greet" in app.identity.get("reasoner_dids", {})
+
+
+def test_sync_execute_202_callback_path(cp_server, greeting_agent):
+    srv, _ = cp_server
+    r = httpx.post(srv.base_url + "/api/v1/execute/greeter.greet",
+                   json={"input": {"name": "world"}}, timeout=20.0)
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert body["status"] == "completed"
+    assert body["result"] == {"greeting": "hello world"}
+    assert body["execution_id"].startswith("exec_")
+    assert body["run_id"].startswith("run_")
+    assert r.headers["X-Execution-ID"] == body["execution_id"]
+    assert body["duration_ms"] is not None
+
+
+def test_sync_execute_failure(cp_server, greeting_agent):
+    srv, _ = cp_server
+    r = httpx.post(srv.base_url + "/api/v1/execute/greeter.fail_always",
+                   json={"input": {"x": 1}}, timeout=20.0)
+    body = r.json()
+    assert body["status"] == "failed"
+    assert "intentional boom" in (body["error_message"] or "")
+
+
+def test_execute_unknown_node_404(cp_server):
+    srv, _ = cp_server
+    r = httpx.post(srv.base_url + "/api/v1/execute/nosuch.thing",
+                   json={"input": {}})
+    assert r.status_code == 404
+
+
+def test_skill_execution_and_target_type(cp_server, greeting_agent):
+    srv, cp = cp_server
+    r = httpx.post(srv.base_url + "/api/v1/execute/greeter.add",
+                   json={"input": {"a": 2, "b": 40}}, timeout=20.0)
+    body = r.json()
+    assert body["status"] == "completed"
+    assert body["result"] == {"result": 42} or body["result"] == 42
+    rec = cp.storage.get_execution(body["execution_id"])
+    assert rec["target_type"] == "skill"
+
+
+def test_async_execute_with_hmac_webhook(cp_server, greeting_agent):
+    srv, _ = cp_server
+    hits = []
+    hook = FastAPI()
+
+    @hook.post("/hook")
+    async def recv(req: Request):
+        body = await req.body()
+        hits.append((dict(req.headers), body))
+        return {"ok": True}
+
+    hook_srv = AppServer(hook).start()
+    try:
+        r = httpx.post(
+            srv.base_url + "/api/v1/execute/async/greeter.greet",
+            json={"input": {"name": "async"},
+                  "webhook": {"url": hook_srv.base_url + "/hook",
+                              "secret": "s3cret",
+                              "headers": {"X-Custom": "yes"}}})
+        assert r.status_code == 202
+        body = r.json()
+        assert body["status"] == "queued" and body["type"] == "reasoner"
+        assert body["workflow_id"] == body["run_id"]
+        wait_until(lambda: hits, timeout=15.0)
+        headers, raw = hits[0]
+        assert headers["x-custom"] == "yes"
+        sig = headers["x-agentfield-signature"]
+        want = "sha256=" + hmac.new(b"s3cret", raw, hashlib.sha256).hexdigest()
+        assert sig == want
+        payload = json.loads(raw)
+        assert payload["event"] == "execution.completed"
+        assert payload["result"] == {"greeting": "hello async"}
+        assert payload["execution_id"] == body["execution_id"]
+        # poll API agrees
+        rec = httpx.get(srv.base_url +
+                        f"/api/v1/executions/{body['execution_id']}").json()
+        assert rec["status"] == "completed"
+        assert rec["webhook_registered"] is True
+    finally:
+        hook_srv.stop()
+
+
+def test_webhook_retry_on_failure(cp_server, greeting_agent):
+    srv, cp = cp_server
+    calls = []
+    hook = FastAPI()
+
+    @hook.post("/flaky")
+    async def recv(req: Request):
+        calls.append(time.time())
+        if len(calls) < 2:
+            from fastapi.responses import JSONResponse
+            return JSONResponse({"err": "nope"}, status_code=500)
+        return {"ok": True}
+
+    hook_srv = AppServer(hook).start()
+    cp.webhooks.poll_interval = 0.2
+    cp.webhooks.backoff_base = 0.1
+    try:
+        r = httpx.post(srv.base_url + "/api/v1/execute/async/greeter.greet",
+                       json={"input": {"name": "retry"},
+                             "webhook": {"url": hook_srv.base_url + "/flaky"}})
+        eid = r.json()["execution_id"]
+        wait_until(lambda: len(calls) >= 2, timeout=20.0)
+        wait_until(lambda: cp.storage.get_webhook(eid)["status"] == "delivered",
+                   timeout=10.0)
+        assert len(cp.storage.webhook_history(eid)) >= 2
+    finally:
+        hook_srv.stop()
+
+
+def test_batch_status(cp_server, greeting_agent):
+    srv, _ = cp_server
+    ids = []
+    for i in range(3):
+        r = httpx.post(srv.base_url + "/api/v1/execute/greeter.greet",
+                       json={"input": {"name": str(i)}}, timeout=20.0)
+        ids.append(r.json()["execution_id"])
+    r = httpx.post(srv.base_url + "/api/v1/executions/batch-status",
+                   json={"execution_ids": ids + ["exec_bogus"]})
+    body = r.json()
+    assert set(body["executions"]) == set(ids)
+    assert body["missing"] == ["exec_bogus"]
+
+
+def test_nested_call_builds_dag(cp_server, greeting_agent):
+    srv, cp = cp_server
+    r = httpx.post(srv.base_url + "/api/v1/execute/greeter.relay",
+                   json={"input": {"name": "dag"}}, timeout=30.0)
+    body = r.json()
+    assert body["status"] == "completed"
+    assert body["result"]["relayed"] == {"greeting": "hello DAG"}
+    run_id = body["run_id"]
+    dag = wait_until(lambda: (lambda d: d if d.status_code == 200 and
+                              len(d.json()["nodes"]) >= 2 else None)(
+        httpx.get(srv.base_url + f"/api/ui/v1/workflows/{run_id}/dag")),
+        timeout=10.0).json()
+    assert dag["total"] >= 2
+    child = [n for n in dag["nodes"] if n["parent_execution_id"]]
+    assert child, "nested call must appear as a DAG edge"
+    assert dag["status"] == "completed"
+    # v2 workflow runs API
+    runs = httpx.get(srv.base_url + "/api/ui/v2/workflow-runs").json()["runs"]
+    assert any(x["run_id"] == run_id for x in runs)
+
+
+def test_memory_scopes_and_vector(cp_server):
+    srv, _ = cp_server
+    base = srv.base_url + "/api/v1/memory"
+    # explicit global scope
+    httpx.post(base + "/set", json={"key": "k1", "value": {"v": 1},
+                                    "scope": "global"})
+    got = httpx.post(base + "/get", json={"key": "k1", "scope": "global"}).json()
+    assert got["found"] and got["value"] == {"v": 1}
+    # workflow scope from header (A.5 priority)
+    hdr = {"X-Run-ID": "run_test123", "X-Session-ID": "sess1"}
+    httpx.post(base + "/set", json={"key": "k2", "value": 7}, headers=hdr)
+    got = httpx.post(base + "/get", json={"key": "k2"}, headers=hdr).json()
+    assert got["scope"] == "workflow" and got["scope_id"] == "run_test123"
+    assert got["value"] == 7
+    miss = httpx.post(base + "/get", json={"key": "k2"},
+                      headers={"X-Session-ID": "sess1"}).json()
+    assert not miss["found"]  # different scope
+    keys = httpx.get(base + "/list", params={"scope": "global",
+                                             "scope_id": "global"}).json()
+    assert "k1" in keys["keys"]
+    # vector memory
+    httpx.post(base + "/vector/set", json={"key": "a", "embedding": [1, 0, 0],
+                                           "scope": "global",
+                                           "metadata": {"kind": "x"}})
+    httpx.post(base + "/vector/set", json={"key": "b", "embedding": [0, 1, 0],
+                                           "scope": "global"})
+    res = httpx.post(base + "/vector/search",
+                     json={"embedding": [0.9, 0.1, 0], "top_k": 2,
+                           "scope": "global"}).json()["results"]
+    assert res[0]["key"] == "a"
+    res = httpx.post(base + "/vector/search",
+                     json={"embedding": [0.9, 0.1, 0], "scope": "global",
+                           "filters": {"kind": "x"}}).json()["results"]
+    assert [r["key"] for r in res] == ["a"]
+
+
+def test_vc_issuance_verify_and_tamper(cp_server, greeting_agent):
+    srv, _ = cp_server
+    r = httpx.post(srv.base_url + "/api/v1/execute/greeter.greet",
+                   json={"input": {"name": "vc"}}, timeout=20.0)
+    eid = r.json()["execution_id"]
+    vc = httpx.post(srv.base_url + "/api/v1/execution/vc",
+                    json={"execution_id": eid}).json()["vc"]
+    assert vc["type"] == ["VerifiableCredential", "AgentFieldExecutionCredential"]
+    assert vc["proof"]["proofValue"]
+    ver = httpx.post(srv.base_url + "/api/v1/did/verify", json=vc).json()
+    assert ver["valid"]
+    # server-side verification incl. output-hash check
+    got = httpx.get(srv.base_url + f"/api/v1/executions/{eid}/vc").json()
+    assert got["verification"]["valid"]
+    assert got["verification"]["checks"]["output_hash_matches"]
+    # tamper -> invalid
+    bad = json.loads(json.dumps(vc))
+    bad["credentialSubject"]["execution"]["output_hash"] = "0" * 64
+    ver = httpx.post(srv.base_url + "/api/v1/did/verify", json=bad).json()
+    assert not ver["valid"]
+    # vc chain for the run
+    chain = httpx.get(srv.base_url +
+                      f"/api/v1/did/workflow/{r.json()['run_id']}/vc-chain").json()
+    assert chain["count"] >= 1 and chain["all_valid"]
+
+
+def test_did_resolve(cp_server, greeting_agent):
+    srv, _ = cp_server
+    st = httpx.get(srv.base_url + "/api/v1/did/status").json()
+    assert st["enabled"] and st["root_did"].startswith("did:key:z")
+    doc = httpx.get(srv.base_url +
+                    f"/api/v1/did/resolve/{st['root_did']}").json()
+    assert doc["document"]["id"] == st["root_did"]
+
+
+def test_node_status_transitions(cp_server, greeting_agent):
+    srv, _ = cp_server
+    ok = httpx.post(srv.base_url + "/api/v1/nodes/greeter/status",
+                    json={"status": "inactive"})
+    assert ok.status_code == 200
+    bad = httpx.post(srv.base_url + "/api/v1/nodes/greeter/status",
+                     json={"status": "bogus"})
+    assert bad.status_code == 400
+    httpx.post(srv.base_url + "/api/v1/nodes/greeter/status",
+               json={"status": "active"})
+
+
+def test_heartbeat_unknown_node_asks_reregister(cp_server):
+    srv, _ = cp_server
+    r = httpx.post(srv.base_url + "/api/v1/nodes/ghost/heartbeat", json={})
+    assert r.status_code == 404 and r.json()["action"] == "re-register"
+
+
+def test_ui_dashboard_and_reasoners(cp_server, greeting_agent):
+    srv, _ = cp_server
+    d = httpx.get(srv.base_url + "/api/ui/v1/dashboard/summary").json()
+    assert d["nodes"]["total"] >= 1
+    rs = httpx.get(srv.base_url + "/api/ui/v1/reasoners").json()["reasoners"]
+    assert any(r["id"] == "greet" for r in rs)
+
+
+def test_execution_sse_stream(cp_server, greeting_agent):
+    srv, _ = cp_server
+    got = []
+
+    def listen():
+        with httpx.stream("GET", srv.base_url + "/api/ui/v1/executions/events",
+                          timeout=10.0) as resp:
+            for line in resp.iter_lines():
+                if line.startswith("data:"):
+                    got.append(json.loads(line[5:]))
+                    return
+
+    t = threading.Thread(target=listen, daemon=True)
+    t.start()
+    time.sleep(0.3)
+    httpx.post(srv.base_url + "/api/v1/execute/greeter.greet",
+               json={"input": {"name": "sse"}}, timeout=20.0)
+    t.join(timeout=10.0)
+    assert got and got[0]["terminal"]
