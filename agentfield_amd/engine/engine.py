@@ -16,7 +16,7 @@ import torch
 from .. import ops
 from ..models.llama import (AttnMetadata, KVCache, LlamaConfig,
                             LlamaForCausalLM)
-from .scheduler import ScheduleBatch, Scheduler, SchedulerConfig
+from .scheduler import ScheduleBatch, SchedulerConfig, make_scheduler
 from .sequence import SamplingParams, Sequence, SeqStatus
 
 DECODE_BUCKETS = (1, 2, 4, 8, 16, 32, 64, 128, 256)
@@ -41,7 +41,7 @@ class LLMEngine:
                  dtype=torch.bfloat16, page_size: int = 16,
                  kv_memory_frac: float = 0.80, num_pages: int | None = None,
                  max_num_seqs: int = 64, max_prefill_tokens: int = 8192,
-                 enable_graphs: bool = True, eos_id: int = 2, seed: int = 0,
+                 max_waiting: int = 4096, enable_graphs: bool = True, eos_id: int = 2, seed: int = 0,
                  model: LlamaForCausalLM | None = None,
                  tp_group=None):
         self.cfg = cfg
@@ -62,9 +62,9 @@ class LLMEngine:
         max_pages_per_seq = (cfg.max_position + page_size - 1) // page_size
         self.max_pages_per_seq = max_pages_per_seq
         self.kv = KVCache(cfg, num_pages, page_size, self.device, dtype)
-        self.sched = Scheduler(SchedulerConfig(
+        self.sched = make_scheduler(SchedulerConfig(
             max_num_seqs=max_num_seqs, max_prefill_tokens=max_prefill_tokens,
-            page_size=page_size, num_pages=num_pages))
+            page_size=page_size, num_pages=num_pages, max_waiting=max_waiting))
         self.page_size = page_size
         self.eos_id = eos_id
         self.max_num_seqs = max_num_seqs
@@ -130,6 +130,7 @@ class LLMEngine:
         for seq, tok in zip(batch.seqs, tokens):
             if not seq.output_ids:
                 seq.first_token_ns = now
+            self.sched.note_token(seq)
             done = seq.append(tok, self.eos_id)
             if done:
                 seq.finish_ns = now

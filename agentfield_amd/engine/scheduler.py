@@ -151,7 +151,90 @@ class Scheduler:
             return None
         return ScheduleBatch(is_prefill=False, seqs=list(self.running))
 
+    def note_token(self, seq: Sequence) -> None:
+        """Interface parity with the native scheduler (token counting is
+        implicit here via seq.num_tokens)."""
+
     def finish(self, seq: Sequence) -> None:
         seq.status = SeqStatus.FINISHED
         self.running.remove(seq)
         self.release(seq)
+
+
+class _AllocShim:
+    def __init__(self, nat, num_pages):
+        self._nat = nat
+        self.num_pages = num_pages
+
+    @property
+    def num_free(self):
+        return self._nat.num_free()
+
+
+class NativeSchedulerAdapter:
+    """Engine-facing adapter over the C++ scheduler (agentfield_amd._native).
+    Keeps the Python Sequence objects; page/queue decisions run natively.
+    Semantics are pinned to the Python Scheduler by test_native_scheduler."""
+
+    def __init__(self, cfg: SchedulerConfig):
+        from agentfield_amd._native import NativeScheduler
+        self.cfg = cfg
+        self.nat = NativeScheduler(cfg.max_num_seqs, cfg.max_prefill_tokens,
+                                   cfg.page_size, cfg.num_pages,
+                                   cfg.max_waiting)
+        self.seqs: dict[int, Sequence] = {}
+        self.alloc = _AllocShim(self.nat, cfg.num_pages)
+
+    @property
+    def n_preempted(self):
+        return self.nat.n_preempted()
+
+    def add(self, seq: Sequence) -> bool:
+        if not self.nat.add(seq.seq_id, len(seq.prompt_ids)):
+            return False
+        self.seqs[seq.seq_id] = seq
+        return True
+
+    def has_work(self) -> bool:
+        return self.nat.has_work()
+
+    def num_queued(self) -> int:
+        return self.nat.num_queued()
+
+    def num_running(self) -> int:
+        return self.nat.num_running()
+
+    def note_token(self, seq: Sequence) -> None:
+        self.nat.note_token(seq.seq_id)
+
+    def schedule(self) -> ScheduleBatch | None:
+        r = self.nat.schedule()
+        for sid in r.preempted:
+            s = self.seqs[sid]
+            s.output_ids.clear()
+            s.pages = []
+            s.status = SeqStatus.WAITING
+        if not r.has_work:
+            return None
+        batch = []
+        for sid in r.seq_ids:
+            s = self.seqs[sid]
+            s.pages = self.nat.pages(sid)
+            s.status = SeqStatus.RUNNING
+            batch.append(s)
+        return ScheduleBatch(is_prefill=r.is_prefill, seqs=batch)
+
+    def finish(self, seq: Sequence) -> None:
+        seq.status = SeqStatus.FINISHED
+        self.nat.finish(seq.seq_id)
+        self.seqs.pop(seq.seq_id, None)
+
+
+def make_scheduler(cfg: SchedulerConfig, prefer_native: bool = True):
+    import os
+    if prefer_native and os.environ.get("AF_NATIVE_SCHED", "1") != "0":
+        try:
+            return NativeSchedulerAdapter(cfg)
+        except ImportError:
+            pass
+    return Scheduler(cfg)

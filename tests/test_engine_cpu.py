@@ -69,7 +69,6 @@ def test_multi_request_interleaving():
 
 
 def test_backpressure_returns_none():
-    eng = make_engine()
-    eng.sched.cfg.max_waiting = 1
+    eng = make_engine(max_waiting=1)
     assert eng.add_request([1], SamplingParams(max_tokens=1)) is not None
     assert eng.add_request([1], SamplingParams(max_tokens=1)) is None
