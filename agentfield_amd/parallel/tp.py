@@ -1,0 +1,172 @@
+"""Tensor parallelism over RCCL/xGMI (one process per GPU).
+
+Sharding (MI355X-first — sized so the only collective per layer is one
+all-reduce after attn-O and one after MLP-down, the pattern that maps onto
+xGMI's 7 p2p links):
+  * qkv:    column-parallel — each rank keeps Hq/tp query heads and Hk/tp
+            KV heads (whole heads, so RoPE/attention stay local)
+  * o:      row-parallel    — partial sums all-reduced
+  * gate_up: column-parallel (gate and up halves sharded separately so the
+            fused silu_mul kernel sees a local [gate|up] layout)
+  * down:   row-parallel    — partial sums all-reduced
+  * embed / lm_head: replicated (~1 GB bf16 for the 128k vocab; vocab-
+            parallel logits are a later optimization)
+
+Weights are deterministic per parameter NAME, so every rank can materialize
+the full tensor (fp32, one at a time), slice its shard and free — no rank
+ever holds the whole model.
+"""
+from __future__ import annotations
+
+import hashlib
+
+import torch
+import torch.distributed as dist
+
+from ..models.llama import LlamaConfig, LlamaForCausalLM
+
+
+def _seed_for(name: str, base_seed: int) -> int:
+    h = hashlib.sha256(f"{base_seed}:{name}".encode()).digest()
+    return int.from_bytes(h[:8], "little") % (2**63)
+
+
+def _full_param(name: str, shape, base_seed: int, device) -> torch.Tensor:
+    if "norm" in name:
+        return torch.ones(shape, dtype=torch.float32, device=device)
+    g = torch.Generator(device=device).manual_seed(_seed_for(name, base_seed))
+    t = torch.empty(shape, dtype=torch.float32, device=device)
+    return t.normal_(0.0, 0.02, generator=g)
+
+
+def shard_llama_weights(model: LlamaForCausalLM, full_cfg: LlamaConfig,
+                        tp: int, rank: int, base_seed: int = 0) -> None:
+    """Fill a shard-shaped model with its slice of deterministic full
+    weights.  Replaces init_random for TP runs (rank-agnostic layout)."""
+    cfg = full_cfg
+    D = cfg.head_dim
+    hq, hk = cfg.num_heads // tp, cfg.num_kv_heads // tp
+    inter = cfg.intermediate_size // tp
+    dev = model.embed.device
+    with torch.no_grad():
+        for name, p in model.named_parameters():
+            if name == "embed" or name == "lm_head":
+                full_shape = (cfg.vocab_size, cfg.hidden_size)
+            elif name.endswith("attn.qkv"):
+                full_shape = (cfg.q_size + 2 * cfg.kv_size, cfg.hidden_size)
+            elif name.endswith("attn.o"):
+                full_shape = (cfg.hidden_size, cfg.q_size)
+            elif name.endswith("mlp.gate_up"):
+                full_shape = (2 * cfg.intermediate_size, cfg.hidden_size)
+            elif name.endswith("mlp.down"):
+                full_shape = (cfg.hidden_size, cfg.intermediate_size)
+            else:  # norms
+                full_shape = tuple(p.shape)
+            full = _full_param(name, full_shape, base_seed, dev)
+            if name.endswith("attn.qkv"):
+                q, k, v = full.split([cfg.q_size, cfg.kv_size, cfg.kv_size], 0)
+                shard = torch.cat([
+                    q[rank * hq * D:(rank + 1) * hq * D],
+                    k[rank * hk * D:(rank + 1) * hk * D],
+                    v[rank * hk * D:(rank + 1) * hk * D]], dim=0)
+            elif name.endswith("attn.o"):
+                shard = full[:, rank * hq * D:(rank + 1) * hq * D]
+            elif name.endswith("mlp.gate_up"):
+                gate, up = full.chunk(2, dim=0)
+                shard = torch.cat([gate[rank * inter:(rank + 1) * inter],
+                                   up[rank * inter:(rank + 1) * inter]], dim=0)
+            elif name.endswith("mlp.down"):
+                shard = full[:, rank * inter:(rank + 1) * inter]
+            else:
+                shard = full
+            p.copy_(shard.to(p.dtype))
+            del full
+
+
+class _TPLayerHook:
+    """Inserts the RCCL all-reduce after attn-O and MLP-down projections."""
+
+    def __init__(self, group):
+        self.group = group
+
+    def __call__(self, module, inputs, output):
+        dist.all_reduce(output, group=self.group)
+        return output
+
+
+def build_tp_model(full_cfg: LlamaConfig, tp: int, rank: int, device,
+                   dtype=torch.bfloat16, group=None,
+                   base_seed: int = 0) -> LlamaForCausalLM:
+    """Construct this rank's shard of the model with collectives attached."""
+    shard_cfg = full_cfg.shard(tp)
+    model = LlamaForCausalLM(shard_cfg, device=device, dtype=dtype)
+    shard_llama_weights(model, full_cfg, tp, rank, base_seed)
+    if tp > 1:
+        hook = _TPLayerHook(group)
+        for layer in model.layers:
+            layer.attn.register_forward_hook(hook)
+            layer.mlp.register_forward_hook(hook)
+    return model
+
+
+class TPEngineGroup:
+    """Drives one LLMEngine across a TP process group.
+
+    All ranks run the identical engine code; rank 0 is the request source
+    and broadcasts each submitted request (prompt ids + sampling) before
+    stepping, so every rank makes identical scheduling/sampling decisions
+    and the only steady-state communication is the per-layer all-reduce.
+    """
+
+    def __init__(self, full_cfg: LlamaConfig, device, dtype=torch.bfloat16,
+                 group=None, base_seed: int = 0, **engine_kw):
+        import os
+        from ..engine import LLMEngine
+        self.group = group
+        self.rank = dist.get_rank(group) if dist.is_initialized() else 0
+        self.tp = dist.get_world_size(group) if dist.is_initialized() else 1
+        if self.tp > 1:
+            # RCCL-inside-hipGraph is opt-in until verified on an 8-GPU box
+            engine_kw.setdefault(
+                "enable_graphs", os.environ.get("AF_TP_GRAPHS", "0") == "1")
+        model = build_tp_model(full_cfg, self.tp, self.rank, device, dtype,
+                               group, base_seed)
+        shard_cfg = model.cfg
+        self.engine = LLMEngine(shard_cfg, device=device, dtype=dtype,
+                                model=model, tp_group=group, **engine_kw)
+        # identical page budget on every rank (min over ranks)
+        if self.tp > 1:
+            t = torch.tensor([self.engine.sched.cfg.num_pages],
+                             dtype=torch.int64)
+            if torch.device(device).type == "cuda":
+                t = t.to(device)
+            dist.all_reduce(t, op=dist.ReduceOp.MIN, group=group)
+
+    # ---- rank-0 request API -------------------------------------------
+    def submit(self, prompt_ids, sampling) -> int | None:
+        """Called with identical args on every rank (bench/test mode), or
+        on rank 0 only followed by sync_requests()."""
+        return self.engine.add_request(prompt_ids, sampling)
+
+    def broadcast_and_submit(self, requests: list | None):
+        """Rank 0 passes its new requests; other ranks pass None."""
+        if self.tp == 1:
+            out = []
+            for (ids, sp) in requests or []:
+                out.append(self.engine.add_request(ids, sp))
+            return out
+        obj = [requests if self.rank == 0 else None]
+        dist.broadcast_object_list(obj, src=0, group=self.group)
+        rids = []
+        for (ids, sp) in obj[0] or []:
+            rids.append(self.engine.add_request(ids, sp))
+        return rids
+
+    def step(self):
+        return self.engine.step()
+
+    def has_work(self):
+        return self.engine.has_work()
+
+    def get_finished(self, rid):
+        return self.engine.get_finished(rid)

@@ -38,6 +38,9 @@ def parse_args():
     p.add_argument("--gen-len", type=int, default=64)
     p.add_argument("--max-num-seqs", type=int, default=64)
     p.add_argument("--no-graphs", action="store_true")
+    p.add_argument("--tp", type=int, default=1,
+                   help="tensor-parallel degree (requires WORLD_SIZE == tp; "
+                        "ranks form one TP group instead of DP replicas)")
     p.add_argument("--device", default=None)
     return p.parse_args()
 
@@ -80,11 +83,21 @@ def main():
     device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
     cfg = CONFIGS[args.model]
     dtype = torch.bfloat16 if device == "cuda" else torch.float32
-    eng = LLMEngine(cfg, device=device, dtype=dtype,
-                    max_num_seqs=args.max_num_seqs,
-                    max_prefill_tokens=args.prompt_len * args.calls,
-                    enable_graphs=not args.no_graphs and device == "cuda",
-                    seed=0)
+    if args.tp > 1:
+        if world != args.tp:
+            raise SystemExit("--tp requires WORLD_SIZE == tp")
+        from agentfield_amd.parallel import TPEngineGroup
+        grp = TPEngineGroup(cfg, device, dtype=dtype,
+                            max_num_seqs=args.max_num_seqs,
+                            max_prefill_tokens=args.prompt_len * args.calls,
+                            enable_graphs=not args.no_graphs and device == "cuda")
+        eng = grp.engine
+    else:
+        eng = LLMEngine(cfg, device=device, dtype=dtype,
+                        max_num_seqs=args.max_num_seqs,
+                        max_prefill_tokens=args.prompt_len * args.calls,
+                        enable_graphs=not args.no_graphs and device == "cuda",
+                        seed=0)
 
     def barrier_sync():
         if dist is not None:
@@ -109,7 +122,8 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t[0])
 
-    total_calls = args.calls * args.steps * world
+    n_replicas = 1 if args.tp > 1 else world
+    total_calls = args.calls * args.steps * n_replicas
     value = total_calls / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
     if rank == 0:
@@ -128,11 +142,11 @@ def main():
             "data": "synthetic",
             "config": {
                 "model": cfg.name,
-                "global_batch": args.calls * world,
+                "global_batch": args.calls * n_replicas,
                 "seq_len": args.prompt_len + args.gen_len,
                 "prompt_len": args.prompt_len,
                 "gen_len": args.gen_len,
-                "parallelism": f"dp{world}",
+                "parallelism": f"tp{world}" if args.tp > 1 else f"dp{world}",
                 "p50_call_ms": round(statistics.median(lats) * 1000, 1) if lats else None,
                 "tokens_per_sec": round(total_calls * (args.prompt_len + args.gen_len) / elapsed, 1),
             },
