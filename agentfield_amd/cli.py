@@ -1,0 +1,190 @@
+"""The `af` CLI (reference parity: C32 — cobra `af` with server/init/run/
+list/vc/... subcommands, reimplemented with typer).
+
+  af server            run the control plane
+  af engine            run a GPU engine server (one per GPU)
+  af init NAME         scaffold an agent project
+  af run FILE          run an agent app (imports FILE, serves the Agent)
+  af list              list registered nodes
+  af executions        recent executions
+  af vc verify FILE    offline verifiable-credential verification
+  af status            control-plane health
+"""
+from __future__ import annotations
+
+import importlib.util
+import json
+import os
+import sys
+from pathlib import Path
+
+import typer
+
+app = typer.Typer(add_completion=False, no_args_is_help=True)
+
+DEFAULT_URL = os.environ.get("AGENTFIELD_URL", "http://127.0.0.1:8520")
+
+
+@app.command()
+def server(host: str = "0.0.0.0", port: int = 8520,
+           db: str = typer.Option("agentfield.db", help="SQLite path"),
+           data_dir: str = typer.Option(".agentfield", help="payload/keys dir"),
+           config: str = typer.Option(None, help="YAML config file"),
+           no_did: bool = False):
+    """Run the control plane."""
+    import uvicorn
+    from .controlplane import ControlPlane, create_app
+    from .controlplane.server import Config
+    kw = {}
+    if config:
+        import yaml
+        kw = yaml.safe_load(Path(config).read_text()) or {}
+    data = Path(kw.pop("data_dir", data_dir))
+    data.mkdir(parents=True, exist_ok=True)
+    cfg = Config(db_path=kw.pop("db_path", db),
+                 payload_dir=str(data / "payloads"),
+                 keystore_path=str(data / "keystore.key"),
+                 did_enabled=not no_did, **kw)
+    cp = ControlPlane(cfg)
+    typer.echo(f"agentfield-amd control plane on {host}:{port} (db={cfg.db_path})")
+    uvicorn.run(create_app(cp), host=host, port=port, log_level="info")
+
+
+@app.command()
+def engine(model: str = "llama-3-8b", device: str = None,
+           host: str = "127.0.0.1", port: int = 8710,
+           max_num_seqs: int = 256):
+    """Run one GPU engine server (start one per GPU for DP)."""
+    from .serving.engine_server import main as engine_main
+    argv = ["--model", model, "--host", host, "--port", str(port),
+            "--max-num-seqs", str(max_num_seqs)]
+    if device:
+        argv += ["--device", device]
+    sys.argv = ["engine_server"] + argv
+    engine_main()
+
+
+AGENT_TEMPLATE = '''"""Agent scaffolded by `af init`."""
+from agentfield_amd.sdk import Agent
+
+app = Agent("{name}")
+
+
+@app.reasoner()
+def greet(name: str):
+    """A minimal reasoner; swap the body for app.ai(...) calls."""
+    return {{"greeting": f"hello {{name}}"}}
+
+
+@app.reasoner()
+def think(question: str):
+    answer = app.ai(question, system="Answer concisely.", max_tokens=128)
+    return {{"answer": answer}}
+
+
+if __name__ == "__main__":
+    app.serve(port=8600)
+'''
+
+
+@app.command()
+def init(name: str, directory: str = "."):
+    """Scaffold a new agent project."""
+    root = Path(directory) / name
+    root.mkdir(parents=True, exist_ok=True)
+    (root / "agent.py").write_text(AGENT_TEMPLATE.format(name=name))
+    (root / "agentfield.yaml").write_text(
+        f"name: {name}\nentrypoint: agent.py\n")
+    typer.echo(f"scaffolded {root}/agent.py — run with: af run {root}/agent.py")
+
+
+@app.command()
+def run(path: str, port: int = 8600, host: str = "127.0.0.1",
+        agentfield_url: str = DEFAULT_URL):
+    """Run an agent app file (finds the Agent instance and serves it)."""
+    p = Path(path)
+    if p.is_dir():
+        p = p / "agent.py"
+    spec = importlib.util.spec_from_file_location("af_user_agent", p)
+    mod = importlib.util.module_from_spec(spec)
+    os.environ.setdefault("AGENTFIELD_URL", agentfield_url)
+    spec.loader.exec_module(mod)
+    from .sdk import Agent
+    agents = [v for v in vars(mod).values() if isinstance(v, Agent)]
+    if not agents:
+        typer.echo("no Agent instance found in module", err=True)
+        raise typer.Exit(1)
+    agents[0].serve(host=host, port=port)
+
+
+def _client():
+    import httpx
+    return httpx.Client(base_url=DEFAULT_URL, timeout=10.0)
+
+
+@app.command("list")
+def list_nodes():
+    """List registered agent nodes."""
+    r = _client().get("/api/v1/nodes").json()
+    for n in r.get("nodes", []):
+        typer.echo(f"{n['id']:24s} {n['status']:10s} {n.get('base_url', '')}"
+                   f"  reasoners={len(n.get('reasoners', []))}")
+
+
+@app.command()
+def executions(limit: int = 20, status: str = None):
+    """Recent executions."""
+    params = {"limit": limit}
+    if status:
+        params["status"] = status
+    r = _client().get("/api/ui/v1/executions", params=params).json()
+    for e in r.get("executions", []):
+        typer.echo(f"{e['id']} {e['status']:10s} "
+                   f"{e.get('node_id')}.{e.get('reasoner_id')} "
+                   f"{e.get('duration_ms') or ''}")
+
+
+@app.command()
+def status():
+    """Control-plane health."""
+    try:
+        r = _client().get("/api/v1/health").json()
+        typer.echo(json.dumps(r, indent=2))
+    except Exception as e:
+        typer.echo(f"control plane unreachable: {e}", err=True)
+        raise typer.Exit(1)
+
+
+vc_app = typer.Typer()
+app.add_typer(vc_app, name="vc", help="Verifiable-credential tools")
+
+
+@vc_app.command("verify")
+def vc_verify(file: str):
+    """Offline verification of a VC JSON document (reference: `af vc verify`)."""
+    from .controlplane.did import VCService
+    doc = json.loads(Path(file).read_text())
+    if "vc" in doc:
+        doc = doc["vc"]
+    res = VCService.verify_document(doc)
+    typer.echo(json.dumps(res, indent=2))
+    raise typer.Exit(0 if res["valid"] else 1)
+
+
+@app.command()
+def bench(steps: int = 4, warmup: int = 1, model: str = "llama-3-8b",
+          calls: int = 64):
+    """Run the serving benchmark (wraps bench.py)."""
+    import subprocess
+    repo = Path(__file__).resolve().parent.parent
+    subprocess.run([sys.executable, str(repo / "bench.py"),
+                    "--steps", str(steps), "--warmup", str(warmup),
+                    "--model", model, "--calls", str(calls)], check=True)
+
+
+def main():
+    app()
+
+
+if __name__ == "__main__":
+    main()

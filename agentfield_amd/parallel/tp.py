@@ -132,15 +132,19 @@ class TPEngineGroup:
         model = build_tp_model(full_cfg, self.tp, self.rank, device, dtype,
                                group, base_seed)
         shard_cfg = model.cfg
+        if self.tp > 1 and "num_pages" not in engine_kw and \
+                torch.device(device).type == "cuda":
+            # identical KV page budget on every rank: min of local estimates
+            from ..models.llama import KVCache
+            free, _ = torch.cuda.mem_get_info(device)
+            page_size = engine_kw.get("page_size", 16)
+            local = max(16, int(free * 0.80) //
+                        KVCache.bytes_per_page(shard_cfg, page_size))
+            t = torch.tensor([local], dtype=torch.int64, device=device)
+            dist.all_reduce(t, op=dist.ReduceOp.MIN, group=group)
+            engine_kw["num_pages"] = int(t.item())
         self.engine = LLMEngine(shard_cfg, device=device, dtype=dtype,
                                 model=model, tp_group=group, **engine_kw)
-        # identical page budget on every rank (min over ranks)
-        if self.tp > 1:
-            t = torch.tensor([self.engine.sched.cfg.num_pages],
-                             dtype=torch.int64)
-            if torch.device(device).type == "cuda":
-                t = t.to(device)
-            dist.all_reduce(t, op=dist.ReduceOp.MIN, group=group)
 
     # ---- rank-0 request API -------------------------------------------
     def submit(self, prompt_ids, sampling) -> int | None:

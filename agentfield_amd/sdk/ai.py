@@ -174,12 +174,20 @@ _runners_lock = threading.Lock()
 
 
 def get_runner(cfg: AIConfig) -> EngineRunner:
-    """Process-wide engine registry: one engine per model name."""
+    """Process-wide engine registry: one engine per model name.  When
+    AGENTFIELD_ENGINE_URLS is set (comma-separated engine servers), calls
+    route to the DP replica fleet instead of an in-process engine."""
     import torch
     key = cfg.model
     with _runners_lock:
         if key in _runners:
             return _runners[key]
+        urls = os.environ.get("AGENTFIELD_ENGINE_URLS")
+        if urls:
+            from ..serving.router import DPRouter, RemoteRunner
+            runner = RemoteRunner(DPRouter(urls.split(",")))
+            _runners[key] = runner
+            return runner
         device = cfg.device or ("cuda" if torch.cuda.is_available() else "cpu")
         model_cfg = CONFIGS[cfg.model]
         kw = {}

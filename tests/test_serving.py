@@ -1,0 +1,115 @@
+"""Engine server + DP router + CLI tests (CPU, tiny model)."""
+import json
+import os
+
+import pytest
+import torch
+
+from agentfield_amd.engine import LLMEngine
+from agentfield_amd.models import CONFIGS
+from agentfield_amd.sdk.ai import AgentAI, AIConfig, EngineRunner
+from agentfield_amd.serving.engine_server import create_engine_app
+from agentfield_amd.serving.router import DPRouter
+
+from helpers import AppServer
+
+
+def make_engine_srv():
+    cfg = CONFIGS["tiny"]
+    eng = LLMEngine(cfg, device="cpu", dtype=torch.float32, page_size=4,
+                    num_pages=128, max_num_seqs=4, enable_graphs=False)
+    runner = EngineRunner(eng)
+    return AppServer(create_engine_app(runner, "tiny")).start(), runner
+
+
+@pytest.fixture(scope="module")
+def replicas():
+    s1, r1 = make_engine_srv()
+    s2, r2 = make_engine_srv()
+    yield [s1, s2]
+    r1.shutdown()
+    r2.shutdown()
+    s1.stop()
+    s2.stop()
+
+
+def test_engine_server_generate(replicas):
+    import httpx
+    r = httpx.post(replicas[0].base_url + "/v1/generate",
+                   json={"prompt_ids": [1, 5, 9], "max_tokens": 4,
+                         "ignore_eos": True}, timeout=60.0)
+    body = r.json()
+    assert len(body["output_ids"]) == 4
+    stats = httpx.get(replicas[0].base_url + "/v1/stats").json()
+    assert stats["decode_tokens"] >= 3 and stats["kv_total_pages"] == 128
+
+
+def test_router_routes_and_balances(replicas):
+    router = DPRouter([s.base_url for s in replicas], refresh_s=0.0)
+    outs = [router.generate(prompt_ids=[1, 2, 3], max_tokens=3,
+                            ignore_eos=True) for _ in range(4)]
+    assert all(len(o["output_ids"]) == 3 for o in outs)
+    st = router.stats()
+    assert all(st["healthy"].values())
+
+
+def test_router_skips_dead_replica(replicas):
+    router = DPRouter(["http://127.0.0.1:9", replicas[0].base_url],
+                      refresh_s=0.0)
+    out = router.generate(prompt_ids=[4, 5], max_tokens=2, ignore_eos=True)
+    assert len(out["output_ids"]) == 2
+
+
+def test_router_streaming(replicas):
+    router = DPRouter([s.base_url for s in replicas])
+    pieces = list(router.stream(prompt_ids=[1, 2], max_tokens=3,
+                                ignore_eos=True))
+    assert len(pieces) >= 1
+
+
+def test_ai_via_remote_engines(replicas, monkeypatch):
+    from agentfield_amd.sdk import ai as ai_mod
+    monkeypatch.setenv("AGENTFIELD_ENGINE_URLS",
+                       ",".join(s.base_url for s in replicas))
+    ai_mod._runners.pop("tiny-remote", None)
+    ai = AgentAI(AIConfig(model="tiny-remote", max_tokens=4, timeout=60))
+    out = ai("hello")
+    assert isinstance(out, str)
+    ai_mod._runners.pop("tiny-remote", None)
+
+
+def test_cli_vc_verify(tmp_path):
+    from typer.testing import CliRunner
+    from agentfield_amd.cli import app as cli_app
+    from agentfield_amd.controlplane.did import DIDService, Keystore, VCService
+    from agentfield_amd.controlplane.storage import Storage
+
+    store = Storage(":memory:")
+    dids = DIDService(store, Keystore(str(tmp_path / "ks.key")))
+    vcs = VCService(store, dids)
+    store.create_execution({"id": "exec_x", "run_id": "run_x",
+                            "input": {"a": 1}})
+    store.update_execution_result("exec_x", "completed", {"ok": True})
+    doc = vcs.issue_execution_vc(store.get_execution("exec_x"))
+    f = tmp_path / "vc.json"
+    f.write_text(json.dumps(doc))
+    res = CliRunner().invoke(cli_app, ["vc", "verify", str(f)])
+    assert res.exit_code == 0, res.output
+    assert '"valid": true' in res.output
+    # tampered doc fails
+    bad = dict(doc)
+    bad["issuanceDate"] = "1999-01-01T00:00:00Z"
+    f.write_text(json.dumps(bad))
+    res = CliRunner().invoke(cli_app, ["vc", "verify", str(f)])
+    assert res.exit_code == 1
+
+
+def test_cli_init(tmp_path):
+    from typer.testing import CliRunner
+    from agentfield_amd.cli import app as cli_app
+    res = CliRunner().invoke(cli_app, ["init", "myagent", "--directory",
+                                       str(tmp_path)])
+    assert res.exit_code == 0
+    assert (tmp_path / "myagent" / "agent.py").exists()
+    text = (tmp_path / "myagent" / "agent.py").read_text()
+    assert 'Agent("myagent")' in text
