@@ -74,6 +74,8 @@ def build_afops(force: bool = False, verbose: bool = True) -> Path:
 
 def build_native(force: bool = False, verbose: bool = True):
     """Build the CPU-side pybind11 extension (scheduler, allocator, crypto)."""
+    import sys as _sys
+    _sys.path.insert(0, str(PKG_DIR.parent))
     from agentfield_amd.native_build import build as _build  # lazy import
     return _build(force=force, verbose=verbose)
 

@@ -1,0 +1,144 @@
+// Skinny-M GEMM for the decode path:  Y[M,N] = X[M,K] @ W[N,K]^T, M <= 64.
+//
+// At decode batch sizes the GEMM is a pure weight-stream (W is N*K*2 bytes,
+// X is L2-resident), so the kernel is built to read W at HBM rate with MFMA
+// doing the math for free:
+//   * no LDS at all — A and B fragments load straight from global memory;
+//     the B-fragment read pattern (lane -> W[n0+(l&15)][k+(l>>4)*8..+8]) is
+//     a coalesced 1 KiB wave read of 16 consecutive W rows
+//   * grid (N/64, SK): 4 waves per workgroup, one 16-row N-tile each;
+//     K is split SK ways so the launch fills 256 CUs even at N=4096
+//     (hipBLASLt's MT16 kernels run at ~1.7-3.7 TB/s here; this targets the
+//     ~6.3 TB/s achievable ceiling)
+//   * fp32 partials [SK, M, N] are folded by a combine kernel that also
+//     carries the fused epilogue: plain bf16 store, or SwiGLU
+//     (out = silu(gate)*up) for the gate_up projection, or residual-add.
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8s;
+
+__device__ __forceinline__ bf16x8s as_bf16x8s(s16x8 v) {
+  union { s16x8 s; bf16x8s b; } u;
+  u.s = v;
+  return u.b;
+}
+
+// MT = number of 16-row M tiles (ceil(M/16)); grid (N/64, SK), block 256.
+template <int MT>
+__global__ void __launch_bounds__(256) gemm_skinny_kernel(
+    float* __restrict__ partial, const u16* __restrict__ X,
+    const u16* __restrict__ W, int M, int N, int K, int Kc) {
+  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int n0 = (blockIdx.x * 4 + wid) * 16;
+  const int sk = blockIdx.y;
+  const int k0 = sk * Kc;
+  const int k1 = min(K, k0 + Kc);
+
+  f32x4 acc[MT];
+#pragma unroll
+  for (int mt = 0; mt < MT; ++mt) acc[mt] = f32x4{0, 0, 0, 0};
+
+  const int arow_base = (lane & 15);
+  const int koff = (lane >> 4) * 8;
+  const u16* wrow = W + (size_t)(n0 + (lane & 15)) * K + koff;
+
+  for (int k = k0; k < k1; k += 32) {
+    const s16x8 bf = *reinterpret_cast<const s16x8*>(wrow + k);
+#pragma unroll
+    for (int mt = 0; mt < MT; ++mt) {
+      const int row = min(mt * 16 + arow_base, M - 1);
+      const s16x8 af =
+          *reinterpret_cast<const s16x8*>(X + (size_t)row * K + k + koff);
+      acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          as_bf16x8s(af), as_bf16x8s(bf), acc[mt], 0, 0, 0);
+    }
+  }
+
+  // C layout: lane l reg r -> C[m=(l>>4)*4+r][n=l&15]
+  float* pbase = partial + (size_t)sk * M * N;
+#pragma unroll
+  for (int mt = 0; mt < MT; ++mt) {
+    const int mrow = mt * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      if (mrow + r < M)
+        pbase[(size_t)(mrow + r) * N + n0 + (lane & 15)] = acc[mt][r];
+    }
+  }
+}
+
+// mode 0: out[M,N] bf16 = sum_sk partial
+// mode 1: out[M,N] bf16 = sum + residual read/write (residual updated in place)
+// mode 2: N = 2I; out[M,I] bf16 = silu(sum[:, :I]) * sum[:, I:]
+__global__ void __launch_bounds__(256) gemm_skinny_combine_kernel(
+    u16* __restrict__ out, u16* __restrict__ residual,
+    const float* __restrict__ partial, int M, int N, int SK, int mode) {
+  const int cols = (mode == 2) ? (N >> 1) : N;
+  const i64 total = (i64)M * (cols >> 2);  // 4 outputs per thread
+  const i64 stride = (i64)gridDim.x * blockDim.x;
+  for (i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    const int m = (int)(i / (cols >> 2));
+    const int c = (int)(i % (cols >> 2)) * 4;
+    float v[4] = {0, 0, 0, 0};
+    for (int s = 0; s < SK; ++s) {
+      const float* p = partial + ((size_t)s * M + m) * N + c;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) v[j] += p[j];
+    }
+    if (mode == 2) {
+      float u[4] = {0, 0, 0, 0};
+      for (int s = 0; s < SK; ++s) {
+        const float* p = partial + ((size_t)s * M + m) * N + (N >> 1) + c;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) u[j] += p[j];
+      }
+      s16x4 o;
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        o[j] = (short)f2bf(v[j] / (1.f + __expf(-v[j])) * u[j]);
+      *reinterpret_cast<s16x4*>(out + (size_t)m * cols + c) = o;
+    } else if (mode == 1) {
+      s16x4 rv = *reinterpret_cast<const s16x4*>(residual + (size_t)m * N + c);
+      s16x4 o;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) o[j] = (short)f2bf(v[j] + bf2f((u16)rv[j]));
+      *reinterpret_cast<s16x4*>(out + (size_t)m * N + c) = o;
+      *reinterpret_cast<s16x4*>(residual + (size_t)m * N + c) = o;
+    } else {
+      s16x4 o;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) o[j] = (short)f2bf(v[j]);
+      *reinterpret_cast<s16x4*>(out + (size_t)m * N + c) = o;
+    }
+  }
+}
+
+AF_EXPORT int af_gemm_skinny(void* out, void* partial, void* residual,
+                             const void* X, const void* W, int M, int N,
+                             int K, int SK, int mode, void* stream) {
+  if (M < 1 || M > 64) return 9005;
+  if (N % 64 || K % 32) return 9006;
+  hipStream_t st = (hipStream_t)stream;
+  int Kc = ((K / SK + 31) / 32) * 32;
+  while ((SK - 1) * Kc >= K) --SK;  // drop empty splits
+  dim3 grid(N / 64, SK), blk(256);
+  const int MT = (M + 15) / 16;
+  switch (MT) {
+    case 1: gemm_skinny_kernel<1><<<grid, blk, 0, st>>>(
+        (float*)partial, (const u16*)X, (const u16*)W, M, N, K, Kc); break;
+    case 2: gemm_skinny_kernel<2><<<grid, blk, 0, st>>>(
+        (float*)partial, (const u16*)X, (const u16*)W, M, N, K, Kc); break;
+    case 3: gemm_skinny_kernel<3><<<grid, blk, 0, st>>>(
+        (float*)partial, (const u16*)X, (const u16*)W, M, N, K, Kc); break;
+    default: gemm_skinny_kernel<4><<<grid, blk, 0, st>>>(
+        (float*)partial, (const u16*)X, (const u16*)W, M, N, K, Kc); break;
+  }
+  const int cols = (mode == 2) ? N / 2 : N;
+  i64 total = (i64)M * (cols / 4);
+  int blocks = (int)((total + 255) / 256);
+  if (blocks > 1024) blocks = 1024;
+  gemm_skinny_combine_kernel<<<blocks, 256, 0, st>>>(
+      (u16*)out, (u16*)residual, (const float*)partial, M, N, SK, mode);
+  return af_last_err();
+}

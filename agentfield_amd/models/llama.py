@@ -124,7 +124,7 @@ class LlamaAttention(nn.Module):
 
     def forward(self, x, positions, rope_tab, kv: KVCache, md: AttnMetadata):
         cfg = self.cfg
-        qkv = x @ self.qkv.t()
+        qkv = ops.linear(x, self.qkv)
         # strided views straight into the fused projection (no copies)
         q = qkv[:, :cfg.q_size]
         k = qkv[:, cfg.q_size:cfg.q_size + cfg.kv_size]
@@ -138,7 +138,7 @@ class LlamaAttention(nn.Module):
             o = ops.attn_decode(q, kv.k[self.layer_idx], kv.v[self.layer_idx],
                                 md.block_table, md.seq_lens_t, self.scale,
                                 nsplit=md.nsplit, scratch=md.decode_scratch)
-        return o @ self.o.t()
+        return ops.linear(o, self.o)
 
 
 class LlamaMLP(nn.Module):
@@ -150,7 +150,8 @@ class LlamaMLP(nn.Module):
                                              cfg.intermediate_size))
 
     def forward(self, x):
-        return ops.silu_and_mul(x @ self.gate_up.t()) @ self.down.t()
+        return ops.linear(ops.linear(x, self.gate_up, silu_fuse=True),
+                          self.down)
 
 
 class LlamaLayer(nn.Module):
@@ -214,4 +215,4 @@ class LlamaForCausalLM(nn.Module):
         h, _ = ops.rmsnorm(h, self.final_norm, self.cfg.rms_eps, residual)
         if logit_rows is not None:
             h = ops.gather_rows(h, logit_rows)
-        return h @ self.lm_head.t()
+        return ops.linear(h, self.lm_head)

@@ -195,6 +195,48 @@ def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
 
 
 # ---------------------------------------------------------------- gemm
+def choose_sk(N: int, K: int) -> int:
+    """K-split for the skinny GEMM: fill >=1024 workgroups, >=256 K/split."""
+    ntiles = N // 64
+    sk = max(1, min(16, 1024 // max(1, ntiles)))
+    sk = min(sk, max(1, K // 256))
+    return sk
+
+
+def linear_skinny(x: torch.Tensor, w: torch.Tensor, mode: int = 0,
+                  residual: torch.Tensor | None = None,
+                  sk: int | None = None) -> torch.Tensor:
+    """Decode-path GEMM (M<=64) via the MFMA weight-streaming kernel.
+    mode 0: plain; mode 1: +residual (in-place update); mode 2: fused SwiGLU
+    (w holds [gate|up] rows, returns [M, N/2])."""
+    M, K = x.shape
+    N = w.shape[0]
+    if sk is None:
+        sk = choose_sk(N, K)
+    cols = N // 2 if mode == 2 else N
+    out = torch.empty(M, cols, dtype=x.dtype, device=x.device)
+    partial = torch.empty(sk, M, N, dtype=torch.float32, device=x.device)
+    rc = _lib.lib().af_gemm_skinny(
+        _lib.ptr(out), _lib.ptr(partial), _lib.ptr(residual), _lib.ptr(x),
+        _lib.ptr(w), M, N, K, sk, mode, _lib.cur_stream())
+    _lib.check(rc, "af_gemm_skinny")
+    return out
+
+
+def linear(x: torch.Tensor, w: torch.Tensor, silu_fuse: bool = False) -> torch.Tensor:
+    """Linear layer dispatch: hand-written skinny MFMA kernel for decode
+    shapes (M<=64), hipBLASLt (torch.matmul) for prefill GEMMs."""
+    M, K = x.shape
+    N = w.shape[0]
+    if (x.is_cuda and M <= 64 and N % 64 == 0 and K % 32 == 0
+            and x.is_contiguous()):
+        return linear_skinny(x, w, mode=2 if silu_fuse else 0)
+    y = x @ w.t()
+    if silu_fuse:
+        return silu_and_mul(y)
+    return y
+
+
 def gemm_bf16(a: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """C[M,N] = a[M,K] @ w[N,K]^T via the hand-written MFMA kernel."""
     if not _on_gpu(a):

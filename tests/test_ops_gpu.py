@@ -209,6 +209,43 @@ def test_gemm():
         assert close(c, want, atol=tol, rtol=5e-2), f"gemm {M}x{N}x{K}"
 
 
+def test_gemm_skinny():
+    for M in (1, 7, 16, 33, 64):
+        for (N, K) in [(6144, 4096), (4096, 14336), (128256, 4096)]:
+            a = rnd(M, K, seed=M + N, scale=0.3)
+            w = rnd(N, K, seed=M + N + 1, scale=0.3)
+            c = ops.linear_skinny(a, w)
+            torch.cuda.synchronize()
+            want = a.float() @ w.float().t()
+            tol = 0.1 + 0.02 * math.sqrt(K)
+            assert close(c, want, atol=tol, rtol=5e-2), f"skinny {M}x{N}x{K}"
+
+
+def test_gemm_skinny_silu_fused():
+    M, I, K = 16, 14336, 4096
+    a = rnd(M, K, seed=5, scale=0.3)
+    w = rnd(2 * I, K, seed=6, scale=0.3)
+    c = ops.linear_skinny(a, w, mode=2)
+    torch.cuda.synchronize()
+    gu = (a.float() @ w.float().t())
+    want = torch.nn.functional.silu(gu[:, :I]) * gu[:, I:]
+    assert close(c, want, atol=3.0, rtol=8e-2)
+
+
+def test_gemm_skinny_residual_mode():
+    M, N, K = 16, 4096, 4096
+    a = rnd(M, K, seed=7, scale=0.3)
+    w = rnd(N, K, seed=8, scale=0.3)
+    res = rnd(M, N, seed=9)
+    res_ref = res.float().cpu().clone()
+    c = ops.linear_skinny(a, w, mode=1, residual=res)
+    torch.cuda.synchronize()
+    want = a.float() @ w.float().t() + res_ref.to(DEV)
+    tol = 0.1 + 0.02 * math.sqrt(K)
+    assert close(c, want, atol=tol, rtol=5e-2)
+    assert close(res, want, atol=tol, rtol=5e-2)  # residual updated in place
+
+
 def test_sample_greedy_matches_argmax():
     torch.manual_seed(99)
     B, V = 5, 128256

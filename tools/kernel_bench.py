@@ -41,6 +41,21 @@ def bench_gemm():
                           "hipblaslt_tflops": round(tf / t_blas, 1)}))
 
 
+def bench_skinny():
+    for (M, N, K) in [(16, 6144, 4096), (16, 4096, 4096), (16, 28672, 4096),
+                      (16, 4096, 14336), (64, 28672, 4096), (16, 128256, 4096)]:
+        a = torch.randn(M, K, dtype=torch.bfloat16, device=DEV)
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV)
+        gb = N * K * 2 / 1e9
+        t_ours = timeit(lambda: ops.linear_skinny(a, w), iters=30)
+        t_blas = timeit(lambda: a @ w.t(), iters=30)
+        print(json.dumps({"op": "gemm_skinny", "MNK": [M, N, K],
+                          "ours_tbps": round(gb / t_ours / 1e3, 2),
+                          "hipblaslt_tbps": round(gb / t_blas / 1e3, 2),
+                          "ours_us": round(t_ours * 1e6, 1),
+                          "blas_us": round(t_blas * 1e6, 1)}))
+
+
 def bench_prefill(Hq=32, Hk=8, D=128):
     for (B, S) in [(16, 512), (4, 2048), (1, 8192)]:
         T = B * S
@@ -101,7 +116,8 @@ if __name__ == "__main__":
     ap.add_argument("--only", default=None)
     args = ap.parse_args()
     torch.manual_seed(0)
-    for name, fn in [("gemm", bench_gemm), ("prefill", bench_prefill),
+    for name, fn in [("gemm", bench_gemm), ("skinny", bench_skinny),
+                     ("prefill", bench_prefill),
                      ("decode", bench_decode), ("norm", bench_norm_rope),
                      ("sample", bench_sample)]:
         if args.only in (None, name):
