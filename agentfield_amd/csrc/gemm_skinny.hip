@@ -1,18 +1,19 @@
 // Skinny-M GEMM for the decode path:  Y[M,N] = X[M,K] @ W[N,K]^T, M <= 64.
 //
-// At decode batch sizes the GEMM is a pure weight-stream (W is N*K*2 bytes,
-// X is L2-resident), so the kernel is built to read W at HBM rate with MFMA
-// doing the math for free:
-//   * no LDS at all — A and B fragments load straight from global memory;
-//     the B-fragment read pattern (lane -> W[n0+(l&15)][k+(l>>4)*8..+8]) is
-//     a coalesced 1 KiB wave read of 16 consecutive W rows
-//   * grid (N/64, SK): 4 waves per workgroup, one 16-row N-tile each;
-//     K is split SK ways so the launch fills 256 CUs even at N=4096
-//     (hipBLASLt's MT16 kernels run at ~1.7-3.7 TB/s here; this targets the
-//     ~6.3 TB/s achievable ceiling)
-//   * fp32 partials [SK, M, N] are folded by a combine kernel that also
-//     carries the fused epilogue: plain bf16 store, or SwiGLU
-//     (out = silu(gate)*up) for the gate_up projection, or residual-add.
+// At decode batch sizes this is a pure weight-stream (W is N*K*2 bytes, X is
+// L2-resident), so the kernel is built to read W at HBM rate with MFMA doing
+// the math for free:
+//   * grid (N/64, SK): block = 4 waves staging one W tile [64 rows x 64 K]
+//     (8 KiB) plus the X tile [<=64 x 64] per K-step through LDS with
+//     __builtin_amdgcn_global_load_lds (16 B lanes) and the T2 XOR swizzle
+//     pre-applied on the *source* address (guide ERRATA #21) so fragment
+//     ds_read_b128s are bank-conflict-free
+//   * K is split SK ways so the launch fills 256 CUs even at N=4096;
+//     fp32 partials [SK, M, N] are folded by a combine kernel that also
+//     carries the fused epilogue: plain bf16 store, SwiGLU
+//     (out = silu(gate)*up) for the gate_up projection, or residual-add
+//   * hipBLASLt's MT16 kernels run at ~1.7-2.6 TB/s on the N=4096/6144
+//     decode shapes; this path targets the ~6 TB/s achievable ceiling
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8s;
@@ -23,13 +24,40 @@ __device__ __forceinline__ bf16x8s as_bf16x8s(s16x8 v) {
   return u.b;
 }
 
+// Stage a 64x64 bf16 tile (8 KiB) into linear LDS with read-side swizzle
+// pre-applied to the global source.  2 x 16 B per thread.
+__device__ __forceinline__ void stage64(const u16* __restrict__ g, size_t ld,
+                                        u16* lds, int row0, int k0,
+                                        int max_row) {
+  const int tid = threadIdx.x;
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    const int e = (it * 256 + tid) * 8;
+    const int row = e >> 6;
+    const int wb = (e & 63) * 2;
+    const int wsw = wb ^ ((row & 7) << 4);
+    const u16* src = g + (size_t)min(row0 + row, max_row - 1) * ld + k0 + (wsw >> 1);
+    u16* dst = lds + (size_t)(it * 256 + (tid & ~63)) * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) uint32_t*)src,
+        (__attribute__((address_space(3))) uint32_t*)dst, 16, 0, 0);
+  }
+}
+
+__device__ __forceinline__ s16x8 frag64(const u16* lds, int row, int byte) {
+  return *reinterpret_cast<const s16x8*>(
+      reinterpret_cast<const char*>(lds) + row * 128 + (byte ^ ((row & 7) << 4)));
+}
+
 // MT = number of 16-row M tiles (ceil(M/16)); grid (N/64, SK), block 256.
 template <int MT>
 __global__ void __launch_bounds__(256) gemm_skinny_kernel(
     float* __restrict__ partial, const u16* __restrict__ X,
     const u16* __restrict__ W, int M, int N, int K, int Kc) {
+  __shared__ u16 sW[2][64 * 64];
+  __shared__ u16 sX[2][64 * 64];
   const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  const int n0 = (blockIdx.x * 4 + wid) * 16;
+  const int nblk = blockIdx.x * 64;
   const int sk = blockIdx.y;
   const int k0 = sk * Kc;
   const int k1 = min(K, k0 + Kc);
@@ -38,31 +66,43 @@ __global__ void __launch_bounds__(256) gemm_skinny_kernel(
 #pragma unroll
   for (int mt = 0; mt < MT; ++mt) acc[mt] = f32x4{0, 0, 0, 0};
 
-  const int arow_base = (lane & 15);
-  const int koff = (lane >> 4) * 8;
-  const u16* wrow = W + (size_t)(n0 + (lane & 15)) * K + koff;
-
-  for (int k = k0; k < k1; k += 32) {
-    const s16x8 bf = *reinterpret_cast<const s16x8*>(wrow + k);
-#pragma unroll
-    for (int mt = 0; mt < MT; ++mt) {
-      const int row = min(mt * 16 + arow_base, M - 1);
-      const s16x8 af =
-          *reinterpret_cast<const s16x8*>(X + (size_t)row * K + k + koff);
-      acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          as_bf16x8s(af), as_bf16x8s(bf), acc[mt], 0, 0, 0);
+  // 2-phase pipeline (guide T3 minimum): stage tile t+1 while computing
+  // tile t; the single __syncthreads (vmcnt(0)+barrier) at loop end drains
+  // the in-flight loads after compute has covered their latency.
+  int cur = 0;
+  stage64(W, K, sW[0], nblk, k0, N);
+  stage64(X, K, sX[0], 0, k0, M);
+  __syncthreads();
+  for (int k = k0; k < k1; k += 64) {
+    if (k + 64 < k1) {
+      stage64(W, K, sW[cur ^ 1], nblk, k + 64, N);
+      stage64(X, K, sX[cur ^ 1], 0, k + 64, M);
     }
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const int byte = (kk * 32 + (lane >> 4) * 8) * 2;
+      const s16x8 bf = frag64(sW[cur], wid * 16 + (lane & 15), byte);
+#pragma unroll
+      for (int mt = 0; mt < MT; ++mt) {
+        const s16x8 af = frag64(sX[cur], mt * 16 + (lane & 15), byte);
+        acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            as_bf16x8s(af), as_bf16x8s(bf), acc[mt], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+    cur ^= 1;
   }
 
   // C layout: lane l reg r -> C[m=(l>>4)*4+r][n=l&15]
   float* pbase = partial + (size_t)sk * M * N;
+  const int ncol = nblk + wid * 16 + (lane & 15);
 #pragma unroll
   for (int mt = 0; mt < MT; ++mt) {
     const int mrow = mt * 16 + (lane >> 4) * 4;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       if (mrow + r < M)
-        pbase[(size_t)(mrow + r) * N + n0 + (lane & 15)] = acc[mt][r];
+        pbase[(size_t)(mrow + r) * N + ncol] = acc[mt][r];
     }
   }
 }
@@ -118,9 +158,9 @@ AF_EXPORT int af_gemm_skinny(void* out, void* partial, void* residual,
                              const void* X, const void* W, int M, int N,
                              int K, int SK, int mode, void* stream) {
   if (M < 1 || M > 64) return 9005;
-  if (N % 64 || K % 32) return 9006;
+  if (N % 64 || K % 64) return 9006;
   hipStream_t st = (hipStream_t)stream;
-  int Kc = ((K / SK + 31) / 32) * 32;
+  int Kc = ((K / SK + 63) / 64) * 64;
   while ((SK - 1) * Kc >= K) --SK;  // drop empty splits
   dim3 grid(N / 64, SK), blk(256);
   const int MT = (M + 15) / 16;

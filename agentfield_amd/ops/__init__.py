@@ -228,8 +228,12 @@ def linear(x: torch.Tensor, w: torch.Tensor, silu_fuse: bool = False) -> torch.T
     shapes (M<=64), hipBLASLt (torch.matmul) for prefill GEMMs."""
     M, K = x.shape
     N = w.shape[0]
-    if (x.is_cuda and M <= 64 and N % 64 == 0 and K % 32 == 0
-            and x.is_contiguous()):
+    # policy from measured MI355X data (profiles/kernel_bench_*): the
+    # hand-written path wins the small-N decode shapes and the fused-SwiGLU
+    # gate_up; hipBLASLt wins K>=14336 (down) and the 128k-vocab lm_head.
+    if (x.is_cuda and M <= 64 and N % 64 == 0 and K % 64 == 0
+            and x.is_contiguous()
+            and (silu_fuse or (N <= 8192 and K <= 8192))):
         return linear_skinny(x, w, mode=2 if silu_fuse else 0)
     y = x @ w.t()
     if silu_fuse:
