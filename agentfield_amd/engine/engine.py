@@ -30,8 +30,9 @@ def _bucket_for(n: int, max_bs: int) -> int:
 
 
 def choose_nsplit(bs: int, hk: int) -> int:
-    """Decode attention split-K factor: fill >=512 workgroups on 256 CUs."""
-    target = 512
+    """Decode attention split-K factor: fill >=1024 workgroups (256 CUs x
+    4 blocks for latency hiding; measured 2.4 TB/s at 512 WGs, bs=64)."""
+    target = 1024
     ns = max(1, min(16, target // max(1, bs * hk)))
     return 1 << (ns.bit_length() - 1)
 

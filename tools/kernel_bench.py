@@ -43,7 +43,8 @@ def bench_gemm():
 
 def bench_skinny():
     for (M, N, K) in [(16, 6144, 4096), (16, 4096, 4096), (16, 28672, 4096),
-                      (16, 4096, 14336), (64, 28672, 4096), (16, 128256, 4096)]:
+                      (16, 4096, 14336), (64, 28672, 4096), (16, 128256, 4096),
+                      (64, 4096, 14336), (64, 128256, 4096), (64, 6144, 4096)]:
         a = torch.randn(M, K, dtype=torch.bfloat16, device=DEV)
         w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV)
         gb = N * K * 2 / 1e9
