@@ -233,7 +233,8 @@ def linear(x: torch.Tensor, w: torch.Tensor, silu_fuse: bool = False) -> torch.T
     # gate_up; hipBLASLt wins K>=14336 (down) and the 128k-vocab lm_head.
     if (x.is_cuda and M <= 64 and N % 64 == 0 and K % 64 == 0
             and x.is_contiguous()
-            and (silu_fuse or (N <= 8192 and K <= 8192))):
+            and (silu_fuse or (N <= 8192 and K <= 8192)
+                 or (M > 32 and N <= 8192))):
         return linear_skinny(x, w, mode=2 if silu_fuse else 0)
     y = x @ w.t()
     if silu_fuse:
