@@ -44,7 +44,9 @@ def server(host: str = "0.0.0.0", port: int = 8520,
     cfg = Config(db_path=kw.pop("db_path", db),
                  payload_dir=str(data / "payloads"),
                  keystore_path=str(data / "keystore.key"),
-                 did_enabled=not no_did, **kw)
+                 did_enabled=not no_did,
+                 admin_grpc_port=int(os.environ.get(
+                     "AGENTFIELD_ADMIN_GRPC_PORT", port + 100)), **kw)
     cp = ControlPlane(cfg)
     typer.echo(f"agentfield-amd control plane on {host}:{port} (db={cfg.db_path})")
     uvicorn.run(create_app(cp), host=host, port=port, log_level="info")
@@ -153,6 +155,83 @@ def status():
     except Exception as e:
         typer.echo(f"control plane unreachable: {e}", err=True)
         raise typer.Exit(1)
+
+
+def _registry():
+    from .controlplane.packages import PackageRegistry
+    return PackageRegistry(os.environ.get("AGENTFIELD_DATA", ".agentfield"))
+
+
+def _procs():
+    from .controlplane.packages import ProcessManager
+    return ProcessManager(os.environ.get("AGENTFIELD_DATA", ".agentfield"))
+
+
+@app.command()
+def install(source: str, name: str = None):
+    """Install an agent package from a directory, archive or git repo."""
+    ent = _registry().install(source, name)
+    typer.echo(f"installed {ent['name']} -> {ent['path']}")
+
+
+@app.command()
+def add(source: str, name: str = None):
+    """Alias of install."""
+    install(source, name)
+
+
+@app.command()
+def uninstall(name: str):
+    if _registry().uninstall(name):
+        typer.echo(f"uninstalled {name}")
+    else:
+        typer.echo("not installed", err=True)
+        raise typer.Exit(1)
+
+
+@app.command("packages")
+def list_packages():
+    for p in _registry().list():
+        typer.echo(f"{p['name']:24s} {p['entrypoint']:12s} {p['path']}")
+
+
+@app.command()
+def start(name: str, agentfield_url: str = DEFAULT_URL, port: int = None):
+    """Start an installed agent package as a managed process."""
+    pkg = _registry().get(name)
+    if pkg is None:
+        typer.echo(f"package '{name}' not installed", err=True)
+        raise typer.Exit(1)
+    pm = _procs()
+    info = pm.start(pkg, agentfield_url, port)
+    ok = pm.wait_ready(name)
+    typer.echo(f"{name}: pid={info['pid']} {info['base_url']} "
+               f"{'ready' if ok else 'NOT READY (see logs)'}")
+
+
+@app.command()
+def stop(name: str):
+    """Stop a managed agent process."""
+    if _procs().stop(name):
+        typer.echo(f"stopped {name}")
+    else:
+        typer.echo(f"{name} was not running")
+
+
+@app.command()
+def logs(name: str, lines: int = 50):
+    """Tail a managed agent's log."""
+    typer.echo(_procs().logs(name, lines))
+
+
+@app.command()
+def ps():
+    """Status of managed agent processes."""
+    pm = _procs()
+    for pkg in _registry().list():
+        st = pm.status(pkg["name"])
+        state = f"pid={st['pid']} {st.get('base_url','')}" if st.get("running") else "stopped"
+        typer.echo(f"{pkg['name']:24s} {state}")
 
 
 vc_app = typer.Typer()

@@ -40,6 +40,7 @@ class Config:
         self.presence_ttl = kw.get("presence_ttl", 300.0)
         self.health_interval = kw.get("health_interval", 60.0)
         self.background_services = kw.get("background_services", True)
+        self.admin_grpc_port = kw.get("admin_grpc_port")
 
 
 class ControlPlane:
@@ -61,6 +62,7 @@ class ControlPlane:
         self.client: httpx.AsyncClient | None = None
         self._async_q: asyncio.Queue | None = None
         self._workers: list[asyncio.Task] = []
+        self._grpc_server = None
         self.started_at = time.time()
 
     async def start(self):
@@ -69,6 +71,10 @@ class ControlPlane:
         for _ in range(self.cfg.async_workers):
             self._workers.append(asyncio.create_task(self._async_worker()))
         await self.webhooks.start()
+        if self.cfg.admin_grpc_port:
+            from .admin_grpc import start_admin_grpc
+            self._grpc_server = start_admin_grpc(self,
+                                                 port=self.cfg.admin_grpc_port)
         if self.cfg.background_services:
             await self.presence.start()
             await self.health.start()
@@ -81,6 +87,8 @@ class ControlPlane:
         await self.presence.stop()
         await self.health.stop()
         await self.cleanup.stop()
+        if self._grpc_server is not None:
+            self._grpc_server.stop(grace=0.5)
         if self.client:
             await self.client.aclose()
 

@@ -31,6 +31,7 @@ from .client import AgentFieldClient
 from .execution_context import (ExecutionContext, current_context,
                                 reset_context, set_context)
 from .memory import MemoryInterface
+from .resilience import ResultCache, StatelessRateLimiter
 
 
 class _FunctionMeta:
@@ -86,6 +87,8 @@ class Agent(FastAPI):
         self.client = AgentFieldClient(self.agentfield_url)
         self.ai = AgentAI(ai_config)
         self.memory = MemoryInterface(self.client, node_id)
+        self.rate_limiter = StatelessRateLimiter()
+        self.result_cache = ResultCache()
         self.auto_register = auto_register
         self.heartbeat_interval = heartbeat_interval
         self.identity: dict = {}
@@ -143,13 +146,16 @@ class Agent(FastAPI):
     def reasoner(self, name: str | None = None, tags=None, vc: bool = False):
         return self._register_fn("reasoner", name, tags, vc)
 
-    def skill(self, name: str | None = None, tags=None):
-        return self._register_fn("skill", name, tags, False)
+    def skill(self, name: str | None = None, tags=None,
+              cache_results: bool = False):
+        return self._register_fn("skill", name, tags, False,
+                                 cache_results=cache_results)
 
-    def _register_fn(self, kind: str, name, tags, vc):
+    def _register_fn(self, kind: str, name, tags, vc, cache_results=False):
         def deco(fn):
             fname = name or fn.__name__
             meta = _FunctionMeta(fn, fname, kind, tags, vc)
+            meta.cache_results = cache_results
             table = self._reasoners if kind == "reasoner" else self._skills
             table[fname] = meta
             path = f"/{kind}s/{fname}"
@@ -183,9 +189,17 @@ class Agent(FastAPI):
             self.add_api_route(path, endpoint, methods=["POST"],
                                name=f"{kind}_{fname}")
 
-            # local call wrapper with workflow tracking
+            # local call wrapper with workflow tracking (+ optional cache)
             def local(*args, **kwargs):
                 bound = self._bind_args(meta, args, kwargs)
+                if getattr(meta, "cache_results", False):
+                    key = ResultCache.key_for(self.node_id, meta.name, bound)
+                    hit = self.result_cache.get(key)
+                    if hit is not None:
+                        return hit
+                    out = self._run_tracked_sync(meta, bound)
+                    self.result_cache.put(key, out)
+                    return out
                 return self._run_tracked_sync(meta, bound)
             local.__name__ = fn.__name__
             local.__wrapped__ = fn
@@ -298,7 +312,9 @@ class Agent(FastAPI):
         headers = ctx.child_headers() if ctx else {}
         if _async:
             return self.client.execute_async(target, kwargs, headers, _webhook)
-        resp = self.client.execute_sync(target, kwargs, headers, _webhook)
+        resp = self.rate_limiter.call(
+            self.client.execute_sync, target, kwargs, headers, _webhook,
+            retries=2)
         if resp.get("status") == "completed":
             result = resp.get("result")
             if isinstance(result, dict) and set(result) == {"result"}:
@@ -306,6 +322,20 @@ class Agent(FastAPI):
             return result
         raise RuntimeError(
             f"call {target} {resp.get('status')}: {resp.get('error_message')}")
+
+    # ---------------------------------------------------------------- MCP
+    def use_mcp(self, project_dir: str = ".", config: dict | None = None):
+        """Start the project's MCP servers and expose their tools as
+        auto-generated skills (reference P17/dynamic_skills)."""
+        from ..mcp import MCPManager
+        if not hasattr(self, "mcp"):
+            self.mcp = MCPManager()
+        if config:
+            for name, spec in config.items():
+                self.mcp.start_server(name, spec)
+        else:
+            self.mcp.start_all(project_dir)
+        return self.mcp.register_as_skills(self)
 
     # ------------------------------------------------- lifecycle / serve
     def register(self) -> bool:
@@ -332,11 +362,23 @@ class Agent(FastAPI):
                                                daemon=True, name="af-heartbeat")
             self._hb_thread.start()
 
+    def _deferred_start(self):
+        """Register once our own HTTP endpoint is up (runs in a thread;
+        newer starlette removed add_event_handler)."""
+        import httpx
+        for _ in range(200):
+            try:
+                httpx.get(self.base_url + "/health", timeout=0.5)
+                break
+            except httpx.HTTPError:
+                time.sleep(0.1)
+        self.start_background()
+
     def serve(self, host: str = "127.0.0.1", port: int = 8600, **uvicorn_kw):
         import uvicorn
         if self.base_url is None:
             self.base_url = f"http://{host}:{port}"
-        self.add_event_handler("startup", self.start_background)
+        threading.Thread(target=self._deferred_start, daemon=True).start()
         uvicorn.run(self, host=host, port=port, log_level="warning",
                     **uvicorn_kw)
 
