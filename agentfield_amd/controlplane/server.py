@@ -146,7 +146,13 @@ class ControlPlane:
             "_node": node,
             "_context": body.get("context"),
         }
-        self.storage.create_execution(rec)
+        inline, uri = self.payloads.maybe_offload(rec["id"], "input",
+                                                  rec["input"])
+        if uri:
+            stored = {**rec, "input": {"$payload_uri": uri}}
+            self.storage.create_execution(stored)
+        else:
+            self.storage.create_execution(rec)
         self.storage.upsert_run(run_id, st.RUNNING, rec["id"])
         if webhook:
             self.storage.register_webhook(rec["id"], webhook["url"],
@@ -187,6 +193,10 @@ class ControlPlane:
                            error: str | None = None,
                            duration_ms: float | None = None) -> dict | None:
         status = st.normalize(status)
+        if result is not None:
+            _, uri = self.payloads.maybe_offload(execution_id, "result", result)
+            if uri:
+                result = {"$payload_uri": uri}
         self.storage.update_execution_result(execution_id, status, result,
                                              error, duration_ms)
         rec = self.storage.get_execution(execution_id)
@@ -231,8 +241,17 @@ class ControlPlane:
             finally:
                 self.metrics.worker_inflight.dec()
 
-    @staticmethod
-    def envelope(rec: dict) -> dict:
+    def _resolve_payload(self, v):
+        if isinstance(v, dict) and "$payload_uri" in v:
+            try:
+                return self.payloads.load(v["$payload_uri"])
+            except Exception:
+                return v
+        return v
+
+    def envelope(self, rec: dict) -> dict:
+        rec = {**rec, "input": self._resolve_payload(rec.get("input")),
+               "result": self._resolve_payload(rec.get("result"))}
         return {
             "execution_id": rec["id"],
             "run_id": rec.get("run_id"),
@@ -715,5 +734,13 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
             return JSONResponse({"error": "not found"}, status_code=404)
         execs = cp.storage.executions_by_run(run_id)
         return {"run": run, "dag": build_dag(execs, lightweight=True)}
+
+    # --------------------------------------------------- embedded web UI
+    from pathlib import Path as _Path
+    _ui = _Path(__file__).parent / "ui" / "index.html"
+
+    @app.get("/")
+    async def ui_index():
+        return Response(_ui.read_text(), media_type="text/html")
 
     return app

@@ -29,6 +29,26 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
     async def health():
         return {"status": "healthy", "model": model_name}
 
+    @app.get("/metrics")
+    async def metrics():
+        from fastapi.responses import Response
+        from prometheus_client import (CollectorRegistry, Gauge,
+                                       generate_latest)
+        eng = runner.engine
+        reg = CollectorRegistry()
+        vals = {
+            "agentfield_engine_queued": eng.sched.num_queued(),
+            "agentfield_engine_running": eng.sched.num_running(),
+            "agentfield_engine_kv_free_pages": eng.sched.alloc.num_free,
+            "agentfield_engine_kv_total_pages": eng.sched.alloc.num_pages,
+            "agentfield_engine_prefill_tokens_total": eng.metrics["prefill_tokens"],
+            "agentfield_engine_decode_tokens_total": eng.metrics["decode_tokens"],
+            "agentfield_engine_steps_total": eng.metrics["steps"],
+        }
+        for name, v in vals.items():
+            Gauge(name, name, registry=reg).set(v)
+        return Response(generate_latest(reg), media_type="text/plain")
+
     @app.get("/v1/stats")
     async def stats():
         eng = runner.engine
