@@ -40,25 +40,44 @@ AF_EXPORT int af_reshape_and_cache(const void* k, const void* v, void* kc, void*
 
 // Gather rows of an i32 embedding-free path is not needed; but the engine uses
 // an embedding gather for input ids -> hidden states.
-// emb: [V, H] bf16, ids: [T] i32, out: [T, H] bf16
+// emb: [V, H] bf16, ids: [T] i32, out: [T, H] bf16.  ss (optional) gets the
+// per-row sum of squares, seeding the fused-norm decode path.
 __global__ void __launch_bounds__(256) embedding_kernel(
     u16* __restrict__ out, const u16* __restrict__ emb, const i32* __restrict__ ids,
-    int T, int H) {
+    float* __restrict__ ss, int T, int H) {
   const int per_row = H / 8;
+  __shared__ float red[4];
   for (int t = blockIdx.x; t < T; t += gridDim.x) {
     const u16* src = emb + (size_t)ids[t] * H;
     u16* dst = out + (size_t)t * H;
-    for (int i = threadIdx.x; i < per_row; i += blockDim.x)
-      *reinterpret_cast<s16x8*>(dst + i * 8) = *reinterpret_cast<const s16x8*>(src + i * 8);
+    float local = 0.f;
+    for (int i = threadIdx.x; i < per_row; i += blockDim.x) {
+      s16x8 v = *reinterpret_cast<const s16x8*>(src + i * 8);
+      *reinterpret_cast<s16x8*>(dst + i * 8) = v;
+      if (ss) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float f = bf2f((u16)v[j]);
+          local += f * f;
+        }
+      }
+    }
+    if (ss) {
+      local = wave_sum_f32(local);
+      if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = local;
+      __syncthreads();
+      if (threadIdx.x == 0) ss[t] = red[0] + red[1] + red[2] + red[3];
+      __syncthreads();
+    }
   }
 }
 
 AF_EXPORT int af_embedding(void* out, const void* emb, const void* ids,
-                           int T, int H, void* stream) {
+                           void* ss, int T, int H, void* stream) {
   if (H % 8) return 9001;
   if (T == 0) return 0;
   int blocks = T < 2048 ? T : 2048;
   embedding_kernel<<<blocks, 256, 0, (hipStream_t)stream>>>(
-      (u16*)out, (const u16*)emb, (const i32*)ids, T, H);
+      (u16*)out, (const u16*)emb, (const i32*)ids, (float*)ss, T, H);
   return af_last_err();
 }

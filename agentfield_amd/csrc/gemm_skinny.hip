@@ -44,16 +44,45 @@ __device__ __forceinline__ void stage64(const u16* __restrict__ g, size_t ld,
   }
 }
 
+// Reg-staged X tile with fused RMSNorm on load: f = x * rsqrt(ss[row]/H+eps)
+// * norm_w[k].  Writes the same swizzled layout stage64 produces.
+__device__ __forceinline__ void stage64_norm(const u16* __restrict__ g,
+                                             size_t ld, u16* lds, int k0,
+                                             int max_row,
+                                             const float* __restrict__ ss,
+                                             const u16* __restrict__ nw,
+                                             float inv_h, float eps) {
+  const int tid = threadIdx.x;
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    const int e = (it * 256 + tid) * 8;
+    const int row = e >> 6;
+    const int wb = (e & 63) * 2;
+    const int r = min(row, max_row - 1);
+    const s16x8 xv = *reinterpret_cast<const s16x8*>(g + (size_t)r * ld + k0 + (wb >> 1));
+    const s16x8 wv = *reinterpret_cast<const s16x8*>(nw + k0 + (wb >> 1));
+    const float rstd = rsqrtf(ss[r] * inv_h + eps);
+    s16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = (short)f2bf(bf2f((u16)xv[j]) * rstd * bf2f((u16)wv[j]));
+    *reinterpret_cast<s16x8*>(
+        reinterpret_cast<char*>(lds) + row * 128 + (wb ^ ((row & 7) << 4))) = o;
+  }
+}
+
 __device__ __forceinline__ s16x8 frag64(const u16* lds, int row, int byte) {
   return *reinterpret_cast<const s16x8*>(
       reinterpret_cast<const char*>(lds) + row * 128 + (byte ^ ((row & 7) << 4)));
 }
 
 // MT = number of 16-row M tiles (ceil(M/16)); grid (N/64, SK), block 256.
-template <int MT>
+template <int MT, bool NORM>
 __global__ void __launch_bounds__(256) gemm_skinny_kernel(
     float* __restrict__ partial, const u16* __restrict__ X,
-    const u16* __restrict__ W, int M, int N, int K, int Kc) {
+    const u16* __restrict__ W, int M, int N, int K, int Kc,
+    const float* __restrict__ norm_ss, const u16* __restrict__ norm_w,
+    float eps) {
   __shared__ u16 sW[2][64 * 64];
   __shared__ u16 sX[2][64 * 64];
   const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
@@ -69,14 +98,21 @@ __global__ void __launch_bounds__(256) gemm_skinny_kernel(
   // 2-phase pipeline (guide T3 minimum): stage tile t+1 while computing
   // tile t; the single __syncthreads (vmcnt(0)+barrier) at loop end drains
   // the in-flight loads after compute has covered their latency.
+  const float inv_h = 1.f / (float)K;
   int cur = 0;
   stage64(W, K, sW[0], nblk, k0, N);
-  stage64(X, K, sX[0], 0, k0, M);
+  if (NORM)
+    stage64_norm(X, K, sX[0], k0, M, norm_ss, norm_w, inv_h, eps);
+  else
+    stage64(X, K, sX[0], 0, k0, M);
   __syncthreads();
   for (int k = k0; k < k1; k += 64) {
     if (k + 64 < k1) {
       stage64(W, K, sW[cur ^ 1], nblk, k + 64, N);
-      stage64(X, K, sX[cur ^ 1], 0, k + 64, M);
+      if (NORM)
+        stage64_norm(X, K, sX[cur ^ 1], k + 64, M, norm_ss, norm_w, inv_h, eps);
+      else
+        stage64(X, K, sX[cur ^ 1], 0, k + 64, M);
     }
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
@@ -154,9 +190,49 @@ __global__ void __launch_bounds__(256) gemm_skinny_combine_kernel(
   }
 }
 
+// mode 4: one block per row: out = sum + residual (residual updated in
+// place) AND ss_out[m] = sum over the row of out^2 — feeds the NEXT skinny
+// GEMM's fused input RMSNorm, eliminating standalone rmsnorm kernels at
+// decode.
+__global__ void __launch_bounds__(256) gemm_skinny_combine_row_kernel(
+    u16* __restrict__ out, u16* __restrict__ residual,
+    const float* __restrict__ partial, float* __restrict__ ss_out,
+    int M, int N, int SK) {
+  const int m = blockIdx.x;
+  __shared__ float red[4];
+  float local = 0.f;
+  for (int c = threadIdx.x * 4; c < N; c += 256 * 4) {
+    float v[4] = {0, 0, 0, 0};
+    for (int s = 0; s < SK; ++s) {
+      const float* p = partial + ((size_t)s * M + m) * N + c;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) v[j] += p[j];
+    }
+    s16x4 rv = *reinterpret_cast<const s16x4*>(residual + (size_t)m * N + c);
+    s16x4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float h = v[j] + bf2f((u16)rv[j]);
+      o[j] = (short)f2bf(h);
+      const float hb = bf2f((u16)o[j]);  // stats on the bf16-rounded value
+      local += hb * hb;
+    }
+    *reinterpret_cast<s16x4*>(out + (size_t)m * N + c) = o;
+    *reinterpret_cast<s16x4*>(residual + (size_t)m * N + c) = o;
+  }
+  local = wave_sum_f32(local);
+  const int wid2 = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) red[wid2] = local;
+  __syncthreads();
+  if (threadIdx.x == 0)
+    ss_out[m] = red[0] + red[1] + red[2] + red[3];
+}
+
 AF_EXPORT int af_gemm_skinny(void* out, void* partial, void* residual,
                              const void* X, const void* W, int M, int N,
-                             int K, int SK, int mode, void* stream) {
+                             int K, int SK, int mode,
+                             const void* norm_ss, const void* norm_w,
+                             float eps, void* ss_out, void* stream) {
   if (M < 1 || M > 64) return 9005;
   if (N % 64 || K % 64) return 9006;
   hipStream_t st = (hipStream_t)stream;
@@ -164,15 +240,27 @@ AF_EXPORT int af_gemm_skinny(void* out, void* partial, void* residual,
   while ((SK - 1) * Kc >= K) --SK;  // drop empty splits
   dim3 grid(N / 64, SK), blk(256);
   const int MT = (M + 15) / 16;
-  switch (MT) {
-    case 1: gemm_skinny_kernel<1><<<grid, blk, 0, st>>>(
-        (float*)partial, (const u16*)X, (const u16*)W, M, N, K, Kc); break;
-    case 2: gemm_skinny_kernel<2><<<grid, blk, 0, st>>>(
-        (float*)partial, (const u16*)X, (const u16*)W, M, N, K, Kc); break;
-    case 3: gemm_skinny_kernel<3><<<grid, blk, 0, st>>>(
-        (float*)partial, (const u16*)X, (const u16*)W, M, N, K, Kc); break;
-    default: gemm_skinny_kernel<4><<<grid, blk, 0, st>>>(
-        (float*)partial, (const u16*)X, (const u16*)W, M, N, K, Kc); break;
+  const bool norm = norm_ss != nullptr;
+#define AF_SK_LAUNCH(MTV, NV)                                                  \
+  gemm_skinny_kernel<MTV, NV><<<grid, blk, 0, st>>>(                           \
+      (float*)partial, (const u16*)X, (const u16*)W, M, N, K, Kc,              \
+      (const float*)norm_ss, (const u16*)norm_w, eps)
+  switch (MT * 2 + (norm ? 1 : 0)) {
+    case 2: AF_SK_LAUNCH(1, false); break;
+    case 3: AF_SK_LAUNCH(1, true); break;
+    case 4: AF_SK_LAUNCH(2, false); break;
+    case 5: AF_SK_LAUNCH(2, true); break;
+    case 6: AF_SK_LAUNCH(3, false); break;
+    case 7: AF_SK_LAUNCH(3, true); break;
+    case 8: AF_SK_LAUNCH(4, false); break;
+    default: AF_SK_LAUNCH(4, true); break;
+  }
+#undef AF_SK_LAUNCH
+  if (mode == 4) {
+    gemm_skinny_combine_row_kernel<<<M, 256, 0, st>>>(
+        (u16*)out, (u16*)residual, (const float*)partial, (float*)ss_out,
+        M, N, SK);
+    return af_last_err();
   }
   const int cols = (mode == 2) ? N / 2 : N;
   i64 total = (i64)M * (cols / 4);

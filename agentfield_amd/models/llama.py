@@ -122,9 +122,9 @@ class LlamaAttention(nn.Module):
         self.o = nn.Parameter(torch.empty(H, cfg.q_size))
         self.scale = 1.0 / math.sqrt(D)
 
-    def forward(self, x, positions, rope_tab, kv: KVCache, md: AttnMetadata):
+    def attend(self, qkv, positions, rope_tab, kv: KVCache, md: AttnMetadata):
+        """RoPE + cache append + attention over the fused QKV buffer."""
         cfg = self.cfg
-        qkv = ops.linear(x, self.qkv)
         # strided views straight into the fused projection (no copies)
         q = qkv[:, :cfg.q_size]
         k = qkv[:, cfg.q_size:cfg.q_size + cfg.kv_size]
@@ -132,12 +132,15 @@ class LlamaAttention(nn.Module):
         ops.rope_cache(q, k, v, positions, rope_tab,
                        kv.k[self.layer_idx], kv.v[self.layer_idx], md.slots)
         if md.is_prefill:
-            o = ops.attn_prefill(q, k, v, md.cu_seqlens, md.seq_lens,
-                                 self.scale, head_dim=cfg.head_dim)
-        else:
-            o = ops.attn_decode(q, kv.k[self.layer_idx], kv.v[self.layer_idx],
-                                md.block_table, md.seq_lens_t, self.scale,
-                                nsplit=md.nsplit, scratch=md.decode_scratch)
+            return ops.attn_prefill(q, k, v, md.cu_seqlens, md.seq_lens,
+                                    self.scale, head_dim=cfg.head_dim)
+        return ops.attn_decode(q, kv.k[self.layer_idx], kv.v[self.layer_idx],
+                               md.block_table, md.seq_lens_t, self.scale,
+                               nsplit=md.nsplit, scratch=md.decode_scratch)
+
+    def forward(self, x, positions, rope_tab, kv: KVCache, md: AttnMetadata):
+        qkv = ops.linear(x, self.qkv)
+        o = self.attend(qkv, positions, rope_tab, kv, md)
         return ops.linear(o, self.o)
 
 
@@ -173,6 +176,22 @@ class LlamaLayer(nn.Module):
         h, residual = ops.rmsnorm(h, self.post_norm, self.eps, residual)
         h = self.mlp(h)
         return h, residual
+
+    def forward_decode_fused(self, residual, ss, ss2, positions, rope_tab,
+                             kv, md):
+        """Decode fast path: RMSNorms fold into the skinny GEMMs — input
+        normalization happens in the GEMM's X staging (from per-row
+        sum-of-squares stats), residual-add + next-norm stats in the GEMM
+        combine.  Zero standalone norm kernels per layer."""
+        at, mlp = self.attn, self.mlp
+        qkv = ops.linear_skinny(residual, at.qkv,
+                                norm=(ss, self.input_norm, self.eps))
+        o = at.attend(qkv, positions, rope_tab, kv, md)
+        ops.linear_skinny(o, at.o, mode=4, residual=residual, ss_out=ss2)
+        act = ops.linear_skinny(residual, mlp.gate_up, mode=2,
+                                norm=(ss2, self.post_norm, self.eps))
+        ops.linear_skinny(act, mlp.down, mode=4, residual=residual, ss_out=ss)
+        return residual, ss
 
 
 class LlamaForCausalLM(nn.Module):

@@ -105,14 +105,19 @@ def reshape_and_cache(k: torch.Tensor, v: torch.Tensor, kcache: torch.Tensor,
     _lib.check(rc, "af_reshape_and_cache")
 
 
-def embedding(ids: torch.Tensor, table: torch.Tensor) -> torch.Tensor:
+def embedding(ids: torch.Tensor, table: torch.Tensor,
+              ss: torch.Tensor | None = None) -> torch.Tensor:
     if not _on_gpu(table):
-        return table[ids.long()]
+        out = table[ids.long()]
+        if ss is not None:
+            ss.copy_(out.float().pow(2).sum(-1))
+        return out
     T = ids.numel()
     V, H = table.shape
     out = torch.empty(T, H, dtype=table.dtype, device=table.device)
     rc = _lib.lib().af_embedding(_lib.ptr(out), _lib.ptr(table),
-                                 _lib.ptr(ids.int()), T, H, _lib.cur_stream())
+                                 _lib.ptr(ids.int()), _lib.ptr(ss), T, H,
+                                 _lib.cur_stream())
     _lib.check(rc, "af_embedding")
     return out
 
@@ -205,10 +210,17 @@ def choose_sk(N: int, K: int) -> int:
 
 def linear_skinny(x: torch.Tensor, w: torch.Tensor, mode: int = 0,
                   residual: torch.Tensor | None = None,
-                  sk: int | None = None) -> torch.Tensor:
+                  sk: int | None = None,
+                  norm: tuple | None = None,
+                  ss_out: torch.Tensor | None = None) -> torch.Tensor:
     """Decode-path GEMM (M<=64) via the MFMA weight-streaming kernel.
+
     mode 0: plain; mode 1: +residual (in-place update); mode 2: fused SwiGLU
-    (w holds [gate|up] rows, returns [M, N/2])."""
+    (w holds [gate|up] rows, returns [M, N/2]); mode 4: +residual AND
+    per-row sum-of-squares into ss_out (feeds the next GEMM's fused norm).
+    norm=(ss, norm_weight, eps) applies RMSNorm to x rows while staging
+    (x is consumed unnormalized — no standalone rmsnorm kernel needed).
+    """
     M, K = x.shape
     N = w.shape[0]
     if sk is None:
@@ -216,9 +228,16 @@ def linear_skinny(x: torch.Tensor, w: torch.Tensor, mode: int = 0,
     cols = N // 2 if mode == 2 else N
     out = torch.empty(M, cols, dtype=x.dtype, device=x.device)
     partial = torch.empty(sk, M, N, dtype=torch.float32, device=x.device)
+    if norm is not None:
+        ss, nw, eps = norm
+        ss_p, nw_p = _lib.ptr(ss), _lib.ptr(nw)
+    else:
+        ss_p = nw_p = _lib.ptr(None)
+        eps = 0.0
     rc = _lib.lib().af_gemm_skinny(
         _lib.ptr(out), _lib.ptr(partial), _lib.ptr(residual), _lib.ptr(x),
-        _lib.ptr(w), M, N, K, sk, mode, _lib.cur_stream())
+        _lib.ptr(w), M, N, K, sk, mode, ss_p, nw_p, eps, _lib.ptr(ss_out),
+        _lib.cur_stream())
     _lib.check(rc, "af_gemm_skinny")
     return out
 

@@ -48,13 +48,14 @@ def _declare(l: ctypes.CDLL) -> None:
     l.af_silu_mul.argtypes = [p, p, i64, i64, p]
     l.af_add.argtypes = [p, p, p, i64, p]
     l.af_reshape_and_cache.argtypes = [p, p, p, p, p, i, i, i, i, p]
-    l.af_embedding.argtypes = [p, p, p, i, i, p]
+    l.af_embedding.argtypes = [p, p, p, p, i, i, p]
     l.af_attn_decode.argtypes = [p, p, p, p, p, p, p, p, f, i, i, i, i, i, i, i,
                                  i64, p]
     l.af_attn_prefill.argtypes = [p, p, p, p, p, p, p, f, i, i, i, i,
                                   i64, i64, i64, p]
     l.af_gemm_bf16.argtypes = [p, p, p, i, i, i, p]
-    l.af_gemm_skinny.argtypes = [p, p, p, p, p, i, i, i, i, i, p]
+    l.af_gemm_skinny.argtypes = [p, p, p, p, p, i, i, i, i, i,
+                                 p, p, f, p, p]
     l.af_sample.argtypes = [p, p, p, p, p, p, u32, i, i, p]
     l.af_gather_rows.argtypes = [p, p, p, i, i, p]
     l.af_mfma_probe.argtypes = [p, p, p, p]

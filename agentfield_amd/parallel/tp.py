@@ -106,6 +106,7 @@ def build_tp_model(full_cfg: LlamaConfig, tp: int, rank: int, device,
         for layer in model.layers:
             layer.attn.register_forward_hook(hook)
             layer.mlp.register_forward_hook(hook)
+        model.no_fused_decode = True  # collectives attach to module forward
     return model
 
 
