@@ -224,9 +224,30 @@ class LlamaForCausalLM(nn.Module):
                 p.normal_(0.0, 0.02, generator=g)
         return self
 
+    def _can_fuse_decode(self, T: int) -> bool:
+        import os
+        if os.environ.get("AF_NO_FUSED_DECODE") == "1":
+            return False
+        cfg = self.cfg
+        return (not getattr(self, "no_fused_decode", False) and T <= 64
+                and cfg.hidden_size % 64 == 0
+                and cfg.intermediate_size % 64 == 0
+                and (cfg.q_size + 2 * cfg.kv_size) % 64 == 0
+                and cfg.q_size % 64 == 0)
+
     def forward(self, ids, positions, kv: KVCache, md: AttnMetadata,
                 logit_rows: torch.Tensor | None = None):
         """ids/positions [T] -> logits [T or len(logit_rows), vocab]."""
+        T = ids.numel()
+        if not md.is_prefill and ids.is_cuda and self._can_fuse_decode(T):
+            ss = torch.empty(T, dtype=torch.float32, device=ids.device)
+            ss2 = torch.empty_like(ss)
+            residual = ops.embedding(ids, self.embed, ss=ss)
+            for layer in self.layers:
+                residual, ss = layer.forward_decode_fused(
+                    residual, ss, ss2, positions, self.rope_tab, kv, md)
+            h = ops.rmsnorm(residual, self.final_norm, self.cfg.rms_eps)
+            return ops.linear(h, self.lm_head)
         h = ops.embedding(ids, self.embed)
         residual = None
         for layer in self.layers:
