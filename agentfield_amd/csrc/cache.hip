@@ -66,7 +66,9 @@ __global__ void __launch_bounds__(256) embedding_kernel(
       local = wave_sum_f32(local);
       if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = local;
       __syncthreads();
-      if (threadIdx.x == 0) ss[t] = red[0] + red[1] + red[2] + red[3];
+      if (threadIdx.x < 8)  // [T][8] column-block layout; total in slot 0
+        ss[t * 8 + threadIdx.x] =
+            threadIdx.x == 0 ? red[0] + red[1] + red[2] + red[3] : 0.f;
       __syncthreads();
     }
   }

@@ -61,7 +61,12 @@ __device__ __forceinline__ void stage64_norm(const u16* __restrict__ g,
     const int r = min(row, max_row - 1);
     const s16x8 xv = *reinterpret_cast<const s16x8*>(g + (size_t)r * ld + k0 + (wb >> 1));
     const s16x8 wv = *reinterpret_cast<const s16x8*>(nw + k0 + (wb >> 1));
-    const float rstd = rsqrtf(ss[r] * inv_h + eps);
+    // ss is [M][8] column-block partials (written without atomics by the
+    // producing combine kernel); tiny and L1-hot
+    float st = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) st += ss[r * 8 + j];
+    const float rstd = rsqrtf(st * inv_h + eps);
     s16x8 o;
 #pragma unroll
     for (int j = 0; j < 8; ++j)
@@ -190,18 +195,21 @@ __global__ void __launch_bounds__(256) gemm_skinny_combine_kernel(
   }
 }
 
-// mode 4: one block per row: out = sum + residual (residual updated in
-// place) AND ss_out[m] = sum over the row of out^2 — feeds the NEXT skinny
-// GEMM's fused input RMSNorm, eliminating standalone rmsnorm kernels at
-// decode.
+// mode 4: grid (M, 8): block (m, cb) combines column block cb of row m:
+// out = sum + residual (residual updated in place) AND
+// ss_out[m*8+cb] = partial sum of out^2 over the block — together the 8
+// partials feed the NEXT skinny GEMM's fused input RMSNorm, eliminating
+// standalone rmsnorm kernels at decode.
 __global__ void __launch_bounds__(256) gemm_skinny_combine_row_kernel(
     u16* __restrict__ out, u16* __restrict__ residual,
     const float* __restrict__ partial, float* __restrict__ ss_out,
     int M, int N, int SK) {
   const int m = blockIdx.x;
+  const int nb = N / 8;          // N % 64 == 0 => nb % 4 == 0... (N/8 cols)
+  const int c0 = blockIdx.y * nb;
   __shared__ float red[4];
   float local = 0.f;
-  for (int c = threadIdx.x * 4; c < N; c += 256 * 4) {
+  for (int c = c0 + threadIdx.x * 4; c < c0 + nb; c += 256 * 4) {
     float v[4] = {0, 0, 0, 0};
     for (int s = 0; s < SK; ++s) {
       const float* p = partial + ((size_t)s * M + m) * N + c;
@@ -225,7 +233,7 @@ __global__ void __launch_bounds__(256) gemm_skinny_combine_row_kernel(
   if ((threadIdx.x & 63) == 0) red[wid2] = local;
   __syncthreads();
   if (threadIdx.x == 0)
-    ss_out[m] = red[0] + red[1] + red[2] + red[3];
+    ss_out[m * 8 + blockIdx.y] = red[0] + red[1] + red[2] + red[3];
 }
 
 AF_EXPORT int af_gemm_skinny(void* out, void* partial, void* residual,
@@ -257,7 +265,9 @@ AF_EXPORT int af_gemm_skinny(void* out, void* partial, void* residual,
   }
 #undef AF_SK_LAUNCH
   if (mode == 4) {
-    gemm_skinny_combine_row_kernel<<<M, 256, 0, st>>>(
+    if (N % 32) return 9007;
+    dim3 g4(M, 8);
+    gemm_skinny_combine_row_kernel<<<g4, 256, 0, st>>>(
         (u16*)out, (u16*)residual, (const float*)partial, (float*)ss_out,
         M, N, SK);
     return af_last_err();

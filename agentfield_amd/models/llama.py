@@ -240,7 +240,7 @@ class LlamaForCausalLM(nn.Module):
         """ids/positions [T] -> logits [T or len(logit_rows), vocab]."""
         T = ids.numel()
         if not md.is_prefill and ids.is_cuda and self._can_fuse_decode(T):
-            ss = torch.empty(T, dtype=torch.float32, device=ids.device)
+            ss = torch.empty(T, 8, dtype=torch.float32, device=ids.device)
             ss2 = torch.empty_like(ss)
             residual = ops.embedding(ids, self.embed, ss=ss)
             for layer in self.layers:

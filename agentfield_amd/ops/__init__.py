@@ -110,7 +110,8 @@ def embedding(ids: torch.Tensor, table: torch.Tensor,
     if not _on_gpu(table):
         out = table[ids.long()]
         if ss is not None:
-            ss.copy_(out.float().pow(2).sum(-1))
+            ss.zero_()
+            ss.view(-1, 8)[:, 0] = out.float().pow(2).sum(-1)
         return out
     T = ids.numel()
     V, H = table.shape
