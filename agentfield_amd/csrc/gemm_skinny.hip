@@ -49,9 +49,8 @@ __device__ __forceinline__ void stage64(const u16* __restrict__ g, size_t ld,
 __device__ __forceinline__ void stage64_norm(const u16* __restrict__ g,
                                              size_t ld, u16* lds, int k0,
                                              int max_row,
-                                             const float* __restrict__ ss,
-                                             const u16* __restrict__ nw,
-                                             float inv_h, float eps) {
+                                             const float* __restrict__ rstd,
+                                             const u16* __restrict__ nw) {
   const int tid = threadIdx.x;
 #pragma unroll
   for (int it = 0; it < 2; ++it) {
@@ -61,16 +60,11 @@ __device__ __forceinline__ void stage64_norm(const u16* __restrict__ g,
     const int r = min(row, max_row - 1);
     const s16x8 xv = *reinterpret_cast<const s16x8*>(g + (size_t)r * ld + k0 + (wb >> 1));
     const s16x8 wv = *reinterpret_cast<const s16x8*>(nw + k0 + (wb >> 1));
-    // ss is [M][8] column-block partials (written without atomics by the
-    // producing combine kernel); tiny and L1-hot
-    float st = 0.f;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) st += ss[r * 8 + j];
-    const float rstd = rsqrtf(st * inv_h + eps);
+    const float rs = rstd[r];
     s16x8 o;
 #pragma unroll
     for (int j = 0; j < 8; ++j)
-      o[j] = (short)f2bf(bf2f((u16)xv[j]) * rstd * bf2f((u16)wv[j]));
+      o[j] = (short)f2bf(bf2f((u16)xv[j]) * rs * bf2f((u16)wv[j]));
     *reinterpret_cast<s16x8*>(
         reinterpret_cast<char*>(lds) + row * 128 + (wb ^ ((row & 7) << 4))) = o;
   }
@@ -103,11 +97,22 @@ __global__ void __launch_bounds__(256) gemm_skinny_kernel(
   // 2-phase pipeline (guide T3 minimum): stage tile t+1 while computing
   // tile t; the single __syncthreads (vmcnt(0)+barrier) at loop end drains
   // the in-flight loads after compute has covered their latency.
-  const float inv_h = 1.f / (float)K;
+  // per-row rstd precomputed once from the [M][8] column-block stats
+  __shared__ float sRstd[64];
+  if (NORM) {
+    if (threadIdx.x < 64) {
+      const int r = min((int)threadIdx.x, M - 1);
+      float st = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) st += norm_ss[r * 8 + j];
+      sRstd[threadIdx.x] = rsqrtf(st / (float)K + eps);
+    }
+    __syncthreads();
+  }
   int cur = 0;
   stage64(W, K, sW[0], nblk, k0, N);
   if (NORM)
-    stage64_norm(X, K, sX[0], k0, M, norm_ss, norm_w, inv_h, eps);
+    stage64_norm(X, K, sX[0], k0, M, sRstd, norm_w);
   else
     stage64(X, K, sX[0], 0, k0, M);
   __syncthreads();
@@ -115,7 +120,7 @@ __global__ void __launch_bounds__(256) gemm_skinny_kernel(
     if (k + 64 < k1) {
       stage64(W, K, sW[cur ^ 1], nblk, k + 64, N);
       if (NORM)
-        stage64_norm(X, K, sX[cur ^ 1], k + 64, M, norm_ss, norm_w, inv_h, eps);
+        stage64_norm(X, K, sX[cur ^ 1], k + 64, M, sRstd, norm_w);
       else
         stage64(X, K, sX[cur ^ 1], 0, k + 64, M);
     }
