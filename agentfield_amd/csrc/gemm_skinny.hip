@@ -44,44 +44,16 @@ __device__ __forceinline__ void stage64(const u16* __restrict__ g, size_t ld,
   }
 }
 
-// Reg-staged X tile with fused RMSNorm on load: f = x * rsqrt(ss[row]/H+eps)
-// * norm_w[k].  Writes the same swizzled layout stage64 produces.
-__device__ __forceinline__ void stage64_norm(const u16* __restrict__ g,
-                                             size_t ld, u16* lds, int k0,
-                                             int max_row,
-                                             const float* __restrict__ rstd,
-                                             const u16* __restrict__ nw) {
-  const int tid = threadIdx.x;
-#pragma unroll
-  for (int it = 0; it < 2; ++it) {
-    const int e = (it * 256 + tid) * 8;
-    const int row = e >> 6;
-    const int wb = (e & 63) * 2;
-    const int r = min(row, max_row - 1);
-    const s16x8 xv = *reinterpret_cast<const s16x8*>(g + (size_t)r * ld + k0 + (wb >> 1));
-    const s16x8 wv = *reinterpret_cast<const s16x8*>(nw + k0 + (wb >> 1));
-    const float rs = rstd[r];
-    s16x8 o;
-#pragma unroll
-    for (int j = 0; j < 8; ++j)
-      o[j] = (short)f2bf(bf2f((u16)xv[j]) * rs * bf2f((u16)wv[j]));
-    *reinterpret_cast<s16x8*>(
-        reinterpret_cast<char*>(lds) + row * 128 + (wb ^ ((row & 7) << 4))) = o;
-  }
-}
-
 __device__ __forceinline__ s16x8 frag64(const u16* lds, int row, int byte) {
   return *reinterpret_cast<const s16x8*>(
       reinterpret_cast<const char*>(lds) + row * 128 + (byte ^ ((row & 7) << 4)));
 }
 
 // MT = number of 16-row M tiles (ceil(M/16)); grid (N/64, SK), block 256.
-template <int MT, bool NORM>
+template <int MT>
 __global__ void __launch_bounds__(256) gemm_skinny_kernel(
     float* __restrict__ partial, const u16* __restrict__ X,
-    const u16* __restrict__ W, int M, int N, int K, int Kc,
-    const float* __restrict__ norm_ss, const u16* __restrict__ norm_w,
-    float eps) {
+    const u16* __restrict__ W, int M, int N, int K, int Kc) {
   __shared__ u16 sW[2][64 * 64];
   __shared__ u16 sX[2][64 * 64];
   const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
@@ -97,32 +69,14 @@ __global__ void __launch_bounds__(256) gemm_skinny_kernel(
   // 2-phase pipeline (guide T3 minimum): stage tile t+1 while computing
   // tile t; the single __syncthreads (vmcnt(0)+barrier) at loop end drains
   // the in-flight loads after compute has covered their latency.
-  // per-row rstd precomputed once from the [M][8] column-block stats
-  __shared__ float sRstd[64];
-  if (NORM) {
-    if (threadIdx.x < 64) {
-      const int r = min((int)threadIdx.x, M - 1);
-      float st = 0.f;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) st += norm_ss[r * 8 + j];
-      sRstd[threadIdx.x] = rsqrtf(st / (float)K + eps);
-    }
-    __syncthreads();
-  }
   int cur = 0;
   stage64(W, K, sW[0], nblk, k0, N);
-  if (NORM)
-    stage64_norm(X, K, sX[0], k0, M, sRstd, norm_w);
-  else
-    stage64(X, K, sX[0], 0, k0, M);
+  stage64(X, K, sX[0], 0, k0, M);
   __syncthreads();
   for (int k = k0; k < k1; k += 64) {
     if (k + 64 < k1) {
       stage64(W, K, sW[cur ^ 1], nblk, k + 64, N);
-      if (NORM)
-        stage64_norm(X, K, sX[cur ^ 1], k + 64, M, sRstd, norm_w);
-      else
-        stage64(X, K, sX[cur ^ 1], 0, k + 64, M);
+      stage64(X, K, sX[cur ^ 1], 0, k + 64, M);
     }
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
@@ -158,7 +112,8 @@ __global__ void __launch_bounds__(256) gemm_skinny_kernel(
 // mode 2: N = 2I; out[M,I] bf16 = silu(sum[:, :I]) * sum[:, I:]
 __global__ void __launch_bounds__(256) gemm_skinny_combine_kernel(
     u16* __restrict__ out, u16* __restrict__ residual,
-    const float* __restrict__ partial, int M, int N, int SK, int mode) {
+    const float* __restrict__ partial, int M, int N, int SK, int mode,
+    const float* __restrict__ scale_ss, float inv_k, float eps) {
   const int cols = (mode == 2) ? (N >> 1) : N;
   const i64 total = (i64)M * (cols >> 2);  // 4 outputs per thread
   const i64 stride = (i64)gridDim.x * blockDim.x;
@@ -172,6 +127,15 @@ __global__ void __launch_bounds__(256) gemm_skinny_combine_kernel(
 #pragma unroll
       for (int j = 0; j < 4; ++j) v[j] += p[j];
     }
+    float rstd = 1.f;
+    if (scale_ss) {
+      float st = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) st += scale_ss[m * 8 + j];
+      rstd = rsqrtf(st * inv_k + eps);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) v[j] *= rstd;
+    }
     if (mode == 2) {
       float u[4] = {0, 0, 0, 0};
       for (int s = 0; s < SK; ++s) {
@@ -179,6 +143,8 @@ __global__ void __launch_bounds__(256) gemm_skinny_combine_kernel(
 #pragma unroll
         for (int j = 0; j < 4; ++j) u[j] += p[j];
       }
+#pragma unroll
+      for (int j = 0; j < 4; ++j) u[j] *= rstd;
       s16x4 o;
 #pragma unroll
       for (int j = 0; j < 4; ++j)
@@ -244,8 +210,8 @@ __global__ void __launch_bounds__(256) gemm_skinny_combine_row_kernel(
 AF_EXPORT int af_gemm_skinny(void* out, void* partial, void* residual,
                              const void* X, const void* W, int M, int N,
                              int K, int SK, int mode,
-                             const void* norm_ss, const void* norm_w,
-                             float eps, void* ss_out, void* stream) {
+                             const void* scale_ss, float eps, void* ss_out,
+                             void* stream) {
   if (M < 1 || M > 64) return 9005;
   if (N % 64 || K % 64) return 9006;
   hipStream_t st = (hipStream_t)stream;
@@ -253,20 +219,14 @@ AF_EXPORT int af_gemm_skinny(void* out, void* partial, void* residual,
   while ((SK - 1) * Kc >= K) --SK;  // drop empty splits
   dim3 grid(N / 64, SK), blk(256);
   const int MT = (M + 15) / 16;
-  const bool norm = norm_ss != nullptr;
-#define AF_SK_LAUNCH(MTV, NV)                                                  \
-  gemm_skinny_kernel<MTV, NV><<<grid, blk, 0, st>>>(                           \
-      (float*)partial, (const u16*)X, (const u16*)W, M, N, K, Kc,              \
-      (const float*)norm_ss, (const u16*)norm_w, eps)
-  switch (MT * 2 + (norm ? 1 : 0)) {
-    case 2: AF_SK_LAUNCH(1, false); break;
-    case 3: AF_SK_LAUNCH(1, true); break;
-    case 4: AF_SK_LAUNCH(2, false); break;
-    case 5: AF_SK_LAUNCH(2, true); break;
-    case 6: AF_SK_LAUNCH(3, false); break;
-    case 7: AF_SK_LAUNCH(3, true); break;
-    case 8: AF_SK_LAUNCH(4, false); break;
-    default: AF_SK_LAUNCH(4, true); break;
+#define AF_SK_LAUNCH(MTV)                                                      \
+  gemm_skinny_kernel<MTV><<<grid, blk, 0, st>>>(                               \
+      (float*)partial, (const u16*)X, (const u16*)W, M, N, K, Kc)
+  switch (MT) {
+    case 1: AF_SK_LAUNCH(1); break;
+    case 2: AF_SK_LAUNCH(2); break;
+    case 3: AF_SK_LAUNCH(3); break;
+    default: AF_SK_LAUNCH(4); break;
   }
 #undef AF_SK_LAUNCH
   if (mode == 4) {
@@ -282,6 +242,7 @@ AF_EXPORT int af_gemm_skinny(void* out, void* partial, void* residual,
   int blocks = (int)((total + 255) / 256);
   if (blocks > 1024) blocks = 1024;
   gemm_skinny_combine_kernel<<<blocks, 256, 0, st>>>(
-      (u16*)out, (u16*)residual, (const float*)partial, M, N, SK, mode);
+      (u16*)out, (u16*)residual, (const float*)partial, M, N, SK, mode,
+      (const float*)scale_ss, 1.f / (float)K, eps);
   return af_last_err();
 }

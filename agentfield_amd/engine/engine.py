@@ -52,6 +52,9 @@ class LLMEngine:
         if model is None:
             model = LlamaForCausalLM(cfg, device=device, dtype=dtype).init_random(seed)
         self.model = model.eval()
+        if self.device.type == "cuda" and tp_group is None and \
+                not getattr(model, "no_fused_decode", False):
+            self.model.fold_norm_weights()
 
         if num_pages is None:
             if self.is_gpu:

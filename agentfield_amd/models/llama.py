@@ -184,12 +184,11 @@ class LlamaLayer(nn.Module):
         sum-of-squares stats), residual-add + next-norm stats in the GEMM
         combine.  Zero standalone norm kernels per layer."""
         at, mlp = self.attn, self.mlp
-        qkv = ops.linear_skinny(residual, at.qkv,
-                                norm=(ss, self.input_norm, self.eps))
+        qkv = ops.linear_skinny(residual, at.qkv, scale=(ss, self.eps))
         o = at.attend(qkv, positions, rope_tab, kv, md)
         ops.linear_skinny(o, at.o, mode=4, residual=residual, ss_out=ss2)
         act = ops.linear_skinny(residual, mlp.gate_up, mode=2,
-                                norm=(ss2, self.post_norm, self.eps))
+                                scale=(ss2, self.eps))
         ops.linear_skinny(act, mlp.down, mode=4, residual=residual, ss_out=ss)
         return residual, ss
 
@@ -224,9 +223,35 @@ class LlamaForCausalLM(nn.Module):
                 p.normal_(0.0, 0.02, generator=g)
         return self
 
+    @torch.no_grad()
+    def fold_norm_weights(self):
+        """Fold each RMSNorm's elementwise weight into the projection that
+        consumes it (qkv <- input_norm, gate_up <- post_norm, lm_head <-
+        final_norm), leaving unit norm weights.  Mathematically identical;
+        lets the fused decode path apply the norm as a per-row output
+        scalar in the GEMM combine, keeping the async weight stream."""
+        if getattr(self, "_norms_folded", False):
+            return self
+        for layer in self.layers:
+            layer.attn.qkv.copy_(
+                (layer.attn.qkv.float() * layer.input_norm.float()
+                 ).to(layer.attn.qkv.dtype))
+            layer.input_norm.fill_(1.0)
+            layer.mlp.gate_up.copy_(
+                (layer.mlp.gate_up.float() * layer.post_norm.float()
+                 ).to(layer.mlp.gate_up.dtype))
+            layer.post_norm.fill_(1.0)
+        self.lm_head.copy_((self.lm_head.float() * self.final_norm.float()
+                            ).to(self.lm_head.dtype))
+        self.final_norm.fill_(1.0)
+        self._norms_folded = True
+        return self
+
     def _can_fuse_decode(self, T: int) -> bool:
         import os
         if os.environ.get("AF_NO_FUSED_DECODE") == "1":
+            return False
+        if not getattr(self, "_norms_folded", False):
             return False
         cfg = self.cfg
         return (not getattr(self, "no_fused_decode", False) and T <= 64

@@ -212,15 +212,17 @@ def choose_sk(N: int, K: int) -> int:
 def linear_skinny(x: torch.Tensor, w: torch.Tensor, mode: int = 0,
                   residual: torch.Tensor | None = None,
                   sk: int | None = None,
-                  norm: tuple | None = None,
+                  scale: tuple | None = None,
                   ss_out: torch.Tensor | None = None) -> torch.Tensor:
     """Decode-path GEMM (M<=64) via the MFMA weight-streaming kernel.
 
     mode 0: plain; mode 1: +residual (in-place update); mode 2: fused SwiGLU
     (w holds [gate|up] rows, returns [M, N/2]); mode 4: +residual AND
-    per-row sum-of-squares into ss_out (feeds the next GEMM's fused norm).
-    norm=(ss, norm_weight, eps) applies RMSNorm to x rows while staging
-    (x is consumed unnormalized — no standalone rmsnorm kernel needed).
+    per-row sum-of-squares stats into ss_out [M,8].
+    scale=(ss, eps) applies the RMSNorm row scalar rsqrt(mean(x^2)+eps) on
+    the OUTPUT (valid when the norm weight is folded into w — see
+    LlamaForCausalLM.fold_norm_weights); x itself stays unnormalized so the
+    weight stream keeps its async global_load_lds pipeline.
     """
     M, K = x.shape
     N = w.shape[0]
@@ -229,15 +231,15 @@ def linear_skinny(x: torch.Tensor, w: torch.Tensor, mode: int = 0,
     cols = N // 2 if mode == 2 else N
     out = torch.empty(M, cols, dtype=x.dtype, device=x.device)
     partial = torch.empty(sk, M, N, dtype=torch.float32, device=x.device)
-    if norm is not None:
-        ss, nw, eps = norm
-        ss_p, nw_p = _lib.ptr(ss), _lib.ptr(nw)
+    if scale is not None:
+        ss, eps = scale
+        ss_p = _lib.ptr(ss)
     else:
-        ss_p = nw_p = _lib.ptr(None)
+        ss_p = _lib.ptr(None)
         eps = 0.0
     rc = _lib.lib().af_gemm_skinny(
         _lib.ptr(out), _lib.ptr(partial), _lib.ptr(residual), _lib.ptr(x),
-        _lib.ptr(w), M, N, K, sk, mode, ss_p, nw_p, eps, _lib.ptr(ss_out),
+        _lib.ptr(w), M, N, K, sk, mode, ss_p, eps, _lib.ptr(ss_out),
         _lib.cur_stream())
     _lib.check(rc, "af_gemm_skinny")
     return out

@@ -55,7 +55,7 @@ def _declare(l: ctypes.CDLL) -> None:
                                   i64, i64, i64, p]
     l.af_gemm_bf16.argtypes = [p, p, p, i, i, i, p]
     l.af_gemm_skinny.argtypes = [p, p, p, p, p, i, i, i, i, i,
-                                 p, p, f, p, p]
+                                 p, f, p, p]
     l.af_sample.argtypes = [p, p, p, p, p, p, u32, i, i, p]
     l.af_gather_rows.argtypes = [p, p, p, i, i, p]
     l.af_mfma_probe.argtypes = [p, p, p, p]

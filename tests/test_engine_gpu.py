@@ -45,6 +45,7 @@ def test_fused_decode_matches_standard_logits():
 
     cfg = CONFIGS["tiny"]
     model = LlamaForCausalLM(cfg, device="cuda").init_random(5)
+    model.fold_norm_weights()
     kv = KVCache(cfg, 64, 16, "cuda")
     T = 6
     torch.manual_seed(0)
@@ -87,3 +88,36 @@ def test_fused_decode_generate_matches():
         del eng
         torch.cuda.empty_cache()
     assert outs[True] == outs[False]
+
+
+def test_norm_folding_preserves_logits():
+    """Folding norm weights into projections must not change the math."""
+    import torch
+    from agentfield_amd.models import CONFIGS
+    from agentfield_amd.models.llama import (AttnMetadata, KVCache,
+                                             LlamaForCausalLM)
+    cfg = CONFIGS["tiny"]
+    T = 5
+    torch.manual_seed(1)
+    ids = torch.randint(0, cfg.vocab_size, (T,), dtype=torch.int32,
+                        device="cuda")
+    pos = torch.arange(T, dtype=torch.int32, device="cuda")
+    md = AttnMetadata(is_prefill=True,
+                      slots=torch.arange(T, dtype=torch.int64, device="cuda"),
+                      cu_seqlens=torch.tensor([0, T], dtype=torch.int32,
+                                              device="cuda"), seq_lens=[T])
+    m = LlamaForCausalLM(cfg, device="cuda").init_random(7)
+    # make norm weights non-trivial so folding is actually exercised
+    with torch.no_grad():
+        for layer in m.layers:
+            layer.input_norm.normal_(1.0, 0.1)
+            layer.post_norm.normal_(1.0, 0.1)
+        m.final_norm.normal_(1.0, 0.1)
+    kv1 = KVCache(cfg, 64, 16, "cuda")
+    base = m(ids, pos, kv1, md).float()
+    m.fold_norm_weights()
+    kv2 = KVCache(cfg, 64, 16, "cuda")
+    folded = m(ids, pos, kv2, md).float()
+    torch.cuda.synchronize()
+    diff = (base - folded).abs().max().item()
+    assert diff < 0.2, f"norm folding changed logits by {diff}"
