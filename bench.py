@@ -78,7 +78,8 @@ def main():
         import torch.distributed as dist_mod
         dist = dist_mod
         dist.init_process_group("nccl" if torch.cuda.is_available() else "gloo")
-        torch.cuda.set_device(local_rank)
+        if torch.cuda.is_available():
+            torch.cuda.set_device(local_rank)
 
     device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
     cfg = CONFIGS[args.model]
