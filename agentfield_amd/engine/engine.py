@@ -89,6 +89,8 @@ class LLMEngine:
             "bt": torch.zeros(B, max_pages_per_seq, dtype=torch.int32, device=dev),
             "lens": torch.ones(B, dtype=torch.int32, device=dev),
             "temps": torch.zeros(B, dtype=torch.float32, device=dev),
+            "topk": torch.zeros(B, dtype=torch.int32, device=dev),
+            "topp": torch.ones(B, dtype=torch.float32, device=dev),
             "tokens": torch.zeros(B, dtype=torch.int32, device=dev),
         }
         self._host = {k: torch.zeros_like(v, device="cpu").pin_memory()
@@ -175,7 +177,14 @@ class LLMEngine:
         logits = self.model(ids_t, pos_t, self.kv, md, logit_rows=last_rows)
         temps = torch.tensor([s.sampling.temperature for s in seqs],
                              dtype=torch.float32, device=dev)
-        toks = ops.sample(logits, temps, self.sampler)
+        kw = {}
+        if any(s.sampling.top_k > 0 or s.sampling.top_p < 1.0 for s in seqs) \
+                and dev.type == "cuda":
+            kw = {"topk": torch.tensor([s.sampling.top_k for s in seqs],
+                                       dtype=torch.int32, device=dev),
+                  "topp": torch.tensor([s.sampling.top_p for s in seqs],
+                                       dtype=torch.float32, device=dev)}
+        toks = ops.sample(logits, temps, self.sampler, **kw)
         return toks.cpu().tolist()
 
     # -- decode path (graph-captured on GPU) --------------------------------
@@ -190,6 +199,8 @@ class LLMEngine:
             npg = len(seq.pages)
             h["bt"][i, :npg] = torch.tensor(seq.pages, dtype=torch.int32)
             h["temps"][i] = seq.sampling.temperature
+            h["topk"][i] = seq.sampling.top_k
+            h["topp"][i] = seq.sampling.top_p
         for i in range(len(seqs), bs):  # dummy lanes -> null page 0
             h["ids"][i] = 0
             h["pos"][i] = 0
@@ -197,19 +208,23 @@ class LLMEngine:
             h["lens"][i] = 1
             h["bt"][i, 0] = 0
             h["temps"][i] = 0.0
+            h["topk"][i] = 0
+            h["topp"][i] = 1.0
         d = self._dec
         nb = self.is_gpu
-        for k in ("ids", "pos", "slots", "lens", "temps"):
+        for k in ("ids", "pos", "slots", "lens", "temps", "topk", "topp"):
             d[k][:bs].copy_(h[k][:bs], non_blocking=nb)
         d["bt"][:bs].copy_(h["bt"][:bs], non_blocking=nb)
 
-    def _decode_forward(self, bs: int, nsplit: int, scratch):
+    def _decode_forward(self, bs: int, nsplit: int, scratch, tkp: bool):
         d = self._dec
         md = AttnMetadata(is_prefill=False, slots=d["slots"][:bs],
                           block_table=d["bt"][:bs], seq_lens_t=d["lens"][:bs],
                           nsplit=nsplit, decode_scratch=scratch)
         logits = self.model(d["ids"][:bs], d["pos"][:bs], self.kv, md)
-        ops.sample(logits, d["temps"][:bs], self.sampler, out=d["tokens"][:bs])
+        kw = {"topk": d["topk"][:bs], "topp": d["topp"][:bs]} if tkp else {}
+        ops.sample(logits, d["temps"][:bs], self.sampler,
+                   out=d["tokens"][:bs], **kw)
 
     def _make_scratch(self, bs: int, nsplit: int):
         if nsplit <= 1:
@@ -222,8 +237,9 @@ class LLMEngine:
                           dtype=torch.float32, device=self.device)
         return po, pml
 
-    def _get_graph(self, bs: int):
-        entry = self._graphs.get(bs)
+    def _get_graph(self, bs: int, tkp: bool):
+        key = (bs, tkp)
+        entry = self._graphs.get(key)
         if entry is not None:
             return entry
         nsplit = choose_nsplit(bs, self.cfg.num_kv_heads)
@@ -233,13 +249,13 @@ class LLMEngine:
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(2):
-                self._decode_forward(bs, nsplit, scratch)
+                self._decode_forward(bs, nsplit, scratch, tkp)
         torch.cuda.current_stream().wait_stream(s)
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
-            self._decode_forward(bs, nsplit, scratch)
+            self._decode_forward(bs, nsplit, scratch, tkp)
         entry = {"graph": g, "nsplit": nsplit, "scratch": scratch}
-        self._graphs[bs] = entry
+        self._graphs[key] = entry
         return entry
 
     def _step_decode(self, batch: ScheduleBatch) -> list[int]:
@@ -247,12 +263,13 @@ class LLMEngine:
         n = len(seqs)
         self.metrics["decode_tokens"] += n
         bs = _bucket_for(n, self.max_num_seqs)
+        tkp = any(s.sampling.top_k > 0 or s.sampling.top_p < 1.0 for s in seqs)
         self._fill_decode_buffers(seqs, bs)
         if self.enable_graphs:
-            self._get_graph(bs)["graph"].replay()
+            self._get_graph(bs, tkp)["graph"].replay()
         else:
             nsplit = choose_nsplit(bs, self.cfg.num_kv_heads) if self.is_gpu else 1
-            self._decode_forward(bs, nsplit, self._make_scratch(bs, nsplit))
+            self._decode_forward(bs, nsplit, self._make_scratch(bs, nsplit), tkp)
         toks = self._dec["tokens"][:n]
         if self.is_gpu:
             self._host["tokens"][:n].copy_(toks, non_blocking=True)

@@ -10,6 +10,8 @@ from typing import Callable
 class SamplingParams:
     max_tokens: int = 128
     temperature: float = 0.0
+    top_k: int = 0          # 0 = disabled
+    top_p: float = 1.0      # >= 1 = disabled
     stop_token_ids: tuple = ()
     ignore_eos: bool = False
 

@@ -286,11 +286,20 @@ class SamplerState:
         self.pi = torch.empty(max_b, 16, dtype=torch.int32, device=device)
         self.step = torch.zeros(1, dtype=torch.int32, device=device)
         self.seed = seed
+        # top-k/top-p scratch (histogram threshold selection)
+        self.rmax = torch.empty(max_b, dtype=torch.float32, device=device)
+        self.thresh = torch.empty(max_b, dtype=torch.float32, device=device)
+        self.hist_n = torch.empty(max_b, 256, dtype=torch.int32, device=device)
+        self.hist_m = torch.empty(max_b, 256, dtype=torch.float32, device=device)
 
 
 def sample(logits: torch.Tensor, temps: torch.Tensor, state: SamplerState,
-           out: torch.Tensor | None = None) -> torch.Tensor:
-    """Greedy (temp==0) or Gumbel-max temperature sampling.  logits [B,V] bf16."""
+           out: torch.Tensor | None = None,
+           topk: torch.Tensor | None = None,
+           topp: torch.Tensor | None = None) -> torch.Tensor:
+    """Greedy (temp==0) or Gumbel-max temperature sampling; optional
+    top-k/top-p nucleus restriction (sort-free histogram threshold).
+    logits [B,V] bf16; topk [B] i32 (0 = off); topp [B] f32 (>=1 = off)."""
     B, V = logits.shape
     if not _on_gpu(logits):
         res = ref.sample_greedy(logits)
@@ -300,6 +309,19 @@ def sample(logits: torch.Tensor, temps: torch.Tensor, state: SamplerState,
         return res
     if out is None:
         out = torch.empty(B, dtype=torch.int32, device=logits.device)
+    if topk is not None or topp is not None:
+        if topk is None:
+            topk = torch.zeros(B, dtype=torch.int32, device=logits.device)
+        if topp is None:
+            topp = torch.ones(B, dtype=torch.float32, device=logits.device)
+        rc = _lib.lib().af_sample_topkp(
+            _lib.ptr(out), _lib.ptr(state.pv), _lib.ptr(state.pi),
+            _lib.ptr(state.rmax), _lib.ptr(state.thresh),
+            _lib.ptr(state.hist_n), _lib.ptr(state.hist_m), _lib.ptr(logits),
+            _lib.ptr(temps), _lib.ptr(topk), _lib.ptr(topp),
+            _lib.ptr(state.step), state.seed, B, V, _lib.cur_stream())
+        _lib.check(rc, "af_sample_topkp")
+        return out
     rc = _lib.lib().af_sample(
         _lib.ptr(out), _lib.ptr(state.pv), _lib.ptr(state.pi), _lib.ptr(logits),
         _lib.ptr(temps), _lib.ptr(state.step), state.seed, B, V, _lib.cur_stream())

@@ -280,3 +280,41 @@ def test_embedding_and_gather():
     g = ops.gather_rows(out, rows)
     torch.cuda.synchronize()
     assert close(g, out.cpu()[rows.cpu().long()], atol=0)
+
+
+def test_sample_topk_topp():
+    torch.manual_seed(7)
+    B, V = 4, 50000
+    logits = rnd(B, V, seed=7, scale=3.0)
+    st = ops.SamplerState(B, DEV)
+    temps = torch.full((B,), 1.0, dtype=torch.float32, device=DEV)
+    # top_k=1 must equal argmax regardless of temperature
+    topk = torch.ones(B, dtype=torch.int32, device=DEV)
+    topp = torch.ones(B, dtype=torch.float32, device=DEV)
+    t = ops.sample(logits, temps, st, topk=topk, topp=topp)
+    torch.cuda.synchronize()
+    assert t.cpu().tolist() == logits.float().argmax(-1).cpu().tolist()
+    # top_k=8: every draw must come from the exact top-8 set
+    topk8 = torch.full((B,), 8, dtype=torch.int32, device=DEV)
+    top8 = logits.float().topk(8, dim=-1).indices.cpu()
+    for _ in range(10):
+        t = ops.sample(logits, temps, st, topk=topk8, topp=topp)
+        torch.cuda.synchronize()
+        for b in range(B):
+            assert int(t[b]) in top8[b].tolist(), f"draw outside top-8 (b={b})"
+    # tiny top_p -> argmax
+    topp_tiny = torch.full((B,), 1e-6, dtype=torch.float32, device=DEV)
+    topk0 = torch.zeros(B, dtype=torch.int32, device=DEV)
+    t = ops.sample(logits, temps, st, topk=topk0, topp=topp_tiny)
+    torch.cuda.synchronize()
+    assert t.cpu().tolist() == logits.float().argmax(-1).cpu().tolist()
+
+
+def test_engine_topk_sampling_runs():
+    from agentfield_amd.engine import LLMEngine, SamplingParams
+    from agentfield_amd.models import CONFIGS
+    eng = LLMEngine(CONFIGS["tiny"], device="cuda", page_size=4, num_pages=128,
+                    max_num_seqs=4, enable_graphs=True, seed=2)
+    outs = eng.generate([[1, 2, 3]], SamplingParams(
+        max_tokens=6, temperature=0.9, top_k=10, top_p=0.9, ignore_eos=True))
+    assert len(outs[0]) == 6
