@@ -45,22 +45,42 @@ def sha256_hex(data: bytes) -> str:
 
 
 class Keystore:
-    """File keystore holding the master seed (0600 perms)."""
+    """File keystore holding the master seed (0600 perms).  When
+    AGENTFIELD_KEYSTORE_KEY (base64url, 32 bytes) is set the seed is
+    AES-256-GCM encrypted at rest via the native libcrypto extension
+    (reference parity: C23 keystore encryption)."""
 
-    def __init__(self, path: str | None):
+    def __init__(self, path: str | None, kek: bytes | None = None):
         self.path = Path(path) if path else None
+        env = os.environ.get("AGENTFIELD_KEYSTORE_KEY")
+        self.kek = kek or (_b64u_dec(env) if env else None)
         self._seed: bytes | None = None
+
+    def _encode(self, seed: bytes) -> str:
+        if self.kek:
+            return "enc:" + _b64u(bytes(_crypto.aes_gcm_encrypt(self.kek, seed)))
+        return _b64u(seed)
+
+    def _decode(self, text: str) -> bytes:
+        text = text.strip()
+        if text.startswith("enc:"):
+            if not self.kek:
+                raise ValueError("keystore is encrypted; set "
+                                 "AGENTFIELD_KEYSTORE_KEY")
+            return bytes(_crypto.aes_gcm_decrypt(self.kek,
+                                                 _b64u_dec(text[4:])))
+        return _b64u_dec(text)
 
     @property
     def seed(self) -> bytes:
         if self._seed is None:
             if self.path and self.path.exists():
-                self._seed = _b64u_dec(self.path.read_text().strip())
+                self._seed = self._decode(self.path.read_text())
             else:
                 self._seed = secrets.token_bytes(32)
                 if self.path:
                     self.path.parent.mkdir(parents=True, exist_ok=True)
-                    self.path.write_text(_b64u(self._seed))
+                    self.path.write_text(self._encode(self._seed))
                     os.chmod(self.path, 0o600)
         return self._seed
 

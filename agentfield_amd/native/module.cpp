@@ -7,6 +7,7 @@
 #include <pybind11/stl.h>
 
 #include <openssl/evp.h>
+#include <openssl/rand.h>
 
 #include <deque>
 #include <stdexcept>
@@ -61,6 +62,58 @@ static bool ed25519_verify(py::bytes pub, py::bytes msg, py::bytes sig) {
   EVP_MD_CTX_free(ctx);
   EVP_PKEY_free(k);
   return ok;
+}
+
+// ------------------------------- AES-256-GCM -------------------------------
+static py::bytes aes_gcm_encrypt(py::bytes key, py::bytes plaintext) {
+  std::string k = key, pt = plaintext;
+  if (k.size() != 32) throw std::invalid_argument("key must be 32 bytes");
+  unsigned char iv[12];
+  if (RAND_bytes(iv, 12) != 1) throw std::runtime_error("RAND_bytes failed");
+  std::string out(12 + pt.size() + 16, '\0');
+  memcpy(&out[0], iv, 12);
+  EVP_CIPHER_CTX* ctx = EVP_CIPHER_CTX_new();
+  int len = 0, total = 0;
+  bool ok = EVP_EncryptInit_ex(ctx, EVP_aes_256_gcm(), nullptr,
+                               (const unsigned char*)k.data(), iv) == 1 &&
+            EVP_EncryptUpdate(ctx, (unsigned char*)&out[12], &len,
+                              (const unsigned char*)pt.data(),
+                              (int)pt.size()) == 1;
+  total = len;
+  ok = ok && EVP_EncryptFinal_ex(ctx, (unsigned char*)&out[12 + total], &len) == 1;
+  total += len;
+  unsigned char tag[16];
+  ok = ok && EVP_CIPHER_CTX_ctrl(ctx, EVP_CTRL_GCM_GET_TAG, 16, tag) == 1;
+  EVP_CIPHER_CTX_free(ctx);
+  if (!ok) throw std::runtime_error("aes-gcm encrypt failed");
+  memcpy(&out[12 + total], tag, 16);
+  out.resize(12 + total + 16);
+  return py::bytes(out);
+}
+
+static py::bytes aes_gcm_decrypt(py::bytes key, py::bytes blob) {
+  std::string k = key, b = blob;
+  if (k.size() != 32) throw std::invalid_argument("key must be 32 bytes");
+  if (b.size() < 28) throw std::invalid_argument("ciphertext too short");
+  const unsigned char* iv = (const unsigned char*)b.data();
+  const unsigned char* ct = iv + 12;
+  size_t ctlen = b.size() - 12 - 16;
+  const unsigned char* tag = (const unsigned char*)b.data() + b.size() - 16;
+  std::string out(ctlen, '\0');
+  EVP_CIPHER_CTX* ctx = EVP_CIPHER_CTX_new();
+  int len = 0;
+  bool ok = EVP_DecryptInit_ex(ctx, EVP_aes_256_gcm(), nullptr,
+                               (const unsigned char*)k.data(), iv) == 1 &&
+            EVP_DecryptUpdate(ctx, (unsigned char*)&out[0], &len, ct,
+                              (int)ctlen) == 1 &&
+            EVP_CIPHER_CTX_ctrl(ctx, EVP_CTRL_GCM_SET_TAG, 16,
+                                (void*)tag) == 1;
+  int fin = 0;
+  ok = ok && EVP_DecryptFinal_ex(ctx, (unsigned char*)&out[0] + len, &fin) == 1;
+  EVP_CIPHER_CTX_free(ctx);
+  if (!ok) throw std::runtime_error("aes-gcm decrypt failed (bad key or tampered)");
+  out.resize(len + fin);
+  return py::bytes(out);
 }
 
 // ------------------------- scheduler / allocator ---------------------------
@@ -227,6 +280,8 @@ PYBIND11_MODULE(_native, m) {
   m.def("ed25519_pubkey", &ed25519_pubkey);
   m.def("ed25519_sign", &ed25519_sign);
   m.def("ed25519_verify", &ed25519_verify);
+  m.def("aes_gcm_encrypt", &aes_gcm_encrypt);
+  m.def("aes_gcm_decrypt", &aes_gcm_decrypt);
 
   py::class_<ScheduleResult>(m, "ScheduleResult")
       .def_readonly("has_work", &ScheduleResult::has_work)

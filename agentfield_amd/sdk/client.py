@@ -78,17 +78,46 @@ class AgentFieldClient:
         return r.json()
 
     def wait_for_result(self, execution_id: str, timeout: float = 300.0,
-                        poll_initial: float = 0.05, poll_max: float = 2.0) -> dict:
-        """Adaptive polling (age-based backoff, reference async_config.py)."""
+                        poll_initial: float = 0.05, poll_max: float = 2.0,
+                        use_sse: bool = True) -> dict:
+        """Adaptive polling with an SSE nudge: a background listener on
+        /api/ui/v1/executions/events wakes the poll loop the moment the
+        terminal event fires (reference async_execution_manager.py:644)."""
+        import threading
         deadline = time.time() + timeout
         poll = poll_initial
-        while time.time() < deadline:
-            rec = self.get_execution(execution_id)
-            if rec and rec.get("status") in ("completed", "failed", "timeout",
-                                             "cancelled"):
-                return rec
-            time.sleep(poll)
-            poll = min(poll * 1.5, poll_max)
+        nudge = threading.Event()
+        stop_sse = threading.Event()
+
+        def sse_listener():
+            try:
+                with httpx.stream(
+                        "GET", f"{self.base_url}/api/ui/v1/executions/events",
+                        timeout=timeout) as resp:
+                    for line in resp.iter_lines():
+                        if stop_sse.is_set():
+                            return
+                        if line.startswith("data:") and execution_id in line:
+                            nudge.set()
+                            return
+            except httpx.HTTPError:
+                pass  # SSE is best-effort; polling still covers us
+
+        if use_sse:
+            threading.Thread(target=sse_listener, daemon=True).start()
+        try:
+            while time.time() < deadline:
+                rec = self.get_execution(execution_id)
+                if rec and rec.get("status") in ("completed", "failed",
+                                                 "timeout", "cancelled"):
+                    return rec
+                if nudge.wait(poll):
+                    nudge.clear()
+                    poll = poll_initial  # event fired: check immediately
+                else:
+                    poll = min(poll * 1.5, poll_max)
+        finally:
+            stop_sse.set()
         raise TimeoutError(f"execution {execution_id} did not finish")
 
     def report_status(self, execution_id: str, status: str, result=None,

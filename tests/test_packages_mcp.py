@@ -221,3 +221,52 @@ def test_admin_grpc():
         assert ns["nodes"][0]["id"] == "n1"
     finally:
         server.stop(grace=0)
+
+
+def test_keystore_aes_gcm_encryption(tmp_path):
+    import base64
+    import secrets as _s
+    from agentfield_amd.controlplane.did import Keystore
+    kek = _s.token_bytes(32)
+    ks = Keystore(str(tmp_path / "k.key"), kek=kek)
+    seed1 = ks.seed
+    raw = (tmp_path / "k.key").read_text()
+    assert raw.startswith("enc:")
+    assert base64.urlsafe_b64encode(seed1).decode().rstrip("=") not in raw
+    ks2 = Keystore(str(tmp_path / "k.key"), kek=kek)
+    assert ks2.seed == seed1
+    ks_bad = Keystore(str(tmp_path / "k.key"), kek=_s.token_bytes(32))
+    import pytest as _pt
+    with _pt.raises(Exception):
+        _ = ks_bad.seed
+
+
+def test_wait_for_result_sse_nudge():
+    from agentfield_amd.controlplane import ControlPlane, create_app
+    from agentfield_amd.controlplane.server import Config
+    from agentfield_amd.sdk.client import AgentFieldClient
+    from helpers import AppServer
+    import threading
+    cp = ControlPlane(Config(background_services=False))
+    srv = AppServer(create_app(cp)).start().wait_healthy()
+    try:
+        cp.storage.create_execution({"id": "exec_sse", "run_id": "run_sse",
+                                     "input": {}})
+        cl = AgentFieldClient(srv.base_url)
+
+        def finish_later():
+            time.sleep(0.5)
+            # completing publishes the SSE event that nudges the waiter
+            import asyncio
+            cp.complete_execution("exec_sse", "completed", {"ok": 1})
+
+        threading.Thread(target=finish_later, daemon=True).start()
+        t0 = time.time()
+        rec = cl.wait_for_result("exec_sse", timeout=20.0, poll_initial=5.0,
+                                 poll_max=5.0)
+        waited = time.time() - t0
+        assert rec["status"] == "completed"
+        # with 5s polls, finishing in ~<2s proves the SSE nudge woke us
+        assert waited < 4.0, f"SSE nudge did not wake poll loop ({waited:.1f}s)"
+    finally:
+        srv.stop()
