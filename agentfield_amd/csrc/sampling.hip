@@ -159,6 +159,10 @@ __global__ void __launch_bounds__(64) tkp_threshold_kernel(
   for (int i = 0; i < TKP_BINS; ++i) total += hist_m[b * TKP_BINS + i];
   const int k = (topk[b] > 0) ? topk[b] : V;
   const float p = (topp[b] > 0.f && topp[b] < 1.f) ? topp[b] : 1.f;
+  if (k == 1) {  // exact: only the argmax survives
+    thresh[b] = mx;
+    return;
+  }
   int cnt = 0;
   float mass = 0.f;
   float th = mx - TKP_SPAN;
