@@ -96,6 +96,8 @@ class LLMEngine:
         self._host = {k: torch.zeros_like(v, device="cpu").pin_memory()
                       if self.is_gpu else torch.zeros_like(v)
                       for k, v in self._dec.items()}
+        # block-table dirty tracking: row -> (seq_id, pages_written)
+        self._bt_rows: dict[int, tuple] = {}
 
     # ------------------------------------------------------------- requests
     def add_request(self, prompt_ids: list[int], sampling: SamplingParams | None = None,
