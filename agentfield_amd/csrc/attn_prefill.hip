@@ -1,7 +1,8 @@
 // Causal varlen prefill attention (flash-style, GQA) with MFMA 16x16x32 bf16.
 //
-// q/k/v : [total_T, H*, D=128] bf16 packed by sequence (cu_seqlens-style),
-//         post-RoPE.  out: [total_T, Hq, D] bf16.
+// q/k/v : rows at strides qs/ks/vs (views into the fused QKV buffer), head h
+//         at offset h*D, D=128, packed by sequence (cu_seqlens), post-RoPE.
+// out   : [total_T, Hq*D] bf16 contiguous.
 // Host precomputes a flat tile map (tile_seq[i], tile_q0[i]): workgroup i on
 // grid.x handles q rows [q0, q0+ROWS) of sequence tile_seq[i]; grid.y = kv head.
 //
@@ -9,25 +10,27 @@
 // query head in the GQA group:
 //   G>=4 : all waves share one 16-row tile; wave w covers heads g=w, w+4, ...
 //   G<4  : 4/G row-tiles per workgroup; wave w -> (g = w%G, tile = w/G)
-// K/V tiles (KVBLK=32 tokens) are staged in LDS once per workgroup and shared.
+// K/V tiles (KVBLK=64 tokens) are staged in LDS once per workgroup and shared
+// by the 4 waves; 64-token tiles halve the barrier count per token vs 32.
 //
 // CDNA4 specifics (see /opt/skills/guides/cdna_hip_programming.md):
 //  - K LDS tile is XOR-swizzled (byte ^= (row&7)<<4) so the B-fragment
-//    ds_read_b128 is bank-conflict-free (guide §6 G4: row-major [32][128]
+//    ds_read_b128 is bank-conflict-free (guide §6 G4: row-major [64][128]
 //    would be a 32-way conflict).
-//  - V is stored transposed [D][KVBLK+pad] so the PV B-fragment read is a
-//    contiguous 16 B ds_read; pad 32->40 spreads banks.
+//  - V is stored transposed [D][KVBLK+pad] with token pairs packed as u32
+//    ds_writes (2 tokens per 4 B slot) so staging is 8 b32 writes per two
+//    16 B loads; pad 64->72 spreads banks on the PV-fragment reads.
 //  - P (scores) round-trips through a small per-wave LDS tile to convert the
 //    MFMA C-layout into the A-fragment layout.
-//  - fragment layouts (guide §3, measured):
+//  - fragment layouts (guide §3, measured, validated by ops.mfma_probe):
 //      A: lane holds A[m=lane&15][k=(lane>>4)*8+j], j=0..7
 //      B: lane holds B[k=(lane>>4)*8+j][n=lane&15]
 //      C: lane reg r holds C[m=(lane>>4)*4+r][n=lane&15]
 #include "common.h"
 
 #define AP_D 128
-#define AP_KVBLK 32
-#define AP_VPAD 40  // 32 tokens padded to 40 (80 B row stride)
+#define AP_KVBLK 64
+#define AP_VPAD 72  // 64 tokens padded to 72 (144 B row stride)
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 
@@ -60,17 +63,12 @@ __global__ void __launch_bounds__(256) attn_prefill_kernel(
   const int my_tile = (G >= 4) ? 0 : (wid / G);
   const int g0 = (G >= 4) ? wid : (wid % G);
   const int q0 = tile_q0[tile] + my_tile * 16;        // abs q row of wave tile
-  if (q0 >= len) {
-    // whole wave's tile out of range; still must participate in staging
-    // barriers, so fall through with masked rows (q0 clamp below).
-  }
 
   __shared__ u16 k_lds[AP_KVBLK * AP_D];              // swizzled rows
-  __shared__ u16 v_lds[AP_D][AP_VPAD];                // transposed
+  __shared__ u16 v_lds[AP_D][AP_VPAD];                // transposed, tok-paired
   __shared__ u16 p_lds[4][16][AP_VPAD];               // per-wave P tile
 
   // ---- load Q fragments (stay in registers for all KV tiles) ----
-  // lane holds Q[q0 + (lane&15)][c*32 + (lane>>4)*8 .. +8] for c=0..3
   const int qrow_frag = min(q0 + (lane & 15), len - 1);
   s16x8 q_frag[HLOOP][4];
 #pragma unroll
@@ -92,39 +90,49 @@ __global__ void __launch_bounds__(256) attn_prefill_kernel(
     for (int dt = 0; dt < 8; ++dt) o_acc[hl][dt] = f32x4{0.f, 0.f, 0.f, 0.f};
   }
 
-  const int kv_end = min(len, tile_q0[tile] + ROWS);  // causal upper bound (max over WG waves)
+  const int kv_end = min(len, tile_q0[tile] + ROWS);  // causal upper bound
   const int n_kv_tiles = (kv_end + AP_KVBLK - 1) / AP_KVBLK;
 
   for (int kt = 0; kt < n_kv_tiles; ++kt) {
     const int kv0 = kt * AP_KVBLK;
     __syncthreads();  // previous iteration's frag reads done
-    // ---- stage K (swizzled) and V (transposed) ----
-    // 32 tokens x 128 dims / 8 per chunk = 512 chunks; 256 threads x 2 iters
+    // ---- stage K (swizzled): 64 tok x 128 d / 8 = 1024 chunks, 4/thread ----
     for (int i = threadIdx.x; i < AP_KVBLK * (AP_D / 8); i += 256) {
       const int tok = i / (AP_D / 8);
       const int d8 = (i % (AP_D / 8)) * 8;
       const int tg = kv0 + tok;
       const size_t kvrow = (size_t)(seq_start + min(tg, len - 1));
-      {  // K: row-major swizzled, garbage beyond len is masked later
-        s16x8 kv8 = lds_read8(k + kvrow * ks + kvh * AP_D + d8);
-        const int byte = tok * (AP_D * 2) + ((d8 * 2) ^ ((tok & 7) << 4));
-        *reinterpret_cast<s16x8*>(reinterpret_cast<char*>(k_lds) + byte) = kv8;
-      }
-      {  // V: transposed, zero-filled beyond len (0 * P avoids NaN)
-        s16x8 vv8 = (tg < len) ? lds_read8(v + kvrow * vs + kvh * AP_D + d8)
-                               : s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      s16x8 kv8 = lds_read8(k + kvrow * ks + kvh * AP_D + d8);
+      const int byte = tok * (AP_D * 2) + ((d8 * 2) ^ ((tok & 7) << 4));
+      *reinterpret_cast<s16x8*>(reinterpret_cast<char*>(k_lds) + byte) = kv8;
+    }
+    // ---- stage V transposed: token pairs -> u32 writes (8 per 2 loads) ----
+    for (int i = threadIdx.x; i < (AP_KVBLK / 2) * (AP_D / 8); i += 256) {
+      const int tp = i / (AP_D / 8);          // token pair
+      const int d8 = (i % (AP_D / 8)) * 8;
+      const int t0g = kv0 + tp * 2;
+      const size_t r0 = (size_t)(seq_start + min(t0g, len - 1));
+      const size_t r1 = (size_t)(seq_start + min(t0g + 1, len - 1));
+      s16x8 a = (t0g < len) ? lds_read8(v + r0 * vs + kvh * AP_D + d8)
+                            : s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      s16x8 b = (t0g + 1 < len) ? lds_read8(v + r1 * vs + kvh * AP_D + d8)
+                                : s16x8{0, 0, 0, 0, 0, 0, 0, 0};
 #pragma unroll
-        for (int j = 0; j < 8; ++j) v_lds[d8 + j][tok] = (u16)vv8[j];
+      for (int j = 0; j < 8; ++j) {
+        const uint32_t packed = (uint32_t)(unsigned short)a[j] |
+                                ((uint32_t)(unsigned short)b[j] << 16);
+        *reinterpret_cast<uint32_t*>(&v_lds[d8 + j][tp * 2]) = packed;
       }
     }
     __syncthreads();
 
 #pragma unroll
     for (int hl = 0; hl < HLOOP; ++hl) {
-      // ---- QK^T: two 16-token sub-tiles ----
-      f32x4 s_acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+      // ---- QK^T: four 16-token sub-tiles ----
+      f32x4 s_acc[4];
 #pragma unroll
-      for (int st = 0; st < 2; ++st) {
+      for (int st = 0; st < 4; ++st) {
+        s_acc[st] = f32x4{0, 0, 0, 0};
         const int tok = st * 16 + (lane & 15);
 #pragma unroll
         for (int c = 0; c < 4; ++c) {
@@ -136,38 +144,53 @@ __global__ void __launch_bounds__(256) attn_prefill_kernel(
               as_bf16x8(q_frag[hl][c]), as_bf16x8(kf), s_acc[st], 0, 0, 0);
         }
       }
-      // ---- mask + online softmax (rows spread: reg r = qrow (lane>>4)*4+r) ----
-      float p[2][4];
+      // ---- mask + online softmax (C-layout: reg r = qrow (lane>>4)*4+r) ----
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int qrow = q0 + (lane >> 4) * 4 + r;
-        float s0 = s_acc[0][r] * scale, s1 = s_acc[1][r] * scale;
-        if (kv0 + (lane & 15) > qrow) s0 = AF_NEG_INF;
-        if (kv0 + 16 + (lane & 15) > qrow) s1 = AF_NEG_INF;
-        const float tmax = group16_max_f32(fmaxf(s0, s1));
+        float sr[4];
+        float tmax = AF_NEG_INF;
+#pragma unroll
+        for (int st = 0; st < 4; ++st) {
+          sr[st] = s_acc[st][r] * scale;
+          if (kv0 + st * 16 + (lane & 15) > qrow) sr[st] = AF_NEG_INF;
+          tmax = fmaxf(tmax, sr[st]);
+        }
+        tmax = group16_max_f32(tmax);
         const float mn = fmaxf(m[hl][r], tmax);
         const float corr = (m[hl][r] <= AF_NEG_INF) ? 0.f : __expf(m[hl][r] - mn);
-        p[0][r] = (s0 <= AF_NEG_INF) ? 0.f : __expf(s0 - mn);
-        p[1][r] = (s1 <= AF_NEG_INF) ? 0.f : __expf(s1 - mn);
-        l[hl][r] = l[hl][r] * corr + group16_sum_f32(p[0][r] + p[1][r]);
+        float psum = 0.f;
+#pragma unroll
+        for (int st = 0; st < 4; ++st) {
+          sr[st] = (sr[st] <= AF_NEG_INF) ? 0.f : __expf(sr[st] - mn);
+          psum += sr[st];
+          s_acc[st][r] = sr[st];  // reuse acc regs to carry P
+        }
+        l[hl][r] = l[hl][r] * corr + group16_sum_f32(psum);
         m[hl][r] = mn;
 #pragma unroll
         for (int dt = 0; dt < 8; ++dt) o_acc[hl][dt][r] *= corr;
       }
       // ---- P -> per-wave LDS (C-layout -> A-fragment layout) ----
 #pragma unroll
-      for (int st = 0; st < 2; ++st)
+      for (int st = 0; st < 4; ++st)
 #pragma unroll
         for (int r = 0; r < 4; ++r)
-          p_lds[wid][(lane >> 4) * 4 + r][st * 16 + (lane & 15)] = f2bf(p[st][r]);
+          p_lds[wid][(lane >> 4) * 4 + r][st * 16 + (lane & 15)] =
+              f2bf(s_acc[st][r]);
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      s16x8 p_frag = lds_read8(&p_lds[wid][lane & 15][(lane >> 4) * 8]);
-      // ---- PV: o_acc[dt] += P(16x32) @ V(32x16) ----
+      s16x8 p_frag0 = lds_read8(&p_lds[wid][lane & 15][(lane >> 4) * 8]);
+      s16x8 p_frag1 = lds_read8(&p_lds[wid][lane & 15][32 + (lane >> 4) * 8]);
+      // ---- PV: o_acc[dt] += P(16x64) @ V(64x16), two K=32 chunks ----
 #pragma unroll
       for (int dt = 0; dt < 8; ++dt) {
-        s16x8 vf = lds_read8(&v_lds[dt * 16 + (lane & 15)][0] + (lane >> 4) * 8);
+        const u16* vrow = &v_lds[dt * 16 + (lane & 15)][0];
+        s16x8 vf0 = lds_read8(vrow + (lane >> 4) * 8);
+        s16x8 vf1 = lds_read8(vrow + 32 + (lane >> 4) * 8);
         o_acc[hl][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            as_bf16x8(p_frag), as_bf16x8(vf), o_acc[hl][dt], 0, 0, 0);
+            as_bf16x8(p_frag0), as_bf16x8(vf0), o_acc[hl][dt], 0, 0, 0);
+        o_acc[hl][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            as_bf16x8(p_frag1), as_bf16x8(vf1), o_acc[hl][dt], 0, 0, 0);
       }
     }
   }
@@ -181,7 +204,8 @@ __global__ void __launch_bounds__(256) attn_prefill_kernel(
       const int qrow = q0 + (lane >> 4) * 4 + r;
       if (qrow >= len) continue;
       const float inv = (l[hl][r] > 0.f) ? 1.f / l[hl][r] : 0.f;
-      u16* op = out + ((size_t)(seq_start + qrow) * Hq + qh) * AP_D + (lane & 15);
+      u16* op = out + (size_t)(seq_start + qrow) * (Hq * AP_D) + qh * AP_D +
+                (lane & 15);
 #pragma unroll
       for (int dt = 0; dt < 8; ++dt) op[dt * 16] = f2bf(o_acc[hl][dt][r] * inv);
     }
