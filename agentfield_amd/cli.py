@@ -234,6 +234,55 @@ def ps():
         typer.echo(f"{pkg['name']:24s} {state}")
 
 
+mcp_app = typer.Typer()
+app.add_typer(mcp_app, name="mcp", help="Model-Context-Protocol tools")
+
+
+@mcp_app.command("discover")
+def mcp_discover(project_dir: str = "."):
+    """Discover MCP server configs and their tools."""
+    from .mcp import MCPManager
+    from .mcp.manager import discover_config
+    cfg = discover_config(project_dir)
+    if not cfg:
+        typer.echo("no MCP config found (mcp.json)")
+        raise typer.Exit(1)
+    mgr = MCPManager()
+    try:
+        for name, tools in mgr.start_all(project_dir).items():
+            typer.echo(f"[{name}]")
+            for t in tools:
+                typer.echo(f"  {t.get('name')}: {t.get('description', '')[:60]}")
+    finally:
+        mgr.stop_all()
+
+
+@mcp_app.command("call")
+def mcp_call(tool: str, args_json: str = "{}", project_dir: str = "."):
+    """Start the project's MCP servers and invoke one tool."""
+    from .mcp import MCPManager
+    mgr = MCPManager()
+    try:
+        mgr.start_all(project_dir)
+        out = mgr.call(tool, json.loads(args_json))
+        typer.echo(json.dumps(out, indent=2))
+    finally:
+        mgr.stop_all()
+
+
+@app.command("config")
+def show_config(config: str = typer.Option(None, help="YAML config file")):
+    """Show the effective control-plane configuration (env > YAML > defaults)."""
+    from .controlplane.server import Config
+    kw = {}
+    if config:
+        import yaml
+        kw = yaml.safe_load(Path(config).read_text()) or {}
+    cfg = Config(**kw)
+    typer.echo(json.dumps({k: v for k, v in vars(cfg).items()
+                           if not k.startswith("_")}, indent=2, default=str))
+
+
 vc_app = typer.Typer()
 app.add_typer(vc_app, name="vc", help="Verifiable-credential tools")
 

@@ -735,6 +735,70 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
         execs = cp.storage.executions_by_run(run_id)
         return {"run": run, "dag": build_dag(execs, lightweight=True)}
 
+    @app.get("/api/ui/v1/executions/{execution_id}/timeline")
+    async def execution_timeline(execution_id: str):
+        """Per-execution timeline: the execution, its ancestors and children
+        (C33 execution_timeline.go)."""
+        rec = cp.storage.get_execution(execution_id)
+        if rec is None:
+            return JSONResponse({"error": "not found"}, status_code=404)
+        chain = []
+        cur = rec
+        seen = set()
+        while cur and cur["id"] not in seen:  # ancestors
+            seen.add(cur["id"])
+            chain.append(cur)
+            pid = cur.get("parent_execution_id")
+            cur = cp.storage.get_execution(pid) if pid else None
+        siblings = cp.storage.executions_by_run(rec.get("run_id")) \
+            if rec.get("run_id") else []
+        children = [e for e in siblings
+                    if e.get("parent_execution_id") == execution_id]
+        events = [{"at": rec.get("started_at"), "event": "started"},
+                  {"at": rec.get("finished_at"), "event": rec.get("status")}]
+        return {"execution": cp.envelope(rec),
+                "ancestors": [cp.envelope(e) for e in chain[1:]],
+                "children": [cp.envelope(e) for e in children],
+                "events": [e for e in events if e["at"]],
+                "webhook_history": cp.storage.webhook_history(execution_id)}
+
+    @app.get("/api/ui/v1/activity/recent")
+    async def recent_activity(req: Request):
+        """Recent cross-entity activity feed (C33 recent_activity.go)."""
+        limit = int(req.query_params.get("limit", 30))
+        items = []
+        for e in cp.storage.list_executions(limit=limit):
+            items.append({
+                "type": "execution", "at": e.get("finished_at")
+                or e.get("created_at"), "id": e["id"],
+                "summary": f"{e.get('node_id')}.{e.get('reasoner_id')} "
+                           f"{e.get('status')}"})
+        for n in cp.storage.list_nodes():
+            items.append({"type": "node", "at": n.get("last_status_change"),
+                          "id": n["id"],
+                          "summary": f"node {n['id']} {n['status']}"})
+        items = [i for i in items if i["at"]]
+        items.sort(key=lambda x: -(x["at"] or 0))
+        return {"activity": items[:limit]}
+
+    @app.get("/api/ui/v1/reasoners/{node_id}/{reasoner_id}/metrics")
+    async def reasoner_metrics(node_id: str, reasoner_id: str):
+        """Per-reasoner performance metrics from execution history
+        (reference GetReasonerPerformanceMetrics, storage.go:117)."""
+        execs = [e for e in cp.storage.list_executions(limit=500,
+                                                       node_id=node_id)
+                 if e.get("reasoner_id") == reasoner_id]
+        durs = sorted(e["duration_ms"] for e in execs
+                      if e.get("duration_ms") is not None)
+        n = len(execs)
+        ok = sum(1 for e in execs if e["status"] == "completed")
+
+        def pct(p):
+            return durs[min(len(durs) - 1, int(p * len(durs)))] if durs else None
+        return {"node_id": node_id, "reasoner_id": reasoner_id,
+                "executions": n, "success_rate": (ok / n) if n else None,
+                "p50_ms": pct(0.50), "p95_ms": pct(0.95), "p99_ms": pct(0.99)}
+
     # --------------------------------------------------- embedded web UI
     from pathlib import Path as _Path
     _ui = _Path(__file__).parent / "ui" / "index.html"

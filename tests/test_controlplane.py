@@ -339,3 +339,20 @@ def test_execution_sse_stream(cp_server, greeting_agent):
                json={"input": {"name": "sse"}}, timeout=20.0)
     t.join(timeout=10.0)
     assert got and got[0]["terminal"]
+
+
+def test_timeline_and_activity_and_metrics(cp_server, greeting_agent):
+    srv, _ = cp_server
+    r = httpx.post(srv.base_url + "/api/v1/execute/greeter.greet",
+                   json={"input": {"name": "tl"}}, timeout=20.0)
+    eid = r.json()["execution_id"]
+    tl = httpx.get(srv.base_url +
+                   f"/api/ui/v1/executions/{eid}/timeline").json()
+    assert tl["execution"]["execution_id"] == eid
+    assert any(e["event"] == "completed" for e in tl["events"])
+    act = httpx.get(srv.base_url + "/api/ui/v1/activity/recent").json()
+    assert any(a["id"] == eid for a in act["activity"])
+    m = httpx.get(srv.base_url +
+                  "/api/ui/v1/reasoners/greeter/greet/metrics").json()
+    assert m["executions"] >= 1 and m["p50_ms"] is not None
+    assert m["success_rate"] > 0
