@@ -132,7 +132,7 @@ AF_EXPORT int af_rmsnorm(void* out, void* res_out, const void* x, const void* re
           (u16*)out, nullptr, (const u16*)x, nullptr, (const u16*)w, eps, T, H);
     return af_last_err();
   }
-  int blocks = T < 2048 ? T : 2048;
+  int blocks = T < 16384 ? T : 16384;  // one row per block fills 256 CUs
   if (res)
     rmsnorm_kernel<true><<<blocks, 256, 0, st>>>(
         (u16*)out, (u16*)res_out, (const u16*)x, (const u16*)res, (const u16*)w, eps, T, H);
