@@ -1,4 +1,4 @@
-// Skinny-M GEMM for the decode path:  Y[M,N] = X[M,K] @ W[N,K]^T, M <= 128.
+// Skinny-M GEMM for the decode path:  Y[M,N] = X[M,K] @ W[N,K]^T, M <= 256.
 //
 // At decode batch sizes this is a pure weight-stream (W is N*K*2 bytes, X is
 // L2-resident), so the kernel is built to read W at HBM rate with MFMA doing
@@ -55,7 +55,7 @@ __global__ void __launch_bounds__(256) gemm_skinny_kernel(
     float* __restrict__ partial, const u16* __restrict__ X,
     const u16* __restrict__ W, int M, int N, int K, int Kc) {
   __shared__ u16 sW[2][64 * 64];
-  __shared__ u16 sX[2][128 * 64];  // up to two stage64 row-blocks (M<=128)
+  __shared__ u16 sX[2][((MT + 3) / 4) * 64 * 64];  // 64-row stage blocks
   const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
   const int nblk = blockIdx.x * 64;
   const int sk = blockIdx.y;
@@ -69,18 +69,21 @@ __global__ void __launch_bounds__(256) gemm_skinny_kernel(
   // 2-phase pipeline (guide T3 minimum): stage tile t+1 while computing
   // tile t; the single __syncthreads (vmcnt(0)+barrier) at loop end drains
   // the in-flight loads after compute has covered their latency.
+  constexpr int XBLKS = (MT + 3) / 4;  // 64-row stage blocks for X
   int cur = 0;
   stage64(W, K, sW[0], nblk, k0, N);
-  stage64(X, K, sX[0], 0, k0, M);
-  if (MT > 4) stage64(X + (size_t)64 * K, K, sX[0] + 64 * 64, 0, k0, M - 64);
+#pragma unroll
+  for (int xb = 0; xb < XBLKS; ++xb)
+    stage64(X + (size_t)(xb * 64) * K, K, sX[0] + xb * 64 * 64, 0, k0,
+            M - xb * 64);
   __syncthreads();
   for (int k = k0; k < k1; k += 64) {
     if (k + 64 < k1) {
       stage64(W, K, sW[cur ^ 1], nblk, k + 64, N);
-      stage64(X, K, sX[cur ^ 1], 0, k + 64, M);
-      if (MT > 4)
-        stage64(X + (size_t)64 * K, K, sX[cur ^ 1] + 64 * 64, 0, k + 64,
-                M - 64);
+#pragma unroll
+      for (int xb = 0; xb < XBLKS; ++xb)
+        stage64(X + (size_t)(xb * 64) * K, K, sX[cur ^ 1] + xb * 64 * 64, 0,
+                k + 64, M - xb * 64);
     }
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
@@ -216,7 +219,7 @@ AF_EXPORT int af_gemm_skinny(void* out, void* partial, void* residual,
                              int K, int SK, int mode,
                              const void* scale_ss, float eps, void* ss_out,
                              void* stream) {
-  if (M < 1 || M > 128) return 9005;
+  if (M < 1 || M > 256) return 9005;
   if (N % 64 || K % 64) return 9006;
   hipStream_t st = (hipStream_t)stream;
   int Kc = ((K / SK + 63) / 64) * 64;
@@ -234,7 +237,15 @@ AF_EXPORT int af_gemm_skinny(void* out, void* partial, void* residual,
     case 5: AF_SK_LAUNCH(5); break;
     case 6: AF_SK_LAUNCH(6); break;
     case 7: AF_SK_LAUNCH(7); break;
-    default: AF_SK_LAUNCH(8); break;
+    case 8: AF_SK_LAUNCH(8); break;
+    case 9: AF_SK_LAUNCH(9); break;
+    case 10: AF_SK_LAUNCH(10); break;
+    case 11: AF_SK_LAUNCH(11); break;
+    case 12: AF_SK_LAUNCH(12); break;
+    case 13: AF_SK_LAUNCH(13); break;
+    case 14: AF_SK_LAUNCH(14); break;
+    case 15: AF_SK_LAUNCH(15); break;
+    default: AF_SK_LAUNCH(16); break;
   }
 #undef AF_SK_LAUNCH
   if (mode == 4) {

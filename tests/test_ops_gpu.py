@@ -210,7 +210,7 @@ def test_gemm():
 
 
 def test_gemm_skinny():
-    for M in (1, 7, 16, 33, 64, 96, 128):
+    for M in (1, 7, 16, 33, 64, 96, 128, 192, 256):
         for (N, K) in [(6144, 4096), (4096, 14336), (128256, 4096)]:
             a = rnd(M, K, seed=M + N, scale=0.3)
             w = rnd(N, K, seed=M + N + 1, scale=0.3)
