@@ -253,7 +253,7 @@ def linear(x: torch.Tensor, w: torch.Tensor, silu_fuse: bool = False) -> torch.T
     # policy from measured MI355X data (profiles/kernel_bench_*): the
     # hand-written path wins the small-N decode shapes and the fused-SwiGLU
     # gate_up; hipBLASLt wins K>=14336 (down) and the 128k-vocab lm_head.
-    if (x.is_cuda and M <= 256 and N % 64 == 0 and K % 64 == 0
+    if (x.is_cuda and M <= 128 and N % 64 == 0 and K % 64 == 0
             and x.is_contiguous()
             and (silu_fuse or (N <= 8192 and K <= 8192)
                  or (M > 32 and N <= 8192))):
