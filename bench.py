@@ -32,11 +32,11 @@ def parse_args():
     p.add_argument("--steps", type=int, default=4)
     p.add_argument("--warmup", type=int, default=1)
     p.add_argument("--model", default="llama-3-8b")
-    p.add_argument("--calls", type=int, default=64,
+    p.add_argument("--calls", type=int, default=128,
                    help="reasoner calls per rank per step")
     p.add_argument("--prompt-len", type=int, default=512)
     p.add_argument("--gen-len", type=int, default=64)
-    p.add_argument("--max-num-seqs", type=int, default=64)
+    p.add_argument("--max-num-seqs", type=int, default=128)
     p.add_argument("--no-graphs", action="store_true")
     p.add_argument("--tp", type=int, default=1,
                    help="tensor-parallel degree (requires WORLD_SIZE == tp; "
