@@ -46,10 +46,16 @@ class Buses:
         self.reasoner = EventBus()
         self.memory = EventBus(buffer=256)
 
-    async def wait_for_execution(self, execution_id: str, timeout: float):
-        """Event-bus wait used by the sync execute path
-        (reference: waitForExecutionCompletion, execute.go:568-629)."""
-        sid, q = self.execution.subscribe()
+    async def wait_for_execution(self, execution_id: str, timeout: float,
+                                 queue: asyncio.Queue | None = None):
+        """Event-bus wait used by the sync execute path (reference:
+        waitForExecutionCompletion, execute.go:568-629).  Pass a queue from
+        an earlier subscribe() to close the subscribe-after-dispatch race:
+        the subscription must exist BEFORE the agent is called, or a fast
+        callback can fire the terminal event with no listener."""
+        sid = None
+        if queue is None:
+            sid, queue = self.execution.subscribe()
         try:
             loop = asyncio.get_running_loop()
             deadline = loop.time() + timeout
@@ -58,10 +64,11 @@ class Buses:
                 if left <= 0:
                     return None
                 try:
-                    ev = await asyncio.wait_for(q.get(), left)
+                    ev = await asyncio.wait_for(queue.get(), left)
                 except asyncio.TimeoutError:
                     return None
                 if ev.get("execution_id") == execution_id and ev.get("terminal"):
                     return ev
         finally:
-            self.execution.unsubscribe(sid)
+            if sid is not None:
+                self.execution.unsubscribe(sid)

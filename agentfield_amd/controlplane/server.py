@@ -197,11 +197,13 @@ class ControlPlane:
             _, uri = self.payloads.maybe_offload(execution_id, "result", result)
             if uri:
                 result = {"$payload_uri": uri}
-        self.storage.update_execution_result(execution_id, status, result,
-                                             error, duration_ms)
+        applied = self.storage.update_execution_result(
+            execution_id, status, result, error, duration_ms)
         rec = self.storage.get_execution(execution_id)
         if rec is None:
             return None
+        if not applied:
+            return rec  # already terminal: no duplicate events/webhooks/VCs
         if rec.get("run_id"):
             sibs = self.storage.executions_by_run(rec["run_id"])
             self.storage.upsert_run(
@@ -374,24 +376,32 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
             return JSONResponse({"error": err["error"]},
                                 status_code=err["status_code"])
         t0 = time.time()
-        code, result, errmsg = await cp.call_agent(rec)
-        if code == 200:
-            final = cp.complete_execution(rec["id"], st.COMPLETED, result=result,
-                                          duration_ms=(time.time() - t0) * 1e3)
-        elif code == 202:
-            cp.metrics.waiters_inflight.inc()
-            try:
-                ev = await cp.buses.wait_for_execution(rec["id"],
-                                                       cp.cfg.sync_timeout)
-            finally:
-                cp.metrics.waiters_inflight.dec()
-            if ev is None:
-                final = cp.complete_execution(rec["id"], st.TIMEOUT,
-                                              error="sync wait timed out")
-            else:
+        # subscribe BEFORE dispatching so a fast agent callback can't race
+        # past the waiter (the terminal event would otherwise be lost and
+        # the request would ride out the full sync timeout)
+        sub_id, sub_q = cp.buses.execution.subscribe()
+        try:
+            code, result, errmsg = await cp.call_agent(rec)
+            if code == 200:
+                final = cp.complete_execution(
+                    rec["id"], st.COMPLETED, result=result,
+                    duration_ms=(time.time() - t0) * 1e3)
+            elif code == 202:
+                cp.metrics.waiters_inflight.inc()
+                try:
+                    ev = await cp.buses.wait_for_execution(
+                        rec["id"], cp.cfg.sync_timeout, queue=sub_q)
+                finally:
+                    cp.metrics.waiters_inflight.dec()
                 final = cp.storage.get_execution(rec["id"])
-        else:
-            final = cp.complete_execution(rec["id"], st.FAILED, error=errmsg)
+                if ev is None and not st.is_terminal(final.get("status", "")):
+                    final = cp.complete_execution(rec["id"], st.TIMEOUT,
+                                                  error="sync wait timed out")
+            else:
+                final = cp.complete_execution(rec["id"], st.FAILED,
+                                              error=errmsg)
+        finally:
+            cp.buses.execution.unsubscribe(sub_id)
         resp = JSONResponse(cp.envelope(final))
         resp.headers["X-Execution-ID"] = rec["id"]
         resp.headers["X-Run-ID"] = rec["run_id"]
@@ -459,7 +469,12 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
         if not ev.get("execution_id"):
             return JSONResponse({"error": "execution_id required"},
                                 status_code=400)
-        cp.storage.upsert_workflow_event(ev)
+        owns = cp.storage.upsert_workflow_event(ev)
+        status = st.normalize(ev.get("status") or "")
+        if owns and st.is_terminal(status):
+            cp.complete_execution(ev["execution_id"], status,
+                                  ev.get("result"), ev.get("error"),
+                                  ev.get("duration_ms"))
         run_id = ev.get("run_id") or ev.get("workflow_id")
         if run_id:
             sibs = cp.storage.executions_by_run(run_id)

@@ -187,16 +187,26 @@ class Storage:
 
     def update_execution_result(self, exec_id: str, status: str,
                                 result=None, error: str | None = None,
-                                duration_ms: float | None = None) -> None:
-        self._exec(
+                                duration_ms: float | None = None) -> bool:
+        """Terminal states are write-once: a late timeout/duplicate callback
+        must not overwrite a completed record.  Returns False if the row was
+        already terminal."""
+        cur = self._exec(
             """UPDATE executions SET status=?, result=?, error_message=?,
                finished_at=?, duration_ms=COALESCE(?, (?-started_at)*1000.0)
-               WHERE id=?""",
+               WHERE id=? AND status NOT IN
+                 ('completed','failed','timeout','cancelled')""",
             (status, json.dumps(result) if result is not None else None,
              error, now(), duration_ms, now(), exec_id))
+        return cur.rowcount > 0
 
-    def upsert_workflow_event(self, ev: dict) -> None:
-        """SDK-pushed nested-call events (A.3): create-if-missing then update."""
+    def upsert_workflow_event(self, ev: dict) -> bool:
+        """SDK-pushed nested-call events (A.3): create-if-missing.  Returns
+        True when this ingestion OWNS the record's lifecycle (it created it,
+        i.e. an in-process nested call): control-plane-dispatched executions
+        (exec_* ids created by prepare_execution) are finalized by the agent
+        status callback, never by trace events — otherwise the trace event
+        races the callback and swallows its result/webhook."""
         if not self._q1("SELECT id FROM executions WHERE id=?", (ev["execution_id"],)):
             self.create_execution({
                 "id": ev["execution_id"], "run_id": ev.get("run_id") or ev.get("workflow_id"),
@@ -204,11 +214,8 @@ class Storage:
                 "node_id": ev.get("agent_node_id"), "reasoner_id": ev.get("reasoner_id"),
                 "status": ev.get("status", "running"), "input": ev.get("input_data"),
             })
-        status = ev.get("status")
-        if status and status not in ("running", "pending", "started"):
-            self.update_execution_result(ev["execution_id"], status,
-                                         ev.get("result"), ev.get("error"),
-                                         ev.get("duration_ms"))
+            return True
+        return not str(ev["execution_id"]).startswith("exec_")
 
     def executions_by_run(self, run_id: str) -> list[dict]:
         return [self._exec_row(r) for r in self._q(

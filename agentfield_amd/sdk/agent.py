@@ -267,9 +267,13 @@ class Agent(FastAPI):
             status, payload, err = "failed", None, \
                 f"{e}\n{traceback.format_exc(limit=3)}"
         duration = (time.time() - t0) * 1e3
-        await asyncio.to_thread(
-            self.client.report_status, ctx.execution_id, status, payload,
-            err, duration)
+        for attempt in range(4):  # a lost callback strands the execution
+            ok = await asyncio.to_thread(
+                self.client.report_status, ctx.execution_id, status, payload,
+                err, duration)
+            if ok:
+                break
+            await asyncio.sleep(0.2 * (2 ** attempt))
         if self.vc_enabled or meta.vc:
             try:
                 await asyncio.to_thread(self.client.create_execution_vc,
@@ -302,8 +306,27 @@ class Agent(FastAPI):
             ev["error"] = error
             ev["duration_ms"] = duration_ms
         ctx.execution_id = ev["execution_id"]
-        threading.Thread(target=self.client.workflow_event, args=(ev,),
-                         daemon=True).start()
+        self._event_queue().put(ev)
+
+    def _event_queue(self):
+        """Single background dispatcher for fire-and-forget workflow events
+        (a thread per event churns under load)."""
+        q = getattr(self, "_wf_queue", None)
+        if q is None:
+            import queue as _queue
+            q = self._wf_queue = _queue.Queue(maxsize=4096)
+
+            def drain():
+                while True:
+                    ev = q.get()
+                    try:
+                        self.client.workflow_event(ev)
+                    except Exception:
+                        pass
+
+            threading.Thread(target=drain, daemon=True,
+                             name="af-wf-events").start()
+        return q
 
     # ------------------------------------------------------- cross-agent
     def call(self, target: str, _async: bool = False, _webhook=None, **kwargs):
