@@ -215,8 +215,19 @@ class Agent(FastAPI):
         return out
 
     # ---------------------------------------------------------- execution
+    def _ensure_executor(self):
+        """Widen the loop's default thread pool (asyncio.to_thread default of
+        ~12 workers serializes concurrent sync reasoners + callbacks)."""
+        if not getattr(self, "_executor_set", False):
+            from concurrent.futures import ThreadPoolExecutor
+            asyncio.get_running_loop().set_default_executor(
+                ThreadPoolExecutor(max_workers=128,
+                                   thread_name_prefix="af-worker"))
+            self._executor_set = True
+
     async def _invoke(self, meta: _FunctionMeta, kwargs: dict,
                       ctx: ExecutionContext):
+        self._ensure_executor()
         token = set_context(ctx)
         self._workflow_event(meta, ctx, "start", kwargs)
         t0 = time.time()
