@@ -8,7 +8,6 @@ loops in hipGraphs).
 """
 from __future__ import annotations
 
-import math
 import time
 
 import torch
@@ -17,7 +16,7 @@ from .. import ops
 from ..models.llama import (AttnMetadata, KVCache, LlamaConfig,
                             LlamaForCausalLM)
 from .scheduler import ScheduleBatch, SchedulerConfig, make_scheduler
-from .sequence import SamplingParams, Sequence, SeqStatus
+from .sequence import SamplingParams, Sequence
 
 DECODE_BUCKETS = (1, 2, 4, 8, 16, 32, 64, 128, 256)
 

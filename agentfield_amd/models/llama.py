@@ -12,7 +12,7 @@ Layout decisions (MI355X-first):
 from __future__ import annotations
 
 import math
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 import torch
 from torch import nn

@@ -9,7 +9,6 @@ from __future__ import annotations
 
 import ctypes
 import functools
-import os
 from pathlib import Path
 
 import torch

@@ -3,7 +3,6 @@ AgentFieldClient + async_execution_manager.py in compact form: submit,
 sync execute, batch-status polling with adaptive backoff, SSE nudge)."""
 from __future__ import annotations
 
-import asyncio
 import json
 import time
 
