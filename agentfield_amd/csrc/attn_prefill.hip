@@ -1,10 +1,16 @@
-// Causal varlen prefill attention (flash-style, GQA) with MFMA 16x16x32 bf16.
+// Paged causal prefill attention (flash-style, GQA) with MFMA 16x16x32 bf16.
 //
-// q/k/v : rows at strides qs/ks/vs (views into the fused QKV buffer), head h
-//         at offset h*D, D=128, packed by sequence (cu_seqlens), post-RoPE.
-// out   : [total_T, Hq*D] bf16 contiguous.
+// q      : chunk rows at stride qs (view into the fused QKV buffer), head h
+//          at offset h*D, D=128, packed by sequence (cu_seqlens of the CHUNK).
+// kc/vc  : the paged KV cache [npages, Hk, page_size, D] — rope_cache has
+//          already appended the chunk, so attention reads history + chunk
+//          uniformly through the block table (chunked prefill falls out).
+// qstart : per-seq absolute position of the chunk's first row (0 for a
+//          full-prompt prefill; >0 when continuing a chunked prefill).
+// out    : [chunk_T, Hq*D] bf16 contiguous.
 // Host precomputes a flat tile map (tile_seq[i], tile_q0[i]): workgroup i on
-// grid.x handles q rows [q0, q0+ROWS) of sequence tile_seq[i]; grid.y = kv head.
+// grid.x handles chunk rows [q0, q0+ROWS) of sequence tile_seq[i];
+// grid.y = kv head.
 //
 // Workgroup = 256 threads = 4 waves.  Each wave owns a 16-row q-tile of one
 // query head in the GQA group:
@@ -46,15 +52,19 @@ __device__ __forceinline__ s16x8 lds_read8(const u16* p) {
 
 template <int G>
 __global__ void __launch_bounds__(256) attn_prefill_kernel(
-    u16* __restrict__ out, const u16* __restrict__ q, const u16* __restrict__ k,
-    const u16* __restrict__ v, const i32* __restrict__ cu_seqlens,
+    u16* __restrict__ out, const u16* __restrict__ q,
+    const u16* __restrict__ kc, const u16* __restrict__ vc,
+    const i32* __restrict__ bt, const i32* __restrict__ qstart,
+    const i32* __restrict__ cu_seqlens,
     const i32* __restrict__ tile_seq, const i32* __restrict__ tile_q0,
-    float scale, int Hq, int Hk, i64 qs, i64 ks, i64 vs) {
+    float scale, int Hq, int Hk, i64 qs, int page_size, int max_pages) {
   const int tile = blockIdx.x, kvh = blockIdx.y;
   const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
   const int seq = tile_seq[tile];
   const int seq_start = cu_seqlens[seq];
-  const int len = cu_seqlens[seq + 1] - seq_start;
+  const int len = cu_seqlens[seq + 1] - seq_start;  // chunk rows
+  const int hist = qstart[seq];                     // cached tokens before chunk
+  const i32* btrow = bt + (size_t)seq * max_pages;
 
   constexpr int ROWTILES = (G >= 4) ? 1 : (4 / G);   // 16-row tiles per WG
   constexpr int ROWS = 16 * ROWTILES;
@@ -90,19 +100,23 @@ __global__ void __launch_bounds__(256) attn_prefill_kernel(
     for (int dt = 0; dt < 8; ++dt) o_acc[hl][dt] = f32x4{0.f, 0.f, 0.f, 0.f};
   }
 
-  const int kv_end = min(len, tile_q0[tile] + ROWS);  // causal upper bound
+  // tokens visible to this WG: full history + chunk rows up to causal bound
+  const int kv_end = min(hist + len, hist + tile_q0[tile] + ROWS);
   const int n_kv_tiles = (kv_end + AP_KVBLK - 1) / AP_KVBLK;
+  const int kv_total = hist + len;
 
   for (int kt = 0; kt < n_kv_tiles; ++kt) {
     const int kv0 = kt * AP_KVBLK;
     __syncthreads();  // previous iteration's frag reads done
-    // ---- stage K (swizzled): 64 tok x 128 d / 8 = 1024 chunks, 4/thread ----
+    // ---- stage K (swizzled) from the paged cache ----
     for (int i = threadIdx.x; i < AP_KVBLK * (AP_D / 8); i += 256) {
       const int tok = i / (AP_D / 8);
       const int d8 = (i % (AP_D / 8)) * 8;
-      const int tg = kv0 + tok;
-      const size_t kvrow = (size_t)(seq_start + min(tg, len - 1));
-      s16x8 kv8 = lds_read8(k + kvrow * ks + kvh * AP_D + d8);
+      const int tg = min(kv0 + tok, kv_total - 1);
+      const i64 page = btrow[tg / page_size];
+      const size_t src =
+          (((size_t)page * Hk + kvh) * page_size + (tg % page_size)) * AP_D + d8;
+      s16x8 kv8 = lds_read8(kc + src);
       const int byte = tok * (AP_D * 2) + ((d8 * 2) ^ ((tok & 7) << 4));
       *reinterpret_cast<s16x8*>(reinterpret_cast<char*>(k_lds) + byte) = kv8;
     }
@@ -111,12 +125,15 @@ __global__ void __launch_bounds__(256) attn_prefill_kernel(
       const int tp = i / (AP_D / 8);          // token pair
       const int d8 = (i % (AP_D / 8)) * 8;
       const int t0g = kv0 + tp * 2;
-      const size_t r0 = (size_t)(seq_start + min(t0g, len - 1));
-      const size_t r1 = (size_t)(seq_start + min(t0g + 1, len - 1));
-      s16x8 a = (t0g < len) ? lds_read8(v + r0 * vs + kvh * AP_D + d8)
-                            : s16x8{0, 0, 0, 0, 0, 0, 0, 0};
-      s16x8 b = (t0g + 1 < len) ? lds_read8(v + r1 * vs + kvh * AP_D + d8)
-                                : s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      const int c0 = min(t0g, kv_total - 1), c1 = min(t0g + 1, kv_total - 1);
+      const size_t s0 = (((size_t)btrow[c0 / page_size] * Hk + kvh) * page_size +
+                         (c0 % page_size)) * AP_D + d8;
+      const size_t s1 = (((size_t)btrow[c1 / page_size] * Hk + kvh) * page_size +
+                         (c1 % page_size)) * AP_D + d8;
+      s16x8 a = (t0g < kv_total) ? lds_read8(vc + s0)
+                                 : s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      s16x8 b = (t0g + 1 < kv_total) ? lds_read8(vc + s1)
+                                     : s16x8{0, 0, 0, 0, 0, 0, 0, 0};
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const uint32_t packed = (uint32_t)(unsigned short)a[j] |
@@ -147,7 +164,7 @@ __global__ void __launch_bounds__(256) attn_prefill_kernel(
       // ---- mask + online softmax (C-layout: reg r = qrow (lane>>4)*4+r) ----
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int qrow = q0 + (lane >> 4) * 4 + r;
+        const int qrow = hist + q0 + (lane >> 4) * 4 + r;  // absolute position
         float sr[4];
         float tmax = AF_NEG_INF;
 #pragma unroll
@@ -212,11 +229,12 @@ __global__ void __launch_bounds__(256) attn_prefill_kernel(
   }
 }
 
-AF_EXPORT int af_attn_prefill(void* out, const void* q, const void* k, const void* v,
+AF_EXPORT int af_attn_prefill(void* out, const void* q, const void* kc,
+                              const void* vc, const void* bt, const void* qstart,
                               const void* cu_seqlens, const void* tile_seq,
                               const void* tile_q0, float scale, int ntiles,
-                              int Hq, int Hk, int D, i64 qs, i64 ks, i64 vs,
-                              void* stream) {
+                              int Hq, int Hk, int D, i64 qs, int page_size,
+                              int max_pages, void* stream) {
   if (D != AP_D) return 9002;
   const int G = Hq / Hk;
   if (G * Hk != Hq) return 9003;
@@ -225,9 +243,10 @@ AF_EXPORT int af_attn_prefill(void* out, const void* q, const void* k, const voi
   hipStream_t st = (hipStream_t)stream;
 #define AF_LAUNCH(GG)                                                           \
   attn_prefill_kernel<GG><<<grid, blk, 0, st>>>(                                \
-      (u16*)out, (const u16*)q, (const u16*)k, (const u16*)v,                   \
+      (u16*)out, (const u16*)q, (const u16*)kc, (const u16*)vc,                 \
+      (const i32*)bt, (const i32*)qstart,                                       \
       (const i32*)cu_seqlens, (const i32*)tile_seq, (const i32*)tile_q0,        \
-      scale, Hq, Hk, qs, ks, vs)
+      scale, Hq, Hk, qs, page_size, max_pages)
   switch (G) {
     case 1: AF_LAUNCH(1); break;
     case 2: AF_LAUNCH(2); break;

@@ -72,6 +72,7 @@ class LLMEngine:
         self.eos_id = eos_id
         self.max_num_seqs = max_num_seqs
         self._next_id = 0
+        self._prefilling: list[Sequence] = []
         self._finished: dict[int, Sequence] = {}
         self.sampler = ops.SamplerState(max_num_seqs, self.device, seed=seed or 0x5EED)
         self.enable_graphs = enable_graphs and self.is_gpu
@@ -115,26 +116,38 @@ class LLMEngine:
         return self._finished.pop(seq_id, None)
 
     def has_work(self) -> bool:
-        return self.sched.has_work()
+        return bool(self._prefilling) or self.sched.has_work()
 
     # ------------------------------------------------------------- stepping
     @torch.inference_mode()
     def step(self) -> list[tuple[int, int, bool]]:
-        """Run one engine iteration.  Returns [(seq_id, token, done)]."""
-        batch = self.sched.schedule()
-        if batch is None:
-            return []
+        """Run one engine iteration.  Returns [(seq_id, token, done)].
+
+        Long prompts prefill in chunks of max_prefill_tokens (paged prefill
+        attention reads history through the block table), so admission cost
+        is bounded regardless of prompt length."""
+        if not self._prefilling:
+            batch = self.sched.schedule()
+            if batch is None:
+                return []
+            if batch.is_prefill:
+                for s in batch.seqs:
+                    s.num_prefilled = 0
+                self._prefilling.extend(batch.seqs)
+            else:
+                self.metrics["steps"] += 1
+                self.metrics["decode_steps"] += 1
+                tokens = self._step_decode(batch)
+                return self._bookkeep(batch.seqs, tokens)
         self.metrics["steps"] += 1
-        if batch.is_prefill:
-            self.metrics["prefill_steps"] += 1
-            tokens = self._step_prefill(batch)
-        else:
-            self.metrics["decode_steps"] += 1
-            tokens = self._step_decode(batch)
-        # host-side bookkeeping
+        self.metrics["prefill_steps"] += 1
+        done_seqs, tokens = self._step_prefill_chunk()
+        return self._bookkeep(done_seqs, tokens)
+
+    def _bookkeep(self, seqs, tokens) -> list[tuple[int, int, bool]]:
         events = []
         now = time.monotonic_ns()
-        for seq, tok in zip(batch.seqs, tokens):
+        for seq, tok in zip(seqs, tokens):
             if not seq.output_ids:
                 seq.first_token_ns = now
             self.sched.note_token(seq)
@@ -152,41 +165,70 @@ class LLMEngine:
         return seq.pages[tok_idx // self.page_size] * self.page_size + \
             tok_idx % self.page_size
 
-    def _step_prefill(self, batch: ScheduleBatch) -> list[int]:
+    def _step_prefill_chunk(self) -> tuple[list[Sequence], list[int]]:
+        """Prefill up to max_prefill_tokens across the pending prompts;
+        sample only for prompts whose last chunk completed."""
         dev = self.device
-        seqs = batch.seqs
-        ids, pos, slots, lens = [], [], [], []
-        for seq in seqs:
-            n = len(seq.prompt_ids)
-            ids.extend(seq.prompt_ids)
-            pos.extend(range(n))
-            slots.extend(self._slot(seq, i) for i in range(n))
-            lens.append(n)
+        budget = self.sched.cfg.max_prefill_tokens
+        seqs, chunks = [], []
+        for seq in self._prefilling:
+            if budget <= 0:
+                break
+            take = min(len(seq.prompt_ids) - seq.num_prefilled, budget)
+            if take <= 0:
+                continue
+            seqs.append(seq)
+            chunks.append(take)
+            budget -= take
+        ids, pos, slots, q_start = [], [], [], []
+        bt_rows = []
+        for seq, take in zip(seqs, chunks):
+            np0 = seq.num_prefilled
+            ids.extend(seq.prompt_ids[np0:np0 + take])
+            pos.extend(range(np0, np0 + take))
+            slots.extend(self._slot(seq, i) for i in range(np0, np0 + take))
+            q_start.append(np0)
+            row = torch.zeros(self.max_pages_per_seq, dtype=torch.int32)
+            row[:len(seq.pages)] = torch.tensor(seq.pages, dtype=torch.int32)
+            bt_rows.append(row)
         self.metrics["prefill_tokens"] += len(ids)
         cu_list = [0]
-        for ln in lens:
+        for ln in chunks:
             cu_list.append(cu_list[-1] + ln)
         cu = torch.tensor(cu_list, dtype=torch.int32, device=dev)
         ids_t = torch.tensor(ids, dtype=torch.int32, device=dev)
         pos_t = torch.tensor(pos, dtype=torch.int32, device=dev)
         slots_t = torch.tensor(slots, dtype=torch.int64, device=dev)
         md = AttnMetadata(is_prefill=True, slots=slots_t, cu_seqlens=cu,
-                          seq_lens=lens)
-        last_rows = torch.tensor(
-            [int(cu[i + 1]) - 1 for i in range(len(seqs))],
-            dtype=torch.int32, device=dev)
+                          seq_lens=chunks,
+                          q_start=torch.tensor(q_start, dtype=torch.int32,
+                                               device=dev),
+                          block_table=torch.stack(bt_rows).to(dev))
+        done, done_rows = [], []
+        for i, (seq, take) in enumerate(zip(seqs, chunks)):
+            seq.num_prefilled += take
+            if seq.num_prefilled >= len(seq.prompt_ids):
+                done.append(seq)
+                done_rows.append(cu_list[i + 1] - 1)
+        for seq in done:
+            self._prefilling.remove(seq)
+        if not done:
+            self.model(ids_t, pos_t, self.kv, md,
+                       logit_rows=torch.zeros(1, dtype=torch.int32, device=dev))
+            return [], []
+        last_rows = torch.tensor(done_rows, dtype=torch.int32, device=dev)
         logits = self.model(ids_t, pos_t, self.kv, md, logit_rows=last_rows)
-        temps = torch.tensor([s.sampling.temperature for s in seqs],
+        temps = torch.tensor([s.sampling.temperature for s in done],
                              dtype=torch.float32, device=dev)
         kw = {}
-        if any(s.sampling.top_k > 0 or s.sampling.top_p < 1.0 for s in seqs) \
+        if any(s.sampling.top_k > 0 or s.sampling.top_p < 1.0 for s in done) \
                 and dev.type == "cuda":
-            kw = {"topk": torch.tensor([s.sampling.top_k for s in seqs],
+            kw = {"topk": torch.tensor([s.sampling.top_k for s in done],
                                        dtype=torch.int32, device=dev),
-                  "topp": torch.tensor([s.sampling.top_p for s in seqs],
+                  "topp": torch.tensor([s.sampling.top_p for s in done],
                                        dtype=torch.float32, device=dev)}
         toks = ops.sample(logits, temps, self.sampler, **kw)
-        return toks.cpu().tolist()
+        return done, toks.cpu().tolist()
 
     # -- decode path (graph-captured on GPU) --------------------------------
     def _fill_decode_buffers(self, seqs: list[Sequence], bs: int) -> None:

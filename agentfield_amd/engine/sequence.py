@@ -29,6 +29,7 @@ class Sequence:
     prompt_ids: list[int]
     sampling: SamplingParams
     status: SeqStatus = SeqStatus.WAITING
+    num_prefilled: int = 0
     output_ids: list[int] = field(default_factory=list)
     pages: list[int] = field(default_factory=list)
     finish_reason: str | None = None

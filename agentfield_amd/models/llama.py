@@ -80,7 +80,9 @@ CONFIGS = {
 class AttnMetadata:
     """Describes the batch for the attention kernels.
 
-    Prefill: cu_seqlens [B+1] i32, seq_lens list[int], slots [T] i64.
+    Prefill: cu_seqlens [B+1] i32 (chunk rows), seq_lens list[int] (chunk
+             lens), q_start [B] i32 (absolute chunk start; 0 unless chunked),
+             block_table [B,maxp] i32, slots [T] i64.
     Decode:  block_table [B,maxp] i32, seq_lens_t [B] i32, slots [B] i64,
              nsplit chosen by the engine.
     """
@@ -88,6 +90,7 @@ class AttnMetadata:
     slots: torch.Tensor
     cu_seqlens: torch.Tensor | None = None
     seq_lens: list[int] | None = None
+    q_start: torch.Tensor | None = None
     block_table: torch.Tensor | None = None
     seq_lens_t: torch.Tensor | None = None
     nsplit: int = 1
@@ -132,7 +135,9 @@ class LlamaAttention(nn.Module):
         ops.rope_cache(q, k, v, positions, rope_tab,
                        kv.k[self.layer_idx], kv.v[self.layer_idx], md.slots)
         if md.is_prefill:
-            return ops.attn_prefill(q, k, v, md.cu_seqlens, md.seq_lens,
+            return ops.attn_prefill(q, kv.k[self.layer_idx],
+                                    kv.v[self.layer_idx], md.block_table,
+                                    md.q_start, md.cu_seqlens, md.seq_lens,
                                     self.scale, head_dim=cfg.head_dim)
         return ops.attn_decode(q, kv.k[self.layer_idx], kv.v[self.layer_idx],
                                md.block_table, md.seq_lens_t, self.scale,

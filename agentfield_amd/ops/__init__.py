@@ -168,22 +168,24 @@ def prefill_tiles(seq_lens: list[int], rows_per_wg: int):
     return ts, tq
 
 
-def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+def attn_prefill(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
+                 block_table: torch.Tensor, q_start: torch.Tensor,
                  cu_seqlens: torch.Tensor, seq_lens: list[int],
                  scale: float | None = None, head_dim: int = 128) -> torch.Tensor:
-    """Causal varlen prefill attention.  q [T, Hq*D], k/v [T, Hk*D]
-    (strided rows ok) -> out [T, Hq*D] contiguous."""
+    """Paged causal prefill attention: chunk rows q [T, Hq*D] (strided ok)
+    attend to cached history + chunk through the block table.  q_start[s] is
+    the chunk's absolute start position (0 = full-prompt prefill); the chunk
+    itself must already be appended to the cache (rope_cache does this)."""
     D = head_dim
+    _, Hk, page, _ = kcache.shape
     T = q.shape[0]
     Hq = q.shape[1] // D
-    Hk = k.shape[1] // D
     if scale is None:
         scale = 1.0 / math.sqrt(D)
     if not _on_gpu(q):
-        return ref.attn_prefill(q.unflatten(-1, (Hq, D)),
-                                k.unflatten(-1, (Hk, D)),
-                                v.unflatten(-1, (Hk, D)),
-                                cu_seqlens, scale).flatten(1)
+        return ref.attn_prefill_paged(q.unflatten(-1, (Hq, D)), kcache, vcache,
+                                      block_table, q_start, cu_seqlens,
+                                      scale).flatten(1)
     G = Hq // Hk
     rows_per_wg = 16 * max(1, 4 // G)
     ts, tq = prefill_tiles(seq_lens, rows_per_wg)
@@ -192,9 +194,10 @@ def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     tile_q0 = torch.tensor(tq, dtype=torch.int32, device=dev)
     out = torch.empty(T, Hq * D, dtype=q.dtype, device=q.device)
     rc = _lib.lib().af_attn_prefill(
-        _lib.ptr(out), _lib.ptr(q), _lib.ptr(k), _lib.ptr(v),
+        _lib.ptr(out), _lib.ptr(q), _lib.ptr(kcache), _lib.ptr(vcache),
+        _lib.ptr(block_table), _lib.ptr(q_start),
         _lib.ptr(cu_seqlens), _lib.ptr(tile_seq), _lib.ptr(tile_q0),
-        scale, len(ts), Hq, Hk, D, q.stride(0), k.stride(0), v.stride(0),
+        scale, len(ts), Hq, Hk, D, q.stride(0), page, block_table.shape[1],
         _lib.cur_stream())
     _lib.check(rc, "af_attn_prefill")
     return out

@@ -50,8 +50,8 @@ def _declare(l: ctypes.CDLL) -> None:
     l.af_embedding.argtypes = [p, p, p, p, i, i, p]
     l.af_attn_decode.argtypes = [p, p, p, p, p, p, p, p, f, i, i, i, i, i, i, i,
                                  i64, p]
-    l.af_attn_prefill.argtypes = [p, p, p, p, p, p, p, f, i, i, i, i,
-                                  i64, i64, i64, p]
+    l.af_attn_prefill.argtypes = [p, p, p, p, p, p, p, p, p, f,
+                                  i, i, i, i, i64, i, i, p]
     l.af_gemm_bf16.argtypes = [p, p, p, i, i, i, p]
     l.af_gemm_skinny.argtypes = [p, p, p, p, p, i, i, i, i, i,
                                  p, f, p, p]

@@ -94,5 +94,35 @@ def attn_prefill(q, k, v, cu_seqlens, scale):
     return out
 
 
+def attn_prefill_paged(q, kcache, vcache, block_table, q_start, cu_seqlens,
+                       scale):
+    """Causal attention of chunk rows against cached history + chunk.
+    q [T,Hq,D]; caches [npages,Hk,page,D]; q_start[s] = absolute position of
+    the chunk's first row."""
+    T, Hq, D = q.shape
+    _, Hk, page, _ = kcache.shape
+    G = Hq // Hk
+    out = torch.empty_like(q)
+    cs = cu_seqlens.tolist()
+    for i in range(len(cs) - 1):
+        s0, s1 = cs[i], cs[i + 1]
+        L = s1 - s0
+        hist = int(q_start[i])
+        total = hist + L
+        npg = (total + page - 1) // page
+        pages = block_table[i, :npg].long()
+        ks = kcache[pages].permute(1, 0, 2, 3).reshape(Hk, -1, D)[:, :total].float()
+        vs = vcache[pages].permute(1, 0, 2, 3).reshape(Hk, -1, D)[:, :total].float()
+        qf = q[s0:s1].float().permute(1, 0, 2)            # [Hq, L, D]
+        kf = ks.repeat_interleave(G, dim=0)
+        vf = vs.repeat_interleave(G, dim=0)
+        att = (qf @ kf.transpose(-1, -2)) * scale         # [Hq, L, total]
+        qpos = torch.arange(hist, hist + L).unsqueeze(1)
+        kpos = torch.arange(total).unsqueeze(0)
+        att = att.masked_fill((kpos > qpos), float("-inf")).softmax(-1)
+        out[s0:s1] = (att @ vf).permute(1, 0, 2).to(q.dtype)
+    return out
+
+
 def sample_greedy(logits):
     return logits.float().argmax(-1).int()

@@ -41,7 +41,9 @@ def test_incremental_decode_matches_full_forward():
         slots = torch.arange(T, dtype=torch.int64)
         md = AttnMetadata(is_prefill=True, slots=slots,
                           cu_seqlens=torch.tensor([0, T], dtype=torch.int32),
-                          seq_lens=[T])
+                          seq_lens=[T],
+                          q_start=torch.zeros(1, dtype=torch.int32),
+                          block_table=torch.arange(64, dtype=torch.int32)[None, :])
         logits = model(torch.tensor(toks, dtype=torch.int32),
                        torch.arange(T, dtype=torch.int32), kv, md)
         toks.append(int(logits[-1].float().argmax()))
@@ -72,3 +74,22 @@ def test_backpressure_returns_none():
     eng = make_engine(max_waiting=1)
     assert eng.add_request([1], SamplingParams(max_tokens=1)) is not None
     assert eng.add_request([1], SamplingParams(max_tokens=1)) is None
+
+
+def test_chunked_prefill_matches_full():
+    """Prompts longer than max_prefill_tokens prefill in chunks via the paged
+    path; greedy tokens must match a single-shot prefill."""
+    cfg = CONFIGS["tiny"]
+    prompts = [[(i * 7 + 3) % cfg.vocab_size for i in range(25)],
+               [(i * 5 + 1) % cfg.vocab_size for i in range(11)]]
+    sp = SamplingParams(max_tokens=5, ignore_eos=True)
+    full = LLMEngine(cfg, device="cpu", dtype=torch.float32, page_size=4,
+                     num_pages=128, max_num_seqs=4, enable_graphs=False,
+                     max_prefill_tokens=512)
+    want = full.generate(prompts, sp)
+    chunked = LLMEngine(cfg, device="cpu", dtype=torch.float32, page_size=4,
+                        num_pages=128, max_num_seqs=4, enable_graphs=False,
+                        max_prefill_tokens=8)  # forces 4+ chunks
+    got = chunked.generate(prompts, sp)
+    assert got == want
+    assert chunked.metrics["prefill_steps"] > full.metrics["prefill_steps"]

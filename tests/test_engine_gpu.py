@@ -55,7 +55,10 @@ def test_fused_decode_matches_standard_logits():
     slots = torch.arange(T, dtype=torch.int64, device="cuda")
     md = AttnMetadata(is_prefill=True, slots=slots,
                       cu_seqlens=torch.tensor([0, T], dtype=torch.int32,
-                                              device="cuda"), seq_lens=[T])
+                                              device="cuda"), seq_lens=[T],
+                      q_start=torch.zeros(1, dtype=torch.int32, device="cuda"),
+                      block_table=torch.arange(64, dtype=torch.int32,
+                                               device="cuda")[None, :])
     model(ids, pos, kv, md)  # prefill fills the cache
 
     dec_ids = torch.randint(0, cfg.vocab_size, (1,), dtype=torch.int32,
@@ -110,7 +113,10 @@ def test_norm_folding_preserves_logits():
     md = AttnMetadata(is_prefill=True,
                       slots=torch.arange(T, dtype=torch.int64, device="cuda"),
                       cu_seqlens=torch.tensor([0, T], dtype=torch.int32,
-                                              device="cuda"), seq_lens=[T])
+                                              device="cuda"), seq_lens=[T],
+                      q_start=torch.zeros(1, dtype=torch.int32, device="cuda"),
+                      block_table=torch.arange(64, dtype=torch.int32,
+                                               device="cuda")[None, :])
     m = LlamaForCausalLM(cfg, device="cuda").init_random(7)
     # make norm weights non-trivial so folding is actually exercised
     with torch.no_grad():

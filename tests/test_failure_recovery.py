@@ -97,7 +97,7 @@ def test_webhook_warm_start_redelivers(tmp_path):
         async def run():
             import asyncio
             await disp.start()
-            for _ in range(100):
+            for _ in range(300):
                 if hits:
                     break
                 await asyncio.sleep(0.1)

@@ -160,7 +160,8 @@ def test_attn_decode_split():
     _decode_case(1, 4, 1, [511], nsplit=8, seed=23)
 
 
-def _prefill_case(Hq, Hk, lens, seed=30):
+def _prefill_case(Hq, Hk, lens, seed=30, page=16):
+    """Full-prompt paged prefill vs the contiguous fp32 reference."""
     D = 128
     T = sum(lens)
     q = rnd(T, Hq * D, seed=seed)
@@ -170,7 +171,24 @@ def _prefill_case(Hq, Hk, lens, seed=30):
     for l in lens:
         cu.append(cu[-1] + l)
     cu_t = torch.tensor(cu, dtype=torch.int32, device=DEV)
-    out = ops.attn_prefill(q, k, v, cu_t, lens)
+    # scatter k/v into a paged cache with per-seq page runs
+    maxp = (max(lens) + page - 1) // page
+    npages = 1 + len(lens) * maxp
+    kc = torch.zeros(npages, Hk, page, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    bt = torch.zeros(len(lens), maxp, dtype=torch.int32, device=DEV)
+    slots = []
+    nextpage = 1
+    for sidx, l in enumerate(lens):
+        n = (l + page - 1) // page
+        bt[sidx, :n] = torch.arange(nextpage, nextpage + n, dtype=torch.int32)
+        slots.extend(nextpage * page + i for i in range(l))
+        nextpage += n
+    slots_t = torch.tensor(slots, dtype=torch.int64, device=DEV)
+    ops.reshape_and_cache(k.unflatten(-1, (Hk, D)), v.unflatten(-1, (Hk, D)),
+                          kc, vc, slots_t)
+    qstart = torch.zeros(len(lens), dtype=torch.int32, device=DEV)
+    out = ops.attn_prefill(q, kc, vc, bt, qstart, cu_t, lens)
     torch.cuda.synchronize()
     want = ref.attn_prefill(q.cpu().unflatten(-1, (Hq, D)),
                             k.cpu().unflatten(-1, (Hk, D)),
@@ -194,6 +212,33 @@ def test_attn_prefill_g2_g8():
 
 def test_attn_prefill_long():
     _prefill_case(8, 2, [1024], seed=70)
+
+
+def test_attn_prefill_chunked_history():
+    """Second chunk attends to cached history: must match the one-shot
+    paged reference over the whole prompt."""
+    D, Hq, Hk, page = 128, 8, 2, 16
+    total, hist = 48, 29
+    chunk = total - hist
+    q_all = rnd(total, Hq * D, seed=80)
+    k_all = rnd(total, Hk * D, seed=81)
+    v_all = rnd(total, Hk * D, seed=82)
+    npages = 1 + (total + page - 1) // page
+    kc = torch.zeros(npages, Hk, page, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    bt = torch.arange(1, npages, dtype=torch.int32, device=DEV)[None, :]
+    slots = torch.arange(page, page + total, dtype=torch.int64, device=DEV)
+    ops.reshape_and_cache(k_all.unflatten(-1, (Hk, D)),
+                          v_all.unflatten(-1, (Hk, D)), kc, vc, slots)
+    q2 = q_all[hist:]
+    cu = torch.tensor([0, chunk], dtype=torch.int32, device=DEV)
+    qstart = torch.tensor([hist], dtype=torch.int32, device=DEV)
+    out = ops.attn_prefill(q2, kc, vc, bt, qstart, cu, [chunk])
+    torch.cuda.synchronize()
+    want = ref.attn_prefill_paged(
+        q2.cpu().unflatten(-1, (Hq, D)), kc.cpu(), vc.cpu(), bt.cpu(),
+        qstart.cpu(), cu.cpu(), 1.0 / math.sqrt(D)).flatten(1)
+    assert close(out, want), "chunked prefill history mismatch"
 
 
 def test_gemm():

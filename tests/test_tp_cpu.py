@@ -62,7 +62,9 @@ def _logits_case(rank, world):
     kv = KVCache(model.cfg, 32, 4, dev, torch.float32)
     md = AttnMetadata(is_prefill=True, slots=torch.arange(T, dtype=torch.int64),
                       cu_seqlens=torch.tensor([0, T], dtype=torch.int32),
-                      seq_lens=[T])
+                      seq_lens=[T],
+                      q_start=torch.zeros(1, dtype=torch.int32),
+                      block_table=torch.arange(32, dtype=torch.int32)[None, :])
     logits = model(ids, pos, kv, md)
 
     # reference: full (TP=1) model, same deterministic weights
