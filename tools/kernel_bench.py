@@ -57,18 +57,23 @@ def bench_skinny():
                           "blas_us": round(t_blas * 1e6, 1)}))
 
 
-def bench_prefill(Hq=32, Hk=8, D=128):
+def bench_prefill(Hq=32, Hk=8, D=128, page=16):
     for (B, S) in [(16, 512), (4, 2048), (1, 8192)]:
         T = B * S
         q = torch.randn(T, Hq * D, dtype=torch.bfloat16, device=DEV)
-        k = torch.randn(T, Hk * D, dtype=torch.bfloat16, device=DEV)
-        v = torch.randn_like(k)
+        maxp = (S + page - 1) // page
+        npages = 1 + B * maxp
+        kc = torch.randn(npages, Hk, page, D, dtype=torch.bfloat16, device=DEV)
+        vc = torch.randn_like(kc)
+        bt = torch.arange(1, npages, dtype=torch.int32,
+                          device=DEV).reshape(B, maxp)
         cu = torch.arange(0, T + 1, S, dtype=torch.int32, device=DEV)
+        qstart = torch.zeros(B, dtype=torch.int32, device=DEV)
         lens = [S] * B
         # causal flops: 2 * (QK + PV) * 0.5 * S^2 * D * Hq per seq
         tf = 2.0 * 2 * 0.5 * S * S * D * Hq * B / 1e12
-        t = timeit(lambda: ops.attn_prefill(q, k, v, cu, lens))
-        print(json.dumps({"op": "attn_prefill", "B": B, "S": S,
+        t = timeit(lambda: ops.attn_prefill(q, kc, vc, bt, qstart, cu, lens))
+        print(json.dumps({"op": "attn_prefill_paged", "B": B, "S": S,
                           "tflops": round(tf / t, 1), "ms": round(t * 1e3, 3)}))
 
 
