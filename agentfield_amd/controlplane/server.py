@@ -449,6 +449,18 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
         return {"executions": {i: cp.envelope(found[i]) for i in found},
                 "missing": [i for i in ids if i not in found]}
 
+    @app.post("/api/v1/executions/{execution_id}/cancel")
+    async def cancel_execution(execution_id: str):
+        rec = cp.storage.get_execution(execution_id)
+        if rec is None:
+            return JSONResponse({"error": "not found"}, status_code=404)
+        if st.is_terminal(rec.get("status", "")):
+            return JSONResponse({"error": "already terminal",
+                                 "status": rec["status"]}, status_code=409)
+        final = cp.complete_execution(execution_id, st.CANCELLED,
+                                      error="cancelled by client")
+        return cp.envelope(final)
+
     @app.post("/api/v1/executions/{execution_id}/status")
     async def execution_status_callback(execution_id: str, req: Request):
         """Agent-side terminal/progress status ingestion (A.2)."""

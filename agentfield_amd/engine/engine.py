@@ -16,7 +16,7 @@ from .. import ops
 from ..models.llama import (AttnMetadata, KVCache, LlamaConfig,
                             LlamaForCausalLM)
 from .scheduler import ScheduleBatch, SchedulerConfig, make_scheduler
-from .sequence import SamplingParams, Sequence
+from .sequence import SamplingParams, Sequence, SeqStatus
 
 DECODE_BUCKETS = (1, 2, 4, 8, 16, 32, 64, 128, 256)
 
@@ -114,6 +114,33 @@ class LLMEngine:
 
     def get_finished(self, seq_id: int) -> Sequence | None:
         return self._finished.pop(seq_id, None)
+
+    def cancel(self, seq_id: int) -> bool:
+        """Cancel a queued or running request; frees its KV pages."""
+        # waiting queue (python scheduler) or native adapter's seq map
+        seqs = getattr(self.sched, "seqs", None)
+        if seqs is not None:  # native adapter
+            seq = seqs.get(seq_id)
+        else:
+            seq = next((s for s in list(self.sched.waiting) +
+                        self.sched.running if s.seq_id == seq_id), None)
+        if seq is None or seq.status == SeqStatus.FINISHED:
+            return False
+        if seq in self._prefilling:
+            self._prefilling.remove(seq)
+        if seqs is not None:
+            self.sched.nat.finish(seq_id)
+            seqs.pop(seq_id, None)
+        else:
+            if seq in self.sched.running:
+                self.sched.running.remove(seq)
+            elif seq in self.sched.waiting:
+                self.sched.waiting.remove(seq)
+            self.sched.release(seq)
+        seq.status = SeqStatus.FINISHED
+        seq.finish_reason = "cancelled"
+        self._finished[seq_id] = seq
+        return True
 
     def has_work(self) -> bool:
         return bool(self._prefilling) or self.sched.has_work()

@@ -177,6 +177,8 @@ class NativeScheduler {
     release(it->second);
     for (size_t i = 0; i < running_.size(); ++i)
       if (running_[i] == seq_id) { running_.erase(running_.begin() + i); break; }
+    for (auto wi = waiting_.begin(); wi != waiting_.end(); ++wi)
+      if (*wi == seq_id) { waiting_.erase(wi); break; }
     seqs_.erase(it);
   }
 

@@ -356,3 +356,13 @@ def test_timeline_and_activity_and_metrics(cp_server, greeting_agent):
                   "/api/ui/v1/reasoners/greeter/greet/metrics").json()
     assert m["executions"] >= 1 and m["p50_ms"] is not None
     assert m["success_rate"] > 0
+
+
+def test_cancel_execution_endpoint(cp_server):
+    srv, cp = cp_server
+    cp.storage.create_execution({"id": "exec_c1", "run_id": "run_c1",
+                                 "status": "running", "input": {}})
+    r = httpx.post(srv.base_url + "/api/v1/executions/exec_c1/cancel")
+    assert r.status_code == 200 and r.json()["status"] == "cancelled"
+    r = httpx.post(srv.base_url + "/api/v1/executions/exec_c1/cancel")
+    assert r.status_code == 409

@@ -93,3 +93,19 @@ def test_chunked_prefill_matches_full():
     got = chunked.generate(prompts, sp)
     assert got == want
     assert chunked.metrics["prefill_steps"] > full.metrics["prefill_steps"]
+
+
+def test_cancel_request_frees_pages():
+    eng = make_engine()
+    free0 = eng.sched.alloc.num_free
+    rid = eng.add_request([1, 2, 3, 4, 5, 6],
+                          SamplingParams(max_tokens=50, ignore_eos=True))
+    eng.step()  # prefill chunk admits + allocates pages
+    assert eng.cancel(rid)
+    fin = eng.get_finished(rid)
+    assert fin is not None and fin.finish_reason == "cancelled"
+    # engine fully drains and pages return
+    while eng.has_work():
+        eng.step()
+    assert eng.sched.alloc.num_free == free0
+    assert not eng.cancel(rid)  # already finished
