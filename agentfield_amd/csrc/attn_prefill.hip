@@ -120,10 +120,15 @@ __global__ void __launch_bounds__(256) attn_prefill_kernel(
       const int byte = tok * (AP_D * 2) + ((d8 * 2) ^ ((tok & 7) << 4));
       *reinterpret_cast<s16x8*>(reinterpret_cast<char*>(k_lds) + byte) = kv8;
     }
-    // ---- stage V transposed: token pairs -> u32 writes (8 per 2 loads) ----
+    // ---- stage V transposed: token pairs -> u32 writes (8 per 2 loads).
+    // Lane decomposition: consecutive lanes take consecutive token pairs at
+    // the SAME d-slice, so the 8 v_lds writes land in consecutive words
+    // (bank-conflict-free).  The d8-major variant put 16 lanes on one bank
+    // (d8 stride x 36-word rows = 0 mod 32): PMC measured 1.7e8
+    // SQ_LDS_BANK_CONFLICT per dispatch (profiles/pmc_r01_counters.txt).
     for (int i = threadIdx.x; i < (AP_KVBLK / 2) * (AP_D / 8); i += 256) {
-      const int tp = i / (AP_D / 8);          // token pair
-      const int d8 = (i % (AP_D / 8)) * 8;
+      const int tp = i % (AP_KVBLK / 2);      // token pair (consecutive lanes)
+      const int d8 = (i / (AP_KVBLK / 2)) * 8;
       const int t0g = kv0 + tp * 2;
       const int c0 = min(t0g, kv_total - 1), c1 = min(t0g + 1, kv_total - 1);
       const size_t s0 = (((size_t)btrow[c0 / page_size] * Hk + kvh) * page_size +
