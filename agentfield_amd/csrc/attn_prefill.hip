@@ -36,7 +36,7 @@
 
 #define AP_D 128
 #define AP_KVBLK 64
-#define AP_VPAD 88  // 64 tokens padded to 88: 44-word rows spread banks (12r+4g)%32
+#define AP_VPAD 72  // 64 tokens padded to 72 (residual 2-way conflicts are free, guide m136)
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 
