@@ -132,3 +132,55 @@ def test_norm_folding_preserves_logits():
     torch.cuda.synchronize()
     diff = (base - folded).abs().max().item()
     assert diff < 0.2, f"norm folding changed logits by {diff}"
+
+
+def test_engine_server_stack_on_gpu():
+    """Serving E2E on the GPU: engine server + router + ai() remote path."""
+    import sys
+    from pathlib import Path
+    sys.path.insert(0, str(Path(__file__).parent))
+    from helpers import AppServer
+    from agentfield_amd.sdk.ai import EngineRunner
+    from agentfield_amd.serving.engine_server import create_engine_app
+    from agentfield_amd.serving.router import DPRouter
+
+    eng = LLMEngine(CONFIGS["debug-1b"], device="cuda", num_pages=256,
+                    max_num_seqs=8, enable_graphs=True, seed=2)
+    runner = EngineRunner(eng)
+    srv = AppServer(create_engine_app(runner, "debug-1b")).start()
+    try:
+        router = DPRouter([srv.base_url])
+        out = router.generate(prompt_ids=[1, 5, 9], max_tokens=6,
+                              ignore_eos=True)
+        assert len(out["output_ids"]) == 6
+        import httpx
+        m = httpx.get(srv.base_url + "/metrics").text
+        assert "agentfield_engine_decode_tokens_total" in m
+        pieces = list(router.stream(prompt_ids=[2, 4], max_tokens=4,
+                                    ignore_eos=True))
+        assert pieces
+    finally:
+        runner.shutdown()
+        srv.stop()
+
+
+def test_checkpoint_load_gpu(tmp_path):
+    """safetensors save on CPU -> load into a GPU model -> identical greedy
+    tokens as the source weights."""
+    from agentfield_amd.models import LlamaForCausalLM
+    from agentfield_amd.models.checkpoint import load_checkpoint, save_checkpoint
+
+    cfg = CONFIGS["tiny"]
+    src = LlamaForCausalLM(cfg, device="cuda").init_random(11)
+    save_checkpoint(src, str(tmp_path / "ck"))
+    eng1 = LLMEngine(cfg, device="cuda", page_size=4, num_pages=128,
+                     max_num_seqs=4, model=src, enable_graphs=False)
+    want = eng1.generate([[1, 2, 3, 4]], SamplingParams(max_tokens=6,
+                                                        ignore_eos=True))
+    loaded = LlamaForCausalLM(cfg, device="cuda")
+    load_checkpoint(loaded, str(tmp_path / "ck"))
+    eng2 = LLMEngine(cfg, device="cuda", page_size=4, num_pages=128,
+                     max_num_seqs=4, model=loaded, enable_graphs=False)
+    got = eng2.generate([[1, 2, 3, 4]], SamplingParams(max_tokens=6,
+                                                       ignore_eos=True))
+    assert got == want
