@@ -156,9 +156,18 @@ def test_engine_server_stack_on_gpu():
         import httpx
         m = httpx.get(srv.base_url + "/metrics").text
         assert "agentfield_engine_decode_tokens_total" in m
-        pieces = list(router.stream(prompt_ids=[2, 4], max_tokens=4,
-                                    ignore_eos=True))
-        assert pieces
+        # stream path: random-vocab tokens may decode to empty text pieces
+        # (byte tokenizer), so assert on the raw SSE events
+        import json as _json
+        events = []
+        with httpx.stream("POST", srv.base_url + "/v1/generate",
+                          json={"prompt_ids": [2, 4], "max_tokens": 4,
+                                "ignore_eos": True, "stream": True},
+                          timeout=120.0) as resp:
+            for line in resp.iter_lines():
+                if line.startswith("data:"):
+                    events.append(_json.loads(line[5:]))
+        assert len(events) == 4 and events[-1]["done"]
     finally:
         runner.shutdown()
         srv.stop()
