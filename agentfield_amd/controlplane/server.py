@@ -24,10 +24,34 @@ from .services import (ExecutionCleanup, HealthMonitor, Metrics,
                        PayloadStore, PresenceManager, WebhookDispatcher)
 from .storage import Storage
 from .workflow import aggregate_status, build_dag
+from ..logging_setup import get_logger
+
+log = get_logger("controlplane")
 
 
 class Config:
+    """Precedence: env (AGENTFIELD_*) > explicit kwargs (YAML) > defaults
+    (reference parity: C27 viper precedence)."""
+
+    _ENV = {
+        "db_path": ("AGENTFIELD_DATABASE_URL", str),
+        "sync_timeout": ("AGENTFIELD_SYNC_TIMEOUT", float),
+        "agent_timeout": ("AGENTFIELD_AGENT_TIMEOUT", float),
+        "async_workers": ("AGENTFIELD_EXEC_ASYNC_WORKERS", int),
+        "async_queue_capacity": ("AGENTFIELD_EXEC_ASYNC_QUEUE_CAPACITY", int),
+        "did_enabled": ("AGENTFIELD_DID_ENABLED", lambda v: v not in ("0", "false")),
+        "auto_vc": ("AGENTFIELD_AUTO_VC", lambda v: v in ("1", "true")),
+        "presence_ttl": ("AGENTFIELD_PRESENCE_TTL", float),
+        "health_interval": ("AGENTFIELD_HEALTH_INTERVAL", float),
+        "admin_grpc_port": ("AGENTFIELD_ADMIN_GRPC_PORT", int),
+    }
+
     def __init__(self, **kw):
+        import os as _os
+        for field, (env, cast) in self._ENV.items():
+            v = _os.environ.get(env)
+            if v is not None:
+                kw[field] = cast(v)
         self.db_path = kw.get("db_path", ":memory:")
         self.payload_dir = kw.get("payload_dir")
         self.keystore_path = kw.get("keystore_path")
@@ -66,6 +90,8 @@ class ControlPlane:
         self.started_at = time.time()
 
     async def start(self):
+        log.info("control plane starting (db=%s, workers=%d)",
+                 self.cfg.db_path, self.cfg.async_workers)
         self.client = httpx.AsyncClient(timeout=self.cfg.agent_timeout)
         self._async_q = asyncio.Queue(self.cfg.async_queue_capacity)
         for _ in range(self.cfg.async_workers):

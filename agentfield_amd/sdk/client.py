@@ -76,13 +76,24 @@ class AgentFieldClient:
         r.raise_for_status()
         return r.json()
 
-    def wait_for_result(self, execution_id: str, timeout: float = 300.0,
-                        poll_initial: float = 0.05, poll_max: float = 2.0,
-                        use_sse: bool = True) -> dict:
+    def wait_for_result(self, execution_id: str, timeout: float | None = None,
+                        poll_initial: float | None = None,
+                        poll_max: float | None = None,
+                        use_sse: bool | None = None) -> dict:
         """Adaptive polling with an SSE nudge: a background listener on
         /api/ui/v1/executions/events wakes the poll loop the moment the
         terminal event fires (reference async_execution_manager.py:644)."""
+        import os
         import threading
+        # env-tunable like the reference's AsyncConfig.from_environment (P8)
+        timeout = timeout if timeout is not None else float(
+            os.environ.get("AGENTFIELD_RESULT_TIMEOUT", 300.0))
+        poll_initial = poll_initial if poll_initial is not None else float(
+            os.environ.get("AGENTFIELD_POLL_INITIAL", 0.05))
+        poll_max = poll_max if poll_max is not None else float(
+            os.environ.get("AGENTFIELD_POLL_MAX", 2.0))
+        if use_sse is None:
+            use_sse = os.environ.get("AGENTFIELD_SSE_NUDGE", "1") != "0"
         deadline = time.time() + timeout
         poll = poll_initial
         nudge = threading.Event()
