@@ -20,14 +20,18 @@ __global__ void __launch_bounds__(256) rmsnorm_kernel(
   const int per_row = H / VEC;
   __shared__ float red[4];
 
+  // statically indexed register cache (a runtime-indexed array spills to
+  // scratch — guide rule #20; the spill measured 0.85 TB/s at prefill T)
   float cache[MAX_IT][VEC];
 
   for (int row = blockIdx.x; row < T; row += gridDim.x) {
     const u16* xrow = x + (size_t)row * H;
     const u16* rrow = HAS_RES ? res + (size_t)row * H : nullptr;
     float ss = 0.f;
-    int it = 0;
-    for (int c = threadIdx.x; c < per_row; c += NTHR, ++it) {
+#pragma unroll
+    for (int it = 0; it < MAX_IT; ++it) {
+      const int c = threadIdx.x + it * NTHR;
+      if (c >= per_row) break;
       s16x8 xv = *reinterpret_cast<const s16x8*>(xrow + c * VEC);
       float* f = cache[it];
 #pragma unroll
@@ -46,8 +50,10 @@ __global__ void __launch_bounds__(256) rmsnorm_kernel(
     __syncthreads();
     const float rstd = rsqrtf((red[0] + red[1] + red[2] + red[3]) / (float)H + eps);
 
-    it = 0;
-    for (int c = threadIdx.x; c < per_row; c += NTHR, ++it) {
+#pragma unroll
+    for (int it = 0; it < MAX_IT; ++it) {
+      const int c = threadIdx.x + it * NTHR;
+      if (c >= per_row) break;
       float* f = cache[it];
       s16x8 wv = *reinterpret_cast<const s16x8*>(w + c * VEC);
       s16x8 ov, hv;

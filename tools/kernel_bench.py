@@ -44,7 +44,9 @@ def bench_gemm():
 def bench_skinny():
     for (M, N, K) in [(16, 6144, 4096), (16, 4096, 4096), (16, 28672, 4096),
                       (16, 4096, 14336), (64, 28672, 4096), (16, 128256, 4096),
-                      (64, 4096, 14336), (64, 128256, 4096), (64, 6144, 4096)]:
+                      (64, 4096, 14336), (64, 128256, 4096), (64, 6144, 4096),
+                      (128, 6144, 4096), (128, 4096, 4096), (128, 28672, 4096),
+                      (128, 4096, 14336)]:
         a = torch.randn(M, K, dtype=torch.bfloat16, device=DEV)
         w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV)
         gb = N * K * 2 / 1e9
@@ -102,6 +104,13 @@ def bench_norm_rope(H=4096):
         t = timeit(lambda: ops.rmsnorm(x, w, 1e-5, residual=res))
         print(json.dumps({"op": "rmsnorm_fused", "T": T, "gbps": round(gb / t, 1),
                           "us": round(t * 1e6, 1)}))
+    x = torch.randn(65536, H, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(H, dtype=torch.bfloat16, device=DEV)
+    res = torch.randn_like(x)
+    gb = 4.0 * 65536 * H * 2 / 1e9
+    t = timeit(lambda: ops.rmsnorm(x, w, 1e-5, residual=res), iters=10)
+    print(json.dumps({"op": "rmsnorm_fused", "T": 65536,
+                      "gbps": round(gb / t, 1), "us": round(t * 1e6, 1)}))
         gu = torch.randn(T, 28672, dtype=torch.bfloat16, device=DEV)
         gb2 = (T * 28672 + T * 14336) * 2 / 1e9
         t2 = timeit(lambda: ops.silu_and_mul(gu))
