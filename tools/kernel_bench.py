@@ -104,6 +104,11 @@ def bench_norm_rope(H=4096):
         t = timeit(lambda: ops.rmsnorm(x, w, 1e-5, residual=res))
         print(json.dumps({"op": "rmsnorm_fused", "T": T, "gbps": round(gb / t, 1),
                           "us": round(t * 1e6, 1)}))
+        gu = torch.randn(T, 28672, dtype=torch.bfloat16, device=DEV)
+        gb2 = (T * 28672 + T * 14336) * 2 / 1e9
+        t2 = timeit(lambda: ops.silu_and_mul(gu))
+        print(json.dumps({"op": "silu_mul", "T": T, "gbps": round(gb2 / t2, 1),
+                          "us": round(t2 * 1e6, 1)}))
     x = torch.randn(65536, H, dtype=torch.bfloat16, device=DEV)
     w = torch.randn(H, dtype=torch.bfloat16, device=DEV)
     res = torch.randn_like(x)
@@ -111,11 +116,6 @@ def bench_norm_rope(H=4096):
     t = timeit(lambda: ops.rmsnorm(x, w, 1e-5, residual=res), iters=10)
     print(json.dumps({"op": "rmsnorm_fused", "T": 65536,
                       "gbps": round(gb / t, 1), "us": round(t * 1e6, 1)}))
-        gu = torch.randn(T, 28672, dtype=torch.bfloat16, device=DEV)
-        gb2 = (T * 28672 + T * 14336) * 2 / 1e9
-        t2 = timeit(lambda: ops.silu_and_mul(gu))
-        print(json.dumps({"op": "silu_mul", "T": T, "gbps": round(gb2 / t2, 1),
-                          "us": round(t2 * 1e6, 1)}))
 
 
 def bench_sample(B=64, V=128256):
