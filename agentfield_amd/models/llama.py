@@ -259,7 +259,7 @@ class LlamaForCausalLM(nn.Module):
         if not getattr(self, "_norms_folded", False):
             return False
         cfg = self.cfg
-        return (not getattr(self, "no_fused_decode", False) and T <= 128
+        return (not getattr(self, "no_fused_decode", False) and T <= 64
                 and cfg.hidden_size % 64 == 0
                 and cfg.intermediate_size % 64 == 0
                 and (cfg.q_size + 2 * cfg.kv_size) % 64 == 0

@@ -253,13 +253,15 @@ def linear(x: torch.Tensor, w: torch.Tensor, silu_fuse: bool = False) -> torch.T
     shapes (M<=64), hipBLASLt (torch.matmul) for prefill GEMMs."""
     M, K = x.shape
     N = w.shape[0]
-    # policy from measured MI355X data (profiles/kernel_bench_*): the
-    # hand-written path wins the small-N decode shapes and the fused-SwiGLU
-    # gate_up; hipBLASLt wins K>=14336 (down) and the 128k-vocab lm_head.
-    if (x.is_cuda and M <= 128 and N % 64 == 0 and K % 64 == 0
-            and x.is_contiguous()
-            and (silu_fuse or (N <= 8192 and K <= 8192)
-                 or (M > 32 and N <= 8192))):
+    # policy from measured MI355X data (profiles/kernel_bench_r01_final.log
+    # + M=128 sweep): the hand-written path wins where work/CU is scarce —
+    # small-M x small-N (qkv/o at M<=64), fused-SwiGLU gate_up at M<=64, and
+    # the K=14336 down-projection at 32<M<=128; hipBLASLt wins larger M and
+    # the 128k-vocab lm_head.
+    ok_shape = (x.is_cuda and N % 64 == 0 and K % 64 == 0
+                and x.is_contiguous() and M <= 128)
+    if ok_shape and ((M <= 64 and (silu_fuse or (N <= 8192 and K <= 8192)))
+                     or (32 < M and K >= 8192 and N <= 8192)):
         return linear_skinny(x, w, mode=2 if silu_fuse else 0)
     y = x @ w.t()
     if silu_fuse:
