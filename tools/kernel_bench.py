@@ -79,9 +79,11 @@ def bench_prefill(Hq=32, Hk=8, D=128, page=16):
                           "tflops": round(tf / t, 1), "ms": round(t * 1e3, 3)}))
 
 
-def bench_decode(Hq=32, Hk=8, D=128, page=16):
-    for (B, L, nsplit) in [(16, 576, 4), (16, 4096, 4), (64, 1024, 1),
-                           (256, 1024, 1), (1, 8192, 16)]:
+def bench_decode(Hq=32, Hk=8, D=128, page=None):
+    import itertools
+    for (B, L, nsplit), page in itertools.product(
+            [(16, 576, 4), (128, 576, 1), (64, 1024, 1), (256, 1024, 1)],
+            (16, 32, 64)):
         npages = B * ((L + page - 1) // page) + 1
         q = torch.randn(B, Hq * D, dtype=torch.bfloat16, device=DEV)
         kc = torch.randn(npages, Hk, page, D, dtype=torch.bfloat16, device=DEV)
@@ -91,8 +93,9 @@ def bench_decode(Hq=32, Hk=8, D=128, page=16):
         lens = torch.full((B,), L, dtype=torch.int32, device=DEV)
         gb = 2.0 * B * L * Hk * D * 2 / 1e9  # K+V bytes read
         t = timeit(lambda: ops.attn_decode(q, kc, vc, bt, lens, nsplit=nsplit))
-        print(json.dumps({"op": "attn_decode", "B": B, "L": L, "nsplit": nsplit,
-                          "gbps": round(gb / t, 1), "us": round(t * 1e6, 1)}))
+        print(json.dumps({"op": "attn_decode", "B": B, "L": L, "page": page,
+                          "nsplit": nsplit, "gbps": round(gb / t, 1),
+                          "us": round(t * 1e6, 1)}))
 
 
 def bench_norm_rope(H=4096):
