@@ -38,7 +38,7 @@ def choose_nsplit(bs: int, hk: int) -> int:
 
 class LLMEngine:
     def __init__(self, cfg: LlamaConfig, device="cuda",
-                 dtype=torch.bfloat16, page_size: int = 16,
+                 dtype=torch.bfloat16, page_size: int = 64,
                  kv_memory_frac: float = 0.80, num_pages: int | None = None,
                  max_num_seqs: int = 64, max_prefill_tokens: int = 8192,
                  max_waiting: int = 4096, enable_graphs: bool = True, eos_id: int = 2, seed: int = 0,
