@@ -15,10 +15,15 @@
 // caches: [npages, Hk, page_size, D];  slot[t] < 0 skips the cache write
 #include "common.h"
 
+// scale_ss (optional): [T,8] column-block sum-of-squares stats; when given,
+// rows are multiplied by rsqrt(sum/K+eps) — the RMSNorm scalar left over
+// after folding the norm weight into the QKV projection.  RoPE rotation is
+// linear, so pre-rotation scaling is exact.
 __global__ void __launch_bounds__(256) rope_cache_kernel(
     u16* __restrict__ q, u16* __restrict__ k, const u16* __restrict__ v,
     const i32* __restrict__ pos, const float* __restrict__ table,
     u16* __restrict__ kc, u16* __restrict__ vc, const i64* __restrict__ slot,
+    const float* __restrict__ scale_ss, float inv_k, float eps,
     int T, int Hq, int Hk, int D, i64 qs, i64 ks, i64 vs, int page_size) {
   const int half = D >> 1;
   const int qh4 = half >> 2;              // 4 rotation pairs per thread
@@ -26,6 +31,13 @@ __global__ void __launch_bounds__(256) rope_cache_kernel(
   const int v_work = Hk * (D >> 3);       // v copy, 8 elems per thread
   for (int t = blockIdx.x; t < T; t += gridDim.x) {
     const int p = pos[t];
+    float rstd = 1.f;
+    if (scale_ss) {
+      float st = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) st += scale_ss[t * 8 + j];
+      rstd = rsqrtf(st * inv_k + eps);
+    }
     const float* cosr = table + (size_t)p * D;
     const float* sinr = cosr + half;
     const i64 s = slot[t];
@@ -44,7 +56,7 @@ __global__ void __launch_bounds__(256) rope_cache_kernel(
       s16x4 oa, ob;
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
-        const float x1 = bf2f((u16)a[j]), x2 = bf2f((u16)b[j]);
+        const float x1 = bf2f((u16)a[j]) * rstd, x2 = bf2f((u16)b[j]) * rstd;
         oa[j] = (short)f2bf(x1 * c[j] - x2 * sn[j]);
         ob[j] = (short)f2bf(x2 * c[j] + x1 * sn[j]);
       }
@@ -62,8 +74,13 @@ __global__ void __launch_bounds__(256) rope_cache_kernel(
       for (int i = threadIdx.x; i < v_work; i += blockDim.x) {
         const int h = (i * 8) / D, d = (i * 8) % D;
         u16* vrow = vc + (((size_t)page * Hk + h) * page_size + off) * D;
-        *reinterpret_cast<s16x8*>(vrow + d) =
-            *reinterpret_cast<const s16x8*>(v + (size_t)t * vs + h * D + d);
+        s16x8 vv = *reinterpret_cast<const s16x8*>(v + (size_t)t * vs + h * D + d);
+        if (scale_ss) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            vv[j] = (short)f2bf(bf2f((u16)vv[j]) * rstd);
+        }
+        *reinterpret_cast<s16x8*>(vrow + d) = vv;
       }
     }
   }
@@ -71,13 +88,16 @@ __global__ void __launch_bounds__(256) rope_cache_kernel(
 
 AF_EXPORT int af_rope_cache(void* q, void* k, const void* v, const void* pos,
                             const void* table, void* kc, void* vc,
-                            const void* slot, int T, int Hq, int Hk, int D,
-                            i64 qs, i64 ks, i64 vs, int page_size, void* stream) {
+                            const void* slot, const void* scale_ss,
+                            float inv_k, float eps, int T, int Hq, int Hk,
+                            int D, i64 qs, i64 ks, i64 vs, int page_size,
+                            void* stream) {
   if (D % 8) return 9001;
   if (T == 0) return 0;
   int blocks = T < 2048 ? T : 2048;
   rope_cache_kernel<<<blocks, 256, 0, (hipStream_t)stream>>>(
       (u16*)q, (u16*)k, (const u16*)v, (const i32*)pos, (const float*)table,
-      (u16*)kc, (u16*)vc, (const i64*)slot, T, Hq, Hk, D, qs, ks, vs, page_size);
+      (u16*)kc, (u16*)vc, (const i64*)slot, (const float*)scale_ss, inv_k,
+      eps, T, Hq, Hk, D, qs, ks, vs, page_size);
   return af_last_err();
 }

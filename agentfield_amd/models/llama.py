@@ -125,7 +125,8 @@ class LlamaAttention(nn.Module):
         self.o = nn.Parameter(torch.empty(H, cfg.q_size))
         self.scale = 1.0 / math.sqrt(D)
 
-    def attend(self, qkv, positions, rope_tab, kv: KVCache, md: AttnMetadata):
+    def attend(self, qkv, positions, rope_tab, kv: KVCache, md: AttnMetadata,
+               scale: tuple | None = None):
         """RoPE + cache append + attention over the fused QKV buffer."""
         cfg = self.cfg
         # strided views straight into the fused projection (no copies)
@@ -133,7 +134,8 @@ class LlamaAttention(nn.Module):
         k = qkv[:, cfg.q_size:cfg.q_size + cfg.kv_size]
         v = qkv[:, cfg.q_size + cfg.kv_size:]
         ops.rope_cache(q, k, v, positions, rope_tab,
-                       kv.k[self.layer_idx], kv.v[self.layer_idx], md.slots)
+                       kv.k[self.layer_idx], kv.v[self.layer_idx], md.slots,
+                       scale=scale)
         if md.is_prefill:
             return ops.attn_prefill(q, kv.k[self.layer_idx],
                                     kv.v[self.layer_idx], md.block_table,
@@ -189,11 +191,25 @@ class LlamaLayer(nn.Module):
         sum-of-squares stats), residual-add + next-norm stats in the GEMM
         combine.  Zero standalone norm kernels per layer."""
         at, mlp = self.attn, self.mlp
-        qkv = ops.linear_skinny(residual, at.qkv, scale=(ss, self.eps))
-        o = at.attend(qkv, positions, rope_tab, kv, md)
+        M = residual.shape[0]
+        K = residual.shape[1]
+        small = M <= 64  # measured crossover vs hipBLASLt (BENCHMARKS.md)
+        if small:
+            qkv = ops.linear_skinny(residual, at.qkv, scale=(ss, self.eps))
+            o = at.attend(qkv, positions, rope_tab, kv, md)
+        else:
+            # blas GEMM on the un-normalized rows; the RMSNorm scalar rides
+            # along in rope_cache (RoPE is linear, so this is exact)
+            qkv = residual @ at.qkv.t()
+            o = at.attend(qkv, positions, rope_tab, kv, md,
+                          scale=(ss, K, self.eps))
         ops.linear_skinny(o, at.o, mode=4, residual=residual, ss_out=ss2)
-        act = ops.linear_skinny(residual, mlp.gate_up, mode=2,
-                                scale=(ss2, self.eps))
+        if small:
+            act = ops.linear_skinny(residual, mlp.gate_up, mode=2,
+                                    scale=(ss2, self.eps))
+        else:
+            act = ops.silu_and_mul(residual @ mlp.gate_up.t(),
+                                   scale=(ss2, K, self.eps))
         ops.linear_skinny(act, mlp.down, mode=4, residual=residual, ss_out=ss)
         return residual, ss
 
@@ -259,7 +275,7 @@ class LlamaForCausalLM(nn.Module):
         if not getattr(self, "_norms_folded", False):
             return False
         cfg = self.cfg
-        return (not getattr(self, "no_fused_decode", False) and T <= 64
+        return (not getattr(self, "no_fused_decode", False) and T <= 128
                 and cfg.hidden_size % 64 == 0
                 and cfg.intermediate_size % 64 == 0
                 and (cfg.q_size + 2 * cfg.kv_size) % 64 == 0

@@ -54,11 +54,14 @@ def rope_table(max_pos: int, head_dim: int, theta: float = 500000.0,
 def rope_cache(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                positions: torch.Tensor, table: torch.Tensor,
                kcache: torch.Tensor, vcache: torch.Tensor,
-               slots: torch.Tensor) -> None:
+               slots: torch.Tensor, scale: tuple | None = None) -> None:
     """Fused in-place RoPE on q/k + paged-cache append of k/v.
 
     q [T, Hq*D], k/v [T, Hk*D] — rows may be strided (views straight out of
     the fused QKV projection); caches [npages, Hk, page, D].
+    scale=(ss [T,8], k_dim, eps): rows multiplied by the RMSNorm scalar
+    rsqrt(mean+eps) — used when the QKV GEMM ran un-normalized (blas path of
+    the fused decode chain; norm weight folded into the projection).
     """
     _, Hk, page, D = kcache.shape
     T = q.shape[0]
@@ -67,26 +70,49 @@ def rope_cache(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     if not _on_gpu(q):
         q3 = q.unflatten(-1, (Hq, D))
         k3 = k.unflatten(-1, (Hk, D))
+        v3 = v.unflatten(-1, (Hk, D))
+        if scale is not None:
+            ss, kd, eps = scale
+            rstd = torch.rsqrt(ss.view(T, 8).sum(-1) / kd + eps)
+            for t3 in (q3, k3, v3):
+                t3.mul_(rstd[:, None, None].to(t3.dtype))
         ref.rope(q3, k3, positions, table)
-        ref.reshape_and_cache(k3, v.unflatten(-1, (Hk, D)), kcache, vcache, slots)
+        ref.reshape_and_cache(k3, v3, kcache, vcache, slots)
         return
+    if scale is not None:
+        ss, kd, eps = scale
+        ss_p, inv_k = _lib.ptr(ss), 1.0 / float(kd)
+    else:
+        ss_p, inv_k, eps = _lib.ptr(None), 0.0, 0.0
     rc = _lib.lib().af_rope_cache(
         _lib.ptr(q), _lib.ptr(k), _lib.ptr(v), _lib.ptr(positions.int()),
         _lib.ptr(table), _lib.ptr(kcache), _lib.ptr(vcache), _lib.ptr(slots),
+        ss_p, inv_k, eps,
         T, Hq, Hk, D, q.stride(0), k.stride(0), v.stride(0), page,
         _lib.cur_stream())
     _lib.check(rc, "af_rope_cache")
 
 
 # ---------------------------------------------------------------- activation
-def silu_and_mul(gate_up: torch.Tensor) -> torch.Tensor:
-    """gate_up [T, 2I] -> silu(gate)*up [T, I]."""
-    if not _on_gpu(gate_up):
-        return ref.silu_and_mul(gate_up)
+def silu_and_mul(gate_up: torch.Tensor,
+                 scale: tuple | None = None) -> torch.Tensor:
+    """gate_up [T, 2I] -> silu(gate*r)*(up*r) [T, I] where r is the optional
+    per-row RMSNorm scalar from scale=(ss [T,8], k_dim, eps)."""
     T, I2 = gate_up.shape
+    if not _on_gpu(gate_up):
+        if scale is not None:
+            ss, kd, eps = scale
+            rstd = torch.rsqrt(ss.view(T, 8).sum(-1) / kd + eps)
+            gate_up = (gate_up.float() * rstd[:, None]).to(gate_up.dtype)
+        return ref.silu_and_mul(gate_up)
+    if scale is not None:
+        ss, kd, eps = scale
+        ss_p, inv_k = _lib.ptr(ss), 1.0 / float(kd)
+    else:
+        ss_p, inv_k, eps = _lib.ptr(None), 0.0, 0.0
     out = torch.empty(T, I2 // 2, dtype=gate_up.dtype, device=gate_up.device)
-    rc = _lib.lib().af_silu_mul(_lib.ptr(out), _lib.ptr(gate_up), T, I2 // 2,
-                                _lib.cur_stream())
+    rc = _lib.lib().af_silu_mul(_lib.ptr(out), _lib.ptr(gate_up), ss_p, inv_k,
+                                eps, T, I2 // 2, _lib.cur_stream())
     _lib.check(rc, "af_silu_mul")
     return out
 

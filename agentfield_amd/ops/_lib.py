@@ -42,9 +42,9 @@ def _declare(l: ctypes.CDLL) -> None:
     f = ctypes.c_float
     u32 = ctypes.c_uint32
     l.af_rmsnorm.argtypes = [p, p, p, p, p, f, i, i, p]
-    l.af_rope_cache.argtypes = [p, p, p, p, p, p, p, p, i, i, i, i,
-                                i64, i64, i64, i, p]
-    l.af_silu_mul.argtypes = [p, p, i64, i64, p]
+    l.af_rope_cache.argtypes = [p, p, p, p, p, p, p, p, p, f, f,
+                                i, i, i, i, i64, i64, i64, i, p]
+    l.af_silu_mul.argtypes = [p, p, p, f, f, i64, i64, p]
     l.af_add.argtypes = [p, p, p, i64, p]
     l.af_reshape_and_cache.argtypes = [p, p, p, p, p, i, i, i, i, p]
     l.af_embedding.argtypes = [p, p, p, p, i, i, p]
