@@ -11,12 +11,13 @@
 //
 // Work decomposition (memory-bound; CDNA4-shaped):
 //   grid (B, Hk, NSPLIT), block = 256 threads = 4 waves.
-//   A wave iteration covers FOUR tokens: lane = tg*16 + dl, token subgroup
-//   tg = lane>>4, dim-group dl = lane&15 owning dims [dl*8, dl*8+8).  Each
-//   K/V row is a 16-lane x 16 B coalesced read; within-page consecutive
-//   tokens make the full wave read 1 KiB contiguous.  The score reduction is
-//   4 xor-shuffles within the 16-lane group + 2 across groups — ~2 shuffles
-//   per token vs 6 for a whole-wave-per-token layout, with 4x the load ILP.
+//   A wave iteration covers EIGHT tokens (two unrolled groups of four):
+//   lane = tg*16 + dl, token subgroup tg = lane>>4, dim-group dl = lane&15
+//   owning dims [dl*8, dl*8+8).  Each K/V row is a 16-lane x 16 B coalesced
+//   read; within-page consecutive tokens make a 4-token group's reads 1 KiB
+//   contiguous.  The score reduction is 4 xor-shuffles within the 16-lane
+//   group + 2 across groups — ~2 shuffles per token vs 6 for a
+//   whole-wave-per-token layout, with 8x the load ILP.
 //   Online softmax state (m, l) is tracked wave-wide over the 4-token tile;
 //   o accumulates per-lane (8 dims) and is tg-reduced once at the end.
 //   Wave partials merge through LDS; split partials merge in a second kernel.

@@ -243,7 +243,7 @@ def linear_skinny(x: torch.Tensor, w: torch.Tensor, mode: int = 0,
                   sk: int | None = None,
                   scale: tuple | None = None,
                   ss_out: torch.Tensor | None = None) -> torch.Tensor:
-    """Decode-path GEMM (M<=64) via the MFMA weight-streaming kernel.
+    """Decode-path GEMM (M<=256) via the MFMA weight-streaming kernel.
 
     mode 0: plain; mode 1: +residual (in-place update); mode 2: fused SwiGLU
     (w holds [gate|up] rows, returns [M, N/2]); mode 4: +residual AND
