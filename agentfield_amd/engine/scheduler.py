@@ -121,6 +121,7 @@ class Scheduler:
                 break
             self.waiting.popleft()
             cand.pages = self.alloc.alloc(need)
+            cand.alloc_epoch += 1
             cand.status = SeqStatus.RUNNING
             batch.append(cand)
             tokens += ntok
@@ -221,6 +222,8 @@ class NativeSchedulerAdapter:
             s = self.seqs[sid]
             s.pages = self.nat.pages(sid)
             s.status = SeqStatus.RUNNING
+            if r.is_prefill:
+                s.alloc_epoch += 1  # fresh allocation: invalidate bt rows
             batch.append(s)
         return ScheduleBatch(is_prefill=r.is_prefill, seqs=batch)
 

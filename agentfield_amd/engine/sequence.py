@@ -30,6 +30,7 @@ class Sequence:
     sampling: SamplingParams
     status: SeqStatus = SeqStatus.WAITING
     num_prefilled: int = 0
+    alloc_epoch: int = 0   # bumped on each (re)admission page allocation
     output_ids: list[int] = field(default_factory=list)
     pages: list[int] = field(default_factory=list)
     finish_reason: str | None = None
