@@ -366,3 +366,48 @@ def test_cancel_execution_endpoint(cp_server):
     assert r.status_code == 200 and r.json()["status"] == "cancelled"
     r = httpx.post(srv.base_url + "/api/v1/executions/exec_c1/cancel")
     assert r.status_code == 409
+
+
+def test_async_queue_backpressure_503():
+    """Bounded async queue: saturation returns 503 + backpressure metric
+    (reference C4: non-blocking submit, 503 'queue is full')."""
+    import asyncio
+    from agentfield_amd.controlplane import ControlPlane, create_app
+    from agentfield_amd.controlplane.server import Config
+
+    cp = ControlPlane(Config(background_services=False,
+                             async_workers=1, async_queue_capacity=2,
+                             agent_timeout=5.0))
+    srv = AppServer(create_app(cp)).start().wait_healthy()
+
+    # an agent that hangs long enough to wedge the single worker
+    slow = FastAPI()
+
+    @slow.post("/reasoners/stall")
+    async def stall(req: Request):
+        await asyncio.sleep(3.0)
+        return {"result": "late"}
+
+    slow_srv = AppServer(slow).start()
+    try:
+        cp.storage.upsert_node({"id": "slowpoke",
+                                "base_url": slow_srv.base_url,
+                                "reasoners": [{"id": "stall"}]})
+        codes = []
+        for _ in range(6):
+            r = httpx.post(srv.base_url + "/api/v1/execute/async/slowpoke.stall",
+                           json={"input": {}}, timeout=5.0)
+            codes.append(r.status_code)
+        assert 202 in codes
+        assert 503 in codes, f"expected backpressure, got {codes}"
+        m = httpx.get(srv.base_url + "/metrics").text
+        assert "agentfield_gateway_backpressure_total" in m
+        line = [l for l in m.splitlines()
+                if l.startswith("agentfield_gateway_backpressure_total ")][0]
+        assert float(line.split()[-1]) >= 1
+        # the rejected executions are failed, not lost
+        failed = [c for c in codes if c == 503]
+        assert len(failed) >= 1
+    finally:
+        slow_srv.stop()
+        srv.stop()
