@@ -411,3 +411,64 @@ def test_async_queue_backpressure_503():
     finally:
         slow_srv.stop()
         srv.stop()
+
+
+def test_memory_sse_and_history(cp_server):
+    srv, _ = cp_server
+    got = []
+
+    def listen():
+        with httpx.stream("GET", srv.base_url + "/api/v1/memory/events/sse",
+                          timeout=10.0) as resp:
+            for line in resp.iter_lines():
+                if line.startswith("data:"):
+                    got.append(json.loads(line[5:]))
+                    return
+
+    t = threading.Thread(target=listen, daemon=True)
+    t.start()
+    time.sleep(0.3)
+    t0 = time.time()
+    httpx.post(srv.base_url + "/api/v1/memory/set",
+               json={"key": "sse_k", "value": 9, "scope": "global"})
+    t.join(timeout=10.0)
+    assert got and got[0]["key"] == "sse_k" and got[0]["op"] == "set"
+    hist = httpx.get(srv.base_url + "/api/v1/memory/events/history",
+                     params={"since": t0 - 1}).json()["events"]
+    assert any(e["key"] == "sse_k" for e in hist)
+    # vector delete roundtrip
+    httpx.post(srv.base_url + "/api/v1/memory/vector/set",
+               json={"key": "vd", "embedding": [1, 2], "scope": "global"})
+    r = httpx.post(srv.base_url + "/api/v1/memory/vector/delete",
+                   json={"key": "vd", "scope": "global"}).json()
+    assert r["deleted"] is True
+
+
+def test_ui_dashboard_served(cp_server):
+    srv, _ = cp_server
+    r = httpx.get(srv.base_url + "/")
+    assert r.status_code == 200
+    assert "agentfield-amd" in r.text and "text/html" in r.headers["content-type"]
+
+
+def test_admin_grpc_execution_and_status(cp_server, greeting_agent):
+    import socket
+    from agentfield_amd.controlplane.admin_grpc import (AdminClient,
+                                                        start_admin_grpc)
+    srv, cp = cp_server
+    r = httpx.post(srv.base_url + "/api/v1/execute/greeter.greet",
+                   json={"input": {"name": "grpc"}}, timeout=20.0)
+    eid = r.json()["execution_id"]
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    server = start_admin_grpc(cp, port=port)
+    try:
+        cl = AdminClient(f"127.0.0.1:{port}")
+        st = cl.server_status()
+        assert st["status"] == "healthy" and st["nodes"] >= 1
+        ex = cl.get_execution(eid)
+        assert ex["execution"]["status"] == "completed"
+    finally:
+        server.stop(grace=0)
