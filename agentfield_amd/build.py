@@ -29,6 +29,7 @@ HIP_SOURCES = [
     "attn_decode.hip",
     "attn_prefill.hip",
     "gemm.hip",
+    "gemm_pipelined.hip",
     "gemm_skinny.hip",
     "sampling.hip",
     "probe.hip",

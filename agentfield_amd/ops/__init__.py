@@ -308,6 +308,20 @@ def gemm_bf16(a: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return c
 
 
+def gemm_bf16_pipelined(a: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """256x256-tile counted-vmcnt pipelined MFMA GEMM (K % 128 == 0)."""
+    if not _on_gpu(a):
+        return (a.float() @ w.float().t()).to(a.dtype)
+    M, K = a.shape
+    N = w.shape[0]
+    c = torch.empty(M, N, dtype=a.dtype, device=a.device)
+    rc = _lib.lib().af_gemm_bf16_pipelined(_lib.ptr(c), _lib.ptr(a),
+                                           _lib.ptr(w), M, N, K,
+                                           _lib.cur_stream())
+    _lib.check(rc, "af_gemm_bf16_pipelined")
+    return c
+
+
 # ---------------------------------------------------------------- sampling
 class SamplerState:
     """Device-side scratch for graph-capturable sampling."""

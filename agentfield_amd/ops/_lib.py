@@ -53,6 +53,7 @@ def _declare(l: ctypes.CDLL) -> None:
     l.af_attn_prefill.argtypes = [p, p, p, p, p, p, p, p, p, f,
                                   i, i, i, i, i64, i, i, p]
     l.af_gemm_bf16.argtypes = [p, p, p, i, i, i, p]
+    l.af_gemm_bf16_pipelined.argtypes = [p, p, p, i, i, i, p]
     l.af_gemm_skinny.argtypes = [p, p, p, p, p, i, i, i, i, i,
                                  p, f, p, p]
     l.af_sample.argtypes = [p, p, p, p, p, p, u32, i, i, p]
@@ -64,7 +65,8 @@ def _declare(l: ctypes.CDLL) -> None:
     l.af_device_sync.argtypes = []
     for fn in ("af_rmsnorm", "af_rope_cache", "af_silu_mul", "af_add",
                "af_reshape_and_cache", "af_embedding", "af_attn_decode",
-               "af_attn_prefill", "af_gemm_bf16", "af_gemm_skinny", "af_sample",
+               "af_attn_prefill", "af_gemm_bf16", "af_gemm_bf16_pipelined",
+               "af_gemm_skinny", "af_sample",
                "af_sample_topkp", "af_gather_rows", "af_mfma_probe", "af_axpy",
                "af_device_sync"):
         getattr(l, fn).restype = ctypes.c_int

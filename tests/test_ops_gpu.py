@@ -254,6 +254,20 @@ def test_gemm():
         assert close(c, want, atol=tol, rtol=5e-2), f"gemm {M}x{N}x{K}"
 
 
+def test_gemm_pipelined():
+    # 256x256-tile pipelined variant: exercises exact-tile, ragged M/N, and
+    # a real model shape (K % 128 == 0 required).
+    for (M, N, K) in [(256, 256, 128), (512, 512, 512), (300, 700, 256),
+                      (512, 4096, 4096)]:
+        a = rnd(M, K, seed=M + N + 7, scale=0.5)
+        w = rnd(N, K, seed=M + N + 8, scale=0.5)
+        c = ops.gemm_bf16_pipelined(a, w)
+        torch.cuda.synchronize()
+        want = (a.float() @ w.float().t())
+        tol = 0.1 + 0.02 * math.sqrt(K)
+        assert close(c, want, atol=tol, rtol=5e-2), f"gemm_pipe {M}x{N}x{K}"
+
+
 def test_gemm_skinny():
     for M in (1, 7, 16, 33, 64, 96, 128, 192, 256):
         for (N, K) in [(6144, 4096), (4096, 14336), (128256, 4096)]:

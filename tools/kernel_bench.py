@@ -36,9 +36,13 @@ def bench_gemm():
         tf = 2.0 * M * N * K / 1e12
         t_ours = timeit(lambda: ops.gemm_bf16(a, w))
         t_blas = timeit(lambda: a @ w.t())
-        print(json.dumps({"op": "gemm_bf16", "MNK": [M, N, K],
-                          "ours_tflops": round(tf / t_ours, 1),
-                          "hipblaslt_tflops": round(tf / t_blas, 1)}))
+        row = {"op": "gemm_bf16", "MNK": [M, N, K],
+               "ours_tflops": round(tf / t_ours, 1),
+               "hipblaslt_tflops": round(tf / t_blas, 1)}
+        if K % 128 == 0 and M >= 256:
+            t_pipe = timeit(lambda: ops.gemm_bf16_pipelined(a, w))
+            row["pipelined_tflops"] = round(tf / t_pipe, 1)
+        print(json.dumps(row))
 
 
 def bench_skinny():
