@@ -117,13 +117,32 @@ class MemoryInterface:
 
     # ------------------------------------------------------ change events
     def on_change(self, pattern: str = "*"):
-        """Decorator: watch memory keys (glob pattern) via history polling
-        over /api/v1/memory/events/history."""
+        """Decorator: watch memory keys (glob pattern).  Events arrive over
+        the server's SSE stream (/api/v1/memory/events/sse, push latency);
+        if the stream drops, missed events are recovered from
+        /api/v1/memory/events/history before reconnecting."""
         def deco(fn):
             self._watchers.append((pattern, fn))
             self._ensure_watch_thread()
             return fn
         return deco
+
+    def _dispatch_event(self, ev: dict, since: float) -> float:
+        since = max(since, ev.get("at", since))
+        for pattern, fn in self._watchers:
+            if fnmatch.fnmatch(ev.get("key", ""), pattern):
+                try:
+                    val = ev.get("value")
+                    if isinstance(val, str):
+                        try:
+                            val = json.loads(val)
+                        except ValueError:
+                            pass
+                    fn({"key": ev.get("key"), "op": ev.get("op"),
+                        "value": val, "scope": ev.get("scope")})
+                except Exception:
+                    pass
+        return since
 
     def _ensure_watch_thread(self):
         if self._watch_thread is not None:
@@ -132,28 +151,38 @@ class MemoryInterface:
             since = time.time()
             base = self.client.base_url
             import httpx
-            while not self._stop.wait(1.0):
+            while not self._stop.is_set():
                 try:
-                    r = httpx.get(f"{base}/api/v1/memory/events/history",
-                                  params={"since": since}, timeout=5.0)
-                    events = r.json().get("events", [])
-                except Exception:
-                    continue
-                for ev in events:
-                    since = max(since, ev.get("at", since))
-                    for pattern, fn in self._watchers:
-                        if fnmatch.fnmatch(ev.get("key", ""), pattern):
+                    with httpx.stream(
+                            "GET", f"{base}/api/v1/memory/events/sse",
+                            timeout=httpx.Timeout(5.0, read=30.0)) as resp:
+                        # catch-up AFTER the stream is subscribed: anything
+                        # published while disconnected comes from history,
+                        # anything newer is already buffered on the stream
+                        # (no window where an event can miss both).
+                        try:
+                            r = httpx.get(
+                                f"{base}/api/v1/memory/events/history",
+                                params={"since": since}, timeout=5.0)
+                            for ev in r.json().get("events", []):
+                                since = self._dispatch_event(ev, since)
+                        except Exception:
+                            pass
+                        for line in resp.iter_lines():
+                            if self._stop.is_set():
+                                return
+                            if not line.startswith("data:"):
+                                continue  # keepalive comments etc.
                             try:
-                                val = ev.get("value")
-                                if isinstance(val, str):
-                                    try:
-                                        val = json.loads(val)
-                                    except ValueError:
-                                        pass
-                                fn({"key": ev.get("key"), "op": ev.get("op"),
-                                    "value": val, "scope": ev.get("scope")})
-                            except Exception:
-                                pass
+                                ev = json.loads(line[5:].strip())
+                            except ValueError:
+                                continue
+                            if ev.get("at", 0) <= since:
+                                continue  # already delivered via catch-up
+                            since = self._dispatch_event(ev, since)
+                except Exception:
+                    pass
+                self._stop.wait(0.5)
         self._watch_thread = threading.Thread(target=loop, daemon=True,
                                               name="af-memory-watch")
         self._watch_thread.start()

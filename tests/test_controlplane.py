@@ -444,6 +444,38 @@ def test_memory_sse_and_history(cp_server):
     assert r["deleted"] is True
 
 
+def test_memory_on_change_sse_push(cp_server):
+    """SDK on_change watcher receives events over the live SSE stream with
+    push latency (not the 1 s polling fallback)."""
+    from agentfield_amd.sdk.client import AgentFieldClient
+    from agentfield_amd.sdk.memory import MemoryInterface
+
+    srv, _ = cp_server
+    mem = MemoryInterface(AgentFieldClient(srv.base_url), "watcher-node")
+    seen = []
+
+    @mem.on_change("watch_*")
+    def _on(ev):
+        seen.append(ev)
+
+    try:
+        # let the watcher connect its SSE stream
+        wait_until(lambda: mem._watch_thread is not None and
+                   mem._watch_thread.is_alive(), 5.0)
+        time.sleep(0.5)
+        httpx.post(srv.base_url + "/api/v1/memory/set",
+                   json={"key": "watch_k1", "value": {"n": 1},
+                         "scope": "global"})
+        httpx.post(srv.base_url + "/api/v1/memory/set",
+                   json={"key": "other_k", "value": 0, "scope": "global"})
+        assert wait_until(lambda: len(seen) >= 1, 5.0)
+        time.sleep(0.3)
+        assert [e["key"] for e in seen] == ["watch_k1"]  # glob filtered
+        assert seen[0]["op"] == "set" and seen[0]["value"] == {"n": 1}
+    finally:
+        mem.stop()
+
+
 def test_ui_dashboard_served(cp_server):
     srv, _ = cp_server
     r = httpx.get(srv.base_url + "/")
