@@ -1,4 +1,5 @@
 from .agent import Agent
+from .agent_router import AgentRouter
 from .ai import AgentAI, AIConfig, ByteTokenizer, EngineRunner
 from .client import AgentFieldClient
 from .execution_context import ExecutionContext, current_context

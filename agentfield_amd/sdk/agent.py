@@ -207,6 +207,24 @@ class Agent(FastAPI):
             return local
         return deco
 
+    def include_router(self, router, prefix: str = ""):
+        """Register an AgentRouter's reasoners/skills on this agent with the
+        prefix dotted into each name (reference: include_router prefix
+        rewriting).  Returns {rewritten_name: local_callable}."""
+        pre = (prefix or router.prefix).strip("./")
+        out = {}
+        for item in router._pending:
+            name = f"{pre}.{item['name']}" if pre else item["name"]
+            if item["kind"] == "reasoner":
+                deco = self.reasoner(name=name, tags=item.get("tags"),
+                                     vc=item.get("vc", False))
+            else:
+                deco = self.skill(name=name, tags=item.get("tags"),
+                                  cache_results=item.get("cache_results",
+                                                         False))
+            out[name] = deco(item["fn"])
+        return out
+
     @staticmethod
     def _bind_args(meta: _FunctionMeta, args, kwargs) -> dict:
         out = dict(kwargs)

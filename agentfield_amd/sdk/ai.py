@@ -242,10 +242,20 @@ class AgentAI:
                 return text  # schema validation failed; return raw text
         return text
 
-    def with_audio(self, *a, **kw):
-        raise NotImplementedError(
-            "multimodal inputs need a multimodal model; the Llama family "
-            "configs shipped with this framework are text-only")
+    def with_multimodal(self, *parts, **kw):
+        """ai() over mixed text/image/audio inputs.  Detection and message
+        assembly are model-independent (sdk/multimodal.py); a text-only
+        model raises UnsupportedModality with the offending part types."""
+        from .multimodal import build_content, require_text
+        cfg = self.config.merged(**{k: v for k, v in kw.items()
+                                    if k in AIConfig.__dataclass_fields__})
+        content = build_content(parts)
+        text = require_text(content, cfg.model)  # text-only model families
+        kw.pop("user", None)
+        return self(user=text, **kw)
 
-    with_vision = with_audio
-    with_multimodal = with_audio
+    def with_vision(self, prompt, *images, **kw):
+        return self.with_multimodal(prompt, *images, **kw)
+
+    def with_audio(self, prompt, *clips, **kw):
+        return self.with_multimodal(prompt, *clips, **kw)

@@ -444,6 +444,44 @@ def test_memory_sse_and_history(cp_server):
     assert r["deleted"] is True
 
 
+def test_include_router_prefix_rewriting(cp_server, greeting_agent):
+    """AgentRouter composition: prefixed names register on the agent, stay
+    locally callable, and resolve through the control plane's
+    node.<dotted.reasoner> target (split on first dot only)."""
+    from agentfield_amd.sdk import AgentRouter
+
+    srv, _ = cp_server
+    _agent_srv, app = greeting_agent
+
+    billing = AgentRouter(prefix="billing")
+
+    @billing.reasoner()
+    def report(month: str):
+        return {"month": month, "total": 42}
+
+    inner = AgentRouter()
+
+    @inner.skill()
+    def audit(x: int):
+        return x * 2
+
+    billing.include_router(inner, prefix="ops")   # nested: billing.ops.audit
+    locals_ = app.include_router(billing)
+    assert set(locals_) == {"billing.report", "billing.ops.audit"}
+    assert locals_["billing.report"]("jan") == {"month": "jan", "total": 42}
+
+    # re-register so the control plane learns the new reasoners
+    assert app.register()
+    r = httpx.post(srv.base_url + "/api/v1/execute/greeter.billing.report",
+                   json={"input": {"month": "feb"}}, timeout=20.0)
+    assert r.status_code == 200, r.text
+    assert r.json()["result"] == {"month": "feb", "total": 42}
+    r = httpx.post(srv.base_url + "/api/v1/execute/greeter.billing.ops.audit",
+                   json={"input": {"x": 21}}, timeout=20.0)
+    assert r.status_code == 200, r.text
+    assert r.json()["result"] == 42
+
+
 def test_memory_on_change_sse_push(cp_server):
     """SDK on_change watcher receives events over the live SSE stream with
     push latency (not the 1 s polling fallback)."""

@@ -90,3 +90,43 @@ def test_runner_direct_submit(tiny_runner):
                                                      ignore_eos=True))
     assert w["done"].wait(60)
     assert len(w["output"]) == 4
+
+
+def test_multimodal_detection():
+    from agentfield_amd.sdk.multimodal import (UnsupportedModality,
+                                               build_content, detect_part,
+                                               require_text, sniff_bytes)
+
+    png = b"\x89PNG\r\n\x1a\n" + b"\x00" * 16
+    assert sniff_bytes(png[:16]) == "image/png"
+    assert sniff_bytes(b"RIFF\x00\x00\x00\x00WAVEfmt ") == "audio/wav"
+    assert sniff_bytes(b"RIFF\x00\x00\x00\x00WEBPVP8 ") == "image/webp"
+
+    p = detect_part(png)
+    assert p["type"] == "image_url"
+    assert p["image_url"]["url"].startswith("data:image/png;base64,")
+    p = detect_part("data:audio/wav;base64,UklGRg==")
+    assert p["type"] == "input_audio" and p["input_audio"]["format"] == "wav"
+    p = detect_part("https://example.com/cat.jpg?s=1")
+    assert p["type"] == "image_url" and p["image_url"]["url"].endswith("?s=1")
+    assert detect_part("plain words")["type"] == "text"
+
+    # all-text collapses to a string
+    assert build_content(["a", "b"]) == "a\nb"
+    mixed = build_content(["look:", png])
+    assert isinstance(mixed, list) and mixed[1]["type"] == "image_url"
+    assert require_text("just text", "tiny") == "just text"
+    with pytest.raises(UnsupportedModality):
+        require_text(mixed, "tiny")
+
+
+def test_ai_with_multimodal_text_only(tiny_runner):
+    ai = AgentAI(AIConfig(model="tiny", max_tokens=4, timeout=120))
+    # pure-text parts route through normally
+    out = ai.with_multimodal("describe", "this scene")
+    assert isinstance(out, str)
+    # an image part against the text-only tiny model raises loudly
+    from agentfield_amd.sdk.multimodal import UnsupportedModality
+    png = b"\x89PNG\r\n\x1a\n" + b"\x00" * 8
+    with pytest.raises(UnsupportedModality):
+        ai.with_vision("describe", png)
