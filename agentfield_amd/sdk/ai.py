@@ -88,6 +88,7 @@ class EngineRunner:
         self.engine = engine
         self.tokenizer = tokenizer or ByteTokenizer(engine.cfg.vocab_size)
         self._submit: queue.Queue = queue.Queue()
+        self._cancel: queue.Queue = queue.Queue()
         self._waiters: dict[int, dict] = {}
         self._lock = threading.Lock()
         self._wake = threading.Event()
@@ -116,20 +117,55 @@ class EngineRunner:
             raise TimeoutError("engine generate timed out")
         if w["error"]:
             raise RuntimeError(w["error"])
-        return self.tokenizer.decode(w["output"])
+        text = self.tokenizer.decode(w["output"])
+        for s in cfg.stop or ():
+            idx = text.find(s)
+            if idx >= 0:
+                text = text[:idx]
+        return text
 
     def stream_text(self, prompt: str, cfg: AIConfig):
         ids = self.tokenizer.encode(prompt)
         sp = SamplingParams(max_tokens=cfg.max_tokens,
                             temperature=cfg.temperature)
         sq: queue.Queue = queue.Queue()
-        self.submit(ids, sp, stream_q=sq)
-        while True:
-            tok, done = sq.get(timeout=cfg.timeout)
-            if tok is not None:
-                yield self.tokenizer.decode([tok])
-            if done:
-                return
+        w = self.submit(ids, sp, stream_q=sq)
+        stops = tuple(s for s in (cfg.stop or ()) if s)
+        # holdback window: never emit the last max(len(stop))-1 chars until
+        # more text arrives, so a stop string split across token pieces is
+        # still caught before any of it reaches the caller
+        hold = max((len(s) for s in stops), default=1) - 1
+        acc = ""
+        try:
+            while True:
+                tok, done = sq.get(timeout=cfg.timeout)
+                if tok is not None:
+                    acc += self.tokenizer.decode([tok])
+                if stops:
+                    cuts = [i for i in (acc.find(s) for s in stops) if i >= 0]
+                    if cuts:
+                        head = acc[:min(cuts)]
+                        if head:
+                            yield head
+                        self.cancel(w)  # free the engine's remaining decode
+                        return
+                if done:
+                    if acc:
+                        yield acc
+                    return
+                if len(acc) > hold:
+                    emit, acc = acc[:len(acc) - hold], acc[len(acc) - hold:]
+                    if emit:
+                        yield emit
+        except GeneratorExit:
+            self.cancel(w)  # caller abandoned the stream
+            raise
+
+    def cancel(self, waiter: dict) -> None:
+        """Cancel a submitted request (engine calls stay on the loop
+        thread; the waiter is resolved as cancelled)."""
+        self._cancel.put(waiter)
+        self._wake.set()
 
     def _loop(self):
         eng = self.engine
@@ -146,7 +182,20 @@ class EngineRunner:
                     waiter["error"] = "engine queue full"
                     waiter["done"].set()
                 else:
+                    waiter["rid"] = rid
                     pending[rid] = waiter
+                moved = True
+            while True:
+                try:
+                    w = self._cancel.get_nowait()
+                except queue.Empty:
+                    break
+                rid = w.get("rid")
+                if rid is not None and rid in pending:
+                    eng.cancel(rid)
+                    pending.pop(rid, None)
+                    w["error"] = w["error"] or "cancelled"
+                    w["done"].set()
                 moved = True
             if eng.has_work():
                 for (rid, tok, done) in eng.step():

@@ -80,9 +80,12 @@ def test_ai_concurrent_calls_batched(tiny_runner):
 
 
 def test_ai_streaming(tiny_runner):
-    ai = AgentAI(AIConfig(model="tiny", max_tokens=5, timeout=120))
+    ai = AgentAI(AIConfig(model="tiny", max_tokens=5, timeout=120,
+                          temperature=0.0))
     pieces = list(ai("stream me", stream=True))
-    assert len(pieces) == 5
+    assert pieces and all(p for p in pieces)  # no empty pieces leak
+    # greedy decode: streamed text == blocking text
+    assert "".join(pieces) == ai("stream me")
 
 
 def test_runner_direct_submit(tiny_runner):
@@ -130,3 +133,18 @@ def test_ai_with_multimodal_text_only(tiny_runner):
     png = b"\x89PNG\r\n\x1a\n" + b"\x00" * 8
     with pytest.raises(UnsupportedModality):
         ai.with_vision("describe", png)
+
+
+def test_stop_sequences(tiny_runner):
+    """String stop sequences truncate both blocking and streaming output;
+    the streaming holdback never leaks any part of the stop string."""
+    ai = AgentAI(AIConfig(model="tiny", max_tokens=24, timeout=120,
+                          temperature=0.0))
+    full = ai("halt on demand")
+    assert isinstance(full, str) and len(full) > 4
+    stop = full[3:5]  # greedy decode -> same text next run
+    want = full[:full.find(stop)]
+    assert ai("halt on demand", stop=(stop,)) == want
+    streamed = "".join(ai("halt on demand", stream=True, stop=(stop,)))
+    assert streamed == want
+    assert stop not in streamed
