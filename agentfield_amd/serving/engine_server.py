@@ -71,6 +71,7 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
             ids = [int(x) for x in body["prompt_ids"]]
         else:
             ids = runner.tokenizer.encode(body.get("prompt", ""))
+        stops = tuple(s for s in body.get("stop", []) if s)
         stream = bool(body.get("stream", False))
         if not stream:
             import anyio
@@ -84,20 +85,48 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
                 return JSONResponse({"error": "timeout"}, status_code=504)
             if w["error"]:
                 return JSONResponse({"error": w["error"]}, status_code=503)
-            return {"output_ids": w["output"],
-                    "text": runner.tokenizer.decode(w["output"])}
+            text = runner.tokenizer.decode(w["output"])
+            for s in stops:
+                idx = text.find(s)
+                if idx >= 0:
+                    text = text[:idx]
+            return {"output_ids": w["output"], "text": text}
 
         sq: queue.Queue = queue.Queue()
-        runner.submit(ids, sp, stream_q=sq)
+        w = runner.submit(ids, sp, stream_q=sq)
 
         async def gen():
             import anyio
+            if not stops:
+                while True:
+                    tok, done = await anyio.to_thread.run_sync(sq.get)
+                    piece = (runner.tokenizer.decode([tok])
+                             if tok is not None else "")
+                    yield f"data: {json.dumps({'token': tok, 'text': piece, 'done': done})}\n\n"
+                    if done:
+                        return
+            # stop-sequence mode: holdback window so a stop split across
+            # token pieces never reaches the client (events coalesce text;
+            # token ids are omitted)
+            hold = max(len(s) for s in stops) - 1
+            acc = ""
             while True:
                 tok, done = await anyio.to_thread.run_sync(sq.get)
-                piece = runner.tokenizer.decode([tok]) if tok is not None else ""
-                yield f"data: {json.dumps({'token': tok, 'text': piece, 'done': done})}\n\n"
-                if done:
+                if tok is not None:
+                    acc += runner.tokenizer.decode([tok])
+                cuts = [i for i in (acc.find(s) for s in stops) if i >= 0]
+                if cuts:
+                    head = acc[:min(cuts)]
+                    yield f"data: {json.dumps({'token': None, 'text': head, 'done': True})}\n\n"
+                    runner.cancel(w)
                     return
+                if done:
+                    yield f"data: {json.dumps({'token': None, 'text': acc, 'done': True})}\n\n"
+                    return
+                if len(acc) > hold:
+                    emit, acc = acc[:len(acc) - hold], acc[len(acc) - hold:]
+                    if emit:
+                        yield f"data: {json.dumps({'token': None, 'text': emit, 'done': False})}\n\n"
         return StreamingResponse(gen(), media_type="text/event-stream")
 
     return app

@@ -27,11 +27,13 @@ class RemoteRunner:
         return self.router.generate(prompt=prompt,
                                     max_tokens=cfg.max_tokens,
                                     temperature=cfg.temperature,
+                                    stop=list(cfg.stop or ()),
                                     timeout=cfg.timeout)["text"]
 
     def stream_text(self, prompt: str, cfg):
         yield from self.router.stream(prompt=prompt, max_tokens=cfg.max_tokens,
                                       temperature=cfg.temperature,
+                                      stop=list(cfg.stop or ()),
                                       timeout=cfg.timeout)
 
 

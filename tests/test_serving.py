@@ -67,6 +67,26 @@ def test_router_streaming(replicas):
     assert len(pieces) >= 1
 
 
+def test_server_stop_sequences(replicas):
+    """Stop strings apply server-side in both modes (greedy => deterministic),
+    so DP-remote behavior matches the in-process EngineRunner."""
+    import httpx
+    body = {"prompt": "stop here", "max_tokens": 96, "ignore_eos": True,
+            "temperature": 0.0}
+    full = httpx.post(replicas[0].base_url + "/v1/generate", json=body,
+                      timeout=60.0).json()["text"]
+    # byte tokenizer drops non-byte ids, so text is sparse; need >=2 chars
+    assert len(full) >= 2
+    stop = full[1:3] if len(full) >= 4 else full[1:2]
+    want = full[:full.find(stop)]
+    got = httpx.post(replicas[0].base_url + "/v1/generate",
+                     json={**body, "stop": [stop]}, timeout=60.0).json()
+    assert got["text"] == want
+    router = DPRouter([replicas[0].base_url])
+    streamed = "".join(router.stream(**{**body, "stop": [stop]}))
+    assert streamed == want and stop not in streamed
+
+
 def test_ai_via_remote_engines(replicas, monkeypatch):
     from agentfield_amd.sdk import ai as ai_mod
     monkeypatch.setenv("AGENTFIELD_ENGINE_URLS",
