@@ -277,10 +277,21 @@ class AgentAI:
                          f"schema:\n{json.dumps(schema_json)}")
         parts.append(f"<|user|>\n{body}\n<|assistant|>\n")
         prompt = "\n".join(parts)
-        runner = get_runner(cfg)
         if stream:
-            return runner.stream_text(prompt, cfg)
-        text = runner.generate_text(prompt, cfg)
+            return get_runner(cfg).stream_text(prompt, cfg)
+        text = None
+        last_err = None
+        for model in (cfg.model,) + tuple(cfg.fallback_models or ()):
+            mcfg = cfg if model == cfg.model else cfg.merged(model=model)
+            try:
+                text = get_runner(mcfg).generate_text(prompt, mcfg)
+                break
+            except (RuntimeError, TimeoutError, KeyError) as e:
+                last_err = e  # engine full / timed out / unknown model
+        if text is None:
+            raise RuntimeError(
+                f"all models failed ({(cfg.model,) + tuple(cfg.fallback_models or ())}): "
+                f"{last_err}") from last_err
         if schema is not None:
             try:
                 data = json.loads(text)

@@ -129,6 +129,146 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
                         yield f"data: {json.dumps({'token': None, 'text': emit, 'done': False})}\n\n"
         return StreamingResponse(gen(), media_type="text/event-stream")
 
+    # ------------------------------------------------- OpenAI-compatible
+    # /v1/chat/completions and /v1/completions speak the OpenAI wire
+    # protocol (including SSE chunk streaming + [DONE]), so any OpenAI
+    # client library can point at an engine replica directly — the serving
+    # analog of the reference's LiteLLM provider seam.
+    def _assemble_chat_prompt(messages) -> str:
+        from ..sdk.multimodal import require_text
+        parts = []
+        for m in messages:
+            role = m.get("role", "user")
+            content = require_text(m.get("content", ""), model_name)
+            parts.append(f"<|{role}|>\n{content}")
+        parts.append("<|assistant|>\n")
+        return "\n".join(parts)
+
+    def _finish_reason(n_out: int, max_tokens: int, stopped: bool) -> str:
+        if stopped:
+            return "stop"
+        return "length" if n_out >= max_tokens else "stop"
+
+    async def _oai_generate(body: dict, prompt: str, kind: str):
+        import time as _time
+        import uuid
+
+        import anyio
+        max_tokens = int(body.get("max_tokens", 128))
+        sp = SamplingParams(max_tokens=max_tokens,
+                            temperature=float(body.get("temperature", 0.0)))
+        stop_in = body.get("stop") or []
+        stops = tuple(s for s in ([stop_in] if isinstance(stop_in, str)
+                                  else stop_in) if s)
+        ids = runner.tokenizer.encode(prompt)
+        rid = f"{'chatcmpl' if kind == 'chat' else 'cmpl'}-{uuid.uuid4().hex[:24]}"
+        created = int(_time.time())
+        base = {"id": rid, "created": created, "model": model_name}
+
+        if not body.get("stream", False):
+            w = runner.submit(ids, sp)
+
+            def wait():
+                w["done"].wait(float(body.get("timeout", 600)))
+                return w
+            w = await anyio.to_thread.run_sync(wait)
+            if not w["done"].is_set():
+                return JSONResponse({"error": {"message": "timeout",
+                                               "type": "timeout"}},
+                                    status_code=504)
+            if w["error"]:
+                return JSONResponse({"error": {"message": w["error"],
+                                               "type": "overloaded"}},
+                                    status_code=503)
+            text = runner.tokenizer.decode(w["output"])
+            stopped = False
+            for s in stops:
+                idx = text.find(s)
+                if idx >= 0:
+                    text, stopped = text[:idx], True
+            fr = _finish_reason(len(w["output"]), max_tokens, stopped)
+            usage = {"prompt_tokens": len(ids),
+                     "completion_tokens": len(w["output"]),
+                     "total_tokens": len(ids) + len(w["output"])}
+            if kind == "chat":
+                choice = {"index": 0, "finish_reason": fr,
+                          "message": {"role": "assistant", "content": text}}
+                return {**base, "object": "chat.completion",
+                        "choices": [choice], "usage": usage}
+            choice = {"index": 0, "finish_reason": fr, "text": text}
+            return {**base, "object": "text_completion",
+                    "choices": [choice], "usage": usage}
+
+        sq: queue.Queue = queue.Queue()
+        w = runner.submit(ids, sp, stream_q=sq)
+        obj = "chat.completion.chunk" if kind == "chat" else "text_completion"
+
+        def chunk(piece: str | None, fr: str | None) -> str:
+            if kind == "chat":
+                delta = {"content": piece} if piece is not None else {}
+                if piece is None and fr is None:
+                    delta = {"role": "assistant"}
+                c = {"index": 0, "delta": delta, "finish_reason": fr}
+            else:
+                c = {"index": 0, "text": piece or "", "finish_reason": fr}
+            return "data: " + json.dumps({**base, "object": obj,
+                                          "choices": [c]}) + "\n\n"
+
+        async def gen():
+            n_out = 0
+            if kind == "chat":
+                yield chunk(None, None)  # leading role delta
+            hold = max((len(s) for s in stops), default=1) - 1
+            acc = ""
+            stopped = False
+            while True:
+                tok, done = await anyio.to_thread.run_sync(sq.get)
+                if tok is not None:
+                    n_out += 1
+                    acc += runner.tokenizer.decode([tok])
+                cuts = [i for i in (acc.find(s) for s in stops) if i >= 0]
+                if cuts:
+                    head = acc[:min(cuts)]
+                    if head:
+                        yield chunk(head, None)
+                    runner.cancel(w)
+                    stopped, done = True, True
+                elif done:
+                    if acc:
+                        yield chunk(acc, None)
+                elif len(acc) > hold:
+                    emit, acc = acc[:len(acc) - hold], acc[len(acc) - hold:]
+                    if emit:
+                        yield chunk(emit, None)
+                if done:
+                    yield chunk(None, _finish_reason(n_out, max_tokens,
+                                                     stopped))
+                    yield "data: [DONE]\n\n"
+                    return
+        return StreamingResponse(gen(), media_type="text/event-stream")
+
+    @app.post("/v1/chat/completions")
+    async def chat_completions(req: Request):
+        body = await req.json()
+        prompt = _assemble_chat_prompt(body.get("messages", []))
+        return await _oai_generate(body, prompt, "chat")
+
+    @app.post("/v1/completions")
+    async def completions(req: Request):
+        body = await req.json()
+        prompt = body.get("prompt", "")
+        if isinstance(prompt, list):
+            prompt = "".join(str(p) for p in prompt)
+        return await _oai_generate(body, prompt, "text")
+
+    @app.get("/v1/models")
+    async def models():
+        import time as _time
+        return {"object": "list",
+                "data": [{"id": model_name, "object": "model",
+                          "created": int(_time.time()),
+                          "owned_by": "agentfield_amd"}]}
+
     return app
 
 

@@ -148,3 +148,17 @@ def test_stop_sequences(tiny_runner):
     streamed = "".join(ai("halt on demand", stream=True, stop=(stop,)))
     assert streamed == want
     assert stop not in streamed
+
+
+def test_fallback_model_chain(tiny_runner):
+    """Primary model failure falls through to the next model in
+    fallback_models (reference P4 chain)."""
+    ai = AgentAI(AIConfig(model="no-such-model", max_tokens=4, timeout=120,
+                          fallback_models=("tiny",)))
+    out = ai("fall back please")
+    assert isinstance(out, str)
+    # all models failing surfaces a collected error
+    ai_bad = AgentAI(AIConfig(model="no-such-model", max_tokens=4,
+                              fallback_models=("also-missing",)))
+    with pytest.raises(RuntimeError, match="all models failed"):
+        ai_bad("nope")

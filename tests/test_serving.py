@@ -133,3 +133,68 @@ def test_cli_init(tmp_path):
     assert (tmp_path / "myagent" / "agent.py").exists()
     text = (tmp_path / "myagent" / "agent.py").read_text()
     assert 'Agent("myagent")' in text
+
+
+def test_openai_chat_completions(replicas):
+    import httpx
+    url = replicas[0].base_url
+    r = httpx.post(url + "/v1/chat/completions", json={
+        "model": "tiny",
+        "messages": [{"role": "system", "content": "be brief"},
+                     {"role": "user", "content": "hi"}],
+        "max_tokens": 6, "temperature": 0.0}, timeout=60.0)
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert body["object"] == "chat.completion"
+    assert body["id"].startswith("chatcmpl-")
+    ch = body["choices"][0]
+    assert ch["message"]["role"] == "assistant"
+    assert isinstance(ch["message"]["content"], str)
+    assert ch["finish_reason"] in ("stop", "length")
+    u = body["usage"]
+    assert u["total_tokens"] == u["prompt_tokens"] + u["completion_tokens"]
+    assert u["completion_tokens"] == 6  # ignore_eos not set but eos unlikely
+    # model listing
+    models = httpx.get(url + "/v1/models").json()
+    assert models["data"][0]["id"] == "tiny"
+
+
+def test_openai_chat_streaming(replicas):
+    import httpx
+    import json as j
+    url = replicas[0].base_url
+    chunks, done_seen = [], False
+    with httpx.stream("POST", url + "/v1/chat/completions", json={
+            "model": "tiny", "messages": [{"role": "user", "content": "go"}],
+            "max_tokens": 5, "temperature": 0.0, "stream": True},
+            timeout=60.0) as resp:
+        for line in resp.iter_lines():
+            if not line.startswith("data:"):
+                continue
+            payload = line[5:].strip()
+            if payload == "[DONE]":
+                done_seen = True
+                break
+            chunks.append(j.loads(payload))
+    assert done_seen
+    assert chunks[0]["object"] == "chat.completion.chunk"
+    assert chunks[0]["choices"][0]["delta"] == {"role": "assistant"}
+    assert chunks[-1]["choices"][0]["finish_reason"] in ("stop", "length")
+    text = "".join(c["choices"][0]["delta"].get("content", "")
+                   for c in chunks)
+    # streamed text equals the blocking result (greedy)
+    blocking = httpx.post(url + "/v1/chat/completions", json={
+        "model": "tiny", "messages": [{"role": "user", "content": "go"}],
+        "max_tokens": 5, "temperature": 0.0}, timeout=60.0).json()
+    assert text == blocking["choices"][0]["message"]["content"]
+
+
+def test_openai_completions_endpoint(replicas):
+    import httpx
+    url = replicas[0].base_url
+    r = httpx.post(url + "/v1/completions", json={
+        "model": "tiny", "prompt": "once upon", "max_tokens": 4,
+        "temperature": 0.0}, timeout=60.0).json()
+    assert r["object"] == "text_completion"
+    assert isinstance(r["choices"][0]["text"], str)
+    assert r["usage"]["completion_tokens"] == 4
