@@ -41,9 +41,12 @@ def _hf_layer_names(i: int) -> dict:
 
 def save_checkpoint(model: LlamaForCausalLM, path: str) -> None:
     """Save in HF-compatible safetensors (single file + config.json)."""
+    cfg = model.cfg
+    if cfg.lm_vocab_rows is not None:
+        raise ValueError("save_checkpoint expects an unsharded model; "
+                         "TP ranks hold weight slices, not the full model")
     root = Path(path)
     root.mkdir(parents=True, exist_ok=True)
-    cfg = model.cfg
     tensors = {}
     sd = {k: v for k, v in model.state_dict().items()}
     for ours, hf in _HF_MAP.items():
@@ -139,10 +142,15 @@ def load_checkpoint(model: LlamaForCausalLM, path: str,
         model.embed.copy_(rd.get(_HF_MAP["embed"]).to(dev, model.embed.dtype))
         model.final_norm.copy_(rd.get(_HF_MAP["final_norm"]).to(dev))
         lm = _HF_MAP["lm_head"]
+        vs_rows = model.lm_head.shape[0]
+        lm_sl = (slice(rank * vs_rows, (rank + 1) * vs_rows)
+                 if vs_rows != cfg_full.vocab_size else None)
         if lm not in rd.weight_map:  # tied embeddings
-            model.lm_head.copy_(model.embed)
+            src = model.embed[lm_sl] if lm_sl is not None else model.embed
+            model.lm_head.copy_(src)
         else:
-            model.lm_head.copy_(rd.get(lm).to(dev, model.lm_head.dtype))
+            model.lm_head.copy_(
+                rd.get(lm, lm_sl).to(dev, model.lm_head.dtype))
         for i in range(cfg.num_layers):
             p = f"model.layers.{i}."
             L = model.layers[i]

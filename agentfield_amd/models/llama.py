@@ -34,6 +34,10 @@ class LlamaConfig:
     max_position: int = 8192
     rms_eps: float = 1e-5
     tie_embeddings: bool = False
+    # vocab-parallel lm_head: rows this rank holds (None -> full vocab).
+    # Embedding stays replicated (lookup is local); logits shards are
+    # all-gathered along the vocab dim so sampling sees full logits.
+    lm_vocab_rows: int | None = None
 
     @property
     def q_size(self):
@@ -54,7 +58,9 @@ class LlamaConfig:
             num_kv_heads=self.num_kv_heads // tp, head_dim=self.head_dim,
             vocab_size=self.vocab_size, rope_theta=self.rope_theta,
             max_position=self.max_position, rms_eps=self.rms_eps,
-            tie_embeddings=self.tie_embeddings)
+            tie_embeddings=self.tie_embeddings,
+            lm_vocab_rows=(self.vocab_size // tp
+                           if self.vocab_size % tp == 0 else None))
 
 
 CONFIGS = {
@@ -226,7 +232,8 @@ class LlamaForCausalLM(nn.Module):
             self.layers = nn.ModuleList(
                 [LlamaLayer(cfg, i) for i in range(cfg.num_layers)])
             self.final_norm = nn.Parameter(torch.empty(cfg.hidden_size))
-            self.lm_head = nn.Parameter(torch.empty(cfg.vocab_size, cfg.hidden_size))
+            self.lm_head = nn.Parameter(torch.empty(
+                cfg.lm_vocab_rows or cfg.vocab_size, cfg.hidden_size))
         self.to(dtype=dtype)
         self.register_buffer(
             "rope_tab",
@@ -293,7 +300,7 @@ class LlamaForCausalLM(nn.Module):
                 residual, ss = layer.forward_decode_fused(
                     residual, ss, ss2, positions, self.rope_tab, kv, md)
             h = ops.rmsnorm(residual, self.final_norm, self.cfg.rms_eps)
-            return ops.linear(h, self.lm_head)
+            return self._project_logits(h)
         h = ops.embedding(ids, self.embed)
         residual = None
         for layer in self.layers:
@@ -301,4 +308,19 @@ class LlamaForCausalLM(nn.Module):
         h, _ = ops.rmsnorm(h, self.final_norm, self.cfg.rms_eps, residual)
         if logit_rows is not None:
             h = ops.gather_rows(h, logit_rows)
-        return ops.linear(h, self.lm_head)
+        return self._project_logits(h)
+
+    def _project_logits(self, h: torch.Tensor) -> torch.Tensor:
+        """lm_head projection.  Under vocab-parallel TP each rank computes
+        its [T, V/tp] shard and the shards are all-gathered along the vocab
+        dim (rank r owns global vocab rows [r*V/tp, (r+1)*V/tp)), so every
+        rank samples from identical full logits."""
+        logits = ops.linear(h, self.lm_head)
+        if getattr(self, "_tp_vocab_parallel", False):
+            import torch.distributed as dist
+            group = getattr(self, "tp_logits_group", None)
+            world = dist.get_world_size(group)
+            shards = [torch.empty_like(logits) for _ in range(world)]
+            dist.all_gather(shards, logits.contiguous(), group=group)
+            logits = torch.cat(shards, dim=-1)
+        return logits

@@ -9,8 +9,12 @@ xGMI's 7 p2p links):
   * gate_up: column-parallel (gate and up halves sharded separately so the
             fused silu_mul kernel sees a local [gate|up] layout)
   * down:   row-parallel    — partial sums all-reduced
-  * embed / lm_head: replicated (~1 GB bf16 for the 128k vocab; vocab-
-            parallel logits are a later optimization)
+  * embed:  replicated (the lookup is local and cheap)
+  * lm_head: vocab-parallel when vocab % tp == 0 — rank r holds global
+            vocab rows [r*V/tp, (r+1)*V/tp), computes its logits shard and
+            all-gathers along the vocab dim (saves the ~1-2 GB replicated
+            head AND tp x the lm_head GEMM FLOPs; sampling still sees full
+            logits on every rank, so scheduling stays identical-decision)
 
 Weights are deterministic per parameter NAME, so every rank can materialize
 the full tensor (fp32, one at a time), slice its shard and free — no rank
@@ -52,6 +56,7 @@ def shard_llama_weights(model: LlamaForCausalLM, full_cfg: LlamaConfig,
         for name, p in model.named_parameters():
             if name == "embed" or name == "lm_head":
                 full_shape = (cfg.vocab_size, cfg.hidden_size)
+                vs_rows = p.shape[0]  # lm_head may be vocab-sharded
             elif name.endswith("attn.qkv"):
                 full_shape = (cfg.q_size + 2 * cfg.kv_size, cfg.hidden_size)
             elif name.endswith("attn.o"):
@@ -77,6 +82,8 @@ def shard_llama_weights(model: LlamaForCausalLM, full_cfg: LlamaConfig,
                                    up[rank * inter:(rank + 1) * inter]], dim=0)
             elif name.endswith("mlp.down"):
                 shard = full[:, rank * inter:(rank + 1) * inter]
+            elif name == "lm_head" and p.shape[0] != cfg.vocab_size:
+                shard = full[rank * vs_rows:(rank + 1) * vs_rows]
             else:
                 shard = full
             p.copy_(shard.to(p.dtype))
@@ -107,6 +114,9 @@ def build_tp_model(full_cfg: LlamaConfig, tp: int, rank: int, device,
             layer.attn.register_forward_hook(hook)
             layer.mlp.register_forward_hook(hook)
         model.no_fused_decode = True  # collectives attach to module forward
+        if shard_cfg.lm_vocab_rows is not None:
+            model._tp_vocab_parallel = True
+            model.tp_logits_group = group
     return model
 
 

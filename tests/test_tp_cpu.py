@@ -65,7 +65,11 @@ def _logits_case(rank, world):
                       seq_lens=[T],
                       q_start=torch.zeros(1, dtype=torch.int32),
                       block_table=torch.arange(32, dtype=torch.int32)[None, :])
+    # vocab-parallel lm_head must actually be sharded...
+    assert model.lm_head.shape[0] == TP_CFG.vocab_size // world
     logits = model(ids, pos, kv, md)
+    # ...while gathered logits cover the full vocab
+    assert logits.shape[-1] == TP_CFG.vocab_size
 
     # reference: full (TP=1) model, same deterministic weights
     full = build_tp_model(TP_CFG, 1, 0, dev, dtype=torch.float32, base_seed=7)
