@@ -145,6 +145,12 @@ def test_checkpoint_tp_slices(tmp_path):
     # rank 1's q rows are full rows [2*D : 4*D)
     want_q = full.layers[0].attn.qkv[2 * D:4 * D]
     assert torch.equal(shard.layers[0].attn.qkv[:2 * D], want_q)
+    # vocab-parallel lm_head: rank 1 holds global vocab rows [128:256)
+    assert shard.lm_head.shape[0] == cfg.vocab_size // 2
+    assert torch.equal(shard.lm_head, full.lm_head[128:256])
+    # sharded models refuse full-model save
+    with pytest.raises(ValueError, match="unsharded"):
+        save_checkpoint(shard, str(tmp_path / "c2"))
 
 
 # ------------------------------------------------------------ resilience
