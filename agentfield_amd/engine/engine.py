@@ -43,7 +43,7 @@ class LLMEngine:
                  max_num_seqs: int = 64, max_prefill_tokens: int = 8192,
                  max_waiting: int = 4096, enable_graphs: bool = True, eos_id: int = 2, seed: int = 0,
                  model: LlamaForCausalLM | None = None,
-                 tp_group=None):
+                 tp_group=None, spec_lookup: int = 0):
         self.cfg = cfg
         self.device = torch.device(device)
         self.is_gpu = self.device.type == "cuda"
@@ -78,7 +78,13 @@ class LLMEngine:
         self.enable_graphs = enable_graphs and self.is_gpu
         self._graphs: dict[int, dict] = {}
         self.metrics = {"prefill_tokens": 0, "decode_tokens": 0, "steps": 0,
-                        "prefill_steps": 0, "decode_steps": 0}
+                        "prefill_steps": 0, "decode_steps": 0,
+                        "spec_steps": 0, "spec_drafted": 0, "spec_accepted": 0}
+        # prompt-lookup speculative decoding (greedy-exact, opt-in):
+        # draft up to spec_lookup tokens from n-gram matches in the
+        # sequence's own context and verify them in ONE chunked-prefill
+        # forward; every emitted token is the model's true greedy token.
+        self.spec_lookup = int(spec_lookup)
         # static decode buffers (shared across graph buckets; sized to max)
         B = max_num_seqs
         dev = self.device
@@ -163,6 +169,15 @@ class LLMEngine:
                 self._prefilling.extend(batch.seqs)
             else:
                 self.metrics["steps"] += 1
+                if self.spec_lookup > 0 and self.tp_group is None and \
+                        all(s.sampling.temperature == 0.0 and
+                            s.sampling.top_k <= 0 and s.sampling.top_p >= 1.0
+                            for s in batch.seqs):
+                    drafts = {s.seq_id: self._draft_for(s)
+                              for s in batch.seqs}
+                    if any(drafts.values()):
+                        self.metrics["spec_steps"] += 1
+                        return self._step_verify(batch, drafts)
                 self.metrics["decode_steps"] += 1
                 tokens = self._step_decode(batch)
                 return self._bookkeep(batch.seqs, tokens)
@@ -256,6 +271,99 @@ class LLMEngine:
                                        dtype=torch.float32, device=dev)}
         toks = ops.sample(logits, temps, self.sampler, **kw)
         return done, toks.cpu().tolist()
+
+    # -- speculative decode (prompt lookup, greedy-exact) -------------------
+    def _draft_for(self, seq: Sequence) -> list[int]:
+        """Propose a continuation from the most recent earlier occurrence of
+        the context's tail bigram.  Capped to (a) remaining generation
+        budget minus the bonus token and (b) KV capacity already allocated
+        by the scheduler — so verification never needs new pages."""
+        k = self.spec_lookup
+        n = seq.num_tokens
+        remaining = len(seq.prompt_ids) + seq.sampling.max_tokens - n
+        capacity = len(seq.pages) * self.page_size - n
+        k = min(k, remaining - 1, capacity)
+        if k <= 0:
+            return []
+        ctx = seq.prompt_ids + seq.output_ids
+        g = 2
+        if len(ctx) <= g:
+            return []
+        tail = ctx[-g:]
+        for i in range(len(ctx) - g - 1, -1, -1):
+            if ctx[i:i + g] == tail:
+                return ctx[i + g:i + g + k]
+        return []
+
+    def _step_verify(self, batch: ScheduleBatch, drafts: dict):
+        """One chunked-prefill forward verifies each sequence's draft: chunk
+        = [last_token] + draft at positions n-1..n-1+k.  Row j's argmax is
+        the model's true greedy token after consuming draft[:j]; the longest
+        matching prefix is accepted plus one bonus token.  Rejected draft
+        positions leave garbage KV beyond the sequence length, which is
+        overwritten before it can ever be read (attention is length-bounded)."""
+        dev = self.device
+        seqs = batch.seqs
+        ids, pos, slots, q_start, bt_rows, chunks = [], [], [], [], [], []
+        for seq in seqs:
+            d = drafts.get(seq.seq_id) or []
+            n = seq.num_tokens
+            chunk = [seq.last_token] + d
+            ids.extend(chunk)
+            pos.extend(range(n - 1, n - 1 + len(chunk)))
+            slots.extend(self._slot(seq, i)
+                         for i in range(n - 1, n - 1 + len(chunk)))
+            q_start.append(n - 1)
+            chunks.append(len(chunk))
+            row = torch.zeros(self.max_pages_per_seq, dtype=torch.int32)
+            row[:len(seq.pages)] = torch.tensor(seq.pages, dtype=torch.int32)
+            bt_rows.append(row)
+            self.metrics["spec_drafted"] += len(d)
+        cu_list = [0]
+        for ln in chunks:
+            cu_list.append(cu_list[-1] + ln)
+        md = AttnMetadata(
+            is_prefill=True,
+            slots=torch.tensor(slots, dtype=torch.int64, device=dev),
+            cu_seqlens=torch.tensor(cu_list, dtype=torch.int32, device=dev),
+            seq_lens=chunks,
+            q_start=torch.tensor(q_start, dtype=torch.int32, device=dev),
+            block_table=torch.stack(bt_rows).to(dev))
+        logits = self.model(torch.tensor(ids, dtype=torch.int32, device=dev),
+                            torch.tensor(pos, dtype=torch.int32, device=dev),
+                            self.kv, md)
+        greedy = logits.argmax(dim=-1).to(torch.int64).cpu().tolist()
+        tok_lists = []
+        for i, seq in enumerate(seqs):
+            t = greedy[cu_list[i]:cu_list[i + 1]]
+            d = drafts.get(seq.seq_id) or []
+            a = 0
+            while a < len(d) and t[a] == d[a]:
+                a += 1
+            tok_lists.append(t[:a + 1])  # accepted prefix + bonus token
+            self.metrics["spec_accepted"] += a
+            self.metrics["decode_tokens"] += a + 1
+        return self._bookkeep_multi(seqs, tok_lists)
+
+    def _bookkeep_multi(self, seqs, tok_lists):
+        events = []
+        now = time.monotonic_ns()
+        for seq, toks in zip(seqs, tok_lists):
+            for tok in toks:
+                if not seq.output_ids:
+                    seq.first_token_ns = now
+                self.sched.note_token(seq)
+                done = seq.append(tok, self.eos_id)
+                if done:
+                    seq.finish_ns = now
+                    self.sched.finish(seq)
+                    self._finished[seq.seq_id] = seq
+                if seq.on_token is not None:
+                    seq.on_token(tok, done)
+                events.append((seq.seq_id, tok, done))
+                if done:
+                    break
+        return events
 
     # -- decode path (graph-captured on GPU) --------------------------------
     def _fill_decode_buffers(self, seqs: list[Sequence], bs: int) -> None:

@@ -280,6 +280,9 @@ def main():
     ap.add_argument("--port", type=int, default=8710)
     ap.add_argument("--max-num-seqs", type=int, default=256)
     ap.add_argument("--no-graphs", action="store_true")
+    ap.add_argument("--spec-lookup", type=int, default=0,
+                    help="prompt-lookup speculative decoding draft length "
+                         "(greedy-exact; 0 disables)")
     args = ap.parse_args()
 
     device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
@@ -292,7 +295,7 @@ def main():
     eng = LLMEngine(cfg, device=device,
                     max_num_seqs=kw.pop("max_num_seqs", args.max_num_seqs),
                     enable_graphs=not args.no_graphs and device.startswith("cuda"),
-                    **kw)
+                    spec_lookup=args.spec_lookup, **kw)
     runner = EngineRunner(eng, load_tokenizer())
     app = create_engine_app(runner, args.model)
     import uvicorn
