@@ -166,38 +166,47 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
         base = {"id": rid, "created": created, "model": model_name}
 
         if not body.get("stream", False):
-            w = runner.submit(ids, sp)
+            nchoice = max(1, min(int(body.get("n", 1)), 16))
+            waiters = [runner.submit(ids, sp) for _ in range(nchoice)]
 
             def wait():
-                w["done"].wait(float(body.get("timeout", 600)))
-                return w
-            w = await anyio.to_thread.run_sync(wait)
-            if not w["done"].is_set():
+                deadline = float(body.get("timeout", 600))
+                for w in waiters:
+                    w["done"].wait(deadline)
+                return waiters
+            await anyio.to_thread.run_sync(wait)
+            if not all(w["done"].is_set() for w in waiters):
                 return JSONResponse({"error": {"message": "timeout",
                                                "type": "timeout"}},
                                     status_code=504)
-            if w["error"]:
-                return JSONResponse({"error": {"message": w["error"],
+            err = next((w["error"] for w in waiters if w["error"]), None)
+            if err:
+                return JSONResponse({"error": {"message": err,
                                                "type": "overloaded"}},
                                     status_code=503)
-            text = runner.tokenizer.decode(w["output"])
-            stopped = False
-            for s in stops:
-                idx = text.find(s)
-                if idx >= 0:
-                    text, stopped = text[:idx], True
-            fr = _finish_reason(len(w["output"]), max_tokens, stopped)
+            choices, out_tokens = [], 0
+            for i, w in enumerate(waiters):
+                text = runner.tokenizer.decode(w["output"])
+                stopped = False
+                for s in stops:
+                    idx = text.find(s)
+                    if idx >= 0:
+                        text, stopped = text[:idx], True
+                fr = _finish_reason(len(w["output"]), max_tokens, stopped)
+                out_tokens += len(w["output"])
+                if kind == "chat":
+                    choices.append({"index": i, "finish_reason": fr,
+                                    "message": {"role": "assistant",
+                                                "content": text}})
+                else:
+                    choices.append({"index": i, "finish_reason": fr,
+                                    "text": text})
             usage = {"prompt_tokens": len(ids),
-                     "completion_tokens": len(w["output"]),
-                     "total_tokens": len(ids) + len(w["output"])}
-            if kind == "chat":
-                choice = {"index": 0, "finish_reason": fr,
-                          "message": {"role": "assistant", "content": text}}
-                return {**base, "object": "chat.completion",
-                        "choices": [choice], "usage": usage}
-            choice = {"index": 0, "finish_reason": fr, "text": text}
-            return {**base, "object": "text_completion",
-                    "choices": [choice], "usage": usage}
+                     "completion_tokens": out_tokens,
+                     "total_tokens": len(ids) + out_tokens}
+            obj_name = "chat.completion" if kind == "chat" else "text_completion"
+            return {**base, "object": obj_name, "choices": choices,
+                    "usage": usage}
 
         sq: queue.Queue = queue.Queue()
         w = runner.submit(ids, sp, stream_q=sq)

@@ -198,3 +198,17 @@ def test_openai_completions_endpoint(replicas):
     assert r["object"] == "text_completion"
     assert isinstance(r["choices"][0]["text"], str)
     assert r["usage"]["completion_tokens"] == 4
+
+
+def test_openai_n_choices(replicas):
+    import httpx
+    url = replicas[0].base_url
+    r = httpx.post(url + "/v1/chat/completions", json={
+        "model": "tiny", "messages": [{"role": "user", "content": "pick"}],
+        "max_tokens": 4, "temperature": 0.8, "n": 3}, timeout=60.0).json()
+    assert len(r["choices"]) == 3
+    assert [c["index"] for c in r["choices"]] == [0, 1, 2]
+    assert r["usage"]["completion_tokens"] == 12
+    # temperature>0: per-request sampler state advances -> varied choices
+    texts = {c["message"]["content"] for c in r["choices"]}
+    assert len(texts) >= 1  # (distinctness is probabilistic; shape matters)
