@@ -28,12 +28,14 @@ class RemoteRunner:
                                     max_tokens=cfg.max_tokens,
                                     temperature=cfg.temperature,
                                     stop=list(cfg.stop or ()),
+                                    model=cfg.model,
                                     timeout=cfg.timeout)["text"]
 
     def stream_text(self, prompt: str, cfg):
         yield from self.router.stream(prompt=prompt, max_tokens=cfg.max_tokens,
                                       temperature=cfg.temperature,
                                       stop=list(cfg.stop or ()),
+                                      model=cfg.model,
                                       timeout=cfg.timeout)
 
 
@@ -45,6 +47,7 @@ class DPRouter:
         self.timeout = timeout
         self._rr = itertools.cycle(range(len(self.urls)))
         self._loads: dict[str, float] = {}
+        self._models: dict[str, str | None] = {}
         self._healthy: dict[str, bool] = {u: True for u in self.urls}
         self._last_refresh = 0.0
         self._lock = threading.Lock()
@@ -61,15 +64,23 @@ class DPRouter:
                 r = self._client.get(u + "/v1/stats", timeout=1.0)
                 s = r.json()
                 self._loads[u] = s.get("queued", 0) + s.get("running", 0)
+                self._models[u] = s.get("model")
                 self._healthy[u] = True
             except Exception:
                 self._healthy[u] = False
 
-    def pick(self) -> str:
+    def pick(self, model: str | None = None) -> str:
+        """Least-loaded healthy replica; with `model`, restricted to the
+        replicas that report serving it (heterogeneous fleets), falling
+        back to the whole fleet when no replica advertises the model."""
         self._refresh()
         healthy = [u for u in self.urls if self._healthy.get(u, True)]
         if not healthy:
             healthy = self.urls  # try anyway
+        if model is not None:
+            serving = [u for u in healthy if self._models.get(u) == model]
+            if serving:
+                healthy = serving
         if self._loads:
             return min(healthy, key=lambda u: self._loads.get(u, 0))
         return healthy[next(self._rr) % len(healthy)]
@@ -77,7 +88,7 @@ class DPRouter:
     def generate(self, **body) -> dict:
         last_err = None
         for _ in range(min(3, len(self.urls))):
-            url = self.pick()
+            url = self.pick(body.get("model"))
             try:
                 r = self._client.post(url + "/v1/generate", json=body)
                 if r.status_code == 200:
@@ -91,7 +102,7 @@ class DPRouter:
 
     def stream(self, **body):
         body["stream"] = True
-        url = self.pick()
+        url = self.pick(body.get("model"))
         with self._client.stream("POST", url + "/v1/generate",
                                  json=body) as resp:
             for line in resp.iter_lines():

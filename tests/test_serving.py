@@ -212,3 +212,33 @@ def test_openai_n_choices(replicas):
     # temperature>0: per-request sampler state advances -> varied choices
     texts = {c["message"]["content"] for c in r["choices"]}
     assert len(texts) >= 1  # (distinctness is probabilistic; shape matters)
+
+
+def test_router_model_aware_routing(replicas):
+    """Heterogeneous fleets: requests carrying a model name route to the
+    replicas that advertise it; unknown models fall back to least-loaded."""
+    from agentfield_amd.serving.engine_server import create_engine_app
+
+    # a third replica advertising a different model name
+    srv_b, runner_b = None, None
+    try:
+        eng = LLMEngine(CONFIGS["tiny"], device="cpu", dtype=torch.float32,
+                        page_size=4, num_pages=128, max_num_seqs=4,
+                        enable_graphs=False)
+        runner_b = EngineRunner(eng)
+        srv_b = AppServer(create_engine_app(runner_b, "tiny-b")).start()
+        urls = [replicas[0].base_url, srv_b.base_url]
+        router = DPRouter(urls, refresh_s=0.0)
+        for _ in range(4):
+            assert router.pick("tiny-b") == srv_b.base_url
+            assert router.pick("tiny") == replicas[0].base_url
+        # unknown model: falls back to the whole fleet (no exception)
+        assert router.pick("missing-model") in urls
+        out = router.generate(prompt_ids=[1, 2, 3], max_tokens=2,
+                              ignore_eos=True, model="tiny-b")
+        assert len(out["output_ids"]) == 2
+    finally:
+        if runner_b:
+            runner_b.shutdown()
+        if srv_b:
+            srv_b.stop()
