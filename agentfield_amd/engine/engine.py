@@ -19,6 +19,7 @@ from .scheduler import ScheduleBatch, SchedulerConfig, make_scheduler
 from .sequence import SamplingParams, Sequence, SeqStatus
 
 DECODE_BUCKETS = (1, 2, 4, 8, 16, 32, 64, 128, 256)
+LP_TOPN = 8  # top-N alternatives reported per token when logprobs requested
 
 
 def _bucket_for(n: int, max_bs: int) -> int:
@@ -98,6 +99,10 @@ class LLMEngine:
             "topk": torch.zeros(B, dtype=torch.int32, device=dev),
             "topp": torch.ones(B, dtype=torch.float32, device=dev),
             "tokens": torch.zeros(B, dtype=torch.int32, device=dev),
+            # logprob reporting (filled only in lp-keyed graphs)
+            "lp_tok": torch.zeros(B, dtype=torch.float32, device=dev),
+            "lp_vals": torch.zeros(B, LP_TOPN, dtype=torch.float32, device=dev),
+            "lp_ids": torch.zeros(B, LP_TOPN, dtype=torch.int32, device=dev),
         }
         self._host = {k: torch.zeros_like(v, device="cpu").pin_memory()
                       if self.is_gpu else torch.zeros_like(v)
@@ -270,7 +275,29 @@ class LLMEngine:
                   "topp": torch.tensor([s.sampling.top_p for s in done],
                                        dtype=torch.float32, device=dev)}
         toks = ops.sample(logits, temps, self.sampler, **kw)
+        if any(s.sampling.logprobs > 0 for s in done):
+            self._attach_row_logprobs(done, logits, toks)
         return done, toks.cpu().tolist()
+
+    def _attach_row_logprobs(self, seqs, logits, toks) -> None:
+        """Per-row logprob reporting for eager paths (prefill-final and
+        speculative verify): rows of `logits` align with `seqs`/`toks`."""
+        lf = logits.float()
+        lse = torch.logsumexp(lf, dim=-1)
+        chosen = (lf.gather(1, toks.long().unsqueeze(1)).squeeze(1) -
+                  lse).cpu().tolist()
+        tv, ti = lf.topk(LP_TOPN, dim=-1)
+        tv = (tv - lse.unsqueeze(1)).cpu().tolist()
+        ti = ti.cpu().tolist()
+        for i, seq in enumerate(seqs):
+            nreq = seq.sampling.logprobs
+            if nreq <= 0:
+                continue
+            if seq.logprobs is None:
+                seq.logprobs = []
+            seq.logprobs.append({"logprob": chosen[i],
+                                 "top": list(zip(ti[i][:nreq],
+                                                 tv[i][:nreq]))})
 
     # -- speculative decode (prompt lookup, greedy-exact) -------------------
     def _draft_for(self, seq: Sequence) -> list[int]:
@@ -332,8 +359,9 @@ class LLMEngine:
         logits = self.model(torch.tensor(ids, dtype=torch.int32, device=dev),
                             torch.tensor(pos, dtype=torch.int32, device=dev),
                             self.kv, md)
-        greedy = logits.argmax(dim=-1).to(torch.int64).cpu().tolist()
-        tok_lists = []
+        greedy_t = logits.argmax(dim=-1).to(torch.int64)
+        greedy = greedy_t.cpu().tolist()
+        tok_lists, lp_rows, lp_seqs = [], [], []
         for i, seq in enumerate(seqs):
             t = greedy[cu_list[i]:cu_list[i + 1]]
             d = drafts.get(seq.seq_id) or []
@@ -343,6 +371,14 @@ class LLMEngine:
             tok_lists.append(t[:a + 1])  # accepted prefix + bonus token
             self.metrics["spec_accepted"] += a
             self.metrics["decode_tokens"] += a + 1
+            if seq.sampling.logprobs > 0:
+                for j in range(a + 1):  # one helper row per emitted token
+                    lp_rows.append(cu_list[i] + j)
+                    lp_seqs.append(seq)
+        if lp_rows:
+            rows = torch.tensor(lp_rows, dtype=torch.int64,
+                                device=logits.device)
+            self._attach_row_logprobs(lp_seqs, logits[rows], greedy_t[rows])
         return self._bookkeep_multi(seqs, tok_lists)
 
     def _bookkeep_multi(self, seqs, tok_lists):
@@ -363,6 +399,9 @@ class LLMEngine:
                 events.append((seq.seq_id, tok, done))
                 if done:
                     break
+            if seq.logprobs is not None:
+                # eos mid-acceptance: drop logprobs for discarded draft rows
+                del seq.logprobs[len(seq.output_ids):]
         return events
 
     # -- decode path (graph-captured on GPU) --------------------------------
@@ -394,7 +433,8 @@ class LLMEngine:
             d[k][:bs].copy_(h[k][:bs], non_blocking=nb)
         d["bt"][:bs].copy_(h["bt"][:bs], non_blocking=nb)
 
-    def _decode_forward(self, bs: int, nsplit: int, scratch, tkp: bool):
+    def _decode_forward(self, bs: int, nsplit: int, scratch, tkp: bool,
+                        lp: bool = False):
         d = self._dec
         md = AttnMetadata(is_prefill=False, slots=d["slots"][:bs],
                           block_table=d["bt"][:bs], seq_lens_t=d["lens"][:bs],
@@ -403,6 +443,14 @@ class LLMEngine:
         kw = {"topk": d["topk"][:bs], "topp": d["topp"][:bs]} if tkp else {}
         ops.sample(logits, d["temps"][:bs], self.sampler,
                    out=d["tokens"][:bs], **kw)
+        if lp:  # graph-capturable logprob reporting
+            lf = logits.float()
+            lse = torch.logsumexp(lf, dim=-1)
+            d["lp_tok"][:bs] = lf.gather(
+                1, d["tokens"][:bs].long().unsqueeze(1)).squeeze(1) - lse
+            tv, ti = lf.topk(LP_TOPN, dim=-1)
+            d["lp_vals"][:bs] = tv - lse.unsqueeze(1)
+            d["lp_ids"][:bs] = ti.to(torch.int32)
 
     def _make_scratch(self, bs: int, nsplit: int):
         if nsplit <= 1:
@@ -415,8 +463,8 @@ class LLMEngine:
                           dtype=torch.float32, device=self.device)
         return po, pml
 
-    def _get_graph(self, bs: int, tkp: bool):
-        key = (bs, tkp)
+    def _get_graph(self, bs: int, tkp: bool, lp: bool = False):
+        key = (bs, tkp, lp)
         entry = self._graphs.get(key)
         if entry is not None:
             return entry
@@ -427,14 +475,35 @@ class LLMEngine:
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(2):
-                self._decode_forward(bs, nsplit, scratch, tkp)
+                self._decode_forward(bs, nsplit, scratch, tkp, lp)
         torch.cuda.current_stream().wait_stream(s)
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
-            self._decode_forward(bs, nsplit, scratch, tkp)
+            self._decode_forward(bs, nsplit, scratch, tkp, lp)
         entry = {"graph": g, "nsplit": nsplit, "scratch": scratch}
         self._graphs[key] = entry
         return entry
+
+    def _attach_decode_logprobs(self, seqs: list[Sequence], n: int) -> None:
+        d, h = self._dec, self._host
+        if self.is_gpu:
+            for k in ("lp_tok", "lp_vals", "lp_ids"):
+                h[k][:n].copy_(d[k][:n], non_blocking=True)
+            torch.cuda.current_stream().synchronize()
+            src = h
+        else:
+            src = d
+        vals, ids_, tokl = (src["lp_vals"][:n].tolist(),
+                            src["lp_ids"][:n].tolist(),
+                            src["lp_tok"][:n].tolist())
+        for i, seq in enumerate(seqs):
+            nreq = seq.sampling.logprobs
+            if nreq <= 0:
+                continue
+            if seq.logprobs is None:
+                seq.logprobs = []
+            top = list(zip(ids_[i][:nreq], vals[i][:nreq]))
+            seq.logprobs.append({"logprob": tokl[i], "top": top})
 
     def _step_decode(self, batch: ScheduleBatch) -> list[int]:
         seqs = batch.seqs
@@ -442,12 +511,16 @@ class LLMEngine:
         self.metrics["decode_tokens"] += n
         bs = _bucket_for(n, self.max_num_seqs)
         tkp = any(s.sampling.top_k > 0 or s.sampling.top_p < 1.0 for s in seqs)
+        lp = any(s.sampling.logprobs > 0 for s in seqs)
         self._fill_decode_buffers(seqs, bs)
         if self.enable_graphs:
-            self._get_graph(bs, tkp)["graph"].replay()
+            self._get_graph(bs, tkp, lp)["graph"].replay()
         else:
             nsplit = choose_nsplit(bs, self.cfg.num_kv_heads) if self.is_gpu else 1
-            self._decode_forward(bs, nsplit, self._make_scratch(bs, nsplit), tkp)
+            self._decode_forward(bs, nsplit, self._make_scratch(bs, nsplit),
+                                 tkp, lp)
+        if lp:
+            self._attach_decode_logprobs(seqs, n)
         toks = self._dec["tokens"][:n]
         if self.is_gpu:
             self._host["tokens"][:n].copy_(toks, non_blocking=True)

@@ -14,6 +14,8 @@ class SamplingParams:
     top_p: float = 1.0      # >= 1 = disabled
     stop_token_ids: tuple = ()
     ignore_eos: bool = False
+    logprobs: int = 0       # >0: report chosen-token logprob + top-N
+                            # alternatives per emitted token (N <= 8)
 
 
 class SeqStatus(enum.Enum):
@@ -34,6 +36,9 @@ class Sequence:
     output_ids: list[int] = field(default_factory=list)
     pages: list[int] = field(default_factory=list)
     finish_reason: str | None = None
+    # per output token, when sampling.logprobs > 0:
+    # {"logprob": float, "top": [(token_id, logprob), ...]}
+    logprobs: list | None = None
     on_token: Callable | None = None     # streaming callback (token_id, done)
     arrival_ns: int = 0
     first_token_ns: int = 0

@@ -208,6 +208,7 @@ class EngineRunner:
                     if fin is not None:
                         w = pending.pop(rid)
                         w["output"] = fin.output_ids
+                        w["logprobs"] = fin.logprobs
                         w["done"].set()
             if not moved:
                 self._wake.wait(0.005)

@@ -66,7 +66,8 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
         sp = SamplingParams(
             max_tokens=int(body.get("max_tokens", 128)),
             temperature=float(body.get("temperature", 0.0)),
-            ignore_eos=bool(body.get("ignore_eos", False)))
+            ignore_eos=bool(body.get("ignore_eos", False)),
+            logprobs=min(int(body.get("logprobs", 0) or 0), 8))
         if "prompt_ids" in body:
             ids = [int(x) for x in body["prompt_ids"]]
         else:
@@ -90,7 +91,13 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
                 idx = text.find(s)
                 if idx >= 0:
                     text = text[:idx]
-            return {"output_ids": w["output"], "text": text}
+            out = {"output_ids": w["output"], "text": text}
+            if w.get("logprobs") is not None:
+                out["logprobs"] = [
+                    {"logprob": e["logprob"],
+                     "top": [[int(t), float(v)] for t, v in e["top"]]}
+                    for e in w["logprobs"]]
+            return out
 
         sq: queue.Queue = queue.Queue()
         w = runner.submit(ids, sp, stream_q=sq)
@@ -155,8 +162,15 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
 
         import anyio
         max_tokens = int(body.get("max_tokens", 128))
+        # chat: logprobs=true + top_logprobs=N; completions: logprobs=N
+        if kind == "chat":
+            lp_n = (min(int(body.get("top_logprobs", 1) or 1), 8)
+                    if body.get("logprobs") else 0)
+        else:
+            lp_n = min(int(body.get("logprobs", 0) or 0), 8)
         sp = SamplingParams(max_tokens=max_tokens,
-                            temperature=float(body.get("temperature", 0.0)))
+                            temperature=float(body.get("temperature", 0.0)),
+                            logprobs=lp_n)
         stop_in = body.get("stop") or []
         stops = tuple(s for s in ([stop_in] if isinstance(stop_in, str)
                                   else stop_in) if s)
@@ -195,12 +209,31 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
                 fr = _finish_reason(len(w["output"]), max_tokens, stopped)
                 out_tokens += len(w["output"])
                 if kind == "chat":
-                    choices.append({"index": i, "finish_reason": fr,
-                                    "message": {"role": "assistant",
-                                                "content": text}})
+                    choice = {"index": i, "finish_reason": fr,
+                              "message": {"role": "assistant",
+                                          "content": text}}
                 else:
-                    choices.append({"index": i, "finish_reason": fr,
-                                    "text": text})
+                    choice = {"index": i, "finish_reason": fr, "text": text}
+                if lp_n and w.get("logprobs") is not None:
+                    dec = runner.tokenizer.decode
+                    if kind == "chat":
+                        choice["logprobs"] = {"content": [
+                            {"token": dec([tok]), "logprob": e["logprob"],
+                             "top_logprobs": [
+                                 {"token": dec([int(t)]),
+                                  "logprob": float(v)}
+                                 for t, v in e["top"]]}
+                            for tok, e in zip(w["output"], w["logprobs"])]}
+                    else:
+                        choice["logprobs"] = {
+                            "tokens": [dec([t]) for t in w["output"]],
+                            "token_logprobs": [e["logprob"]
+                                               for e in w["logprobs"]],
+                            "top_logprobs": [
+                                {dec([int(t)]): float(v)
+                                 for t, v in e["top"]}
+                                for e in w["logprobs"]]}
+                choices.append(choice)
             usage = {"prompt_tokens": len(ids),
                      "completion_tokens": out_tokens,
                      "total_tokens": len(ids) + out_tokens}

@@ -166,3 +166,52 @@ def test_spec_respects_max_tokens_and_eos():
     base = make_engine().generate(prompts, SamplingParams(max_tokens=3,
                                                           ignore_eos=True))
     assert outs == base
+
+
+# ----------------------------------------------------------- logprobs
+def test_logprobs_reporting():
+    """Chosen-token logprob matches log_softmax of an equivalent forward;
+    greedy chosen token is the top-1 alternative; list aligns 1:1."""
+    eng = make_engine()
+    rid = eng.add_request([1, 5, 9, 20],
+                          SamplingParams(max_tokens=5, ignore_eos=True,
+                                         logprobs=3))
+    fin = None
+    for _ in range(50):
+        eng.step()
+        fin = fin or eng.get_finished(rid)
+        if fin:
+            break
+    assert fin is not None and fin.logprobs is not None
+    assert len(fin.logprobs) == len(fin.output_ids) == 5
+    for tok, e in zip(fin.output_ids, fin.logprobs):
+        assert e["logprob"] <= 0.0
+        assert len(e["top"]) == 3
+        # greedy: chosen == top-1 and logprob equal
+        assert e["top"][0][0] == tok
+        assert abs(e["top"][0][1] - e["logprob"]) < 1e-5
+    # normalization: top-3 mass < 1
+    import math
+    assert sum(math.exp(v) for _, v in fin.logprobs[0]["top"]) < 1.0 + 1e-6
+
+
+def test_logprobs_with_spec_lookup():
+    """Logprob lists stay 1:1 with output tokens through the speculative
+    verify path (including eos/max_tokens truncation)."""
+    prompts = [[7, 8, 9, 7, 8, 9, 7, 8, 9, 7, 8]]
+    sp = SamplingParams(max_tokens=8, ignore_eos=True, logprobs=2)
+    eng = make_engine(spec_lookup=4)
+    rid = eng.add_request(prompts[0], sp)
+    fin = None
+    for _ in range(80):
+        eng.step()
+        fin = fin or eng.get_finished(rid)
+        if fin:
+            break
+    assert fin is not None
+    assert len(fin.logprobs) == len(fin.output_ids) == 8
+    assert all(e["top"][0][0] == t
+               for t, e in zip(fin.output_ids, fin.logprobs))
+    # same tokens as the non-speculative engine
+    base = make_engine().generate(prompts, sp)
+    assert fin.output_ids == base[0]

@@ -242,3 +242,32 @@ def test_router_model_aware_routing(replicas):
             runner_b.shutdown()
         if srv_b:
             srv_b.stop()
+
+
+def test_openai_logprobs(replicas):
+    import math
+    import httpx
+    url = replicas[0].base_url
+    # chat format
+    r = httpx.post(url + "/v1/chat/completions", json={
+        "model": "tiny", "messages": [{"role": "user", "content": "lp"}],
+        "max_tokens": 4, "temperature": 0.0,
+        "logprobs": True, "top_logprobs": 2}, timeout=60.0).json()
+    lp = r["choices"][0]["logprobs"]["content"]
+    assert len(lp) == 4
+    for e in lp:
+        assert e["logprob"] <= 0.0 and len(e["top_logprobs"]) == 2
+        assert abs(e["top_logprobs"][0]["logprob"] - e["logprob"]) < 1e-5
+    # completions format
+    r = httpx.post(url + "/v1/completions", json={
+        "model": "tiny", "prompt": "x", "max_tokens": 3,
+        "temperature": 0.0, "logprobs": 2}, timeout=60.0).json()
+    lp = r["choices"][0]["logprobs"]
+    assert len(lp["token_logprobs"]) == 3 and len(lp["top_logprobs"]) == 3
+    assert all(math.exp(v) <= 1.0 + 1e-6 for v in lp["token_logprobs"])
+    # raw engine endpoint
+    r = httpx.post(url + "/v1/generate", json={
+        "prompt_ids": [1, 2, 3], "max_tokens": 2, "ignore_eos": True,
+        "logprobs": 2}, timeout=60.0).json()
+    assert len(r["logprobs"]) == 2
+    assert r["logprobs"][0]["top"][0][0] == r["output_ids"][0]
