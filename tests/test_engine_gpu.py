@@ -244,3 +244,28 @@ def test_spec_lookup_gpu_verify_path():
     # drafts from the plain run's own tokens: acceptance must occur
     assert m1["spec_drafted"] > 0 and m1["spec_accepted"] > 0, m1
     assert m1["spec_steps"] > 0
+
+
+def test_logprobs_in_graph_decode():
+    """logprob computation (logsumexp + topk + gather) is captured inside
+    the decode hipGraph; chosen == top-1 for greedy and lists align."""
+    cfg = CONFIGS["tiny"]
+    eng = LLMEngine(cfg, device="cuda", page_size=4, num_pages=128,
+                    max_num_seqs=4, enable_graphs=True, seed=3)
+    rid = eng.add_request([1, 5, 9, 20],
+                          SamplingParams(max_tokens=6, ignore_eos=True,
+                                         logprobs=3))
+    fin = None
+    for _ in range(60):
+        eng.step()
+        fin = fin or eng.get_finished(rid)
+        if fin:
+            break
+    assert fin is not None and fin.logprobs is not None
+    assert len(fin.logprobs) == len(fin.output_ids) == 6
+    for tok, e in zip(fin.output_ids, fin.logprobs):
+        assert e["top"][0][0] == tok
+        assert abs(e["top"][0][1] - e["logprob"]) < 1e-4
+        assert e["logprob"] <= 0.0
+    # an lp-keyed graph was captured (distinct from the plain key)
+    assert any(k[2] for k in eng._graphs), eng._graphs.keys()
