@@ -51,8 +51,11 @@ def save_checkpoint(model: LlamaForCausalLM, path: str) -> None:
     sd = {k: v for k, v in model.state_dict().items()}
     for ours, hf in _HF_MAP.items():
         tensors[hf] = sd[ours].cpu().contiguous()
+    moe = cfg.num_experts > 1
     for i in range(cfg.num_layers):
         for ours, hf in _hf_layer_names(i).items():
+            if moe and (".mlp.gate_up" in ours or ".mlp.down" in ours):
+                continue  # emitted in Mixtral layout below
             t = sd[ours].cpu()
             if isinstance(hf, tuple):
                 if "qkv" in ours:
@@ -65,9 +68,21 @@ def save_checkpoint(model: LlamaForCausalLM, path: str) -> None:
                     tensors[hf[1]] = u.contiguous()
             else:
                 tensors[hf] = t.contiguous()
+        if moe:  # HF Mixtral layout: block_sparse_moe.gate + experts.N.w1/2/3
+            p = f"model.layers.{i}.block_sparse_moe."
+            tensors[p + "gate.weight"] = \
+                sd[f"layers.{i}.mlp.router"].cpu().contiguous()
+            gu = sd[f"layers.{i}.mlp.gate_up"].cpu()
+            dn = sd[f"layers.{i}.mlp.down"].cpu()
+            for e in range(cfg.num_experts):
+                g, u = gu[e].chunk(2, 0)
+                tensors[p + f"experts.{e}.w1.weight"] = g.contiguous()
+                tensors[p + f"experts.{e}.w3.weight"] = u.contiguous()
+                tensors[p + f"experts.{e}.w2.weight"] = dn[e].contiguous()
     save_file(tensors, str(root / "model.safetensors"))
-    (root / "config.json").write_text(json.dumps({
-        "architectures": ["LlamaForCausalLM"],
+    cfg_json = {
+        "architectures": ["MixtralForCausalLM" if moe else
+                          "LlamaForCausalLM"],
         "hidden_size": cfg.hidden_size,
         "intermediate_size": cfg.intermediate_size,
         "num_hidden_layers": cfg.num_layers,
@@ -78,7 +93,11 @@ def save_checkpoint(model: LlamaForCausalLM, path: str) -> None:
         "rope_theta": cfg.rope_theta,
         "max_position_embeddings": cfg.max_position,
         "rms_norm_eps": cfg.rms_eps,
-    }, indent=2))
+    }
+    if moe:
+        cfg_json["num_local_experts"] = cfg.num_experts
+        cfg_json["num_experts_per_tok"] = cfg.num_experts_per_tok
+    (root / "config.json").write_text(json.dumps(cfg_json, indent=2))
 
 
 def config_from_dir(path: str) -> LlamaConfig:
@@ -95,7 +114,9 @@ def config_from_dir(path: str) -> LlamaConfig:
         vocab_size=c["vocab_size"],
         rope_theta=c.get("rope_theta", 500000.0),
         max_position=c.get("max_position_embeddings", 8192),
-        rms_eps=c.get("rms_norm_eps", 1e-5))
+        rms_eps=c.get("rms_norm_eps", 1e-5),
+        num_experts=c.get("num_local_experts", 1),
+        num_experts_per_tok=c.get("num_experts_per_tok", 2))
 
 
 class _ShardedReader:
@@ -162,9 +183,24 @@ def load_checkpoint(model: LlamaForCausalLM, path: str,
             L.attn.qkv.copy_(torch.cat([q, k, v], 0).to(dev, L.attn.qkv.dtype))
             o = rd.get(p + "self_attn.o_proj.weight")[:, rank * hq * D:(rank + 1) * hq * D]
             L.attn.o.copy_(o.to(dev, L.attn.o.dtype))
-            g = rows(p + "mlp.gate_proj.weight", rank * inter, (rank + 1) * inter)
-            u = rows(p + "mlp.up_proj.weight", rank * inter, (rank + 1) * inter)
-            L.mlp.gate_up.copy_(torch.cat([g, u], 0).to(dev, L.mlp.gate_up.dtype))
-            d = rd.get(p + "mlp.down_proj.weight")[:, rank * inter:(rank + 1) * inter]
-            L.mlp.down.copy_(d.to(dev, L.mlp.down.dtype))
+            if cfg_full.num_experts > 1:
+                m = p + "block_sparse_moe."
+                L.mlp.router.copy_(
+                    rd.get(m + "gate.weight").to(dev, L.mlp.router.dtype))
+                for e in range(cfg_full.num_experts):
+                    g = rows(m + f"experts.{e}.w1.weight",
+                             rank * inter, (rank + 1) * inter)
+                    u = rows(m + f"experts.{e}.w3.weight",
+                             rank * inter, (rank + 1) * inter)
+                    L.mlp.gate_up[e].copy_(
+                        torch.cat([g, u], 0).to(dev, L.mlp.gate_up.dtype))
+                    d = rd.get(m + f"experts.{e}.w2.weight")[
+                        :, rank * inter:(rank + 1) * inter]
+                    L.mlp.down[e].copy_(d.to(dev, L.mlp.down.dtype))
+            else:
+                g = rows(p + "mlp.gate_proj.weight", rank * inter, (rank + 1) * inter)
+                u = rows(p + "mlp.up_proj.weight", rank * inter, (rank + 1) * inter)
+                L.mlp.gate_up.copy_(torch.cat([g, u], 0).to(dev, L.mlp.gate_up.dtype))
+                d = rd.get(p + "mlp.down_proj.weight")[:, rank * inter:(rank + 1) * inter]
+                L.mlp.down.copy_(d.to(dev, L.mlp.down.dtype))
     return model

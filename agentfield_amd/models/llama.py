@@ -38,6 +38,10 @@ class LlamaConfig:
     # Embedding stays replicated (lookup is local); logits shards are
     # all-gathered along the vocab dim so sampling sees full logits.
     lm_vocab_rows: int | None = None
+    # sparse MoE (Mixtral-style): >1 experts replaces the dense FFN with a
+    # softmax-routed top-k mixture; intermediate_size is PER EXPERT.
+    num_experts: int = 1
+    num_experts_per_tok: int = 2
 
     @property
     def q_size(self):
@@ -60,7 +64,9 @@ class LlamaConfig:
             max_position=self.max_position, rms_eps=self.rms_eps,
             tie_embeddings=self.tie_embeddings,
             lm_vocab_rows=(self.vocab_size // tp
-                           if self.vocab_size % tp == 0 else None))
+                           if self.vocab_size % tp == 0 else None),
+            num_experts=self.num_experts,
+            num_experts_per_tok=self.num_experts_per_tok)
 
 
 CONFIGS = {
@@ -79,6 +85,18 @@ CONFIGS = {
                             intermediate_size=8192, num_layers=16,
                             num_heads=16, num_kv_heads=8, vocab_size=32000,
                             max_position=8192),
+    # sparse MoE family (Mixtral-style; intermediate_size is per expert)
+    "mixtral-8x7b": LlamaConfig(name="mixtral-8x7b", hidden_size=4096,
+                                intermediate_size=14336, num_layers=32,
+                                num_heads=32, num_kv_heads=8,
+                                vocab_size=32000, rope_theta=1e6,
+                                max_position=8192, num_experts=8,
+                                num_experts_per_tok=2),
+    "tiny-moe": LlamaConfig(name="tiny-moe", hidden_size=256,
+                            intermediate_size=512, num_layers=2,
+                            num_heads=2, num_kv_heads=1, vocab_size=512,
+                            max_position=512, num_experts=4,
+                            num_experts_per_tok=2),
 }
 
 
@@ -170,13 +188,50 @@ class LlamaMLP(nn.Module):
                           self.down)
 
 
+class MoEMLP(nn.Module):
+    """Mixtral-style sparse FFN (reference family scope: the upstream
+    framework serves arbitrary provider models through LiteLLM; in-process
+    we add MoE as a second model family next to dense Llama).
+
+    Softmax router over num_experts, top-k per token with renormalized
+    gates, token-dropless eager dispatch: each expert runs the SAME fused
+    gate_up(+SwiGLU)/down ops as the dense MLP on its token subset, so the
+    skinny-GEMM decode kernels serve MoE decode shapes unchanged.  Routing
+    shapes are data-dependent, so MoE decode is not hipGraph-captured
+    (static-capacity capture is the round-2 optimization)."""
+
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        E, H, inter = cfg.num_experts, cfg.hidden_size, cfg.intermediate_size
+        self.top_k = cfg.num_experts_per_tok
+        self.router = nn.Parameter(torch.empty(E, H))
+        self.gate_up = nn.Parameter(torch.empty(E, 2 * inter, H))
+        self.down = nn.Parameter(torch.empty(E, H, inter))
+
+    def forward(self, x):
+        # fp32 routing (bf16 softmax near-ties destabilize expert choice)
+        probs = torch.softmax(x.float() @ self.router.float().t(), dim=-1)
+        topv, topi = probs.topk(self.top_k, dim=-1)
+        topv = topv / topv.sum(dim=-1, keepdim=True)
+        out = torch.zeros(x.shape, dtype=torch.float32, device=x.device)
+        for e in range(self.router.shape[0]):
+            sel, slot = (topi == e).nonzero(as_tuple=True)
+            if sel.numel() == 0:
+                continue
+            xe = x[sel]
+            ye = ops.linear(ops.linear(xe, self.gate_up[e], silu_fuse=True),
+                            self.down[e])
+            out.index_add_(0, sel, ye.float() * topv[sel, slot].unsqueeze(1))
+        return out.to(x.dtype)
+
+
 class LlamaLayer(nn.Module):
     def __init__(self, cfg: LlamaConfig, layer_idx: int):
         super().__init__()
         self.input_norm = nn.Parameter(torch.empty(cfg.hidden_size))
         self.post_norm = nn.Parameter(torch.empty(cfg.hidden_size))
         self.attn = LlamaAttention(cfg, layer_idx)
-        self.mlp = LlamaMLP(cfg)
+        self.mlp = MoEMLP(cfg) if cfg.num_experts > 1 else LlamaMLP(cfg)
         self.eps = cfg.rms_eps
 
     def forward(self, h, residual, positions, rope_tab, kv, md):
@@ -267,7 +322,12 @@ class LlamaForCausalLM(nn.Module):
             layer.input_norm.fill_(1.0)
             layer.mlp.gate_up.copy_(
                 (layer.mlp.gate_up.float() * layer.post_norm.float()
-                 ).to(layer.mlp.gate_up.dtype))
+                 ).to(layer.mlp.gate_up.dtype))  # broadcasts over MoE [E,2I,H]
+            if isinstance(layer.mlp, MoEMLP):
+                # the router consumes the same post-norm output: fold there too
+                layer.mlp.router.copy_(
+                    (layer.mlp.router.float() * layer.post_norm.float()
+                     ).to(layer.mlp.router.dtype))
             layer.post_norm.fill_(1.0)
         self.lm_head.copy_((self.lm_head.float() * self.final_norm.float()
                             ).to(self.lm_head.dtype))
@@ -283,6 +343,7 @@ class LlamaForCausalLM(nn.Module):
             return False
         cfg = self.cfg
         return (not getattr(self, "no_fused_decode", False) and T <= 128
+                and cfg.num_experts == 1  # MoE routes through the eager path
                 and cfg.hidden_size % 64 == 0
                 and cfg.intermediate_size % 64 == 0
                 and (cfg.q_size + 2 * cfg.kv_size) % 64 == 0

@@ -63,9 +63,13 @@ def shard_llama_weights(model: LlamaForCausalLM, full_cfg: LlamaConfig,
                 full_shape = (cfg.hidden_size, cfg.q_size)
             elif name.endswith("mlp.gate_up"):
                 full_shape = (2 * cfg.intermediate_size, cfg.hidden_size)
+                if p.dim() == 3:  # MoE: per-expert FFN, same col sharding
+                    full_shape = (cfg.num_experts,) + full_shape
             elif name.endswith("mlp.down"):
                 full_shape = (cfg.hidden_size, cfg.intermediate_size)
-            else:  # norms
+                if p.dim() == 3:
+                    full_shape = (cfg.num_experts,) + full_shape
+            else:  # norms, MoE router (replicated)
                 full_shape = tuple(p.shape)
             full = _full_param(name, full_shape, base_seed, dev)
             if name.endswith("attn.qkv"):
@@ -77,11 +81,13 @@ def shard_llama_weights(model: LlamaForCausalLM, full_cfg: LlamaConfig,
             elif name.endswith("attn.o"):
                 shard = full[:, rank * hq * D:(rank + 1) * hq * D]
             elif name.endswith("mlp.gate_up"):
-                gate, up = full.chunk(2, dim=0)
-                shard = torch.cat([gate[rank * inter:(rank + 1) * inter],
-                                   up[rank * inter:(rank + 1) * inter]], dim=0)
+                cdim = full.dim() - 2  # expert dim (if any) leads
+                gate, up = full.chunk(2, dim=cdim)
+                sl = slice(rank * inter, (rank + 1) * inter)
+                shard = torch.cat([gate.narrow(cdim, sl.start, inter),
+                                   up.narrow(cdim, sl.start, inter)], dim=cdim)
             elif name.endswith("mlp.down"):
-                shard = full[:, rank * inter:(rank + 1) * inter]
+                shard = full.narrow(full.dim() - 1, rank * inter, inter)
             elif name == "lm_head" and p.shape[0] != cfg.vocab_size:
                 shard = full[rank * vs_rows:(rank + 1) * vs_rows]
             else:

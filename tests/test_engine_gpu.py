@@ -270,3 +270,20 @@ def test_logprobs_in_graph_decode():
         assert e["logprob"] <= 0.0
     # an lp-keyed graph was captured (distinct from the plain key)
     assert any(k[2] for k in eng._graphs), eng._graphs.keys()
+
+
+def test_moe_engine_on_gpu():
+    """tiny-moe end-to-end on the HIP kernels (eager decode — MoE routing
+    shapes are data-dependent, so no graph capture)."""
+    cfg = CONFIGS["tiny-moe"]
+    outs = []
+    for _ in range(2):
+        eng = LLMEngine(cfg, device="cuda", page_size=4, num_pages=128,
+                        max_num_seqs=4, enable_graphs=True, seed=5)
+        assert not eng.enable_graphs
+        outs.append(eng.generate([[1, 5, 9, 20], [3, 7, 2]],
+                                 SamplingParams(max_tokens=6,
+                                                ignore_eos=True)))
+    assert outs[0] == outs[1]
+    assert all(len(o) == 6 for o in outs[0])
+    assert all(0 <= t < cfg.vocab_size for o in outs[0] for t in o)
