@@ -179,6 +179,7 @@ class LLMEngine:
                 if self.spec_lookup > 0 and self.tp_group is None and \
                         all(s.sampling.temperature == 0.0 and
                             s.sampling.top_k <= 0 and s.sampling.top_p >= 1.0
+                            and not s.sampling.json_mode
                             for s in batch.seqs):
                     drafts = {s.seq_id: self._draft_for(s)
                               for s in batch.seqs}
@@ -267,6 +268,8 @@ class LLMEngine:
             return [], []
         last_rows = torch.tensor(done_rows, dtype=torch.int32, device=dev)
         logits = self.model(ids_t, pos_t, self.kv, md, logit_rows=last_rows)
+        if any(s.sampling.json_mode for s in done):
+            logits = logits + self._json_mask(done).to(logits.dtype)
         temps = torch.tensor([s.sampling.temperature for s in done],
                              dtype=torch.float32, device=dev)
         kw = {}
@@ -300,6 +303,38 @@ class LLMEngine:
             seq.logprobs.append({"logprob": chosen[i],
                                  "top": list(zip(ti[i][:nreq],
                                                  tv[i][:nreq]))})
+
+    # -- grammar-constrained JSON decoding ----------------------------------
+    def _json_allowed(self, seq: Sequence) -> list[int]:
+        """Token ids legal for seq's next token under the JSON grammar,
+        completable within its remaining budget.  The FSM is cached on the
+        sequence and rebuilt after preemption (output_ids reset)."""
+        from .jsonfsm import EOS_ID, JsonFSM
+        fsm = getattr(seq, "_fsm", None)
+        pos = getattr(seq, "_fsm_pos", 0)
+        if fsm is None or pos > len(seq.output_ids):
+            fsm, pos = JsonFSM(), 0
+        for tid in seq.output_ids[pos:]:
+            b = tid - 4  # ByteTokenizer offset
+            if 0 <= b < 256:
+                fsm.advance(b)
+        seq._fsm, seq._fsm_pos = fsm, len(seq.output_ids)
+        remaining = seq.sampling.max_tokens - len(seq.output_ids)
+        ids = fsm.allowed_token_ids(remaining)
+        return [self.eos_id if i == EOS_ID else i for i in ids]
+
+    def _json_mask(self, seqs: list[Sequence]) -> torch.Tensor:
+        """Additive [len(seqs), V] mask: 0 for allowed, -inf elsewhere for
+        json_mode rows; all-0 for unconstrained rows."""
+        V = self.cfg.vocab_size
+        mask = torch.zeros(len(seqs), V)
+        for i, seq in enumerate(seqs):
+            if seq.sampling.json_mode:
+                # -1e30 not -inf: stays finite through the bf16 top-k/p
+                # histogram sampler while still contributing zero mass
+                mask[i] = -1e30
+                mask[i, self._json_allowed(seq)] = 0.0
+        return mask.to(self.device)
 
     # -- speculative decode (prompt lookup, greedy-exact) -------------------
     def _draft_for(self, seq: Sequence) -> list[int]:
@@ -436,12 +471,14 @@ class LLMEngine:
         d["bt"][:bs].copy_(h["bt"][:bs], non_blocking=nb)
 
     def _decode_forward(self, bs: int, nsplit: int, scratch, tkp: bool,
-                        lp: bool = False):
+                        lp: bool = False, mask: torch.Tensor | None = None):
         d = self._dec
         md = AttnMetadata(is_prefill=False, slots=d["slots"][:bs],
                           block_table=d["bt"][:bs], seq_lens_t=d["lens"][:bs],
                           nsplit=nsplit, decode_scratch=scratch)
         logits = self.model(d["ids"][:bs], d["pos"][:bs], self.kv, md)
+        if mask is not None:  # grammar-constrained rows (eager only)
+            logits = logits + mask.to(logits.dtype)
         kw = {"topk": d["topk"][:bs], "topp": d["topp"][:bs]} if tkp else {}
         ops.sample(logits, d["temps"][:bs], self.sampler,
                    out=d["tokens"][:bs], **kw)
@@ -514,13 +551,20 @@ class LLMEngine:
         bs = _bucket_for(n, self.max_num_seqs)
         tkp = any(s.sampling.top_k > 0 or s.sampling.top_p < 1.0 for s in seqs)
         lp = any(s.sampling.logprobs > 0 for s in seqs)
+        jm = any(s.sampling.json_mode for s in seqs)
         self._fill_decode_buffers(seqs, bs)
-        if self.enable_graphs:
+        if self.enable_graphs and not jm:
             self._get_graph(bs, tkp, lp)["graph"].replay()
         else:
+            # json grammar masks are data-dependent -> eager decode
+            mask = None
+            if jm:
+                mask = torch.zeros(bs, self.cfg.vocab_size,
+                                   device=self.device)
+                mask[:n] = self._json_mask(seqs)
             nsplit = choose_nsplit(bs, self.cfg.num_kv_heads) if self.is_gpu else 1
             self._decode_forward(bs, nsplit, self._make_scratch(bs, nsplit),
-                                 tkp, lp)
+                                 tkp, lp, mask)
         if lp:
             self._attach_decode_logprobs(seqs, n)
         toks = self._dec["tokens"][:n]

@@ -16,6 +16,8 @@ class SamplingParams:
     ignore_eos: bool = False
     logprobs: int = 0       # >0: report chosen-token logprob + top-N
                             # alternatives per emitted token (N <= 8)
+    json_mode: bool = False  # grammar-constrained valid-JSON decoding
+                             # (byte tokenizer; engine/jsonfsm.py)
 
 
 class SeqStatus(enum.Enum):

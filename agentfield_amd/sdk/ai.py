@@ -28,6 +28,7 @@ class AIConfig:
     timeout: float = 600.0
     device: str | None = None           # None -> cuda if available
     fallback_models: tuple = ()
+    json_only: bool = False             # grammar-constrained valid JSON
     extra: dict = field(default_factory=dict)
 
     def merged(self, **overrides) -> "AIConfig":
@@ -111,7 +112,8 @@ class EngineRunner:
         if len(ids) + cfg.max_tokens > limit:
             ids = ids[-(limit - cfg.max_tokens):]  # token-aware trim
         sp = SamplingParams(max_tokens=cfg.max_tokens,
-                            temperature=cfg.temperature)
+                            temperature=cfg.temperature,
+                            json_mode=cfg.json_only)
         w = self.submit(ids, sp)
         if not w["done"].wait(cfg.timeout):
             raise TimeoutError("engine generate timed out")
@@ -127,7 +129,8 @@ class EngineRunner:
     def stream_text(self, prompt: str, cfg: AIConfig):
         ids = self.tokenizer.encode(prompt)
         sp = SamplingParams(max_tokens=cfg.max_tokens,
-                            temperature=cfg.temperature)
+                            temperature=cfg.temperature,
+                            json_mode=cfg.json_only)
         sq: queue.Queue = queue.Queue()
         w = self.submit(ids, sp, stream_q=sq)
         stops = tuple(s for s in (cfg.stop or ()) if s)
@@ -266,6 +269,11 @@ class AgentAI:
                  user: str | None = None, schema=None, stream: bool = False,
                  **overrides):
         cfg = self.config.merged(**overrides)
+        if schema is not None and not cfg.json_only:
+            # schema requests get grammar-constrained JSON decoding: the
+            # engine can GUARANTEE syntactic validity (engine/jsonfsm.py),
+            # unlike the reference's prompt-and-validate seam
+            cfg = cfg.merged(json_only=True)
         parts = []
         sys_p = system or cfg.system_prompt
         if sys_p:

@@ -67,7 +67,8 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
             max_tokens=int(body.get("max_tokens", 128)),
             temperature=float(body.get("temperature", 0.0)),
             ignore_eos=bool(body.get("ignore_eos", False)),
-            logprobs=min(int(body.get("logprobs", 0) or 0), 8))
+            logprobs=min(int(body.get("logprobs", 0) or 0), 8),
+            json_mode=bool(body.get("json_mode", False)))
         if "prompt_ids" in body:
             ids = [int(x) for x in body["prompt_ids"]]
         else:
@@ -168,9 +169,11 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
                     if body.get("logprobs") else 0)
         else:
             lp_n = min(int(body.get("logprobs", 0) or 0), 8)
+        rf = body.get("response_format") or {}
         sp = SamplingParams(max_tokens=max_tokens,
                             temperature=float(body.get("temperature", 0.0)),
-                            logprobs=lp_n)
+                            logprobs=lp_n,
+                            json_mode=rf.get("type") == "json_object")
         stop_in = body.get("stop") or []
         stops = tuple(s for s in ([stop_in] if isinstance(stop_in, str)
                                   else stop_in) if s)

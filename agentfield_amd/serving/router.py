@@ -29,6 +29,7 @@ class RemoteRunner:
                                     temperature=cfg.temperature,
                                     stop=list(cfg.stop or ()),
                                     model=cfg.model,
+                                    json_mode=getattr(cfg, "json_only", False),
                                     timeout=cfg.timeout)["text"]
 
     def stream_text(self, prompt: str, cfg):

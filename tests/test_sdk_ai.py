@@ -162,3 +162,20 @@ def test_fallback_model_chain(tiny_runner):
                               fallback_models=("also-missing",)))
     with pytest.raises(RuntimeError, match="all models failed"):
         ai_bad("nope")
+
+
+def test_ai_json_only_always_parses(tiny_runner):
+    """json_only=True (set automatically by ai(schema=...)) grammar-
+    constrains decoding: raw output text is ALWAYS parseable JSON, at
+    sampling temperature, regardless of the (random-init) model."""
+    import json as _json
+    ai = AgentAI(AIConfig(model="tiny", max_tokens=24, timeout=120,
+                          temperature=0.9, json_only=True))
+    for i in range(4):
+        out = ai(f"emit some json please, variant {i}")
+        _json.loads(out.strip())
+    # ai(schema=...) flips json_only on implicitly: the document always
+    # parses, so the schema path returns the PARSED value (which may be a
+    # plain string if the model emitted a JSON string) — never raises
+    merged = AgentAI(AIConfig(model="tiny", max_tokens=8, timeout=120))
+    merged("obj", schema={"type": "object"})

@@ -271,3 +271,19 @@ def test_openai_logprobs(replicas):
         "logprobs": 2}, timeout=60.0).json()
     assert len(r["logprobs"]) == 2
     assert r["logprobs"][0]["top"][0][0] == r["output_ids"][0]
+
+
+def test_openai_json_mode(replicas):
+    import json as _json
+    import httpx
+    url = replicas[0].base_url
+    r = httpx.post(url + "/v1/chat/completions", json={
+        "model": "tiny", "messages": [{"role": "user", "content": "emit"}],
+        "max_tokens": 16, "temperature": 0.9,
+        "response_format": {"type": "json_object"}}, timeout=60.0).json()
+    _json.loads(r["choices"][0]["message"]["content"].strip())
+    # raw engine endpoint
+    r = httpx.post(url + "/v1/generate", json={
+        "prompt": "data:", "max_tokens": 12, "temperature": 1.1,
+        "json_mode": True}, timeout=60.0).json()
+    _json.loads(r["text"].strip())
