@@ -287,3 +287,30 @@ def test_moe_engine_on_gpu():
     assert outs[0] == outs[1]
     assert all(len(o) == 6 for o in outs[0])
     assert all(0 <= t < cfg.vocab_size for o in outs[0] for t in o)
+
+
+def test_json_mode_on_gpu():
+    """Constrained decoding over the HIP kernels + HIP Gumbel sampler:
+    masked (-1e30 in bf16) logits must never win, so every output parses."""
+    import json as _json
+    from agentfield_amd.sdk.ai import ByteTokenizer
+    cfg = CONFIGS["tiny"]
+    eng = LLMEngine(cfg, device="cuda", page_size=4, num_pages=128,
+                    max_num_seqs=4, enable_graphs=True, seed=11)
+    rids = [eng.add_request([1, 5 + i, 9], SamplingParams(
+        max_tokens=6 + 4 * i, temperature=0.9, json_mode=True))
+        for i in range(3)]
+    fins = {}
+    for _ in range(200):
+        eng.step()
+        for r in rids:
+            if r not in fins:
+                f = eng.get_finished(r)
+                if f:
+                    fins[r] = f
+        if len(fins) == 3:
+            break
+    assert len(fins) == 3
+    tok = ByteTokenizer()
+    for r in rids:
+        _json.loads(tok.decode(fins[r].output_ids).strip())
