@@ -17,6 +17,15 @@ __device__ __forceinline__ float u32_to_unit(uint32_t x) {
   return (x >> 8) * (1.f / 16777216.f) + (0.5f / 16777216.f);
 }
 
+// Gumbel(0,1) sample from u in (0,1).  The inner fast-math __logf(u) can
+// round to EXACTLY 0 for u near 1, making -__logf(0) = +inf — which lets
+// any token (including grammar-masked -1e30 rows) win the argmax.  Floor
+// the inner result at 1e-30 (gumbel <= ~69, finite).
+__device__ __forceinline__ float gumbel_of(float u) {
+  const float nl = fmaxf(-__logf(u), 1e-30f);
+  return -__logf(nl);
+}
+
 __device__ __forceinline__ uint32_t hash3(uint32_t a, uint32_t b, uint32_t c) {
   // Wang/xxhash-style avalanche mix
   uint32_t h = a * 0x9E3779B1u + b * 0x85EBCA77u + c * 0xC2B2AE3Du + 0x27220A95u;
@@ -44,7 +53,7 @@ __global__ void __launch_bounds__(256) sample_partial_kernel(
     float x = bf2f(row[v]);
     if (t > 0.f) {
       const float u = u32_to_unit(hash3(seed ^ st, (uint32_t)b, (uint32_t)v));
-      x = x * invt - __logf(-__logf(u));
+      x = x * invt + gumbel_of(u);
     }
     if (x > best) { best = x; bidx = v; }
   }
@@ -197,7 +206,7 @@ __global__ void __launch_bounds__(256) tkp_sample_kernel(
     if (x < th) continue;
     if (t > 0.f) {
       const float u = u32_to_unit(hash3(seed ^ st, (uint32_t)b, (uint32_t)v));
-      x -= __logf(-__logf(u));
+      x += gumbel_of(u);
     }
     if (x > best) { best = x; bidx = v; }
   }
