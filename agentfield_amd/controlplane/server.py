@@ -119,6 +119,18 @@ class ControlPlane:
             await self.client.aclose()
 
     # ----------------------------------------------------- execution core
+    async def probe_node(self, node: dict) -> bool:
+        """One active GET /health probe (C9's single-node variant)."""
+        import httpx
+        if not node.get("base_url"):
+            return False
+        try:
+            async with httpx.AsyncClient(timeout=5.0) as client:
+                r = await client.get(node["base_url"].rstrip("/") + "/health")
+                return r.status_code == 200
+        except Exception:
+            return False
+
     @staticmethod
     def parse_target(target: str) -> tuple[str, str]:
         if "." not in target:
@@ -241,6 +253,11 @@ class ControlPlane:
             "terminal": st.is_terminal(status), "run_id": rec.get("run_id"),
         })
         if st.is_terminal(status):
+            self.buses.reasoner.publish({
+                "type": "reasoner.execution", "node_id": rec.get("node_id"),
+                "reasoner_id": rec.get("reasoner_id"), "status": status,
+                "duration_ms": rec.get("duration_ms"),
+            })
             self.webhooks.notify(rec)
             if self.cfg.auto_vc and self.vcs:
                 try:
@@ -390,6 +407,86 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
         cp.buses.node.publish({"type": "node.removed", "node_id": node_id})
         return {"status": "ok"}
 
+    @app.post("/api/v1/nodes/status/bulk")
+    async def nodes_status_bulk(req: Request):
+        body = await req.json()
+        out = {}
+        for nid in body.get("ids", []):
+            node = cp.storage.get_node(nid)
+            out[nid] = ({"status": node["status"],
+                         "last_heartbeat": node.get("last_heartbeat")}
+                        if node else None)
+        return {"statuses": out}
+
+    @app.post("/api/v1/nodes/{node_id}/status/refresh")
+    async def node_status_refresh(node_id: str):
+        """Probe the agent's /health NOW and reconcile its status."""
+        node = cp.storage.get_node(node_id)
+        if node is None:
+            return JSONResponse({"error": "not found"}, status_code=404)
+        healthy = await cp.probe_node(node)
+        nxt = "active" if healthy else "unhealthy"
+        if node["status"] != nxt:
+            cp.storage.set_node_status(node_id, nxt)
+            cp.buses.node.publish({"type": "node.status",
+                                   "node_id": node_id, "status": nxt})
+        return {"node_id": node_id, "healthy": healthy, "status": nxt}
+
+    @app.get("/api/v1/nodes/{node_id}/lifecycle")
+    async def node_lifecycle(node_id: str):
+        node = cp.storage.get_node(node_id)
+        if node is None:
+            return JSONResponse({"error": "not found"}, status_code=404)
+        return {"node_id": node_id, "status": node["status"],
+                "registered_at": node.get("registered_at"),
+                "last_heartbeat": node.get("last_heartbeat"),
+                "last_status_change": node.get("last_status_change"),
+                "pending_actions": cp.storage.pending_actions(node_id)}
+
+    def _lifecycle_action(node_id: str, action: str, status: str | None):
+        node = cp.storage.get_node(node_id)
+        if node is None:
+            return JSONResponse({"error": "not found"}, status_code=404)
+        aid = cp.storage.enqueue_action(node_id, action)
+        if status:
+            cp.storage.set_node_status(node_id, status)
+            cp.buses.node.publish({"type": "node.status",
+                                   "node_id": node_id, "status": status})
+        return {"status": "ok", "action_id": aid}
+
+    @app.post("/api/v1/nodes/{node_id}/start")
+    async def node_start(node_id: str):
+        return _lifecycle_action(node_id, "start", "starting")
+
+    @app.post("/api/v1/nodes/{node_id}/stop")
+    async def node_stop(node_id: str):
+        return _lifecycle_action(node_id, "stop", "stopping")
+
+    @app.post("/api/v1/nodes/{node_id}/shutdown")
+    async def node_shutdown(node_id: str):
+        return _lifecycle_action(node_id, "shutdown", "stopping")
+
+    @app.post("/api/v1/nodes/{node_id}/actions/claim")
+    async def actions_claim(node_id: str, req: Request):
+        try:
+            body = await req.json()
+        except Exception:
+            body = {}
+        acts = cp.storage.claim_actions(
+            node_id, lease_s=float(body.get("lease_s", 30.0)),
+            limit=int(body.get("max", 16)))
+        return {"actions": acts}
+
+    @app.post("/api/v1/nodes/{node_id}/actions/ack")
+    async def actions_ack(node_id: str, req: Request):
+        body = await req.json()
+        ok = cp.storage.ack_action(int(body["action_id"]),
+                                   body.get("status", "done"))
+        if not ok:
+            return JSONResponse({"error": "unknown or unclaimed action"},
+                                status_code=409)
+        return {"status": "ok"}
+
     # ------------------------------------------------------------ execute
     @app.post("/api/v1/execute/{target}")
     async def execute_sync(target: str, req: Request):
@@ -432,6 +529,19 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
         resp.headers["X-Execution-ID"] = rec["id"]
         resp.headers["X-Run-ID"] = rec["run_id"]
         return resp
+
+    # legacy direct-execution routes (reference: POST /reasoners/:id and
+    # /skills/:id with workflow headers, reasoners.go:45+): same sync
+    # pipeline, target addressed by path segments instead of node.reasoner
+    @app.post("/api/v1/reasoners/{node_id}/{reasoner_id}")
+    async def execute_reasoner_direct(node_id: str, reasoner_id: str,
+                                      req: Request):
+        return await execute_sync(f"{node_id}.{reasoner_id}", req)
+
+    @app.post("/api/v1/skills/{node_id}/{skill_id}")
+    async def execute_skill_direct(node_id: str, skill_id: str,
+                                   req: Request):
+        return await execute_sync(f"{node_id}.{skill_id}", req)
 
     @app.post("/api/v1/execute/async/{target}")
     async def execute_async(target: str, req: Request):
@@ -501,6 +611,19 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
         return {"status": "ok"}
 
     # ------------------------------------------------- workflow events/DAG
+    @app.post("/api/v1/executions/{execution_id}/notes")
+    async def add_execution_note(execution_id: str, req: Request):
+        if cp.storage.get_execution(execution_id) is None:
+            return JSONResponse({"error": "not found"}, status_code=404)
+        body = await req.json()
+        nid = cp.storage.add_note(execution_id, str(body.get("note", "")),
+                                  author=str(body.get("author", "")))
+        return {"status": "ok", "note_id": nid}
+
+    @app.get("/api/v1/executions/{execution_id}/notes")
+    async def list_execution_notes(execution_id: str):
+        return {"notes": cp.storage.notes_for(execution_id)}
+
     @app.post("/api/v1/workflow/executions/events")
     async def workflow_events(req: Request):
         ev = await req.json()
@@ -766,6 +889,33 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
             finally:
                 cp.buses.node.unsubscribe(sid)
         return StreamingResponse(gen(), media_type="text/event-stream")
+
+    @app.get("/api/ui/v1/reasoners/events")
+    async def reasoner_sse():
+        sid, q = cp.buses.reasoner.subscribe()
+
+        async def gen():
+            try:
+                while True:
+                    try:
+                        ev = await asyncio.wait_for(q.get(), 15.0)
+                        yield _sse(ev)
+                    except asyncio.TimeoutError:
+                        yield ": keepalive\n\n"
+            finally:
+                cp.buses.reasoner.unsubscribe(sid)
+        return StreamingResponse(gen(), media_type="text/event-stream")
+
+    @app.get("/api/ui/v1/mcp")
+    async def ui_mcp():
+        """MCP visibility: servers advertised by registered nodes'
+        metadata (the SDK's MCPManager reports them at registration)."""
+        servers = []
+        for node in cp.storage.list_nodes():
+            for s in (node.get("metadata") or {}).get("mcp_servers", []):
+                servers.append({"node_id": node["id"], **(
+                    s if isinstance(s, dict) else {"name": s})})
+        return {"servers": servers}
 
     @app.get("/api/ui/v1/workflows/{run_id}/dag")
     async def workflow_dag(run_id: str, req: Request):

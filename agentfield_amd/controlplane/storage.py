@@ -75,6 +75,18 @@ CREATE TABLE IF NOT EXISTS workflow_runs (
   run_id TEXT PRIMARY KEY, status TEXT, root_execution_id TEXT,
   started_at REAL, finished_at REAL, metadata TEXT DEFAULT '{}'
 );
+CREATE TABLE IF NOT EXISTS node_actions (
+  id INTEGER PRIMARY KEY AUTOINCREMENT, node_id TEXT, action TEXT,
+  payload TEXT DEFAULT '{}', status TEXT DEFAULT 'pending',
+  lease_expires_at REAL DEFAULT 0, created_at REAL, acked_at REAL,
+  ack_status TEXT
+);
+CREATE INDEX IF NOT EXISTS idx_actions_node ON node_actions(node_id, status);
+CREATE TABLE IF NOT EXISTS execution_notes (
+  id INTEGER PRIMARY KEY AUTOINCREMENT, execution_id TEXT, author TEXT,
+  note TEXT, created_at REAL
+);
+CREATE INDEX IF NOT EXISTS idx_notes_exec ON execution_notes(execution_id);
 """
 
 
@@ -473,6 +485,70 @@ class Storage:
         return rows
 
     # ---------------- workflow runs (v2) ----------------
+    # ------------------------------------------- node action lease queue
+    # (reference: nodes actions claim/ack lease protocol, server.go:865-867)
+    def enqueue_action(self, node_id: str, action: str,
+                       payload: dict | None = None) -> int:
+        with self._lock:
+            cur = self._db.execute(
+                "INSERT INTO node_actions (node_id, action, payload, "
+                "created_at) VALUES (?,?,?,?)",
+                (node_id, action, json.dumps(payload or {}), now()))
+            self._db.commit()
+            return int(cur.lastrowid)
+
+    def claim_actions(self, node_id: str, lease_s: float = 30.0,
+                      limit: int = 16) -> list[dict]:
+        """Claim pending (or lease-expired claimed) actions for a node.
+        Claims are leases: unacked actions become claimable again after
+        lease_s, so a crashed agent never strands an action."""
+        t = now()
+        with self._lock:
+            rows = self._q(
+                "SELECT * FROM node_actions WHERE node_id=? AND "
+                "(status='pending' OR (status='claimed' AND "
+                "lease_expires_at<?)) ORDER BY id LIMIT ?",
+                (node_id, t, limit))
+            for r in rows:
+                self._db.execute(
+                    "UPDATE node_actions SET status='claimed', "
+                    "lease_expires_at=? WHERE id=?",
+                    (t + lease_s, r["id"]))
+            self._db.commit()
+        for r in rows:
+            r["payload"] = json.loads(r["payload"] or "{}")
+            r["status"] = "claimed"
+        return rows
+
+    def ack_action(self, action_id: int, status: str = "done") -> bool:
+        with self._lock:
+            cur = self._db.execute(
+                "UPDATE node_actions SET status='acked', acked_at=?, "
+                "ack_status=? WHERE id=? AND status='claimed'",
+                (now(), status, action_id))
+            self._db.commit()
+            return cur.rowcount > 0
+
+    def pending_actions(self, node_id: str) -> int:
+        r = self._q("SELECT COUNT(*) AS n FROM node_actions WHERE node_id=? "
+                    "AND status!='acked'", (node_id,))
+        return int(r[0]["n"])
+
+    # ------------------------------------------------- execution notes
+    def add_note(self, execution_id: str, note: str,
+                 author: str = "") -> int:
+        with self._lock:
+            cur = self._db.execute(
+                "INSERT INTO execution_notes (execution_id, author, note, "
+                "created_at) VALUES (?,?,?,?)",
+                (execution_id, author, note, now()))
+            self._db.commit()
+            return int(cur.lastrowid)
+
+    def notes_for(self, execution_id: str) -> list[dict]:
+        return self._q("SELECT * FROM execution_notes WHERE execution_id=? "
+                       "ORDER BY id", (execution_id,))
+
     def upsert_run(self, run_id: str, status: str,
                    root_execution_id: str | None = None) -> None:
         self._exec(

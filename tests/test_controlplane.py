@@ -542,3 +542,98 @@ def test_admin_grpc_execution_and_status(cp_server, greeting_agent):
         assert ex["execution"]["status"] == "completed"
     finally:
         server.stop(grace=0)
+
+
+def test_node_actions_lease_protocol(cp_server, greeting_agent):
+    """Action claim/ack leases (reference C7): pending actions are claimed
+    with a lease, re-claimable after expiry, acked exactly once."""
+    srv, cp = cp_server
+    base = srv.base_url + "/api/v1/nodes/greeter"
+    r = httpx.post(base + "/stop")
+    assert r.status_code == 200
+    aid = r.json()["action_id"]
+    lc = httpx.get(base + "/lifecycle").json()
+    assert lc["status"] == "stopping" and lc["pending_actions"] >= 1
+    # claim with a short lease
+    acts = httpx.post(base + "/actions/claim",
+                      json={"lease_s": 0.2}).json()["actions"]
+    assert any(a["id"] == aid and a["action"] == "stop" for a in acts)
+    # second claim within the lease: nothing new
+    assert httpx.post(base + "/actions/claim",
+                      json={"lease_s": 0.2}).json()["actions"] == []
+    time.sleep(0.3)  # lease expires -> claimable again
+    acts2 = httpx.post(base + "/actions/claim",
+                       json={"lease_s": 30}).json()["actions"]
+    assert any(a["id"] == aid for a in acts2)
+    assert httpx.post(base + "/actions/ack",
+                      json={"action_id": aid}).status_code == 200
+    # double-ack rejected
+    assert httpx.post(base + "/actions/ack",
+                      json={"action_id": aid}).status_code == 409
+    # restore status for other tests
+    httpx.post(base + "/start")
+    httpx.post(srv.base_url + "/api/v1/nodes/greeter/status",
+               json={"status": "active"})
+
+
+def test_node_bulk_status_and_refresh(cp_server, greeting_agent):
+    srv, _ = cp_server
+    r = httpx.post(srv.base_url + "/api/v1/nodes/status/bulk",
+                   json={"ids": ["greeter", "missing"]}).json()
+    assert r["statuses"]["greeter"]["status"]
+    assert r["statuses"]["missing"] is None
+    # live refresh probes the real agent server -> healthy
+    r = httpx.post(srv.base_url +
+                   "/api/v1/nodes/greeter/status/refresh").json()
+    assert r["healthy"] is True and r["status"] == "active"
+
+
+def test_legacy_direct_execution_routes(cp_server, greeting_agent):
+    srv, _ = cp_server
+    r = httpx.post(srv.base_url + "/api/v1/reasoners/greeter/greet",
+                   json={"input": {"name": "direct"}}, timeout=20.0)
+    assert r.status_code == 200, r.text
+    assert r.json()["result"] == {"greeting": "hello direct"}
+    r = httpx.post(srv.base_url + "/api/v1/skills/greeter/add",
+                   json={"input": {"a": 2, "b": 5}}, timeout=20.0)
+    assert r.status_code == 200 and r.json()["result"] == 7
+
+
+def test_execution_notes(cp_server, greeting_agent):
+    srv, _ = cp_server
+    r = httpx.post(srv.base_url + "/api/v1/execute/greeter.greet",
+                   json={"input": {"name": "note-me"}}, timeout=20.0)
+    eid = r.json()["execution_id"]
+    r = httpx.post(srv.base_url + f"/api/v1/executions/{eid}/notes",
+                   json={"note": "checked by ops", "author": "alice"})
+    assert r.status_code == 200
+    notes = httpx.get(srv.base_url +
+                      f"/api/v1/executions/{eid}/notes").json()["notes"]
+    assert len(notes) == 1 and notes[0]["note"] == "checked by ops"
+    assert notes[0]["author"] == "alice"
+    # unknown execution 404s
+    assert httpx.post(srv.base_url + "/api/v1/executions/exec_nope/notes",
+                      json={"note": "x"}).status_code == 404
+
+
+def test_reasoner_sse_stream(cp_server, greeting_agent):
+    srv, _ = cp_server
+    got = []
+
+    def listen():
+        with httpx.stream("GET", srv.base_url + "/api/ui/v1/reasoners/events",
+                          timeout=10.0) as resp:
+            for line in resp.iter_lines():
+                if line.startswith("data:"):
+                    got.append(json.loads(line[5:]))
+                    return
+
+    t = threading.Thread(target=listen, daemon=True)
+    t.start()
+    time.sleep(0.3)
+    httpx.post(srv.base_url + "/api/v1/execute/greeter.greet",
+               json={"input": {"name": "sse"}}, timeout=20.0)
+    t.join(timeout=10.0)
+    assert got and got[0]["type"] == "reasoner.execution"
+    assert got[0]["reasoner_id"] == "greet"
+    assert got[0]["status"] == "completed"
