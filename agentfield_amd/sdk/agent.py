@@ -96,6 +96,7 @@ class Agent(FastAPI):
         self._skills: dict[str, _FunctionMeta] = {}
         self._hb_thread: threading.Thread | None = None
         self._hb_stop = threading.Event()
+        self._action_handlers: dict[str, callable] = {}
         self._registered = False
         self._setup_builtin_routes()
 
@@ -405,6 +406,40 @@ class Agent(FastAPI):
             ok = self.client.heartbeat(self.node_id, {"status": "active"})
             if not ok:
                 self.register()  # resilient re-register (P10)
+            else:
+                self._drain_actions()
+
+    def _drain_actions(self):
+        """Claim and execute pending control-plane lifecycle actions
+        (reference C7 claim/ack lease protocol).  Stop/shutdown actions
+        end the heartbeat loop; custom actions go to @on_action handlers."""
+        try:
+            acts = self.client.claim_actions(self.node_id)
+        except Exception:
+            return
+        for a in acts:
+            status = "done"
+            try:
+                if a["action"] in ("stop", "shutdown"):
+                    self._hb_stop.set()
+                elif a["action"] in self._action_handlers:
+                    self._action_handlers[a["action"]](a.get("payload", {}))
+                else:
+                    status = "ignored"
+            except Exception as e:
+                status = f"error: {e}"
+            try:
+                self.client.ack_action(self.node_id, a["id"], status)
+            except Exception:
+                pass
+
+    def on_action(self, name: str):
+        """Decorator: handle a named control-plane action delivered via the
+        claim/ack lease queue."""
+        def deco(fn):
+            self._action_handlers[name] = fn
+            return fn
+        return deco
 
     def start_background(self):
         if self.auto_register:

@@ -23,6 +23,21 @@ class AgentFieldClient:
         r.raise_for_status()
         return r.json()
 
+    def claim_actions(self, node_id: str, lease_s: float = 30.0,
+                      max_n: int = 16) -> list[dict]:
+        r = self._client.post(
+            f"{self.base_url}/api/v1/nodes/{node_id}/actions/claim",
+            json={"lease_s": lease_s, "max": max_n})
+        r.raise_for_status()
+        return r.json().get("actions", [])
+
+    def ack_action(self, node_id: str, action_id: int,
+                   status: str = "done") -> bool:
+        r = self._client.post(
+            f"{self.base_url}/api/v1/nodes/{node_id}/actions/ack",
+            json={"action_id": action_id, "status": status})
+        return r.status_code == 200
+
     def heartbeat(self, node_id: str, payload: dict | None = None) -> bool:
         try:
             r = self._client.post(

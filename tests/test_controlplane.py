@@ -637,3 +637,27 @@ def test_reasoner_sse_stream(cp_server, greeting_agent):
     assert got and got[0]["type"] == "reasoner.execution"
     assert got[0]["reasoner_id"] == "greet"
     assert got[0]["status"] == "completed"
+
+
+def test_sdk_action_handler_roundtrip(cp_server, greeting_agent):
+    """Agent-side claim/ack: a custom @on_action handler runs on heartbeat
+    and the action is acked with its outcome."""
+    srv, cp = cp_server
+    _agent_srv, app = greeting_agent
+    seen = []
+
+    @app.on_action("rotate-keys")
+    def rotate(payload):
+        seen.append(payload)
+
+    cp.storage.enqueue_action("greeter", "rotate-keys", {"reason": "test"})
+    cp.storage.enqueue_action("greeter", "unknown-action")
+    app._drain_actions()  # what the heartbeat loop runs
+    assert seen == [{"reason": "test"}]
+    rows = cp.storage._q(
+        "SELECT action, status, ack_status FROM node_actions "
+        "WHERE node_id='greeter' AND action IN ('rotate-keys','unknown-action')")
+    by = {r["action"]: r for r in rows}
+    assert by["rotate-keys"]["status"] == "acked"
+    assert by["rotate-keys"]["ack_status"] == "done"
+    assert by["unknown-action"]["ack_status"] == "ignored"
