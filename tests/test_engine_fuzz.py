@@ -73,3 +73,61 @@ def test_engine_fuzz_add_step_cancel():
                 f"seq {rid} diverged after scheduling churn"
             checked += 1
     assert checked > 0
+
+
+def test_engine_fuzz_with_speculation():
+    """Same churn with prompt-lookup speculation enabled: multi-token
+    acceptance must stay greedy-exact through preemption and recompute.
+    Repetitive prompts make drafts fire; page pressure forces preemption."""
+    rng = random.Random(7)
+    eng = LLMEngine(CONFIGS["tiny"], device="cpu", dtype=torch.float32,
+                    page_size=4, num_pages=20, max_num_seqs=4,
+                    max_prefill_tokens=16, enable_graphs=False,
+                    max_waiting=64, spec_lookup=4)
+    total_pages = eng.sched.alloc.num_pages - 1
+    live: dict[int, dict] = {}
+    finished = {}
+    submitted = 0
+    for step in range(400):
+        if rng.random() < 0.3 and submitted < 40:
+            pat = [rng.randrange(500) for _ in range(rng.randint(2, 4))]
+            reps = rng.randint(2, 6)
+            prompt = (pat * reps)[:rng.randint(4, 18)]
+            mt = rng.randint(8, 24)
+            rid = eng.add_request(prompt, SamplingParams(max_tokens=mt,
+                                                         ignore_eos=True))
+            if rid is not None:
+                live[rid] = True
+                submitted += 1
+        eng.step()
+        for rid in list(live):
+            fin = eng.get_finished(rid)
+            if fin is not None:
+                finished[rid] = fin
+                del live[rid]
+    for _ in range(2000):
+        if not eng.has_work():
+            break
+        eng.step()
+        for rid in list(live):
+            fin = eng.get_finished(rid)
+            if fin is not None:
+                finished[rid] = fin
+                del live[rid]
+    assert not live
+    assert eng.sched.alloc.num_free == total_pages, "KV pages leaked"
+    assert eng.metrics["spec_drafted"] > 0, "speculation never fired"
+    assert eng.metrics["spec_accepted"] > 0, "no draft ever accepted"
+    assert eng.sched.n_preempted > 0, "churn config should preempt"
+    ref_eng = LLMEngine(CONFIGS["tiny"], device="cpu", dtype=torch.float32,
+                        page_size=4, num_pages=128, max_num_seqs=4,
+                        max_prefill_tokens=64, enable_graphs=False)
+    checked = 0
+    for rid, fin in finished.items():
+        assert len(fin.output_ids) == fin.sampling.max_tokens
+        if checked < 10:
+            want = ref_eng.generate([fin.prompt_ids], fin.sampling)[0]
+            assert fin.output_ids == want, \
+                f"seq {rid} diverged under speculation + churn"
+            checked += 1
+    assert checked > 0
