@@ -44,7 +44,8 @@ class LLMEngine:
                  max_num_seqs: int = 64, max_prefill_tokens: int = 8192,
                  max_waiting: int = 4096, enable_graphs: bool = True, eos_id: int = 2, seed: int = 0,
                  model: LlamaForCausalLM | None = None,
-                 tp_group=None, spec_lookup: int = 0):
+                 tp_group=None, spec_lookup: int = 0,
+                 prefix_cache: bool = False):
         self.cfg = cfg
         self.device = torch.device(device)
         self.is_gpu = self.device.type == "cuda"
@@ -66,9 +67,14 @@ class LLMEngine:
         max_pages_per_seq = (cfg.max_position + page_size - 1) // page_size
         self.max_pages_per_seq = max_pages_per_seq
         self.kv = KVCache(cfg, num_pages, page_size, self.device, dtype)
-        self.sched = make_scheduler(SchedulerConfig(
+        sched_cfg = SchedulerConfig(
             max_num_seqs=max_num_seqs, max_prefill_tokens=max_prefill_tokens,
-            page_size=page_size, num_pages=num_pages, max_waiting=max_waiting))
+            page_size=page_size, num_pages=num_pages, max_waiting=max_waiting)
+        if prefix_cache:
+            from .prefix_cache import PrefixCachingScheduler
+            self.sched = PrefixCachingScheduler(sched_cfg)
+        else:
+            self.sched = make_scheduler(sched_cfg)
         self.page_size = page_size
         self.eos_id = eos_id
         self.max_num_seqs = max_num_seqs
@@ -172,7 +178,8 @@ class LLMEngine:
                 return []
             if batch.is_prefill:
                 for s in batch.seqs:
-                    s.num_prefilled = 0
+                    # prefix cache: cached full pages skip their prefill
+                    s.num_prefilled = getattr(s, "cached_prefix", 0)
                 self._prefilling.extend(batch.seqs)
             else:
                 self.metrics["steps"] += 1

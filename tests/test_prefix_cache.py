@@ -1,0 +1,87 @@
+"""Automatic prefix caching (engine/prefix_cache.py): shared prompt pages,
+exactness, refcount hygiene, eviction, and preemption interplay."""
+import random
+
+import torch
+
+from agentfield_amd.engine import LLMEngine, SamplingParams
+from agentfield_amd.models import CONFIGS
+
+
+def make(prefix_cache=True, num_pages=64, **kw):
+    return LLMEngine(CONFIGS["tiny"], device="cpu", dtype=torch.float32,
+                     page_size=4, num_pages=num_pages, max_num_seqs=4,
+                     enable_graphs=False, prefix_cache=prefix_cache, **kw)
+
+
+PROMPT = [7, 3, 9, 1, 8, 2, 6, 4, 5, 9, 2, 7, 1, 3, 8, 6, 4, 2, 9, 5, 7, 1]
+SP = SamplingParams(max_tokens=6, ignore_eos=True)
+
+
+def test_prefix_cache_hits_and_exact_tokens():
+    base = make(prefix_cache=False).generate([PROMPT], SP)[0]
+    eng = make()
+    out1 = eng.generate([PROMPT], SP)[0]
+    cold_prefill = eng.metrics["prefill_tokens"]
+    assert out1 == base and eng.sched.cached_tokens == 0
+    out2 = eng.generate([PROMPT], SP)[0]
+    assert out2 == base
+    # 22-token prompt, page 4: (22-1)//4 = 5 full pages = 20 tokens cached
+    assert eng.sched.cached_tokens == 20
+    assert eng.metrics["prefill_tokens"] == cold_prefill + 2  # only tail
+    # shared prefix with a different tail also hits
+    p2 = PROMPT[:20] + [9, 9, 9]
+    base2 = make(prefix_cache=False).generate([p2], SP)[0]
+    assert eng.generate([p2], SP)[0] == base2
+    assert eng.sched.cached_tokens == 40  # +20 for p2's 5 shared pages
+
+
+def test_prefix_cache_page_accounting():
+    eng = make()
+    for _ in range(3):
+        eng.generate([PROMPT, PROMPT[:17]], SP)
+    alloc = eng.sched.alloc
+    # all sequence refs released; cache holds exactly one ref per entry
+    assert len(alloc.refs) == len(eng.sched._cache)
+    assert alloc.num_free == alloc.num_pages - 1 - len(eng.sched._cache)
+    assert all(alloc.refs[p] == 1 for p in eng.sched._cache.values())
+
+
+def test_prefix_cache_eviction_under_pressure():
+    rng = random.Random(3)
+    eng = make(num_pages=16)  # tiny pool forces eviction
+    for i in range(12):
+        p = [rng.randrange(500) for _ in range(rng.randint(8, 14))]
+        out = eng.generate([p], SP)[0]
+        assert len(out) == SP.max_tokens
+    # pool never deadlocks and cache stays within the pool
+    assert len(eng.sched._cache) <= 15
+
+
+def test_prefix_cache_with_preemption_exact():
+    """Preempted sequences re-admit THROUGH the cache (their own published
+    pages) and still produce exact greedy tokens."""
+    rng = random.Random(9)
+    eng = make(num_pages=18, max_prefill_tokens=16, max_waiting=64)
+    sp = SamplingParams(max_tokens=12, ignore_eos=True)
+    prompts = [PROMPT, PROMPT[:18], PROMPT[:14], PROMPT]
+    rids = {}
+    for p in prompts:
+        rids[eng.add_request(p, sp)] = p
+    fins = {}
+    for _ in range(2000):
+        if len(fins) == len(rids):
+            break
+        eng.step()
+        for r in list(rids):
+            if r not in fins:
+                f = eng.get_finished(r)
+                if f:
+                    fins[r] = f
+    assert len(fins) == len(rids)
+    ref = make(prefix_cache=False, num_pages=128)
+    for r, p in rids.items():
+        want = ref.generate([p], sp)[0]
+        assert fins[r].output_ids == want, "prefix cache broke exactness"
+    assert eng.sched.n_preempted > 0  # churn really happened
+    assert eng.sched.cached_tokens > 0
