@@ -314,3 +314,26 @@ def test_json_mode_on_gpu():
     tok = ByteTokenizer()
     for r in rids:
         _json.loads(tok.decode(fins[r].output_ids).strip())
+
+
+def test_prefix_cache_on_gpu():
+    """Prefix-cached admission runs the HIP chunked-prefill kernels with a
+    shared-block-table history.  Structural assertions only: cross-shape
+    bf16 prefill numerics can legitimately flip greedy near-ties, so token
+    equality with the uncached engine is a CPU(fp32) property
+    (tests/test_prefix_cache.py)."""
+    cfg = CONFIGS["tiny"]
+    eng = LLMEngine(cfg, device="cuda", page_size=4, num_pages=128,
+                    max_num_seqs=4, enable_graphs=True, seed=3,
+                    prefix_cache=True)
+    prompt = list(range(40, 62))
+    sp = SamplingParams(max_tokens=6, ignore_eos=True)
+    out1 = eng.generate([prompt], sp)[0]
+    pt_cold = eng.metrics["prefill_tokens"]
+    out2 = eng.generate([prompt], sp)[0]
+    assert eng.sched.cached_tokens == 20          # 5 of 6 pages shared
+    assert eng.metrics["prefill_tokens"] == pt_cold + 2
+    assert out1 == out2                           # warm hit is deterministic
+    assert len(out2) == 6
+    alloc = eng.sched.alloc
+    assert len(alloc.refs) == len(eng.sched._cache)  # only cache refs remain
