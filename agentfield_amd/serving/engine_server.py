@@ -328,6 +328,9 @@ def main():
     ap.add_argument("--spec-lookup", type=int, default=0,
                     help="prompt-lookup speculative decoding draft length "
                          "(greedy-exact; 0 disables)")
+    ap.add_argument("--prefix-cache", action="store_true",
+                    help="share paged KV across requests with a common "
+                         "prompt prefix (refcounted pages, LRU eviction)")
     args = ap.parse_args()
 
     device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
@@ -340,7 +343,8 @@ def main():
     eng = LLMEngine(cfg, device=device,
                     max_num_seqs=kw.pop("max_num_seqs", args.max_num_seqs),
                     enable_graphs=not args.no_graphs and device.startswith("cuda"),
-                    spec_lookup=args.spec_lookup, **kw)
+                    spec_lookup=args.spec_lookup,
+                    prefix_cache=args.prefix_cache, **kw)
     runner = EngineRunner(eng, load_tokenizer())
     app = create_engine_app(runner, args.model)
     import uvicorn
