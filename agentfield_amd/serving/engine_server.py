@@ -51,7 +51,7 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
     @app.get("/v1/stats")
     async def stats():
         eng = runner.engine
-        return {
+        out = {
             "model": model_name,
             "queued": eng.sched.num_queued(),
             "running": eng.sched.num_running(),
@@ -59,6 +59,10 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
             "kv_total_pages": eng.sched.alloc.num_pages,
             **eng.metrics,
         }
+        if hasattr(eng.sched, "cached_tokens"):  # prefix cache enabled
+            out["prefix_cached_tokens"] = eng.sched.cached_tokens
+            out["prefix_cache_pages"] = len(eng.sched._cache)
+        return out
 
     @app.post("/v1/generate")
     async def generate(req: Request):
