@@ -20,13 +20,28 @@ from ..models import CONFIGS
 from ..sdk.ai import EngineRunner, load_tokenizer
 
 
-def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
+def create_engine_app(runner: EngineRunner, model_name: str,
+                      extra_models: dict[str, EngineRunner] | None = None
+                      ) -> FastAPI:
+    """One HTTP server over one or more in-process engines.  288 GB of
+    HBM3E comfortably co-resides several model families (8B + 70B fit
+    together), so a replica can serve multiple models; requests route by
+    their `model` field, defaulting to the primary."""
+    runners: dict[str, EngineRunner] = {model_name: runner,
+                                        **(extra_models or {})}
     app = FastAPI(title=f"agentfield-engine:{model_name}")
     app.state.runner = runner
 
+    def pick(body: dict) -> tuple[EngineRunner, str]:
+        m = body.get("model")
+        if m and m in runners:
+            return runners[m], m
+        return runners[model_name], model_name
+
     @app.get("/v1/health")
     async def health():
-        return {"status": "healthy", "model": model_name}
+        return {"status": "healthy", "model": model_name,
+                "models": list(runners)}
 
     @app.get("/metrics")
     async def metrics():
@@ -62,11 +77,18 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
         if hasattr(eng.sched, "cached_tokens"):  # prefix cache enabled
             out["prefix_cached_tokens"] = eng.sched.cached_tokens
             out["prefix_cache_pages"] = len(eng.sched._cache)
+        if len(runners) > 1:
+            out["models"] = {
+                name: {"queued": r.engine.sched.num_queued(),
+                       "running": r.engine.sched.num_running(),
+                       **r.engine.metrics}
+                for name, r in runners.items()}
         return out
 
     @app.post("/v1/generate")
     async def generate(req: Request):
         body = await req.json()
+        runner, _mname = pick(body)
         sp = SamplingParams(
             max_tokens=int(body.get("max_tokens", 128)),
             temperature=float(body.get("temperature", 0.0)),
@@ -146,12 +168,12 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
     # protocol (including SSE chunk streaming + [DONE]), so any OpenAI
     # client library can point at an engine replica directly — the serving
     # analog of the reference's LiteLLM provider seam.
-    def _assemble_chat_prompt(messages) -> str:
+    def _assemble_chat_prompt(messages, mname: str = None) -> str:
         from ..sdk.multimodal import require_text
         parts = []
         for m in messages:
             role = m.get("role", "user")
-            content = require_text(m.get("content", ""), model_name)
+            content = require_text(m.get("content", ""), mname or model_name)
             parts.append(f"<|{role}|>\n{content}")
         parts.append("<|assistant|>\n")
         return "\n".join(parts)
@@ -166,6 +188,7 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
         import uuid
 
         import anyio
+        runner, _mname = pick(body)
         max_tokens = int(body.get("max_tokens", 128))
         # chat: logprobs=true + top_logprobs=N; completions: logprobs=N
         if kind == "chat":
@@ -184,7 +207,7 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
         ids = runner.tokenizer.encode(prompt)
         rid = f"{'chatcmpl' if kind == 'chat' else 'cmpl'}-{uuid.uuid4().hex[:24]}"
         created = int(_time.time())
-        base = {"id": rid, "created": created, "model": model_name}
+        base = {"id": rid, "created": created, "model": _mname}
 
         if not body.get("stream", False):
             nchoice = max(1, min(int(body.get("n", 1)), 16))
@@ -299,7 +322,8 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
     @app.post("/v1/chat/completions")
     async def chat_completions(req: Request):
         body = await req.json()
-        prompt = _assemble_chat_prompt(body.get("messages", []))
+        prompt = _assemble_chat_prompt(body.get("messages", []),
+                                       pick(body)[1])
         return await _oai_generate(body, prompt, "chat")
 
     @app.post("/v1/completions")
@@ -314,16 +338,20 @@ def create_engine_app(runner: EngineRunner, model_name: str) -> FastAPI:
     async def models():
         import time as _time
         return {"object": "list",
-                "data": [{"id": model_name, "object": "model",
+                "data": [{"id": name, "object": "model",
                           "created": int(_time.time()),
-                          "owned_by": "agentfield_amd"}]}
+                          "owned_by": "agentfield_amd"}
+                         for name in runners]}
 
     return app
 
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--model", default="llama-3-8b")
+    ap.add_argument("--model", action="append", default=None,
+                    help="model to serve (repeatable: the first is the "
+                         "default, extras co-reside in HBM and route by "
+                         "the request's model field)")
     ap.add_argument("--device", default=None)
     ap.add_argument("--host", default="127.0.0.1")
     ap.add_argument("--port", type=int, default=8710)
@@ -340,17 +368,26 @@ def main():
     device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
     if device.startswith("cuda:"):
         torch.cuda.set_device(device)
-    cfg = CONFIGS[args.model]
-    kw = {}
-    if not device.startswith("cuda"):
-        kw = {"num_pages": 512, "max_num_seqs": 8, "dtype": torch.float32}
-    eng = LLMEngine(cfg, device=device,
-                    max_num_seqs=kw.pop("max_num_seqs", args.max_num_seqs),
-                    enable_graphs=not args.no_graphs and device.startswith("cuda"),
-                    spec_lookup=args.spec_lookup,
-                    prefix_cache=args.prefix_cache, **kw)
-    runner = EngineRunner(eng, load_tokenizer())
-    app = create_engine_app(runner, args.model)
+    models = args.model or ["llama-3-8b"]
+
+    def build(name: str) -> EngineRunner:
+        kw = {}
+        if not device.startswith("cuda"):
+            kw = {"num_pages": 512, "max_num_seqs": 8,
+                  "dtype": torch.float32}
+        eng = LLMEngine(
+            CONFIGS[name], device=device,
+            max_num_seqs=kw.pop("max_num_seqs", args.max_num_seqs),
+            enable_graphs=not args.no_graphs and device.startswith("cuda"),
+            spec_lookup=args.spec_lookup,
+            prefix_cache=args.prefix_cache, **kw)
+        return EngineRunner(eng, load_tokenizer())
+
+    runners = {name: build(name) for name in models}
+    primary = models[0]
+    app = create_engine_app(runners[primary], primary,
+                            extra_models={n: r for n, r in runners.items()
+                                          if n != primary})
     import uvicorn
     uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
 

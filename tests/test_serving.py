@@ -287,3 +287,52 @@ def test_openai_json_mode(replicas):
         "prompt": "data:", "max_tokens": 12, "temperature": 1.1,
         "json_mode": True}, timeout=60.0).json()
     _json.loads(r["text"].strip())
+
+
+def test_multi_model_single_server():
+    """One server process serving two models; requests route by the model
+    field and default to the primary."""
+    import httpx
+    from agentfield_amd.serving.engine_server import create_engine_app
+
+    def mk(vocab_seed):
+        eng = LLMEngine(CONFIGS["tiny"], device="cpu", dtype=torch.float32,
+                        page_size=4, num_pages=128, max_num_seqs=4,
+                        enable_graphs=False, seed=vocab_seed)
+        return EngineRunner(eng)
+
+    ra, rb = mk(1), mk(2)
+    srv = AppServer(create_engine_app(ra, "tiny-a",
+                                      extra_models={"tiny-b": rb})).start()
+    try:
+        url = srv.base_url
+        h = httpx.get(url + "/v1/health").json()
+        assert set(h["models"]) == {"tiny-a", "tiny-b"}
+        models = {m["id"] for m in
+                  httpx.get(url + "/v1/models").json()["data"]}
+        assert models == {"tiny-a", "tiny-b"}
+        body = {"prompt_ids": [1, 5, 9], "max_tokens": 4, "ignore_eos": True}
+        oa = httpx.post(url + "/v1/generate",
+                        json={**body, "model": "tiny-a"},
+                        timeout=60.0).json()["output_ids"]
+        ob = httpx.post(url + "/v1/generate",
+                        json={**body, "model": "tiny-b"},
+                        timeout=60.0).json()["output_ids"]
+        od = httpx.post(url + "/v1/generate", json=body,
+                        timeout=60.0).json()["output_ids"]
+        assert od == oa          # default routes to primary
+        assert oa != ob          # different weights -> different greedy
+        # per-model stats visible
+        stats = httpx.get(url + "/v1/stats").json()
+        assert set(stats["models"]) == {"tiny-a", "tiny-b"}
+        assert stats["models"]["tiny-b"]["decode_tokens"] >= 3
+        # OpenAI path routes and stamps the routed model name
+        r = httpx.post(url + "/v1/chat/completions", json={
+            "model": "tiny-b",
+            "messages": [{"role": "user", "content": "hi"}],
+            "max_tokens": 3}, timeout=60.0).json()
+        assert r["model"] == "tiny-b"
+    finally:
+        ra.shutdown()
+        rb.shutdown()
+        srv.stop()
