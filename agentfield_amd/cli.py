@@ -53,15 +53,18 @@ def server(host: str = "0.0.0.0", port: int = 8520,
 
 
 @app.command()
-def engine(model: str = "llama-3-8b", device: str = None,
+def engine(model: list[str] = ["llama-3-8b"], device: str = None,
            host: str = "127.0.0.1", port: int = 8710,
            max_num_seqs: int = 256, spec_lookup: int = 0,
            prefix_cache: bool = False):
-    """Run one GPU engine server (start one per GPU for DP)."""
+    """Run one GPU engine server (start one per GPU for DP; repeat
+    --model to co-serve several models from one replica)."""
     from .serving.engine_server import main as engine_main
-    argv = ["--model", model, "--host", host, "--port", str(port),
+    argv = ["--host", host, "--port", str(port),
             "--max-num-seqs", str(max_num_seqs),
             "--spec-lookup", str(spec_lookup)]
+    for m in model:
+        argv += ["--model", m]
     if prefix_cache:
         argv.append("--prefix-cache")
     if device:
