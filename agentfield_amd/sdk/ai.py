@@ -106,6 +106,16 @@ class EngineRunner:
         self._wake.set()
         return waiter
 
+    def _check_json_mode(self, cfg: AIConfig) -> bool:
+        """Grammar-constrained decoding masks BYTE token ids: exact for
+        the byte tokenizer, nonsense for multi-byte HF vocabularies
+        (token-level grammar compilation is round-2) — fail loudly."""
+        if cfg.json_only and not isinstance(self.tokenizer, ByteTokenizer):
+            raise ValueError(
+                "json_only requires the byte tokenizer; the configured HF "
+                "tokenizer needs token-level grammar support (roadmap)")
+        return cfg.json_only
+
     def generate_text(self, prompt: str, cfg: AIConfig) -> str:
         ids = self.tokenizer.encode(prompt)
         limit = self.engine.cfg.max_position - 8
@@ -113,7 +123,7 @@ class EngineRunner:
             ids = ids[-(limit - cfg.max_tokens):]  # token-aware trim
         sp = SamplingParams(max_tokens=cfg.max_tokens,
                             temperature=cfg.temperature,
-                            json_mode=cfg.json_only)
+                            json_mode=self._check_json_mode(cfg))
         w = self.submit(ids, sp)
         if not w["done"].wait(cfg.timeout):
             raise TimeoutError("engine generate timed out")
@@ -130,7 +140,7 @@ class EngineRunner:
         ids = self.tokenizer.encode(prompt)
         sp = SamplingParams(max_tokens=cfg.max_tokens,
                             temperature=cfg.temperature,
-                            json_mode=cfg.json_only)
+                            json_mode=self._check_json_mode(cfg))
         sq: queue.Queue = queue.Queue()
         w = self.submit(ids, sp, stream_q=sq)
         stops = tuple(s for s in (cfg.stop or ()) if s)

@@ -17,7 +17,11 @@ from fastapi.responses import JSONResponse, StreamingResponse
 
 from ..engine import LLMEngine, SamplingParams
 from ..models import CONFIGS
-from ..sdk.ai import EngineRunner, load_tokenizer
+from ..sdk.ai import ByteTokenizer, EngineRunner, load_tokenizer
+
+
+def _byte_tok(runner: EngineRunner) -> bool:
+    return isinstance(runner.tokenizer, ByteTokenizer)
 
 
 def create_engine_app(runner: EngineRunner, model_name: str,
@@ -89,12 +93,17 @@ def create_engine_app(runner: EngineRunner, model_name: str,
     async def generate(req: Request):
         body = await req.json()
         runner, _mname = pick(body)
+        jm = bool(body.get("json_mode", False))
+        if jm and not _byte_tok(runner):
+            return JSONResponse(
+                {"error": "json_mode requires the byte tokenizer"},
+                status_code=400)
         sp = SamplingParams(
             max_tokens=int(body.get("max_tokens", 128)),
             temperature=float(body.get("temperature", 0.0)),
             ignore_eos=bool(body.get("ignore_eos", False)),
             logprobs=min(int(body.get("logprobs", 0) or 0), 8),
-            json_mode=bool(body.get("json_mode", False)))
+            json_mode=jm)
         if "prompt_ids" in body:
             ids = [int(x) for x in body["prompt_ids"]]
         else:
@@ -197,10 +206,17 @@ def create_engine_app(runner: EngineRunner, model_name: str,
         else:
             lp_n = min(int(body.get("logprobs", 0) or 0), 8)
         rf = body.get("response_format") or {}
+        jm = rf.get("type") == "json_object"
+        if jm and not _byte_tok(runner):
+            return JSONResponse(
+                {"error": {"message": "response_format json_object "
+                           "requires the byte tokenizer",
+                           "type": "invalid_request_error"}},
+                status_code=400)
         sp = SamplingParams(max_tokens=max_tokens,
                             temperature=float(body.get("temperature", 0.0)),
                             logprobs=lp_n,
-                            json_mode=rf.get("type") == "json_object")
+                            json_mode=jm)
         stop_in = body.get("stop") or []
         stops = tuple(s for s in ([stop_in] if isinstance(stop_in, str)
                                   else stop_in) if s)
