@@ -71,8 +71,17 @@ class LLMEngine:
             max_num_seqs=max_num_seqs, max_prefill_tokens=max_prefill_tokens,
             page_size=page_size, num_pages=num_pages, max_waiting=max_waiting)
         if prefix_cache:
-            from .prefix_cache import PrefixCachingScheduler
-            self.sched = PrefixCachingScheduler(sched_cfg)
+            # Python impl is the round-1 default (GPU-validated); the C++
+            # NativeScheduler's prefix mode is lockstep-pinned to it
+            # (tests/test_native_scheduler.py) and opted in via env.
+            import os
+            if os.environ.get("AF_NATIVE_PREFIX") == "1":
+                from .scheduler import NativeSchedulerAdapter
+                self.sched = NativeSchedulerAdapter(sched_cfg,
+                                                    prefix_cache=True)
+            else:
+                from .prefix_cache import PrefixCachingScheduler
+                self.sched = PrefixCachingScheduler(sched_cfg)
         else:
             self.sched = make_scheduler(sched_cfg)
         self.page_size = page_size

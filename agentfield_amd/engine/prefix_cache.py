@@ -22,6 +22,16 @@ from .scheduler import ScheduleBatch, Scheduler, SchedulerConfig
 from .sequence import Sequence, SeqStatus
 
 
+def prefix_hashes(prompt: list[int], page_size: int) -> list[int]:
+    """Chained content hash per FULL prompt page (shared by the Python
+    scheduler and the C++ NativeScheduler's prefix mode)."""
+    out, h = [], 0
+    for i in range(len(prompt) // page_size):
+        h = hash((h, tuple(prompt[i * page_size:(i + 1) * page_size])))
+        out.append(h)
+    return out
+
+
 class RefcountAllocator:
     """Free-list allocator with refcounted sharing.  Page 0 stays reserved
     (null page for padded graph lanes)."""
@@ -67,12 +77,7 @@ class PrefixCachingScheduler(Scheduler):
 
     # ------------------------------------------------------------ hashing
     def _prefix_hashes(self, prompt: list[int]) -> list[int]:
-        ps = self.cfg.page_size
-        out, h = [], 0
-        for i in range(len(prompt) // ps):
-            h = hash((h, tuple(prompt[i * ps:(i + 1) * ps])))
-            out.append(h)
-        return out
+        return prefix_hashes(prompt, self.cfg.page_size)
 
     # ------------------------------------------------------------- cache
     def _evict_one(self) -> bool:
@@ -129,13 +134,16 @@ class PrefixCachingScheduler(Scheduler):
             ntok = len(prompt) - cached_tok
             if batch and tokens + ntok > cfg.max_prefill_tokens:
                 break
-            need = self._pages_needed(len(prompt)) - len(matched)
-            if not self._ensure_free(need):
-                break
-            self.waiting.popleft()
+            # PIN matched pages before any eviction: _ensure_free must not
+            # reclaim a page this candidate is about to share
             for p in matched:
                 self.alloc.ref(p)
                 self._cache.move_to_end(self._page_hash[p])
+            need = self._pages_needed(len(prompt)) - len(matched)
+            if not self._ensure_free(need):
+                self.alloc.free(matched)  # unpin; candidate stays queued
+                break
+            self.waiting.popleft()
             cand.pages = matched + self.alloc.alloc(need)
             cand.cached_prefix = cached_tok
             cand.alloc_epoch += 1

@@ -177,21 +177,34 @@ class NativeSchedulerAdapter:
     Keeps the Python Sequence objects; page/queue decisions run natively.
     Semantics are pinned to the Python Scheduler by test_native_scheduler."""
 
-    def __init__(self, cfg: SchedulerConfig):
+    def __init__(self, cfg: SchedulerConfig, prefix_cache: bool = False):
         from agentfield_amd._native import NativeScheduler
         self.cfg = cfg
+        self.prefix_cache = prefix_cache
         self.nat = NativeScheduler(cfg.max_num_seqs, cfg.max_prefill_tokens,
                                    cfg.page_size, cfg.num_pages,
-                                   cfg.max_waiting)
+                                   cfg.max_waiting, prefix_cache)
         self.seqs: dict[int, Sequence] = {}
         self.alloc = _AllocShim(self.nat, cfg.num_pages)
+
+    @property
+    def cached_tokens(self) -> int:
+        return self.nat.cached_tokens() if self.prefix_cache else 0
+
+    @property
+    def cache_hits(self) -> int:
+        return self.nat.cache_hits() if self.prefix_cache else 0
 
     @property
     def n_preempted(self):
         return self.nat.n_preempted()
 
     def add(self, seq: Sequence) -> bool:
-        if not self.nat.add(seq.seq_id, len(seq.prompt_ids)):
+        hashes = []
+        if self.prefix_cache:
+            from .prefix_cache import prefix_hashes
+            hashes = prefix_hashes(seq.prompt_ids, self.cfg.page_size)
+        if not self.nat.add(seq.seq_id, len(seq.prompt_ids), hashes):
             return False
         self.seqs[seq.seq_id] = seq
         return True
@@ -224,6 +237,8 @@ class NativeSchedulerAdapter:
             s.status = SeqStatus.RUNNING
             if r.is_prefill:
                 s.alloc_epoch += 1  # fresh allocation: invalidate bt rows
+                if self.prefix_cache:
+                    s.cached_prefix = self.nat.cached_prefix(sid)
             batch.append(s)
         return ScheduleBatch(is_prefill=r.is_prefill, seqs=batch)
 

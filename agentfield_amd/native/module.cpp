@@ -10,6 +10,7 @@
 #include <openssl/rand.h>
 
 #include <deque>
+#include <list>
 #include <stdexcept>
 #include <string>
 #include <unordered_map>
@@ -121,7 +122,9 @@ static py::bytes aes_gcm_decrypt(py::bytes key, py::bytes blob) {
 struct SeqState {
   int prompt_len = 0;
   int num_tokens = 0;   // prompt + generated
+  int cached_prefix = 0;  // prompt tokens served from the prefix cache
   std::vector<int> pages;
+  std::vector<long> hashes;  // chained per-full-page prompt hashes
 };
 
 struct ScheduleResult {
@@ -133,18 +136,24 @@ struct ScheduleResult {
 
 class NativeScheduler {
  public:
+  // prefix_cache=true enables refcounted shared prompt pages keyed by the
+  // chained per-page hashes the Python side computes from prompt tokens
+  // (semantics lockstep-pinned to engine/prefix_cache.py).
   NativeScheduler(int max_num_seqs, int max_prefill_tokens, int page_size,
-                  int num_pages, int max_waiting)
+                  int num_pages, int max_waiting, bool prefix_cache = false)
       : max_num_seqs_(max_num_seqs), max_prefill_tokens_(max_prefill_tokens),
-        page_size_(page_size), num_pages_(num_pages), max_waiting_(max_waiting) {
+        page_size_(page_size), num_pages_(num_pages),
+        max_waiting_(max_waiting), prefix_cache_(prefix_cache) {
     for (int p = num_pages - 1; p >= 1; --p) free_list_.push_back(p);
   }
 
-  bool add(long seq_id, int prompt_len) {
+  bool add(long seq_id, int prompt_len,
+           const std::vector<long>& page_hashes = {}) {
     if ((int)waiting_.size() >= max_waiting_) return false;
     SeqState st;
     st.prompt_len = prompt_len;
     st.num_tokens = prompt_len;
+    st.hashes = page_hashes;
     seqs_[seq_id] = st;
     waiting_.push_back(seq_id);
     return true;
@@ -168,7 +177,11 @@ class NativeScheduler {
 
   void note_token(long seq_id) {  // engine appended one generated token
     auto it = seqs_.find(seq_id);
-    if (it != seqs_.end()) it->second.num_tokens += 1;
+    if (it == seqs_.end()) return;
+    SeqState& st = it->second;
+    if (prefix_cache_ && st.num_tokens == st.prompt_len)
+      register_pages(st);  // prompt fully prefilled: publish full pages
+    st.num_tokens += 1;
   }
 
   void finish(long seq_id) {
@@ -182,6 +195,14 @@ class NativeScheduler {
     seqs_.erase(it);
   }
 
+  int cached_prefix(long seq_id) const {
+    auto it = seqs_.find(seq_id);
+    return it == seqs_.end() ? 0 : it->second.cached_prefix;
+  }
+  long cache_hits() const { return cache_hits_; }
+  long cached_tokens() const { return cached_tokens_; }
+  int cache_pages() const { return (int)lru_.size(); }
+
   ScheduleResult schedule() {
     ScheduleResult r;
     // 1) admit prefills
@@ -191,12 +212,36 @@ class NativeScheduler {
            (int)(running_.size() + batch.size()) < max_num_seqs_) {
       long cand = waiting_.front();
       SeqState& st = seqs_[cand];
-      int ntok = st.prompt_len;
+      // cached full-page prefix (>=1 prompt token always recomputed)
+      std::vector<int> matched;
+      if (prefix_cache_) {
+        int max_full = (st.prompt_len - 1) / page_size_;
+        for (int k = 0; k < (int)st.hashes.size() && k < max_full; ++k) {
+          auto it = cache_.find(st.hashes[k]);
+          if (it == cache_.end()) break;
+          matched.push_back(it->second->second);
+        }
+      }
+      int cached_tok = (int)matched.size() * page_size_;
+      int ntok = st.prompt_len - cached_tok;
       if (!batch.empty() && tokens + ntok > max_prefill_tokens_) break;
-      int need = pages_needed(ntok);
-      if (need > (int)free_list_.size()) break;
+      // PIN matched pages before eviction can touch them (lockstep with
+      // the Python oracle's fix for the same hazard)
+      for (size_t k = 0; k < matched.size(); ++k) {
+        refs_[matched[k]] += 1;
+        lru_touch(st.hashes[k]);
+      }
+      int need = pages_needed(st.prompt_len) - (int)matched.size();
+      if (!ensure_free(need)) {
+        for (int p : matched) unref(p);  // unpin; candidate stays queued
+        break;
+      }
       waiting_.pop_front();
+      st.pages = matched;
       alloc_into(st, need);
+      st.cached_prefix = cached_tok;
+      cache_hits_ += (long)matched.size();
+      cached_tokens_ += cached_tok;
       batch.push_back(cand);
       tokens += ntok;
     }
@@ -241,23 +286,77 @@ class NativeScheduler {
 
   void alloc_into(SeqState& st, int n) {
     for (int j = 0; j < n; ++j) {
-      st.pages.push_back(free_list_.back());
+      int p = free_list_.back();
       free_list_.pop_back();
+      refs_[p] = 1;
+      st.pages.push_back(p);
     }
   }
 
   bool grow(SeqState& st) {
     int need = pages_needed(st.num_tokens + 1);
     if (need > (int)st.pages.size()) {
-      if (free_list_.empty()) return false;
+      if (!ensure_free(1)) return false;
       alloc_into(st, 1);
     }
     return true;
   }
 
   void release(SeqState& st) {
-    for (int p : st.pages) free_list_.push_back(p);
+    for (int p : st.pages) unref(p);
     st.pages.clear();
+    st.cached_prefix = 0;
+  }
+
+  void unref(int p) {
+    auto it = refs_.find(p);
+    if (it == refs_.end()) return;  // pre-refcount pages (never happens)
+    if (--it->second == 0) {
+      refs_.erase(it);
+      free_list_.push_back(p);
+    }
+  }
+
+  bool ensure_free(int need) {
+    while ((int)free_list_.size() < need) {
+      if (!evict_one()) return false;
+    }
+    return true;
+  }
+
+  bool evict_one() {
+    // LRU-first cached page that only the cache itself references
+    for (auto it = lru_.begin(); it != lru_.end(); ++it) {
+      auto rit = refs_.find(it->second);
+      if (rit != refs_.end() && rit->second == 1) {
+        page_hash_.erase(it->second);
+        cache_.erase(it->first);
+        unref(it->second);
+        lru_.erase(it);
+        return true;
+      }
+    }
+    return false;
+  }
+
+  void lru_touch(long h) {
+    auto it = cache_.find(h);
+    if (it == cache_.end()) return;
+    lru_.splice(lru_.end(), lru_, it->second);  // move to MRU end
+  }
+
+  void register_pages(SeqState& st) {
+    for (size_t k = 0; k < st.hashes.size() && k < st.pages.size(); ++k) {
+      long h = st.hashes[k];
+      if (cache_.count(h)) {
+        lru_touch(h);
+        continue;
+      }
+      lru_.emplace_back(h, st.pages[k]);
+      cache_[h] = std::prev(lru_.end());
+      page_hash_[st.pages[k]] = h;
+      refs_[st.pages[k]] += 1;
+    }
   }
 
   void preempt(long sid, ScheduleResult& r) {
@@ -270,11 +369,18 @@ class NativeScheduler {
   }
 
   int max_num_seqs_, max_prefill_tokens_, page_size_, num_pages_, max_waiting_;
+  bool prefix_cache_ = false;
   std::vector<int> free_list_;
   std::deque<long> waiting_;
   std::vector<long> running_;
   std::unordered_map<long, SeqState> seqs_;
   long n_preempted_ = 0;
+  // prefix cache state: LRU list of (hash, page); maps for O(1) lookups
+  std::list<std::pair<long, int>> lru_;
+  std::unordered_map<long, std::list<std::pair<long, int>>::iterator> cache_;
+  std::unordered_map<int, long> page_hash_;
+  std::unordered_map<int, int> refs_;
+  long cache_hits_ = 0, cached_tokens_ = 0;
 };
 
 PYBIND11_MODULE(_native, m) {
@@ -292,10 +398,17 @@ PYBIND11_MODULE(_native, m) {
       .def_readonly("preempted", &ScheduleResult::preempted);
 
   py::class_<NativeScheduler>(m, "NativeScheduler")
-      .def(py::init<int, int, int, int, int>(), py::arg("max_num_seqs"),
+      .def(py::init<int, int, int, int, int, bool>(), py::arg("max_num_seqs"),
            py::arg("max_prefill_tokens"), py::arg("page_size"),
-           py::arg("num_pages"), py::arg("max_waiting") = 4096)
-      .def("add", &NativeScheduler::add)
+           py::arg("num_pages"), py::arg("max_waiting") = 4096,
+           py::arg("prefix_cache") = false)
+      .def("add", &NativeScheduler::add, py::arg("seq_id"),
+           py::arg("prompt_len"),
+           py::arg("page_hashes") = std::vector<long>{})
+      .def("cached_prefix", &NativeScheduler::cached_prefix)
+      .def("cache_hits", &NativeScheduler::cache_hits)
+      .def("cached_tokens", &NativeScheduler::cached_tokens)
+      .def("cache_pages", &NativeScheduler::cache_pages)
       .def("schedule", &NativeScheduler::schedule)
       .def("finish", &NativeScheduler::finish)
       .def("note_token", &NativeScheduler::note_token)

@@ -2,6 +2,8 @@
 exactness, refcount hygiene, eviction, and preemption interplay."""
 import random
 
+import pytest
+
 import torch
 
 from agentfield_amd.engine import LLMEngine, SamplingParams
@@ -85,3 +87,18 @@ def test_prefix_cache_with_preemption_exact():
         assert fins[r].output_ids == want, "prefix cache broke exactness"
     assert eng.sched.n_preempted > 0  # churn really happened
     assert eng.sched.cached_tokens > 0
+
+
+def test_native_prefix_engine_end_to_end(monkeypatch):
+    """The C++ prefix mode drives a full engine run: same exact tokens as
+    the Python-scheduler engine, cache hits recorded."""
+    pytest.importorskip("agentfield_amd._native")
+    ref = make(prefix_cache=False)
+    base = [ref.generate([PROMPT], SP)[0], ref.generate([PROMPT], SP)[0]]
+    monkeypatch.setenv("AF_NATIVE_PREFIX", "1")
+    eng = make()
+    from agentfield_amd.engine.scheduler import NativeSchedulerAdapter
+    assert isinstance(eng.sched, NativeSchedulerAdapter)
+    got = [eng.generate([PROMPT], SP)[0], eng.generate([PROMPT], SP)[0]]
+    assert got == base
+    assert eng.sched.cached_tokens == 20  # second request hit 5 pages
