@@ -75,6 +75,10 @@ class PrefixCachingScheduler(Scheduler):
         self.cache_hits = 0       # pages served from cache
         self.cached_tokens = 0    # prompt tokens whose prefill was skipped
 
+    @property
+    def cache_pages(self) -> int:
+        return len(self._cache)
+
     # ------------------------------------------------------------ hashing
     def _prefix_hashes(self, prompt: list[int]) -> list[int]:
         return prefix_hashes(prompt, self.cfg.page_size)

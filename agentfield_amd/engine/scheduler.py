@@ -196,6 +196,10 @@ class NativeSchedulerAdapter:
         return self.nat.cache_hits() if self.prefix_cache else 0
 
     @property
+    def cache_pages(self) -> int:
+        return self.nat.cache_pages() if self.prefix_cache else 0
+
+    @property
     def n_preempted(self):
         return self.nat.n_preempted()
 

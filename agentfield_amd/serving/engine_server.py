@@ -78,9 +78,10 @@ def create_engine_app(runner: EngineRunner, model_name: str,
             "kv_total_pages": eng.sched.alloc.num_pages,
             **eng.metrics,
         }
-        if hasattr(eng.sched, "cached_tokens"):  # prefix cache enabled
-            out["prefix_cached_tokens"] = eng.sched.cached_tokens
-            out["prefix_cache_pages"] = len(eng.sched._cache)
+        ct = getattr(eng.sched, "cached_tokens", None)
+        if ct is not None:  # prefix-cache-capable scheduler
+            out["prefix_cached_tokens"] = ct
+            out["prefix_cache_pages"] = getattr(eng.sched, "cache_pages", 0)
         if len(runners) > 1:
             out["models"] = {
                 name: {"queued": r.engine.sched.num_queued(),
