@@ -890,6 +890,22 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
                 cp.buses.node.unsubscribe(sid)
         return StreamingResponse(gen(), media_type="text/event-stream")
 
+    @app.get("/api/ui/v1/config")
+    async def ui_config():
+        """Sanitized server configuration (reference C33 env/config
+        handlers): secrets and key material are never exposed."""
+        cfg = cp.cfg
+        hidden = {"keystore_key", "webhook_secret"}
+        out = {}
+        for k, v in vars(cfg).items():
+            if k.startswith("_") or any(s in k.lower() for s in
+                                        ("secret", "key", "token", "pass")):
+                continue
+            if isinstance(v, (str, int, float, bool, type(None))):
+                out[k] = v
+        return {"config": out, "hidden_fields": sorted(
+            k for k in vars(cfg) if k not in out and not k.startswith("_"))}
+
     @app.get("/api/ui/v1/reasoners/events")
     async def reasoner_sse():
         sid, q = cp.buses.reasoner.subscribe()

@@ -661,3 +661,11 @@ def test_sdk_action_handler_roundtrip(cp_server, greeting_agent):
     assert by["rotate-keys"]["status"] == "acked"
     assert by["rotate-keys"]["ack_status"] == "done"
     assert by["unknown-action"]["ack_status"] == "ignored"
+
+
+def test_ui_config_sanitized(cp_server):
+    srv, _ = cp_server
+    r = httpx.get(srv.base_url + "/api/ui/v1/config").json()
+    assert "sync_timeout" in r["config"]
+    joined = json.dumps(r["config"]).lower()
+    assert "secret" not in joined and "keystore" not in joined
