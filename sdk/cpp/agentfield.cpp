@@ -574,7 +574,11 @@ void Agent::heartbeat_loop() {
     auto resp = http_request(
         "POST", cfg_.agentfield_url + "/api/v1/nodes/" + cfg_.node_id +
                     "/heartbeat", hb.dump());
-    if (resp.status == 404) do_register();  // resilient re-register
+    if (resp.status == 404) {
+      do_register();  // resilient re-register
+    } else if (resp.status == 200) {
+      drain_actions();  // claim/ack lease queue (G2)
+    }
   }
 }
 
@@ -632,6 +636,79 @@ std::string Agent::ai(const std::string& prompt, int max_tokens,
   if (resp.status != 200)
     throw std::runtime_error("engine error: " + resp.body);
   return Json::parse(resp.body).get_str("text");
+}
+
+std::string Agent::chat(
+    const std::vector<std::pair<std::string, std::string>>& messages,
+    int max_tokens, double temperature) {
+  std::string url = cfg_.engine_url;
+  if (url.empty()) {
+    const char* env = getenv("AGENTFIELD_ENGINE_URLS");
+    if (env) {
+      url = env;
+      size_t comma = url.find(',');
+      if (comma != std::string::npos) url = url.substr(0, comma);
+    }
+  }
+  if (url.empty()) throw std::runtime_error("no engine url configured");
+  Json body = Json::object();
+  Json msgs = Json::array();
+  for (auto& m : messages) {
+    Json one = Json::object();
+    one["role"] = m.first;
+    one["content"] = m.second;
+    msgs.arr.push_back(one);
+  }
+  body["messages"] = msgs;
+  body["max_tokens"] = max_tokens;
+  body["temperature"] = temperature;
+  auto resp = http_request("POST", url + "/v1/chat/completions", body.dump(),
+                           {}, 600000);
+  if (resp.status != 200)
+    throw std::runtime_error("engine error: " + resp.body);
+  Json doc = Json::parse(resp.body);
+  const Json* choices = doc.find("choices");
+  if (!choices || choices->arr.empty()) return "";
+  const Json* msg = choices->arr[0].find("message");
+  return msg ? msg->get_str("content") : "";
+}
+
+void Agent::on_action(const std::string& name,
+                      std::function<void(const Json&)> fn) {
+  action_handlers_[name] = fn;
+}
+
+void Agent::drain_actions() {
+  Json req = Json::object();
+  req["lease_s"] = 30.0;
+  auto resp = http_request(
+      "POST", cfg_.agentfield_url + "/api/v1/nodes/" + cfg_.node_id +
+                  "/actions/claim", req.dump());
+  if (resp.status != 200) return;
+  Json doc = Json::parse(resp.body);  // keep alive: find() borrows
+  const Json* acts = doc.find("actions");
+  if (!acts) return;
+  for (const Json& a : acts->arr) {
+    std::string action = a.get_str("action");
+    std::string status = "done";
+    if (action == "stop" || action == "shutdown") {
+      stopping_ = true;
+    } else {
+      auto it = action_handlers_.find(action);
+      if (it != action_handlers_.end()) {
+        const Json* payload = a.find("payload");
+        Json empty = Json::object();
+        it->second(payload ? *payload : empty);
+      } else {
+        status = "ignored";
+      }
+    }
+    Json ack = Json::object();
+    ack["action_id"] = a.get_num("id");
+    ack["status"] = status;
+    http_request("POST", cfg_.agentfield_url + "/api/v1/nodes/" +
+                             cfg_.node_id + "/actions/ack", ack.dump());
+  }
 }
 
 }  // namespace agentfield

@@ -122,17 +122,30 @@ class Agent {
   std::string ai(const std::string& prompt, int max_tokens = 128,
                  double temperature = 0.0);
 
+  // OpenAI-style chat completion against the engine fleet
+  // (/v1/chat/completions); messages are (role, content) pairs.
+  // Returns the assistant message content ("" on error).
+  std::string chat(
+      const std::vector<std::pair<std::string, std::string>>& messages,
+      int max_tokens = 128, double temperature = 0.0);
+
+  // Handle a named control-plane action delivered through the claim/ack
+  // lease queue (reference G2 AcknowledgeAction; drained each heartbeat).
+  void on_action(const std::string& name, std::function<void(const Json&)> fn);
+
   bool registered() const { return registered_; }
 
  private:
   bool do_register();
   void heartbeat_loop();
+  void drain_actions();
   HttpResponse handle_invoke(const std::string& kind, const std::string& name,
                              const HttpRequest& req);
 
   AgentConfig cfg_;
   std::map<std::string, ReasonerFn> reasoners_;
   std::map<std::string, ReasonerFn> skills_;
+  std::map<std::string, std::function<void(const Json&)>> action_handlers_;
   std::unique_ptr<HttpServer> server_;
   std::thread hb_thread_;
   std::atomic<bool> stopping_{false};

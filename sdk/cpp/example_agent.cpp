@@ -43,6 +43,11 @@ int main(int argc, char** argv) {
     return out;
   });
 
+  app.on_action("ping", [](const Json& payload) {
+    printf("ACTION ping %s\n", payload.dump().c_str());
+    fflush(stdout);
+  });
+
   printf("PORT=%d\n", app.port());
   fflush(stdout);
   app.run(true);

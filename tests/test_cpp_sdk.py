@@ -76,3 +76,19 @@ def test_cpp_agent_health_direct(stack):
     node = cp.storage.get_node("cppagent")
     h = httpx.get(node["base_url"] + "/health", timeout=5.0).json()
     assert h["node_id"] == "cppagent"
+
+
+def test_cpp_action_claim_ack(stack):
+    """The C++ agent drains the control plane's action lease queue on its
+    heartbeat (5 s in the example binary) and acks with the outcome."""
+    _srv, cp = stack
+    aid = cp.storage.enqueue_action("cppagent", "ping", {"n": 1})
+    bad = cp.storage.enqueue_action("cppagent", "no-handler")
+
+    def acked(i, want=None):
+        rows = cp.storage._q("SELECT * FROM node_actions WHERE id=?", (i,))
+        return rows and rows[0]["status"] == "acked" and \
+            (want is None or rows[0]["ack_status"] == want)
+
+    assert wait_until(lambda: acked(aid, "done"), timeout=15.0)
+    assert wait_until(lambda: acked(bad, "ignored"), timeout=15.0)
