@@ -76,6 +76,12 @@ def load_tokenizer(path: str | None = None):
 
             def decode(self, ids):
                 return self.tok.decode(ids, skip_special_tokens=True)
+
+            def apply_chat_template(self, messages):
+                """Model-native chat formatting (e.g. Llama-3 header
+                tokens) for the OpenAI chat endpoint."""
+                return self.tok.apply_chat_template(
+                    messages, tokenize=False, add_generation_prompt=True)
         return _HF(AutoTokenizer.from_pretrained(path))
     return ByteTokenizer()
 

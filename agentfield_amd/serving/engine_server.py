@@ -178,13 +178,20 @@ def create_engine_app(runner: EngineRunner, model_name: str,
     # protocol (including SSE chunk streaming + [DONE]), so any OpenAI
     # client library can point at an engine replica directly — the serving
     # analog of the reference's LiteLLM provider seam.
-    def _assemble_chat_prompt(messages, mname: str = None) -> str:
+    def _assemble_chat_prompt(messages, mname: str = None,
+                              rnr: EngineRunner = None) -> str:
         from ..sdk.multimodal import require_text
-        parts = []
+        flat = []
         for m in messages:
-            role = m.get("role", "user")
-            content = require_text(m.get("content", ""), mname or model_name)
-            parts.append(f"<|{role}|>\n{content}")
+            flat.append({"role": m.get("role", "user"),
+                         "content": require_text(m.get("content", ""),
+                                                 mname or model_name)})
+        tok = (rnr or runner).tokenizer
+        if hasattr(tok, "apply_chat_template"):
+            # HF tokenizer: the model's own chat template (Llama-3 header
+            # tokens etc.) instead of the generic fallback format
+            return tok.apply_chat_template(flat)
+        parts = [f"<|{m['role']}|>\n{m['content']}" for m in flat]
         parts.append("<|assistant|>\n")
         return "\n".join(parts)
 
@@ -339,8 +346,8 @@ def create_engine_app(runner: EngineRunner, model_name: str,
     @app.post("/v1/chat/completions")
     async def chat_completions(req: Request):
         body = await req.json()
-        prompt = _assemble_chat_prompt(body.get("messages", []),
-                                       pick(body)[1])
+        rnr, mname = pick(body)
+        prompt = _assemble_chat_prompt(body.get("messages", []), mname, rnr)
         return await _oai_generate(body, prompt, "chat")
 
     @app.post("/v1/completions")

@@ -336,3 +336,44 @@ def test_multi_model_single_server():
         ra.shutdown()
         rb.shutdown()
         srv.stop()
+
+
+def test_chat_uses_tokenizer_template():
+    """When the tokenizer provides apply_chat_template (HF path), the chat
+    endpoint formats messages with the MODEL'S template instead of the
+    generic fallback."""
+    import httpx
+    from agentfield_amd.sdk.ai import ByteTokenizer
+    from agentfield_amd.serving.engine_server import create_engine_app
+
+    class TemplTok(ByteTokenizer):
+        def __init__(self):
+            super().__init__()
+            self.last_encoded = None
+
+        def encode(self, text, add_bos=True):
+            self.last_encoded = text
+            return super().encode(text, add_bos)
+
+        def apply_chat_template(self, messages):
+            return "".join(f"<<{m['role']}>>{m['content']}"
+                           for m in messages) + "<<assistant>>"
+
+    eng = LLMEngine(CONFIGS["tiny"], device="cpu", dtype=torch.float32,
+                    page_size=4, num_pages=128, max_num_seqs=4,
+                    enable_graphs=False)
+    tok = TemplTok()
+    runner = EngineRunner(eng, tok)
+    srv = AppServer(create_engine_app(runner, "tiny")).start()
+    try:
+        r = httpx.post(srv.base_url + "/v1/chat/completions", json={
+            "model": "tiny",
+            "messages": [{"role": "system", "content": "be nice"},
+                         {"role": "user", "content": "hello"}],
+            "max_tokens": 3}, timeout=60.0)
+        assert r.status_code == 200
+        assert tok.last_encoded == \
+            "<<system>>be nice<<user>>hello<<assistant>>"
+    finally:
+        runner.shutdown()
+        srv.stop()
