@@ -102,3 +102,19 @@ def test_native_prefix_engine_end_to_end(monkeypatch):
     got = [eng.generate([PROMPT], SP)[0], eng.generate([PROMPT], SP)[0]]
     assert got == base
     assert eng.sched.cached_tokens == 20  # second request hit 5 pages
+
+
+def test_prefix_cache_with_speculation_exact():
+    """Prefix cache + speculative decoding combined: draft verification
+    writes KV only at positions >= prompt_len, which always land in the
+    sequence's own fresh pages (shared pages cover strictly fewer full
+    prompt pages), so the combination stays greedy-exact."""
+    rep = [5, 9, 5, 9, 5, 9, 5, 9, 5, 9, 5, 9, 5, 9, 5, 9, 5, 9, 5, 9, 5]
+    sp = SamplingParams(max_tokens=10, ignore_eos=True)
+    base = make(prefix_cache=False).generate([rep], sp)[0]
+    eng = make(spec_lookup=4)
+    out1 = eng.generate([rep], sp)[0]
+    out2 = eng.generate([rep], sp)[0]   # warm: cached prefix + speculation
+    assert out1 == base and out2 == base
+    assert eng.sched.cached_tokens > 0, "prefix cache never hit"
+    assert eng.metrics["spec_drafted"] > 0, "speculation never fired"
