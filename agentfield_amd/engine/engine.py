@@ -330,10 +330,23 @@ class LLMEngine:
         pos = getattr(seq, "_fsm_pos", 0)
         if fsm is None or pos > len(seq.output_ids):
             fsm, pos = JsonFSM(), 0
-        for tid in seq.output_ids[pos:]:
-            b = tid - 4  # ByteTokenizer offset
-            if 0 <= b < 256:
-                fsm.advance(b)
+        try:
+            for tid in seq.output_ids[pos:]:
+                b = tid - 4  # ByteTokenizer offset
+                if 0 <= b < 256:
+                    fsm.advance(b)
+        except (ValueError, AssertionError):
+            # a grammar-invalid token in history means the constraint was
+            # already violated (should be impossible; the sampler-overflow
+            # bug fixed in csrc/sampling.hip was the one known cause).
+            # Fail OPEN for this sequence rather than killing the engine
+            # loop: drop the constraint and let it finish unconstrained.
+            import logging
+            logging.getLogger("agentfield.engine").warning(
+                "seq %d: grammar-invalid token in history; disabling "
+                "json_mode for it", seq.seq_id)
+            seq.sampling.json_mode = False
+            return list(range(self.cfg.vocab_size))
         seq._fsm, seq._fsm_pos = fsm, len(seq.output_ids)
         remaining = seq.sampling.max_tokens - len(seq.output_ids)
         ids = fsm.allowed_token_ids(remaining)

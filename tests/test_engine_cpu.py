@@ -269,3 +269,22 @@ def test_json_mode_mixed_batch():
             break
     _json.loads(ByteTokenizer().decode(fins[r1].output_ids).strip())
     assert len(fins[r2].output_ids) == 10  # unconstrained ran to budget
+
+
+def test_json_mode_fails_open_on_corrupt_history():
+    """A grammar-invalid token in a sequence's history (defense-in-depth;
+    the only known cause was the fixed sampler overflow) drops the
+    constraint for that sequence instead of crashing the engine loop."""
+    eng = make_engine()
+    rid = eng.add_request([1, 5, 9], SamplingParams(max_tokens=8,
+                                                    json_mode=True,
+                                                    ignore_eos=True))
+    eng.step()  # prefill: first token, FSM synced
+    seq = eng.sched.seqs[rid] if hasattr(eng.sched, "seqs") else \
+        next(s for s in eng.sched.running if s.seq_id == rid)
+    seq.output_ids.append(4 + ord("@"))  # '@' is invalid everywhere
+    for _ in range(40):
+        eng.step()  # must not raise
+        if eng.get_finished(rid):
+            break
+    assert not seq.sampling.json_mode  # constraint was dropped
