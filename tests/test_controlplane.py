@@ -669,3 +669,34 @@ def test_ui_config_sanitized(cp_server):
     assert "sync_timeout" in r["config"]
     joined = json.dumps(r["config"]).lower()
     assert "secret" not in joined and "keystore" not in joined
+
+
+def test_ui_dashboard_data_contracts(cp_server, greeting_agent):
+    """Every endpoint+field the embedded dashboard's JS reads must exist
+    (pinned so UI and API can't drift apart)."""
+    srv, _ = cp_server
+    base = srv.base_url
+    r = httpx.post(base + "/api/v1/execute/greeter.relay",
+                   json={"input": {"name": "ui"}}, timeout=30.0).json()
+    run_id = r["run_id"]
+    html = httpx.get(base + "/").text
+    for ep in ("/api/ui/v1/dashboard/summary", "/api/ui/v1/nodes",
+               "/api/ui/v1/executions", "/api/ui/v2/workflow-runs",
+               "/api/ui/v1/config", "/api/v1/did/status",
+               "/api/ui/v1/executions/events"):
+        assert ep in html, ep
+    d = httpx.get(base + "/api/ui/v1/dashboard/summary").json()
+    assert {"nodes", "executions", "uptime_s"} <= set(d)
+    nodes = httpx.get(base + "/api/ui/v1/nodes").json()["nodes"]
+    assert {"id", "status", "reasoners", "skills",
+            "last_heartbeat"} <= set(nodes[0])
+    ex = httpx.get(base + "/api/ui/v1/executions?limit=5").json()["executions"]
+    assert {"id", "node_id", "reasoner_id", "status", "run_id"} <= set(ex[0])
+    dag = httpx.get(base + f"/api/ui/v1/workflows/{run_id}/dag").json()
+    assert {"execution_id", "node_id", "reasoner_id", "status",
+            "parent_execution_id"} <= set(dag["nodes"][0])
+    tl = httpx.get(base +
+                   f"/api/ui/v1/executions/{ex[0]['id']}/timeline").json()
+    assert all({"at", "event"} <= set(e) for e in tl["events"])
+    m = httpx.get(base + "/api/ui/v1/reasoners/greeter/greet/metrics").json()
+    assert {"executions", "success_rate", "p50_ms"} <= set(m)
