@@ -170,7 +170,7 @@ class PrefixCachingScheduler(Scheduler):
                 if self.running[-1] is seq:
                     self.running.pop()
                     self.release(seq)
-                    seq.output_ids.clear()
+                    seq.reset_generation()
                     seq.status = SeqStatus.WAITING
                     self.waiting.appendleft(seq)
                     self.n_preempted += 1

@@ -100,7 +100,7 @@ class Scheduler:
     def _preempt_last(self) -> None:
         victim = self.running.pop()
         self.release(victim)
-        victim.output_ids.clear()       # recompute from scratch on readmit
+        victim.reset_generation()       # recompute from scratch on readmit
         victim.status = SeqStatus.WAITING
         self.waiting.appendleft(victim)
         self.n_preempted += 1
@@ -140,7 +140,7 @@ class Scheduler:
                     # can't preempt self and nothing else to free: defer
                     self.running.pop()
                     self.release(seq)
-                    seq.output_ids.clear()
+                    seq.reset_generation()
                     seq.status = SeqStatus.WAITING
                     self.waiting.appendleft(seq)
                     self.n_preempted += 1
@@ -229,7 +229,7 @@ class NativeSchedulerAdapter:
         r = self.nat.schedule()
         for sid in r.preempted:
             s = self.seqs[sid]
-            s.output_ids.clear()
+            s.reset_generation()
             s.pages = []
             s.status = SeqStatus.WAITING
         if not r.has_work:

@@ -54,6 +54,13 @@ class Sequence:
     def last_token(self) -> int:
         return self.output_ids[-1] if self.output_ids else self.prompt_ids[-1]
 
+    def reset_generation(self) -> None:
+        """Preemption: drop all generated state so the sequence recomputes
+        from scratch on re-admission (tokens AND per-token logprobs — a
+        stale logprobs list would misalign with the regenerated output)."""
+        self.output_ids.clear()
+        self.logprobs = None
+
     def append(self, tok: int, eos_id: int) -> bool:
         """Append a generated token; returns True when the sequence finished."""
         self.output_ids.append(tok)

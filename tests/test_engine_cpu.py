@@ -288,3 +288,29 @@ def test_json_mode_fails_open_on_corrupt_history():
         if eng.get_finished(rid):
             break
     assert not seq.sampling.json_mode  # constraint was dropped
+
+
+def test_logprobs_survive_preemption():
+    """Preemption resets generated state; a stale logprobs list must not
+    survive into the recomputed output (found by the round-1 soak)."""
+    from agentfield_amd.models import CONFIGS as _C
+    eng = LLMEngine(_C["tiny"], device="cpu", dtype=torch.float32,
+                    page_size=4, num_pages=10, max_num_seqs=4,
+                    enable_graphs=False, max_prefill_tokens=16)
+    sp = SamplingParams(max_tokens=12, ignore_eos=True, logprobs=2)
+    rids = [eng.add_request([1, 5, 9, 20, 7, 3, 11, 2], sp)
+            for _ in range(3)]
+    fins = {}
+    for _ in range(500):
+        eng.step()
+        for r in rids:
+            if r not in fins:
+                f = eng.get_finished(r)
+                if f:
+                    fins[r] = f
+        if len(fins) == len(rids):
+            break
+    assert len(fins) == len(rids)
+    assert eng.sched.n_preempted > 0, "pool must force preemption"
+    for f in fins.values():
+        assert len(f.logprobs) == len(f.output_ids) == 12
