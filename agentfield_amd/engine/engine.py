@@ -222,10 +222,23 @@ class LLMEngine:
                 seq.finish_ns = now
                 self.sched.finish(seq)
                 self._finished[seq.seq_id] = seq
-            if seq.on_token is not None:
-                seq.on_token(tok, done)
-            events.append((seq.seq_id, tok, done))
+            if self._emit(seq, done):
+                if seq.on_token is not None:
+                    seq.on_token(tok, done)
+                events.append((seq.seq_id, tok, done))
         return events
+
+    @staticmethod
+    def _emit(seq: Sequence, done: bool) -> bool:
+        """Stream exactly-once under preemption: a preempted sequence
+        recomputes tokens it already delivered (identical for greedy); a
+        high-water mark suppresses the re-walk so streaming consumers of
+        step() events and on_token never see duplicates."""
+        pos = len(seq.output_ids)
+        if pos > getattr(seq, "_streamed", 0):
+            seq._streamed = pos
+            return True
+        return done  # terminal event always delivered
 
     def _slot(self, seq: Sequence, tok_idx: int) -> int:
         return seq.pages[tok_idx // self.page_size] * self.page_size + \
@@ -460,9 +473,10 @@ class LLMEngine:
                     seq.finish_ns = now
                     self.sched.finish(seq)
                     self._finished[seq.seq_id] = seq
-                if seq.on_token is not None:
-                    seq.on_token(tok, done)
-                events.append((seq.seq_id, tok, done))
+                if self._emit(seq, done):
+                    if seq.on_token is not None:
+                        seq.on_token(tok, done)
+                    events.append((seq.seq_id, tok, done))
                 if done:
                     break
             if seq.logprobs is not None:

@@ -131,3 +131,37 @@ def test_engine_fuzz_with_speculation():
                 f"seq {rid} diverged under speculation + churn"
             checked += 1
     assert checked > 0
+
+
+def test_stream_events_exactly_once_under_preemption():
+    """step() events (the streaming feed) never replay a token position
+    after preemption: positions are strictly increasing per sequence."""
+    rng = random.Random(11)
+    eng = LLMEngine(CONFIGS["tiny"], device="cpu", dtype=torch.float32,
+                    page_size=4, num_pages=14, max_num_seqs=4,
+                    max_prefill_tokens=16, enable_graphs=False,
+                    max_waiting=64)
+    counts = {}
+    rids = []
+    for i in range(6):
+        rid = eng.add_request([rng.randrange(500) for _ in
+                               range(rng.randint(6, 14))],
+                              SamplingParams(max_tokens=10, ignore_eos=True))
+        counts[rid] = 0
+        rids.append(rid)
+    fins = {}
+    for _ in range(1500):
+        for (rid, tok, done) in eng.step():
+            counts[rid] += 1
+        for rid in rids:
+            if rid not in fins:
+                f = eng.get_finished(rid)
+                if f:
+                    fins[rid] = f
+        if len(fins) == len(rids):
+            break
+    assert len(fins) == len(rids)
+    assert eng.sched.n_preempted > 0, "pool must force preemption"
+    for rid, f in fins.items():
+        assert counts[rid] == len(f.output_ids) == 10, \
+            f"seq {rid}: {counts[rid]} events for {len(f.output_ids)} tokens"
