@@ -392,13 +392,18 @@ class LLMEngine:
         if k <= 0:
             return []
         ctx = seq.prompt_ids + seq.output_ids
-        g = 2
-        if len(ctx) <= g:
-            return []
-        tail = ctx[-g:]
-        for i in range(len(ctx) - g - 1, -1, -1):
-            if ctx[i:i + g] == tail:
-                return ctx[i + g:i + g + k]
+        # longest-suffix match first: a 4-gram match predicts the
+        # continuation better than a bigram (acceptance is what pays;
+        # correctness never depends on draft quality)
+        for g in (4, 3, 2):
+            if len(ctx) <= g:
+                continue
+            tail = ctx[-g:]
+            for i in range(len(ctx) - g - 1, -1, -1):
+                if ctx[i:i + g] == tail:
+                    d = ctx[i + g:i + g + k]
+                    if d:
+                        return d
         return []
 
     def _step_verify(self, batch: ScheduleBatch, drafts: dict):
