@@ -77,6 +77,24 @@ CONFIGS = {
     "llama-3-70b": LlamaConfig(name="llama-3-70b", hidden_size=8192,
                                intermediate_size=28672, num_layers=80,
                                num_heads=64, num_kv_heads=8, vocab_size=128256),
+    # same architecture, different shapes: loadable from HF checkpoints
+    "llama-2-7b": LlamaConfig(name="llama-2-7b", hidden_size=4096,
+                              intermediate_size=11008, num_layers=32,
+                              num_heads=32, num_kv_heads=32,
+                              vocab_size=32000, rope_theta=10000.0,
+                              max_position=4096),
+    "llama-2-13b": LlamaConfig(name="llama-2-13b", hidden_size=5120,
+                               intermediate_size=13824, num_layers=40,
+                               num_heads=40, num_kv_heads=40,
+                               vocab_size=32000, rope_theta=10000.0,
+                               max_position=4096),
+    # NOTE: full-causal attention; Mistral's 4096 sliding window is not
+    # implemented, so outputs match the real model only for ctx <= 4096.
+    "mistral-7b": LlamaConfig(name="mistral-7b", hidden_size=4096,
+                              intermediate_size=14336, num_layers=32,
+                              num_heads=32, num_kv_heads=8,
+                              vocab_size=32000, rope_theta=10000.0,
+                              max_position=4096),
     # small configs for tests / smoke
     "tiny": LlamaConfig(name="tiny", hidden_size=256, intermediate_size=512,
                         num_layers=2, num_heads=2, num_kv_heads=1,

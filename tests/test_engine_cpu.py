@@ -314,3 +314,20 @@ def test_logprobs_survive_preemption():
     assert eng.sched.n_preempted > 0, "pool must force preemption"
     for f in fins.values():
         assert len(f.logprobs) == len(f.output_ids) == 12
+
+
+def test_model_config_family_shapes():
+    """Config-level sanity for the extra families (no weights built)."""
+    for name, heads, kv, hd in [("llama-2-7b", 32, 32, 128),
+                                ("llama-2-13b", 40, 40, 128),
+                                ("mistral-7b", 32, 8, 128),
+                                ("mixtral-8x7b", 32, 8, 128)]:
+        c = CONFIGS[name]
+        assert c.head_dim == hd and c.num_heads == heads
+        assert c.hidden_size == heads * hd
+        assert c.num_kv_heads == kv
+        for tp in (2, 4, 8):
+            if heads % tp == 0 and kv % tp == 0:
+                s = c.shard(tp)
+                assert s.num_heads * tp == heads
+                assert s.intermediate_size * tp == c.intermediate_size
