@@ -52,12 +52,20 @@ class PackageRegistry:
         elif src.is_file() and src.suffix == ".zip":
             staged = self.pkg_dir / (name or src.stem)
             with zipfile.ZipFile(src) as z:
+                root = staged.resolve()
+                for m in z.namelist():
+                    # zip-slip guard: every member must resolve inside the
+                    # target directory (reject ../ traversal and abs paths)
+                    if not (root / m).resolve().is_relative_to(root):
+                        raise ValueError(f"unsafe archive member: {m}")
                 z.extractall(staged)
         elif src.is_file() and (src.name.endswith(".tar.gz")
                                 or src.suffix == ".tgz"):
             staged = self.pkg_dir / (name or src.name.split(".")[0])
             with tarfile.open(src) as t:
-                t.extractall(staged)
+                # 'data' filter rejects traversal, absolute names, devices
+                # and out-of-tree links (tar-slip)
+                t.extractall(staged, filter="data")
         else:  # treat as git URL / repo path
             target = self.pkg_dir / (name or Path(source).stem)
             if target.exists():

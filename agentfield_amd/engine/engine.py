@@ -230,10 +230,10 @@ class LLMEngine:
 
     @staticmethod
     def _emit(seq: Sequence, done: bool) -> bool:
-        """Stream exactly-once under preemption: a preempted sequence
-        recomputes tokens it already delivered (identical for greedy); a
-        high-water mark suppresses the re-walk so streaming consumers of
-        step() events and on_token never see duplicates."""
+        """Stream exactly-once: preemption retains generated tokens (their
+        KV recomputes as prefill), so new tokens always advance the
+        high-water mark; the mark stays as a safety net against any future
+        path that re-walks delivered tokens."""
         pos = len(seq.output_ids)
         if pos > getattr(seq, "_streamed", 0):
             seq._streamed = pos
@@ -246,24 +246,31 @@ class LLMEngine:
 
     def _step_prefill_chunk(self) -> tuple[list[Sequence], list[int]]:
         """Prefill up to max_prefill_tokens across the pending prompts;
-        sample only for prompts whose last chunk completed."""
+        sample only for prompts whose last chunk completed.
+
+        A sequence preempted mid-generation re-enters here with its
+        output_ids retained: the recompute covers prompt AND outputs, so
+        decode resumes exactly where it stopped (no resampling — streamed
+        tokens stay the final tokens even at temperature>0)."""
         dev = self.device
         budget = self.sched.cfg.max_prefill_tokens
-        seqs, chunks = [], []
+        seqs, chunks, known = [], [], []
         for seq in self._prefilling:
             if budget <= 0:
                 break
-            take = min(len(seq.prompt_ids) - seq.num_prefilled, budget)
+            all_ids = seq.prompt_ids + seq.output_ids
+            take = min(len(all_ids) - seq.num_prefilled, budget)
             if take <= 0:
                 continue
             seqs.append(seq)
             chunks.append(take)
+            known.append(all_ids)
             budget -= take
         ids, pos, slots, q_start = [], [], [], []
         bt_rows = []
-        for seq, take in zip(seqs, chunks):
+        for seq, take, all_ids in zip(seqs, chunks, known):
             np0 = seq.num_prefilled
-            ids.extend(seq.prompt_ids[np0:np0 + take])
+            ids.extend(all_ids[np0:np0 + take])
             pos.extend(range(np0, np0 + take))
             slots.extend(self._slot(seq, i) for i in range(np0, np0 + take))
             q_start.append(np0)
@@ -284,9 +291,9 @@ class LLMEngine:
                                                device=dev),
                           block_table=torch.stack(bt_rows).to(dev))
         done, done_rows = [], []
-        for i, (seq, take) in enumerate(zip(seqs, chunks)):
+        for i, (seq, take, all_ids) in enumerate(zip(seqs, chunks, known)):
             seq.num_prefilled += take
-            if seq.num_prefilled >= len(seq.prompt_ids):
+            if seq.num_prefilled >= len(all_ids):
                 done.append(seq)
                 done_rows.append(cu_list[i + 1] - 1)
         for seq in done:

@@ -135,7 +135,8 @@ class PrefixCachingScheduler(Scheduler):
                     break
                 matched.append(page)
             cached_tok = len(matched) * cfg.page_size
-            ntok = len(prompt) - cached_tok
+            # num_tokens includes outputs retained across preemption
+            ntok = cand.num_tokens - cached_tok
             if batch and tokens + ntok > cfg.max_prefill_tokens:
                 break
             # PIN matched pages before any eviction: _ensure_free must not
@@ -143,7 +144,7 @@ class PrefixCachingScheduler(Scheduler):
             for p in matched:
                 self.alloc.ref(p)
                 self._cache.move_to_end(self._page_hash[p])
-            need = self._pages_needed(len(prompt)) - len(matched)
+            need = self._pages_needed(cand.num_tokens) - len(matched)
             if not self._ensure_free(need):
                 self.alloc.free(matched)  # unpin; candidate stays queued
                 break
@@ -170,7 +171,7 @@ class PrefixCachingScheduler(Scheduler):
                 if self.running[-1] is seq:
                     self.running.pop()
                     self.release(seq)
-                    seq.reset_generation()
+                    seq.on_preempt()
                     seq.status = SeqStatus.WAITING
                     self.waiting.appendleft(seq)
                     self.n_preempted += 1

@@ -100,7 +100,7 @@ class Scheduler:
     def _preempt_last(self) -> None:
         victim = self.running.pop()
         self.release(victim)
-        victim.reset_generation()       # recompute from scratch on readmit
+        victim.on_preempt()   # keep outputs; recompute their KV on readmit
         victim.status = SeqStatus.WAITING
         self.waiting.appendleft(victim)
         self.n_preempted += 1
@@ -113,7 +113,9 @@ class Scheduler:
         tokens = 0
         while (self.waiting and len(self.running) + len(batch) < cfg.max_num_seqs):
             cand = self.waiting[0]
-            ntok = len(cand.prompt_ids)
+            # num_tokens includes outputs retained across preemption: their
+            # KV recomputes as prefill on re-admission
+            ntok = cand.num_tokens
             if batch and tokens + ntok > cfg.max_prefill_tokens:
                 break
             need = self._pages_needed(ntok)
@@ -140,7 +142,7 @@ class Scheduler:
                     # can't preempt self and nothing else to free: defer
                     self.running.pop()
                     self.release(seq)
-                    seq.reset_generation()
+                    seq.on_preempt()
                     seq.status = SeqStatus.WAITING
                     self.waiting.appendleft(seq)
                     self.n_preempted += 1
@@ -229,7 +231,7 @@ class NativeSchedulerAdapter:
         r = self.nat.schedule()
         for sid in r.preempted:
             s = self.seqs[sid]
-            s.reset_generation()
+            s.on_preempt()
             s.pages = []
             s.status = SeqStatus.WAITING
         if not r.has_work:

@@ -54,12 +54,15 @@ class Sequence:
     def last_token(self) -> int:
         return self.output_ids[-1] if self.output_ids else self.prompt_ids[-1]
 
-    def reset_generation(self) -> None:
-        """Preemption: drop all generated state so the sequence recomputes
-        from scratch on re-admission (tokens AND per-token logprobs — a
-        stale logprobs list would misalign with the regenerated output)."""
-        self.output_ids.clear()
-        self.logprobs = None
+    def on_preempt(self) -> None:
+        """Preemption keeps already-generated tokens (vLLM-style recompute):
+        on re-admission the prompt AND retained outputs re-prefill, then
+        decoding continues from where it stopped.  Streamed output therefore
+        never diverges from final output_ids, even for sampled sequences —
+        dropping tokens and resampling would splice old-sample prefix with
+        new-sample suffix for temperature>0 streams."""
+        self.num_prefilled = 0
+        self.cached_prefix = 0
 
     def append(self, tok: int, eos_id: int) -> bool:
         """Append a generated token; returns True when the sequence finished."""

@@ -131,7 +131,7 @@ struct ScheduleResult {
   bool has_work = false;
   bool is_prefill = false;
   std::vector<long> seq_ids;
-  std::vector<long> preempted;  // seqs whose generated tokens must be dropped
+  std::vector<long> preempted;  // seqs to re-prefill (outputs retained)
 };
 
 class NativeScheduler {
@@ -223,7 +223,9 @@ class NativeScheduler {
         }
       }
       int cached_tok = (int)matched.size() * page_size_;
-      int ntok = st.prompt_len - cached_tok;
+      // num_tokens includes outputs retained across preemption — their KV
+      // recomputes as prefill on re-admission
+      int ntok = st.num_tokens - cached_tok;
       if (!batch.empty() && tokens + ntok > max_prefill_tokens_) break;
       // PIN matched pages before eviction can touch them (lockstep with
       // the Python oracle's fix for the same hazard)
@@ -231,7 +233,7 @@ class NativeScheduler {
         refs_[matched[k]] += 1;
         lru_touch(st.hashes[k]);
       }
-      int need = pages_needed(st.prompt_len) - (int)matched.size();
+      int need = pages_needed(st.num_tokens) - (int)matched.size();
       if (!ensure_free(need)) {
         for (int p : matched) unref(p);  // unpin; candidate stays queued
         break;
@@ -362,7 +364,9 @@ class NativeScheduler {
   void preempt(long sid, ScheduleResult& r) {
     SeqState& st = seqs_[sid];
     release(st);
-    st.num_tokens = st.prompt_len;  // recompute from scratch
+    // num_tokens keeps the generated tokens: on re-admission their KV
+    // recomputes as prefill and decode resumes (vLLM-style recompute;
+    // resampling would splice sampled streams)
     waiting_.push_front(sid);
     n_preempted_ += 1;
     r.preempted.push_back(sid);

@@ -196,7 +196,16 @@ class EngineRunner:
                     ids, sp, waiter = self._submit.get_nowait()
                 except queue.Empty:
                     break
-                rid = eng.add_request(ids, sp)
+                # Bad requests (prompt+max_tokens over the context window,
+                # malformed sampling params) must resolve this waiter, not
+                # kill the driver thread and wedge every other caller.
+                try:
+                    rid = eng.add_request(ids, sp)
+                except Exception as e:  # noqa: BLE001 — loop must survive
+                    waiter["error"] = f"rejected: {e}"
+                    waiter["done"].set()
+                    moved = True
+                    continue
                 if rid is None:
                     waiter["error"] = "engine queue full"
                     waiter["done"].set()
@@ -217,7 +226,17 @@ class EngineRunner:
                     w["done"].set()
                 moved = True
             if eng.has_work():
-                for (rid, tok, done) in eng.step():
+                try:
+                    events = eng.step()
+                except Exception as e:  # noqa: BLE001 — resolve, don't hang
+                    for w in pending.values():
+                        w["error"] = f"engine step failed: {e}"
+                        w["done"].set()
+                        if w["stream"] is not None:
+                            w["stream"].put((None, True))
+                    pending.clear()
+                    continue
+                for (rid, tok, done) in events:
                     w = pending.get(rid)
                     if w and w["stream"] is not None:
                         w["stream"].put((tok, done))

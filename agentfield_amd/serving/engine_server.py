@@ -42,6 +42,17 @@ def create_engine_app(runner: EngineRunner, model_name: str,
             return runners[m], m
         return runners[model_name], model_name
 
+    def clamp_prompt(rnr: EngineRunner, ids: list[int],
+                     max_tokens: int) -> list[int]:
+        """Token-aware head trim so prompt+max_tokens always fits the
+        context window (mirrors the reference's client-side trimming,
+        agent_ai.py:267) — an oversized prompt must degrade, not 500."""
+        limit = rnr.engine.cfg.max_position - 8
+        if len(ids) + max_tokens > limit:
+            keep = max(1, limit - max_tokens)
+            ids = ids[-keep:]
+        return ids
+
     @app.get("/v1/health")
     async def health():
         return {"status": "healthy", "model": model_name,
@@ -102,6 +113,8 @@ def create_engine_app(runner: EngineRunner, model_name: str,
         sp = SamplingParams(
             max_tokens=int(body.get("max_tokens", 128)),
             temperature=float(body.get("temperature", 0.0)),
+            top_k=int(body.get("top_k", 0) or 0),
+            top_p=float(body.get("top_p", 1.0) or 1.0),
             ignore_eos=bool(body.get("ignore_eos", False)),
             logprobs=min(int(body.get("logprobs", 0) or 0), 8),
             json_mode=jm)
@@ -109,6 +122,7 @@ def create_engine_app(runner: EngineRunner, model_name: str,
             ids = [int(x) for x in body["prompt_ids"]]
         else:
             ids = runner.tokenizer.encode(body.get("prompt", ""))
+        ids = clamp_prompt(runner, ids, sp.max_tokens)
         stops = tuple(s for s in body.get("stop", []) if s)
         stream = bool(body.get("stream", False))
         if not stream:
@@ -120,6 +134,7 @@ def create_engine_app(runner: EngineRunner, model_name: str,
                 return w
             w = await anyio.to_thread.run_sync(wait)
             if not w["done"].is_set():
+                runner.cancel(w)  # free KV pages + batch slot
                 return JSONResponse({"error": "timeout"}, status_code=504)
             if w["error"]:
                 return JSONResponse({"error": w["error"]}, status_code=503)
@@ -223,12 +238,15 @@ def create_engine_app(runner: EngineRunner, model_name: str,
                 status_code=400)
         sp = SamplingParams(max_tokens=max_tokens,
                             temperature=float(body.get("temperature", 0.0)),
+                            top_p=float(body.get("top_p", 1.0) or 1.0),
+                            top_k=int(body.get("top_k", 0) or 0),
                             logprobs=lp_n,
                             json_mode=jm)
         stop_in = body.get("stop") or []
         stops = tuple(s for s in ([stop_in] if isinstance(stop_in, str)
                                   else stop_in) if s)
-        ids = runner.tokenizer.encode(prompt)
+        ids = clamp_prompt(runner, runner.tokenizer.encode(prompt),
+                           max_tokens)
         rid = f"{'chatcmpl' if kind == 'chat' else 'cmpl'}-{uuid.uuid4().hex[:24]}"
         created = int(_time.time())
         base = {"id": rid, "created": created, "model": _mname}
@@ -244,6 +262,9 @@ def create_engine_app(runner: EngineRunner, model_name: str,
                 return waiters
             await anyio.to_thread.run_sync(wait)
             if not all(w["done"].is_set() for w in waiters):
+                for w in waiters:
+                    if not w["done"].is_set():
+                        runner.cancel(w)  # free KV pages + batch slot
                 return JSONResponse({"error": {"message": "timeout",
                                                "type": "timeout"}},
                                     status_code=504)

@@ -291,8 +291,8 @@ def test_json_mode_fails_open_on_corrupt_history():
 
 
 def test_logprobs_survive_preemption():
-    """Preemption resets generated state; a stale logprobs list must not
-    survive into the recomputed output (found by the round-1 soak)."""
+    """Logprobs stay aligned with output_ids across preemption (outputs are
+    retained and their KV recomputed; found by the round-1 soak)."""
     from agentfield_amd.models import CONFIGS as _C
     eng = LLMEngine(_C["tiny"], device="cpu", dtype=torch.float32,
                     page_size=4, num_pages=10, max_num_seqs=4,
@@ -314,6 +314,46 @@ def test_logprobs_survive_preemption():
     assert eng.sched.n_preempted > 0, "pool must force preemption"
     for f in fins.values():
         assert len(f.logprobs) == len(f.output_ids) == 12
+
+
+def test_preemption_keeps_sampled_stream_exact():
+    """Preemption retains generated tokens (vLLM-style recompute): for
+    temperature>0 the streamed token sequence must equal the final
+    output_ids exactly — no old-sample/new-sample splice."""
+    from agentfield_amd.models import CONFIGS as _C
+    eng = LLMEngine(_C["tiny"], device="cpu", dtype=torch.float32,
+                    page_size=4, num_pages=10, max_num_seqs=4,
+                    enable_graphs=False, max_prefill_tokens=16)
+    sp = SamplingParams(max_tokens=12, ignore_eos=True, temperature=0.9)
+    streamed = {}
+
+    def mk(rid_box):
+        def cb(tok, done):
+            streamed.setdefault(rid_box[0], []).append(tok)
+        return cb
+
+    rids = []
+    for _ in range(3):
+        box = [None]
+        rid = eng.add_request([1, 5, 9, 20, 7, 3, 11, 2], sp, on_token=mk(box))
+        box[0] = rid
+        rids.append(rid)
+    fins = {}
+    for _ in range(500):
+        eng.step()
+        for r in rids:
+            if r not in fins:
+                f = eng.get_finished(r)
+                if f:
+                    fins[r] = f
+        if len(fins) == len(rids):
+            break
+    assert len(fins) == len(rids)
+    assert eng.sched.n_preempted > 0, "pool must force preemption"
+    for r, f in fins.items():
+        assert streamed[r] == f.output_ids, \
+            "streamed tokens diverged from final output across preemption"
+        assert len(f.output_ids) == 12
 
 
 def test_model_config_family_shapes():
