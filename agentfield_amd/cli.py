@@ -30,8 +30,51 @@ def server(host: str = "0.0.0.0", port: int = 8520,
            db: str = typer.Option("agentfield.db", help="SQLite path"),
            data_dir: str = typer.Option(".agentfield", help="payload/keys dir"),
            config: str = typer.Option(None, help="YAML config file"),
+           workers: int = typer.Option(1, help="control-plane worker "
+                                       "processes (ports port..port+N-1 "
+                                       "over one shared WAL database; "
+                                       "status callbacks stick to the "
+                                       "dispatching worker)"),
            no_did: bool = False):
     """Run the control plane."""
+    if workers > 1:
+        # multi-worker plane (reference parity: the Go plane's NumCPU
+        # worker pool, execute.go:1373 — here realized as N processes
+        # because Python serializes a single process on the GIL).
+        # Worker i listens on port+i; worker 0 runs the background
+        # services (webhook poller, presence, cleanup); all workers
+        # share the SQLite WAL file.  Agents may register at any worker;
+        # X-AgentField-Callback pins status callbacks to the worker
+        # holding the sync waiter.
+        import subprocess
+        cb_host = "127.0.0.1" if host == "0.0.0.0" else host
+        urls = ",".join(f"http://{cb_host}:{port + i}"
+                        for i in range(workers))
+        procs = []
+        for i in range(workers):
+            env = {**os.environ,
+                   "AGENTFIELD_PUBLIC_URL": f"http://{cb_host}:{port + i}",
+                   "AGENTFIELD_WORKER_URLS": urls,
+                   "AGENTFIELD_ADMIN_GRPC_PORT": str(port + workers + 100 + i)}
+            if i > 0:
+                env["AGENTFIELD_BACKGROUND_SERVICES"] = "0"
+            cmd = [sys.executable, "-m", "agentfield_amd", "server",
+                   "--host", host, "--port", str(port + i), "--db", db,
+                   "--data-dir", data_dir, "--workers", "1"]
+            if config:
+                cmd += ["--config", config]
+            if no_did:
+                cmd += ["--no-did"]
+            procs.append(subprocess.Popen(cmd, env=env))
+        typer.echo(f"agentfield-amd control plane x{workers} on "
+                   f"{host}:{port}-{port + workers - 1} (db={db})")
+        try:
+            for p in procs:
+                p.wait()
+        except KeyboardInterrupt:
+            for p in procs:
+                p.terminate()
+        return
     import uvicorn
     from .controlplane import ControlPlane, create_app
     from .controlplane.server import Config
@@ -49,7 +92,8 @@ def server(host: str = "0.0.0.0", port: int = 8520,
                      "AGENTFIELD_ADMIN_GRPC_PORT", port + 100)), **kw)
     cp = ControlPlane(cfg)
     typer.echo(f"agentfield-amd control plane on {host}:{port} (db={cfg.db_path})")
-    uvicorn.run(create_app(cp), host=host, port=port, log_level="info")
+    uvicorn.run(create_app(cp), host=host, port=port, log_level="warning",
+                access_log=False)
 
 
 @app.command()

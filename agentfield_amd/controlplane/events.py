@@ -38,37 +38,56 @@ class EventBus:
 
 
 class Buses:
-    """The control plane's bus set (execution / node / reasoner / memory)."""
+    """The control plane's bus set (execution / node / reasoner / memory).
+
+    Sync-execute waiters do NOT ride the broadcast bus: with C concurrent
+    waiters a broadcast is O(C) per event / O(C²) per batch and bounded
+    subscriber queues drop the very terminal event a waiter needs (measured:
+    p99 = the full 90 s sync timeout under concurrency 64).  Waiters are a
+    targeted {execution_id -> [Future]} registry resolved in O(1); the bus
+    stays for SSE/WS consumers that want the whole stream."""
 
     def __init__(self):
         self.execution = EventBus()
         self.node = EventBus()
         self.reasoner = EventBus()
         self.memory = EventBus(buffer=256)
+        self._exec_waiters: dict[str, list[asyncio.Future]] = {}
+
+    def register_waiter(self, execution_id: str) -> asyncio.Future:
+        """Register BEFORE dispatching to the agent so a fast callback
+        cannot race past the waiter."""
+        fut = asyncio.get_running_loop().create_future()
+        self._exec_waiters.setdefault(execution_id, []).append(fut)
+        return fut
+
+    def discard_waiter(self, execution_id: str, fut: asyncio.Future) -> None:
+        lst = self._exec_waiters.get(execution_id)
+        if lst is not None:
+            try:
+                lst.remove(fut)
+            except ValueError:
+                pass
+            if not lst:
+                self._exec_waiters.pop(execution_id, None)
+
+    def publish_execution(self, event: dict) -> None:
+        """Publish to SSE subscribers and resolve any sync waiters."""
+        self.execution.publish(event)
+        if event.get("terminal"):
+            for fut in self._exec_waiters.pop(event["execution_id"], ()):
+                if not fut.done():
+                    fut.set_result(event)
 
     async def wait_for_execution(self, execution_id: str, timeout: float,
-                                 queue: asyncio.Queue | None = None):
-        """Event-bus wait used by the sync execute path (reference:
-        waitForExecutionCompletion, execute.go:568-629).  Pass a queue from
-        an earlier subscribe() to close the subscribe-after-dispatch race:
-        the subscription must exist BEFORE the agent is called, or a fast
-        callback can fire the terminal event with no listener."""
-        sid = None
-        if queue is None:
-            sid, queue = self.execution.subscribe()
+                                 fut: asyncio.Future | None = None):
+        """Targeted wait for an execution's terminal event (reference:
+        waitForExecutionCompletion, execute.go:568-629)."""
+        if fut is None:
+            fut = self.register_waiter(execution_id)
         try:
-            loop = asyncio.get_running_loop()
-            deadline = loop.time() + timeout
-            while True:
-                left = deadline - loop.time()
-                if left <= 0:
-                    return None
-                try:
-                    ev = await asyncio.wait_for(queue.get(), left)
-                except asyncio.TimeoutError:
-                    return None
-                if ev.get("execution_id") == execution_id and ev.get("terminal"):
-                    return ev
+            return await asyncio.wait_for(fut, timeout)
+        except asyncio.TimeoutError:
+            return None
         finally:
-            if sid is not None:
-                self.execution.unsubscribe(sid)
+            self.discard_waiter(execution_id, fut)

@@ -44,6 +44,11 @@ class Config:
         "presence_ttl": ("AGENTFIELD_PRESENCE_TTL", float),
         "health_interval": ("AGENTFIELD_HEALTH_INTERVAL", float),
         "admin_grpc_port": ("AGENTFIELD_ADMIN_GRPC_PORT", int),
+        "public_url": ("AGENTFIELD_PUBLIC_URL", str),
+        "background_services": ("AGENTFIELD_BACKGROUND_SERVICES",
+                                lambda v: v not in ("0", "false")),
+        "worker_urls": ("AGENTFIELD_WORKER_URLS",
+                        lambda v: [u for u in v.split(",") if u]),
     }
 
     def __init__(self, **kw):
@@ -65,6 +70,13 @@ class Config:
         self.health_interval = kw.get("health_interval", 60.0)
         self.background_services = kw.get("background_services", True)
         self.admin_grpc_port = kw.get("admin_grpc_port")
+        # this worker's own externally-reachable URL; sent to agents as
+        # X-AgentField-Callback so status callbacks stick to the worker
+        # holding the sync waiter (enables multi-worker planes)
+        self.public_url = kw.get("public_url")
+        # the whole worker fleet (advertised via /api/v1/health so load
+        # clients can round-robin without a front proxy)
+        self.worker_urls = kw.get("worker_urls") or []
 
 
 class ControlPlane:
@@ -83,7 +95,7 @@ class ControlPlane:
         self.dids = DIDService(self.storage, Keystore(self.cfg.keystore_path)) \
             if self.cfg.did_enabled else None
         self.vcs = VCService(self.storage, self.dids) if self.dids else None
-        self.client: httpx.AsyncClient | None = None
+        self.client = None  # aiohttp.ClientSession (created in start())
         self._async_q: asyncio.Queue | None = None
         self._workers: list[asyncio.Task] = []
         self._grpc_server = None
@@ -92,7 +104,13 @@ class ControlPlane:
     async def start(self):
         log.info("control plane starting (db=%s, workers=%d)",
                  self.cfg.db_path, self.cfg.async_workers)
-        self.client = httpx.AsyncClient(timeout=self.cfg.agent_timeout)
+        # aiohttp for the agent-dispatch hot path: its C http stack costs a
+        # fraction of httpx per request (the control plane makes one
+        # outbound call per execution)
+        import aiohttp
+        self.client = aiohttp.ClientSession(
+            timeout=aiohttp.ClientTimeout(total=self.cfg.agent_timeout),
+            connector=aiohttp.TCPConnector(limit=512))
         self._async_q = asyncio.Queue(self.cfg.async_queue_capacity)
         for _ in range(self.cfg.async_workers):
             self._workers.append(asyncio.create_task(self._async_worker()))
@@ -116,7 +134,7 @@ class ControlPlane:
         if self._grpc_server is not None:
             self._grpc_server.stop(grace=0.5)
         if self.client:
-            await self.client.aclose()
+            await self.client.close()
 
     # ----------------------------------------------------- execution core
     async def probe_node(self, node: dict) -> bool:
@@ -186,16 +204,8 @@ class ControlPlane:
         }
         inline, uri = self.payloads.maybe_offload(rec["id"], "input",
                                                   rec["input"])
-        if uri:
-            stored = {**rec, "input": {"$payload_uri": uri}}
-            self.storage.create_execution(stored)
-        else:
-            self.storage.create_execution(rec)
-        self.storage.upsert_run(run_id, st.RUNNING, rec["id"])
-        if webhook:
-            self.storage.register_webhook(rec["id"], webhook["url"],
-                                          webhook.get("secret", ""),
-                                          webhook.get("headers"))
+        stored = {**rec, "input": {"$payload_uri": uri}} if uri else rec
+        self.storage.create_execution_full(stored, run_id, webhook)
         return rec, None
 
     async def call_agent(self, rec: dict) -> tuple[int, dict | None, str | None]:
@@ -213,19 +223,26 @@ class ControlPlane:
             headers["X-Session-ID"] = rec["session_id"]
         if rec.get("actor_id"):
             headers["X-Actor-ID"] = rec["actor_id"]
+        if self.cfg.public_url:
+            # sticky callback routing: the terminal status callback must
+            # land on THIS worker (it holds the sync waiter's future)
+            headers["X-AgentField-Callback"] = self.cfg.public_url
         payload = dict(rec.get("input") or {})
         try:
-            resp = await self.client.post(url, json=payload, headers=headers)
+            async with self.client.post(url, json=payload,
+                                        headers=headers) as resp:
+                if resp.status == 202:
+                    return 202, None, None
+                if resp.status == 200:
+                    try:
+                        return 200, await resp.json(content_type=None), None
+                    except ValueError:
+                        return 200, {"raw": await resp.text()}, None
+                text = await resp.text()
+                return resp.status, None, \
+                    f"agent HTTP {resp.status}: {text[:300]}"
         except Exception as e:
             return 0, None, f"agent unreachable: {e}"
-        if resp.status_code == 202:
-            return 202, None, None
-        if resp.status_code == 200:
-            try:
-                return 200, resp.json(), None
-            except ValueError:
-                return 200, {"raw": resp.text}, None
-        return resp.status_code, None, f"agent HTTP {resp.status_code}: {resp.text[:300]}"
 
     def complete_execution(self, execution_id: str, status: str, result=None,
                            error: str | None = None,
@@ -235,20 +252,16 @@ class ControlPlane:
             _, uri = self.payloads.maybe_offload(execution_id, "result", result)
             if uri:
                 result = {"$payload_uri": uri}
-        applied = self.storage.update_execution_result(
-            execution_id, status, result, error, duration_ms)
-        rec = self.storage.get_execution(execution_id)
+        applied, rec = self.storage.finalize_execution(
+            execution_id, status, result, error, duration_ms,
+            aggregate=aggregate_status)
         if rec is None:
             return None
         if not applied:
             return rec  # already terminal: no duplicate events/webhooks/VCs
-        if rec.get("run_id"):
-            sibs = self.storage.executions_by_run(rec["run_id"])
-            self.storage.upsert_run(
-                rec["run_id"], aggregate_status([e["status"] for e in sibs]))
         if rec.get("duration_ms"):
             self.metrics.step_duration.observe(rec["duration_ms"] / 1000.0)
-        self.buses.execution.publish({
+        self.buses.publish_execution({
             "execution_id": execution_id, "status": status,
             "terminal": st.is_terminal(status), "run_id": rec.get("run_id"),
         })
@@ -294,6 +307,83 @@ class ControlPlane:
                 return v
         return v
 
+    # ------------------------------------------------- hot-path handlers
+    # Plain async functions (no framework Request/Response) so both the
+    # FastAPI routes and the raw-ASGI fast path share one implementation.
+    async def h_execute_sync(self, target: str, body: dict,
+                             headers) -> tuple[int, dict, dict]:
+        rec, err = self.prepare_execution(target, body, headers)
+        if err:
+            return err["status_code"], {"error": err["error"]}, {}
+        t0 = time.time()
+        # register the waiter BEFORE dispatching so a fast agent callback
+        # can't race past it (the terminal event would otherwise be lost
+        # and the request would ride out the full sync timeout)
+        fut = self.buses.register_waiter(rec["id"])
+        try:
+            code, result, errmsg = await self.call_agent(rec)
+            if code == 200:
+                final = self.complete_execution(
+                    rec["id"], st.COMPLETED, result=result,
+                    duration_ms=(time.time() - t0) * 1e3)
+            elif code == 202:
+                self.metrics.waiters_inflight.inc()
+                try:
+                    ev = await self.buses.wait_for_execution(
+                        rec["id"], self.cfg.sync_timeout, fut=fut)
+                finally:
+                    self.metrics.waiters_inflight.dec()
+                final = self.storage.get_execution(rec["id"])
+                if ev is None and not st.is_terminal(final.get("status", "")):
+                    final = self.complete_execution(
+                        rec["id"], st.TIMEOUT, error="sync wait timed out")
+            else:
+                final = self.complete_execution(rec["id"], st.FAILED,
+                                                error=errmsg)
+        finally:
+            self.buses.discard_waiter(rec["id"], fut)
+        return 200, self.envelope(final), {"X-Execution-ID": rec["id"],
+                                           "X-Run-ID": rec["run_id"]}
+
+    async def h_execute_async(self, target: str, body: dict,
+                              headers) -> tuple[int, dict, dict]:
+        rec, err = self.prepare_execution(target, body, headers)
+        if err:
+            return err["status_code"], {"error": err["error"]}, {}
+        try:
+            self._async_q.put_nowait(rec)
+        except asyncio.QueueFull:
+            self.metrics.backpressure.inc()
+            self.complete_execution(rec["id"], st.FAILED,
+                                    error="async queue is full")
+            return 503, {"error": "queue is full"}, {}
+        self.metrics.queue_depth.set(self._async_q.qsize())
+        ts = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
+        return 202, {
+            "execution_id": rec["id"], "run_id": rec["run_id"],
+            "workflow_id": rec["run_id"], "status": "queued",
+            "target": target, "type": rec["target_type"],
+            "created_at": ts, "enqueued_at": ts,
+        }, {"X-Execution-ID": rec["id"], "X-Run-ID": rec["run_id"]}
+
+    def h_status_callback(self, execution_id: str,
+                          body: dict) -> tuple[int, dict]:
+        status = st.normalize(body.get("status", ""))
+        if status not in (st.RUNNING, st.PENDING) and \
+                not st.is_terminal(status):
+            return 400, {"error": f"unknown status '{body.get('status')}'"}
+        if st.is_terminal(status):
+            self.complete_execution(execution_id, status, body.get("result"),
+                                    body.get("error"),
+                                    body.get("duration_ms"))
+        return 200, {"status": "ok"}
+
+    def h_get_execution(self, execution_id: str) -> tuple[int, dict]:
+        rec = self.storage.get_execution(execution_id)
+        if rec is None:
+            return 404, {"error": "not found"}
+        return 200, self.envelope(rec)
+
     def envelope(self, rec: dict) -> dict:
         rec = {**rec, "input": self._resolve_payload(rec.get("input")),
                "result": self._resolve_payload(rec.get("result"))}
@@ -313,6 +403,82 @@ def _sse(event: dict) -> str:
     return f"data: {json.dumps(event)}\n\n"
 
 
+class FastPathASGI:
+    """Raw-ASGI dispatch for the execute hot loop.
+
+    Starlette matches routes by scanning ~50 compiled regexes per request
+    and FastAPI adds dependency resolution on top (profiled: ~45% of
+    control-plane CPU under load, tools/stress.py).  The four hot
+    endpoints — sync/async execute, the agent status callback and
+    execution GET — are dispatched here with exact prefix checks and
+    hand-rolled request/response handling; everything else falls through
+    to the full FastAPI app unchanged."""
+
+    _EXEC = "/api/v1/execute/"
+    _EXEC_ASYNC = "/api/v1/execute/async/"
+    _EXECUTIONS = "/api/v1/executions/"
+
+    def __init__(self, app, cp: "ControlPlane"):
+        self.app = app
+        self.cp = cp
+
+    async def __call__(self, scope, receive, send):
+        if scope["type"] != "http":
+            return await self.app(scope, receive, send)
+        path, method = scope["path"], scope["method"]
+        cp = self.cp
+        if method == "POST" and path.startswith(self._EXEC):
+            headers = {k.decode("latin1"): v.decode("latin1")
+                       for k, v in scope["headers"]}
+            body = await self._read_json(receive)
+            if path.startswith(self._EXEC_ASYNC):
+                target = path[len(self._EXEC_ASYNC):]
+                code, payload, hdrs = await cp.h_execute_async(
+                    target, body, headers)
+            else:
+                target = path[len(self._EXEC):]
+                code, payload, hdrs = await cp.h_execute_sync(
+                    target, body, headers)
+            return await self._respond(send, code, payload, hdrs)
+        if path.startswith(self._EXECUTIONS):
+            rest = path[len(self._EXECUTIONS):]
+            if method == "POST" and rest.endswith("/status") and \
+                    "/" not in rest[:-7]:
+                body = await self._read_json(receive)
+                code, payload = cp.h_status_callback(rest[:-7], body)
+                return await self._respond(send, code, payload)
+            if method == "GET" and "/" not in rest and rest:
+                code, payload = cp.h_get_execution(rest)
+                return await self._respond(send, code, payload)
+        await self.app(scope, receive, send)
+
+    @staticmethod
+    async def _read_json(receive) -> dict:
+        chunks = []
+        while True:
+            msg = await receive()
+            chunks.append(msg.get("body", b""))
+            if not msg.get("more_body"):
+                break
+        try:
+            out = json.loads(b"".join(chunks))
+            return out if isinstance(out, dict) else {}
+        except (ValueError, TypeError):
+            return {}
+
+    @staticmethod
+    async def _respond(send, code: int, payload: dict,
+                       extra_headers: dict | None = None):
+        body = json.dumps(payload).encode()
+        headers = [(b"content-type", b"application/json"),
+                   (b"content-length", str(len(body)).encode())]
+        for k, v in (extra_headers or {}).items():
+            headers.append((k.encode("latin1"), str(v).encode("latin1")))
+        await send({"type": "http.response.start", "status": code,
+                    "headers": headers})
+        await send({"type": "http.response.body", "body": body})
+
+
 def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
     cp = cp or ControlPlane(Config(**cfg_kw))
 
@@ -328,8 +494,11 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
     # ------------------------------------------------------------- basics
     @app.get("/api/v1/health")
     async def health():
-        return {"status": "healthy", "uptime_s": time.time() - cp.started_at,
-                "version": "0.1.0"}
+        out = {"status": "healthy", "uptime_s": time.time() - cp.started_at,
+               "version": "0.1.0"}
+        if cp.cfg.worker_urls:
+            out["workers"] = cp.cfg.worker_urls
+        return out
 
     @app.get("/metrics")
     async def metrics():
@@ -494,41 +663,9 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
             body = await req.json()
         except Exception:
             body = {}
-        rec, err = cp.prepare_execution(target, body, req.headers)
-        if err:
-            return JSONResponse({"error": err["error"]},
-                                status_code=err["status_code"])
-        t0 = time.time()
-        # subscribe BEFORE dispatching so a fast agent callback can't race
-        # past the waiter (the terminal event would otherwise be lost and
-        # the request would ride out the full sync timeout)
-        sub_id, sub_q = cp.buses.execution.subscribe()
-        try:
-            code, result, errmsg = await cp.call_agent(rec)
-            if code == 200:
-                final = cp.complete_execution(
-                    rec["id"], st.COMPLETED, result=result,
-                    duration_ms=(time.time() - t0) * 1e3)
-            elif code == 202:
-                cp.metrics.waiters_inflight.inc()
-                try:
-                    ev = await cp.buses.wait_for_execution(
-                        rec["id"], cp.cfg.sync_timeout, queue=sub_q)
-                finally:
-                    cp.metrics.waiters_inflight.dec()
-                final = cp.storage.get_execution(rec["id"])
-                if ev is None and not st.is_terminal(final.get("status", "")):
-                    final = cp.complete_execution(rec["id"], st.TIMEOUT,
-                                                  error="sync wait timed out")
-            else:
-                final = cp.complete_execution(rec["id"], st.FAILED,
-                                              error=errmsg)
-        finally:
-            cp.buses.execution.unsubscribe(sub_id)
-        resp = JSONResponse(cp.envelope(final))
-        resp.headers["X-Execution-ID"] = rec["id"]
-        resp.headers["X-Run-ID"] = rec["run_id"]
-        return resp
+        code, payload, hdrs = await cp.h_execute_sync(target, body,
+                                                      req.headers)
+        return JSONResponse(payload, status_code=code, headers=hdrs)
 
     # legacy direct-execution routes (reference: POST /reasoners/:id and
     # /skills/:id with workflow headers, reasoners.go:45+): same sync
@@ -549,33 +686,14 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
             body = await req.json()
         except Exception:
             body = {}
-        rec, err = cp.prepare_execution(target, body, req.headers)
-        if err:
-            return JSONResponse({"error": err["error"]},
-                                status_code=err["status_code"])
-        try:
-            cp._async_q.put_nowait(rec)
-        except asyncio.QueueFull:
-            cp.metrics.backpressure.inc()
-            cp.complete_execution(rec["id"], st.FAILED,
-                                  error="async queue is full")
-            return JSONResponse({"error": "queue is full"}, status_code=503)
-        cp.metrics.queue_depth.set(cp._async_q.qsize())
-        return JSONResponse({
-            "execution_id": rec["id"], "run_id": rec["run_id"],
-            "workflow_id": rec["run_id"], "status": "queued",
-            "target": target, "type": rec["target_type"],
-            "created_at": time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime()),
-            "enqueued_at": time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime()),
-        }, status_code=202,
-            headers={"X-Execution-ID": rec["id"], "X-Run-ID": rec["run_id"]})
+        code, payload, hdrs = await cp.h_execute_async(target, body,
+                                                       req.headers)
+        return JSONResponse(payload, status_code=code, headers=hdrs)
 
     @app.get("/api/v1/executions/{execution_id}")
     async def get_execution(execution_id: str):
-        rec = cp.storage.get_execution(execution_id)
-        if rec is None:
-            return JSONResponse({"error": "not found"}, status_code=404)
-        return cp.envelope(rec)
+        code, payload = cp.h_get_execution(execution_id)
+        return JSONResponse(payload, status_code=code)
 
     @app.post("/api/v1/executions/batch-status")
     async def batch_status(req: Request):
@@ -601,14 +719,8 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
     async def execution_status_callback(execution_id: str, req: Request):
         """Agent-side terminal/progress status ingestion (A.2)."""
         body = await req.json()
-        status = st.normalize(body.get("status", ""))
-        if status not in (st.RUNNING, st.PENDING) and not st.is_terminal(status):
-            return JSONResponse({"error": f"unknown status '{body.get('status')}'"},
-                                status_code=400)
-        if st.is_terminal(status):
-            cp.complete_execution(execution_id, status, body.get("result"),
-                                  body.get("error"), body.get("duration_ms"))
-        return {"status": "ok"}
+        code, payload = cp.h_status_callback(execution_id, body)
+        return JSONResponse(payload, status_code=code)
 
     # ------------------------------------------------- workflow events/DAG
     @app.post("/api/v1/executions/{execution_id}/notes")
@@ -1026,4 +1138,6 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
     async def ui_index():
         return Response(_ui.read_text(), media_type="text/html")
 
-    return app
+    wrapped = FastPathASGI(app, cp)
+    wrapped.state = app.state  # convenience passthrough (app.state.cp)
+    return wrapped

@@ -94,12 +94,17 @@ class Storage:
     def __init__(self, path: str = ":memory:"):
         if path != ":memory:":
             Path(path).parent.mkdir(parents=True, exist_ok=True)
-        self._db = sqlite3.connect(path, check_same_thread=False)
+        self._db = sqlite3.connect(path, check_same_thread=False,
+                                   timeout=10.0)
         self._db.row_factory = sqlite3.Row
         self._lock = threading.RLock()
         with self._lock:
             if path != ":memory:":
                 self._db.execute("PRAGMA journal_mode=WAL")
+                # multi-worker planes share one WAL file: writers from
+                # other processes serialize on the WAL write lock; wait
+                # instead of surfacing SQLITE_BUSY
+                self._db.execute("PRAGMA busy_timeout=10000")
             self._db.execute("PRAGMA synchronous=NORMAL")
             self._db.executescript(_SCHEMA)
             self._db.commit()
@@ -557,6 +562,80 @@ class Storage:
                finished_at=CASE WHEN excluded.status IN
                  ('completed','failed','timeout','cancelled') THEN ? ELSE NULL END""",
             (run_id, status, root_execution_id, now(), now()))
+
+    # ---------------- hot-path composites (single lock + single commit) --
+    # The execute hot loop (SURVEY.md §3.1) is 2-4 SQL round trips per
+    # request when issued one-by-one; grouping them under one lock/commit
+    # mirrors the reference's serialized completion queue batching
+    # (execute.go:1404-1439) without a worker hop.
+    def create_execution_full(self, rec: dict, run_id: str,
+                              webhook: dict | None = None) -> None:
+        with self._lock:
+            self._db.execute(
+                """INSERT OR REPLACE INTO executions
+                   (id, run_id, parent_execution_id, node_id, reasoner_id,
+                    target_type, status, input, session_id, actor_id, depth,
+                    created_at, started_at, webhook_registered)
+                   VALUES (?,?,?,?,?,?,?,?,?,?,?,?,?,?)""",
+                (rec["id"], rec.get("run_id"), rec.get("parent_execution_id"),
+                 rec.get("node_id"), rec.get("reasoner_id"),
+                 rec.get("target_type", "reasoner"),
+                 rec.get("status", "running"),
+                 json.dumps(rec.get("input")), rec.get("session_id"),
+                 rec.get("actor_id"), rec.get("depth", 0), now(), now(),
+                 1 if rec.get("webhook_registered") else 0))
+            self._db.execute(
+                """INSERT INTO workflow_runs
+                   (run_id, status, root_execution_id, started_at)
+                   VALUES (?,?,?,?) ON CONFLICT(run_id) DO UPDATE SET
+                   status=excluded.status, finished_at=NULL""",
+                (run_id, "running", rec["id"], now()))
+            if webhook:
+                self._db.execute(
+                    """INSERT OR REPLACE INTO execution_webhooks
+                       (execution_id, url, secret, headers, status, attempts,
+                        next_attempt_at, created_at)
+                       VALUES (?,?,?,?, 'pending', 0, ?, ?)""",
+                    (rec["id"], webhook["url"], webhook.get("secret", ""),
+                     json.dumps(webhook.get("headers") or {}), now(), now()))
+            self._db.commit()
+
+    def finalize_execution(self, exec_id: str, status: str, result=None,
+                           error: str | None = None,
+                           duration_ms: float | None = None,
+                           aggregate=None) -> tuple[bool, dict | None]:
+        """Terminal write + row fetch + run-status roll-up in ONE
+        lock/commit.  `aggregate` folds sibling statuses into the run
+        status (workflow.aggregate_status).  Returns (applied, record);
+        applied=False when the row was already terminal (write-once)."""
+        with self._lock:
+            cur = self._db.execute(
+                """UPDATE executions SET status=?, result=?, error_message=?,
+                   finished_at=?, duration_ms=COALESCE(?, (?-started_at)*1000.0)
+                   WHERE id=? AND status NOT IN
+                     ('completed','failed','timeout','cancelled')""",
+                (status, json.dumps(result) if result is not None else None,
+                 error, now(), duration_ms, now(), exec_id))
+            applied = cur.rowcount > 0
+            r = self._db.execute("SELECT * FROM executions WHERE id=?",
+                                 (exec_id,)).fetchone()
+            rec = self._exec_row(dict(r)) if r else None
+            if applied and rec and rec.get("run_id") and aggregate:
+                sibs = self._db.execute(
+                    "SELECT status FROM executions WHERE run_id=?",
+                    (rec["run_id"],)).fetchall()
+                run_status = aggregate([s["status"] for s in sibs])
+                self._db.execute(
+                    """INSERT INTO workflow_runs (run_id, status, started_at)
+                       VALUES (?,?,?) ON CONFLICT(run_id) DO UPDATE SET
+                       status=excluded.status,
+                       finished_at=CASE WHEN excluded.status IN
+                         ('completed','failed','timeout','cancelled')
+                         THEN ? ELSE NULL END""",
+                    (rec["run_id"], run_status, now(), now()))
+            if applied:
+                self._db.commit()
+        return applied, rec
 
     def get_run(self, run_id: str) -> dict | None:
         return self._q1("SELECT * FROM workflow_runs WHERE run_id=?", (run_id,))

@@ -248,19 +248,30 @@ class Agent(FastAPI):
                       ctx: ExecutionContext):
         self._ensure_executor()
         token = set_context(ctx)
-        self._workflow_event(meta, ctx, "start", kwargs)
+        # Control-plane-invoked executions (ctx carries the CP-minted
+        # execution id) already have their record created by
+        # prepare_execution and their terminal state delivered via the
+        # status callback — SDK workflow events would be duplicate rows
+        # and double the CP's HTTP load.  Events fire only for in-process
+        # nested calls, which are otherwise invisible to the DAG (A.3).
+        cp_invoked = bool(ctx.execution_id)
+        if not cp_invoked:
+            self._workflow_event(meta, ctx, "start", kwargs)
         t0 = time.time()
         try:
             if meta.is_async:
                 result = await meta.fn(**kwargs)
             else:
                 result = await asyncio.to_thread(meta.fn, **kwargs)
-            self._workflow_event(meta, ctx, "complete", kwargs, result=result,
-                                 duration_ms=(time.time() - t0) * 1e3)
+            if not cp_invoked:
+                self._workflow_event(meta, ctx, "complete", kwargs,
+                                     result=result,
+                                     duration_ms=(time.time() - t0) * 1e3)
             return result
         except Exception as e:
-            self._workflow_event(meta, ctx, "error", kwargs, error=str(e),
-                                 duration_ms=(time.time() - t0) * 1e3)
+            if not cp_invoked:
+                self._workflow_event(meta, ctx, "error", kwargs, error=str(e),
+                                     duration_ms=(time.time() - t0) * 1e3)
             raise
         finally:
             reset_context(token)
@@ -287,6 +298,32 @@ class Agent(FastAPI):
         finally:
             reset_context(token)
 
+    def _callback_session(self):
+        """Shared aiohttp session for status callbacks: posting on the loop
+        avoids a thread hop per callback, and aiohttp's C http stack costs
+        a fraction of the sync-httpx-in-thread path under load."""
+        sess = getattr(self, "_cb_session", None)
+        if sess is None or sess.closed:
+            import aiohttp
+            sess = self._cb_session = aiohttp.ClientSession(
+                timeout=aiohttp.ClientTimeout(total=30),
+                connector=aiohttp.TCPConnector(limit=256))
+        return sess
+
+    async def _report_status(self, callback_url: str | None,
+                             execution_id: str, status: str, result,
+                             err, duration_ms) -> bool:
+        base = (callback_url or self.agentfield_url).rstrip("/")
+        try:
+            async with self._callback_session().post(
+                    f"{base}/api/v1/executions/{execution_id}/status",
+                    json={"execution_id": execution_id, "status": status,
+                          "result": result, "error": err,
+                          "duration_ms": duration_ms}) as r:
+                return r.status == 200
+        except Exception:
+            return False
+
     async def _run_with_callback(self, meta: _FunctionMeta, kwargs: dict,
                                  ctx: ExecutionContext):
         t0 = time.time()
@@ -298,9 +335,9 @@ class Agent(FastAPI):
                 f"{e}\n{traceback.format_exc(limit=3)}"
         duration = (time.time() - t0) * 1e3
         for attempt in range(4):  # a lost callback strands the execution
-            ok = await asyncio.to_thread(
-                self.client.report_status, ctx.execution_id, status, payload,
-                err, duration)
+            ok = await self._report_status(ctx.callback_url,
+                                           ctx.execution_id, status, payload,
+                                           err, duration)
             if ok:
                 break
             await asyncio.sleep(0.2 * (2 ** attempt))
@@ -466,6 +503,7 @@ class Agent(FastAPI):
         if self.base_url is None:
             self.base_url = f"http://{host}:{port}"
         threading.Thread(target=self._deferred_start, daemon=True).start()
+        uvicorn_kw.setdefault("access_log", False)
         uvicorn.run(self, host=host, port=port, log_level="warning",
                     **uvicorn_kw)
 

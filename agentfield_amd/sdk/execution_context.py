@@ -15,6 +15,11 @@ class ExecutionContext:
     session_id: str | None = None
     actor_id: str | None = None
     agent_did: str | None = None
+    # where to deliver the terminal status callback; set by the control
+    # plane (X-AgentField-Callback) so callbacks land on the SAME control
+    # plane worker that holds the sync waiter (sticky routing across a
+    # multi-worker plane)
+    callback_url: str | None = None
     extras: dict = field(default_factory=dict)
 
     def to_headers(self) -> dict[str, str]:
@@ -49,6 +54,8 @@ class ExecutionContext:
                                  or get("X-Parent-Execution-ID")),
             session_id=get("x-session-id") or get("X-Session-ID"),
             actor_id=get("x-actor-id") or get("X-Actor-ID"),
+            callback_url=(get("x-agentfield-callback")
+                          or get("X-AgentField-Callback")),
         )
 
     def child(self, execution_id: str | None = None) -> "ExecutionContext":

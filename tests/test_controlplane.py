@@ -554,22 +554,31 @@ def test_node_actions_lease_protocol(cp_server, greeting_agent):
     aid = r.json()["action_id"]
     lc = httpx.get(base + "/lifecycle").json()
     assert lc["status"] == "stopping" and lc["pending_actions"] >= 1
-    # claim with a short lease
+    # claim with a LONG lease so the within-lease assertion can't race
+    # lease expiry on a loaded machine
     acts = httpx.post(base + "/actions/claim",
-                      json={"lease_s": 0.2}).json()["actions"]
+                      json={"lease_s": 60}).json()["actions"]
     assert any(a["id"] == aid and a["action"] == "stop" for a in acts)
     # second claim within the lease: nothing new
     assert httpx.post(base + "/actions/claim",
-                      json={"lease_s": 0.2}).json()["actions"] == []
-    time.sleep(0.3)  # lease expires -> claimable again
-    acts2 = httpx.post(base + "/actions/claim",
-                       json={"lease_s": 30}).json()["actions"]
-    assert any(a["id"] == aid for a in acts2)
+                      json={"lease_s": 60}).json()["actions"] == []
     assert httpx.post(base + "/actions/ack",
                       json={"action_id": aid}).status_code == 200
     # double-ack rejected
     assert httpx.post(base + "/actions/ack",
                       json={"action_id": aid}).status_code == 409
+    # expiry: a SECOND action claimed with a short lease becomes
+    # claimable again once the lease lapses
+    aid2 = httpx.post(base + "/stop").json()["action_id"]
+    acts = httpx.post(base + "/actions/claim",
+                      json={"lease_s": 0.2}).json()["actions"]
+    assert any(a["id"] == aid2 for a in acts)
+    time.sleep(0.5)  # lease expires -> claimable again
+    acts2 = httpx.post(base + "/actions/claim",
+                       json={"lease_s": 30}).json()["actions"]
+    assert any(a["id"] == aid2 for a in acts2)
+    assert httpx.post(base + "/actions/ack",
+                      json={"action_id": aid2}).status_code == 200
     # restore status for other tests
     httpx.post(base + "/start")
     httpx.post(srv.base_url + "/api/v1/nodes/greeter/status",

@@ -34,6 +34,7 @@ def pct(values, p):
 
 async def run_load(url: str, target: str, requests: int, concurrency: int,
                    mode: str, payload_bytes: int, depth: int) -> dict:
+    import aiohttp
     sem = asyncio.Semaphore(concurrency)
     latencies: list[float] = []
     statuses: dict[str, int] = {}
@@ -42,40 +43,58 @@ async def run_load(url: str, target: str, requests: int, concurrency: int,
     payload = {"name": "x" * max(1, payload_bytes)}
     if depth > 1:
         payload["depth"] = depth
+    targets = target.split(",")  # round-robin across agent replicas
+    # round-robin across control-plane workers when the plane advertises a
+    # fleet (multi-worker mode)
+    urls = [url]
+    try:
+        h = httpx.get(f"{url}/api/v1/health", timeout=2.0).json()
+        if h.get("workers"):
+            urls = h["workers"]
+    except Exception:
+        pass
 
-    async with httpx.AsyncClient(timeout=120.0) as client:
+    async with aiohttp.ClientSession(
+            timeout=aiohttp.ClientTimeout(total=120.0),
+            connector=aiohttp.TCPConnector(limit=concurrency + 32)) as client:
         async def one(i: int):
             nonlocal backpressure
             async with sem:
                 t0 = time.perf_counter()
                 try:
                     if mode == "sync":
-                        r = await client.post(f"{url}/api/v1/execute/{target}",
-                                              json={"input": payload})
-                        http_codes[r.status_code] = http_codes.get(r.status_code, 0) + 1
-                        body = r.json() if r.status_code == 200 else {}
-                        st = body.get("status", f"http_{r.status_code}")
+                        async with client.post(
+                                f"{urls[i % len(urls)]}/api/v1/execute/{targets[i % len(targets)]}",
+                                json={"input": payload}) as r:
+                            http_codes[r.status] = http_codes.get(r.status, 0) + 1
+                            body = (await r.json(content_type=None)
+                                    if r.status == 200 else {})
+                        st = body.get("status", f"http_{r.status}")
                     else:
-                        r = await client.post(
-                            f"{url}/api/v1/execute/async/{target}",
-                            json={"input": payload})
-                        http_codes[r.status_code] = http_codes.get(r.status_code, 0) + 1
-                        if r.status_code == 503:
+                        async with client.post(
+                                f"{urls[i % len(urls)]}/api/v1/execute/async/{targets[i % len(targets)]}",
+                                json={"input": payload}) as r:
+                            http_codes[r.status] = http_codes.get(r.status, 0) + 1
+                            code = r.status
+                            body = await r.json(content_type=None) \
+                                if code in (200, 202) else {}
+                        if code == 503:
                             backpressure += 1
                             st = "backpressure"
-                        elif r.status_code == 202:
-                            eid = r.json()["execution_id"]
+                        elif code == 202:
+                            eid = body["execution_id"]
                             st = "queued"
                             for _ in range(600):
-                                g = await client.get(
-                                    f"{url}/api/v1/executions/{eid}")
-                                st = g.json().get("status", "unknown")
+                                async with client.get(
+                                        f"{urls[i % len(urls)]}/api/v1/executions/{eid}") as g:
+                                    st = (await g.json(content_type=None)
+                                          ).get("status", "unknown")
                                 if st in ("completed", "failed", "timeout",
                                           "cancelled"):
                                     break
                                 await asyncio.sleep(0.05)
                         else:
-                            st = f"http_{r.status_code}"
+                            st = f"http_{code}"
                     statuses[st] = statuses.get(st, 0) + 1
                     latencies.append(time.perf_counter() - t0)
                 except Exception as e:
@@ -119,6 +138,18 @@ async def scrape_metrics(url: str) -> dict:
     return out
 
 
+def _wait_http(url: str, timeout: float = 20.0):
+    deadline = time.time() + timeout
+    while time.time() < deadline:
+        try:
+            if httpx.get(url, timeout=1.0).status_code == 200:
+                return
+        except httpx.HTTPError:
+            pass
+        time.sleep(0.05)
+    raise TimeoutError(f"{url} not up")
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--url", default="http://127.0.0.1:8520")
@@ -131,9 +162,59 @@ def main():
                     help="nested call depth (needs the relay reasoner)")
     ap.add_argument("--self-contained", action="store_true",
                     help="spin up an in-process control plane + echo agent")
+    ap.add_argument("--procs", action="store_true",
+                    help="spin up the control plane and echo agent as "
+                         "SEPARATE processes (no shared GIL — the real "
+                         "deployment topology)")
+    ap.add_argument("--cp-port", type=int, default=8520)
+    ap.add_argument("--agent-port", type=int, default=8601)
+    ap.add_argument("--agents", type=int, default=1,
+                    help="number of echo agent processes (--procs mode); "
+                         "load round-robins across them")
+    ap.add_argument("--cp-workers", type=int, default=1,
+                    help="control-plane worker processes (--procs mode)")
     args = ap.parse_args()
 
     servers = []
+    procs = []
+    if args.procs:
+        import os
+        import subprocess
+        root = Path(__file__).resolve().parent.parent
+        env = {**os.environ, "PYTHONPATH": str(root)}
+        cp_url = f"http://127.0.0.1:{args.cp_port}"
+        import tempfile
+        tmp = tempfile.mkdtemp(prefix="af-stress-")
+        procs.append(subprocess.Popen(
+            [sys.executable, "-m", "agentfield_amd", "server",
+             "--host", "127.0.0.1", "--port", str(args.cp_port),
+             "--db", f"{tmp}/af.db", "--data-dir", tmp,
+             "--workers", str(args.cp_workers)],
+            env=env, cwd=root, stdout=subprocess.DEVNULL,
+            stderr=subprocess.DEVNULL))
+        for i in range(args.cp_workers):
+            _wait_http(f"http://127.0.0.1:{args.cp_port + i}/api/v1/health")
+        names = (["echo"] if args.agents == 1
+                 else [f"echo{i}" for i in range(args.agents)])
+        for i, name in enumerate(names):
+            procs.append(subprocess.Popen(
+                [sys.executable, str(root / "tools" / "echo_agent.py"),
+                 "--port", str(args.agent_port + i), "--cp", cp_url,
+                 "--node-id", name],
+                env=env, cwd=root, stdout=subprocess.DEVNULL,
+                stderr=subprocess.DEVNULL))
+        for i in range(args.agents):
+            _wait_http(f"http://127.0.0.1:{args.agent_port + i}/health")
+        # wait for registration
+        for _ in range(200):
+            r = httpx.get(f"{cp_url}/api/v1/nodes", timeout=2.0)
+            ids = {n.get("id") for n in r.json().get("nodes", [])}
+            if all(n in ids for n in names):
+                break
+            time.sleep(0.05)
+        args.url = cp_url
+        if args.agents > 1:
+            args.target = ",".join(f"{n}.greet" for n in names)
     if args.self_contained:
         from agentfield_amd.controlplane import ControlPlane, create_app
         from agentfield_amd.controlplane.server import Config
@@ -169,6 +250,13 @@ def main():
     print(json.dumps(result, indent=2))
     for s in servers:
         s.stop()
+    for p in procs:
+        p.terminate()
+    for p in procs:
+        try:
+            p.wait(timeout=5)
+        except Exception:
+            p.kill()
 
 
 if __name__ == "__main__":
