@@ -29,6 +29,9 @@ class AIConfig:
     device: str | None = None           # None -> cuda if available
     fallback_models: tuple = ()
     json_only: bool = False             # grammar-constrained valid JSON
+    ignore_eos: bool = False            # benchmark mode: always decode
+                                        # max_tokens (random-init weights
+                                        # emit EOS at chance rate)
     extra: dict = field(default_factory=dict)
 
     def merged(self, **overrides) -> "AIConfig":
@@ -129,6 +132,7 @@ class EngineRunner:
             ids = ids[-(limit - cfg.max_tokens):]  # token-aware trim
         sp = SamplingParams(max_tokens=cfg.max_tokens,
                             temperature=cfg.temperature,
+                            ignore_eos=cfg.ignore_eos,
                             json_mode=self._check_json_mode(cfg))
         w = self.submit(ids, sp)
         if not w["done"].wait(cfg.timeout):

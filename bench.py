@@ -1,12 +1,19 @@
 #!/usr/bin/env python3
-"""Flagship serving benchmark: reasoner calls/sec on Llama-3-8B (bf16,
-synthetic prompts, random-init weights).
+"""Flagship serving benchmark: reasoner calls/sec + p50 REST latency on
+Llama-3-8B (bf16, synthetic prompts, random-init weights).
 
-One rank per GPU (DP replicas over RCCL when launched via torchrun).  A
-"step" = each rank completes a fixed batch of `--calls` reasoner calls
-end-to-end through the serving engine (continuous-batch prefill of
-`--prompt-len` synthetic tokens + `--gen-len` decoded tokens each, sampling
-included).  value = whole-job completed calls/sec across all N GPUs.
+Default mode (`--rest`, BASELINE config 2/4) measures the WHOLE stack:
+REST client -> control plane (/api/v1/execute/:target) -> agent
+@reasoner -> app.ai() -> in-process MI355X engine.  One rank per GPU
+(DP replicas when launched via torchrun); rank 0 additionally hosts the
+control-plane worker fleet (subprocess) and the load client (subprocess).
+A "step" = `--calls` reasoner calls per rank completed end-to-end
+(continuous-batch prefill of `--prompt-len` synthetic tokens + `--gen-len`
+decoded tokens each, sampling included).  value = whole-job completed
+calls/sec across all N GPUs; p50_call_ms is CLIENT-OBSERVED REST latency.
+
+`--engine-only` drives LLMEngine directly (no HTTP), isolating GPU-side
+throughput; `--tp N` shards one engine across N ranks instead.
 """
 from __future__ import annotations
 
@@ -14,7 +21,10 @@ import argparse
 import json
 import os
 import statistics
+import subprocess
 import sys
+import tempfile
+import threading
 import time
 from pathlib import Path
 
@@ -42,7 +52,204 @@ def parse_args():
                    help="tensor-parallel degree (requires WORLD_SIZE == tp; "
                         "ranks form one TP group instead of DP replicas)")
     p.add_argument("--device", default=None)
+    p.add_argument("--rest", dest="rest", action="store_true", default=None,
+                   help="measure through the REST path (default)")
+    p.add_argument("--engine-only", dest="rest", action="store_false",
+                   help="drive the engine directly, no HTTP")
+    p.add_argument("--cp-workers", type=int, default=0,
+                   help="control-plane worker processes (0 = auto)")
     return p.parse_args()
+
+
+# --------------------------------------------------------------- REST mode
+def _wait_http(url: str, timeout: float = 60.0) -> None:
+    import httpx
+    deadline = time.time() + timeout
+    while time.time() < deadline:
+        try:
+            if httpx.get(url, timeout=2.0).status_code == 200:
+                return
+        except Exception:
+            pass
+        time.sleep(0.1)
+    raise TimeoutError(f"{url} not up")
+
+
+class _AgentServer:
+    """uvicorn-in-a-thread hosting this rank's Agent (the engine stays in
+    this process: app.ai() is served by the in-process runner)."""
+
+    def __init__(self, app, port: int):
+        import uvicorn
+        self.config = uvicorn.Config(app, host="127.0.0.1", port=port,
+                                     log_level="error", access_log=False,
+                                     lifespan="off")
+        self.server = uvicorn.Server(self.config)
+        self.thread = threading.Thread(target=self.server.run, daemon=True)
+
+    def start(self):
+        self.thread.start()
+        deadline = time.time() + 30
+        while time.time() < deadline and not self.server.started:
+            time.sleep(0.02)
+        if not self.server.started:
+            raise TimeoutError("agent server did not start")
+        return self
+
+
+def run_rest(args, world: int, rank: int, dist, device: str, cfg) -> None:
+    from agentfield_amd.sdk import Agent
+    from agentfield_amd.sdk.ai import ByteTokenizer, EngineRunner, set_runner
+
+    dtype = torch.bfloat16 if device.startswith("cuda") else torch.float32
+    kw = {}
+    if not device.startswith("cuda"):
+        kw = {"num_pages": 512, "max_num_seqs": 8, "dtype": torch.float32}
+    eng = LLMEngine(cfg, device=device,
+                    max_num_seqs=kw.pop("max_num_seqs", args.max_num_seqs),
+                    max_prefill_tokens=args.prompt_len * args.calls,
+                    enable_graphs=not args.no_graphs and
+                    device.startswith("cuda"),
+                    dtype=kw.pop("dtype", dtype), seed=0, **kw)
+    runner = EngineRunner(eng, ByteTokenizer(cfg.vocab_size))
+    set_runner(args.model, runner)
+
+    port_base = 17000 + int(os.environ.get("MASTER_PORT", "29500")) % 1000
+    cp_port = port_base
+    # +500: clear of the CP worker ports (cp_port..+workers) and their
+    # admin gRPC ports (cp_port+100..)
+    agent_port = port_base + 500 + rank
+    cp_url = f"http://127.0.0.1:{cp_port}"
+    cp_workers = args.cp_workers or max(1, min(4, (os.cpu_count() or 8) // 8))
+    root = Path(__file__).resolve().parent
+
+    cp_proc = None
+    if rank == 0:
+        tmp = tempfile.mkdtemp(prefix="af-bench-")
+        cp_proc = subprocess.Popen(
+            [sys.executable, "-m", "agentfield_amd", "server",
+             "--host", "127.0.0.1", "--port", str(cp_port),
+             "--db", f"{tmp}/af.db", "--data-dir", tmp,
+             "--workers", str(cp_workers)],
+            env={**os.environ, "PYTHONPATH": str(root)}, cwd=root,
+            stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+        for i in range(cp_workers):
+            _wait_http(f"http://127.0.0.1:{cp_port + i}/api/v1/health")
+    if dist is not None:
+        dist.barrier()
+
+    from agentfield_amd.sdk.ai import AgentAI, AIConfig
+    agent = Agent(f"gpu{rank}", agentfield_url=cp_url, auto_register=False,
+                  base_url=f"http://127.0.0.1:{agent_port}",
+                  ai_config=AIConfig(model=args.model))
+    assert isinstance(agent.ai, AgentAI)
+
+    @agent.reasoner()
+    def reason(prompt: str, gen: int = 64):
+        return {"text": agent.ai(prompt, max_tokens=gen, ignore_eos=True,
+                                 temperature=0.0)}
+
+    _AgentServer(agent, agent_port).start()
+    if not agent.register():
+        raise RuntimeError(f"rank {rank}: agent registration failed")
+    if dist is not None:
+        dist.barrier()
+
+    if rank != 0:
+        # serve: step barriers pace us with rank 0's load client
+        for _ in range(args.warmup):
+            dist.barrier()
+        if device.startswith("cuda"):
+            torch.cuda.synchronize()
+        dist.barrier()  # timing start
+        for _ in range(args.steps):
+            dist.barrier()
+        if device.startswith("cuda"):
+            torch.cuda.synchronize()
+        dist.barrier()  # timing end
+        dist.barrier()  # teardown gate
+        return
+
+    # ---- rank 0: drive the load via the client subprocess
+    import httpx
+    health = httpx.get(f"{cp_url}/api/v1/health", timeout=5.0).json()
+    urls = health.get("workers") or [cp_url]
+    targets = [f"gpu{r}.reason" for r in range(world)]
+    client = subprocess.Popen(
+        [sys.executable, str(root / "tools" / "rest_load.py")],
+        stdin=subprocess.PIPE, stdout=subprocess.PIPE, text=True,
+        env={**os.environ, "PYTHONPATH": str(root)}, cwd=root)
+    spec = {"calls": args.calls * world, "targets": targets, "urls": urls,
+            "prompt_chars": max(8, args.prompt_len - 25),
+            "gen": args.gen_len, "concurrency": args.calls * world}
+
+    def one_step() -> dict:
+        client.stdin.write(json.dumps(spec) + "\n")
+        client.stdin.flush()
+        out = json.loads(client.stdout.readline())
+        if out["ok"] != spec["calls"]:
+            raise RuntimeError(f"step had failures: {out['errors']}")
+        return out
+
+    for _ in range(args.warmup):
+        one_step()
+        if dist is not None:
+            dist.barrier()
+    if device.startswith("cuda"):
+        torch.cuda.synchronize()
+    if dist is not None:
+        dist.barrier()
+    t0 = time.perf_counter()
+    lats: list[float] = []
+    for _ in range(args.steps):
+        out = one_step()
+        lats.extend(out["latencies_ms"])
+        if dist is not None:
+            dist.barrier()
+    if device.startswith("cuda"):
+        torch.cuda.synchronize()
+    if dist is not None:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    total_calls = args.calls * world * args.steps
+    value = total_calls / elapsed
+    out = {
+        "metric": "reasoner_calls_per_sec",
+        "value": round(value, 3),
+        "unit": "calls/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(elapsed / args.steps * 1000.0, 2),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16" if device.startswith("cuda") else "fp32",
+        "data": "synthetic",
+        "config": {
+            "model": cfg.name,
+            "global_batch": args.calls * world,
+            "seq_len": args.prompt_len + args.gen_len,
+            "prompt_len": args.prompt_len,
+            "gen_len": args.gen_len,
+            "parallelism": f"dp{world}",
+            "path": "rest",
+            "cp_workers": cp_workers,
+            "p50_call_ms": round(statistics.median(lats), 1) if lats else None,
+            "p95_call_ms": round(sorted(lats)[int(0.95 * len(lats))], 1)
+            if lats else None,
+            "tokens_per_sec": round(
+                total_calls * (args.prompt_len + args.gen_len) / elapsed, 1),
+        },
+    }
+    print(json.dumps(out))
+    client.stdin.close()
+    client.terminate()
+    if dist is not None:
+        dist.barrier()  # teardown gate
+    if cp_proc is not None:
+        cp_proc.terminate()
 
 
 def run_step(eng: LLMEngine, rank: int, step: int, args) -> list[float]:
@@ -73,16 +280,30 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    use_rest = (args.rest is None or args.rest) and args.tp <= 1
     dist = None
     if world > 1:
         import torch.distributed as dist_mod
         dist = dist_mod
-        dist.init_process_group("nccl" if torch.cuda.is_available() else "gloo")
+        if use_rest:
+            # DP-over-REST has no GPU collectives: gloo barriers pace the
+            # steps without touching the engine's CUDA streams
+            dist.init_process_group("gloo")
+        else:
+            dist.init_process_group(
+                "nccl" if torch.cuda.is_available() else "gloo")
         if torch.cuda.is_available():
             torch.cuda.set_device(local_rank)
 
     device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
     cfg = CONFIGS[args.model]
+    if use_rest:
+        if device == "cuda":
+            device = f"cuda:{local_rank}"
+        run_rest(args, world, rank, dist, device, cfg)
+        if dist is not None:
+            dist.destroy_process_group()
+        return
     dtype = torch.bfloat16 if device == "cuda" else torch.float32
     if args.tp > 1:
         if world != args.tp:
