@@ -30,6 +30,8 @@ HIP_SOURCES = [
     "attn_prefill.hip",
     "gemm.hip",
     "gemm_pipelined.hip",
+    "gemm_ring.hip",
+    "gemm_q8.hip",
     "gemm_skinny.hip",
     "sampling.hip",
     "probe.hip",

@@ -268,6 +268,23 @@ def test_gemm_pipelined():
         assert close(c, want, atol=tol, rtol=5e-2), f"gemm_pipe {M}x{N}x{K}"
 
 
+def test_gemm_ring():
+    # deep-pipelined ring variant (K % 32 == 0): exact tiles, ragged M/N,
+    # K shorter than the ring depth, and a real model shape
+    for (M, N, K) in [(256, 256, 32), (256, 256, 96), (512, 512, 512),
+                      (300, 700, 160), (512, 4096, 4096)]:
+        a = rnd(M, K, seed=M + N + 11, scale=0.5)
+        w = rnd(N, K, seed=M + N + 12, scale=0.5)
+        c = ops.gemm_bf16_ring(a, w)
+        torch.cuda.synchronize()
+        want = (a.float() @ w.float().t())
+        tol = 0.1 + 0.02 * math.sqrt(K)
+        assert close(c, want, atol=tol, rtol=5e-2), f"gemm_ring {M}x{N}x{K}"
+        c32 = ops.gemm_bf16_ring32(a, w)
+        torch.cuda.synchronize()
+        assert close(c32, want, atol=tol, rtol=5e-2), f"gemm_ring32 {M}x{N}x{K}"
+
+
 def test_gemm_skinny():
     for M in (1, 7, 16, 33, 64, 96, 128, 192, 256):
         for (N, K) in [(6144, 4096), (4096, 14336), (128256, 4096)]:
