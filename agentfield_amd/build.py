@@ -29,7 +29,6 @@ HIP_SOURCES = [
     "attn_decode.hip",
     "attn_prefill.hip",
     "gemm.hip",
-    "gemm_pipelined.hip",
     "gemm_ring.hip",
     "gemm_q8.hip",
     "gemm_skinny.hip",

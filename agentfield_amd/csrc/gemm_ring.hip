@@ -210,153 +210,6 @@ __global__ void __launch_bounds__(512, 1) gemm_bf16_ring_kernel(
   }
 }
 
-// ---- 32x32x16 variant ------------------------------------------------
-// Same ring/sub-phase schedule, MFMA shape 32x32x16 (measured ceiling
-// 2495 TF vs 2176 for 16x16x32 — ~8.07 cyc per 32.8 kFLOP).  Per-wave
-// output 128x64 = 4(M) x 2(N) fragments of 32x32; per chunk (K=32) two
-// K=16 steps.  A/B operand layout: lane L holds 8 contiguous K elements
-// of row (L&31), K-group (L>>5); C/D: col=lane&31,
-// row=(reg&3)+8*(reg>>2)+4*(lane>>5)  (guide §3, m74/m101-verified).
-typedef __attribute__((ext_vector_type(16))) float f32x16;
-
-__global__ void __launch_bounds__(512, 1) gemm_bf16_ring32_kernel(
-    u16* __restrict__ C, const u16* __restrict__ A, const u16* __restrict__ W,
-    int M, int N, int K, int tiles_m, int tiles_n) {
-  __shared__ u16 sA[GR_SLOTS][GR_CHUNK_U16];
-  __shared__ u16 sB[GR_SLOTS][GR_CHUNK_U16];
-
-  const int bid = xcd_swizzle(blockIdx.x, tiles_m * tiles_n);
-  const int tm = bid / tiles_n, tn = bid % tiles_n;
-  const int m0 = tm * GR_BM, n0 = tn * GR_BM;
-  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  const int wm = (wid >> 2) * 128;
-  const int wn = (wid & 3) * 64;
-
-  // frag byte offsets: row (lane&31) of m/n-block, K-slab 2*kk + (lane>>5)
-  int offA[4][2], offB[2][2];
-#pragma unroll
-  for (int mb = 0; mb < 4; ++mb)
-#pragma unroll
-    for (int kk = 0; kk < 2; ++kk)
-      offA[mb][kk] = gr_off(wm + mb * 32 + (lane & 31),
-                            kk * 2 + (lane >> 5));
-#pragma unroll
-  for (int nb = 0; nb < 2; ++nb)
-#pragma unroll
-    for (int kk = 0; kk < 2; ++kk)
-      offB[nb][kk] = gr_off(wn + nb * 32 + (lane & 31),
-                            kk * 2 + (lane >> 5));
-
-  f32x16 acc[4][2];
-#pragma unroll
-  for (int i = 0; i < 4; ++i)
-#pragma unroll
-    for (int j = 0; j < 2; ++j) acc[i][j] = (f32x16)(0.0f);
-
-  const int nchunks = K / GR_KC;
-  for (int c = 0; c < 3 && c < nchunks; ++c) {
-    gr_stage(A, K, sA[c], m0, c * GR_KC, M);
-    gr_stage(W, K, sB[c], n0, c * GR_KC, N);
-  }
-  if (nchunks >= 3) {
-    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-  } else if (nchunks == 2) {
-    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-  } else {
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  }
-  __builtin_amdgcn_s_barrier();
-
-  for (int c = 0; c < nchunks; ++c) {
-    const u16* a = sA[c & (GR_SLOTS - 1)];
-    const u16* b = sB[c & (GR_SLOTS - 1)];
-    const bool prefetch = c + 3 < nchunks;
-    const int slot = (c + 3) & (GR_SLOTS - 1);
-    s16x8 af[4], bf[4];
-    // ---- sub-phase 1: m-blocks 0,1 (8 MFMA between barriers) ----
-#pragma unroll
-    for (int mb = 0; mb < 2; ++mb)
-#pragma unroll
-      for (int kk = 0; kk < 2; ++kk) af[mb * 2 + kk] = gr_read(a, offA[mb][kk]);
-#pragma unroll
-    for (int nb = 0; nb < 2; ++nb)
-#pragma unroll
-      for (int kk = 0; kk < 2; ++kk) bf[nb * 2 + kk] = gr_read(b, offB[nb][kk]);
-    if (prefetch) gr_stage(A, K, sA[slot], m0, (c + 3) * GR_KC, M);
-    __builtin_amdgcn_s_barrier();
-    __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-    for (int mb = 0; mb < 2; ++mb)
-#pragma unroll
-      for (int nb = 0; nb < 2; ++nb)
-#pragma unroll
-        for (int kk = 0; kk < 2; ++kk)
-          acc[mb][nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              as_bfr(af[mb * 2 + kk]), as_bfr(bf[nb * 2 + kk]), acc[mb][nb],
-              0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
-    __builtin_amdgcn_s_barrier();
-    // ---- sub-phase 2: m-blocks 2,3 (B frags stay in registers) ----
-#pragma unroll
-    for (int mb = 0; mb < 2; ++mb)
-#pragma unroll
-      for (int kk = 0; kk < 2; ++kk)
-        af[mb * 2 + kk] = gr_read(a, offA[2 + mb][kk]);
-    if (prefetch) gr_stage(W, K, sB[slot], n0, (c + 3) * GR_KC, N);
-    {
-      const int ahead = min(nchunks - 2 - c, 2);
-      if (ahead >= 2) {
-        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-      } else if (ahead == 1) {
-        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-      } else {
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      }
-    }
-    __builtin_amdgcn_s_barrier();
-    __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-    for (int mb = 0; mb < 2; ++mb)
-#pragma unroll
-      for (int nb = 0; nb < 2; ++nb)
-#pragma unroll
-        for (int kk = 0; kk < 2; ++kk)
-          acc[2 + mb][nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              as_bfr(af[mb * 2 + kk]), as_bfr(bf[nb * 2 + kk]),
-              acc[2 + mb][nb], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
-    __builtin_amdgcn_s_barrier();
-  }
-
-  // epilogue: C/D layout col=lane&31, row=(reg&3)+8*(reg>>2)+4*(lane>>5)
-#pragma unroll
-  for (int mb = 0; mb < 4; ++mb) {
-#pragma unroll
-    for (int nb = 0; nb < 2; ++nb) {
-      const int ncol = n0 + wn + nb * 32 + (lane & 31);
-      if (ncol >= N) continue;
-#pragma unroll
-      for (int reg = 0; reg < 16; ++reg) {
-        const int mrow = m0 + wm + mb * 32 +
-                         (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
-        if (mrow >= M) continue;
-        C[(size_t)mrow * N + ncol] = f2bf(acc[mb][nb][reg]);
-      }
-    }
-  }
-}
-
-AF_EXPORT int af_gemm_bf16_ring32(void* C, const void* A, const void* W,
-                                  int M, int N, int K, void* stream) {
-  if (K % GR_KC || K < GR_KC) return 9004;
-  if (M == 0) return 0;
-  const int tiles_m = (M + GR_BM - 1) / GR_BM;
-  const int tiles_n = (N + GR_BM - 1) / GR_BM;
-  gemm_bf16_ring32_kernel<<<tiles_m * tiles_n, 512, 0, (hipStream_t)stream>>>(
-      (u16*)C, (const u16*)A, (const u16*)W, M, N, K, tiles_m, tiles_n);
-  return af_last_err();
-}
-
 AF_EXPORT int af_gemm_bf16_ring(void* C, const void* A, const void* W,
                                 int M, int N, int K, void* stream) {
   if (K % GR_KC || K < GR_KC) return 9004;

@@ -308,20 +308,6 @@ def gemm_bf16(a: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return c
 
 
-def gemm_bf16_pipelined(a: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
-    """256x256-tile counted-vmcnt pipelined MFMA GEMM (K % 128 == 0)."""
-    if not _on_gpu(a):
-        return (a.float() @ w.float().t()).to(a.dtype)
-    M, K = a.shape
-    N = w.shape[0]
-    c = torch.empty(M, N, dtype=a.dtype, device=a.device)
-    rc = _lib.lib().af_gemm_bf16_pipelined(_lib.ptr(c), _lib.ptr(a),
-                                           _lib.ptr(w), M, N, K,
-                                           _lib.cur_stream())
-    _lib.check(rc, "af_gemm_bf16_pipelined")
-    return c
-
-
 def gemm_bf16_ring(a: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """Deep-pipelined 256x256 ring-buffered MFMA GEMM (K % 32 == 0):
     4-slot LDS K-chunk ring, counted vmcnt (never drained in-loop),
@@ -338,18 +324,20 @@ def gemm_bf16_ring(a: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return c
 
 
-def gemm_bf16_ring32(a: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
-    """Ring GEMM with 32x32x16 MFMA fragments (higher matrix-pipe
-    ceiling: 2495 vs 2176 TF measured on the shape ubench)."""
+def gemm_bf16_q8(a: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """Block-quadrant 8-phase deep-pipelined MFMA GEMM (K % 64 == 0,
+    K >= 128): 8-slot LDS granule ring, one granule staged per phase two
+    K-tiles ahead, counted vmcnt(8), barrier-paired 16-MFMA clusters
+    under s_setprio.  1.04-1.15 PF measured (vs 0.83 for gemm.hip)."""
     if not _on_gpu(a):
         return (a.float() @ w.float().t()).to(a.dtype)
     M, K = a.shape
     N = w.shape[0]
     c = torch.empty(M, N, dtype=a.dtype, device=a.device)
-    rc = _lib.lib().af_gemm_bf16_ring32(_lib.ptr(c), _lib.ptr(a),
-                                        _lib.ptr(w), M, N, K,
-                                        _lib.cur_stream())
-    _lib.check(rc, "af_gemm_bf16_ring32")
+    rc = _lib.lib().af_gemm_bf16_q8(_lib.ptr(c), _lib.ptr(a),
+                                    _lib.ptr(w), M, N, K,
+                                    _lib.cur_stream())
+    _lib.check(rc, "af_gemm_bf16_q8")
     return c
 
 

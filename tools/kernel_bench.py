@@ -39,22 +39,12 @@ def bench_gemm():
         row = {"op": "gemm_bf16", "MNK": [M, N, K],
                "ours_tflops": round(tf / t_ours, 1),
                "hipblaslt_tflops": round(tf / t_blas, 1)}
-        if K % 128 == 0 and M >= 256:
-            t_pipe = timeit(lambda: ops.gemm_bf16_pipelined(a, w))
-            row["pipelined_tflops"] = round(tf / t_pipe, 1)
         if K % 32 == 0 and M >= 256:
             t_ring = timeit(lambda: ops.gemm_bf16_ring(a, w))
             row["ring_tflops"] = round(tf / t_ring, 1)
-            t_r32 = timeit(lambda: ops.gemm_bf16_ring32(a, w))
-            row["ring32_tflops"] = round(tf / t_r32, 1)
         if K % 64 == 0 and K >= 128 and M >= 256:
-            from agentfield_amd.ops import _lib as L
-            def q8():
-                c = torch.empty(M, N, dtype=a.dtype, device=a.device)
-                L.check(L.lib().af_gemm_bf16_q8(L.ptr(c), L.ptr(a), L.ptr(w),
-                                                M, N, K, L.cur_stream()), "q8")
-                return c
-            row["q8_tflops"] = round(tf / timeit(q8), 1)
+            t_q8 = timeit(lambda: ops.gemm_bf16_q8(a, w))
+            row["q8_tflops"] = round(tf / t_q8, 1)
         print(json.dumps(row))
 
 
