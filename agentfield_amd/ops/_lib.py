@@ -55,6 +55,11 @@ def _declare(l: ctypes.CDLL) -> None:
     l.af_gemm_bf16.argtypes = [p, p, p, i, i, i, p]
     l.af_gemm_bf16_ring.argtypes = [p, p, p, i, i, i, p]
     l.af_gemm_bf16_q8.argtypes = [p, p, p, i, i, i, p]
+    l.af_ipc_get_handle.argtypes = [p, p]
+    l.af_ipc_open_handle.argtypes = [p, p]
+    l.af_ipc_close_handle.argtypes = [p]
+    l.af_oneshot_allreduce.argtypes = [p, p, p, i, i, ctypes.c_long,
+                                       ctypes.c_long, ctypes.c_ulonglong, p]
     l.af_gemm_skinny.argtypes = [p, p, p, p, p, i, i, i, i, i,
                                  p, f, p, p]
     l.af_sample.argtypes = [p, p, p, p, p, p, u32, i, i, p]
@@ -66,7 +71,7 @@ def _declare(l: ctypes.CDLL) -> None:
     l.af_device_sync.argtypes = []
     for fn in ("af_rmsnorm", "af_rope_cache", "af_silu_mul", "af_add",
                "af_reshape_and_cache", "af_embedding", "af_attn_decode",
-               "af_attn_prefill", "af_gemm_bf16", "af_gemm_bf16_ring", "af_gemm_bf16_q8",
+               "af_attn_prefill", "af_gemm_bf16", "af_gemm_bf16_ring", "af_gemm_bf16_q8", "af_oneshot_allreduce",
                "af_gemm_skinny", "af_sample",
                "af_sample_topkp", "af_gather_rows", "af_mfma_probe", "af_axpy",
                "af_device_sync"):

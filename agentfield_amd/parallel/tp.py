@@ -97,25 +97,33 @@ def shard_llama_weights(model: LlamaForCausalLM, full_cfg: LlamaConfig,
 
 
 class _TPLayerHook:
-    """Inserts the RCCL all-reduce after attn-O and MLP-down projections."""
+    """Inserts the all-reduce after attn-O and MLP-down projections.
 
-    def __init__(self, group):
+    Small bf16 decode tensors go through the one-shot xGMI kernel when
+    its init self-test passed (parallel/oneshot.py); everything else —
+    prefill-sized tensors, capture-time calls — uses RCCL."""
+
+    def __init__(self, group, oneshot=None):
         self.group = group
+        self.oneshot = oneshot
 
     def __call__(self, module, inputs, output):
+        if self.oneshot is not None and self.oneshot.eligible(output):
+            return self.oneshot.allreduce(output)
         dist.all_reduce(output, group=self.group)
         return output
 
 
 def build_tp_model(full_cfg: LlamaConfig, tp: int, rank: int, device,
                    dtype=torch.bfloat16, group=None,
-                   base_seed: int = 0) -> LlamaForCausalLM:
+                   base_seed: int = 0,
+                   oneshot=None) -> LlamaForCausalLM:
     """Construct this rank's shard of the model with collectives attached."""
     shard_cfg = full_cfg.shard(tp)
     model = LlamaForCausalLM(shard_cfg, device=device, dtype=dtype)
     shard_llama_weights(model, full_cfg, tp, rank, base_seed)
     if tp > 1:
-        hook = _TPLayerHook(group)
+        hook = _TPLayerHook(group, oneshot)
         for layer in model.layers:
             layer.attn.register_forward_hook(hook)
             layer.mlp.register_forward_hook(hook)
@@ -143,11 +151,28 @@ class TPEngineGroup:
         self.rank = dist.get_rank(group) if dist.is_initialized() else 0
         self.tp = dist.get_world_size(group) if dist.is_initialized() else 1
         if self.tp > 1:
-            # RCCL-inside-hipGraph is opt-in until verified on an 8-GPU box
-            engine_kw.setdefault(
-                "enable_graphs", os.environ.get("AF_TP_GRAPHS", "0") == "1")
+            # RCCL-inside-hipGraph: AF_TP_GRAPHS=0/1 forces; otherwise a
+            # runtime self-test captures+replays a real all-reduce in a
+            # hipGraph and enables graphs only when every rank verifies
+            # the replayed result (graph-capture support differs across
+            # RCCL builds — probe, don't assume)
+            env = os.environ.get("AF_TP_GRAPHS")
+            if env is not None:
+                engine_kw.setdefault("enable_graphs", env == "1")
+            elif torch.device(device).type == "cuda":
+                engine_kw.setdefault("enable_graphs",
+                                     self._rccl_graph_selftest(device))
+            else:
+                engine_kw.setdefault("enable_graphs", False)
+        oneshot = None
+        if self.tp > 1 and torch.device(device).type == "cuda" and \
+                os.environ.get("AF_ONESHOT_AR", "1") != "0":
+            from .oneshot import OneShotAllReduce
+            cand = OneShotAllReduce(group, device)
+            oneshot = cand if cand.ok else None
+        self.oneshot = oneshot
         model = build_tp_model(full_cfg, self.tp, self.rank, device, dtype,
-                               group, base_seed)
+                               group, base_seed, oneshot=oneshot)
         shard_cfg = model.cfg
         if self.tp > 1 and "num_pages" not in engine_kw and \
                 torch.device(device).type == "cuda":
@@ -163,23 +188,73 @@ class TPEngineGroup:
         self.engine = LLMEngine(shard_cfg, device=device, dtype=dtype,
                                 model=model, tp_group=group, **engine_kw)
 
+    def _rccl_graph_selftest(self, device) -> bool:
+        """Capture one RCCL all-reduce in a hipGraph and verify a replay
+        on every rank; all ranks must agree (MIN-reduce of the verdict)
+        before graphs are enabled."""
+        ok = 0
+        try:
+            x = torch.ones(64, device=device)
+            dist.all_reduce(x, group=self.group)  # warm RCCL comms
+            torch.cuda.synchronize()
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                dist.all_reduce(x, group=self.group)
+            torch.cuda.current_stream().wait_stream(s)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                dist.all_reduce(x, group=self.group)
+            x.fill_(1.0)
+            g.replay()
+            torch.cuda.synchronize()
+            ok = int(abs(x[0].item() - self.tp) < 1e-3)
+        except Exception:
+            ok = 0
+        verdict = torch.tensor([ok], dtype=torch.int32, device=device)
+        dist.all_reduce(verdict, op=dist.ReduceOp.MIN, group=self.group)
+        return bool(int(verdict.item()))
+
     # ---- rank-0 request API -------------------------------------------
     def submit(self, prompt_ids, sampling) -> int | None:
         """Called with identical args on every rank (bench/test mode), or
         on rank 0 only followed by sync_requests()."""
         return self.engine.add_request(prompt_ids, sampling)
 
+    def _bcast_device(self):
+        """Device RCCL payloads must live on (gloo broadcasts CPU)."""
+        if dist.get_backend(self.group) == "nccl":
+            return torch.device("cuda", torch.cuda.current_device())
+        return torch.device("cpu")
+
     def broadcast_and_submit(self, requests: list | None):
-        """Rank 0 passes its new requests; other ranks pass None."""
+        """Rank 0 passes its new requests; other ranks pass None.
+
+        No pickle on the hot path: requests ride a flat
+        int32 tensor (parallel/wire.py) — one length broadcast, and a
+        payload broadcast only when there are new requests."""
         if self.tp == 1:
             out = []
             for (ids, sp) in requests or []:
                 out.append(self.engine.add_request(ids, sp))
             return out
-        obj = [requests if self.rank == 0 else None]
-        dist.broadcast_object_list(obj, src=0, group=self.group)
+        from .wire import decode_requests, encode_requests
+        dev = self._bcast_device()
+        if self.rank == 0:
+            payload = encode_requests(requests or []).to(dev)
+            size = torch.tensor([payload.numel()], dtype=torch.int64,
+                                device=dev)
+        else:
+            size = torch.zeros(1, dtype=torch.int64, device=dev)
+        dist.broadcast(size, src=0, group=self.group)
+        n = int(size.item())
+        if n <= 1:  # encode([]) == [0]: nothing new this step
+            return []
+        if self.rank != 0:
+            payload = torch.empty(n, dtype=torch.int32, device=dev)
+        dist.broadcast(payload, src=0, group=self.group)
         rids = []
-        for (ids, sp) in obj[0] or []:
+        for (ids, sp) in decode_requests(payload.cpu()):
             rids.append(self.engine.add_request(ids, sp))
         return rids
 

@@ -1,0 +1,60 @@
+"""Fixed-layout tensor codec for TP request sync.
+
+The TP group must make identical scheduling decisions, so rank 0
+broadcasts each batch of newly-submitted requests before stepping.
+Pickle (`broadcast_object_list`) on the decode hot path costs Python
+serialization per step at TP=8; this codec packs requests into one int32
+tensor so the broadcast is a plain RCCL payload.
+
+Layout: [n_requests] then per request a 9-word header
+  [prompt_len, max_tokens, temp_u, top_k, top_p_u, ignore_eos,
+   logprobs, json_mode, n_stop]
+followed by prompt ids then stop token ids.  Float fields are scaled by
+1e6 (temperature/top_p resolution far below sampling noise).
+"""
+from __future__ import annotations
+
+import torch
+
+from ..engine import SamplingParams
+
+_SCALE = 1_000_000
+_HDR = 9
+
+
+def encode_requests(requests: list[tuple[list[int], SamplingParams]]
+                    ) -> torch.Tensor:
+    words: list[int] = [len(requests)]
+    for ids, sp in requests:
+        stop = list(sp.stop_token_ids or ())
+        words += [len(ids), sp.max_tokens, int(sp.temperature * _SCALE),
+                  sp.top_k, int(sp.top_p * _SCALE), int(sp.ignore_eos),
+                  sp.logprobs, int(sp.json_mode), len(stop)]
+        words += list(ids)
+        words += stop
+    return torch.tensor(words, dtype=torch.int32)
+
+
+def decode_requests(t: torch.Tensor
+                    ) -> list[tuple[list[int], SamplingParams]]:
+    w = t.tolist()
+    n = w[0]
+    out = []
+    at = 1
+    for _ in range(n):
+        (plen, max_tokens, temp_u, top_k, top_p_u, ignore_eos,
+         logprobs, json_mode, n_stop) = w[at:at + _HDR]
+        at += _HDR
+        ids = w[at:at + plen]
+        at += plen
+        stop = tuple(w[at:at + n_stop])
+        at += n_stop
+        sp = SamplingParams(max_tokens=max_tokens,
+                            temperature=temp_u / _SCALE,
+                            top_k=top_k, top_p=top_p_u / _SCALE,
+                            stop_token_ids=stop,
+                            ignore_eos=bool(ignore_eos),
+                            logprobs=logprobs,
+                            json_mode=bool(json_mode))
+        out.append((ids, sp))
+    return out
