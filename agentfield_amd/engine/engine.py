@@ -71,15 +71,21 @@ class LLMEngine:
             max_num_seqs=max_num_seqs, max_prefill_tokens=max_prefill_tokens,
             page_size=page_size, num_pages=num_pages, max_waiting=max_waiting)
         if prefix_cache:
-            # Python impl is the round-1 default (GPU-validated); the C++
-            # NativeScheduler's prefix mode is lockstep-pinned to it
-            # (tests/test_native_scheduler.py) and opted in via env.
+            # The C++ NativeScheduler's prefix mode is the default (round-2:
+            # lockstep-pinned to the Python oracle in
+            # tests/test_native_scheduler.py and GPU-soaked by the DAG
+            # bench); AF_NATIVE_PREFIX=0 falls back to the Python impl.
             import os
-            if os.environ.get("AF_NATIVE_PREFIX") == "1":
-                from .scheduler import NativeSchedulerAdapter
-                self.sched = NativeSchedulerAdapter(sched_cfg,
-                                                    prefix_cache=True)
-            else:
+            use_native = os.environ.get("AF_NATIVE_PREFIX", "1") != "0"
+            self.sched = None
+            if use_native:
+                try:
+                    from .scheduler import NativeSchedulerAdapter
+                    self.sched = NativeSchedulerAdapter(sched_cfg,
+                                                        prefix_cache=True)
+                except ImportError:
+                    pass  # extension not built: Python fallback
+            if self.sched is None:
                 from .prefix_cache import PrefixCachingScheduler
                 self.sched = PrefixCachingScheduler(sched_cfg)
         else:
