@@ -97,9 +97,9 @@ class LLMEngine:
         self._prefilling: list[Sequence] = []
         self._finished: dict[int, Sequence] = {}
         self.sampler = ops.SamplerState(max_num_seqs, self.device, seed=seed or 0x5EED)
-        # MoE routing shapes are data-dependent -> decode stays eager
-        self.enable_graphs = (enable_graphs and self.is_gpu
-                              and cfg.num_experts == 1)
+        # MoE decode graph-captures too: the static-capacity dispatch
+        # (models/llama.py MoEMLP) keeps every decode shape fixed
+        self.enable_graphs = enable_graphs and self.is_gpu
         self._graphs: dict[int, dict] = {}
         self.metrics = {"prefill_tokens": 0, "decode_tokens": 0, "steps": 0,
                         "prefill_steps": 0, "decode_steps": 0,

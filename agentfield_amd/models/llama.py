@@ -212,11 +212,17 @@ class MoEMLP(nn.Module):
     we add MoE as a second model family next to dense Llama).
 
     Softmax router over num_experts, top-k per token with renormalized
-    gates, token-dropless eager dispatch: each expert runs the SAME fused
-    gate_up(+SwiGLU)/down ops as the dense MLP on its token subset, so the
-    skinny-GEMM decode kernels serve MoE decode shapes unchanged.  Routing
-    shapes are data-dependent, so MoE decode is not hipGraph-captured
-    (static-capacity capture is the round-2 optimization)."""
+    gates.  Two dispatch modes:
+
+    * prefill (exact, eager): each expert runs the fused gate_up/down on
+      its routed token subset — FLOP-proportional, shapes data-dependent.
+    * decode (static, hipGraph-capturable): every expert runs on every
+      token with routing weights zeroed for unrouted pairs.  Decode
+      GEMMs are weight-streaming-bound and a top-2-of-8 batch >= ~16
+      touches essentially every expert anyway, so streaming all expert
+      weights costs what exact dispatch costs — while every shape stays
+      static: one grouped gate_up GEMM [T, E*2I], one fused SwiGLU over
+      [T*E, 2I], one batched down GEMM, one weighted sum."""
 
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
@@ -226,13 +232,25 @@ class MoEMLP(nn.Module):
         self.gate_up = nn.Parameter(torch.empty(E, 2 * inter, H))
         self.down = nn.Parameter(torch.empty(E, H, inter))
 
-    def forward(self, x):
+    def forward(self, x, static: bool = False):
         # fp32 routing (bf16 softmax near-ties destabilize expert choice)
+        E = self.router.shape[0]
         probs = torch.softmax(x.float() @ self.router.float().t(), dim=-1)
         topv, topi = probs.topk(self.top_k, dim=-1)
         topv = topv / topv.sum(dim=-1, keepdim=True)
+        if static:
+            T, H = x.shape
+            I2 = self.gate_up.shape[1]
+            w = torch.zeros(T, E, dtype=torch.float32, device=x.device)
+            w.scatter_(1, topi, topv)
+            gu = ops.linear(x, self.gate_up.reshape(E * I2, H))
+            act = ops.silu_and_mul(gu.reshape(T * E, I2))
+            act = act.reshape(T, E, I2 // 2).transpose(0, 1)  # [E, T, I]
+            ye = torch.bmm(act, self.down.transpose(1, 2))    # [E, T, H]
+            out = (ye.float() * w.t().unsqueeze(-1)).sum(0)
+            return out.to(x.dtype)
         out = torch.zeros(x.shape, dtype=torch.float32, device=x.device)
-        for e in range(self.router.shape[0]):
+        for e in range(E):
             sel, slot = (topi == e).nonzero(as_tuple=True)
             if sel.numel() == 0:
                 continue
@@ -260,7 +278,14 @@ class LlamaLayer(nn.Module):
             h, residual = ops.rmsnorm(h, self.input_norm, self.eps, residual)
         h = self.attn(h, positions, rope_tab, kv, md)
         h, residual = ops.rmsnorm(h, self.post_norm, self.eps, residual)
-        h = self.mlp(h)
+        if isinstance(self.mlp, MoEMLP):
+            # decode always uses the static-capacity dispatch: measured on
+            # MI355X it beats the exact per-expert loop at every decode
+            # batch up to 256 (the loop's nonzero() host syncs dominate),
+            # and it is the only hipGraph-capturable form
+            h = self.mlp(h, static=not md.is_prefill)
+        else:
+            h = self.mlp(h)
         return h, residual
 
     def forward_decode_fused(self, residual, ss, ss2, positions, rope_tab,

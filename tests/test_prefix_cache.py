@@ -10,10 +10,23 @@ from agentfield_amd.engine import LLMEngine, SamplingParams
 from agentfield_amd.models import CONFIGS
 
 
-def make(prefix_cache=True, num_pages=64, **kw):
-    return LLMEngine(CONFIGS["tiny"], device="cpu", dtype=torch.float32,
-                     page_size=4, num_pages=num_pages, max_num_seqs=4,
-                     enable_graphs=False, prefix_cache=prefix_cache, **kw)
+def make(prefix_cache=True, num_pages=64, native=False, **kw):
+    """White-box tests of the PYTHON PrefixCachingScheduler (the oracle
+    the C++ NativeScheduler is lockstep-pinned to) — force it explicitly
+    now that the native scheduler is the default.  native=True keeps the
+    default C++ path."""
+    import os
+    old = os.environ.get("AF_NATIVE_PREFIX")
+    os.environ["AF_NATIVE_PREFIX"] = "1" if native else "0"
+    try:
+        return LLMEngine(CONFIGS["tiny"], device="cpu", dtype=torch.float32,
+                         page_size=4, num_pages=num_pages, max_num_seqs=4,
+                         enable_graphs=False, prefix_cache=prefix_cache, **kw)
+    finally:
+        if old is None:
+            os.environ.pop("AF_NATIVE_PREFIX", None)
+        else:
+            os.environ["AF_NATIVE_PREFIX"] = old
 
 
 PROMPT = [7, 3, 9, 1, 8, 2, 6, 4, 5, 9, 2, 7, 1, 3, 8, 6, 4, 2, 9, 5, 7, 1]
@@ -96,7 +109,7 @@ def test_native_prefix_engine_end_to_end(monkeypatch):
     ref = make(prefix_cache=False)
     base = [ref.generate([PROMPT], SP)[0], ref.generate([PROMPT], SP)[0]]
     monkeypatch.setenv("AF_NATIVE_PREFIX", "1")
-    eng = make()
+    eng = make(native=True)
     from agentfield_amd.engine.scheduler import NativeSchedulerAdapter
     assert isinstance(eng.sched, NativeSchedulerAdapter)
     got = [eng.generate([PROMPT], SP)[0], eng.generate([PROMPT], SP)[0]]
