@@ -347,17 +347,27 @@ class LLMEngine:
                                                  tv[i][:nreq]))})
 
     # -- grammar-constrained JSON decoding ----------------------------------
+    # Byte tokenizer: the byte PDA masks directly (jsonfsm.py).  HF/BPE
+    # vocabularies: attach a TokenJsonGrammar (set_token_grammar) and the
+    # same seam masks whole tokens (token_grammar.py).
+    def set_token_grammar(self, grammar) -> None:
+        self.token_grammar = grammar
+
     def _json_allowed(self, seq: Sequence) -> list[int]:
         """Token ids legal for seq's next token under the JSON grammar,
         completable within its remaining budget.  The FSM is cached on the
         sequence and rebuilt after preemption (output_ids reset)."""
         from .jsonfsm import EOS_ID, JsonFSM
+        grammar = getattr(self, "token_grammar", None)
         fsm = getattr(seq, "_fsm", None)
         pos = getattr(seq, "_fsm_pos", 0)
         if fsm is None or pos > len(seq.output_ids):
             fsm, pos = JsonFSM(), 0
         try:
             for tid in seq.output_ids[pos:]:
+                if grammar is not None:
+                    grammar.advance_token(fsm, tid)
+                    continue
                 b = tid - 4  # ByteTokenizer offset
                 if 0 <= b < 256:
                     fsm.advance(b)
@@ -375,6 +385,8 @@ class LLMEngine:
             return list(range(self.cfg.vocab_size))
         seq._fsm, seq._fsm_pos = fsm, len(seq.output_ids)
         remaining = seq.sampling.max_tokens - len(seq.output_ids)
+        if grammar is not None:
+            return grammar.allowed_token_ids(fsm, remaining)
         ids = fsm.allowed_token_ids(remaining)
         return [self.eos_id if i == EOS_ID else i for i in ids]
 

@@ -240,6 +240,8 @@ class JsonFSM:
             if c in _WS:
                 return
             if b == ord(","):
+                if not self.stack:
+                    raise ValueError("',' outside any container")
                 top = self.stack[-1]
                 self.state = OBJ_KEY if top == ord("{") else VAL
             elif b == ord("}") and self.stack and self.stack[-1] == ord("{"):

@@ -115,14 +115,28 @@ class EngineRunner:
         self._wake.set()
         return waiter
 
+    def ensure_json_support(self) -> None:
+        """Make grammar-constrained JSON decoding available: byte
+        tokenizers mask bytes directly; HF/BPE tokenizers get a
+        token-level grammar compiled (once) from their vocabulary and
+        attached to the engine (engine/token_grammar.py)."""
+        if isinstance(self.tokenizer, ByteTokenizer):
+            return
+        if getattr(self.engine, "token_grammar", None) is not None:
+            return
+        from ..engine.token_grammar import (TokenJsonGrammar,
+                                            vocab_bytes_from_hf)
+        tok = getattr(self.tokenizer, "tok", None)
+        if tok is None:
+            raise ValueError("json_only needs a byte or HF tokenizer")
+        vocab = vocab_bytes_from_hf(tok, self.engine.cfg.vocab_size)
+        eos = self.tokenizer.eos_id
+        eos = self.engine.eos_id if eos is None else eos
+        self.engine.set_token_grammar(TokenJsonGrammar(vocab, eos_id=eos))
+
     def _check_json_mode(self, cfg: AIConfig) -> bool:
-        """Grammar-constrained decoding masks BYTE token ids: exact for
-        the byte tokenizer, nonsense for multi-byte HF vocabularies
-        (token-level grammar compilation is round-2) — fail loudly."""
-        if cfg.json_only and not isinstance(self.tokenizer, ByteTokenizer):
-            raise ValueError(
-                "json_only requires the byte tokenizer; the configured HF "
-                "tokenizer needs token-level grammar support (roadmap)")
+        if cfg.json_only:
+            self.ensure_json_support()
         return cfg.json_only
 
     def generate_text(self, prompt: str, cfg: AIConfig) -> str:

@@ -20,8 +20,14 @@ from ..models import CONFIGS
 from ..sdk.ai import ByteTokenizer, EngineRunner, load_tokenizer
 
 
-def _byte_tok(runner: EngineRunner) -> bool:
-    return isinstance(runner.tokenizer, ByteTokenizer)
+def _json_ready(runner: EngineRunner) -> bool:
+    """Grammar-constrained JSON: byte tokenizers mask bytes; HF
+    tokenizers get a token-level grammar compiled on first use."""
+    try:
+        runner.ensure_json_support()
+        return True
+    except Exception:
+        return False
 
 
 def create_engine_app(runner: EngineRunner, model_name: str,
@@ -106,9 +112,9 @@ def create_engine_app(runner: EngineRunner, model_name: str,
         body = await req.json()
         runner, _mname = pick(body)
         jm = bool(body.get("json_mode", False))
-        if jm and not _byte_tok(runner):
+        if jm and not _json_ready(runner):
             return JSONResponse(
-                {"error": "json_mode requires the byte tokenizer"},
+                {"error": "json_mode unsupported for this tokenizer"},
                 status_code=400)
         sp = SamplingParams(
             max_tokens=int(body.get("max_tokens", 128)),
@@ -230,10 +236,10 @@ def create_engine_app(runner: EngineRunner, model_name: str,
             lp_n = min(int(body.get("logprobs", 0) or 0), 8)
         rf = body.get("response_format") or {}
         jm = rf.get("type") == "json_object"
-        if jm and not _byte_tok(runner):
+        if jm and not _json_ready(runner):
             return JSONResponse(
                 {"error": {"message": "response_format json_object "
-                           "requires the byte tokenizer",
+                           "unsupported for this tokenizer",
                            "type": "invalid_request_error"}},
                 status_code=400)
         sp = SamplingParams(max_tokens=max_tokens,

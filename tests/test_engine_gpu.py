@@ -340,3 +340,38 @@ def test_prefix_cache_on_gpu():
     assert len(out2) == 6
     # native scheduler (default): pages held only by the cache remain
     assert eng.sched.cache_pages == 5
+
+
+def test_json_mode_bpe_vocab_on_gpu():
+    """Token-level grammar over a multi-byte (BPE-style) vocabulary on
+    the HIP kernels: every sampled output must decode to valid JSON
+    (VERDICT r1 #8 — the guarantee now extends beyond the byte
+    tokenizer)."""
+    import json as _json
+    import random
+    from agentfield_amd.engine.token_grammar import TokenJsonGrammar
+    from test_token_grammar import FakeBPE
+    tok = FakeBPE()
+    cfg = CONFIGS["tiny"]
+    eng = LLMEngine(cfg, device="cuda", page_size=4, num_pages=128,
+                    max_num_seqs=4, enable_graphs=True, seed=6)
+    eng.set_token_grammar(TokenJsonGrammar(tok.vocab, eos_id=2))
+    rng = random.Random(1)
+    rids = [eng.add_request([1, 4 + rng.randrange(256), 9],
+                            SamplingParams(max_tokens=24, temperature=0.9,
+                                           json_mode=True))
+            for _ in range(6)]
+    outs = {}
+    for _ in range(600):
+        eng.step()
+        for r in rids:
+            if r not in outs:
+                f = eng.get_finished(r)
+                if f:
+                    outs[r] = f.output_ids
+        if len(outs) == len(rids):
+            break
+    assert len(outs) == len(rids)
+    for ids in outs.values():
+        body = ids[:-1] if ids and ids[-1] == 2 else ids
+        _json.loads(tok.decode(body))
