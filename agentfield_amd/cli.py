@@ -339,15 +339,68 @@ app.add_typer(vc_app, name="vc", help="Verifiable-credential tools")
 
 
 @vc_app.command("verify")
-def vc_verify(file: str):
-    """Offline verification of a VC JSON document (reference: `af vc verify`)."""
+def vc_verify(file: str, report: bool = typer.Option(
+        False, "--report", help="comprehensive scored report (chain-aware)")):
+    """Offline verification of a VC JSON document or an exported chain
+    (reference: `af vc verify` + vc_verification_enhanced.go)."""
     from .controlplane.did import VCService
     doc = json.loads(Path(file).read_text())
     if "vc" in doc:
         doc = doc["vc"]
+    if report or "credentials" in doc or isinstance(doc, list):
+        res = offline_chain_report(doc)
+        typer.echo(json.dumps(res, indent=2))
+        raise typer.Exit(0 if res["valid"] else 1)
     res = VCService.verify_document(doc)
     typer.echo(json.dumps(res, indent=2))
     raise typer.Exit(0 if res["valid"] else 1)
+
+
+def offline_chain_report(doc) -> dict:
+    """Comprehensive offline report over one VC or an exported chain
+    ({"credentials": [...]} from /api/v1/did/export/vcs): per-credential
+    signature + structure + compliance scoring without any control-plane
+    state (stored-record cross-checks require the online endpoint
+    POST /api/ui/v1/executions/:id/verify-vc)."""
+    import time as _t
+    from .controlplane.did import VCService
+
+    creds = doc.get("credentials") if isinstance(doc, dict) else doc
+    if creds is None:
+        creds = [doc]
+    svc = VCService.__new__(VCService)  # offline: no storage/dids needed
+    components = {}
+    for c in creds:
+        subj = c.get("credentialSubject", {})
+        eid = subj.get("execution_id") or c.get("id", "?")
+        rec = {"issuer_did": c.get("issuer"),
+               "execution_id": subj.get("execution_id"),
+               "run_id": subj.get("workflow_id")}
+        result = {"verification_timestamp":
+                  _t.strftime("%Y-%m-%dT%H:%M:%SZ", _t.gmtime()),
+                  "critical_issues": [], "warnings": []}
+        result["integrity_checks"] = svc._integrity_checks(rec, c, None)
+        result["security_analysis"] = svc._security_analysis(rec, c)
+        result["compliance_checks"] = svc._compliance_checks(c)
+        for sec in ("integrity_checks", "security_analysis",
+                    "compliance_checks"):
+            for issue in result[sec]["issues"]:
+                if issue["severity"] == "critical":
+                    result["critical_issues"].append(issue)
+                elif issue["severity"] == "warning":
+                    result["warnings"].append(issue)
+        result["valid"] = not result["critical_issues"]
+        result["overall_score"] = svc._score(result)
+        components[eid] = result
+    scores = [c["overall_score"] for c in components.values()]
+    return {
+        "components": components,
+        "count": len(components),
+        "valid": bool(components) and all(c["valid"]
+                                          for c in components.values()),
+        "overall_score": round(sum(scores) / len(scores), 2)
+        if scores else 0.0,
+    }
 
 
 @app.command()

