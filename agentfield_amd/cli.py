@@ -285,27 +285,151 @@ def ps():
         typer.echo(f"{pkg['name']:24s} {state}")
 
 
+@app.command("config")
+def show_config(config: str = typer.Option(None, help="YAML config file")):
+    """Show the effective control-plane configuration (env > YAML > defaults)."""
+    from .controlplane.server import Config
+    kw = {}
+    if config:
+        import yaml
+        kw = yaml.safe_load(Path(config).read_text()) or {}
+    cfg = Config(**kw)
+    typer.echo(json.dumps({k: v for k, v in vars(cfg).items()
+                           if not k.startswith("_")}, indent=2, default=str))
+
+
 mcp_app = typer.Typer()
-app.add_typer(mcp_app, name="mcp", help="Model-Context-Protocol tools")
+app.add_typer(mcp_app, name="mcp",
+              help="MCP server tools (reference C32 `af mcp` verbs)")
+
+
+def _mcp_config(project: str) -> dict:
+    from .mcp.manager import discover_config
+    return discover_config(project)
 
 
 @mcp_app.command("discover")
-def mcp_discover(project_dir: str = "."):
-    """Discover MCP server configs and their tools."""
-    from .mcp import MCPManager
-    from .mcp.manager import discover_config
-    cfg = discover_config(project_dir)
+def mcp_discover(project: str = "."):
+    """Run the discovery fallback chain (stdio -> HTTP -> static
+    analysis) for every configured server and cache capabilities."""
+    from .mcp.discovery import CapabilityCache, discover_server
+    cache = CapabilityCache(project)
+    cfg = _mcp_config(project)
     if not cfg:
-        typer.echo("no MCP config found (mcp.json)")
+        typer.echo("no MCP config (mcp.json) found", err=True)
         raise typer.Exit(1)
-    mgr = MCPManager()
+    for alias, spec in cfg.items():
+        entry = discover_server(alias, spec, project, cache)
+        typer.echo(f"{alias:20s} source={entry['source']:6s} "
+                   f"tools={[t['name'] for t in entry['tools']]}")
+
+
+@mcp_app.command("status")
+def mcp_status(project: str = "."):
+    """Configured servers, their cached capabilities, and liveness of
+    CLI-managed processes."""
+    from .mcp.discovery import CapabilityCache
+    cache = CapabilityCache(project)
+    cfg = _mcp_config(project)
+    aliases = sorted(set(cfg) | set(cache.aliases()))
+    if not aliases:
+        typer.echo("no MCP servers configured or cached")
+        return
+    for alias in aliases:
+        entry = cache.get(alias) or {}
+        pid = _mcp_pid(project, alias)
+        state = "running" if pid else "stopped"
+        tools = [t["name"] for t in entry.get("tools", [])]
+        typer.echo(f"{alias:20s} {state:8s} "
+                   f"source={entry.get('source', '-'):6s} tools={tools}")
+
+
+def _mcp_dir(project: str, alias: str) -> Path:
+    return Path(project) / ".agentfield" / "mcp" / alias
+
+
+def _mcp_pid(project: str, alias: str) -> int | None:
+    p = _mcp_dir(project, alias) / "pid"
+    if not p.exists():
+        return None
     try:
-        for name, tools in mgr.start_all(project_dir).items():
-            typer.echo(f"[{name}]")
-            for t in tools:
-                typer.echo(f"  {t.get('name')}: {t.get('description', '')[:60]}")
-    finally:
-        mgr.stop_all()
+        pid = int(p.read_text().strip())
+        os.kill(pid, 0)
+        return pid
+    except (ValueError, ProcessLookupError, PermissionError):
+        return None
+
+
+@mcp_app.command("start")
+def mcp_start(name: str, project: str = "."):
+    """Start a configured MCP server as a detached managed process
+    (pidfile + log under .agentfield/mcp/<name>/)."""
+    import subprocess
+    spec = _mcp_config(project).get(name)
+    if spec is None or not spec.get("command"):
+        typer.echo(f"no startable MCP server '{name}' configured", err=True)
+        raise typer.Exit(1)
+    if _mcp_pid(project, name):
+        typer.echo(f"{name} already running")
+        return
+    d = _mcp_dir(project, name)
+    d.mkdir(parents=True, exist_ok=True)
+    log = open(d / "server.log", "ab")
+    proc = subprocess.Popen(
+        [spec["command"], *spec.get("args", [])],
+        stdin=subprocess.PIPE, stdout=log, stderr=log,
+        cwd=spec.get("cwd"), env={**os.environ, **spec.get("env", {})},
+        start_new_session=True)
+    (d / "pid").write_text(str(proc.pid))
+    typer.echo(f"{name}: pid={proc.pid} log={d / 'server.log'}")
+
+
+@mcp_app.command("stop")
+def mcp_stop(name: str, project: str = "."):
+    pid = _mcp_pid(project, name)
+    if pid is None:
+        typer.echo(f"{name} not running")
+        return
+    import signal as _signal
+    os.kill(pid, _signal.SIGTERM)
+    (_mcp_dir(project, name) / "pid").unlink(missing_ok=True)
+    typer.echo(f"stopped {name} (pid {pid})")
+
+
+@mcp_app.command("restart")
+def mcp_restart(name: str, project: str = "."):
+    mcp_stop(name, project)
+    mcp_start(name, project)
+
+
+@mcp_app.command("logs")
+def mcp_logs(name: str, lines: int = 50, project: str = "."):
+    p = _mcp_dir(project, name) / "server.log"
+    if not p.exists():
+        typer.echo("no log yet")
+        return
+    content = p.read_text(errors="replace").splitlines()
+    typer.echo("\n".join(content[-lines:]))
+
+
+@mcp_app.command("skills")
+def mcp_skills(project: str = ".", out: str = "mcp_skills"):
+    """Generate importable skill modules from cached capabilities
+    (reference skill_generator.go)."""
+    from .mcp.discovery import CapabilityCache, generate_skill_file
+    cache = CapabilityCache(project)
+    wrote = []
+    for alias in cache.aliases():
+        entry = cache.get(alias)
+        if entry and entry.get("tools"):
+            wrote.append(str(generate_skill_file(alias, entry["tools"],
+                                                 Path(project) / out)))
+    if not wrote:
+        typer.echo("no cached capabilities — run `af mcp discover` first",
+                   err=True)
+        raise typer.Exit(1)
+    for w in wrote:
+        typer.echo(w)
 
 
 @mcp_app.command("call")
@@ -319,19 +443,6 @@ def mcp_call(tool: str, args_json: str = "{}", project_dir: str = "."):
         typer.echo(json.dumps(out, indent=2))
     finally:
         mgr.stop_all()
-
-
-@app.command("config")
-def show_config(config: str = typer.Option(None, help="YAML config file")):
-    """Show the effective control-plane configuration (env > YAML > defaults)."""
-    from .controlplane.server import Config
-    kw = {}
-    if config:
-        import yaml
-        kw = yaml.safe_load(Path(config).read_text()) or {}
-    cfg = Config(**kw)
-    typer.echo(json.dumps({k: v for k, v in vars(cfg).items()
-                           if not k.startswith("_")}, indent=2, default=str))
 
 
 vc_app = typer.Typer()

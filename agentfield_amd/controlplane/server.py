@@ -547,6 +547,13 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
         status = st.normalize(body.get("status") or "active")
         cp.storage.touch_heartbeat(node_id, "active" if status in
                                    ("active", "healthy", "completed") else None)
+        if "mcp_servers" in body:
+            # per-node MCP server health rides the enhanced heartbeat
+            # (reference: HeartbeatHandler nodes.go:646 + health_monitor
+            # MCP polling); surfaces at /api/ui/v1/mcp
+            meta = dict(node.get("metadata") or {})
+            meta["mcp_servers"] = body["mcp_servers"]
+            cp.storage.set_node_metadata(node_id, meta)
         return {"status": "ok"}
 
     @app.post("/api/v1/nodes/{node_id}/status")

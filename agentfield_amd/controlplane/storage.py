@@ -164,6 +164,10 @@ class Storage:
         self._exec("UPDATE agent_nodes SET status=?, last_status_change=? WHERE id=?",
                    (status, now(), node_id))
 
+    def set_node_metadata(self, node_id: str, metadata: dict) -> None:
+        self._exec("UPDATE agent_nodes SET metadata=? WHERE id=?",
+                   (json.dumps(metadata), node_id))
+
     def touch_heartbeat(self, node_id: str, status: str | None = None) -> None:
         if status:
             self._exec("UPDATE agent_nodes SET last_heartbeat=?, status=? WHERE id=?",

@@ -23,18 +23,48 @@ def discover_config(project_dir: str) -> dict:
 
 class MCPManager:
     def __init__(self):
-        self.servers: dict[str, MCPStdioClient] = {}
+        self.servers: dict[str, object] = {}
+        self.specs: dict[str, dict] = {}
         self.tools: dict[str, tuple[str, dict]] = {}  # tool -> (server, schema)
 
     def start_server(self, name: str, spec: dict) -> list[dict]:
-        cmd = [spec["command"], *spec.get("args", [])]
-        client = MCPStdioClient(cmd, env=spec.get("env"), cwd=spec.get("cwd"))
+        if spec.get("url"):
+            # HTTP JSON-RPC transport (reference tryHTTPDiscovery)
+            from .discovery import MCPHttpClient
+            client = MCPHttpClient(spec["url"])
+        else:
+            cmd = [spec["command"], *spec.get("args", [])]
+            client = MCPStdioClient(cmd, env=spec.get("env"),
+                                    cwd=spec.get("cwd"))
         client.initialize()
         self.servers[name] = client
+        self.specs[name] = spec
         tools = client.list_tools()
         for t in tools:
             self.tools[t["name"]] = (name, t)
         return tools
+
+    def stop_server(self, name: str) -> bool:
+        c = self.servers.pop(name, None)
+        if c is None:
+            return False
+        c.close()
+        self.tools = {t: v for t, v in self.tools.items() if v[0] != name}
+        return True
+
+    def restart_server(self, name: str) -> list[dict]:
+        spec = self.specs[name]
+        self.stop_server(name)
+        return self.start_server(name, spec)
+
+    def status(self) -> list[dict]:
+        """Per-server health for the enhanced heartbeat / af mcp status."""
+        return [{"name": n, "alive": bool(getattr(c, "alive", False)),
+                 "transport": "http" if self.specs.get(n, {}).get("url")
+                 else "stdio",
+                 "tools": sorted(t for t, (srv, _s) in self.tools.items()
+                                 if srv == n)}
+                for n, c in self.servers.items()]
 
     def start_all(self, project_dir: str) -> dict[str, list[dict]]:
         out = {}

@@ -440,7 +440,13 @@ class Agent(FastAPI):
 
     def _heartbeat_loop(self):
         while not self._hb_stop.wait(self.heartbeat_interval):
-            ok = self.client.heartbeat(self.node_id, {"status": "active"})
+            payload = {"status": "active"}
+            mcp = getattr(self, "mcp", None)
+            if mcp is not None:
+                # enhanced heartbeat: per-server MCP health rides along
+                # (reference P10 agent_field_handler.py:227-264)
+                payload["mcp_servers"] = mcp.status()
+            ok = self.client.heartbeat(self.node_id, payload)
             if not ok:
                 self.register()  # resilient re-register (P10)
             else:
