@@ -140,12 +140,87 @@ if __name__ == "__main__":
 '''
 
 
+FULL_TEMPLATE = '''"""Agent scaffolded by `af init --template full`.
+
+Shows the main SDK surfaces: reasoners, deterministic skills with result
+caching, app.ai() structured output, cross-agent calls, shared memory
+and lifecycle actions.  Start the control plane (`af server`), then
+`af dev {name}/` for the watch-reload loop.
+"""
+from agentfield_amd.sdk import Agent
+
+app = Agent("{name}", vc_enabled=False)
+
+
+@app.reasoner(tags=["demo"])
+def answer(question: str):
+    """LLM-backed reasoner served by the in-process MI355X engine."""
+    text = app.ai(question, system="Answer concisely.", max_tokens=128)
+    app.memory.workflow.set("last_question", question)
+    return {{"answer": text}}
+
+
+@app.reasoner()
+def extract(text: str):
+    """Structured output: the engine grammar-guarantees valid JSON."""
+    data = app.ai(f"Extract entities from: {{text}}",
+                  schema={{"type": "object",
+                           "properties": {{"entities": {{"type": "array"}}}}}})
+    return {{"data": data}}
+
+
+@app.skill(cache_results=True)
+def word_count(text: str):
+    """Deterministic skill; repeated inputs hit the result cache."""
+    return {{"words": len(text.split())}}
+
+
+@app.reasoner()
+def delegate(task: str):
+    """Nested cross-agent call — traced into the workflow DAG."""
+    return {{"handled_by": "{name}",
+             "note": "use app.call('other.reasoner', ...) to fan out"}}
+
+
+@app.on_action("reload-config")
+def reload_config(payload):
+    """Handle control-plane lifecycle actions (claim/ack lease queue)."""
+    return {{"reloaded": True}}
+
+
+if __name__ == "__main__":
+    app.serve(port=8600)
+'''
+
+FULL_README = """# {name}
+
+Scaffolded by `af init {name} --template full`.
+
+    af server                 # control plane
+    af dev {name}/            # watch-reload dev loop
+    af run {name}/agent.py    # plain run
+
+    curl -X POST localhost:8520/api/v1/execute/{name}.answer \\
+         -d '{{"input": {{"question": "hello"}}}}'
+
+Add MCP servers in `mcp.json`; discover them with `af mcp discover`.
+"""
+
+
 @app.command()
-def init(name: str, directory: str = "."):
+def init(name: str, directory: str = ".",
+         template: str = typer.Option(
+             "minimal", help="minimal | full (reference init.go templates)")):
     """Scaffold a new agent project."""
     root = Path(directory) / name
     root.mkdir(parents=True, exist_ok=True)
-    (root / "agent.py").write_text(AGENT_TEMPLATE.format(name=name))
+    if template == "full":
+        (root / "agent.py").write_text(FULL_TEMPLATE.format(name=name))
+        (root / "README.md").write_text(FULL_README.format(name=name))
+        (root / "mcp.json").write_text(json.dumps(
+            {"mcpServers": {}}, indent=2))
+    else:
+        (root / "agent.py").write_text(AGENT_TEMPLATE.format(name=name))
     (root / "agentfield.yaml").write_text(
         f"name: {name}\nentrypoint: agent.py\n")
     typer.echo(f"scaffolded {root}/agent.py — run with: af run {root}/agent.py")
@@ -168,6 +243,69 @@ def run(path: str, port: int = 8600, host: str = "127.0.0.1",
         typer.echo("no Agent instance found in module", err=True)
         raise typer.Exit(1)
     agents[0].serve(host=host, port=port)
+
+
+@app.command()
+def dev(path: str, port: int = 8600, host: str = "127.0.0.1",
+        agentfield_url: str = DEFAULT_URL,
+        poll: float = typer.Option(0.5, help="file-watch poll interval")):
+    """Watch-reload dev loop (reference C32: `af dev`, dev.go:37): run
+    the agent and restart it whenever a .py/.yaml/.json file under its
+    directory changes."""
+    import subprocess
+    import time as _t
+    p = Path(path)
+    watch_root = p if p.is_dir() else p.parent
+
+    def snapshot():
+        out = {}
+        for pat in ("**/*.py", "**/*.yaml", "**/*.json"):
+            for f in watch_root.glob(pat):
+                if ".agentfield" in f.parts or "__pycache__" in f.parts:
+                    continue
+                try:
+                    out[str(f)] = f.stat().st_mtime
+                except OSError:
+                    pass
+        return out
+
+    def spawn():
+        return subprocess.Popen(
+            [sys.executable, "-m", "agentfield_amd", "run", str(p),
+             "--port", str(port), "--host", host,
+             "--agentfield-url", agentfield_url],
+            env={**os.environ})
+
+    typer.echo(f"dev: watching {watch_root} (restart on change, ctrl-c to"
+               " stop)")
+    state = snapshot()
+    proc = spawn()
+    try:
+        while True:
+            _t.sleep(poll)
+            if proc.poll() is not None:
+                typer.echo(f"dev: agent exited rc={proc.returncode}; "
+                           "waiting for a change to restart")
+            cur = snapshot()
+            if cur != state:
+                changed = [k for k in cur
+                           if state.get(k) != cur[k]] or \
+                    [k for k in state if k not in cur]
+                typer.echo(f"dev: change in {Path(changed[0]).name} — "
+                           "restarting")
+                state = cur
+                if proc.poll() is None:
+                    proc.terminate()
+                    try:
+                        proc.wait(timeout=5)
+                    except subprocess.TimeoutExpired:
+                        proc.kill()
+                proc = spawn()
+    except KeyboardInterrupt:
+        pass
+    finally:
+        if proc.poll() is None:
+            proc.terminate()
 
 
 def _client():

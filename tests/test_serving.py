@@ -377,3 +377,89 @@ def test_chat_uses_tokenizer_template():
     finally:
         runner.shutdown()
         srv.stop()
+
+
+def test_init_full_template(tmp_path):
+    import subprocess
+    import sys as _sys
+    from pathlib import Path as _P
+    r = subprocess.run(
+        [_sys.executable, "-m", "agentfield_amd", "init", "demo",
+         "--directory", str(tmp_path), "--template", "full"],
+        capture_output=True, text=True,
+        cwd=_P(__file__).resolve().parent.parent)
+    assert r.returncode == 0, r.stderr
+    root = tmp_path / "demo"
+    assert (root / "README.md").exists() and (root / "mcp.json").exists()
+    import importlib.util
+    spec = importlib.util.spec_from_file_location("demo_agent",
+                                                  root / "agent.py")
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    assert set(mod.app._reasoners) == {"answer", "extract", "delegate"}
+    assert "word_count" in mod.app._skills
+    assert "reload-config" in mod.app._action_handlers
+
+
+def test_af_dev_watch_reload(tmp_path):
+    """`af dev` restarts the agent process when a watched file changes
+    (reference C32 dev.go watch loop)."""
+    import httpx
+    import subprocess
+    import sys as _sys
+    import time as _t
+    from pathlib import Path as _P
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    root = tmp_path / "watched"
+    root.mkdir()
+    (root / "agent.py").write_text(
+        "from agentfield_amd.sdk import Agent\n"
+        "app = Agent('dv', auto_register=False)\n"
+        "@app.reasoner()\n"
+        "def ping():\n    return {'v': 1}\n"
+        "if __name__ == '__main__':\n"
+        f"    app.serve(port={port})\n")
+    proc = subprocess.Popen(
+        [_sys.executable, "-m", "agentfield_amd", "dev", str(root),
+         "--port", str(port), "--poll", "0.2"],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        cwd=_P(__file__).resolve().parent.parent)
+    try:
+        deadline = _t.time() + 30
+        up = False
+        while _t.time() < deadline:
+            try:
+                if httpx.get(f"http://127.0.0.1:{port}/health",
+                             timeout=0.5).status_code == 200:
+                    up = True
+                    break
+            except httpx.HTTPError:
+                _t.sleep(0.2)
+        assert up, "agent never came up under af dev"
+        # change the file -> watcher must restart the server
+        _t.sleep(0.3)
+        (root / "agent.py").write_text(
+            (root / "agent.py").read_text().replace("'v': 1", "'v': 2"))
+        restarted = False
+        deadline = _t.time() + 30
+        while _t.time() < deadline:
+            try:
+                r = httpx.post(f"http://127.0.0.1:{port}/reasoners/ping",
+                               json={}, timeout=0.5)
+                if r.status_code == 200 and r.json()["result"]["v"] == 2:
+                    restarted = True
+                    break
+            except httpx.HTTPError:
+                pass
+            _t.sleep(0.3)
+        assert restarted, "agent was not reloaded with the new code"
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=5)
+        except subprocess.TimeoutExpired:
+            proc.kill()
