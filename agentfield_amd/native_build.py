@@ -12,6 +12,9 @@ EXT = sysconfig.get_config_var("EXT_SUFFIX") or ".so"
 OUT = PKG / f"_native{EXT}"
 
 
+ASAN_OUT = PKG / f"_native_asan{EXT}"
+
+
 def build(force: bool = False, verbose: bool = True) -> Path:
     if not force and OUT.exists() and OUT.stat().st_mtime > SRC.stat().st_mtime:
         return OUT
@@ -26,6 +29,27 @@ def build(force: bool = False, verbose: bool = True) -> Path:
         print("[native]", " ".join(cmd), file=sys.stderr)
     subprocess.run(cmd, check=True)
     return OUT
+
+
+def build_asan(force: bool = False, verbose: bool = True) -> Path:
+    """AddressSanitizer build of the native extension (the
+    new framework adds the sanitizer tier the reference never had).
+    Import it with LD_PRELOAD=libasan.so — see tests/test_sanitizer.py."""
+    if not force and ASAN_OUT.exists() and \
+            ASAN_OUT.stat().st_mtime > SRC.stat().st_mtime:
+        return ASAN_OUT
+    import pybind11
+    cmd = [
+        "g++", "-O1", "-g", "-fsanitize=address", "-fno-omit-frame-pointer",
+        "-shared", "-fPIC", "-std=c++17",
+        f"-I{pybind11.get_include()}",
+        f"-I{sysconfig.get_paths()['include']}",
+        str(SRC), "-lcrypto", "-o", str(ASAN_OUT),
+    ]
+    if verbose:
+        print("[native-asan]", " ".join(cmd), file=sys.stderr)
+    subprocess.run(cmd, check=True)
+    return ASAN_OUT
 
 
 if __name__ == "__main__":

@@ -10,8 +10,11 @@ python agentfield_amd/build.py
 echo "== C++ SDK =="
 make -s -C sdk/cpp
 
+echo "== sanitizer + soak tier (ASAN native build; threaded CP soak) =="
+python -m pytest tests/test_sanitizer.py -q
+
 echo "== CPU test tier (incl. gloo multi-process) =="
-python -m pytest tests/ -q -m "not gpu"
+python -m pytest tests/ -q -m "not gpu" --deselect tests/test_sanitizer.py
 
 if python -c "import torch; raise SystemExit(0 if torch.cuda.is_available() else 1)" 2>/dev/null; then
   echo "== GPU test tier (MI355X) =="
