@@ -187,25 +187,38 @@ class JsonFSM:
             else:
                 raise ValueError("expected ':'")
             return
+        # NOTE: advance() must be STRICT — the token-level grammar walk
+        # (token_grammar.py) discovers legality by calling advance() on a
+        # clone and catching ValueError, so any byte advance() silently
+        # accepts becomes a byte multi-byte tokens may smuggle into the
+        # output (found on GPU: raw control bytes inside strings).
         if s == STR:
             if b == ord('"'):
                 self.state = OBJ_COLON if self.key_str else AFTER_VAL
                 self.key_str = False
             elif b == ord("\\"):
                 self.state = STR_ESC
+            elif b < 0x20:
+                raise ValueError("raw control byte in string")
             return
         if s == STR_ESC:
             if b == ord("u"):
                 self.hex_left, self.state = 4, STR_U
-            else:
+            elif c in _ESCAPABLE:
                 self.state = STR
+            else:
+                raise ValueError(f"invalid escape {c!r}")
             return
         if s == STR_U:
+            if c not in _HEX:
+                raise ValueError(f"invalid \\u hex digit {c!r}")
             self.hex_left -= 1
             if self.hex_left == 0:
                 self.state = STR
             return
         if s == NUM_INT_START:
+            if c not in _DIGITS:
+                raise ValueError("digit required after '-'")
             self.state = NUM_ZERO if b == ord("0") else NUM_INT
             return
         if s in (NUM_INT, NUM_ZERO):
@@ -216,6 +229,8 @@ class JsonFSM:
             # NUM_INT digit: stay
             return
         if s == NUM_FRAC_START:
+            if c not in _DIGITS:
+                raise ValueError("digit required after '.'")
             self.state = NUM_FRAC
             return
         if s == NUM_FRAC:
@@ -223,15 +238,20 @@ class JsonFSM:
                 self.state = NUM_EXP_START
             return
         if s == NUM_EXP_START:
+            if c not in b"+-" + _DIGITS:
+                raise ValueError("sign or digit required after exponent")
             self.state = NUM_EXP_SIGN if c in b"+-" else NUM_EXP
             return
         if s == NUM_EXP_SIGN:
+            if c not in _DIGITS:
+                raise ValueError("digit required after exponent sign")
             self.state = NUM_EXP
             return
         if s == NUM_EXP:
             return  # digits stay
         if s == LIT:
-            assert c == self.lit[:1]
+            if c != self.lit[:1]:
+                raise ValueError(f"literal expects {self.lit[:1]!r}")
             self.lit = self.lit[1:]
             if not self.lit:
                 self.state = AFTER_VAL

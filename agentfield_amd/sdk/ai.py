@@ -296,17 +296,22 @@ def get_runner(cfg: AIConfig) -> EngineRunner:
             return runner
         device = cfg.device or ("cuda" if torch.cuda.is_available() else "cpu")
         model_cfg = CONFIGS[cfg.model]  # KeyError -> fallback chain
+        tokenizer = load_tokenizer()
         kw = {}
         if device == "cpu":
             kw = {"num_pages": 512, "max_num_seqs": 8, "enable_graphs": False,
                   "dtype": torch.float32}
+        if tokenizer.eos_id is not None:
+            # the engine's stop check must use THIS vocabulary's EOS (an
+            # HF tokenizer's differs from the byte tokenizer's 2)
+            kw["eos_id"] = tokenizer.eos_id
         # agent workloads repeat system prompts on every ai() call: prefix
         # caching turns that repeated prefill into a block-table lookup
         # (opt out with AGENTFIELD_NO_PREFIX_CACHE=1)
         kw["prefix_cache"] = os.environ.get(
             "AGENTFIELD_NO_PREFIX_CACHE") != "1"
         eng = LLMEngine(model_cfg, device=device, **kw)
-        runner = EngineRunner(eng, load_tokenizer())
+        runner = EngineRunner(eng, tokenizer)
         _runners[key] = runner
         return runner
 

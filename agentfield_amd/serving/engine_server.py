@@ -426,13 +426,16 @@ def main():
         if not device.startswith("cuda"):
             kw = {"num_pages": 512, "max_num_seqs": 8,
                   "dtype": torch.float32}
+        tokenizer = load_tokenizer()
+        if tokenizer.eos_id is not None:
+            kw["eos_id"] = tokenizer.eos_id  # HF vocab EOS != byte EOS
         eng = LLMEngine(
             CONFIGS[name], device=device,
             max_num_seqs=kw.pop("max_num_seqs", args.max_num_seqs),
             enable_graphs=not args.no_graphs and device.startswith("cuda"),
             spec_lookup=args.spec_lookup,
             prefix_cache=args.prefix_cache, **kw)
-        return EngineRunner(eng, load_tokenizer())
+        return EngineRunner(eng, tokenizer)
 
     runners = {name: build(name) for name in models}
     primary = models[0]

@@ -8,7 +8,7 @@ import pytest
 import torch
 
 from agentfield_amd.engine import LLMEngine, SamplingParams
-from agentfield_amd.engine.jsonfsm import JsonFSM
+from agentfield_amd.engine.jsonfsm import JsonFSM  # noqa: F401
 from agentfield_amd.engine.token_grammar import TokenJsonGrammar
 from agentfield_amd.models import CONFIGS
 
@@ -121,6 +121,33 @@ def test_engine_json_mode_with_bpe_vocab():
         body = ids[:-1] if ids and ids[-1] == 2 else ids
         text = tok.decode(body)
         json.loads(text)  # must parse — the whole point
+
+
+def test_grammar_fuzz_always_parses():
+    """Random walks sampling ONLY from allowed sets must always produce
+    parseable JSON within budget — including multi-byte tokens that
+    smuggle bytes the byte-mask path never offers (the GPU-found
+    control-chars-in-string bug class)."""
+    import random
+    tok = FakeBPE()
+    g = TokenJsonGrammar(tok.vocab, eos_id=2)
+    rng = random.Random(7)
+    for trial in range(200):
+        f = JsonFSM()
+        out = []
+        remaining = rng.choice([6, 12, 24])
+        budget = remaining
+        while budget > 0:
+            allowed = g.allowed_token_ids(f, budget)
+            assert allowed, (trial, out, f.state, f.stack)
+            t = rng.choice(allowed)
+            if t == 2:
+                break
+            g.advance_token(f, t)
+            out.append(t)
+            budget -= 1
+        text = tok.decode(out)
+        json.loads(text)  # must parse, every time
 
 
 def test_runner_attaches_grammar_for_hf_tokenizer():
