@@ -68,12 +68,26 @@ def server(host: str = "0.0.0.0", port: int = 8520,
             procs.append(subprocess.Popen(cmd, env=env))
         typer.echo(f"agentfield-amd control plane x{workers} on "
                    f"{host}:{port}-{port + workers - 1} (db={db})")
+        import signal as _signal
+
+        def _stop(*_a):
+            raise SystemExit(0)
+
+        _signal.signal(_signal.SIGTERM, _stop)
         try:
             for p in procs:
                 p.wait()
-        except KeyboardInterrupt:
+        finally:
+            # the parent dying must take the worker fleet with it
+            # (bench/stress terminate() the parent)
             for p in procs:
-                p.terminate()
+                if p.poll() is None:
+                    p.terminate()
+            for p in procs:
+                try:
+                    p.wait(timeout=5)
+                except Exception:
+                    p.kill()
         return
     import uvicorn
     from .controlplane import ControlPlane, create_app

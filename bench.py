@@ -114,13 +114,22 @@ def run_rest(args, world: int, rank: int, dist, device: str, cfg) -> None:
     runner = EngineRunner(eng, ByteTokenizer(cfg.vocab_size))
     set_runner(args.model, runner)
 
-    port_base = 17000 + int(os.environ.get("MASTER_PORT", "29500")) % 1000
+    # ports isolated per world size so the driver's back-to-back
+    # N=1,2,4,8 SCALE runs can never collide with a straggler from the
+    # previous run
+    port_base = (17000 + int(os.environ.get("MASTER_PORT", "29500")) % 1000
+                 + world * 23)
     cp_port = port_base
     # +500: clear of the CP worker ports (cp_port..+workers) and their
     # admin gRPC ports (cp_port+100..)
     agent_port = port_base + 500 + rank
     cp_url = f"http://127.0.0.1:{cp_port}"
-    cp_workers = args.cp_workers or max(1, min(4, (os.cpu_count() or 8) // 8))
+    # control-plane worker fleet sized for the DP load: ~88 calls/s per
+    # GPU, ~450+ calls/s per worker (measured) — one worker per 4 GPUs
+    # would do; give headroom without oversubscribing small CI boxes
+    auto = max(1, min(4, (os.cpu_count() or 8) // 8)) if world == 1 else \
+        max(2, min(8, world, (os.cpu_count() or 8) // 4))
+    cp_workers = args.cp_workers or auto
     root = Path(__file__).resolve().parent
 
     cp_proc = None
@@ -250,6 +259,10 @@ def run_rest(args, world: int, rank: int, dist, device: str, cfg) -> None:
         dist.barrier()  # teardown gate
     if cp_proc is not None:
         cp_proc.terminate()
+        try:
+            cp_proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            cp_proc.kill()
 
 
 def run_step(eng: LLMEngine, rank: int, step: int, args) -> list[float]:
