@@ -9,6 +9,8 @@ pat = sys.argv[2]
 rows = []
 for f in glob.glob(pat, recursive=True):
     for r in csv.DictReader(open(f)):
+        if r.get("Name") in ("KERNEL_DISPATCH", "MEMORY_COPY", "TOTAL"):
+            continue  # aggregate rows double the total
         rows.append(r)
 key = "TotalDurationNs" if rows and "TotalDurationNs" in rows[0] else "DurationNs"
 rows.sort(key=lambda r: -float(r.get(key, 0)))
