@@ -27,7 +27,7 @@
 #define AD_MAXG 8
 
 // Partial layout: po [B, Hk, NSPLIT, G, D] f32; pml [B, Hk, NSPLIT, G, 2] f32.
-template <int G>
+template <int G, bool WIDE>
 __global__ void __launch_bounds__(256) attn_decode_kernel(
     u16* __restrict__ out, float* __restrict__ po, float* __restrict__ pml,
     const u16* __restrict__ q, const u16* __restrict__ kc, const u16* __restrict__ vc,
@@ -63,40 +63,105 @@ __global__ void __launch_bounds__(256) attn_decode_kernel(
   }
 
   const i32* btrow = bt + (size_t)b * max_pages;
-  // wave w handles tokens t0 + i*16 + w*4 + tg
-  for (int tb = t0 + wid * 4; tb < t1; tb += 16) {
-    const int t = tb + tg;
-    const bool valid = t < t1;
-    const int tc = valid ? t : (t1 - 1);
-    const i64 page = btrow[tc / page_size];
-    const size_t base =
-        (((size_t)page * Hk + kvh) * page_size + (tc % page_size)) * AD_D + dl * 8;
-    const s16x8 kv8 = *reinterpret_cast<const s16x8*>(kc + base);
-    const s16x8 vv8 = *reinterpret_cast<const s16x8*>(vc + base);
-    float kf[8], vf[8];
+  if (!WIDE) {
+    // one 4-token tile per wave iteration: wins at LOW grid occupancy
+    // (<= 2 blocks/CU) where the wider loop loses throughput (measured)
+    for (int tb = t0 + wid * 4; tb < t1; tb += 16) {
+      const int t = tb + tg;
+      const bool valid = t < t1;
+      const int tc = valid ? t : (t1 - 1);
+      const i64 page = btrow[tc / page_size];
+      const size_t base = (((size_t)page * Hk + kvh) * page_size +
+                           (tc % page_size)) * AD_D + dl * 8;
+      const s16x8 kv8 = *reinterpret_cast<const s16x8*>(kc + base);
+      const s16x8 vv8 = *reinterpret_cast<const s16x8*>(vc + base);
+      float kf[8], vf[8];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) { kf[j] = bf2f((u16)kv8[j]); vf[j] = bf2f((u16)vv8[j]); }
+      for (int j = 0; j < 8; ++j) {
+        kf[j] = bf2f((u16)kv8[j]);
+        vf[j] = bf2f((u16)vv8[j]);
+      }
 #pragma unroll
-    for (int g = 0; g < G; ++g) {
-      float d = 0.f;
+      for (int g = 0; g < G; ++g) {
+        float d = 0.f;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) d += qr[g][j] * kf[j];
-      // 16-lane reduce -> s uniform within token group
-      float s = group16_sum_f32(d);
-      if (!valid) s = AF_NEG_INF;
-      // tile max across the 4 token groups
-      float tm = fmaxf(s, __shfl_xor(s, 16, 64));
-      tm = fmaxf(tm, __shfl_xor(tm, 32, 64));
-      const float mn = fmaxf(m[g], tm);
-      const float corr = (m[g] <= AF_NEG_INF) ? 0.f : __expf(m[g] - mn);
-      const float p = (s <= AF_NEG_INF) ? 0.f : __expf(s - mn);
-      float psum = p + __shfl_xor(p, 16, 64);
-      psum += __shfl_xor(psum, 32, 64);
-      l[g] = l[g] * corr + psum;
-      m[g] = mn;
+        for (int j = 0; j < 8; ++j) d += qr[g][j] * kf[j];
+        float s = group16_sum_f32(d);
+        if (!valid) s = AF_NEG_INF;
+        float tm = fmaxf(s, __shfl_xor(s, 16, 64));
+        tm = fmaxf(tm, __shfl_xor(tm, 32, 64));
+        const float mn = fmaxf(m[g], tm);
+        const float corr = (m[g] <= AF_NEG_INF) ? 0.f : __expf(m[g] - mn);
+        const float p = (s <= AF_NEG_INF) ? 0.f : __expf(s - mn);
+        float psum = p + __shfl_xor(p, 16, 64);
+        psum += __shfl_xor(psum, 32, 64);
+        l[g] = l[g] * corr + psum;
+        m[g] = mn;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) acc[g][j] = acc[g][j] * corr + p * vf[j];
+        for (int j = 0; j < 8; ++j)
+          acc[g][j] = acc[g][j] * corr + p * vf[j];
+      }
     }
+  } else {
+  // wave w handles tokens {i*32 + w*4 + tg} and {i*32 + 16 + w*4 + tg}:
+  // TWO 4-token tiles per iteration with all four 16-B K/V loads issued
+  // before any compute — the kernel is gather-latency-bound, so doubling
+  // the in-flight bytes per wave is the lever (guide: memory-level
+  // parallelism, not arithmetic, sets gather throughput)
+  auto tile_base = [&](int t) {
+    const i64 page = btrow[t / page_size];
+    return (((size_t)page * Hk + kvh) * page_size + (t % page_size)) * AD_D +
+           dl * 8;
+  };
+  for (int tb = t0 + wid * 4; tb < t1; tb += 32) {
+    const int ta = tb + tg, tb2 = tb + 16 + tg;
+    const bool va = ta < t1, vb = tb2 < t1;
+    // unconditional clamped loads: a branch around the second tile's
+    // loads stops the compiler issuing them early, which defeats the
+    // whole point (measured -25% on short chunks)
+    const size_t base_a = tile_base(va ? ta : (t1 - 1));
+    const size_t base_b = tile_base(vb ? tb2 : (t1 - 1));
+    const s16x8 ka8 = *reinterpret_cast<const s16x8*>(kc + base_a);
+    const s16x8 va8 = *reinterpret_cast<const s16x8*>(vc + base_a);
+    const s16x8 kb8 = *reinterpret_cast<const s16x8*>(kc + base_b);
+    const s16x8 vb8 = *reinterpret_cast<const s16x8*>(vc + base_b);
+    float kfa[8], vfa[8], kfb[8], vfb[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      kfa[j] = bf2f((u16)ka8[j]);
+      vfa[j] = bf2f((u16)va8[j]);
+      kfb[j] = bf2f((u16)kb8[j]);
+      vfb[j] = bf2f((u16)vb8[j]);
+    }
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const bool valid = half ? vb : va;
+      const float* kf = half ? kfb : kfa;
+      const float* vf = half ? vfb : vfa;
+#pragma unroll
+      for (int g = 0; g < G; ++g) {
+        float d = 0.f;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) d += qr[g][j] * kf[j];
+        // 16-lane reduce -> s uniform within token group
+        float s = group16_sum_f32(d);
+        if (!valid) s = AF_NEG_INF;
+        // tile max across the 4 token groups
+        float tm = fmaxf(s, __shfl_xor(s, 16, 64));
+        tm = fmaxf(tm, __shfl_xor(tm, 32, 64));
+        if (tm <= AF_NEG_INF) continue;  // whole tile invalid (tail)
+        const float mn = fmaxf(m[g], tm);
+        const float corr = (m[g] <= AF_NEG_INF) ? 0.f : __expf(m[g] - mn);
+        const float p = (s <= AF_NEG_INF) ? 0.f : __expf(s - mn);
+        float psum = p + __shfl_xor(p, 16, 64);
+        psum += __shfl_xor(psum, 32, 64);
+        l[g] = l[g] * corr + psum;
+        m[g] = mn;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[g][j] = acc[g][j] * corr + p * vf[j];
+      }
+    }
+  }
   }
   // fold the 4 token-groups' partial o (same dims, disjoint tokens)
 #pragma unroll
@@ -196,11 +261,20 @@ AF_EXPORT int af_attn_decode(void* out, void* po, void* pml, const void* q,
   if (B == 0) return 0;
   dim3 grid(B, Hk, nsplit), blk(256);
   hipStream_t st = (hipStream_t)stream;
+  const bool wide = (size_t)B * Hk * nsplit >= 1024;  // >= 4 blocks/CU
 #define AF_LAUNCH(GG)                                                            \
-  attn_decode_kernel<GG><<<grid, blk, 0, st>>>(                                  \
-      (u16*)out, (float*)po, (float*)pml, (const u16*)q, (const u16*)kc,         \
-      (const u16*)vc, (const i32*)bt, (const i32*)len, scale, Hk, page_size,     \
-      max_pages, nsplit, qs)
+  do {                                                                           \
+    if (wide)                                                                    \
+      attn_decode_kernel<GG, true><<<grid, blk, 0, st>>>(                        \
+          (u16*)out, (float*)po, (float*)pml, (const u16*)q, (const u16*)kc,     \
+          (const u16*)vc, (const i32*)bt, (const i32*)len, scale, Hk,            \
+          page_size, max_pages, nsplit, qs);                                     \
+    else                                                                         \
+      attn_decode_kernel<GG, false><<<grid, blk, 0, st>>>(                       \
+          (u16*)out, (float*)po, (float*)pml, (const u16*)q, (const u16*)kc,     \
+          (const u16*)vc, (const i32*)bt, (const i32*)len, scale, Hk,            \
+          page_size, max_pages, nsplit, qs);                                     \
+  } while (0)
   switch (G) {
     case 1: AF_LAUNCH(1); break;
     case 2: AF_LAUNCH(2); break;
