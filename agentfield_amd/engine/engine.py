@@ -362,7 +362,16 @@ class LLMEngine:
         fsm = getattr(seq, "_fsm", None)
         pos = getattr(seq, "_fsm_pos", 0)
         if fsm is None or pos > len(seq.output_ids):
-            fsm, pos = JsonFSM(), 0
+            spec = seq.sampling.json_schema
+            if spec is not None:
+                from .schemafsm import SchemaFSM, SchemaSpec
+                if not isinstance(spec, SchemaSpec):
+                    spec = SchemaSpec(spec)
+                    seq.sampling.json_schema = spec  # compile once
+                fsm = SchemaFSM(spec)
+            else:
+                fsm = JsonFSM()
+            pos = 0
         try:
             for tid in seq.output_ids[pos:]:
                 if grammar is not None:

@@ -45,8 +45,35 @@ _HEX = b"0123456789abcdefABCDEF"
 _NUM_END_STATES = (NUM_INT, NUM_ZERO, NUM_FRAC, NUM_EXP)
 
 
+# UTF-8 lead-byte table: lead -> (n_continuations, first-cont lo, hi);
+# continuations after the first are always 0x80-0xBF.  The restricted
+# first-continuation ranges reject overlong encodings and surrogates, so
+# masked output is always decodable UTF-8 (what json.loads requires).
+def _utf8_lead(b: int):
+    if 0xC2 <= b <= 0xDF:
+        return (1, 0x80, 0xBF)
+    if b == 0xE0:
+        return (2, 0xA0, 0xBF)
+    if b == 0xED:
+        return (2, 0x80, 0x9F)
+    if 0xE1 <= b <= 0xEF:
+        return (2, 0x80, 0xBF)
+    if b == 0xF0:
+        return (3, 0x90, 0xBF)
+    if b == 0xF4:
+        return (3, 0x80, 0x8F)
+    if 0xF1 <= b <= 0xF3:
+        return (3, 0x80, 0xBF)
+    return None  # ASCII handled elsewhere; C0/C1/F5+ invalid
+
+
+_UTF8_LEADS = bytes(b for b in range(0xC2, 0xF5) if _utf8_lead(b))
+_STR_ASCII = bytes(b for b in range(0x20, 0x80) if b not in b'"\\')
+
+
 class JsonFSM:
-    __slots__ = ("state", "stack", "key_str", "lit", "hex_left")
+    __slots__ = ("state", "stack", "key_str", "lit", "hex_left",
+                 "utf_left", "utf_lo", "utf_hi")
 
     def __init__(self):
         self.state = VAL
@@ -54,6 +81,9 @@ class JsonFSM:
         self.key_str = False  # current STR is an object key
         self.lit = b""
         self.hex_left = 0
+        self.utf_left = 0     # pending UTF-8 continuation bytes
+        self.utf_lo = 0x80
+        self.utf_hi = 0xBF
 
     def clone(self) -> "JsonFSM":
         f = JsonFSM.__new__(JsonFSM)
@@ -62,6 +92,9 @@ class JsonFSM:
         f.key_str = self.key_str
         f.lit = self.lit
         f.hex_left = self.hex_left
+        f.utf_left = self.utf_left
+        f.utf_lo = self.utf_lo
+        f.utf_hi = self.utf_hi
         return f
 
     # ---------------------------------------------------------- grammar
@@ -78,10 +111,9 @@ class JsonFSM:
         if s == OBJ_COLON:
             return b":" + _WS
         if s == STR:
-            # any byte except the control range, '"' and '\' end/escape
-            ordinary = bytes(b for b in range(0x20, 0x100)
-                             if b not in b'"\\')
-            return b'"\\' + ordinary
+            if self.utf_left:
+                return bytes(range(self.utf_lo, self.utf_hi + 1))
+            return b'"\\' + _STR_ASCII + _UTF8_LEADS
         if s == STR_ESC:
             return _ESCAPABLE
         if s == STR_U:
@@ -193,6 +225,12 @@ class JsonFSM:
         # accepts becomes a byte multi-byte tokens may smuggle into the
         # output (found on GPU: raw control bytes inside strings).
         if s == STR:
+            if self.utf_left:
+                if not (self.utf_lo <= b <= self.utf_hi):
+                    raise ValueError("invalid UTF-8 continuation")
+                self.utf_left -= 1
+                self.utf_lo, self.utf_hi = 0x80, 0xBF
+                return
             if b == ord('"'):
                 self.state = OBJ_COLON if self.key_str else AFTER_VAL
                 self.key_str = False
@@ -200,6 +238,11 @@ class JsonFSM:
                 self.state = STR_ESC
             elif b < 0x20:
                 raise ValueError("raw control byte in string")
+            elif b >= 0x80:
+                lead = _utf8_lead(b)
+                if lead is None:
+                    raise ValueError("invalid UTF-8 lead byte")
+                self.utf_left, self.utf_lo, self.utf_hi = lead
             return
         if s == STR_ESC:
             if b == ord("u"):
@@ -305,7 +348,7 @@ class JsonFSM:
             return 2 + closers          # ':','0'
         if s == STR:
             extra = 3 if self.key_str else 0  # '"' then ':','0'
-            return 1 + extra + closers
+            return self.utf_left + 1 + extra + closers
         if s == STR_ESC:
             extra = 3 if self.key_str else 0
             return 2 + extra + closers

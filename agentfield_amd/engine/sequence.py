@@ -17,7 +17,10 @@ class SamplingParams:
     logprobs: int = 0       # >0: report chosen-token logprob + top-N
                             # alternatives per emitted token (N <= 8)
     json_mode: bool = False  # grammar-constrained valid-JSON decoding
-                             # (byte tokenizer; engine/jsonfsm.py)
+                             # (engine/jsonfsm.py / token_grammar.py)
+    json_schema: object = None  # SchemaSpec (or raw dict): constrain to a
+                                # JSON-Schema subset, not just syntax
+                                # (engine/schemafsm.py)
 
 
 class SeqStatus(enum.Enum):

@@ -72,14 +72,25 @@ class TokenJsonGrammar:
         for b in bs:
             fsm.advance(b)
 
-    @staticmethod
-    def _sig(fsm: JsonFSM, remaining: int) -> tuple:
-        # the budget stops binding once it exceeds any reachable min_close
+    SCHEMA_BUDGET_CAP = 96
+
+    @classmethod
+    def _sig(cls, fsm, remaining: int) -> tuple:
+        if hasattr(fsm, "sig"):
+            # SchemaFSM: min_close is not stack-bounded, so masks for
+            # remaining >= CAP are computed AT the cap (over-strict for
+            # larger budgets, never under-strict) and shared
+            return (fsm.sig(), min(remaining, cls.SCHEMA_BUDGET_CAP))
+        # JsonFSM: the budget stops binding once it exceeds any
+        # reachable min_close
         cap = len(fsm.stack) + 12
         return (fsm.state, tuple(fsm.stack), fsm.key_str, fsm.lit,
-                fsm.hex_left, min(remaining, cap))
+                fsm.hex_left, fsm.utf_left, fsm.utf_lo, fsm.utf_hi,
+                min(remaining, cap))
 
-    def allowed_token_ids(self, fsm: JsonFSM, remaining: int) -> list[int]:
+    def allowed_token_ids(self, fsm, remaining: int) -> list[int]:
+        if hasattr(fsm, "sig"):
+            remaining = min(remaining, self.SCHEMA_BUDGET_CAP)
         sig = self._sig(fsm, remaining)
         hit = self._mask_cache.get(sig)
         if hit is not None:

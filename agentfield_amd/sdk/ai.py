@@ -29,6 +29,7 @@ class AIConfig:
     device: str | None = None           # None -> cuda if available
     fallback_models: tuple = ()
     json_only: bool = False             # grammar-constrained valid JSON
+    json_schema: dict | None = None     # schema-constrained (keys/types)
     ignore_eos: bool = False            # benchmark mode: always decode
                                         # max_tokens (random-init weights
                                         # emit EOS at chance rate)
@@ -135,9 +136,15 @@ class EngineRunner:
         self.engine.set_token_grammar(TokenJsonGrammar(vocab, eos_id=eos))
 
     def _check_json_mode(self, cfg: AIConfig) -> bool:
-        if cfg.json_only:
+        if cfg.json_only or cfg.json_schema is not None:
             self.ensure_json_support()
-        return cfg.json_only
+        return cfg.json_only or cfg.json_schema is not None
+
+    def _compiled_schema(self, cfg: AIConfig):
+        if cfg.json_schema is None:
+            return None
+        from ..engine.schemafsm import SchemaSpec
+        return SchemaSpec(cfg.json_schema)
 
     def generate_text(self, prompt: str, cfg: AIConfig) -> str:
         ids = self.tokenizer.encode(prompt)
@@ -147,7 +154,8 @@ class EngineRunner:
         sp = SamplingParams(max_tokens=cfg.max_tokens,
                             temperature=cfg.temperature,
                             ignore_eos=cfg.ignore_eos,
-                            json_mode=self._check_json_mode(cfg))
+                            json_mode=self._check_json_mode(cfg),
+                            json_schema=self._compiled_schema(cfg))
         w = self.submit(ids, sp)
         if not w["done"].wait(cfg.timeout):
             raise TimeoutError("engine generate timed out")
@@ -164,7 +172,8 @@ class EngineRunner:
         ids = self.tokenizer.encode(prompt)
         sp = SamplingParams(max_tokens=cfg.max_tokens,
                             temperature=cfg.temperature,
-                            json_mode=self._check_json_mode(cfg))
+                            json_mode=self._check_json_mode(cfg),
+                            json_schema=self._compiled_schema(cfg))
         sq: queue.Queue = queue.Queue()
         w = self.submit(ids, sp, stream_q=sq)
         stops = tuple(s for s in (cfg.stop or ()) if s)
@@ -332,11 +341,15 @@ class AgentAI:
                  user: str | None = None, schema=None, stream: bool = False,
                  **overrides):
         cfg = self.config.merged(**overrides)
-        if schema is not None and not cfg.json_only:
-            # schema requests get grammar-constrained JSON decoding: the
-            # engine can GUARANTEE syntactic validity (engine/jsonfsm.py),
-            # unlike the reference's prompt-and-validate seam
-            cfg = cfg.merged(json_only=True)
+        if schema is not None:
+            # schema requests get SCHEMA-constrained decoding: the engine
+            # GUARANTEES the output parses AND matches the schema subset
+            # (keys/types/enums — engine/schemafsm.py), where the
+            # reference can only prompt-and-validate
+            schema_json = (schema.model_json_schema()
+                           if hasattr(schema, "model_json_schema")
+                           else schema)
+            cfg = cfg.merged(json_only=True, json_schema=schema_json)
         parts = []
         sys_p = system or cfg.system_prompt
         if sys_p:

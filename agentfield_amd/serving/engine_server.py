@@ -111,11 +111,15 @@ def create_engine_app(runner: EngineRunner, model_name: str,
     async def generate(req: Request):
         body = await req.json()
         runner, _mname = pick(body)
-        jm = bool(body.get("json_mode", False))
+        schema = body.get("json_schema")
+        jm = bool(body.get("json_mode", False)) or schema is not None
         if jm and not _json_ready(runner):
             return JSONResponse(
                 {"error": "json_mode unsupported for this tokenizer"},
                 status_code=400)
+        if schema is not None and not isinstance(schema, dict):
+            return JSONResponse(
+                {"error": "json_schema must be an object"}, status_code=400)
         sp = SamplingParams(
             max_tokens=int(body.get("max_tokens", 128)),
             temperature=float(body.get("temperature", 0.0)),
@@ -123,7 +127,7 @@ def create_engine_app(runner: EngineRunner, model_name: str,
             top_p=float(body.get("top_p", 1.0) or 1.0),
             ignore_eos=bool(body.get("ignore_eos", False)),
             logprobs=min(int(body.get("logprobs", 0) or 0), 8),
-            json_mode=jm)
+            json_mode=jm, json_schema=schema)
         if "prompt_ids" in body:
             ids = [int(x) for x in body["prompt_ids"]]
         else:
@@ -235,10 +239,22 @@ def create_engine_app(runner: EngineRunner, model_name: str,
         else:
             lp_n = min(int(body.get("logprobs", 0) or 0), 8)
         rf = body.get("response_format") or {}
-        jm = rf.get("type") == "json_object"
+        schema = None
+        if rf.get("type") == "json_schema":
+            # OpenAI structured outputs: {"type": "json_schema",
+            #   "json_schema": {"name": ..., "schema": {...}, "strict": ...}}
+            js = rf.get("json_schema") or {}
+            schema = js.get("schema") if isinstance(js, dict) else None
+            if not isinstance(schema, dict):
+                return JSONResponse(
+                    {"error": {"message": "response_format.json_schema"
+                               ".schema must be an object",
+                               "type": "invalid_request_error"}},
+                    status_code=400)
+        jm = rf.get("type") in ("json_object", "json_schema")
         if jm and not _json_ready(runner):
             return JSONResponse(
-                {"error": {"message": "response_format json_object "
+                {"error": {"message": f"response_format {rf.get('type')} "
                            "unsupported for this tokenizer",
                            "type": "invalid_request_error"}},
                 status_code=400)
@@ -247,7 +263,7 @@ def create_engine_app(runner: EngineRunner, model_name: str,
                             top_p=float(body.get("top_p", 1.0) or 1.0),
                             top_k=int(body.get("top_k", 0) or 0),
                             logprobs=lp_n,
-                            json_mode=jm)
+                            json_mode=jm, json_schema=schema)
         stop_in = body.get("stop") or []
         stops = tuple(s for s in ([stop_in] if isinstance(stop_in, str)
                                   else stop_in) if s)

@@ -375,3 +375,45 @@ def test_json_mode_bpe_vocab_on_gpu():
     for ids in outs.values():
         body = ids[:-1] if ids and ids[-1] == 2 else ids
         _json.loads(tok.decode(body))
+
+
+@pytest.mark.gpu
+def test_schema_constrained_decoding_on_gpu():
+    """Schema-constrained sampling on the HIP kernels: outputs parse AND
+    conform (required keys, types, enum membership)."""
+    import json as _json
+    from agentfield_amd.engine.schemafsm import SchemaSpec
+    cfg = CONFIGS["tiny"]
+    eng = LLMEngine(cfg, device="cuda", page_size=4, num_pages=128,
+                    max_num_seqs=4, enable_graphs=True, seed=11)
+    spec = SchemaSpec({"type": "object",
+                       "properties": {"x": {"type": "integer"},
+                                      "s": {"enum": ["a", "bb"]},
+                                      "t": {"type": "array",
+                                            "items": {"type": "string"}}},
+                       "required": ["x"]})
+    rids = [eng.add_request([1, 30 + i, 9],
+                            SamplingParams(max_tokens=28, temperature=0.95,
+                                           json_mode=True, json_schema=spec))
+            for i in range(6)]
+    outs = {}
+    for _ in range(800):
+        eng.step()
+        for r in rids:
+            if r not in outs:
+                f = eng.get_finished(r)
+                if f:
+                    outs[r] = f.output_ids
+        if len(outs) == len(rids):
+            break
+    assert len(outs) == len(rids)
+    for ids in outs.values():
+        body = ids[:-1] if ids and ids[-1] == 2 else ids
+        data = _json.loads(bytes(b - 4 for b in body).decode(
+            "utf-8", errors="replace"))
+        assert isinstance(data["x"], int)
+        assert set(data) <= {"x", "s", "t"}
+        if "s" in data:
+            assert data["s"] in ("a", "bb")
+        if "t" in data:
+            assert all(isinstance(v, str) for v in data["t"])

@@ -231,3 +231,27 @@ def test_tp2_moe_logits_match_tp1():
 
 def test_tp2_generate_matches_tp1():
     _spawn("_generate_case")
+
+
+def test_request_wire_codec_carries_schema():
+    """json_schema travels on the TP wire so every rank masks identically
+    (ranks sample with the same seed; divergent masks would fork the
+    token stream)."""
+    from agentfield_amd.engine import SamplingParams
+    from agentfield_amd.engine.schemafsm import SchemaSpec
+    from agentfield_amd.parallel.wire import decode_requests, encode_requests
+    schema = {"type": "object",
+              "properties": {"a": {"type": "integer"},
+                             "b": {"enum": ["x", 2]}},
+              "required": ["a"]}
+    reqs = [([1, 2, 3], SamplingParams(max_tokens=8, json_mode=True,
+                                       json_schema=schema)),
+            ([4], SamplingParams(max_tokens=4)),
+            ([5, 6], SamplingParams(max_tokens=4, json_mode=True,
+                                    json_schema=SchemaSpec(schema)))]
+    out = decode_requests(encode_requests(reqs))
+    assert out[0][1].json_schema == schema
+    assert out[1][1].json_schema is None
+    assert out[2][1].json_schema == schema  # SchemaSpec round-trips via source
+    # property order (mask construction depends on it) survives the wire
+    assert list(out[0][1].json_schema["properties"]) == ["a", "b"]
