@@ -32,15 +32,18 @@ __global__ void __launch_bounds__(256) attn_decode_kernel(
     u16* __restrict__ out, float* __restrict__ po, float* __restrict__ pml,
     const u16* __restrict__ q, const u16* __restrict__ kc, const u16* __restrict__ vc,
     const i32* __restrict__ bt, const i32* __restrict__ len,
-    float scale, int Hk, int page_size, int max_pages, int nsplit, i64 qs) {
+    float scale, int Hk, int page_size, int max_pages, int nsplit, i64 qs,
+    int win) {
   const int b = blockIdx.x, kvh = blockIdx.y, split = blockIdx.z;
   const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
   const int tg = lane >> 4, dl = lane & 15;
   const int Hq = Hk * G;
 
   const int L = len[b];
-  const int chunk = (L + nsplit - 1) / nsplit;
-  const int t0 = split * chunk;
+  // sliding window (Mistral): only the last `win` tokens are visible
+  const int s0 = (win > 0 && L > win) ? L - win : 0;
+  const int chunk = (L - s0 + nsplit - 1) / nsplit;
+  const int t0 = s0 + split * chunk;
   const int t1 = min(L, t0 + chunk);
 
   // Q (8 dims per lane per head), pre-scaled
@@ -254,7 +257,7 @@ AF_EXPORT int af_attn_decode(void* out, void* po, void* pml, const void* q,
                              const void* kc, const void* vc, const void* bt,
                              const void* len, float scale, int B, int Hq, int Hk,
                              int D, int page_size, int max_pages, int nsplit,
-                             i64 qs, void* stream) {
+                             i64 qs, int win, void* stream) {
   if (D != AD_D) return 9002;
   const int G = Hq / Hk;
   if (G < 1 || G > AD_MAXG || G * Hk != Hq) return 9003;
@@ -268,12 +271,12 @@ AF_EXPORT int af_attn_decode(void* out, void* po, void* pml, const void* q,
       attn_decode_kernel<GG, true><<<grid, blk, 0, st>>>(                        \
           (u16*)out, (float*)po, (float*)pml, (const u16*)q, (const u16*)kc,     \
           (const u16*)vc, (const i32*)bt, (const i32*)len, scale, Hk,            \
-          page_size, max_pages, nsplit, qs);                                     \
+          page_size, max_pages, nsplit, qs, win);                                \
     else                                                                         \
       attn_decode_kernel<GG, false><<<grid, blk, 0, st>>>(                       \
           (u16*)out, (float*)po, (float*)pml, (const u16*)q, (const u16*)kc,     \
           (const u16*)vc, (const i32*)bt, (const i32*)len, scale, Hk,            \
-          page_size, max_pages, nsplit, qs);                                     \
+          page_size, max_pages, nsplit, qs, win);                                \
   } while (0)
   switch (G) {
     case 1: AF_LAUNCH(1); break;

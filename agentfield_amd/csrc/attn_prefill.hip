@@ -57,7 +57,8 @@ __global__ void __launch_bounds__(256) attn_prefill_kernel(
     const i32* __restrict__ bt, const i32* __restrict__ qstart,
     const i32* __restrict__ cu_seqlens,
     const i32* __restrict__ tile_seq, const i32* __restrict__ tile_q0,
-    float scale, int Hq, int Hk, i64 qs, int page_size, int max_pages) {
+    float scale, int Hq, int Hk, i64 qs, int page_size, int max_pages,
+    int win) {
   const int tile = blockIdx.x, kvh = blockIdx.y;
   const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
   const int seq = tile_seq[tile];
@@ -100,12 +101,15 @@ __global__ void __launch_bounds__(256) attn_prefill_kernel(
     for (int dt = 0; dt < 8; ++dt) o_acc[hl][dt] = f32x4{0.f, 0.f, 0.f, 0.f};
   }
 
-  // tokens visible to this WG: full history + chunk rows up to causal bound
+  // tokens visible to this WG: history + chunk rows up to the causal bound,
+  // and (sliding window) nothing below qrow_min - win + 1
   const int kv_end = min(hist + len, hist + tile_q0[tile] + ROWS);
   const int n_kv_tiles = (kv_end + AP_KVBLK - 1) / AP_KVBLK;
   const int kv_total = hist + len;
+  const int kt_first =
+      (win > 0) ? max(0, hist + tile_q0[tile] - win + 1) / AP_KVBLK : 0;
 
-  for (int kt = 0; kt < n_kv_tiles; ++kt) {
+  for (int kt = kt_first; kt < n_kv_tiles; ++kt) {
     const int kv0 = kt * AP_KVBLK;
     __syncthreads();  // previous iteration's frag reads done
     // ---- stage K (swizzled) from the paged cache ----
@@ -175,7 +179,9 @@ __global__ void __launch_bounds__(256) attn_prefill_kernel(
 #pragma unroll
         for (int st = 0; st < 4; ++st) {
           sr[st] = s_acc[st][r] * scale;
-          if (kv0 + st * 16 + (lane & 15) > qrow) sr[st] = AF_NEG_INF;
+          const int kpos = kv0 + st * 16 + (lane & 15);
+          if (kpos > qrow || (win > 0 && kpos <= qrow - win))
+            sr[st] = AF_NEG_INF;
           tmax = fmaxf(tmax, sr[st]);
         }
         tmax = group16_max_f32(tmax);
@@ -239,7 +245,7 @@ AF_EXPORT int af_attn_prefill(void* out, const void* q, const void* kc,
                               const void* cu_seqlens, const void* tile_seq,
                               const void* tile_q0, float scale, int ntiles,
                               int Hq, int Hk, int D, i64 qs, int page_size,
-                              int max_pages, void* stream) {
+                              int max_pages, int win, void* stream) {
   if (D != AP_D) return 9002;
   const int G = Hq / Hk;
   if (G * Hk != Hq) return 9003;
@@ -251,7 +257,7 @@ AF_EXPORT int af_attn_prefill(void* out, const void* q, const void* kc,
       (u16*)out, (const u16*)q, (const u16*)kc, (const u16*)vc,                 \
       (const i32*)bt, (const i32*)qstart,                                       \
       (const i32*)cu_seqlens, (const i32*)tile_seq, (const i32*)tile_q0,        \
-      scale, Hq, Hk, qs, page_size, max_pages)
+      scale, Hq, Hk, qs, page_size, max_pages, win)
   switch (G) {
     case 1: AF_LAUNCH(1); break;
     case 2: AF_LAUNCH(2); break;

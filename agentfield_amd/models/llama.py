@@ -42,6 +42,9 @@ class LlamaConfig:
     # softmax-routed top-k mixture; intermediate_size is PER EXPERT.
     num_experts: int = 1
     num_experts_per_tok: int = 2
+    # sliding-window attention (Mistral): 0 = full causal; >0 = each token
+    # attends to at most the last `sliding_window` positions (all layers)
+    sliding_window: int = 0
 
     @property
     def q_size(self):
@@ -66,7 +69,8 @@ class LlamaConfig:
             lm_vocab_rows=(self.vocab_size // tp
                            if self.vocab_size % tp == 0 else None),
             num_experts=self.num_experts,
-            num_experts_per_tok=self.num_experts_per_tok)
+            num_experts_per_tok=self.num_experts_per_tok,
+            sliding_window=self.sliding_window)
 
 
 CONFIGS = {
@@ -88,13 +92,14 @@ CONFIGS = {
                                num_heads=40, num_kv_heads=40,
                                vocab_size=32000, rope_theta=10000.0,
                                max_position=4096),
-    # NOTE: full-causal attention; Mistral's 4096 sliding window is not
-    # implemented, so outputs match the real model only for ctx <= 4096.
+    # Mistral v0.1: 4096-token sliding-window attention (both kernels take
+    # the band; KV pages behind the window stay allocated — rolling-buffer
+    # page reuse is a scheduler change queued in docs/ROADMAP.md)
     "mistral-7b": LlamaConfig(name="mistral-7b", hidden_size=4096,
                               intermediate_size=14336, num_layers=32,
                               num_heads=32, num_kv_heads=8,
                               vocab_size=32000, rope_theta=10000.0,
-                              max_position=4096),
+                              max_position=8192, sliding_window=4096),
     # small configs for tests / smoke
     "tiny": LlamaConfig(name="tiny", hidden_size=256, intermediate_size=512,
                         num_layers=2, num_heads=2, num_kv_heads=1,
@@ -182,10 +187,12 @@ class LlamaAttention(nn.Module):
             return ops.attn_prefill(q, kv.k[self.layer_idx],
                                     kv.v[self.layer_idx], md.block_table,
                                     md.q_start, md.cu_seqlens, md.seq_lens,
-                                    self.scale, head_dim=cfg.head_dim)
+                                    self.scale, head_dim=cfg.head_dim,
+                                    window=cfg.sliding_window)
         return ops.attn_decode(q, kv.k[self.layer_idx], kv.v[self.layer_idx],
                                md.block_table, md.seq_lens_t, self.scale,
-                               nsplit=md.nsplit, scratch=md.decode_scratch)
+                               nsplit=md.nsplit, scratch=md.decode_scratch,
+                               window=cfg.sliding_window)
 
     def forward(self, x, positions, rope_tab, kv: KVCache, md: AttnMetadata):
         qkv = ops.linear(x, self.qkv)

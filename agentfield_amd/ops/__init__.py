@@ -154,8 +154,11 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
                 block_table: torch.Tensor, seq_lens: torch.Tensor,
                 scale: float | None = None, nsplit: int = 1,
                 scratch: tuple[torch.Tensor, torch.Tensor] | None = None,
-                out: torch.Tensor | None = None) -> torch.Tensor:
-    """Paged decode attention. q [B, Hq*D] (strided rows ok) -> out [B, Hq*D]."""
+                out: torch.Tensor | None = None,
+                window: int = 0) -> torch.Tensor:
+    """Paged decode attention. q [B, Hq*D] (strided rows ok) -> out [B, Hq*D].
+    window > 0: sliding-window (Mistral) — only the last `window` cached
+    tokens are attended."""
     _, Hk, page, D = kcache.shape
     B = q.shape[0]
     Hq = q.shape[1] // D
@@ -163,7 +166,8 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
         scale = 1.0 / math.sqrt(D)
     if not _on_gpu(q):
         return ref.attn_decode(q.unflatten(-1, (Hq, D)), kcache, vcache,
-                               block_table, seq_lens, scale).flatten(1)
+                               block_table, seq_lens, scale,
+                               window=window).flatten(1)
     if out is None:
         out = torch.empty(B, Hq * D, dtype=q.dtype, device=q.device)
     G = Hq // Hk
@@ -179,7 +183,7 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
         _lib.ptr(out), _lib.ptr(po), _lib.ptr(pml), _lib.ptr(q),
         _lib.ptr(kcache), _lib.ptr(vcache), _lib.ptr(block_table),
         _lib.ptr(seq_lens), scale, B, Hq, Hk, D, page, block_table.shape[1],
-        nsplit, q.stride(0), _lib.cur_stream())
+        nsplit, q.stride(0), window, _lib.cur_stream())
     _lib.check(rc, "af_attn_decode")
     return out
 
@@ -197,7 +201,8 @@ def prefill_tiles(seq_lens: list[int], rows_per_wg: int):
 def attn_prefill(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
                  block_table: torch.Tensor, q_start: torch.Tensor,
                  cu_seqlens: torch.Tensor, seq_lens: list[int],
-                 scale: float | None = None, head_dim: int = 128) -> torch.Tensor:
+                 scale: float | None = None, head_dim: int = 128,
+                 window: int = 0) -> torch.Tensor:
     """Paged causal prefill attention: chunk rows q [T, Hq*D] (strided ok)
     attend to cached history + chunk through the block table.  q_start[s] is
     the chunk's absolute start position (0 = full-prompt prefill); the chunk
@@ -211,7 +216,7 @@ def attn_prefill(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
     if not _on_gpu(q):
         return ref.attn_prefill_paged(q.unflatten(-1, (Hq, D)), kcache, vcache,
                                       block_table, q_start, cu_seqlens,
-                                      scale).flatten(1)
+                                      scale, window=window).flatten(1)
     G = Hq // Hk
     rows_per_wg = 16 * max(1, 4 // G)
     ts, tq = prefill_tiles(seq_lens, rows_per_wg)
@@ -224,7 +229,7 @@ def attn_prefill(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
         _lib.ptr(block_table), _lib.ptr(q_start),
         _lib.ptr(cu_seqlens), _lib.ptr(tile_seq), _lib.ptr(tile_q0),
         scale, len(ts), Hq, Hk, D, q.stride(0), page, block_table.shape[1],
-        _lib.cur_stream())
+        window, _lib.cur_stream())
     _lib.check(rc, "af_attn_prefill")
     return out
 

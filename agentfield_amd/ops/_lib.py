@@ -49,9 +49,9 @@ def _declare(l: ctypes.CDLL) -> None:
     l.af_reshape_and_cache.argtypes = [p, p, p, p, p, i, i, i, i, p]
     l.af_embedding.argtypes = [p, p, p, p, i, i, p]
     l.af_attn_decode.argtypes = [p, p, p, p, p, p, p, p, f, i, i, i, i, i, i, i,
-                                 i64, p]
+                                 i64, i, p]
     l.af_attn_prefill.argtypes = [p, p, p, p, p, p, p, p, p, f,
-                                  i, i, i, i, i64, i, i, p]
+                                  i, i, i, i, i64, i, i, i, p]
     l.af_gemm_bf16.argtypes = [p, p, p, i, i, i, p]
     l.af_gemm_bf16_ring.argtypes = [p, p, p, i, i, i, p]
     l.af_gemm_bf16_q8.argtypes = [p, p, p, i, i, i, p]

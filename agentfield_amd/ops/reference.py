@@ -53,17 +53,19 @@ def reshape_and_cache(k, v, kcache, vcache, slots):
         vcache[p, :, o] = v[t]
 
 
-def attn_decode(q, kcache, vcache, block_table, seq_lens, scale):
-    """q [B,Hq,D] -> out [B,Hq,D], gathering K/V from pages."""
+def attn_decode(q, kcache, vcache, block_table, seq_lens, scale, window=0):
+    """q [B,Hq,D] -> out [B,Hq,D], gathering K/V from pages.
+    window > 0: only the last `window` tokens are attended."""
     B, Hq, D = q.shape
     _, Hk, page, _ = kcache.shape
     G = Hq // Hk
     out = torch.empty_like(q)
     for b in range(B):
         L = int(seq_lens[b])
+        w0 = max(0, L - window) if window > 0 else 0
         pages = block_table[b, : (L + page - 1) // page].long()
-        ks = kcache[pages].permute(1, 0, 2, 3).reshape(Hk, -1, D)[:, :L].float()
-        vs = vcache[pages].permute(1, 0, 2, 3).reshape(Hk, -1, D)[:, :L].float()
+        ks = kcache[pages].permute(1, 0, 2, 3).reshape(Hk, -1, D)[:, w0:L].float()
+        vs = vcache[pages].permute(1, 0, 2, 3).reshape(Hk, -1, D)[:, w0:L].float()
         for h in range(Hq):
             kh = ks[h // G]
             vh = vs[h // G]
@@ -95,7 +97,7 @@ def attn_prefill(q, k, v, cu_seqlens, scale):
 
 
 def attn_prefill_paged(q, kcache, vcache, block_table, q_start, cu_seqlens,
-                       scale):
+                       scale, window=0):
     """Causal attention of chunk rows against cached history + chunk.
     q [T,Hq,D]; caches [npages,Hk,page,D]; q_start[s] = absolute position of
     the chunk's first row."""
@@ -119,7 +121,10 @@ def attn_prefill_paged(q, kcache, vcache, block_table, q_start, cu_seqlens,
         att = (qf @ kf.transpose(-1, -2)) * scale         # [Hq, L, total]
         qpos = torch.arange(hist, hist + L).unsqueeze(1)
         kpos = torch.arange(total).unsqueeze(0)
-        att = att.masked_fill((kpos > qpos), float("-inf")).softmax(-1)
+        bad = kpos > qpos
+        if window > 0:
+            bad |= kpos <= qpos - window
+        att = att.masked_fill(bad, float("-inf")).softmax(-1)
         out[s0:s1] = (att @ vf).permute(1, 0, 2).to(q.dtype)
     return out
 

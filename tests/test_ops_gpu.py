@@ -122,7 +122,7 @@ def test_reshape_and_cache():
     assert close(kc, kc_ref, atol=0) and close(vc, vc_ref, atol=0)
 
 
-def _decode_case(B, Hq, Hk, lens, page=16, nsplit=1, seed=13):
+def _decode_case(B, Hq, Hk, lens, page=16, nsplit=1, seed=13, window=0):
     D = 128
     torch.manual_seed(seed)
     maxp = (max(lens) + page - 1) // page
@@ -139,11 +139,13 @@ def _decode_case(B, Hq, Hk, lens, page=16, nsplit=1, seed=13):
         at += n
     bt = bt.to(DEV)
     lens_t = torch.tensor(lens, dtype=torch.int32, device=DEV)
-    out = ops.attn_decode(q, kc, vc, bt, lens_t, nsplit=nsplit)
+    out = ops.attn_decode(q, kc, vc, bt, lens_t, nsplit=nsplit, window=window)
     torch.cuda.synchronize()
     want = ref.attn_decode(q.cpu().unflatten(-1, (Hq, D)), kc.cpu(), vc.cpu(),
-                           bt.cpu(), lens_t.cpu(), 1.0 / math.sqrt(D)).flatten(1)
-    assert close(out, want), f"decode mismatch B={B} G={Hq//Hk} nsplit={nsplit}"
+                           bt.cpu(), lens_t.cpu(), 1.0 / math.sqrt(D),
+                           window=window).flatten(1)
+    assert close(out, want), \
+        f"decode mismatch B={B} G={Hq//Hk} nsplit={nsplit} win={window}"
 
 
 def test_attn_decode_g4():
@@ -160,7 +162,7 @@ def test_attn_decode_split():
     _decode_case(1, 4, 1, [511], nsplit=8, seed=23)
 
 
-def _prefill_case(Hq, Hk, lens, seed=30, page=16):
+def _prefill_case(Hq, Hk, lens, seed=30, page=16, window=0):
     """Full-prompt paged prefill vs the contiguous fp32 reference."""
     D = 128
     T = sum(lens)
@@ -188,13 +190,14 @@ def _prefill_case(Hq, Hk, lens, seed=30, page=16):
     ops.reshape_and_cache(k.unflatten(-1, (Hk, D)), v.unflatten(-1, (Hk, D)),
                           kc, vc, slots_t)
     qstart = torch.zeros(len(lens), dtype=torch.int32, device=DEV)
-    out = ops.attn_prefill(q, kc, vc, bt, qstart, cu_t, lens)
+    out = ops.attn_prefill(q, kc, vc, bt, qstart, cu_t, lens, window=window)
     torch.cuda.synchronize()
-    want = ref.attn_prefill(q.cpu().unflatten(-1, (Hq, D)),
-                            k.cpu().unflatten(-1, (Hk, D)),
-                            v.cpu().unflatten(-1, (Hk, D)), cu_t.cpu(),
-                            1.0 / math.sqrt(D)).flatten(1)
-    assert close(out, want), f"prefill mismatch G={Hq//Hk} lens={lens}"
+    want = ref.attn_prefill_paged(q.cpu().unflatten(-1, (Hq, D)), kc.cpu(),
+                                  vc.cpu(), bt.cpu(), qstart.cpu(), cu_t.cpu(),
+                                  1.0 / math.sqrt(D),
+                                  window=window).flatten(1)
+    assert close(out, want), \
+        f"prefill mismatch G={Hq//Hk} lens={lens} win={window}"
 
 
 def test_attn_prefill_g4():
@@ -212,6 +215,19 @@ def test_attn_prefill_g2_g8():
 
 def test_attn_prefill_long():
     _prefill_case(8, 2, [1024], seed=70)
+
+
+def test_attn_decode_sliding_window():
+    # window < L (band active), window >= L (equivalent to full causal)
+    _decode_case(3, 8, 2, [5, 33, 200], seed=91, window=48)
+    _decode_case(2, 8, 2, [100, 230], nsplit=4, seed=92, window=64)
+    _decode_case(2, 4, 4, [20, 64], seed=93, window=128)
+
+
+def test_attn_prefill_sliding_window():
+    _prefill_case(8, 2, [48, 170], seed=94, window=32)
+    _prefill_case(4, 2, [300], seed=95, window=96)
+    _prefill_case(8, 2, [40], seed=96, window=64)  # win >= len: full causal
 
 
 def test_attn_prefill_chunked_history():
