@@ -75,4 +75,32 @@ __device__ __forceinline__ int xcd_swizzle(int bid, int nwg) {
   return base + idx;
 }
 
+// Supertile variant: map the 32 CONCURRENT workgroups of one XCD (1 WG/CU,
+// 32 CUs/XCD) onto an 8x4 tile block instead of one 32-wide tile row.  The
+// concurrent set then touches 8 A-stripes + 4 B-stripes per K-slice instead
+// of 1 + 32, cutting L2-miss traffic ~2.7x for K-synchronized walks.
+// Bijective when nwg % 256 == 0 and the grid divides 8x4; callers fall back
+// to xcd_swizzle otherwise.
+__device__ __forceinline__ void xcd_tile_map(int bid, int tiles_m,
+                                             int tiles_n, int order,
+                                             int* tm, int* tn) {
+  const int nwg = tiles_m * tiles_n;
+  // order encodes the supertile aspect: 2 = 8x4, 3 = 4x8, 4 = 16x2
+  const int sm = (order == 3) ? 4 : (order == 4) ? 16 : 8;
+  const int sn = 32 / sm;
+  if (order >= 2 && tiles_m % sm == 0 && tiles_n % sn == 0 &&
+      (nwg & 255) == 0) {
+    const int xcd = bid & 7, idx = bid >> 3;
+    const int within = idx & 31, super_seq = idx >> 5;
+    const int sid = super_seq * 8 + xcd;      // supertile of sm x sn tiles
+    const int nsup_n = tiles_n / sn;
+    *tm = (sid / nsup_n) * sm + within % sm;
+    *tn = (sid % nsup_n) * sn + within / sm;
+    return;
+  }
+  const int b = xcd_swizzle(bid, nwg);
+  *tm = b / tiles_n;
+  *tn = b % tiles_n;
+}
+
 static inline int af_last_err() { return (int)hipGetLastError(); }

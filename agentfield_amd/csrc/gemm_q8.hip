@@ -24,6 +24,7 @@
 // Per phase: { ds_read frags | stage 1 granule | vmcnt(8) | barrier |
 //              setprio(1) 16 MFMA setprio(0) | barrier }.
 #include "common.h"
+#include <cstdlib>
 
 #define GQ_BM 256
 #define GQ_BK 64
@@ -70,15 +71,22 @@ __device__ __forceinline__ s16x8 gq_read(const u16* lds, int off) {
       reinterpret_cast<const char*>(lds) + off);
 }
 
-#define GQ_WAIT8 asm volatile("s_waitcnt vmcnt(8)" ::: "memory")
+// AF_GEMM_WAIT selects the per-phase counted wait depth (A/B lever):
+// vmcnt(N) admits N outstanding global_load_lds per thread; 8 = 4 granules.
+#if !defined(GQ_WAITN)
+#define GQ_WAITN 8
+#endif
+#define GQ_STR2(x) #x
+#define GQ_STR(x) GQ_STR2(x)
+#define GQ_WAIT8   asm volatile("s_waitcnt vmcnt(" GQ_STR(GQ_WAITN) ")" ::: "memory")
 
 __global__ void __launch_bounds__(512, 1) gemm_bf16_q8_kernel(
     u16* __restrict__ C, const u16* __restrict__ A, const u16* __restrict__ W,
-    int M, int N, int K, int tiles_m, int tiles_n) {
+    int M, int N, int K, int tiles_m, int tiles_n, int order) {
   __shared__ u16 ring[8][GQ_GRAN];
 
-  const int bid = xcd_swizzle(blockIdx.x, tiles_m * tiles_n);
-  const int tm = bid / tiles_n, tn = bid % tiles_n;
+  int tm, tn;
+  xcd_tile_map(blockIdx.x, tiles_m, tiles_n, order, &tm, &tn);
   const int m0 = tm * GQ_BM, n0 = tn * GQ_BM;
   const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
   // 8 waves tile each 128x128 quadrant as 2x4 of 64x32 slices
@@ -250,7 +258,13 @@ AF_EXPORT int af_gemm_bf16_q8(void* C, const void* A, const void* W,
   if (M == 0) return 0;
   const int tiles_m = (M + GQ_BM - 1) / GQ_BM;
   const int tiles_n = (N + GQ_BM - 1) / GQ_BM;
+  static int order = -1;  // AF_GEMM_ORDER: 1 = row-chunk, 2 = 8x4 supertile
+  if (order < 0) {
+    const char* e = getenv("AF_GEMM_ORDER");
+    order = e ? atoi(e) : 2;
+  }
   gemm_bf16_q8_kernel<<<tiles_m * tiles_n, 512, 0, (hipStream_t)stream>>>(
-      (u16*)C, (const u16*)A, (const u16*)W, M, N, K, tiles_m, tiles_n);
+      (u16*)C, (const u16*)A, (const u16*)W, M, N, K, tiles_m, tiles_n,
+      order);
   return af_last_err();
 }
