@@ -113,7 +113,8 @@ def server(host: str = "0.0.0.0", port: int = 8520,
 @app.command()
 def engine(model: list[str] = ["llama-3-8b"], device: str = None,
            host: str = "127.0.0.1", port: int = 8710,
-           max_num_seqs: int = 256, spec_lookup: int = 0,
+           max_num_seqs: int = 256, max_prefill_tokens: int = 0,
+           spec_lookup: int = 0, no_graphs: bool = False,
            prefix_cache: bool = False):
     """Run one GPU engine server (start one per GPU for DP; repeat
     --model to co-serve several models from one replica)."""
@@ -121,6 +122,10 @@ def engine(model: list[str] = ["llama-3-8b"], device: str = None,
     argv = ["--host", host, "--port", str(port),
             "--max-num-seqs", str(max_num_seqs),
             "--spec-lookup", str(spec_lookup)]
+    if max_prefill_tokens:
+        argv += ["--max-prefill-tokens", str(max_prefill_tokens)]
+    if no_graphs:
+        argv.append("--no-graphs")
     for m in model:
         argv += ["--model", m]
     if prefix_cache:

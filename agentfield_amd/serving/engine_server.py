@@ -423,6 +423,9 @@ def main():
     ap.add_argument("--host", default="127.0.0.1")
     ap.add_argument("--port", type=int, default=8710)
     ap.add_argument("--max-num-seqs", type=int, default=256)
+    ap.add_argument("--max-prefill-tokens", type=int, default=None,
+                    help="prefill admission budget per step (engine default "
+                         "when unset)")
     ap.add_argument("--no-graphs", action="store_true")
     ap.add_argument("--spec-lookup", type=int, default=0,
                     help="prompt-lookup speculative decoding draft length "
@@ -445,6 +448,8 @@ def main():
         tokenizer = load_tokenizer()
         if tokenizer.eos_id is not None:
             kw["eos_id"] = tokenizer.eos_id  # HF vocab EOS != byte EOS
+        if args.max_prefill_tokens:
+            kw["max_prefill_tokens"] = args.max_prefill_tokens
         eng = LLMEngine(
             CONFIGS[name], device=device,
             max_num_seqs=kw.pop("max_num_seqs", args.max_num_seqs),

@@ -24,13 +24,18 @@ class RemoteRunner:
         self.router = router
 
     def generate_text(self, prompt: str, cfg) -> str:
-        return self.router.generate(prompt=prompt,
-                                    max_tokens=cfg.max_tokens,
-                                    temperature=cfg.temperature,
-                                    stop=list(cfg.stop or ()),
-                                    model=cfg.model,
-                                    json_mode=getattr(cfg, "json_only", False),
-                                    timeout=cfg.timeout)["text"]
+        body = dict(prompt=prompt,
+                    max_tokens=cfg.max_tokens,
+                    temperature=cfg.temperature,
+                    stop=list(cfg.stop or ()),
+                    model=cfg.model,
+                    json_mode=getattr(cfg, "json_only", False),
+                    ignore_eos=getattr(cfg, "ignore_eos", False),
+                    timeout=cfg.timeout)
+        schema = getattr(cfg, "json_schema", None)
+        if schema is not None:
+            body["json_schema"] = schema
+        return self.router.generate(**body)["text"]
 
     def stream_text(self, prompt: str, cfg):
         yield from self.router.stream(prompt=prompt, max_tokens=cfg.max_tokens,
@@ -52,7 +57,12 @@ class DPRouter:
         self._healthy: dict[str, bool] = {u: True for u in self.urls}
         self._last_refresh = 0.0
         self._lock = threading.Lock()
-        self._client = httpx.Client(timeout=timeout)
+        # hundreds of agent threads share this client under DP load; the
+        # httpx default pool (100) would throttle them
+        self._client = httpx.Client(
+            timeout=timeout,
+            limits=httpx.Limits(max_connections=1024,
+                                max_keepalive_connections=1024))
 
     def _refresh(self):
         nowt = time.time()
