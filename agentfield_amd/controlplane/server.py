@@ -776,6 +776,28 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
                 return scope, v
         return "global", "global"
 
+    @app.post("/api/v1/locks/acquire")
+    async def lock_acquire(req: Request):
+        """Cross-process lease lock (SQLite-atomic UPSERT): reentrant for
+        the same owner, steals only expired leases."""
+        body = await req.json()
+        ok = cp.storage.lock_acquire(body["name"], body["owner"],
+                                     float(body.get("ttl_s", 30.0)))
+        return {"acquired": ok,
+                "holder": cp.storage.lock_holder(body["name"])}
+
+    @app.post("/api/v1/locks/release")
+    async def lock_release(req: Request):
+        body = await req.json()
+        return {"released": cp.storage.lock_release(body["name"],
+                                                    body["owner"])}
+
+    @app.post("/api/v1/locks/refresh")
+    async def lock_refresh(req: Request):
+        body = await req.json()
+        return {"refreshed": cp.storage.lock_refresh(
+            body["name"], body["owner"], float(body.get("ttl_s", 30.0)))}
+
     @app.post("/api/v1/memory/set")
     async def memory_set(req: Request):
         body = await req.json()

@@ -87,6 +87,9 @@ CREATE TABLE IF NOT EXISTS execution_notes (
   note TEXT, created_at REAL
 );
 CREATE INDEX IF NOT EXISTS idx_notes_exec ON execution_notes(execution_id);
+CREATE TABLE IF NOT EXISTS dist_locks (
+  name TEXT PRIMARY KEY, owner TEXT, expires_at REAL
+);
 """
 
 
@@ -647,6 +650,50 @@ class Storage:
     def list_runs(self, limit: int = 50) -> list[dict]:
         return self._q("SELECT * FROM workflow_runs ORDER BY started_at DESC LIMIT ?",
                        (limit,))
+
+    # ---------------------------------------------------- distributed locks
+    # Lease-based mutual exclusion across processes sharing the DB
+    # (reference: storage locks with expiry — local.go's lock table).
+    # acquire is a single atomic UPSERT guarded on expiry/ownership, so
+    # concurrent workers over the same WAL cannot both win.
+    def lock_acquire(self, name: str, owner: str, ttl_s: float = 30.0) -> bool:
+        now = time.time()
+        with self._lock:
+            cur = self._db.execute(
+                """INSERT INTO dist_locks(name, owner, expires_at)
+                   VALUES(?,?,?)
+                   ON CONFLICT(name) DO UPDATE
+                     SET owner=excluded.owner, expires_at=excluded.expires_at
+                     WHERE dist_locks.expires_at < ?
+                        OR dist_locks.owner = excluded.owner""",
+                (name, owner, now + ttl_s, now))
+            self._db.commit()
+            if cur.rowcount > 0:
+                return True
+        return False
+
+    def lock_release(self, name: str, owner: str) -> bool:
+        with self._lock:
+            cur = self._db.execute(
+                "DELETE FROM dist_locks WHERE name=? AND owner=?",
+                (name, owner))
+            self._db.commit()
+            return cur.rowcount > 0
+
+    def lock_refresh(self, name: str, owner: str, ttl_s: float = 30.0) -> bool:
+        with self._lock:
+            cur = self._db.execute(
+                "UPDATE dist_locks SET expires_at=? WHERE name=? AND owner=?",
+                (time.time() + ttl_s, name, owner))
+            self._db.commit()
+            return cur.rowcount > 0
+
+    def lock_holder(self, name: str) -> str | None:
+        row = self._q1("SELECT owner, expires_at FROM dist_locks WHERE name=?",
+                       (name,))
+        if row and row["expires_at"] > time.time():
+            return row["owner"]
+        return None
 
     def close(self):
         with self._lock:

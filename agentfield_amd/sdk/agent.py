@@ -30,7 +30,7 @@ from .ai import AgentAI, AIConfig
 from .client import AgentFieldClient
 from .execution_context import (ExecutionContext, current_context,
                                 reset_context, set_context)
-from .memory import MemoryInterface
+from .memory import DistributedLock, MemoryInterface
 from .resilience import ResultCache, StatelessRateLimiter
 
 
@@ -396,6 +396,13 @@ class Agent(FastAPI):
         return q
 
     # ------------------------------------------------------- cross-agent
+    def lock(self, name: str, ttl_s: float = 30.0,
+             timeout_s: float = 60.0) -> DistributedLock:
+        """Cross-process lease lock via the control plane:
+        `with app.lock("migrate-db"): ...`"""
+        return DistributedLock(self.client, name, owner=self.node_id,
+                               ttl_s=ttl_s, timeout_s=timeout_s)
+
     def call(self, target: str, _async: bool = False, _webhook=None, **kwargs):
         """Nested cross-agent call via the control plane (builds the DAG)."""
         ctx = current_context()

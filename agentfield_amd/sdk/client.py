@@ -179,6 +179,11 @@ class AgentFieldClient:
         r.raise_for_status()
         return r.json()
 
+    def lock_op(self, op: str, body: dict) -> dict:
+        r = self._client.post(f"{self.base_url}/api/v1/locks/{op}", json=body)
+        r.raise_for_status()
+        return r.json()
+
     # --------------------------------------------------------------- DID/VC
     def did_register(self, node_id: str, reasoners: list[str],
                      skills: list[str]) -> dict:

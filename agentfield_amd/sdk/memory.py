@@ -58,6 +58,53 @@ class ScopedMemory:
                                             self._body(key=key))["deleted"]
 
 
+class DistributedLock:
+    """Cross-process lease lock served by the control plane (storage-level
+    atomic UPSERT; reference: the storage provider's lock table).  Usable
+    as a context manager; the lease auto-refreshes on `refresh()` and
+    expires server-side if the holder dies."""
+
+    def __init__(self, client, name: str, owner: str, ttl_s: float = 30.0,
+                 timeout_s: float = 60.0, poll_s: float = 0.2):
+        self.client = client
+        self.name = name
+        self.owner = owner
+        self.ttl_s = ttl_s
+        self.timeout_s = timeout_s
+        self.poll_s = poll_s
+
+    def acquire(self, block: bool = True) -> bool:
+        deadline = time.time() + self.timeout_s
+        while True:
+            r = self.client.lock_op("acquire", {"name": self.name,
+                                                "owner": self.owner,
+                                                "ttl_s": self.ttl_s})
+            if r.get("acquired"):
+                return True
+            if not block or time.time() >= deadline:
+                return False
+            time.sleep(self.poll_s)
+
+    def refresh(self) -> bool:
+        return self.client.lock_op("refresh", {
+            "name": self.name, "owner": self.owner,
+            "ttl_s": self.ttl_s}).get("refreshed", False)
+
+    def release(self) -> bool:
+        return self.client.lock_op("release", {
+            "name": self.name, "owner": self.owner}).get("released", False)
+
+    def __enter__(self) -> "DistributedLock":
+        if not self.acquire():
+            raise TimeoutError(f"lock {self.name!r} not acquired "
+                               f"within {self.timeout_s}s")
+        return self
+
+    def __exit__(self, *exc):
+        self.release()
+        return False
+
+
 class MemoryInterface:
     """app.memory — default scope follows the current execution context
     (workflow > session > actor > global)."""

@@ -775,3 +775,45 @@ def test_ui_dashboard_data_contracts(cp_server, greeting_agent):
     assert all({"at", "event"} <= set(e) for e in tl["events"])
     m = httpx.get(base + "/api/ui/v1/reasoners/greeter/greet/metrics").json()
     assert {"executions", "success_rate", "p50_ms"} <= set(m)
+
+
+def test_distributed_locks(cp_server):
+    """Lease locks: exclusive across owners, reentrant for the holder,
+    stealable only after expiry, releasable only by the holder."""
+    import time as _t
+
+    import httpx
+    srv, _cp = cp_server
+    url = srv.base_url
+
+    def op(o, **body):
+        return httpx.post(f"{url}/api/v1/locks/{o}", json=body,
+                          timeout=10.0).json()
+
+    assert op("acquire", name="L", owner="a", ttl_s=5.0)["acquired"]
+    assert op("acquire", name="L", owner="a", ttl_s=5.0)["acquired"]  # reentrant
+    r = op("acquire", name="L", owner="b", ttl_s=5.0)
+    assert not r["acquired"] and r["holder"] == "a"
+    assert not op("release", name="L", owner="b")["released"]
+    assert op("refresh", name="L", owner="a", ttl_s=5.0)["refreshed"]
+    assert op("release", name="L", owner="a")["released"]
+    assert op("acquire", name="L", owner="b", ttl_s=0.2)["acquired"]
+    _t.sleep(0.35)  # expire b's lease
+    assert op("acquire", name="L", owner="c", ttl_s=5.0)["acquired"]
+    assert not op("refresh", name="L", owner="b")["refreshed"]
+    op("release", name="L", owner="c")
+
+
+def test_sdk_lock_context_manager(cp_server):
+    from agentfield_amd.sdk.client import AgentFieldClient
+    from agentfield_amd.sdk.memory import DistributedLock
+    srv, _cp = cp_server
+    c1 = AgentFieldClient(srv.base_url)
+    c2 = AgentFieldClient(srv.base_url)
+    with DistributedLock(c1, "job", owner="n1", ttl_s=5.0, timeout_s=1.0):
+        l2 = DistributedLock(c2, "job", owner="n2", ttl_s=5.0,
+                             timeout_s=0.3, poll_s=0.05)
+        assert not l2.acquire()
+    # released -> n2 can take it
+    assert DistributedLock(c2, "job", owner="n2", ttl_s=5.0,
+                           timeout_s=1.0).acquire()
