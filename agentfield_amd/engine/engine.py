@@ -67,9 +67,15 @@ class LLMEngine:
         max_pages_per_seq = (cfg.max_position + page_size - 1) // page_size
         self.max_pages_per_seq = max_pages_per_seq
         self.kv = KVCache(cfg, num_pages, page_size, self.device, dtype)
+        import os as _os
+        # rolling KV reclamation for sliding-window models (AF_KV_ROLL=0
+        # keeps every page allocated — A/B/debug escape hatch)
+        roll = (getattr(cfg, "sliding_window", 0)
+                if _os.environ.get("AF_KV_ROLL", "1") != "0" else 0)
         sched_cfg = SchedulerConfig(
             max_num_seqs=max_num_seqs, max_prefill_tokens=max_prefill_tokens,
-            page_size=page_size, num_pages=num_pages, max_waiting=max_waiting)
+            page_size=page_size, num_pages=num_pages, max_waiting=max_waiting,
+            window_tokens=roll)
         if prefix_cache:
             # The C++ NativeScheduler's prefix mode is the default (round-2:
             # lockstep-pinned to the Python oracle in

@@ -166,6 +166,7 @@ class PrefixCachingScheduler(Scheduler):
         i = 0
         while i < len(self.running):
             seq = self.running[i]
+            self._roll(seq)  # refcounted free: shared prefix pages survive
             while not (self._pages_needed(seq.num_tokens + 1) <=
                        len(seq.pages) or self._grow_cached(seq)):
                 if self.running[-1] is seq:
@@ -200,6 +201,7 @@ class PrefixCachingScheduler(Scheduler):
         self.release(seq)
 
     def release(self, seq: Sequence) -> None:
-        self.alloc.free(seq.pages)
+        self.alloc.free(seq.pages[seq.freed_pages:])
         seq.pages = []
+        seq.freed_pages = 0
         seq.cached_prefix = 0

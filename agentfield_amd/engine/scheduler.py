@@ -47,6 +47,14 @@ class SchedulerConfig:
     page_size: int = 16
     num_pages: int = 1024
     max_waiting: int = 4096
+    # sliding-window models (cfg.sliding_window): pages wholly behind the
+    # attention band are reclaimed during decode (rolling KV buffer).
+    # 0 disables.  The 64-token slack covers the prefill kernel's KV-tile
+    # staging below the band start (attn_prefill.hip kt_first).
+    window_tokens: int = 0
+
+
+ROLL_SLACK = 64
 
 
 @dataclass
@@ -84,6 +92,19 @@ class Scheduler:
     def _pages_needed(self, ntokens: int) -> int:
         return (ntokens + self.cfg.page_size - 1) // self.cfg.page_size
 
+    def _roll(self, seq: Sequence) -> None:
+        """Reclaim pages wholly behind the attention window.  The page ID
+        stays in seq.pages (block-table rows keep their slot; the band
+        mask guarantees those positions are never scored) but returns to
+        the pool — long windowed sequences hold O(window) pages."""
+        w = self.cfg.window_tokens
+        if not w:
+            return
+        lim = seq.num_tokens - w - ROLL_SLACK
+        while (seq.freed_pages + 1) * self.cfg.page_size <= lim:
+            self.alloc.free([seq.pages[seq.freed_pages]])
+            seq.freed_pages += 1
+
     def _grow(self, seq: Sequence) -> bool:
         """Ensure capacity for one more token; returns False if OOM."""
         need = self._pages_needed(seq.num_tokens + 1)
@@ -94,8 +115,9 @@ class Scheduler:
         return True
 
     def release(self, seq: Sequence) -> None:
-        self.alloc.free(seq.pages)
+        self.alloc.free(seq.pages[seq.freed_pages:])
         seq.pages = []
+        seq.freed_pages = 0
 
     def _preempt_last(self) -> None:
         victim = self.running.pop()
@@ -137,6 +159,7 @@ class Scheduler:
         i = 0
         while i < len(self.running):
             seq = self.running[i]
+            self._roll(seq)
             while not self._grow(seq):
                 if self.running[-1] is seq:
                     # can't preempt self and nothing else to free: defer
@@ -185,7 +208,8 @@ class NativeSchedulerAdapter:
         self.prefix_cache = prefix_cache
         self.nat = NativeScheduler(cfg.max_num_seqs, cfg.max_prefill_tokens,
                                    cfg.page_size, cfg.num_pages,
-                                   cfg.max_waiting, prefix_cache)
+                                   cfg.max_waiting, prefix_cache,
+                                   cfg.window_tokens)
         self.seqs: dict[int, Sequence] = {}
         self.alloc = _AllocShim(self.nat, cfg.num_pages)
 

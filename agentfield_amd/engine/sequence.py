@@ -40,6 +40,7 @@ class Sequence:
     alloc_epoch: int = 0   # bumped on each (re)admission page allocation
     output_ids: list[int] = field(default_factory=list)
     pages: list[int] = field(default_factory=list)
+    freed_pages: int = 0   # leading pages reclaimed by the rolling window
     finish_reason: str | None = None
     # per output token, when sampling.logprobs > 0:
     # {"logprob": float, "top": [(token_id, logprob), ...]}

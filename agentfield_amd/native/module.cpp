@@ -123,6 +123,7 @@ struct SeqState {
   int prompt_len = 0;
   int num_tokens = 0;   // prompt + generated
   int cached_prefix = 0;  // prompt tokens served from the prefix cache
+  int freed_pages = 0;  // leading pages reclaimed by the rolling window
   std::vector<int> pages;
   std::vector<long> hashes;  // chained per-full-page prompt hashes
 };
@@ -140,10 +141,12 @@ class NativeScheduler {
   // chained per-page hashes the Python side computes from prompt tokens
   // (semantics lockstep-pinned to engine/prefix_cache.py).
   NativeScheduler(int max_num_seqs, int max_prefill_tokens, int page_size,
-                  int num_pages, int max_waiting, bool prefix_cache = false)
+                  int num_pages, int max_waiting, bool prefix_cache = false,
+                  int window_tokens = 0)
       : max_num_seqs_(max_num_seqs), max_prefill_tokens_(max_prefill_tokens),
         page_size_(page_size), num_pages_(num_pages),
-        max_waiting_(max_waiting), prefix_cache_(prefix_cache) {
+        max_waiting_(max_waiting), prefix_cache_(prefix_cache),
+        window_tokens_(window_tokens) {
     for (int p = num_pages - 1; p >= 1; --p) free_list_.push_back(p);
   }
 
@@ -260,6 +263,7 @@ class NativeScheduler {
     while (i < running_.size()) {
       long sid = running_[i];
       SeqState& st = seqs_[sid];
+      roll(st);
       while (!grow(st)) {
         long victim = running_.back();
         if (victim == sid) {
@@ -295,6 +299,20 @@ class NativeScheduler {
     }
   }
 
+  // Rolling KV buffer (sliding-window models): reclaim pages wholly
+  // behind the attention band.  The ID stays in st.pages so block-table
+  // slots keep their position (the band mask guarantees those tokens are
+  // never scored); unref handles prefix-shared pages.  The 64-token
+  // slack covers attn_prefill.hip's KV-tile staging below the band.
+  void roll(SeqState& st) {
+    if (!window_tokens_) return;
+    const int lim = st.num_tokens - window_tokens_ - 64;
+    while ((st.freed_pages + 1) * page_size_ <= lim) {
+      unref(st.pages[st.freed_pages]);
+      st.freed_pages += 1;
+    }
+  }
+
   bool grow(SeqState& st) {
     int need = pages_needed(st.num_tokens + 1);
     if (need > (int)st.pages.size()) {
@@ -305,8 +323,10 @@ class NativeScheduler {
   }
 
   void release(SeqState& st) {
-    for (int p : st.pages) unref(p);
+    for (size_t k = st.freed_pages; k < st.pages.size(); ++k)
+      unref(st.pages[k]);
     st.pages.clear();
+    st.freed_pages = 0;
     st.cached_prefix = 0;
   }
 
@@ -374,6 +394,7 @@ class NativeScheduler {
 
   int max_num_seqs_, max_prefill_tokens_, page_size_, num_pages_, max_waiting_;
   bool prefix_cache_ = false;
+  int window_tokens_ = 0;
   std::vector<int> free_list_;
   std::deque<long> waiting_;
   std::vector<long> running_;
@@ -402,10 +423,11 @@ PYBIND11_MODULE(_native, m) {
       .def_readonly("preempted", &ScheduleResult::preempted);
 
   py::class_<NativeScheduler>(m, "NativeScheduler")
-      .def(py::init<int, int, int, int, int, bool>(), py::arg("max_num_seqs"),
+      .def(py::init<int, int, int, int, int, bool, int>(),
+           py::arg("max_num_seqs"),
            py::arg("max_prefill_tokens"), py::arg("page_size"),
            py::arg("num_pages"), py::arg("max_waiting") = 4096,
-           py::arg("prefix_cache") = false)
+           py::arg("prefix_cache") = false, py::arg("window_tokens") = 0)
       .def("add", &NativeScheduler::add, py::arg("seq_id"),
            py::arg("prompt_len"),
            py::arg("page_hashes") = std::vector<long>{})

@@ -5,6 +5,7 @@ import math
 import sys
 from pathlib import Path
 
+import pytest
 import torch
 
 sys.path.insert(0, str(Path(__file__).resolve().parent))
@@ -111,3 +112,92 @@ def test_engine_sliding_window_end_to_end():
 def test_mistral_config_has_window():
     assert CONFIGS["mistral-7b"].sliding_window == 4096
     assert CONFIGS["mistral-7b"].shard(2).sliding_window == 4096
+
+
+def _mk_seq(sid, n, sp=None):
+    from agentfield_amd.engine.sequence import Sequence
+    return Sequence(seq_id=sid, prompt_ids=list(range(1, n + 1)),
+                    sampling=sp or SamplingParams(max_tokens=64,
+                                                  ignore_eos=True))
+
+
+def test_rolling_window_reclaims_pages():
+    """Pages wholly behind the band (minus the 64-token staging slack)
+    return to the pool during decode; release() must not double-free."""
+    from agentfield_amd.engine.scheduler import Scheduler, SchedulerConfig
+    cfg = SchedulerConfig(max_num_seqs=2, max_prefill_tokens=4096,
+                          page_size=4, num_pages=128, window_tokens=16)
+    s = Scheduler(cfg)
+    seq = _mk_seq(1, 40)
+    assert s.add(seq)
+    b = s.schedule()
+    assert b.is_prefill
+    free_after_prefill = s.alloc.num_free
+    for _ in range(60):
+        seq.output_ids.append(7)
+        s.schedule()
+    # lim = 100 - 16 - 64 = 20 tokens -> 5 pages reclaimed
+    assert seq.freed_pages == 5
+    # grow() keeps capacity for num_tokens+1: 40 -> 101 tokens = 10 -> 26
+    # pages (+16), 5 reclaimed
+    assert s.alloc.num_free == free_after_prefill - 16 + 5
+    total_free_before_finish = s.alloc.num_free
+    s.finish(seq)
+    assert s.alloc.num_free == 127  # all pages back exactly once
+    assert s.alloc.num_free > total_free_before_finish
+
+
+def test_rolling_window_native_lockstep():
+    """C++ scheduler reclaims identically to the Python oracle."""
+    pytest.importorskip("agentfield_amd._native")
+    from agentfield_amd._native import NativeScheduler
+    from agentfield_amd.engine.scheduler import Scheduler, SchedulerConfig
+    cfg = SchedulerConfig(max_num_seqs=2, max_prefill_tokens=4096,
+                          page_size=4, num_pages=64, window_tokens=16)
+    py = Scheduler(cfg)
+    nat = NativeScheduler(2, 4096, 4, 64, 4096, False, 16)
+    seq = _mk_seq(1, 40)
+    py.add(seq)
+    nat.add(1, 40)
+    py.schedule()
+    nat.schedule()
+    for step in range(50):
+        seq.output_ids.append(9)
+        nat.note_token(1)
+        py.schedule()
+        nat.schedule()
+        assert nat.num_free() == py.alloc.num_free, step
+    py.finish(seq)
+    nat.finish(1)
+    assert nat.num_free() == py.alloc.num_free == 63
+
+
+def test_engine_rolling_outputs_identical(monkeypatch):
+    """AF_KV_ROLL on/off: identical outputs (reclaimed pages are never
+    scored), strictly more free pages with rolling."""
+    base = CONFIGS["tiny"]
+    swa = LlamaConfig(name="tiny-swa2", hidden_size=base.hidden_size,
+                      intermediate_size=base.intermediate_size,
+                      num_layers=base.num_layers, num_heads=base.num_heads,
+                      num_kv_heads=base.num_kv_heads,
+                      vocab_size=base.vocab_size,
+                      max_position=base.max_position, sliding_window=16)
+
+    def run(roll: str):
+        monkeypatch.setenv("AF_KV_ROLL", roll)
+        eng = LLMEngine(swa, device="cpu", dtype=torch.float32, page_size=4,
+                        num_pages=128, max_num_seqs=2, enable_graphs=False,
+                        seed=5)
+        rid = eng.add_request(list(range(1, 61)),
+                              SamplingParams(max_tokens=48, ignore_eos=True))
+        for _ in range(300):
+            eng.step()
+            f = eng.get_finished(rid)
+            if f:
+                return f.output_ids, eng.sched.alloc.num_free
+        raise AssertionError("did not finish")
+
+    out_roll, free_roll = run("1")
+    out_flat, free_flat = run("0")
+    assert out_roll == out_flat
+    assert len(out_roll) == 48
