@@ -417,3 +417,32 @@ def test_schema_constrained_decoding_on_gpu():
             assert data["s"] in ("a", "bb")
         if "t" in data:
             assert all(isinstance(v, str) for v in data["t"])
+
+
+@pytest.mark.gpu
+def test_sliding_window_rolling_on_gpu(monkeypatch):
+    """Windowed decode on the HIP kernels: rolling page reclamation must
+    not change outputs (greedy), and must free pages."""
+    from agentfield_amd.models.llama import LlamaConfig
+    base = CONFIGS["tiny"]
+    swa = LlamaConfig(name="tiny-swa-gpu", hidden_size=base.hidden_size,
+                      intermediate_size=base.intermediate_size,
+                      num_layers=base.num_layers, num_heads=base.num_heads,
+                      num_kv_heads=base.num_kv_heads,
+                      vocab_size=base.vocab_size,
+                      max_position=base.max_position, sliding_window=16)
+
+    def run(roll):
+        monkeypatch.setenv("AF_KV_ROLL", roll)
+        eng = LLMEngine(swa, device="cuda", page_size=4, num_pages=256,
+                        max_num_seqs=2, enable_graphs=True, seed=5)
+        rid = eng.add_request(list(range(1, 81)),
+                              SamplingParams(max_tokens=64, ignore_eos=True))
+        for _ in range(400):
+            eng.step()
+            f = eng.get_finished(rid)
+            if f:
+                return f.output_ids
+        raise AssertionError("did not finish")
+
+    assert run("1") == run("0")
