@@ -120,6 +120,7 @@ class PrefixCachingScheduler(Scheduler):
         cfg = self.cfg
         batch: list[Sequence] = []
         tokens = 0
+        claimed: set[int] = set()  # first-uncached-page hashes this batch
         while (self.waiting and
                len(self.running) + len(batch) < cfg.max_num_seqs):
             cand = self.waiting[0]
@@ -134,6 +135,14 @@ class PrefixCachingScheduler(Scheduler):
                 if page is None:
                     break
                 matched.append(page)
+            # same-batch dedup: if another candidate in THIS batch is about
+            # to compute the page this one needs next, defer — one round
+            # later the publisher's pages are in the cache and this prompt
+            # prefills as a hit instead of a duplicate (FIFO preserved)
+            nxt = (hashes[len(matched)] if len(matched) < max_full
+                   else None)
+            if nxt is not None and nxt in claimed:
+                break
             cached_tok = len(matched) * cfg.page_size
             # num_tokens includes outputs retained across preemption
             ntok = cand.num_tokens - cached_tok
@@ -155,6 +164,8 @@ class PrefixCachingScheduler(Scheduler):
             cand.status = SeqStatus.RUNNING
             self.cache_hits += len(matched)
             self.cached_tokens += cached_tok
+            if nxt is not None:
+                claimed.add(nxt)
             batch.append(cand)
             tokens += ntok
         if batch:

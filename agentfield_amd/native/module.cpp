@@ -14,6 +14,7 @@
 #include <stdexcept>
 #include <string>
 #include <unordered_map>
+#include <unordered_set>
 #include <vector>
 
 namespace py = pybind11;
@@ -211,20 +212,28 @@ class NativeScheduler {
     // 1) admit prefills
     int tokens = 0;
     std::vector<long> batch;
+    std::unordered_set<long> claimed;  // first-uncached hashes this batch
     while (!waiting_.empty() &&
            (int)(running_.size() + batch.size()) < max_num_seqs_) {
       long cand = waiting_.front();
       SeqState& st = seqs_[cand];
       // cached full-page prefix (>=1 prompt token always recomputed)
       std::vector<int> matched;
+      int max_full = (st.prompt_len - 1) / page_size_;
       if (prefix_cache_) {
-        int max_full = (st.prompt_len - 1) / page_size_;
         for (int k = 0; k < (int)st.hashes.size() && k < max_full; ++k) {
           auto it = cache_.find(st.hashes[k]);
           if (it == cache_.end()) break;
           matched.push_back(it->second->second);
         }
       }
+      // same-batch dedup (lockstep with the Python oracle): defer a
+      // candidate whose next page another batch member will compute
+      bool has_next = prefix_cache_ &&
+                      (int)matched.size() < max_full &&
+                      (int)matched.size() < (int)st.hashes.size();
+      long nxt = has_next ? st.hashes[matched.size()] : 0;
+      if (has_next && claimed.count(nxt)) break;
       int cached_tok = (int)matched.size() * page_size_;
       // num_tokens includes outputs retained across preemption — their KV
       // recomputes as prefill on re-admission
@@ -247,6 +256,7 @@ class NativeScheduler {
       st.cached_prefix = cached_tok;
       cache_hits_ += (long)matched.size();
       cached_tokens_ += cached_tok;
+      if (has_next) claimed.insert(nxt);
       batch.push_back(cand);
       tokens += ntok;
     }

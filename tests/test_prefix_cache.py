@@ -131,3 +131,57 @@ def test_prefix_cache_with_speculation_exact():
     assert out1 == base and out2 == base
     assert eng.sched.cached_tokens > 0, "prefix cache never hit"
     assert eng.metrics["spec_drafted"] > 0, "speculation never fired"
+
+
+def test_same_batch_duplicate_prompts_dedup():
+    """Two identical prompts submitted together, cold cache: the second
+    defers one round and then prefills as a cache hit instead of
+    computing the same pages twice."""
+    from agentfield_amd.engine.prefix_cache import PrefixCachingScheduler
+    from agentfield_amd.engine.scheduler import SchedulerConfig
+    from agentfield_amd.engine.sequence import Sequence
+
+    def mkseq(sid, ids):
+        return Sequence(seq_id=sid, prompt_ids=ids, sampling=SP)
+
+    cfg = SchedulerConfig(max_num_seqs=8, max_prefill_tokens=4096,
+                          page_size=4, num_pages=64)
+    sched = PrefixCachingScheduler(cfg)
+    a = mkseq(1, list(range(1, 21)))
+    b = mkseq(2, list(range(1, 21)))
+    sched.add(a)
+    sched.add(b)
+    r = sched.schedule()
+    assert r.is_prefill and [s.seq_id for s in r.seqs] == [1]  # b deferred
+    # prompt of a completes -> pages published
+    a.num_prefilled = len(a.prompt_ids)
+    sched.note_token(a)  # engine notes BEFORE appending the sampled token
+    a.output_ids.append(5)
+    r2 = sched.schedule()
+    assert r2.is_prefill and [s.seq_id for s in r2.seqs] == [2]
+    assert b.cached_prefix == 16  # 4 full pages shared, tail recomputed
+    assert sched.cache_hits >= 4
+    # distinct prompts are NOT deferred
+    sched2 = PrefixCachingScheduler(cfg)
+    c = mkseq(3, list(range(1, 21)))
+    d = mkseq(4, list(range(100, 120)))
+    sched2.add(c)
+    sched2.add(d)
+    r3 = sched2.schedule()
+    assert [s.seq_id for s in r3.seqs] == [3, 4]
+
+
+def test_same_batch_dedup_native_lockstep():
+    pytest.importorskip("agentfield_amd._native")
+    from agentfield_amd._native import NativeScheduler
+    from agentfield_amd.engine.prefix_cache import prefix_hashes
+    nat = NativeScheduler(8, 4096, 4, 64, 4096, True)
+    h = prefix_hashes(list(range(1, 21)), 4)
+    nat.add(1, 20, h)
+    nat.add(2, 20, h)
+    r = nat.schedule()
+    assert r.is_prefill and list(r.seq_ids) == [1]
+    nat.note_token(1)  # prompt prefilled: publish
+    r2 = nat.schedule()
+    assert list(r2.seq_ids) == [2]
+    assert nat.cached_prefix(2) == 16
