@@ -31,6 +31,61 @@ AF_EXPORT int af_mfma_probe(void* D, const void* A, const void* B, void* stream)
   return af_last_err();
 }
 
+// MX-fp8 layout probe: single-wave 16x16x128 e4m3 matmul via the scaled
+// MFMA (hardware-fused dequant).  Layouts mapped EMPIRICALLY on gfx950
+// (tools/mx_probe*.py, block-one-hot + single-byte scale perturbation):
+//   C: reg r -> C[m = (lane>>4)*4 + r][n = lane&15]  (shape-determined)
+//   scale: byte 0 (op_sel 0) of scale lane (m|n, s = lane>>4) covers
+//     hw K-block s (32 consecutive hw k) of that row/col; E8M0.
+//   data: lane (m, g = lane>>4), byte j maps to
+//     hw k = 32*(2*(j>>4) + (g>>1)) + 16*(g&1) + (j&15)
+//   i.e. a lane's 32 bytes SPAN TWO hw blocks (halves j<16 and j>=16) —
+//   the naive "lane g = block g" guess mispairs data with scales (it
+//   cancels for uniform scales, which is why data-only probes passed).
+// Loading with the inverse mapping below makes memory-k == hw-k, so a
+// per-32-memory-block scale vector works: sa[row][s] at lane (row, s).
+// A: [16,128] row-major e4m3 bytes, sa: [16,4] E8M0 per (row, k-block);
+// B: [128,16] col-read,         sb: [16,4] per (col, k-block); D [16,16] f32.
+#include <hip/hip_fp8.h>
+
+typedef __attribute__((ext_vector_type(8))) int i32x8p;
+
+__device__ __forceinline__ int mx_hw_k(int g, int j) {
+  return 32 * (2 * (j >> 4) + (g >> 1)) + 16 * (g & 1) + (j & 15);
+}
+
+__global__ void mfma_mx_probe_kernel(float* __restrict__ D,
+                                     const unsigned char* __restrict__ A,
+                                     const unsigned char* __restrict__ B,
+                                     const int* __restrict__ SA,
+                                     const int* __restrict__ SB) {
+  const int lane = threadIdx.x & 63;
+  const int m = lane & 15, kb = lane >> 4;
+  union { unsigned char u8[32]; i32x8p v; } a, b;
+#pragma unroll
+  for (int j = 0; j < 32; ++j) {
+    const int k = mx_hw_k(kb, j);
+    a.u8[j] = A[m * 128 + k];
+    b.u8[j] = B[k * 16 + m];
+  }
+  const int sa = SA[lane];  // full i32 scale operand, host-crafted per lane
+  const int sb = SB[lane];
+  f32x4 acc = {0, 0, 0, 0};
+  acc = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+      a.v, b.v, acc, 0, 0, 0, sa, 0, sb);
+#pragma unroll
+  for (int r = 0; r < 4; ++r)
+    D[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = acc[r];
+}
+
+AF_EXPORT int af_mfma_mx_probe(void* D, const void* A, const void* B,
+                               const void* SA, const void* SB, void* stream) {
+  mfma_mx_probe_kernel<<<1, 64, 0, (hipStream_t)stream>>>(
+      (float*)D, (const unsigned char*)A, (const unsigned char*)B,
+      (const int*)SA, (const int*)SB);
+  return af_last_err();
+}
+
 // Tiny helper so tests can verify the ctypes plumbing end-to-end without MFMA.
 __global__ void axpy_kernel(float* y, const float* x, float a, int n) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;

@@ -410,6 +410,19 @@ def gather_rows(x: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
     return out
 
 
+def mfma_mx_probe(a8: torch.Tensor, b8: torch.Tensor, sa: torch.Tensor,
+                  sb: torch.Tensor) -> torch.Tensor:
+    """Single-wave 16x16x128 MX-fp8 scaled MFMA with the assumed layouts.
+    a8 [16,128] u8 (e4m3 bytes), b8 [128,16] u8, sa/sb [16,4] u8 (E8M0
+    per 32-element K-block)."""
+    d = torch.empty(16, 16, dtype=torch.float32, device=a8.device)
+    rc = _lib.lib().af_mfma_mx_probe(_lib.ptr(d), _lib.ptr(a8), _lib.ptr(b8),
+                                     _lib.ptr(sa), _lib.ptr(sb),
+                                     _lib.cur_stream())
+    _lib.check(rc, "af_mfma_mx_probe")
+    return d
+
+
 def mfma_probe(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     """Single-wave 16x16x32 MFMA with the assumed layouts (GPU layout test)."""
     d = torch.empty(16, 16, dtype=torch.float32, device=a.device)
