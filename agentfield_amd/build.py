@@ -31,6 +31,7 @@ HIP_SOURCES = [
     "gemm.hip",
     "gemm_ring.hip",
     "gemm_q8.hip",
+    "gemm_mxfp8.hip",
     "allreduce.hip",
     "gemm_skinny.hip",
     "sampling.hip",

@@ -86,6 +86,40 @@ AF_EXPORT int af_mfma_mx_probe(void* D, const void* A, const void* B,
   return af_last_err();
 }
 
+// global_load_lds sub-dword semantics probe: each lane loads `size`
+// bytes from src + lane*size with a wave-uniform LDS base; dump the LDS
+// region so the host can see the actual per-lane LDS stride.
+__global__ void lds_stride_probe_kernel(unsigned char* __restrict__ out,
+                                        const unsigned char* __restrict__ src,
+                                        int size) {
+  __shared__ unsigned char lds[1024];
+  for (int i = threadIdx.x; i < 1024; i += 64) lds[i] = 0xEE;
+  __syncthreads();
+  const unsigned char* s = src + threadIdx.x * size;
+  if (size == 1)
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) uint32_t*)s,
+        (__attribute__((address_space(3))) uint32_t*)lds, 1, 0, 0);
+  else if (size == 2)
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) uint32_t*)s,
+        (__attribute__((address_space(3))) uint32_t*)lds, 2, 0, 0);
+  else
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) uint32_t*)s,
+        (__attribute__((address_space(3))) uint32_t*)lds, 4, 0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  for (int i = threadIdx.x; i < 1024; i += 64) out[i] = lds[i];
+}
+
+AF_EXPORT int af_lds_stride_probe(void* out, const void* src, int size,
+                                  void* stream) {
+  lds_stride_probe_kernel<<<1, 64, 0, (hipStream_t)stream>>>(
+      (unsigned char*)out, (const unsigned char*)src, size);
+  return af_last_err();
+}
+
 // Tiny helper so tests can verify the ctypes plumbing end-to-end without MFMA.
 __global__ void axpy_kernel(float* y, const float* x, float a, int n) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;

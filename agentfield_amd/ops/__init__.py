@@ -346,6 +346,28 @@ def gemm_bf16_q8(a: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return c
 
 
+def gemm_mxfp8(a8: torch.Tensor, sa: torch.Tensor, w8: torch.Tensor,
+               sw: torch.Tensor, M: int | None = None,
+               N: int | None = None, K: int | None = None) -> torch.Tensor:
+    """C[M,N] bf16 = dequant(a8,sa) @ dequant(w8,sw)^T on the CDNA4
+    block-scaled MFMA (MX-fp8: OCP e4m3 + E8M0 per-32-K-block scales,
+    quantize with agentfield_amd.quant.quantize_mx).  K % 128 == 0,
+    K >= 256."""
+    M = M or a8.shape[0]
+    K = K or a8.shape[1]
+    N = N or w8.shape[0]
+    if not _on_gpu(a8):
+        from ..quant import dequantize_mx
+        return (dequantize_mx(a8, sa) @ dequantize_mx(w8, sw).t()).to(
+            torch.bfloat16)
+    c = torch.empty(M, N, dtype=torch.bfloat16, device=a8.device)
+    rc = _lib.lib().af_gemm_mxfp8(_lib.ptr(c), _lib.ptr(a8), _lib.ptr(sa),
+                                  _lib.ptr(w8), _lib.ptr(sw), M, N, K,
+                                  _lib.cur_stream())
+    _lib.check(rc, "af_gemm_mxfp8")
+    return c
+
+
 # ---------------------------------------------------------------- sampling
 class SamplerState:
     """Device-side scratch for graph-capturable sampling."""

@@ -407,3 +407,25 @@ def test_engine_topk_sampling_runs():
     outs = eng.generate([[1, 2, 3]], SamplingParams(
         max_tokens=6, temperature=0.9, top_k=10, top_p=0.9, ignore_eos=True))
     assert len(outs[0]) == 6
+
+
+def test_gemm_mxfp8():
+    """MX-fp8 block-scaled GEMM vs the exact dequantized fp32 product
+    (hardware dequant is exact for power-of-two E8M0 scales; only fp32
+    accumulation order differs)."""
+    from agentfield_amd.quant import dequantize_mx, quantize_mx
+    for (M, N, K) in [(256, 256, 256), (512, 512, 1024), (300, 256, 512)]:
+        torch.manual_seed(M + K)
+        A = torch.randn(M, K) * 2.0
+        W = torch.randn(N, K) * 2.0
+        A[min(3, M - 1)] *= 29.0   # exercise non-uniform scales
+        W[min(7, N - 1)] *= 0.02
+        a8, sa = quantize_mx(A)
+        w8, sw = quantize_mx(W)
+        want = (dequantize_mx(a8, sa) @ dequantize_mx(w8, sw).t())
+        c = ops.gemm_mxfp8(a8.to(DEV), sa.to(DEV), w8.to(DEV), sw.to(DEV),
+                           M=M, N=N, K=K)
+        torch.cuda.synchronize()
+        rel = (c.float().cpu() - want).abs().max().item() / \
+            want.abs().max().item()
+        assert rel < 5e-3, (M, N, K, rel)

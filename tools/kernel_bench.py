@@ -45,6 +45,27 @@ def bench_gemm():
         if K % 64 == 0 and K >= 128 and M >= 256:
             t_q8 = timeit(lambda: ops.gemm_bf16_q8(a, w))
             row["q8_tflops"] = round(tf / t_q8, 1)
+        if K % 128 == 0 and K >= 256 and M >= 256:
+            from agentfield_amd.quant import quantize_mx
+            a8, sa = quantize_mx(a.float().cpu())
+            w8, sw = quantize_mx(w.float().cpu())
+            a8, sa, w8, sw = (t.to(DEV) for t in (a8, sa, w8, sw))
+            t_mx = timeit(lambda: ops.gemm_mxfp8(a8, sa, w8, sw))
+            row["mxfp8_tflops"] = round(tf / t_mx, 1)
+            # torch fp8 rowwise-scaled baseline when available
+            try:
+                af = a.float()
+                wf = w.float()
+                sa_r = af.abs().amax(1, keepdim=True) / 448.0
+                sw_r = wf.abs().amax(1, keepdim=True) / 448.0
+                a_f8 = (af / sa_r).to(torch.float8_e4m3fn)
+                w_f8 = (wf / sw_r).to(torch.float8_e4m3fn)
+                t_s = timeit(lambda: torch._scaled_mm(
+                    a_f8, w_f8.t(), scale_a=sa_r, scale_b=sw_r.t(),
+                    out_dtype=torch.bfloat16))
+                row["scaled_mm_tflops"] = round(tf / t_s, 1)
+            except Exception as e:
+                row["scaled_mm_tflops"] = str(e)[:60]
         print(json.dumps(row))
 
 
