@@ -86,6 +86,51 @@ AF_EXPORT int af_mfma_mx_probe(void* D, const void* A, const void* B,
   return af_last_err();
 }
 
+// MX-fp8 32x32x64 layout probe (analog of the 16x16x128 probe): assumed
+//   A: lane (m = lane&31, g = lane>>5) byte j -> hw k with the same
+//      half-interleave: k = 16*(2*(j>>4) + ???)  — start from the naive
+//      guess k = 32*g + j and let the host diagnose via scales.
+//   C: reg r -> C[row = (r&3) + 8*(r>>2) + 4*(lane>>5)][col = lane&31]
+//   scale: lane (m, s = lane>>5) byte0 covers hw K-block s (2 blocks)
+__global__ void mfma_mx32_probe_kernel(float* __restrict__ D,
+                                       const unsigned char* __restrict__ A,
+                                       const unsigned char* __restrict__ B,
+                                       const int* __restrict__ SA,
+                                       const int* __restrict__ SB,
+                                       int layout) {
+  const int lane = threadIdx.x & 63;
+  const int m = lane & 31, g = lane >> 5;
+  union { unsigned char u8[32]; i32x8p v; } a, b;
+#pragma unroll
+  for (int j = 0; j < 32; ++j) {
+    int k;
+    if (layout == 0) k = g * 32 + j;                       // naive
+    else k = 16 * (2 * (j >> 4) + g) + (j & 15);           // interleave
+    a.u8[j] = A[m * 64 + k];
+    b.u8[j] = B[k * 32 + m];
+  }
+  typedef __attribute__((ext_vector_type(16))) float f32x16p;
+  f32x16p acc;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) acc[r] = 0.f;
+  acc = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+      a.v, b.v, acc, 0, 0, 0, SA[lane], 0, SB[lane]);
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+    D[row * 32 + (lane & 31)] = acc[r];
+  }
+}
+
+AF_EXPORT int af_mfma_mx32_probe(void* D, const void* A, const void* B,
+                                 const void* SA, const void* SB, int layout,
+                                 void* stream) {
+  mfma_mx32_probe_kernel<<<1, 64, 0, (hipStream_t)stream>>>(
+      (float*)D, (const unsigned char*)A, (const unsigned char*)B,
+      (const int*)SA, (const int*)SB, layout);
+  return af_last_err();
+}
+
 // global_load_lds sub-dword semantics probe: each lane loads `size`
 // bytes from src + lane*size with a wave-uniform LDS base; dump the LDS
 // region so the host can see the actual per-lane LDS stride.

@@ -70,13 +70,14 @@ def _declare(l: ctypes.CDLL) -> None:
     l.af_mfma_probe.argtypes = [p, p, p, p]
     l.af_mfma_mx_probe.argtypes = [p, p, p, p, p, p]
     l.af_lds_stride_probe.argtypes = [p, p, i, p]
+    l.af_mfma_mx32_probe.argtypes = [p, p, p, p, p, i, p]
     l.af_axpy.argtypes = [p, p, f, i, p]
     l.af_device_sync.argtypes = []
     for fn in ("af_rmsnorm", "af_rope_cache", "af_silu_mul", "af_add",
                "af_reshape_and_cache", "af_embedding", "af_attn_decode",
                "af_attn_prefill", "af_gemm_bf16", "af_gemm_bf16_ring", "af_gemm_bf16_q8", "af_gemm_mxfp8", "af_oneshot_allreduce",
                "af_gemm_skinny", "af_sample",
-               "af_sample_topkp", "af_gather_rows", "af_mfma_probe", "af_mfma_mx_probe", "af_lds_stride_probe", "af_axpy",
+               "af_sample_topkp", "af_gather_rows", "af_mfma_probe", "af_mfma_mx_probe", "af_lds_stride_probe", "af_mfma_mx32_probe", "af_axpy",
                "af_device_sync"):
         getattr(l, fn).restype = ctypes.c_int
 
