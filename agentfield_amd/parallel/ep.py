@@ -1,0 +1,158 @@
+"""Expert parallelism (EP) for the sparse-MoE family: experts are
+sharded across ranks and tokens travel to their experts over an
+all-to-all exchange (RCCL over xGMI on the GPU; gloo lacks all_to_all,
+so the same exchange runs over batched isend/irecv there — identical
+semantics, which is what the CPU tests pin).
+
+Topology: rank r owns experts [r*E/ep, (r+1)*E/ep).  The router is
+replicated (tiny), so every rank computes identical top-k choices for
+ITS tokens; each forward is:
+
+  1. route local tokens, sort the (token, expert-slot) pairs by owner
+     rank, count per-destination
+  2. all-to-all the counts, then the activations (one [sum, H] tensor
+     split by destination — xGMI is point-to-point, so one large
+     all_to_all_single beats per-peer sends)
+  3. run the LOCAL experts' fused gate_up/SwiGLU/down on the received
+     rows (grouped per expert — exact dispatch, FLOP-proportional)
+  4. all-to-all the results back and combine with the renormalized
+     gate weights
+
+Capacity note: the exchange is exact (variable splits), not
+static-capacity — hipGraph capture of EP decode would need the static
+variant (engine MoE decode already has one for the single-rank path).
+"""
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+from torch import nn
+
+from .. import ops
+
+
+def _exchange(t: torch.Tensor, out_splits: list[int], in_splits: list[int],
+              group) -> torch.Tensor:
+    """all_to_all_single with a gloo fallback (batched isend/irecv)."""
+    out = torch.empty(sum(out_splits), *t.shape[1:], dtype=t.dtype,
+                      device=t.device)
+    backend = dist.get_backend(group)
+    if backend != "gloo":
+        dist.all_to_all_single(out, t.contiguous(),
+                               output_split_sizes=out_splits,
+                               input_split_sizes=in_splits, group=group)
+        return out
+    rank = dist.get_rank(group)
+    world = dist.get_world_size(group)
+    in_off = [0]
+    for s in in_splits:
+        in_off.append(in_off[-1] + s)
+    out_off = [0]
+    for s in out_splits:
+        out_off.append(out_off[-1] + s)
+    reqs = []
+    tc = t.contiguous()
+    for peer in range(world):
+        if peer == rank:
+            out[out_off[peer]:out_off[peer + 1]] = \
+                tc[in_off[peer]:in_off[peer + 1]]
+            continue
+        if in_splits[peer]:
+            reqs.append(dist.isend(
+                tc[in_off[peer]:in_off[peer + 1]].contiguous(), peer,
+                group=group))
+        if out_splits[peer]:
+            buf = torch.empty(out_splits[peer], *t.shape[1:], dtype=t.dtype,
+                              device=t.device)
+            reqs.append((dist.irecv(buf, peer, group=group), peer, buf))
+    for r in reqs:
+        if isinstance(r, tuple):
+            req, peer, buf = r
+            req.wait()
+            out[out_off[peer]:out_off[peer + 1]] = buf
+        else:
+            r.wait()
+    return out
+
+
+class EPMoE(nn.Module):
+    """Expert-parallel Mixtral-style sparse FFN.  Construct from a full
+    MoEMLP's weights via `shard_from` (each rank keeps its expert slice
+    plus the replicated router)."""
+
+    def __init__(self, num_experts: int, top_k: int, hidden: int,
+                 inter: int, group=None):
+        super().__init__()
+        self.group = group
+        self.world = dist.get_world_size(group)
+        self.rank = dist.get_rank(group)
+        assert num_experts % self.world == 0, \
+            "num_experts must divide the EP group"
+        self.E = num_experts
+        self.e_local = num_experts // self.world
+        self.top_k = top_k
+        self.router = nn.Parameter(torch.empty(num_experts, hidden))
+        self.gate_up = nn.Parameter(torch.empty(self.e_local, 2 * inter,
+                                                hidden))
+        self.down = nn.Parameter(torch.empty(self.e_local, hidden, inter))
+
+    @classmethod
+    def shard_from(cls, moe, group=None) -> "EPMoE":
+        """Slice a full MoEMLP (models.llama) onto this rank."""
+        E, H = moe.router.shape
+        inter = moe.down.shape[2]
+        ep = cls(E, moe.top_k, H, inter, group=group)
+        lo = ep.rank * ep.e_local
+        with torch.no_grad():
+            ep.router.copy_(moe.router)
+            ep.gate_up.copy_(moe.gate_up[lo:lo + ep.e_local])
+            ep.down.copy_(moe.down[lo:lo + ep.e_local])
+        return ep
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        T, H = x.shape
+        dev = x.device
+        # 1) replicated fp32 routing (identical to MoEMLP.forward)
+        probs = torch.softmax(x.float() @ self.router.float().t(), dim=-1)
+        topv, topi = probs.topk(self.top_k, dim=-1)
+        topv = topv / topv.sum(dim=-1, keepdim=True)
+
+        # flatten (token, slot) pairs and sort by owning rank (stable so
+        # the inverse permutation reassembles results deterministically)
+        flat_e = topi.reshape(-1)                      # [T*k]
+        owner = flat_e // self.e_local
+        order = torch.argsort(owner, stable=True)
+        inv = torch.empty_like(order)
+        inv[order] = torch.arange(order.numel(), device=dev)
+        send_tok = (torch.arange(T, device=dev)
+                    .repeat_interleave(self.top_k))[order]
+        send_e = flat_e[order]
+        send_x = x[send_tok]
+
+        in_splits = torch.bincount(owner, minlength=self.world).tolist()
+        # 2) exchange counts, then rows (+ expert ids alongside)
+        counts = torch.tensor(in_splits, dtype=torch.int64)
+        all_counts = [torch.zeros_like(counts) for _ in range(self.world)]
+        dist.all_gather(all_counts, counts, group=self.group)
+        out_splits = [int(c[self.rank]) for c in all_counts]
+        rx = _exchange(send_x, out_splits, in_splits, self.group)
+        re = _exchange(send_e.unsqueeze(1).to(torch.int64), out_splits,
+                       in_splits, self.group).squeeze(1)
+
+        # 3) local experts on received rows (exact grouped dispatch)
+        ry = torch.zeros(rx.shape[0], H, dtype=torch.float32, device=dev)
+        lo = self.rank * self.e_local
+        for le in range(self.e_local):
+            sel = (re == lo + le).nonzero(as_tuple=True)[0]
+            if sel.numel() == 0:
+                continue
+            xe = rx[sel]
+            ye = ops.linear(ops.linear(xe, self.gate_up[le], silu_fuse=True),
+                            self.down[le])
+            ry[sel] = ye.float()
+
+        # 4) results return to the sender; combine with gate weights
+        back = _exchange(ry.to(x.dtype), in_splits, out_splits, self.group)
+        back = back[inv].reshape(T, self.top_k, H)
+        out = (back.float() * topv.unsqueeze(-1)).sum(dim=1)
+        return out.to(x.dtype)
