@@ -45,6 +45,8 @@ class LLMEngine:
                  max_waiting: int = 4096, enable_graphs: bool = True, eos_id: int = 2, seed: int = 0,
                  model: LlamaForCausalLM | None = None,
                  tp_group=None, spec_lookup: int = 0,
+                 spec_draft=None, spec_draft_k: int = 4,
+                 draft_model: "LlamaForCausalLM | None" = None,
                  prefix_cache: bool = False):
         self.cfg = cfg
         self.device = torch.device(device)
@@ -67,6 +69,29 @@ class LLMEngine:
         max_pages_per_seq = (cfg.max_position + page_size - 1) // page_size
         self.max_pages_per_seq = max_pages_per_seq
         self.kv = KVCache(cfg, num_pages, page_size, self.device, dtype)
+        # draft-model speculation (greedy-exact): a small model proposes
+        # spec_draft_k tokens per step; the main model verifies them in
+        # one chunked forward (_step_verify — shared with prompt-lookup).
+        # The draft KV cache REUSES the sequences' page indices (its own
+        # pool, same geometry), so preemption/allocation stay in sync for
+        # free; seq.draft_len tracks how far the draft's KV is valid.
+        self.draft = None
+        self.spec_draft_k = int(spec_draft_k)
+        if spec_draft is not None or draft_model is not None:
+            from ..models import CONFIGS as _CFGS
+            if draft_model is not None:
+                dcfg = draft_model.cfg
+            else:
+                dcfg = (_CFGS[spec_draft] if isinstance(spec_draft, str)
+                        else spec_draft)
+                draft_model = LlamaForCausalLM(
+                    dcfg, device=device, dtype=dtype).init_random(seed + 1)
+            assert dcfg.vocab_size == cfg.vocab_size, \
+                "draft and target must share a vocabulary"
+            self.draft = draft_model.eval()
+            self.draft_cfg = dcfg
+            self.draft_kv = KVCache(dcfg, num_pages, page_size, self.device,
+                                    dtype)
         import os as _os
         # rolling KV reclamation for sliding-window models (AF_KV_ROLL=0
         # keeps every page allocated — A/B/debug escape hatch)
@@ -111,6 +136,26 @@ class LLMEngine:
                         "prefill_steps": 0, "decode_steps": 0,
                         "spec_steps": 0, "spec_drafted": 0, "spec_accepted": 0}
         # prompt-lookup speculative decoding (greedy-exact, opt-in):
+        # draft-model speculation (greedy-exact): a small model proposes
+        # spec_draft_k tokens per step; the main model verifies them in
+        # one chunked forward (_step_verify — shared with prompt-lookup).
+        # The draft KV cache REUSES the sequences' page indices (its own
+        # pool, same geometry), so preemption/allocation stay in sync
+        # for free; seq.draft_len tracks how far the draft's KV is valid.
+        self.draft = None
+        self.spec_draft_k = int(spec_draft_k)
+        if spec_draft is not None or draft_model is not None:
+            from ..models import CONFIGS as _CFGS
+            if draft_model is not None:
+                dcfg = draft_model.cfg
+            else:
+                dcfg = (_CFGS[spec_draft] if isinstance(spec_draft, str)
+                        else spec_draft)
+                draft_model = LlamaForCausalLM(
+                    dcfg, device=device, dtype=dtype).init_random(seed + 1)
+            assert dcfg.vocab_size == cfg.vocab_size,                 "draft and target must share a vocabulary"
+            self.draft = draft_model.eval()
+            self.draft_cfg = dcfg
         # draft up to spec_lookup tokens from n-gram matches in the
         # sequence's own context and verify them in ONE chunked-prefill
         # forward; every emitted token is the model's true greedy token.
@@ -204,13 +249,17 @@ class LLMEngine:
                 self._prefilling.extend(batch.seqs)
             else:
                 self.metrics["steps"] += 1
-                if self.spec_lookup > 0 and self.tp_group is None and \
+                if (self.spec_lookup > 0 or self.draft is not None) and \
+                        self.tp_group is None and \
                         all(s.sampling.temperature == 0.0 and
                             s.sampling.top_k <= 0 and s.sampling.top_p >= 1.0
                             and not s.sampling.json_mode
                             for s in batch.seqs):
-                    drafts = {s.seq_id: self._draft_for(s)
-                              for s in batch.seqs}
+                    if self.draft is not None:
+                        drafts = self._draft_model_propose(batch.seqs)
+                    else:
+                        drafts = {s.seq_id: self._draft_for(s)
+                                  for s in batch.seqs}
                     if any(drafts.values()):
                         self.metrics["spec_steps"] += 1
                         return self._step_verify(batch, drafts)
@@ -418,6 +467,92 @@ class LLMEngine:
                 mask[i, self._json_allowed(seq)] = 0.0
         return mask.to(self.device)
 
+    # -- speculative decode: draft model (greedy-exact) ---------------------
+    def _spec_cap(self, seq: Sequence, k: int) -> int:
+        """Cap a draft to the generation budget (minus the bonus token)
+        and to KV capacity the scheduler already allocated."""
+        n = seq.num_tokens
+        remaining = len(seq.prompt_ids) + seq.sampling.max_tokens - n
+        capacity = len(seq.pages) * self.page_size - n
+        return max(0, min(k, remaining - 1, capacity))
+
+    @torch.no_grad()
+    def _draft_model_propose(self, seqs: list[Sequence]) -> dict:
+        """Autoregressive k-token proposals from the draft model.  First a
+        varlen catch-up chunk brings each sequence's draft KV up to its
+        current length (tokens the target emitted without the draft —
+        bonus tokens, re-admissions after preemption), then k batched
+        decode rounds propose greedily."""
+        dev = self.device
+        ks = {s.seq_id: self._spec_cap(s, self.spec_draft_k) for s in seqs}
+        live = [s for s in seqs if ks[s.seq_id] > 0]
+        if not live:
+            return {s.seq_id: [] for s in seqs}
+        # ---- catch-up: feed ctx[draft_len : n-1] (KV only) ----
+        cu_ids, cu_pos, cu_slots, cu_qs, cu_lens, cu_bt = [], [], [], [], [], []
+        for s in live:
+            dl = getattr(s, "draft_len", 0)
+            n = s.num_tokens
+            if dl >= n - 1:
+                continue
+            ctx = s.prompt_ids + s.output_ids
+            chunk = ctx[dl:n - 1]
+            cu_ids.extend(chunk)
+            cu_pos.extend(range(dl, n - 1))
+            cu_slots.extend(self._slot(s, i) for i in range(dl, n - 1))
+            cu_qs.append(dl)
+            cu_lens.append(len(chunk))
+            row = torch.zeros(self.max_pages_per_seq, dtype=torch.int32)
+            row[:len(s.pages)] = torch.tensor(s.pages, dtype=torch.int32)
+            cu_bt.append(row)
+            s.draft_len = n - 1
+        if cu_ids:
+            cu = [0]
+            for ln in cu_lens:
+                cu.append(cu[-1] + ln)
+            md = AttnMetadata(
+                is_prefill=True,
+                slots=torch.tensor(cu_slots, dtype=torch.int64, device=dev),
+                cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
+                seq_lens=cu_lens,
+                q_start=torch.tensor(cu_qs, dtype=torch.int32, device=dev),
+                block_table=torch.stack(cu_bt).to(dev))
+            self.draft(torch.tensor(cu_ids, dtype=torch.int32, device=dev),
+                       torch.tensor(cu_pos, dtype=torch.int32, device=dev),
+                       self.draft_kv, md)
+        # ---- k decode rounds over the live batch ----
+        B = len(live)
+        bt = torch.zeros(B, self.max_pages_per_seq, dtype=torch.int32)
+        for i, s in enumerate(live):
+            bt[i, :len(s.pages)] = torch.tensor(s.pages, dtype=torch.int32)
+        bt = bt.to(dev)
+        cur = torch.tensor([s.last_token for s in live], dtype=torch.int32,
+                           device=dev)
+        base = torch.tensor([s.num_tokens for s in live], dtype=torch.int32,
+                            device=dev)
+        out: dict = {s.seq_id: [] for s in seqs}
+        kmax = max(ks[s.seq_id] for s in live)
+        for j in range(kmax):
+            pos = base - 1 + j
+            # lanes past their own cap keep decoding (batch stays static)
+            # but write KV to the reserved null page; their outputs are
+            # discarded below
+            slots = torch.tensor(
+                [self._slot(s, s.num_tokens - 1 + j)
+                 if j < ks[s.seq_id] else 0 for s in live],
+                dtype=torch.int64, device=dev)
+            md = AttnMetadata(is_prefill=False, slots=slots,
+                              block_table=bt, seq_lens_t=base + j,
+                              nsplit=choose_nsplit(B,
+                                                   self.draft_cfg.num_kv_heads))
+            logits = self.draft(cur, pos, self.draft_kv, md)
+            cur = logits.argmax(dim=-1).to(torch.int32)
+            toks = cur.cpu().tolist()
+            for i, s in enumerate(live):
+                if j < ks[s.seq_id]:
+                    out[s.seq_id].append(int(toks[i]))
+        return out
+
     # -- speculative decode (prompt lookup, greedy-exact) -------------------
     def _draft_for(self, seq: Sequence) -> list[int]:
         """Propose a continuation from the most recent earlier occurrence of
@@ -493,6 +628,11 @@ class LLMEngine:
             while a < len(d) and t[a] == d[a]:
                 a += 1
             tok_lists.append(t[:a + 1])  # accepted prefix + bonus token
+            if self.draft is not None:
+                # draft KV valid through position n-1+a (last_token +
+                # the accepted prefix); the bonus token catches up next
+                # round
+                seq.draft_len = seq.num_tokens + a
             self.metrics["spec_accepted"] += a
             self.metrics["decode_tokens"] += a + 1
             if seq.sampling.logprobs > 0:

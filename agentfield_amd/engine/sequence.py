@@ -41,6 +41,7 @@ class Sequence:
     output_ids: list[int] = field(default_factory=list)
     pages: list[int] = field(default_factory=list)
     freed_pages: int = 0   # leading pages reclaimed by the rolling window
+    draft_len: int = 0     # tokens whose KV the spec DRAFT model holds
     finish_reason: str | None = None
     # per output token, when sampling.logprobs > 0:
     # {"logprob": float, "top": [(token_id, logprob), ...]}
@@ -67,6 +68,7 @@ class Sequence:
         new-sample suffix for temperature>0 streams."""
         self.num_prefilled = 0
         self.cached_prefix = 0
+        self.draft_len = 0  # draft KV pages were released with the rest
 
     def append(self, tok: int, eos_id: int) -> bool:
         """Append a generated token; returns True when the sequence finished."""

@@ -438,6 +438,11 @@ def main():
     ap.add_argument("--spec-lookup", type=int, default=0,
                     help="prompt-lookup speculative decoding draft length "
                          "(greedy-exact; 0 disables)")
+    ap.add_argument("--spec-draft", default=None,
+                    help="draft MODEL for speculative decoding (a config "
+                         "name, e.g. debug-1b for a llama-3-8b target; "
+                         "greedy-exact verify)")
+    ap.add_argument("--spec-draft-k", type=int, default=4)
     ap.add_argument("--prefix-cache", action="store_true",
                     help="share paged KV across requests with a common "
                          "prompt prefix (refcounted pages, LRU eviction)")
@@ -463,6 +468,7 @@ def main():
             max_num_seqs=kw.pop("max_num_seqs", args.max_num_seqs),
             enable_graphs=not args.no_graphs and device.startswith("cuda"),
             spec_lookup=args.spec_lookup,
+            spec_draft=args.spec_draft, spec_draft_k=args.spec_draft_k,
             prefix_cache=args.prefix_cache, **kw)
         return EngineRunner(eng, tokenizer)
 
