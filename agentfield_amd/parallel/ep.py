@@ -75,6 +75,24 @@ def _exchange(t: torch.Tensor, out_splits: list[int], in_splits: list[int],
     return out
 
 
+def build_ep_model(cfg, device="cpu", dtype=torch.float32,
+                   base_seed: int = 0, group=None):
+    """Full MoE model with every sparse FFN replaced by its EP shard.
+    Every rank builds the SAME full model (same seed) and slices its
+    experts — attention/embeddings stay replicated (EP-only layout; the
+    reference placement for MoE where expert weights dominate memory).
+    Returns a LlamaForCausalLM whose MoEMLP modules are EPMoE."""
+    from ..models.llama import LlamaForCausalLM, MoEMLP
+    model = LlamaForCausalLM(cfg, device=device,
+                             dtype=dtype).init_random(base_seed)
+    for layer in model.layers:
+        if isinstance(layer.mlp, MoEMLP):
+            layer.mlp = EPMoE.shard_from(layer.mlp, group=group).to(
+                device=device, dtype=dtype)
+    model.no_fused_decode = True  # collectives attach to module forward
+    return model
+
+
 class EPMoE(nn.Module):
     """Expert-parallel Mixtral-style sparse FFN.  Construct from a full
     MoEMLP's weights via `shard_from` (each rank keeps its expert slice

@@ -98,3 +98,41 @@ def _ep_empty_rank_case(rank, world):
 
 def test_ep_handles_empty_ranks():
     assert _spawn("_ep_empty_rank_case")
+
+
+def _ep_model_case(rank, world):
+    """Full MoE model with EP-sharded FFNs matches the single-rank model
+    logits exactly (attention replicated, experts exchanged)."""
+    from agentfield_amd.models.llama import (AttnMetadata, KVCache,
+                                             LlamaForCausalLM)
+    from agentfield_amd.parallel.ep import build_ep_model
+
+    full = LlamaForCausalLM(CFG, device="cpu",
+                            dtype=torch.float32).init_random(7)
+    ep = build_ep_model(CFG, device="cpu", dtype=torch.float32, base_seed=7)
+
+    T = 10
+    torch.manual_seed(2)
+    ids = torch.randint(0, CFG.vocab_size, (T,), dtype=torch.int32)
+    pos = torch.arange(T, dtype=torch.int32)
+
+    def fwd(model):
+        kv = KVCache(CFG, num_pages=16, page_size=8, device="cpu",
+                     dtype=torch.float32)
+        bt = torch.arange(1, 3, dtype=torch.int32)[None, :]
+        slots = (8 + torch.arange(T)).to(torch.int64)
+        md = AttnMetadata(is_prefill=True, slots=slots,
+                          cu_seqlens=torch.tensor([0, T], dtype=torch.int32),
+                          seq_lens=[T],
+                          q_start=torch.zeros(1, dtype=torch.int32),
+                          block_table=bt)
+        return model(ids, pos, kv, md)
+
+    a = fwd(full)
+    b = fwd(ep)
+    assert torch.allclose(a, b, atol=1e-4), (a - b).abs().max().item()
+    return True
+
+
+def test_ep_full_model_matches():
+    assert _spawn("_ep_model_case")
