@@ -23,8 +23,17 @@ import statistics
 import subprocess
 import sys
 import tempfile
+import threading
 import time
 from pathlib import Path
+
+
+def log(msg: str) -> None:
+    print(f"[cfg4 +{time.monotonic() - T0:.1f}s] {msg}", file=sys.stderr,
+          flush=True)
+
+
+T0 = time.monotonic()
 
 ROOT = Path(__file__).resolve().parent.parent
 sys.path.insert(0, str(ROOT))
@@ -46,6 +55,7 @@ def main():
 
     import torch
 
+    log("importing engine")
     from agentfield_amd.engine import LLMEngine
     from agentfield_amd.models import CONFIGS
     from agentfield_amd.sdk import Agent
@@ -59,6 +69,7 @@ def main():
     kw = {}
     if device == "cpu":
         kw = {"num_pages": 512, "max_num_seqs": 8, "dtype": torch.float32}
+    log(f"building engine ({args.model}, {device})")
     eng = LLMEngine(cfg, device=device,
                     max_num_seqs=kw.pop("max_num_seqs", 128),
                     max_prefill_tokens=args.prompt_chars * 160,
@@ -66,6 +77,7 @@ def main():
                     seed=0, **kw)
     runner = EngineRunner(eng, ByteTokenizer(cfg.vocab_size))
     set_runner(args.model, runner)
+    log("engine ready; starting control plane")
 
     cp_port = args.port
     agent_port = args.port + 500
@@ -92,8 +104,10 @@ def main():
             return {"text": agent.ai(prompt, max_tokens=gen,
                                      ignore_eos=True, temperature=0.0)}
 
+        log("control plane healthy; starting agent")
         _AgentServer(agent, agent_port).start()
         assert agent.register()
+        log("agent registered; driving load")
 
         out = asyncio.run(drive(args, cp_url, hook_port))
         print(json.dumps(out))
@@ -153,24 +167,38 @@ async def drive(args, cp_url: str, hook_port: int) -> dict:
                 body = await r.json()
                 submit_t[body["execution_id"]] = t0
 
+    stop_watch = threading.Event()
+
+    def watchdog():
+        while not stop_watch.wait(20.0):
+            log(f"watchdog: submitted={len(submit_t)} done={len(done)} "
+                f"expected={expected['n']}")
+
+    threading.Thread(target=watchdog, daemon=True).start()
     async with aiohttp.ClientSession(
             connector=aiohttp.TCPConnector(limit=0),
             timeout=aiohttp.ClientTimeout(total=900)) as session:
         # warmup (untimed): fills graphs/caches
         expected["n"] = args.warmup
+        log(f"warmup: submitting {args.warmup}")
         await asyncio.gather(*(submit(session, i)
                                for i in range(args.warmup)))
-        await asyncio.wait_for(all_done.wait(), timeout=600)
+        log("warmup submitted; waiting for webhooks")
+        await asyncio.wait_for(all_done.wait(), timeout=300)
+        log("warmup complete")
         done.clear()
         submit_t.clear()
         all_done.clear()
 
         expected["n"] = args.requests
         t0 = time.perf_counter()
+        log(f"timed: submitting {args.requests}")
         await asyncio.gather(*(submit(session, i)
                                for i in range(args.requests)))
-        await asyncio.wait_for(all_done.wait(), timeout=900)
+        log("timed submitted; waiting for webhooks")
+        await asyncio.wait_for(all_done.wait(), timeout=420)
         elapsed = time.perf_counter() - t0
+        stop_watch.set()
 
     lats = sorted(done[e] - submit_t[e] for e in done if e in submit_t)
     await runner.cleanup()
