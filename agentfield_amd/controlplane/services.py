@@ -18,6 +18,9 @@ from prometheus_client import (CollectorRegistry, Counter, Gauge, Histogram,
                                generate_latest)
 
 from . import status as st
+from ..logging_setup import get_logger
+
+log = get_logger("controlplane.webhooks")
 
 
 class Metrics:
@@ -160,11 +163,20 @@ class WebhookDispatcher:
     async def _worker(self):
         while True:
             execution_id = await self._queue.get()
-            await self._deliver(execution_id)
+            try:
+                await self._deliver(execution_id)
+            except Exception as e:  # a worker must never die w/ claims
+                log.warning("webhook delivery crashed",
+                            execution_id=execution_id, error=str(e))
 
     async def _deliver(self, execution_id: str):
         wh = self.storage.get_webhook(execution_id)
-        if not wh or not wh.get("payload"):
+        if not wh:
+            return
+        if not wh.get("payload"):
+            # claimed before the execution completed: release, the
+            # completion-time notify() re-drives it
+            self.storage.release_webhook_claim(execution_id)
             return
         body = wh["payload"].encode() if isinstance(wh["payload"], str) \
             else json.dumps(wh["payload"]).encode()

@@ -305,13 +305,29 @@ class Storage:
         self._exec("UPDATE execution_webhooks SET payload=? WHERE execution_id=?",
                    (json.dumps(payload), execution_id))
 
-    def try_mark_webhook_inflight(self, execution_id: str) -> bool:
-        """Idempotent in-flight claim (webhook_dispatcher.go:196)."""
+    def try_mark_webhook_inflight(self, execution_id: str,
+                                  stale_s: float = 60.0) -> bool:
+        """Idempotent in-flight claim (webhook_dispatcher.go:196).  The
+        claim stamps next_attempt_at = now + stale_s so a claim whose
+        worker died (process crash, orphaned queue item) becomes
+        reclaimable after the window instead of wedging forever."""
+        t = now()
         cur = self._exec(
-            """UPDATE execution_webhooks SET status='inflight'
-               WHERE execution_id=? AND status IN ('pending','retry')""",
-            (execution_id,))
+            """UPDATE execution_webhooks
+               SET status='inflight', next_attempt_at=?
+               WHERE execution_id=? AND
+                     (status IN ('pending','retry')
+                      OR (status='inflight' AND next_attempt_at <= ?))""",
+            (t + stale_s, execution_id, t))
         return cur.rowcount > 0
+
+    def release_webhook_claim(self, execution_id: str) -> None:
+        """Undo an in-flight claim that cannot proceed (e.g. the
+        execution has not completed yet, so no payload is staged)."""
+        self._exec(
+            """UPDATE execution_webhooks SET status='pending',
+               next_attempt_at=? WHERE execution_id=? AND status='inflight'""",
+            (now(), execution_id))
 
     def webhook_attempted(self, execution_id: str, ok: bool, status_code: int,
                           error: str | None, backoff_s: float,
@@ -335,9 +351,16 @@ class Storage:
             " VALUES (?,?,?,?,?)", (execution_id, attempts, status_code, error, now()))
 
     def due_webhooks(self, batch: int = 64) -> list[dict]:
+        # payload IS NOT NULL: a pending webhook of a still-RUNNING
+        # execution is not deliverable — claiming it pre-completion
+        # orphaned it forever (notify() skips inflight rows; measured:
+        # exactly one poller batch of 64 wedged under a 384-burst).
+        # Stale inflight claims (worker death) become due again via the
+        # claim's next_attempt_at stamp.
         rows = self._q(
             """SELECT * FROM execution_webhooks
-               WHERE status IN ('pending','retry') AND next_attempt_at <= ?
+               WHERE payload IS NOT NULL AND next_attempt_at <= ?
+                 AND status IN ('pending','retry','inflight')
                ORDER BY next_attempt_at LIMIT ?""", (now(), batch))
         for r in rows:
             r["headers"] = json.loads(r["headers"] or "{}")

@@ -817,3 +817,55 @@ def test_sdk_lock_context_manager(cp_server):
     # released -> n2 can take it
     assert DistributedLock(c2, "job", owner="n2", ttl_s=5.0,
                            timeout_s=1.0).acquire()
+
+
+def test_webhook_burst_poller_race():
+    """Burst of async executions with webhooks while the retry POLLER
+    runs: the poller must never claim a webhook whose execution has not
+    completed (that orphaned one whole batch as 'inflight' forever —
+    notify() skips inflight rows), and every webhook must deliver."""
+    from agentfield_amd.controlplane import ControlPlane, create_app
+    from agentfield_amd.controlplane.server import Config as CPConfig
+
+    cp = ControlPlane(CPConfig(background_services=True, sync_timeout=30.0,
+                               db_path=":memory:"))
+    srv = AppServer(create_app(cp)).start().wait_healthy()
+    # make the poller race hard: poll every 0.05s instead of 5s
+    cp.webhooks.poll_interval = 0.05
+    agent = Agent("burst", agentfield_url=srv.base_url, auto_register=False)
+
+    @agent.reasoner()
+    def slow(x: int = 0):
+        time.sleep(0.2)  # executions outlive several poller cycles
+        return {"x": x}
+
+    asrv = AppServer(agent).start()
+    agent.base_url = asrv.base_url
+    assert agent.register()
+
+    hits = []
+    hook = FastAPI()
+
+    @hook.post("/hook")
+    async def recv(req: Request):
+        hits.append(await req.body())
+        return {"ok": True}
+
+    hsrv = AppServer(hook).start()
+    N = 96
+    with httpx.Client(timeout=30.0) as c:
+        for i in range(N):
+            r = c.post(srv.base_url + "/api/v1/execute/async/burst.slow",
+                       json={"input": {"x": i},
+                             "webhook": {"url": hsrv.base_url + "/hook",
+                                         "secret": "s"}})
+            assert r.status_code == 202
+    wait_until(lambda: len(hits) >= N, timeout=90.0)
+
+    def all_marked():
+        rows = cp.storage._q("SELECT status, COUNT(*) c FROM "
+                             "execution_webhooks GROUP BY status", ())
+        return rows == [{"status": "delivered", "c": N}]
+
+    wait_until(all_marked, timeout=10.0)  # status write can lag the hook
+    srv.stop()
