@@ -419,11 +419,11 @@ class LLMEngine:
         if fsm is None or pos > len(seq.output_ids):
             spec = seq.sampling.json_schema
             if spec is not None:
-                from .schemafsm import SchemaFSM, SchemaSpec
-                if not isinstance(spec, SchemaSpec):
-                    spec = SchemaSpec(spec)
-                    seq.sampling.json_schema = spec  # compile once
-                fsm = SchemaFSM(spec)
+                from .schemafsm import SchemaFSM, SchemaSpec, make_fsm
+                if isinstance(spec, SchemaSpec):
+                    fsm = SchemaFSM(spec)
+                else:
+                    fsm = make_fsm(spec)  # dict (incl. root anyOf)
             else:
                 fsm = JsonFSM()
             pos = 0

@@ -619,3 +619,78 @@ class SchemaFSM:
             if nxt.min_close() <= remaining - 1:
                 out.append(b + BYTE_OFFSET)
         return out
+
+
+class MultiFSM:
+    """Root-level `anyOf`: an NFA of SchemaFSM alternatives.  A byte is
+    legal if ANY alive alternative accepts it; alternatives that reject
+    die.  Same surface as SchemaFSM/JsonFSM so the byte-mask path and
+    the token-trie walk drive it unchanged.  (Nested anyOf inside a
+    schema is not supported — compile fails loudly via make_fsm.)"""
+
+    __slots__ = ("alts",)
+
+    def __init__(self, alts):
+        self.alts = list(alts)
+        if not self.alts:
+            raise ValueError("anyOf needs at least one alternative")
+
+    def clone(self) -> "MultiFSM":
+        m = MultiFSM.__new__(MultiFSM)
+        m.alts = [a.clone() for a in self.alts]
+        return m
+
+    def sig(self) -> tuple:
+        return ("anyof",) + tuple(a.sig() for a in self.alts)
+
+    def complete(self) -> bool:
+        return any(a.complete() for a in self.alts)
+
+    def min_close(self) -> int:
+        return min(a.min_close() for a in self.alts)
+
+    def advance(self, b: int) -> None:
+        alive = []
+        for a in self.alts:
+            try:
+                c = a.clone()
+                c.advance(b)
+                alive.append(c)
+            except ValueError:
+                pass
+        if not alive:
+            raise ValueError(f"byte {bytes([b])!r} rejected by every "
+                             "anyOf alternative")
+        self.alts = alive
+
+    def _allowed_bytes(self) -> set:
+        out: set = set()
+        for a in self.alts:
+            out.update(a._allowed_raw())  # candidates; advance() re-checks
+        return out
+
+    def allowed_token_ids(self, remaining: int) -> list[int]:
+        from .jsonfsm import BYTE_OFFSET, EOS_ID
+        out = []
+        if self.complete():
+            out.append(EOS_ID)
+        budget = remaining - 1
+        for b in sorted(self._allowed_bytes()):
+            m = self.clone()
+            try:
+                m.advance(b)
+            except ValueError:
+                continue
+            if m.min_close() <= budget:
+                out.append(b + BYTE_OFFSET)
+        return out
+
+
+def make_fsm(schema):
+    """Compile a JSON value into the right automaton: a root-level anyOf
+    becomes a MultiFSM of alternatives, anything else a SchemaFSM."""
+    if isinstance(schema, SchemaSpec):
+        return SchemaFSM(schema)
+    if isinstance(schema, dict) and isinstance(schema.get("anyOf"), list):
+        return MultiFSM(SchemaFSM(SchemaSpec(s)) for s in schema["anyOf"])
+    return SchemaFSM(SchemaSpec(schema))

@@ -259,7 +259,43 @@ def create_engine_app(runner: EngineRunner, model_name: str,
                                ".schema must be an object",
                                "type": "invalid_request_error"}},
                     status_code=400)
-        jm = rf.get("type") in ("json_object", "json_schema")
+        # OpenAI tool calling: with tool_choice "required" or a named
+        # function, the call is GUARANTEED valid by schema-constrained
+        # decoding (anyOf over the tool schemas).  "auto"/"none" generate
+        # unconstrained (random-init models have no trained tool tokens;
+        # real checkpoints would use their chat template's tool markers).
+        tools = body.get("tools") or []
+        tool_choice = body.get("tool_choice", "auto" if tools else "none")
+        tool_mode = False
+        if kind == "chat" and tools and tool_choice not in ("none", "auto"):
+            chosen = tools
+            if isinstance(tool_choice, dict):
+                want = tool_choice.get("function", {}).get("name")
+                chosen = [t for t in tools
+                          if t.get("function", {}).get("name") == want]
+                if not chosen:
+                    return JSONResponse(
+                        {"error": {"message": f"unknown tool {want!r}",
+                                   "type": "invalid_request_error"}},
+                        status_code=400)
+            alts = []
+            for t in chosen:
+                fn = t.get("function", {})
+                alts.append({
+                    "type": "object",
+                    "properties": {
+                        "name": {"enum": [fn.get("name", "fn")]},
+                        "arguments": fn.get("parameters") or {},
+                    },
+                    "required": ["name", "arguments"]})
+            schema = alts[0] if len(alts) == 1 else {"anyOf": alts}
+            tool_mode = True
+            # surface the tool signatures to the model (a real checkpoint
+            # would get these through its chat template's tool section)
+            prompt = ("Available tools:\n" +
+                      json.dumps([t.get("function", {}) for t in chosen]) +
+                      "\nRespond with a single JSON tool call.\n" + prompt)
+        jm = rf.get("type") in ("json_object", "json_schema") or tool_mode
         if jm and not _json_ready(runner):
             return JSONResponse(
                 {"error": {"message": f"response_format {rf.get('type')} "
@@ -313,7 +349,29 @@ def create_engine_app(runner: EngineRunner, model_name: str,
                         text, stopped = text[:idx], True
                 fr = _finish_reason(len(w["output"]), max_tokens, stopped)
                 out_tokens += len(w["output"])
-                if kind == "chat":
+                if kind == "chat" and tool_mode:
+                    import uuid as _uuid
+                    try:
+                        call = json.loads(text.strip())
+                        args = call.get("arguments", {})
+                        choice = {"index": i, "finish_reason": "tool_calls",
+                                  "message": {"role": "assistant",
+                                              "content": None,
+                                              "tool_calls": [{
+                                                  "id": "call_" +
+                                                  _uuid.uuid4().hex[:24],
+                                                  "type": "function",
+                                                  "function": {
+                                                      "name": call.get("name"),
+                                                      "arguments":
+                                                      json.dumps(args)}}]}}
+                    except Exception:
+                        # truncated by max_tokens before the close — the
+                        # grammar guarantees a prefix of valid JSON only
+                        choice = {"index": i, "finish_reason": "length",
+                                  "message": {"role": "assistant",
+                                              "content": text}}
+                elif kind == "chat":
                     choice = {"index": i, "finish_reason": fr,
                               "message": {"role": "assistant",
                                           "content": text}}

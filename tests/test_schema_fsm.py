@@ -206,3 +206,57 @@ def test_ai_schema_guarantee():
     data = json.loads(out)
     assert set(data) == {"answer"} and isinstance(data["answer"], str)
     runner.shutdown()
+
+
+def test_root_anyof_tool_call_shapes():
+    """Root-level anyOf (NFA of SchemaFSM alternatives): masks admit the
+    union, advance prunes dead branches, completion/min_close take the
+    best alternative — the shape tool-calling compiles to."""
+    from agentfield_amd.engine.schemafsm import MultiFSM, make_fsm
+    schema = {"anyOf": [
+        {"type": "object",
+         "properties": {"name": {"enum": ["get_time"]},
+                        "arguments": {"type": "object",
+                                      "properties": {"tz": {"type": "string"}},
+                                      "required": ["tz"]}},
+         "required": ["name", "arguments"]},
+        {"type": "object",
+         "properties": {"name": {"enum": ["add"]},
+                        "arguments": {"type": "object",
+                                      "properties": {"a": {"type": "integer"},
+                                                     "b": {"type": "integer"}},
+                                      "required": ["a", "b"]}},
+         "required": ["name", "arguments"]},
+    ]}
+    f = make_fsm(schema)
+    assert isinstance(f, MultiFSM)
+    doc = b'{"name": "add", "arguments": {"a": 1, "b": -2}}'
+    for b in doc:
+        f.advance(b)
+    assert f.complete()
+    # pruning: after the discriminator only the matching branch survives
+    f2 = make_fsm(schema)
+    for b in b'{"name": "g':
+        f2.advance(b)
+    assert len(f2.alts) == 1
+    with pytest.raises(ValueError):
+        f2.advance(ord("x"))  # 'gx' matches no enum
+    # fuzz: random mask walks always produce a doc matching ONE branch
+    rng = random.Random(9)
+    for _ in range(40):
+        g = make_fsm(schema)
+        out = bytearray()
+        budget = 70
+        while budget > 0:
+            ids = g.allowed_token_ids(budget)
+            assert ids
+            t = rng.choice(ids)
+            if t == 2:
+                break
+            g.advance(t - 4)
+            out.append(t - 4)
+            budget -= 1
+        data = json.loads(bytes(out).decode("utf-8", errors="replace"))
+        assert data["name"] in ("get_time", "add")
+        if data["name"] == "add":
+            assert set(data["arguments"]) <= {"a", "b"}

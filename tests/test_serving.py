@@ -495,3 +495,61 @@ def test_openai_json_schema_response_format(replicas):
         "json_schema": schema}, timeout=60.0).json()
     data = _json.loads(r["text"].strip())
     assert isinstance(data["x"], int)
+
+
+def test_openai_tool_calling(replicas):
+    """tools + tool_choice=required: schema-constrained decoding makes
+    the tool call GUARANTEED well-formed — name from the tool set,
+    arguments matching the parameter schema."""
+    import json as _json
+    import httpx
+    url = replicas[0].base_url
+    tools = [
+        {"type": "function", "function": {
+            "name": "get_time",
+            "description": "current time",
+            "parameters": {"type": "object",
+                           "properties": {"tz": {"type": "string"}},
+                           "required": ["tz"]}}},
+        {"type": "function", "function": {
+            "name": "add",
+            "description": "add two ints",
+            "parameters": {"type": "object",
+                           "properties": {"a": {"type": "integer"},
+                                          "b": {"type": "integer"}},
+                           "required": ["a", "b"]}}},
+    ]
+    r = httpx.post(url + "/v1/chat/completions", json={
+        "model": "tiny", "messages": [{"role": "user", "content": "call"}],
+        "max_tokens": 48, "temperature": 0.9,
+        "tools": tools, "tool_choice": "required"}, timeout=60.0).json()
+    ch = r["choices"][0]
+    assert ch["finish_reason"] == "tool_calls"
+    tc = ch["message"]["tool_calls"][0]
+    assert tc["type"] == "function"
+    assert tc["function"]["name"] in ("get_time", "add")
+    args = _json.loads(tc["function"]["arguments"])
+    if tc["function"]["name"] == "add":
+        assert all(isinstance(v, int) for v in args.values())
+    else:
+        assert set(args) <= {"tz"}
+    # named tool_choice pins the function
+    r = httpx.post(url + "/v1/chat/completions", json={
+        "model": "tiny", "messages": [{"role": "user", "content": "call"}],
+        "max_tokens": 48, "temperature": 0.9, "tools": tools,
+        "tool_choice": {"type": "function",
+                        "function": {"name": "add"}}}, timeout=60.0).json()
+    tc = r["choices"][0]["message"]["tool_calls"][0]
+    assert tc["function"]["name"] == "add"
+    # unknown named tool -> 400
+    r = httpx.post(url + "/v1/chat/completions", json={
+        "model": "tiny", "messages": [{"role": "user", "content": "x"}],
+        "tools": tools,
+        "tool_choice": {"type": "function",
+                        "function": {"name": "nope"}}}, timeout=60.0)
+    assert r.status_code == 400
+    # tool_choice auto (default): plain text generation still works
+    r = httpx.post(url + "/v1/chat/completions", json={
+        "model": "tiny", "messages": [{"role": "user", "content": "hi"}],
+        "max_tokens": 8, "tools": tools}, timeout=60.0).json()
+    assert r["choices"][0]["message"]["content"] is not None
