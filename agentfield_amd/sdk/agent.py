@@ -86,6 +86,7 @@ class Agent(FastAPI):
         self.base_url = base_url  # resolved at serve() if None
         self.client = AgentFieldClient(self.agentfield_url)
         self.ai = AgentAI(ai_config)
+        self.ai._agent = self  # skills-as-tools loop (ai(tools=...))
         self.memory = MemoryInterface(self.client, node_id)
         self.rate_limiter = StatelessRateLimiter()
         self.result_cache = ResultCache()

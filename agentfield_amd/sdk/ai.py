@@ -143,6 +143,8 @@ class EngineRunner:
     def _compiled_schema(self, cfg: AIConfig):
         if cfg.json_schema is None:
             return None
+        if isinstance(cfg.json_schema, dict) and "anyOf" in cfg.json_schema:
+            return cfg.json_schema  # engine compiles root anyOf (make_fsm)
         from ..engine.schemafsm import SchemaSpec
         return SchemaSpec(cfg.json_schema)
 
@@ -336,10 +338,16 @@ class AgentAI:
     def __init__(self, default_config: AIConfig | None = None):
         self.config = default_config or AIConfig(
             model=os.environ.get("AGENTFIELD_AI_MODEL", "llama-3-8b"))
+        self._agent = None  # set by Agent: enables ai(tools=...)
 
     def __call__(self, *prompt_parts, system: str | None = None,
                  user: str | None = None, schema=None, stream: bool = False,
-                 **overrides):
+                 tools=None, max_tool_rounds: int = 4, **overrides):
+        if tools is not None:
+            body = (user if user is not None
+                    else "\n".join(str(p) for p in prompt_parts))
+            return self._tool_loop(body, system, tools, max_tool_rounds,
+                                   self.config.merged(**overrides))
         cfg = self.config.merged(**overrides)
         if schema is not None:
             # schema requests get SCHEMA-constrained decoding: the engine
@@ -386,6 +394,81 @@ class AgentAI:
             except (ValueError, TypeError):
                 return text  # schema validation failed; return raw text
         return text
+
+    # ------------------------------------------------- skills-as-tools
+    _FINAL = {"type": "object",
+              "properties": {"name": {"enum": ["final_answer"]},
+                             "arguments": {
+                                 "type": "object",
+                                 "properties": {"text": {"type": "string"}},
+                                 "required": ["text"]}},
+              "required": ["name", "arguments"]}
+
+    def _tool_loop(self, body: str, system, tools, max_rounds: int,
+                   cfg: "AIConfig") -> str:
+        """Agentic tool use over the agent's OWN @skill functions: every
+        round the model emits a schema-GUARANTEED tool call (anyOf over
+        the skill signatures + a built-in final_answer), the skill runs
+        locally, and its result joins the transcript.  The last round is
+        constrained to final_answer only, so the loop always terminates
+        with an answer.  (The reference routes tool use through external
+        providers' function calling; here the constrained decoder makes
+        malformed calls impossible.)"""
+        import asyncio as _aio
+
+        if self._agent is None:
+            raise RuntimeError("ai(tools=...) needs an Agent-bound ai")
+        skills = self._agent._skills
+        names = list(skills) if tools is True else list(tools)
+        missing = [n for n in names if n not in skills]
+        if missing:
+            raise KeyError(f"unknown skills for tools=: {missing}")
+
+        def call_schema(nm):
+            m = skills[nm]
+            return {"type": "object",
+                    "properties": {"name": {"enum": [nm]},
+                                   "arguments": m.input_schema()},
+                    "required": ["name", "arguments"]}
+
+        alts = [call_schema(n) for n in names] + [self._FINAL]
+        catalog = json.dumps(
+            [{"name": n, "parameters": skills[n].input_schema()}
+             for n in names] +
+            [{"name": "final_answer",
+              "parameters": {"text": "the answer"}}])
+        transcript: list[str] = []
+        for rnd in range(max_rounds):
+            last = rnd == max_rounds - 1
+            schema = self._FINAL if last else                 (alts[0] if len(alts) == 1 else {"anyOf": alts})
+            rcfg = cfg.merged(json_only=True, json_schema=schema)
+            parts = []
+            if system or cfg.system_prompt:
+                parts.append(f"<|system|>\n{system or cfg.system_prompt}")
+            parts.append("<|system|>\nTools available:\n" + catalog +
+                         "\nRespond with a single JSON tool call; use "
+                         "final_answer to finish.")
+            parts.append(f"<|user|>\n{body}")
+            parts.extend(transcript)
+            parts.append("<|assistant|>\n")
+            out = get_runner(rcfg).generate_text("\n".join(parts), rcfg)
+            try:
+                call = json.loads(out)
+            except ValueError:  # truncated by max_tokens
+                return out
+            if call.get("name") == "final_answer":
+                return str(call.get("arguments", {}).get("text", ""))
+            meta = skills.get(call.get("name"))
+            args = call.get("arguments") or {}
+            try:
+                result = (_aio.run(meta.fn(**args)) if meta.is_async
+                          else meta.fn(**args))
+            except Exception as e:  # the model sees the failure and adapts
+                result = {"error": str(e)}
+            transcript.append(
+                f"<|assistant|>\n{out}\n<|tool|>\n"
+                f"{json.dumps(result, default=str)}")
+        raise AssertionError("unreachable: last round forces final_answer")
 
     def with_multimodal(self, *parts, **kw):
         """ai() over mixed text/image/audio inputs.  Detection and message

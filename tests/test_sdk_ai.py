@@ -179,3 +179,57 @@ def test_ai_json_only_always_parses(tiny_runner):
     # plain string if the model emitted a JSON string) — never raises
     merged = AgentAI(AIConfig(model="tiny", max_tokens=8, timeout=120))
     merged("obj", schema={"type": "object"})
+
+
+def test_ai_tool_loop_with_skills():
+    """ai(tools=...) runs the agentic loop over the agent's @skill
+    functions: every call is schema-guaranteed (name from the skill set,
+    typed arguments), skills execute locally, and the built-in
+    final_answer always terminates the loop."""
+    import torch
+
+    from agentfield_amd.engine import LLMEngine
+    from agentfield_amd.models import CONFIGS
+    from agentfield_amd.sdk import Agent
+    from agentfield_amd.sdk.ai import ByteTokenizer, EngineRunner, set_runner
+
+    eng = LLMEngine(CONFIGS["tiny"], device="cpu", dtype=torch.float32,
+                    page_size=4, num_pages=256, max_num_seqs=4,
+                    enable_graphs=False, seed=6)
+    runner = EngineRunner(eng, ByteTokenizer(CONFIGS["tiny"].vocab_size))
+    set_runner("tiny", runner)
+    try:
+        app = Agent("tooluser", auto_register=False)
+        calls = []
+
+        @app.skill()
+        def add(a: int = 0, b: int = 0):
+            calls.append(("add", a, b))
+            return {"sum": a + b}
+
+        @app.skill()
+        def shout(text: str = ""):
+            calls.append(("shout", text))
+            return {"text": text.upper()}
+
+        out = app.ai("use a tool", tools=True, max_tool_rounds=3,
+                     model="tiny", max_tokens=48, temperature=0.9)
+        assert isinstance(out, str)
+        # a random-init model still produced only WELL-FORMED calls:
+        # every recorded invocation got typed arguments
+        for c in calls:
+            if c[0] == "add":
+                assert isinstance(c[1], int) and isinstance(c[2], int)
+            else:
+                assert isinstance(c[1], str)
+        # subset + unknown-name validation
+        out2 = app.ai("x", tools=["add"], max_tool_rounds=2, model="tiny",
+                      max_tokens=40, temperature=0.9)
+        assert isinstance(out2, str)
+        try:
+            app.ai("x", tools=["nope"], model="tiny")
+            raise AssertionError("unknown skill must raise")
+        except KeyError:
+            pass
+    finally:
+        runner.shutdown()
