@@ -165,3 +165,57 @@ def test_stream_events_exactly_once_under_preemption():
     for rid, f in fins.items():
         assert counts[rid] == len(f.output_ids) == 10, \
             f"seq {rid}: {counts[rid]} events for {len(f.output_ids)} tokens"
+
+
+def test_engine_fuzz_with_draft_model():
+    """Preemption churn with the DRAFT-MODEL speculator: the draft KV
+    (which reuses the sequences' page indices) must resync through
+    release/re-admission, and every output stays greedy-exact."""
+    rng = random.Random(11)
+    eng = LLMEngine(CONFIGS["tiny"], device="cpu", dtype=torch.float32,
+                    page_size=4, num_pages=16, max_num_seqs=3,
+                    max_prefill_tokens=16, enable_graphs=False,
+                    max_waiting=64, spec_draft=CONFIGS["tiny"],
+                    spec_draft_k=3)
+    total_pages = eng.sched.alloc.num_pages - 1
+    live, finished, submitted = {}, {}, 0
+    for step in range(400):
+        if rng.random() < 0.3 and submitted < 30:
+            prompt = [rng.randrange(500) for _ in range(rng.randint(4, 16))]
+            mt = rng.randint(8, 20)
+            rid = eng.add_request(prompt, SamplingParams(max_tokens=mt,
+                                                         ignore_eos=True))
+            if rid is not None:
+                live[rid] = True
+                submitted += 1
+        eng.step()
+        for rid in list(live):
+            fin = eng.get_finished(rid)
+            if fin is not None:
+                finished[rid] = fin
+                del live[rid]
+    for _ in range(3000):
+        if not eng.has_work():
+            break
+        eng.step()
+        for rid in list(live):
+            fin = eng.get_finished(rid)
+            if fin is not None:
+                finished[rid] = fin
+                del live[rid]
+    assert not live
+    assert eng.sched.alloc.num_free == total_pages, "KV pages leaked"
+    assert eng.metrics["spec_steps"] > 0
+    assert eng.sched.n_preempted > 0, "churn config should preempt"
+    ref_eng = LLMEngine(CONFIGS["tiny"], device="cpu", dtype=torch.float32,
+                        page_size=4, num_pages=128, max_num_seqs=4,
+                        max_prefill_tokens=64, enable_graphs=False)
+    checked = 0
+    for rid, fin in finished.items():
+        assert len(fin.output_ids) == fin.sampling.max_tokens
+        if checked < 8:
+            want = ref_eng.generate([fin.prompt_ids], fin.sampling)[0]
+            assert fin.output_ids == want, \
+                f"seq {rid} diverged under draft-spec + churn"
+            checked += 1
+    assert checked > 0
