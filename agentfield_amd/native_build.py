@@ -52,5 +52,31 @@ def build_asan(force: bool = False, verbose: bool = True) -> Path:
     return ASAN_OUT
 
 
+TSAN_OUT = ASAN_OUT.with_name(ASAN_OUT.name.replace("asan", "tsan")) \
+    if "asan" in ASAN_OUT.name else ASAN_OUT.with_suffix(".tsan.so")
+
+
+def build_tsan(force: bool = False, verbose: bool = True) -> Path:
+    """ThreadSanitizer build of the native extension.  The engine calls
+    the scheduler from one driver thread, but the Ed25519/AES-GCM
+    helpers run under FastAPI worker THREADS concurrently — that is the
+    surface TSAN guards (see tests/test_sanitizer.py)."""
+    if not force and TSAN_OUT.exists() and \
+            TSAN_OUT.stat().st_mtime > SRC.stat().st_mtime:
+        return TSAN_OUT
+    import pybind11
+    cmd = [
+        "g++", "-O1", "-g", "-fsanitize=thread", "-fno-omit-frame-pointer",
+        "-shared", "-fPIC", "-std=c++17",
+        f"-I{pybind11.get_include()}",
+        f"-I{sysconfig.get_paths()['include']}",
+        str(SRC), "-lcrypto", "-o", str(TSAN_OUT),
+    ]
+    if verbose:
+        print("[native-tsan]", " ".join(cmd), file=sys.stderr)
+    subprocess.run(cmd, check=True)
+    return TSAN_OUT
+
+
 if __name__ == "__main__":
     build(force="--force" in sys.argv)

@@ -166,3 +166,61 @@ def test_controlplane_threaded_soak():
     assert not remaining, f"stuck async executions: {list(remaining)[:5]}"
     a_srv.stop()
     srv.stop()
+
+
+TSAN_DRIVER = r"""
+import sys, threading
+sys.path.insert(0, {root!r})
+import importlib.util
+spec = importlib.util.spec_from_file_location("agentfield_amd._native",
+                                              {so!r})
+mod = importlib.util.module_from_spec(spec)
+spec.loader.exec_module(mod)
+
+import os
+seed = os.urandom(32)
+pub = mod.ed25519_pubkey(seed)
+key = os.urandom(32)
+
+def worker(i):
+    # the crypto helpers run under FastAPI worker threads in production;
+    # hammer them concurrently (GIL releases inside OpenSSL calls)
+    for j in range(300):
+        msg = (f"m{{i}}-{{j}}").encode() * 8
+        sig = mod.ed25519_sign(seed, msg)
+        assert mod.ed25519_verify(pub, msg, sig)
+        ct = mod.aes_gcm_encrypt(key, msg)
+        assert mod.aes_gcm_decrypt(key, ct) == msg
+
+threads = [threading.Thread(target=worker, args=(i,)) for i in range(8)]
+for t in threads:
+    t.start()
+for t in threads:
+    t.join()
+print("TSAN_DRIVER_OK")
+"""
+
+
+def test_native_crypto_under_tsan(tmp_path):
+    """ThreadSanitizer over the Ed25519/AES-GCM helpers hammered from 8
+    threads.  CPython itself is not TSAN-annotated, so only reports
+    whose stacks hit OUR extension fail the test."""
+    libtsan = subprocess.run(["gcc", "-print-file-name=libtsan.so"],
+                             capture_output=True, text=True).stdout.strip()
+    if not libtsan or not Path(libtsan).exists():
+        pytest.skip("libtsan not available")
+    from agentfield_amd.native_build import build_tsan
+    so = build_tsan(verbose=False)
+    script = tmp_path / "drv.py"
+    script.write_text(TSAN_DRIVER.format(root=str(ROOT), so=str(so)))
+    log = tmp_path / "tsan.log"
+    r = subprocess.run(
+        [sys.executable, str(script)],
+        env={**os.environ, "LD_PRELOAD": libtsan,
+             "TSAN_OPTIONS": f"log_path={log} exitcode=0 halt_on_error=0"},
+        capture_output=True, text=True, timeout=300)
+    assert "TSAN_DRIVER_OK" in r.stdout, r.stderr[-2000:]
+    reports = "".join(p.read_text() for p in tmp_path.glob("tsan.log.*"))
+    ours = [blk for blk in reports.split("==================")
+            if "_native_tsan" in blk and "data race" in blk.lower()]
+    assert not ours, ours[0][:2000]
