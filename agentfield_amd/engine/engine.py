@@ -77,6 +77,11 @@ class LLMEngine:
         # free; seq.draft_len tracks how far the draft's KV is valid.
         self.draft = None
         self.spec_draft_k = int(spec_draft_k)
+        # rejection-sampled speculation (temperature > 0) draws from its
+        # own stream: deterministic per seed, distinct from the sampler's
+        self._spec_gen = torch.Generator(
+            device=self.device if self.is_gpu else "cpu")
+        self._spec_gen.manual_seed((seed or 0x5EED) ^ 0x5A5A5A)
         if spec_draft is not None or draft_model is not None:
             from ..models import CONFIGS as _CFGS
             if draft_model is not None:
@@ -144,6 +149,11 @@ class LLMEngine:
         # for free; seq.draft_len tracks how far the draft's KV is valid.
         self.draft = None
         self.spec_draft_k = int(spec_draft_k)
+        # rejection-sampled speculation (temperature > 0) draws from its
+        # own stream: deterministic per seed, distinct from the sampler's
+        self._spec_gen = torch.Generator(
+            device=self.device if self.is_gpu else "cpu")
+        self._spec_gen.manual_seed((seed or 0x5EED) ^ 0x5A5A5A)
         if spec_draft is not None or draft_model is not None:
             from ..models import CONFIGS as _CFGS
             if draft_model is not None:
@@ -251,18 +261,21 @@ class LLMEngine:
                 self.metrics["steps"] += 1
                 if (self.spec_lookup > 0 or self.draft is not None) and \
                         self.tp_group is None and \
-                        all(s.sampling.temperature == 0.0 and
-                            s.sampling.top_k <= 0 and s.sampling.top_p >= 1.0
+                        all(s.sampling.top_k <= 0 and
+                            s.sampling.top_p >= 1.0
                             and not s.sampling.json_mode
+                            and (s.sampling.temperature == 0.0
+                                 or s.sampling.logprobs == 0)
                             for s in batch.seqs):
                     if self.draft is not None:
-                        drafts = self._draft_model_propose(batch.seqs)
+                        drafts, qs = self._draft_model_propose(batch.seqs)
                     else:
                         drafts = {s.seq_id: self._draft_for(s)
                                   for s in batch.seqs}
+                        qs = {}
                     if any(drafts.values()):
                         self.metrics["spec_steps"] += 1
-                        return self._step_verify(batch, drafts)
+                        return self._step_verify(batch, drafts, qs)
                 self.metrics["decode_steps"] += 1
                 tokens = self._step_decode(batch)
                 return self._bookkeep(batch.seqs, tokens)
@@ -477,17 +490,21 @@ class LLMEngine:
         return max(0, min(k, remaining - 1, capacity))
 
     @torch.no_grad()
-    def _draft_model_propose(self, seqs: list[Sequence]) -> dict:
+    def _draft_model_propose(self, seqs: list[Sequence]) -> tuple:
         """Autoregressive k-token proposals from the draft model.  First a
         varlen catch-up chunk brings each sequence's draft KV up to its
         current length (tokens the target emitted without the draft —
         bonus tokens, re-admissions after preemption), then k batched
-        decode rounds propose greedily."""
+        decode rounds propose — argmax for greedy sequences, a sample
+        from the draft's temperature-scaled distribution otherwise (the
+        q the rejection-sampling verify divides by).  Returns
+        (drafts, q_rows): q_rows[seq_id] is a [k, V] tensor of draft
+        probabilities for sampled sequences, None entries for greedy."""
         dev = self.device
         ks = {s.seq_id: self._spec_cap(s, self.spec_draft_k) for s in seqs}
         live = [s for s in seqs if ks[s.seq_id] > 0]
         if not live:
-            return {s.seq_id: [] for s in seqs}
+            return {s.seq_id: [] for s in seqs}, {}
         # ---- catch-up: feed ctx[draft_len : n-1] (KV only) ----
         cu_ids, cu_pos, cu_slots, cu_qs, cu_lens, cu_bt = [], [], [], [], [], []
         for s in live:
@@ -531,6 +548,10 @@ class LLMEngine:
         base = torch.tensor([s.num_tokens for s in live], dtype=torch.int32,
                             device=dev)
         out: dict = {s.seq_id: [] for s in seqs}
+        temps = [s.sampling.temperature for s in live]
+        any_sampled = any(t > 0.0 for t in temps)
+        q_rows: dict = {s.seq_id: [] for s in live
+                        if s.sampling.temperature > 0.0}
         kmax = max(ks[s.seq_id] for s in live)
         for j in range(kmax):
             pos = base - 1 + j
@@ -546,12 +567,27 @@ class LLMEngine:
                               nsplit=choose_nsplit(B,
                                                    self.draft_cfg.num_kv_heads))
             logits = self.draft(cur, pos, self.draft_kv, md)
-            cur = logits.argmax(dim=-1).to(torch.int32)
+            nxt = logits.argmax(dim=-1).to(torch.int32)
+            if any_sampled:
+                nl = [int(x) for x in nxt.cpu()]
+                for i, s in enumerate(live):
+                    t = temps[i]
+                    if t <= 0.0:
+                        continue
+                    q = torch.softmax(logits[i].float() / t, dim=-1)
+                    tok = int(torch.multinomial(q, 1,
+                                                generator=self._spec_gen))
+                    nl[i] = tok
+                    if j < ks[s.seq_id]:
+                        q_rows[s.seq_id].append(q)
+                nxt = torch.tensor(nl, dtype=torch.int32, device=dev)
+            cur = nxt
             toks = cur.cpu().tolist()
             for i, s in enumerate(live):
                 if j < ks[s.seq_id]:
                     out[s.seq_id].append(int(toks[i]))
-        return out
+        qs = {sid: torch.stack(rows) for sid, rows in q_rows.items() if rows}
+        return out, qs
 
     # -- speculative decode (prompt lookup, greedy-exact) -------------------
     def _draft_for(self, seq: Sequence) -> list[int]:
@@ -581,13 +617,18 @@ class LLMEngine:
                         return d
         return []
 
-    def _step_verify(self, batch: ScheduleBatch, drafts: dict):
+    def _step_verify(self, batch: ScheduleBatch, drafts: dict,
+                     qs: dict | None = None):
         """One chunked-prefill forward verifies each sequence's draft: chunk
-        = [last_token] + draft at positions n-1..n-1+k.  Row j's argmax is
-        the model's true greedy token after consuming draft[:j]; the longest
-        matching prefix is accepted plus one bonus token.  Rejected draft
-        positions leave garbage KV beyond the sequence length, which is
-        overwritten before it can ever be read (attention is length-bounded)."""
+        = [last_token] + draft at positions n-1..n-1+k.  Greedy sequences:
+        row j's argmax is the model's true greedy token after consuming
+        draft[:j]; the longest matching prefix is accepted plus one bonus
+        token.  Sampled sequences (temperature > 0) go through rejection
+        sampling against the draft's q (spec_sampling.accept_resample),
+        which keeps the emitted distribution EXACTLY the target's.
+        Rejected draft positions leave garbage KV beyond the sequence
+        length, which is overwritten before it can ever be read
+        (attention is length-bounded)."""
         dev = self.device
         seqs = batch.seqs
         ids, pos, slots, q_start, bt_rows, chunks = [], [], [], [], [], []
@@ -624,6 +665,20 @@ class LLMEngine:
         for i, seq in enumerate(seqs):
             t = greedy[cu_list[i]:cu_list[i + 1]]
             d = drafts.get(seq.seq_id) or []
+            temp = seq.sampling.temperature
+            if temp > 0.0 and d:
+                from .spec_sampling import accept_resample
+                rows = logits[cu_list[i]:cu_list[i + 1]]
+                p_rows = torch.softmax(rows.float() / temp, dim=-1)
+                toks = accept_resample(p_rows, d,
+                                       (qs or {}).get(seq.seq_id),
+                                       self._spec_gen)
+                tok_lists.append(toks)
+                if self.draft is not None:
+                    seq.draft_len = seq.num_tokens + len(toks) - 1
+                self.metrics["spec_accepted"] += len(toks) - 1
+                self.metrics["decode_tokens"] += len(toks)
+                continue
             a = 0
             while a < len(d) and t[a] == d[a]:
                 a += 1
