@@ -483,3 +483,38 @@ def test_spec_draft_model_on_gpu():
     got = run(spec)
     assert got == base
     assert spec.metrics["spec_steps"] > 0
+
+
+@pytest.mark.gpu
+def test_sampled_speculation_on_gpu():
+    """Rejection-sampled speculation on the HIP kernels: sampled
+    sequences complete at full length while a greedy neighbor stays
+    exactly equal to the non-speculative engine."""
+    cfg = CONFIGS["tiny"]
+
+    def mk(**kw):
+        return LLMEngine(cfg, device="cuda", page_size=4, num_pages=256,
+                         max_num_seqs=4, enable_graphs=True, seed=7, **kw)
+
+    base = mk()
+    b0 = base.generate([list(range(1, 30))],
+                       SamplingParams(max_tokens=16, ignore_eos=True))[0]
+    eng = mk(spec_draft=cfg, spec_draft_k=3)
+    rg = eng.add_request(list(range(1, 30)),
+                         SamplingParams(max_tokens=16, ignore_eos=True))
+    rs = eng.add_request([7, 9, 2, 44] * 5,
+                         SamplingParams(max_tokens=16, temperature=0.9,
+                                        ignore_eos=True))
+    outs = {}
+    for _ in range(400):
+        eng.step()
+        for r in (rg, rs):
+            if r not in outs:
+                f = eng.get_finished(r)
+                if f:
+                    outs[r] = f.output_ids
+        if len(outs) == 2:
+            break
+    assert outs[rg] == b0
+    assert len(outs[rs]) == 16
+    assert eng.metrics["spec_steps"] > 0
