@@ -421,6 +421,39 @@ def create_engine_app(runner: EngineRunner, model_name: str,
 
         async def gen():
             n_out = 0
+            if kind == "chat" and tool_mode:
+                # tool-call streaming: the call JSON is buffered (its
+                # name field only exists once generated) and emitted as
+                # one OpenAI-shaped tool_calls delta
+                import uuid as _uuid
+                buf = []
+                while True:
+                    tok, done = await anyio.to_thread.run_sync(sq.get)
+                    if tok is not None:
+                        buf.append(tok)
+                    if done:
+                        break
+                text = runner.tokenizer.decode(buf)
+                try:
+                    call = json.loads(text.strip())
+                    delta = {"role": "assistant", "tool_calls": [{
+                        "index": 0,
+                        "id": "call_" + _uuid.uuid4().hex[:24],
+                        "type": "function",
+                        "function": {
+                            "name": call.get("name"),
+                            "arguments": json.dumps(
+                                call.get("arguments", {}))}}]}
+                    fr = "tool_calls"
+                except Exception:  # truncated mid-call
+                    delta = {"role": "assistant", "content": text}
+                    fr = "length"
+                c = {"index": 0, "delta": delta, "finish_reason": None}
+                yield "data: " + json.dumps({**base, "object": obj,
+                                             "choices": [c]}) + "\n\n"
+                yield chunk(None, fr)
+                yield "data: [DONE]\n\n"
+                return
             if kind == "chat":
                 yield chunk(None, None)  # leading role delta
             hold = max((len(s) for s in stops), default=1) - 1

@@ -553,3 +553,30 @@ def test_openai_tool_calling(replicas):
         "model": "tiny", "messages": [{"role": "user", "content": "hi"}],
         "max_tokens": 8, "tools": tools}, timeout=60.0).json()
     assert r["choices"][0]["message"]["content"] is not None
+
+
+def test_openai_tool_calling_streaming(replicas):
+    """stream=true with tool_choice=required: one OpenAI-shaped
+    tool_calls delta, finish_reason tool_calls, then [DONE]."""
+    import json as _json
+    import httpx
+    url = replicas[0].base_url
+    tools = [{"type": "function", "function": {
+        "name": "ping",
+        "parameters": {"type": "object",
+                       "properties": {"n": {"type": "integer"}},
+                       "required": ["n"]}}}]
+    deltas = []
+    with httpx.stream("POST", url + "/v1/chat/completions", json={
+            "model": "tiny", "messages": [{"role": "user", "content": "go"}],
+            "max_tokens": 56, "temperature": 0.9, "stream": True,
+            "tools": tools, "tool_choice": "required"},
+            timeout=60.0) as r:
+        for line in r.iter_lines():
+            if line.startswith("data:") and "[DONE]" not in line:
+                deltas.append(_json.loads(line[5:]))
+    tc = deltas[0]["choices"][0]["delta"]["tool_calls"][0]
+    assert tc["function"]["name"] == "ping"
+    args = _json.loads(tc["function"]["arguments"])
+    assert set(args) <= {"n"}
+    assert deltas[-1]["choices"][0]["finish_reason"] == "tool_calls"

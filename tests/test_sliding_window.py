@@ -201,3 +201,50 @@ def test_engine_rolling_outputs_identical(monkeypatch):
     out_flat, free_flat = run("0")
     assert out_roll == out_flat
     assert len(out_roll) == 48
+
+
+def test_rolling_window_respects_prefix_sharing():
+    """A windowed sequence whose early pages are SHARED via the prefix
+    cache: rolling reclamation unrefs them, but the cache (and any other
+    sequence reading them) keeps the pages alive until eviction."""
+    from agentfield_amd.engine.prefix_cache import PrefixCachingScheduler
+    from agentfield_amd.engine.scheduler import SchedulerConfig
+    from agentfield_amd.engine.sequence import Sequence
+
+    cfg = SchedulerConfig(max_num_seqs=4, max_prefill_tokens=4096,
+                          page_size=4, num_pages=64, window_tokens=16)
+    s = PrefixCachingScheduler(cfg)
+    sp = SamplingParams(max_tokens=64, ignore_eos=True)
+    a = Sequence(seq_id=1, prompt_ids=list(range(1, 33)), sampling=sp)
+    s.add(a)
+    s.schedule()
+    a.num_prefilled = 32
+    s.note_token(a)          # publish prompt pages into the cache
+    a.output_ids.append(5)
+    cached_pages = list(a.pages[:8])
+    # duplicate prompt admits THROUGH the shared pages
+    b = Sequence(seq_id=2, prompt_ids=list(range(1, 33)), sampling=sp)
+    s.add(b)
+    r = s.schedule()
+    assert r.is_prefill and b.cached_prefix == 28
+    b.num_prefilled = 32
+    s.note_token(b)
+    b.output_ids.append(5)
+    # decode a far past the window: its early pages roll
+    for _ in range(60):
+        a.output_ids.append(7)
+        b.output_ids.append(7)
+        s.schedule()
+    assert a.freed_pages > 0 and b.freed_pages > 0
+    # the shared pages are still referenced by the CACHE: none returned
+    # to the free list while the cache holds them
+    for p in cached_pages:
+        assert p in s.alloc.refs, p
+        assert p not in s.alloc.free_list
+    # both sequences finish; cache still owns the shared pages
+    s.finish(a)
+    s.finish(b)
+    for p in cached_pages[:7]:  # 7 full prefix pages were published
+        assert p in s.alloc.refs
+    # total conservation: free + cache-held == all pages
+    assert s.alloc.num_free + len(s.alloc.refs) == 63
