@@ -481,6 +481,14 @@ def create_engine_app(runner: EngineRunner, model_name: str,
                 if done:
                     yield chunk(None, _finish_reason(n_out, max_tokens,
                                                      stopped))
+                    if (body.get("stream_options") or {}).get(
+                            "include_usage"):
+                        yield "data: " + json.dumps({
+                            **base, "object": obj, "choices": [],
+                            "usage": {"prompt_tokens": len(ids),
+                                      "completion_tokens": n_out,
+                                      "total_tokens": len(ids) + n_out},
+                        }) + "\n\n"
                     yield "data: [DONE]\n\n"
                     return
         return StreamingResponse(gen(), media_type="text/event-stream")

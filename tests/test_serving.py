@@ -580,3 +580,20 @@ def test_openai_tool_calling_streaming(replicas):
     args = _json.loads(tc["function"]["arguments"])
     assert set(args) <= {"n"}
     assert deltas[-1]["choices"][0]["finish_reason"] == "tool_calls"
+
+
+def test_stream_include_usage(replicas):
+    import json as _json
+    import httpx
+    url = replicas[0].base_url
+    events = []
+    with httpx.stream("POST", url + "/v1/chat/completions", json={
+            "model": "tiny", "messages": [{"role": "user", "content": "x"}],
+            "max_tokens": 4, "stream": True,
+            "stream_options": {"include_usage": True}}, timeout=60.0) as r:
+        for line in r.iter_lines():
+            if line.startswith("data:") and "[DONE]" not in line:
+                events.append(_json.loads(line[5:]))
+    usage = events[-1].get("usage")
+    assert usage and usage["completion_tokens"] == 4
+    assert events[-1]["choices"] == []

@@ -176,3 +176,36 @@ def test_runner_attaches_grammar_for_hf_tokenizer():
     json.loads(out)
     assert eng.token_grammar is not None
     runner.shutdown()
+
+
+def test_token_grammar_with_root_anyof():
+    """MultiFSM (root anyOf) through the BPE token-trie walk: every walk
+    yields a document matching one alternative."""
+    import random as _random
+
+    from agentfield_amd.engine.schemafsm import make_fsm
+    tok = FakeBPE()
+    g = TokenJsonGrammar(tok.vocab, eos_id=2)
+    schema = {"anyOf": [
+        {"type": "object", "properties": {"a": {"type": "integer"}},
+         "required": ["a"]},
+        {"type": "object", "properties": {"s": {"enum": ["x", "yy"]}},
+         "required": ["s"]},
+    ]}
+    rng = _random.Random(13)
+    for trial in range(30):
+        f = make_fsm(schema)
+        out = []
+        budget = 24
+        while budget > 0:
+            ids = g.allowed_token_ids(f, budget)
+            assert ids, (trial, out)
+            t = rng.choice(ids)
+            if t == 2:
+                break
+            g.advance_token(f, t)
+            out.append(t)
+            budget -= 1
+        data = json.loads(tok.decode(out))
+        assert ("a" in data and isinstance(data["a"], int)) or \
+               ("s" in data and data["s"] in ("x", "yy"))
