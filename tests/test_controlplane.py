@@ -869,3 +869,24 @@ def test_webhook_burst_poller_race():
 
     wait_until(all_marked, timeout=10.0)  # status write can lag the hook
     srv.stop()
+
+
+def test_webhook_stale_inflight_reclaimed():
+    """A claim whose worker died (claimed inflight, never delivered)
+    becomes due again after the stale window and CAN be re-claimed —
+    crash-safe delivery."""
+    from agentfield_amd.controlplane.storage import Storage
+    st = Storage(":memory:")
+    st.register_webhook("e1", "http://x/hook", "sec", {})
+    st.stage_webhook_payload("e1", {"event": "execution.completed"})
+    # claim with a tiny stale window, simulate worker death (no attempt)
+    assert st.try_mark_webhook_inflight("e1", stale_s=0.05)
+    assert not st.try_mark_webhook_inflight("e1", stale_s=0.05)  # held
+    assert st.due_webhooks() == []          # not yet stale
+    time.sleep(0.1)
+    due = st.due_webhooks()
+    assert [d["execution_id"] for d in due] == ["e1"]  # stale -> due
+    assert st.try_mark_webhook_inflight("e1")          # re-claimable
+    st.webhook_attempted("e1", True, 200, None, 0.0, 5)
+    assert st.get_webhook("e1")["status"] == "delivered"
+    st.close()
