@@ -96,3 +96,58 @@ def test_engine_sampled_speculation_end_to_end():
     assert all(0 <= t < CONFIGS["tiny"].vocab_size for t in outs[r_samp])
     assert eng.metrics["spec_steps"] > 0
     assert eng.metrics["spec_accepted"] > 0
+
+
+def test_spec_gate_mixed_constraints():
+    """json_mode or top-k/p sequences in the batch disable the spec path
+    for that step (guard), but everything still completes correctly."""
+    eng = LLMEngine(CONFIGS["tiny"], device="cpu", dtype=torch.float32,
+                    page_size=4, num_pages=256, max_num_seqs=4,
+                    enable_graphs=False, seed=4,
+                    spec_draft=CONFIGS["tiny"], spec_draft_k=3)
+    import json as _json
+    rids = [
+        eng.add_request(list(range(1, 20)),
+                        SamplingParams(max_tokens=12, ignore_eos=True)),
+        eng.add_request(list(range(1, 20)),
+                        SamplingParams(max_tokens=12, temperature=0.8,
+                                       json_mode=True)),
+        eng.add_request(list(range(30, 50)),
+                        SamplingParams(max_tokens=12, temperature=0.8,
+                                       top_k=4, ignore_eos=True)),
+    ]
+    outs = {}
+    for _ in range(400):
+        eng.step()
+        for r in rids:
+            if r not in outs:
+                f = eng.get_finished(r)
+                if f:
+                    outs[r] = f
+        if len(outs) == len(rids):
+            break
+    assert len(outs) == len(rids)
+    body = outs[rids[1]].output_ids
+    if body and body[-1] == 2:
+        body = body[:-1]
+    _json.loads(bytes(b - 4 for b in body).decode("utf-8",
+                                                  errors="replace"))
+    assert len(outs[rids[2]].output_ids) == 12
+
+
+def test_prompt_lookup_with_sampled_sequences():
+    """Prompt-lookup drafts (point-mass q) now serve sampled sequences
+    too: repetitive prompts fire the speculator at temperature > 0."""
+    eng = LLMEngine(CONFIGS["tiny"], device="cpu", dtype=torch.float32,
+                    page_size=4, num_pages=256, max_num_seqs=4,
+                    enable_graphs=False, seed=5, spec_lookup=4)
+    rid = eng.add_request([7, 3, 9, 1] * 6,
+                          SamplingParams(max_tokens=20, temperature=0.7,
+                                         ignore_eos=True))
+    for _ in range(400):
+        eng.step()
+        f = eng.get_finished(rid)
+        if f:
+            break
+    assert f and len(f.output_ids) == 20
+    assert eng.metrics["spec_steps"] > 0
