@@ -418,6 +418,60 @@ def start(name: str, agentfield_url: str = DEFAULT_URL, port: int = None):
 
 
 @app.command()
+def doctor():
+    """Environment diagnostics: ROCm toolchain, GPU visibility, native
+    extensions, engine smoke.  Run this first when anything misbehaves."""
+    import shutil
+    import subprocess as sp
+
+    def row(label, ok, detail=""):
+        mark = "ok " if ok else "MISSING"
+        typer.echo(f"  [{mark:7s}] {label}{': ' + detail if detail else ''}")
+
+    typer.echo("toolchain:")
+    hipcc = shutil.which("hipcc")
+    row("hipcc", bool(hipcc), hipcc or "")
+    for tool in ("rocm-smi", "rocprofv3", "g++", "cmake"):
+        row(tool, bool(shutil.which(tool)))
+    typer.echo("python:")
+    import torch
+    row("torch", True, torch.__version__)
+    cuda = torch.cuda.is_available()
+    row("GPU visible", cuda,
+        torch.cuda.get_device_name(0) if cuda else "CPU-only mode")
+    if cuda:
+        arch = torch.cuda.get_device_properties(0).gcnArchName
+        row("gfx950", "gfx950" in arch, arch)
+    typer.echo("extensions:")
+    from pathlib import Path as _P
+    pkg = _P(__file__).resolve().parent
+    so = pkg / "libafops.so"
+    row("libafops.so (HIP kernels)", so.exists(),
+        f"{so.stat().st_size // 1024} KiB" if so.exists() else
+        "run python agentfield_amd/build.py")
+    try:
+        import agentfield_amd._native  # noqa: F401
+        row("_native (C++ scheduler/crypto)", True)
+    except ImportError as e:
+        row("_native (C++ scheduler/crypto)", False, str(e)[:60])
+    typer.echo("engine:")
+    try:
+        import torch as _t
+        from .engine import LLMEngine, SamplingParams
+        from .models import CONFIGS
+        dev = "cuda" if cuda else "cpu"
+        kw = {} if cuda else {"dtype": _t.float32, "num_pages": 64,
+                              "page_size": 4, "enable_graphs": False}
+        eng = LLMEngine(CONFIGS["tiny"], device=dev, max_num_seqs=2,
+                        seed=0, **kw)
+        out = eng.generate([[1, 2, 3]],
+                           SamplingParams(max_tokens=4, ignore_eos=True))[0]
+        row("tiny-model decode", len(out) == 4, f"{dev}: {out}")
+    except Exception as e:
+        row("tiny-model decode", False, str(e)[:80])
+
+
+@app.command()
 def stop(name: str):
     """Stop a managed agent process."""
     if _procs().stop(name):

@@ -597,3 +597,13 @@ def test_stream_include_usage(replicas):
     usage = events[-1].get("usage")
     assert usage and usage["completion_tokens"] == 4
     assert events[-1]["choices"] == []
+
+
+def test_cli_doctor(capsys):
+    from typer.testing import CliRunner
+
+    from agentfield_amd.cli import app as cli_app
+    r = CliRunner().invoke(cli_app, ["doctor"])
+    assert r.exit_code == 0, r.output
+    assert "hipcc" in r.output and "tiny-model decode" in r.output
+    assert "MISSING] libafops" not in r.output
