@@ -607,3 +607,29 @@ def test_cli_doctor(capsys):
     assert r.exit_code == 0, r.output
     assert "hipcc" in r.output and "tiny-model decode" in r.output
     assert "MISSING] libafops" not in r.output
+
+
+def test_ai_tool_loop_via_remote_engines(replicas, monkeypatch):
+    """ai(tools=...) against a REMOTE replica fleet: the schema (incl.
+    root anyOf) rides /v1/generate, so the guarantee holds end-to-end
+    over HTTP exactly as in-process."""
+    from agentfield_amd.sdk import Agent
+    from agentfield_amd.sdk import ai as ai_mod
+    monkeypatch.setenv("AGENTFIELD_ENGINE_URLS",
+                       ",".join(s.base_url for s in replicas))
+    ai_mod._runners.pop("tiny-rfleet", None)
+    app = Agent("remotetool", auto_register=False,
+                ai_config=AIConfig(model="tiny-rfleet", max_tokens=48,
+                                   temperature=0.9, timeout=60))
+    calls = []
+
+    @app.skill()
+    def note(text: str = ""):
+        calls.append(text)
+        return {"ok": True}
+
+    out = app.ai("use the tool", tools=True, max_tool_rounds=2)
+    assert isinstance(out, str)
+    for c in calls:  # every call that happened was schema-well-formed
+        assert isinstance(c, str)
+    ai_mod._runners.pop("tiny-rfleet", None)
