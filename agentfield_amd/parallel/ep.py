@@ -85,6 +85,8 @@ def build_ep_model(cfg, device="cpu", dtype=torch.float32,
     from ..models.llama import LlamaForCausalLM, MoEMLP
     model = LlamaForCausalLM(cfg, device=device,
                              dtype=dtype).init_random(base_seed)
+    if not dist.is_initialized() or dist.get_world_size(group) == 1:
+        return model  # single rank: the plain MoE IS the EP=1 layout
     for layer in model.layers:
         if isinstance(layer.mlp, MoEMLP):
             layer.mlp = EPMoE.shard_from(layer.mlp, group=group).to(

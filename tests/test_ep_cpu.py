@@ -136,3 +136,16 @@ def _ep_model_case(rank, world):
 
 def test_ep_full_model_matches():
     assert _spawn("_ep_model_case")
+
+
+def test_build_ep_model_single_rank_fallback():
+    """Without a process group (or world 1) the plain MoE model comes
+    back unchanged — EP=1 is the identity layout."""
+    from agentfield_amd.models.llama import LlamaForCausalLM, MoEMLP
+    from agentfield_amd.parallel.ep import build_ep_model
+    m = build_ep_model(CFG, device="cpu", dtype=torch.float32, base_seed=7)
+    assert isinstance(m.layers[0].mlp, MoEMLP)
+    ref = LlamaForCausalLM(CFG, device="cpu",
+                           dtype=torch.float32).init_random(7)
+    x = torch.randn(5, CFG.hidden_size)
+    assert torch.allclose(m.layers[0].mlp(x), ref.layers[0].mlp(x))
