@@ -666,10 +666,14 @@ class LLMEngine:
             t = greedy[cu_list[i]:cu_list[i + 1]]
             d = drafts.get(seq.seq_id) or []
             temp = seq.sampling.temperature
-            if temp > 0.0 and d:
+            if temp > 0.0:
                 from .spec_sampling import accept_resample
                 rows = logits[cu_list[i]:cu_list[i + 1]]
                 p_rows = torch.softmax(rows.float() / temp, dim=-1)
+                # d may be empty (capacity/budget-capped seq riding a
+                # spec step): accept_resample degenerates to one sample
+                # from p_0 — NOT the greedy path, which would silently
+                # freeze this sequence's temperature to 0
                 toks = accept_resample(p_rows, d,
                                        (qs or {}).get(seq.seq_id),
                                        self._spec_gen)

@@ -151,3 +151,15 @@ def test_prompt_lookup_with_sampled_sequences():
             break
     assert f and len(f.output_ids) == 20
     assert eng.metrics["spec_steps"] > 0
+
+
+def test_sampled_seq_with_empty_draft_still_samples():
+    """A sampled sequence that gets no draft during a spec step must
+    still draw from its temperature distribution (not argmax)."""
+    # accept_resample with k=0: exactly one sample from p_0
+    gen = torch.Generator().manual_seed(2)
+    p = torch.tensor([[0.5, 0.5, 0.0, 0.0]])
+    seen = set()
+    for _ in range(64):
+        seen.add(accept_resample(p, [], None, gen)[0])
+    assert seen == {0, 1}  # both modes appear: it samples, not argmax
