@@ -37,6 +37,8 @@ class Config:
         "db_path": ("AGENTFIELD_DATABASE_URL", str),
         "sync_timeout": ("AGENTFIELD_SYNC_TIMEOUT", float),
         "agent_timeout": ("AGENTFIELD_AGENT_TIMEOUT", float),
+        "step_retries": ("AGENTFIELD_STEP_RETRIES", int),
+        "step_retry_backoff_s": ("AGENTFIELD_STEP_RETRY_BACKOFF", float),
         "async_workers": ("AGENTFIELD_EXEC_ASYNC_WORKERS", int),
         "async_queue_capacity": ("AGENTFIELD_EXEC_ASYNC_QUEUE_CAPACITY", int),
         "did_enabled": ("AGENTFIELD_DID_ENABLED", lambda v: v not in ("0", "false")),
@@ -62,6 +64,9 @@ class Config:
         self.keystore_path = kw.get("keystore_path")
         self.sync_timeout = kw.get("sync_timeout", 90.0)
         self.agent_timeout = kw.get("agent_timeout", 90.0)
+        # transient-failure step retries (502/503/504 + connect errors)
+        self.step_retries = kw.get("step_retries", 2)
+        self.step_retry_backoff_s = kw.get("step_retry_backoff_s", 0.2)
         self.async_workers = kw.get("async_workers", 8)
         self.async_queue_capacity = kw.get("async_queue_capacity", 1024)
         self.did_enabled = kw.get("did_enabled", True)
@@ -228,21 +233,34 @@ class ControlPlane:
             # land on THIS worker (it holds the sync waiter's future)
             headers["X-AgentField-Callback"] = self.cfg.public_url
         payload = dict(rec.get("input") or {})
-        try:
-            async with self.client.post(url, json=payload,
-                                        headers=headers) as resp:
-                if resp.status == 202:
-                    return 202, None, None
-                if resp.status == 200:
-                    try:
-                        return 200, await resp.json(content_type=None), None
-                    except ValueError:
-                        return 200, {"raw": await resp.text()}, None
-                text = await resp.text()
-                return resp.status, None, \
-                    f"agent HTTP {resp.status}: {text[:300]}"
-        except Exception as e:
-            return 0, None, f"agent unreachable: {e}"
+        # bounded step retry on TRANSIENT failures (connect errors, 502/
+        # 503/504 — an agent restarting under the process manager): the
+        # reference's worker retries steps and counts them
+        # (agentfield_step_retries_total).  4xx and agent-side exceptions
+        # (500) are NOT transient and fail immediately.
+        last_err = "agent unreachable"
+        for attempt in range(self.cfg.step_retries + 1):
+            if attempt:
+                self.metrics.step_retries.inc()
+                await asyncio.sleep(self.cfg.step_retry_backoff_s * attempt)
+            try:
+                async with self.client.post(url, json=payload,
+                                            headers=headers) as resp:
+                    if resp.status == 202:
+                        return 202, None, None
+                    if resp.status == 200:
+                        try:
+                            return 200, \
+                                await resp.json(content_type=None), None
+                        except ValueError:
+                            return 200, {"raw": await resp.text()}, None
+                    text = await resp.text()
+                    last_err = f"agent HTTP {resp.status}: {text[:300]}"
+                    if resp.status not in (502, 503, 504):
+                        return resp.status, None, last_err
+            except Exception as e:
+                last_err = f"agent unreachable: {e}"
+        return 0, None, last_err
 
     def complete_execution(self, execution_id: str, status: str, result=None,
                            error: str | None = None,

@@ -890,3 +890,50 @@ def test_webhook_stale_inflight_reclaimed():
     st.webhook_attempted("e1", True, 200, None, 0.0, 5)
     assert st.get_webhook("e1")["status"] == "delivered"
     st.close()
+
+
+def test_step_retry_on_transient_failure():
+    """A 503 from the agent (restarting under the process manager) is
+    retried up to step_retries times with backoff and counted in
+    agentfield_step_retries_total; a 500 (agent-side bug) is NOT."""
+    from agentfield_amd.controlplane import ControlPlane, create_app
+    from agentfield_amd.controlplane.server import Config as CPConfig
+
+    cp = ControlPlane(CPConfig(background_services=False, sync_timeout=20.0,
+                               step_retries=3, step_retry_backoff_s=0.01))
+    srv = AppServer(create_app(cp)).start().wait_healthy()
+    flaky_hits = {"n": 0}
+    buggy_hits = {"n": 0}
+    agent = FastAPI()
+
+    from fastapi.responses import Response as FResponse
+
+    @agent.post("/reasoners/flaky")
+    async def flaky(req: Request):
+        flaky_hits["n"] += 1
+        if flaky_hits["n"] < 3:
+            return FResponse(status_code=503)
+        return {"result": {"ok": True}}
+
+    @agent.post("/reasoners/buggy")
+    async def buggy(req: Request):
+        buggy_hits["n"] += 1
+        return FResponse(status_code=500)
+
+    asrv = AppServer(agent).start()
+    httpx.post(srv.base_url + "/api/v1/nodes/register", json={
+        "node_id": "fl", "base_url": asrv.base_url,
+        "reasoners": [{"id": "flaky"}, {"id": "buggy"}]}, timeout=10.0)
+
+    r = httpx.post(srv.base_url + "/api/v1/execute/fl.flaky",
+                   json={"input": {}}, timeout=30.0)
+    assert r.status_code == 200, r.text
+    assert flaky_hits["n"] == 3  # two transient 503s, then success
+
+    before = buggy_hits["n"]
+    r = httpx.post(srv.base_url + "/api/v1/execute/fl.buggy",
+                   json={"input": {}}, timeout=30.0)
+    body = r.json()
+    assert r.status_code >= 400 or body.get("status") == "failed", body
+    assert buggy_hits["n"] == before + 1  # 500 is terminal, no retry
+    srv.stop()
