@@ -572,6 +572,10 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
             meta = dict(node.get("metadata") or {})
             meta["mcp_servers"] = body["mcp_servers"]
             cp.storage.set_node_metadata(node_id, meta)
+        if isinstance(body.get("engine"), dict):
+            # agents hosting an in-process engine report it: feeds the
+            # agentfield_engine_* gauges on /metrics, per node
+            cp.metrics.record_engine_heartbeat(node_id, body["engine"])
         return {"status": "ok"}
 
     @app.post("/api/v1/nodes/{node_id}/status")

@@ -44,15 +44,31 @@ class Metrics:
                                        "execution duration", registry=self.registry)
         self.step_retries = Counter("agentfield_step_retries_total",
                                     "execution retries", registry=self.registry)
-        # engine-side metrics (new surface: GPU serving)
+        # engine-side metrics, aggregated from agent heartbeats (each
+        # agent with an in-process EngineRunner reports its engine's
+        # counters; labeled by node so DP fleets don't clobber each other)
         self.engine_tokens = Counter("agentfield_engine_tokens_total",
-                                     "tokens generated", ["kind"],
+                                     "tokens generated", ["node", "kind"],
                                      registry=self.registry)
         self.engine_batch = Gauge("agentfield_engine_batch_occupancy",
                                   "sequences in the running batch",
-                                  registry=self.registry)
+                                  ["node"], registry=self.registry)
         self.engine_kv_pages = Gauge("agentfield_engine_kv_free_pages",
-                                     "free KV pages", registry=self.registry)
+                                     "free KV pages", ["node"],
+                                     registry=self.registry)
+
+    def record_engine_heartbeat(self, node_id: str, eng: dict) -> None:
+        """Fold an agent heartbeat's engine snapshot into the gauges;
+        token counters advance by the reported deltas."""
+        for kind in ("prefill", "decode"):
+            d = eng.get(f"{kind}_tokens_delta")
+            if d:
+                self.engine_tokens.labels(node=node_id, kind=kind).inc(d)
+        if "running" in eng:
+            self.engine_batch.labels(node=node_id).set(eng["running"])
+        if "kv_free_pages" in eng:
+            self.engine_kv_pages.labels(node=node_id).set(
+                eng["kv_free_pages"])
 
     def render(self) -> bytes:
         return generate_latest(self.registry)

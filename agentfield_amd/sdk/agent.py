@@ -454,11 +454,40 @@ class Agent(FastAPI):
                 # enhanced heartbeat: per-server MCP health rides along
                 # (reference P10 agent_field_handler.py:227-264)
                 payload["mcp_servers"] = mcp.status()
+            eng = self._engine_heartbeat()
+            if eng:
+                payload["engine"] = eng
             ok = self.client.heartbeat(self.node_id, payload)
             if not ok:
                 self.register()  # resilient re-register (P10)
             else:
                 self._drain_actions()
+
+    def _engine_heartbeat(self) -> dict | None:
+        """Snapshot of the in-process engine (if this agent hosts one)
+        for the control plane's agentfield_engine_* metrics: token
+        counters as DELTAS since the previous heartbeat."""
+        from .ai import _runners
+        eng = None
+        for r in _runners.values():
+            eng = getattr(r, "engine", None)
+            if eng is not None:
+                break
+        if eng is None:
+            return None
+        m = eng.metrics
+        prev = getattr(self, "_hb_engine_prev", {})
+        out = {
+            "running": eng.sched.num_running(),
+            "kv_free_pages": eng.sched.alloc.num_free,
+            "prefill_tokens_delta":
+                m["prefill_tokens"] - prev.get("prefill_tokens", 0),
+            "decode_tokens_delta":
+                m["decode_tokens"] - prev.get("decode_tokens", 0),
+        }
+        self._hb_engine_prev = {"prefill_tokens": m["prefill_tokens"],
+                                "decode_tokens": m["decode_tokens"]}
+        return out
 
     def _drain_actions(self):
         """Claim and execute pending control-plane lifecycle actions

@@ -937,3 +937,30 @@ def test_step_retry_on_transient_failure():
     assert r.status_code >= 400 or body.get("status") == "failed", body
     assert buggy_hits["n"] == before + 1  # 500 is terminal, no retry
     srv.stop()
+
+
+def test_engine_metrics_ride_heartbeats():
+    """Agents hosting an in-process engine report it in heartbeats; the
+    control plane folds the snapshot into agentfield_engine_* metrics
+    (token counters advance by deltas, gauges labeled per node)."""
+    from agentfield_amd.controlplane import ControlPlane, create_app
+    from agentfield_amd.controlplane.server import Config as CPConfig
+
+    cp = ControlPlane(CPConfig(background_services=False))
+    srv = AppServer(create_app(cp)).start().wait_healthy()
+    httpx.post(srv.base_url + "/api/v1/nodes/register", json={
+        "node_id": "gpu0", "base_url": "http://127.0.0.1:1",
+        "reasoners": []}, timeout=10.0)
+    for delta in (100, 50):
+        r = httpx.post(srv.base_url + "/api/v1/nodes/gpu0/heartbeat", json={
+            "status": "active",
+            "engine": {"running": 7, "kv_free_pages": 1234,
+                       "prefill_tokens_delta": delta * 8,
+                       "decode_tokens_delta": delta}}, timeout=10.0)
+        assert r.status_code == 200
+    text = httpx.get(srv.base_url + "/metrics", timeout=10.0).text
+    assert 'agentfield_engine_tokens_total{kind="decode",node="gpu0"} 150' \
+        in text
+    assert 'agentfield_engine_batch_occupancy{node="gpu0"} 7' in text
+    assert 'agentfield_engine_kv_free_pages{node="gpu0"} 1234' in text
+    srv.stop()
