@@ -264,8 +264,6 @@ class LLMEngine:
                         all(s.sampling.top_k <= 0 and
                             s.sampling.top_p >= 1.0
                             and not s.sampling.json_mode
-                            and (s.sampling.temperature == 0.0
-                                 or s.sampling.logprobs == 0)
                             for s in batch.seqs):
                     if self.draft is not None:
                         drafts, qs = self._draft_model_propose(batch.seqs)
@@ -662,6 +660,7 @@ class LLMEngine:
         greedy_t = logits.argmax(dim=-1).to(torch.int64)
         greedy = greedy_t.cpu().tolist()
         tok_lists, lp_rows, lp_seqs = [], [], []
+        sp_rows, sp_chosen, sp_seqs = [], [], []  # sampled-branch logprobs
         for i, seq in enumerate(seqs):
             t = greedy[cu_list[i]:cu_list[i + 1]]
             d = drafts.get(seq.seq_id) or []
@@ -682,6 +681,11 @@ class LLMEngine:
                     seq.draft_len = seq.num_tokens + len(toks) - 1
                 self.metrics["spec_accepted"] += len(toks) - 1
                 self.metrics["decode_tokens"] += len(toks)
+                if seq.sampling.logprobs > 0:
+                    for j, tk in enumerate(toks):
+                        sp_rows.append(cu_list[i] + j)
+                        sp_chosen.append(tk)
+                        sp_seqs.append(seq)
                 continue
             a = 0
             while a < len(d) and t[a] == d[a]:
@@ -702,6 +706,12 @@ class LLMEngine:
             rows = torch.tensor(lp_rows, dtype=torch.int64,
                                 device=logits.device)
             self._attach_row_logprobs(lp_seqs, logits[rows], greedy_t[rows])
+        if sp_rows:
+            rows = torch.tensor(sp_rows, dtype=torch.int64,
+                                device=logits.device)
+            chosen = torch.tensor(sp_chosen, dtype=torch.int64,
+                                  device=logits.device)
+            self._attach_row_logprobs(sp_seqs, logits[rows], chosen)
         return self._bookkeep_multi(seqs, tok_lists)
 
     def _bookkeep_multi(self, seqs, tok_lists):

@@ -163,3 +163,27 @@ def test_sampled_seq_with_empty_draft_still_samples():
     for _ in range(64):
         seen.add(accept_resample(p, [], None, gen)[0])
     assert seen == {0, 1}  # both modes appear: it samples, not argmax
+
+
+def test_logprobs_under_sampled_speculation():
+    """Sampled sequences requesting logprobs no longer disable the spec
+    path: every emitted token carries its raw-logit logprob + top-k,
+    aligned with output_ids (eos truncation included)."""
+    eng = LLMEngine(CONFIGS["tiny"], device="cpu", dtype=torch.float32,
+                    page_size=4, num_pages=256, max_num_seqs=4,
+                    enable_graphs=False, seed=6,
+                    spec_draft=CONFIGS["tiny"], spec_draft_k=3)
+    rid = eng.add_request([5, 9, 2, 44] * 5,
+                          SamplingParams(max_tokens=14, temperature=0.8,
+                                         ignore_eos=True, logprobs=3))
+    for _ in range(400):
+        eng.step()
+        f = eng.get_finished(rid)
+        if f:
+            break
+    assert f and len(f.output_ids) == 14
+    assert eng.metrics["spec_steps"] > 0
+    assert len(f.logprobs) == 14
+    for e, tok in zip(f.logprobs, f.output_ids):
+        assert e["logprob"] <= 0.0
+        assert len(e["top"]) == 3
