@@ -574,8 +574,14 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
             cp.storage.set_node_metadata(node_id, meta)
         if isinstance(body.get("engine"), dict):
             # agents hosting an in-process engine report it: feeds the
-            # agentfield_engine_* gauges on /metrics, per node
+            # agentfield_engine_* gauges on /metrics, per node, and the
+            # dashboard's engines panel (latest snapshot in node meta)
             cp.metrics.record_engine_heartbeat(node_id, body["engine"])
+            meta = dict(node.get("metadata") or {})
+            meta["engine"] = {k: body["engine"][k]
+                              for k in ("running", "kv_free_pages")
+                              if k in body["engine"]}
+            cp.storage.set_node_metadata(node_id, meta)
         return {"status": "ok"}
 
     @app.post("/api/v1/nodes/{node_id}/status")
@@ -1009,10 +1015,14 @@ def create_app(cp: ControlPlane | None = None, **cfg_kw) -> FastAPI:
         by_status: dict[str, int] = {}
         for e in recents:
             by_status[e["status"]] = by_status.get(e["status"], 0) + 1
+        engines = {n["id"]: (n.get("metadata") or {}).get("engine")
+                   for n in nodes
+                   if (n.get("metadata") or {}).get("engine")}
         return {
             "nodes": {"total": len(nodes),
                       "active": sum(1 for n in nodes if n["status"] == "active")},
             "executions": {"recent": len(recents), "by_status": by_status},
+            "engines": engines,  # latest heartbeat snapshot per GPU node
             "uptime_s": time.time() - cp.started_at,
         }
 

@@ -963,4 +963,9 @@ def test_engine_metrics_ride_heartbeats():
         in text
     assert 'agentfield_engine_batch_occupancy{node="gpu0"} 7' in text
     assert 'agentfield_engine_kv_free_pages{node="gpu0"} 1234' in text
+    # dashboard surfaces the latest snapshot per node
+    summ = httpx.get(srv.base_url + "/api/ui/v1/dashboard/summary",
+                     timeout=10.0).json()
+    assert summ["engines"]["gpu0"] == {"running": 7,
+                                       "kv_free_pages": 1234}
     srv.stop()
