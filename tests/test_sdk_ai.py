@@ -233,3 +233,35 @@ def test_ai_tool_loop_with_skills():
             pass
     finally:
         runner.shutdown()
+
+
+def test_ai_tool_loop_skill_failure_feeds_back():
+    """A skill that raises becomes an {"error": ...} observation in the
+    transcript (the model sees the failure); the loop still terminates
+    with final_answer."""
+    import torch
+
+    from agentfield_amd.engine import LLMEngine
+    from agentfield_amd.models import CONFIGS
+    from agentfield_amd.sdk import Agent
+    from agentfield_amd.sdk.ai import ByteTokenizer, EngineRunner, set_runner
+
+    eng = LLMEngine(CONFIGS["tiny"], device="cpu", dtype=torch.float32,
+                    page_size=4, num_pages=256, max_num_seqs=4,
+                    enable_graphs=False, seed=12)
+    runner = EngineRunner(eng, ByteTokenizer(CONFIGS["tiny"].vocab_size))
+    set_runner("tiny", runner)
+    try:
+        app = Agent("failtool", auto_register=False)
+        attempts = {"n": 0}
+
+        @app.skill()
+        def explode(x: int = 0):
+            attempts["n"] += 1
+            raise RuntimeError("boom")
+
+        out = app.ai("go", tools=["explode"], max_tool_rounds=3,
+                     model="tiny", max_tokens=48, temperature=0.9)
+        assert isinstance(out, str)  # loop terminated despite failures
+    finally:
+        runner.shutdown()
