@@ -633,3 +633,34 @@ def test_ai_tool_loop_via_remote_engines(replicas, monkeypatch):
     for c in calls:  # every call that happened was schema-well-formed
         assert isinstance(c, str)
     ai_mod._runners.pop("tiny-rfleet", None)
+
+
+def test_engine_server_with_spec_draft():
+    """A replica built with a draft-model speculator serves sampled and
+    greedy requests; spec metrics surface on /v1/stats."""
+    import httpx
+    import torch
+
+    from agentfield_amd.engine import LLMEngine
+    from agentfield_amd.models import CONFIGS
+    from agentfield_amd.sdk.ai import ByteTokenizer, EngineRunner
+    from agentfield_amd.serving.engine_server import create_engine_app
+
+    eng = LLMEngine(CONFIGS["tiny"], device="cpu", dtype=torch.float32,
+                    page_size=4, num_pages=256, max_num_seqs=4,
+                    enable_graphs=False, seed=2,
+                    spec_draft=CONFIGS["tiny"], spec_draft_k=3)
+    runner = EngineRunner(eng, ByteTokenizer(CONFIGS["tiny"].vocab_size))
+    srv = AppServer(create_engine_app(runner, "tiny")).start()
+    try:
+        for temp in (0.0, 0.9):
+            r = httpx.post(srv.base_url + "/v1/generate", json={
+                "prompt_ids": [7, 9, 2, 44] * 5, "max_tokens": 12,
+                "temperature": temp, "ignore_eos": True},
+                timeout=60.0).json()
+            assert len(r["output_ids"]) == 12
+        stats = httpx.get(srv.base_url + "/v1/stats", timeout=10.0).json()
+        assert stats["spec_steps"] > 0
+    finally:
+        srv.stop()
+        runner.shutdown()
