@@ -95,6 +95,50 @@ def build_ep_model(cfg, device="cpu", dtype=torch.float32,
     return model
 
 
+class EPEngineGroup:
+    """Drives one LLMEngine per rank with an expert-parallel MoE model:
+    attention/embeddings replicated, experts sharded, tokens exchanged
+    inside each MoE forward.  Request sync, stepping and the no-pickle
+    wire codec are TPEngineGroup's (identical scheduling + seeds on all
+    ranks -> identical outputs); only the model construction differs —
+    the full config is kept (no head/FFN sharding) and the per-layer
+    collective is the MoE all-to-all instead of TP all-reduces."""
+
+    def __init__(self, cfg, device, dtype=torch.bfloat16, group=None,
+                 base_seed: int = 0, **engine_kw):
+        from ..engine import LLMEngine
+        from .tp import TPEngineGroup
+        self.group = group
+        self.rank = dist.get_rank(group) if dist.is_initialized() else 0
+        self.tp = dist.get_world_size(group) if dist.is_initialized() else 1
+        if torch.device(device).type != "cuda":
+            engine_kw.setdefault("enable_graphs", False)
+        model = build_ep_model(cfg, device=device, dtype=dtype,
+                               base_seed=base_seed, group=group)
+        self.engine = LLMEngine(cfg, device=device, dtype=dtype,
+                                model=model, tp_group=group, **engine_kw)
+
+    def _bcast_device(self):
+        from .tp import TPEngineGroup
+        return TPEngineGroup._bcast_device(self)
+
+    def broadcast_and_submit(self, requests=None):
+        from .tp import TPEngineGroup
+        return TPEngineGroup.broadcast_and_submit(self, requests)
+
+    def submit(self, prompt_ids, sampling):
+        return self.engine.add_request(prompt_ids, sampling)
+
+    def step(self):
+        return self.engine.step()
+
+    def has_work(self):
+        return self.engine.has_work()
+
+    def get_finished(self, rid):
+        return self.engine.get_finished(rid)
+
+
 class EPMoE(nn.Module):
     """Expert-parallel Mixtral-style sparse FFN.  Construct from a full
     MoEMLP's weights via `shard_from` (each rank keeps its expert slice
@@ -129,7 +173,10 @@ class EPMoE(nn.Module):
             ep.down.copy_(moe.down[lo:lo + ep.e_local])
         return ep
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, static: bool = False) -> torch.Tensor:
+        # `static` is the single-rank MoE's graph-capture mode; the EP
+        # exchange has data-dependent splits, so EP decode runs the same
+        # exact dispatch either way (static EP capacity: docs/ROADMAP.md)
         T, H = x.shape
         dev = x.device
         # 1) replicated fp32 routing (identical to MoEMLP.forward)

@@ -149,3 +149,52 @@ def test_build_ep_model_single_rank_fallback():
                            dtype=torch.float32).init_random(7)
     x = torch.randn(5, CFG.hidden_size)
     assert torch.allclose(m.layers[0].mlp(x), ref.layers[0].mlp(x))
+
+
+def _ep_engine_case(rank, world):
+    """EP serving end-to-end on gloo: an EPEngineGroup (experts sharded,
+    requests broadcast rank0 -> all) must emit exactly the single-rank
+    MoE engine's greedy tokens."""
+    from agentfield_amd.engine import LLMEngine, SamplingParams
+    from agentfield_amd.parallel.ep import EPEngineGroup
+
+    kw = dict(device="cpu", dtype=torch.float32, page_size=4,
+              num_pages=128, max_num_seqs=4, enable_graphs=False, seed=3)
+    grp = EPEngineGroup(CFG, base_seed=7, **kw)
+
+    prompts = [list(range(1, 14)), [7, 3, 9, 1] * 3]
+    sp = SamplingParams(max_tokens=8, ignore_eos=True)
+    if rank == 0:
+        rids = grp.broadcast_and_submit([(p, sp) for p in prompts])
+    else:
+        rids = grp.broadcast_and_submit(None)
+    outs = {}
+    for _ in range(200):
+        grp.step()
+        for r in rids:
+            if r not in outs:
+                f = grp.get_finished(r)
+                if f:
+                    outs[r] = f.output_ids
+        if len(outs) == len(rids):
+            break
+        if not grp.has_work():
+            if rank == 0:
+                grp.broadcast_and_submit([])
+            else:
+                grp.broadcast_and_submit(None)
+    assert len(outs) == len(rids), outs
+
+    # single-rank reference with identical weights (same base_seed)
+    from agentfield_amd.models.llama import LlamaForCausalLM
+    ref_model = LlamaForCausalLM(CFG, device="cpu",
+                                 dtype=torch.float32).init_random(7)
+    ref = LLMEngine(CFG, model=ref_model, **kw)
+    want = ref.generate(prompts, sp)
+    got = [outs[r] for r in rids]
+    assert got == want, (got, want)
+    return True
+
+
+def test_ep_engine_group_matches_single_rank():
+    assert _spawn("_ep_engine_case")
