@@ -187,3 +187,21 @@ def test_logprobs_under_sampled_speculation():
     for e, tok in zip(f.logprobs, f.output_ids):
         assert e["logprob"] <= 0.0
         assert len(e["top"]) == 3
+
+
+def test_spec_bench_tool():
+    """tools/spec_bench.py: algorithmic speedup in engine steps, with
+    greedy-exactness asserted inside the tool itself."""
+    import json as _json
+    import subprocess
+    import sys
+    from pathlib import Path
+    root = Path(__file__).resolve().parent.parent
+    r = subprocess.run(
+        [sys.executable, str(root / "tools" / "spec_bench.py"),
+         "--gen", "24", "--prompts", "4"],
+        capture_output=True, text=True, timeout=300, cwd=root)
+    assert r.returncode == 0, r.stderr[-1500:]
+    out = _json.loads(r.stdout.strip().splitlines()[-1])
+    assert out["lookup_speedup"] > 1.15
+    assert out["ceiling_speedup"] > 1.15
