@@ -174,9 +174,8 @@ class EPMoE(nn.Module):
         return ep
 
     def forward(self, x: torch.Tensor, static: bool = False) -> torch.Tensor:
-        # `static` is the single-rank MoE's graph-capture mode; the EP
-        # exchange has data-dependent splits, so EP decode runs the same
-        # exact dispatch either way (static EP capacity: docs/ROADMAP.md)
+        if static:
+            return self._forward_static(x)
         T, H = x.shape
         dev = x.device
         # 1) replicated fp32 routing (identical to MoEMLP.forward)
@@ -222,4 +221,59 @@ class EPMoE(nn.Module):
         back = _exchange(ry.to(x.dtype), in_splits, out_splits, self.group)
         back = back[inv].reshape(T, self.top_k, H)
         out = (back.float() * topv.unsqueeze(-1)).sum(dim=1)
+        return out.to(x.dtype)
+
+    def _forward_static(self, x: torch.Tensor) -> torch.Tensor:
+        """Static-capacity exchange for decode: every rank sends a FIXED
+        [world, T*top_k, H] block (slots padded with expert-id -1), so
+        every tensor shape is data-INDEPENDENT — the property hipGraph
+        capture needs.  Padding-only (capacity = T*top_k per
+        destination), so unlike trained capacity-factor routing nothing
+        is ever dropped and the result is EXACTLY the dynamic
+        dispatch's.  Costs world x the exchange volume of the exact
+        path; decode batches are small, so the static shape is the
+        better trade (same reasoning as the single-rank static MoE)."""
+        T, H = x.shape
+        dev = x.device
+        cap = T * self.top_k
+        probs = torch.softmax(x.float() @ self.router.float().t(), dim=-1)
+        topv, topi = probs.topk(self.top_k, dim=-1)
+        topv = topv / topv.sum(dim=-1, keepdim=True)
+
+        flat_e = topi.reshape(-1)                       # [T*k]
+        owner = flat_e // self.e_local
+        # slot s of destination block d holds pair index s if owned by d
+        send_x = torch.zeros(self.world * cap, H, dtype=x.dtype, device=dev)
+        send_e = torch.full((self.world * cap,), -1, dtype=torch.int64,
+                            device=dev)
+        tok_of_pair = (torch.arange(T, device=dev)
+                       .repeat_interleave(self.top_k))
+        idx = owner * cap + torch.arange(cap, device=dev)
+        send_x[idx] = x[tok_of_pair]
+        send_e[idx] = flat_e
+
+        splits = [cap] * self.world
+        rx = _exchange(send_x, splits, splits, self.group)
+        re = _exchange(send_e.unsqueeze(1), splits, splits,
+                       self.group).squeeze(1)
+
+        ry = torch.zeros(self.world * cap, H, dtype=torch.float32,
+                         device=dev)
+        lo = self.rank * self.e_local
+        for le in range(self.e_local):
+            sel = (re == lo + le).nonzero(as_tuple=True)[0]
+            if sel.numel() == 0:
+                continue
+            ye = ops.linear(ops.linear(rx[sel], self.gate_up[le],
+                                       silu_fuse=True), self.down[le])
+            ry[sel] = ye.float()
+
+        back = _exchange(ry.to(x.dtype), splits, splits, self.group)
+        # this rank's block d came back at block d; slot s = pair s
+        mine = back[self.rank * cap:(self.rank + 1) * cap]             if False else None
+        # reassemble: pair s was sent in block owner[s], slot s
+        pair_rows = back.reshape(self.world, cap, H)[
+            owner, torch.arange(cap, device=dev)]
+        pair_rows = pair_rows.reshape(T, self.top_k, H)
+        out = (pair_rows.float() * topv.unsqueeze(-1)).sum(dim=1)
         return out.to(x.dtype)

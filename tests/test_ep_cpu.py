@@ -70,6 +70,32 @@ def _ep_case(rank, world):
     return outs
 
 
+def _ep_static_case(rank, world):
+    """Static-capacity exchange (every shape data-independent) must be
+    EXACTLY the dynamic dispatch — padding-only, nothing dropped."""
+    from agentfield_amd.models.llama import MoEMLP
+    from agentfield_amd.parallel.ep import EPMoE
+
+    torch.manual_seed(3)
+    moe = MoEMLP(CFG).float()
+    for p in moe.parameters():
+        torch.nn.init.normal_(p, std=0.1)
+    ep = EPMoE.shard_from(moe)
+    torch.manual_seed(23)
+    for T in (1, 5, 16):
+        x = torch.randn(T, CFG.hidden_size) * 0.5
+        want = moe(x)
+        dyn = ep(x)
+        stat = ep(x, static=True)
+        assert torch.allclose(dyn, want, atol=1e-4)
+        assert torch.allclose(stat, want, atol=1e-4),             (T, (stat - want).abs().max().item())
+    return True
+
+
+def test_ep_static_capacity_exact():
+    assert _spawn("_ep_static_case")
+
+
 def test_ep_matches_single_rank():
     errs = _spawn("_ep_case")
     assert all(e < 1e-4 for e in errs)
