@@ -20,6 +20,19 @@ from pathlib import Path
 
 import typer
 
+def _die_with_parent():
+    """preexec_fn: the child receives SIGTERM if its parent dies — no
+    orphaned control-plane workers / dev children when the launcher is
+    killed uncleanly (measured: an orphaned worker squatting on the
+    bench's derived ports 404'd every later run)."""
+    try:
+        import ctypes
+        libc = ctypes.CDLL("libc.so.6", use_errno=True)
+        libc.prctl(1, 15)  # PR_SET_PDEATHSIG = 1, SIGTERM = 15
+    except Exception:
+        pass
+
+
 app = typer.Typer(add_completion=False, no_args_is_help=True)
 
 DEFAULT_URL = os.environ.get("AGENTFIELD_URL", "http://127.0.0.1:8520")
@@ -65,7 +78,8 @@ def server(host: str = "0.0.0.0", port: int = 8520,
                 cmd += ["--config", config]
             if no_did:
                 cmd += ["--no-did"]
-            procs.append(subprocess.Popen(cmd, env=env))
+            procs.append(subprocess.Popen(cmd, env=env,
+                                          preexec_fn=_die_with_parent))
         typer.echo(f"agentfield-amd control plane x{workers} on "
                    f"{host}:{port}-{port + workers - 1} (db={db})")
         import signal as _signal
@@ -293,7 +307,7 @@ def dev(path: str, port: int = 8600, host: str = "127.0.0.1",
             [sys.executable, "-m", "agentfield_amd", "run", str(p),
              "--port", str(port), "--host", host,
              "--agentfield-url", agentfield_url],
-            env={**os.environ})
+            env={**os.environ}, preexec_fn=_die_with_parent)
 
     typer.echo(f"dev: watching {watch_root} (restart on change, ctrl-c to"
                " stop)")

@@ -25,6 +25,18 @@ import subprocess
 import sys
 import tempfile
 import threading
+
+
+def _die_with_parent():
+    """preexec_fn for subprocesses (control plane, engine replica): take
+    SIGTERM when the bench rank dies, so an aborted run can never leave
+    a stale control plane squatting on the derived ports (an orphan
+    there makes every subsequent run on the same ports fail with 404s)."""
+    try:
+        import ctypes
+        ctypes.CDLL("libc.so.6").prctl(1, 15)  # PR_SET_PDEATHSIG, SIGTERM
+    except Exception:
+        pass
 import time
 from pathlib import Path
 
@@ -152,9 +164,11 @@ def run_rest(args, world: int, rank: int, dist, device: str, cfg) -> None:
             eng_argv.append("--no-graphs")
         eng_proc = subprocess.Popen(
             eng_argv, env={**os.environ, "PYTHONPATH": str(root)}, cwd=root,
-            stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+            stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,
+            preexec_fn=_die_with_parent)
         _wait_http(f"{stats_url}/v1/health", timeout=600.0)
         os.environ["AGENTFIELD_ENGINE_URLS"] = stats_url
+        runner = None
     else:
         dtype = (torch.bfloat16 if device.startswith("cuda")
                  else torch.float32)
@@ -181,7 +195,8 @@ def run_rest(args, world: int, rank: int, dist, device: str, cfg) -> None:
              "--db", f"{tmp}/af.db", "--data-dir", tmp,
              "--workers", str(cp_workers)],
             env={**os.environ, "PYTHONPATH": str(root)}, cwd=root,
-            stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+            stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,
+            preexec_fn=_die_with_parent)
         for i in range(cp_workers):
             _wait_http(f"http://127.0.0.1:{cp_port + i}/api/v1/health")
     if dist is not None:
@@ -233,7 +248,26 @@ def run_rest(args, world: int, rank: int, dist, device: str, cfg) -> None:
     client = subprocess.Popen(
         [sys.executable, str(root / "tools" / "rest_load.py")],
         stdin=subprocess.PIPE, stdout=subprocess.PIPE, text=True,
-        env={**os.environ, "PYTHONPATH": str(root)}, cwd=root)
+        env={**os.environ, "PYTHONPATH": str(root)}, cwd=root,
+        preexec_fn=_die_with_parent)
+    try:
+        _drive_and_report(args, world, device, cfg, dist, client,
+                          urls, targets, cp_workers, stats_url,
+                          runner if stats_url is None else None)
+    finally:
+        client.terminate()
+        for prc in (eng_proc, cp_proc):
+            if prc is not None:
+                prc.terminate()
+                try:
+                    prc.wait(timeout=10)
+                except subprocess.TimeoutExpired:
+                    prc.kill()
+
+
+def _drive_and_report(args, world, device, cfg, dist, client, urls,
+                      targets, cp_workers, stats_url, runner):
+    import httpx
     spec = {"calls": args.calls * world, "targets": targets, "urls": urls,
             "prompt_chars": max(8, args.prompt_len - 25),
             "gen": args.gen_len, "concurrency": args.calls * world}
@@ -308,16 +342,8 @@ def run_rest(args, world: int, rank: int, dist, device: str, cfg) -> None:
     }
     print(json.dumps(out))
     client.stdin.close()
-    client.terminate()
     if dist is not None:
         dist.barrier()  # teardown gate
-    for prc in (eng_proc, cp_proc):
-        if prc is not None:
-            prc.terminate()
-            try:
-                prc.wait(timeout=10)
-            except subprocess.TimeoutExpired:
-                prc.kill()
 
 
 def run_step(eng: LLMEngine, rank: int, step: int, args) -> list[float]:
